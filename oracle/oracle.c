@@ -1,0 +1,389 @@
+/* oracle.c — CPU restatement of the hot-path operator semantics.
+ * See oracle.h for scope and citations.  C99 + OpenMP, built by oracle/Makefile
+ * into liboracle.so (test infrastructure only). */
+#include "oracle.h"
+#include "../presto_amd/csrc/fixed128.h"
+#include <stdlib.h>
+#include <string.h>
+
+/* ---------------- fixed-tree f64 schedule (must match GPU kernels) ----
+ * Virtual lanes: VL = NBLOCKS*NTHREADS.  Lane v owns row pairs
+ * {2v, 2v+1} + k*2*VL.  Within a lane: sequential accumulation in row order.
+ * Reduction: within block, pairwise stride-halving over the 256 lane
+ * partials (for s=128..1: p[i] += p[i+s]); then the same stride-halving over
+ * the NBLOCKS block partials.  Documented in DESIGN.md §determinism. */
+#define FT_NBLOCKS 4096
+#define FT_NTHREADS 256
+#define FT_VL ((int64_t)FT_NBLOCKS * FT_NTHREADS)
+
+/* ---------------- Q1 ---------------- */
+
+#define Q1_SHIP_MAX 10471 /* DATE '1998-12-01' - 90 days, epoch days */
+#define NKEY 6            /* (A,N,R) x (F,O) */
+
+static inline int q1_key(uint8_t rf, uint8_t ls)
+{
+    /* rf in {A,N,R} -> 0,1,2 ; ls in {F,O} -> 0,1 */
+    int r = rf == 'A' ? 0 : (rf == 'N' ? 1 : 2);
+    return r * 2 + (ls == 'O' ? 1 : 0);
+}
+
+typedef struct {
+    int64_t cnt, qty, base, disc4, disc_c;
+    __int128 charge6;
+    double f[5]; /* fixed-tree f64 partials: qty, base, disc_price, charge,
+                    disc */
+} q1_acc_t;
+
+int32_t oracle_q1(int64_t n, const double* qty, const double* eprice,
+                  const double* disc, const double* tax, const int32_t* sdate,
+                  const uint8_t* rflag, const uint8_t* lstat,
+                  q1_group_t* groups)
+{
+    /* ---- decimal mode: order-independent exact integer sums ---- */
+    q1_acc_t total[NKEY];
+    memset(total, 0, sizeof(total));
+#pragma omp parallel
+    {
+        q1_acc_t loc[NKEY];
+        memset(loc, 0, sizeof(loc));
+#pragma omp for schedule(static)
+        for (int64_t i = 0; i < n; i++) {
+            if (sdate[i] > Q1_SHIP_MAX) continue;
+            int k = q1_key(rflag[i], lstat[i]);
+            /* exact decimal ticks (values are decimal-representable:
+             * qty units, eprice cents, disc/tax hundredths) */
+            int64_t q = (int64_t)(qty[i] + 0.5);
+            int64_t cents = (int64_t)(eprice[i] * 100.0 + 0.5);
+            int64_t d = (int64_t)(disc[i] * 100.0 + 0.5);
+            int64_t t = (int64_t)(tax[i] * 100.0 + 0.5);
+            loc[k].cnt++;
+            loc[k].qty += q;
+            loc[k].base += cents;
+            loc[k].disc4 += cents * (100 - d);
+            loc[k].charge6 += (__int128)(cents * (100 - d)) * (100 + t);
+            loc[k].disc_c += d;
+        }
+#pragma omp critical
+        for (int k = 0; k < NKEY; k++) {
+            total[k].cnt += loc[k].cnt;
+            total[k].qty += loc[k].qty;
+            total[k].base += loc[k].base;
+            total[k].disc4 += loc[k].disc4;
+            total[k].charge6 += loc[k].charge6;
+            total[k].disc_c += loc[k].disc_c;
+        }
+    }
+
+    /* ---- f64 mode: fixed-tree deterministic schedule ---- */
+    static double* blockp = 0; /* FT_NBLOCKS * NKEY * 5 */
+    double* bp = (double*)malloc((size_t)FT_NBLOCKS * NKEY * 5 * sizeof(double));
+#pragma omp parallel for schedule(static)
+    for (int b = 0; b < FT_NBLOCKS; b++) {
+        /* lane partials for this block */
+        double lane[FT_NTHREADS][NKEY][5];
+        memset(lane, 0, sizeof(lane));
+        for (int t = 0; t < FT_NTHREADS; t++) {
+            int64_t v = (int64_t)b * FT_NTHREADS + t;
+            for (int64_t base = 2 * v; base < n; base += 2 * FT_VL) {
+                for (int64_t i = base; i < base + 2 && i < n; i++) {
+                    if (sdate[i] > Q1_SHIP_MAX) continue;
+                    int k = q1_key(rflag[i], lstat[i]);
+                    double dp = eprice[i] * (1.0 - disc[i]);
+                    double ch = dp * (1.0 + tax[i]);
+                    lane[t][k][0] += qty[i];
+                    lane[t][k][1] += eprice[i];
+                    lane[t][k][2] += dp;
+                    lane[t][k][3] += ch;
+                    lane[t][k][4] += disc[i];
+                }
+            }
+        }
+        for (int s = FT_NTHREADS / 2; s >= 1; s >>= 1)
+            for (int t = 0; t < s; t++)
+                for (int k = 0; k < NKEY; k++)
+                    for (int j = 0; j < 5; j++)
+                        lane[t][k][j] += lane[t + s][k][j];
+        memcpy(&bp[(size_t)b * NKEY * 5], &lane[0][0][0],
+               NKEY * 5 * sizeof(double));
+    }
+    /* stride-halving over block partials */
+    for (int s = FT_NBLOCKS / 2; s >= 1; s >>= 1)
+        for (int b = 0; b < s; b++)
+            for (int j = 0; j < NKEY * 5; j++)
+                bp[(size_t)b * NKEY * 5 + j] += bp[(size_t)(b + s) * NKEY * 5 + j];
+    (void)blockp;
+
+    /* emit groups sorted by (returnflag, linestatus) == key order */
+    static const uint8_t RF[3] = {'A', 'N', 'R'}, LS[2] = {'F', 'O'};
+    int32_t ng = 0;
+    for (int k = 0; k < NKEY; k++) {
+        if (total[k].cnt == 0) continue;
+        q1_group_t* g = &groups[ng++];
+        g->returnflag = RF[k / 2];
+        g->linestatus = LS[k % 2];
+        g->count_order = total[k].cnt;
+        g->sum_qty_units = total[k].qty;
+        g->sum_base_cents = total[k].base;
+        g->sum_disc_1e4 = total[k].disc4;
+        g->sum_charge_1e6_hi = (int64_t)(total[k].charge6 >> 64);
+        g->sum_charge_1e6_lo = (uint64_t)total[k].charge6;
+        g->sum_disc_cents = total[k].disc_c;
+        g->f64_sum_qty = bp[k * 5 + 0];
+        g->f64_sum_base = bp[k * 5 + 1];
+        g->f64_sum_disc_price = bp[k * 5 + 2];
+        g->f64_sum_charge = bp[k * 5 + 3];
+        g->f64_sum_disc = bp[k * 5 + 4];
+    }
+    free(bp);
+    return ng;
+}
+
+/* ---------------- primitives ---------------- */
+
+uint64_t oracle_murmur3_finalize(uint64_t h) { return pg_murmur3_finalize(h); }
+uint64_t oracle_bigint_hash(int64_t v) { return pg_bigint_hash(v); }
+int32_t oracle_partition(uint64_t h, int32_t n) { return pg_partition(h, n); }
+
+/* next power of two >= ceil(x / 0.75), min 2 — fastutil arraySize semantics
+ * used by BigintGroupByHash.java:49 / PagesHash.java:67 */
+static int64_t hash_capacity(int64_t expected)
+{
+    int64_t need = (int64_t)(expected / 0.75) + 1;
+    int64_t c = 2;
+    while (c < need) c <<= 1;
+    return c;
+}
+
+int64_t oracle_bigint_group_by(int64_t n, const int64_t* keys,
+                               int32_t* group_ids)
+{
+    /* BigintGroupByHash.java:222-332: open-address linear probe, fill 0.75,
+     * bucket = murmur3_finalize(key) & mask, x2 rehash, dense first-seen
+     * group ids.  Start capacity from a small hint like the operator does
+     * (expected 10_000 in LocalExecutionPlanner); rehash reproduces the
+     * same final assignment regardless, since ids are insertion-ordered. */
+    int64_t cap = hash_capacity(1024);
+    int64_t* values = (int64_t*)malloc(cap * sizeof(int64_t));
+    int32_t* gids = (int32_t*)malloc(cap * sizeof(int32_t));
+    int64_t* by_gid = (int64_t*)malloc(cap * sizeof(int64_t));
+    memset(gids, -1, cap * sizeof(int32_t));
+    int64_t next_gid = 0, max_fill = (int64_t)(cap * 0.75 + 0.9999);
+    for (int64_t i = 0; i < n; i++) {
+        int64_t key = keys[i];
+        int64_t pos = (int64_t)(pg_murmur3_finalize((uint64_t)key) & (cap - 1));
+        int32_t gid;
+        for (;;) {
+            if (gids[pos] == -1) {
+                gid = (int32_t)next_gid++;
+                gids[pos] = gid;
+                values[pos] = key;
+                by_gid[gid] = key;
+                if (next_gid >= max_fill) {
+                    /* rehash x2, reinsert in group-id order
+                     * (BigintGroupByHash.java:271-311) */
+                    int64_t ncap = cap * 2;
+                    int64_t* nv = (int64_t*)malloc(ncap * sizeof(int64_t));
+                    int32_t* ng = (int32_t*)malloc(ncap * sizeof(int32_t));
+                    memset(ng, -1, ncap * sizeof(int32_t));
+                    for (int64_t g = 0; g < next_gid; g++) {
+                        int64_t p = (int64_t)(pg_murmur3_finalize(
+                                        (uint64_t)by_gid[g]) & (ncap - 1));
+                        while (ng[p] != -1) p = (p + 1) & (ncap - 1);
+                        ng[p] = (int32_t)g;
+                        nv[p] = by_gid[g];
+                    }
+                    free(values);
+                    free(gids);
+                    values = nv;
+                    gids = ng;
+                    cap = ncap;
+                    max_fill = (int64_t)(cap * 0.75 + 0.9999);
+                    by_gid = (int64_t*)realloc(by_gid, cap * sizeof(int64_t));
+                }
+                break;
+            }
+            if (values[pos] == key) {
+                gid = gids[pos];
+                break;
+            }
+            pos = (pos + 1) & (cap - 1);
+        }
+        if (group_ids) group_ids[i] = gid;
+    }
+    free(values);
+    free(gids);
+    free(by_gid);
+    return next_gid;
+}
+
+int64_t oracle_join_bigint(int64_t n_build, const int64_t* bkeys,
+                           int64_t n_probe, const int64_t* pkeys,
+                           int64_t* out_probe, int64_t* out_build,
+                           int64_t out_cap)
+{
+    /* PagesHash.java:81-125 build (bucket = murmur3_finalize(bigint_hash),
+     * linear probe) + ArrayPositionLinks head-insert chains
+     * (ArrayPositionLinks.java:25-30), probe PagesHash.getAddressIndex:169. */
+    int64_t cap = hash_capacity(n_build);
+    int64_t* slot_row = (int64_t*)malloc(cap * sizeof(int64_t));
+    int64_t* links = (int64_t*)malloc(n_build * sizeof(int64_t));
+    memset(slot_row, -1, cap * sizeof(int64_t));
+    for (int64_t i = 0; i < n_build; i++) links[i] = -1;
+    for (int64_t i = 0; i < n_build; i++) {
+        uint64_t h = pg_murmur3_finalize(pg_bigint_hash(bkeys[i]));
+        int64_t pos = (int64_t)(h & (cap - 1));
+        for (;;) {
+            int64_t r = slot_row[pos];
+            if (r == -1) {
+                slot_row[pos] = i;
+                break;
+            }
+            if (bkeys[r] == bkeys[i]) {
+                /* head-insert: new row becomes chain head */
+                links[i] = r;
+                slot_row[pos] = i;
+                break;
+            }
+            pos = (pos + 1) & (cap - 1);
+        }
+    }
+    int64_t out = 0;
+    for (int64_t i = 0; i < n_probe; i++) {
+        uint64_t h = pg_murmur3_finalize(pg_bigint_hash(pkeys[i]));
+        int64_t pos = (int64_t)(h & (cap - 1));
+        int64_t r = -1;
+        for (;;) {
+            int64_t s = slot_row[pos];
+            if (s == -1) break;
+            if (bkeys[s] == pkeys[i]) {
+                r = s;
+                break;
+            }
+            pos = (pos + 1) & (cap - 1);
+        }
+        while (r != -1) {
+            if (out < out_cap) {
+                out_probe[out] = i;
+                out_build[out] = r;
+            }
+            out++;
+            r = links[r];
+        }
+    }
+    free(slot_row);
+    free(links);
+    return out;
+}
+
+/* ---------------- Q3 ---------------- */
+
+#define Q3_DATE 9204 /* DATE '1995-03-15', epoch days */
+
+int32_t oracle_q3(int64_t n_cust, const int64_t* ck, const uint8_t* cseg,
+                  int64_t n_ord, const int64_t* ook, const int64_t* ock,
+                  const int32_t* odate, int64_t n_li, const int64_t* lok,
+                  const double* lep, const double* ldisc, const int32_t* lsd,
+                  int32_t limit, q3_row_t* out)
+{
+    /* 1. customer filter: mktsegment == 'BUILDING' (id 1) -> key set.
+     * custkey values are dense 1..n_cust (tpch), use a byte map. */
+    int64_t max_ck = 0;
+    for (int64_t i = 0; i < n_cust; i++)
+        if (ck[i] > max_ck) max_ck = ck[i];
+    uint8_t* in_seg = (uint8_t*)calloc(max_ck + 1, 1);
+    for (int64_t i = 0; i < n_cust; i++)
+        if (cseg[i] == 1) in_seg[ck[i]] = 1;
+
+    /* 2. orders filter (odate < 9204) + semijoin custkey -> build table
+     * keyed by orderkey: open-address, murmur3(bigint_hash), linear probe
+     * (join build semantics, keys unique). */
+    int64_t* b_ok;
+    int32_t* b_od;
+    int64_t n_b = 0;
+    b_ok = (int64_t*)malloc(n_ord * sizeof(int64_t));
+    b_od = (int32_t*)malloc(n_ord * sizeof(int32_t));
+    for (int64_t i = 0; i < n_ord; i++) {
+        if (odate[i] < Q3_DATE && ock[i] <= max_ck && in_seg[ock[i]]) {
+            b_ok[n_b] = ook[i];
+            b_od[n_b] = odate[i];
+            n_b++;
+        }
+    }
+    int64_t cap = hash_capacity(n_b < 2 ? 2 : n_b);
+    int64_t* slot = (int64_t*)malloc(cap * sizeof(int64_t));
+    memset(slot, -1, cap * sizeof(int64_t));
+    for (int64_t i = 0; i < n_b; i++) {
+        uint64_t h = pg_murmur3_finalize(pg_bigint_hash(b_ok[i]));
+        int64_t pos = (int64_t)(h & (cap - 1));
+        while (slot[pos] != -1) pos = (pos + 1) & (cap - 1);
+        slot[pos] = i;
+    }
+    /* revenue accumulators per build row: exact ticks + exact fx128 sum of
+     * the f64 per-row products (order-independent; see fixed128.h) */
+    int64_t* rev4 = (int64_t*)calloc(n_b, sizeof(int64_t));
+    uint64_t* fhi = (uint64_t*)calloc(n_b, sizeof(uint64_t));
+    uint64_t* flo = (uint64_t*)calloc(n_b, sizeof(uint64_t));
+
+    /* 3. lineitem filter (shipdate > 9204) + probe + grouped sum */
+#pragma omp parallel for schedule(static)
+    for (int64_t i = 0; i < n_li; i++) {
+        if (lsd[i] <= Q3_DATE) continue;
+        uint64_t h = pg_murmur3_finalize(pg_bigint_hash(lok[i]));
+        int64_t pos = (int64_t)(h & (cap - 1));
+        int64_t r = -1;
+        for (;;) {
+            int64_t s = slot[pos];
+            if (s == -1) break;
+            if (b_ok[s] == lok[i]) {
+                r = s;
+                break;
+            }
+            pos = (pos + 1) & (cap - 1);
+        }
+        if (r == -1) continue;
+        int64_t cents = (int64_t)(lep[i] * 100.0 + 0.5);
+        int64_t d = (int64_t)(ldisc[i] * 100.0 + 0.5);
+        int64_t ticks = cents * (100 - d);
+        double p = lep[i] * (1.0 - ldisc[i]);
+        uint64_t phi, plo;
+        fx128_from_f64(p, &phi, &plo);
+#pragma omp critical
+        {
+            rev4[r] += ticks;
+            fx128_add(&fhi[r], &flo[r], phi, plo);
+        }
+    }
+
+    /* 4. TopN by (revenue desc, orderdate asc, orderkey asc) — insertion
+     * into a bounded sorted list (TopNOperator.java:90-111 semantics) */
+    int32_t n_out = 0;
+    for (int64_t i = 0; i < n_b; i++) {
+        if (rev4[i] == 0) continue;
+        q3_row_t row = {b_ok[i], rev4[i], b_od[i], 0,
+                        fx128_to_f64(fhi[i], flo[i])};
+        int32_t pos = n_out;
+        while (pos > 0) {
+            q3_row_t* prev = &out[pos - 1];
+            int better = row.revenue_1e4 > prev->revenue_1e4 ||
+                         (row.revenue_1e4 == prev->revenue_1e4 &&
+                          (row.orderdate < prev->orderdate ||
+                           (row.orderdate == prev->orderdate &&
+                            row.orderkey < prev->orderkey)));
+            if (!better) break;
+            pos--;
+        }
+        if (pos >= limit) continue;
+        if (n_out < limit) n_out++;
+        for (int32_t j = n_out - 1; j > pos; j--) out[j] = out[j - 1];
+        out[pos] = row;
+    }
+    free(in_seg);
+    free(b_ok);
+    free(b_od);
+    free(slot);
+    free(rev4);
+    free(fhi);
+    free(flo);
+    return n_out;
+}

@@ -1,0 +1,128 @@
+/* oracle.h — CPU restatement of the reference's per-Page operator hot path.
+ *
+ * TEST INFRASTRUCTURE ONLY.  Only tests/, __graft_entry__.smoke() and
+ * bench.py's cpu_baseline leg may call this library; the product path
+ * (presto_amd + libpresto_gpu.so) never links or loads it.
+ *
+ * Each function follows the cited Java implementation in /root/reference
+ * (prestodb/presto 0.300-SNAPSHOT).  The oracle is pinned by the reference's
+ * golden SF1 result vectors (see tpchgen.h header) via oracle_q1 / oracle_q3
+ * in decimal mode, which reproduce
+ *   presto-product-tests/.../hive_tpch/q01.result and q03.result
+ * digit-for-digit (exact decimal arithmetic).
+ */
+#ifndef ORACLE_H
+#define ORACLE_H
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* ---------------- TPC-H Q1 ----------------
+ * SQL: presto-benchto-benchmarks/.../tpch/q01.sql
+ * Pipeline restated: ScanFilterAndProjectOperator (filter shipdate <= 10471,
+ * PageProcessor.java:299-343 selection semantics) -> HashAggregationOperator
+ * (HashAggregationOperator.java:413,451; grouping on (returnflag,linestatus),
+ * MultiChannelGroupByHash.java:300-380 assigns dense group ids in first-seen
+ * order) -> aggregates per DoubleSumAggregation.java (running f64 sum),
+ * LongSumAggregation.java:33-37 (overflow-checked), CountAggregation.java:34,
+ * AverageAggregations.java:34-63 ((count, sum) state, output sum/count).
+ *
+ * Groups are keyed by (returnflag<<8)|linestatus; results are emitted sorted
+ * by (returnflag, linestatus) like the query's ORDER BY.
+ *
+ * Decimal outputs are exact integer ticks:
+ *   sum_qty        : quantity summed in units (quantity is integral)
+ *   sum_base_cents : extendedprice summed in cents            (scale 2)
+ *   sum_disc_1e4   : sum extprice*(1-disc), ticks of 1e-4     (scale 4)
+ *   sum_charge_1e6 : sum extprice*(1-disc)*(1+tax), 1e-6      (scale 6)
+ *                    (returned as hi/lo int128 halves)
+ *   sum_disc_cents : discount summed in hundredths            (scale 2)
+ * f64 outputs follow the fixed-tree deterministic schedule the GPU kernels
+ * use (documented in DESIGN.md): per virtual-lane sequential accumulation
+ * over rows lane, lane+NLANES, ... then pairwise reduction — bit-exact
+ * reproducible on CPU and GPU.
+ */
+typedef struct {
+    uint8_t returnflag, linestatus;
+    int64_t count_order;
+    int64_t sum_qty_units;
+    int64_t sum_base_cents;
+    int64_t sum_disc_1e4;
+    int64_t sum_charge_1e6_hi; /* int128 = hi*2^64 + lo (two's complement) */
+    uint64_t sum_charge_1e6_lo;
+    int64_t sum_disc_cents;
+    /* f64 fixed-tree sums (same schedule as GPU kernels) */
+    double f64_sum_qty, f64_sum_base, f64_sum_disc_price, f64_sum_charge,
+        f64_sum_disc;
+} q1_group_t;
+
+/* returns number of groups (<=6), fills groups[] sorted by key */
+int32_t oracle_q1(int64_t n_rows, const double* quantity,
+                  const double* extendedprice, const double* discount,
+                  const double* tax, const int32_t* shipdate,
+                  const uint8_t* returnflag, const uint8_t* linestatus,
+                  q1_group_t* groups);
+
+/* ---------------- TPC-H Q3 ----------------
+ * SQL: presto-benchto-benchmarks/.../tpch/q03.sql
+ * Pipeline restated: build customer set (mktsegment='BUILDING'), hash join
+ * orders (orderdate < 9204) x customer (HashBuilderOperator.java:333,
+ * PagesHash.java:81-125 open-address linear probe with murmur3-finalizer
+ * bucket, PagesHash.java:236-252), then lineitem (shipdate > 9204) probe
+ * (LookupJoinOperator.java:481-604) with group-by on orderkey
+ * (BigintGroupByHash.java:222-252) summing revenue; TopN 10 by
+ * (revenue desc, orderdate asc, orderkey asc) (TopNOperator.java:90-111;
+ * orderkey added as deterministic final tiebreak — revenue ties beyond the
+ * SQL ORDER BY are unspecified in the reference).
+ * revenue is exact decimal: ticks of 1e-4 = cents*(100-d). */
+typedef struct {
+    int64_t orderkey;
+    int64_t revenue_1e4;
+    int32_t orderdate;
+    int32_t shippriority;
+    double f64_revenue; /* fixed-tree f64 mode (same GPU schedule) */
+} q3_row_t;
+
+int32_t oracle_q3(int64_t n_cust, const int64_t* c_custkey,
+                  const uint8_t* c_mktseg_id, int64_t n_ord,
+                  const int64_t* o_orderkey, const int64_t* o_custkey,
+                  const int32_t* o_orderdate, int64_t n_li,
+                  const int64_t* l_orderkey, const double* l_extendedprice,
+                  const double* l_discount, const int32_t* l_shipdate,
+                  int32_t limit, q3_row_t* out);
+
+/* ---------------- operator-level primitives (parity targets) ---------- */
+
+/* murmur3 finalizer bucket — PagesHash.java:236-252 /
+ * BigintGroupByHash.getHashPosition */
+uint64_t oracle_murmur3_finalize(uint64_t h);
+/* bigint type hash — AbstractLongType.java:137-140 (xxhash64 mix) */
+uint64_t oracle_bigint_hash(int64_t v);
+/* partition id — HashGenerator.java:22-29:
+ * (u32(Long.hashCode(rawHash)) * partitionCount) >> 32 */
+int32_t oracle_partition(uint64_t raw_hash, int32_t partition_count);
+
+/* Group-by over a single bigint key column, BigintGroupByHash.java:222-332:
+ * open-address linear probe, fill 0.75, capacity = next pow2(ceil(hint/.75)),
+ * group ids dense in first-seen order.  Writes group_ids[i] per row and
+ * returns group count.  (No null handling: key column non-null.) */
+int64_t oracle_bigint_group_by(int64_t n, const int64_t* keys,
+                               int32_t* group_ids);
+
+/* Join build+probe over bigint keys with duplicate chains,
+ * PagesHash.java:81-181 + ArrayPositionLinks.java:25-56 (head-insert; probe
+ * visits the LATEST inserted duplicate first, then walks to earlier ones).
+ * For each probe row i, appends matched build-row indexes (in chain order)
+ * to out_build_idx/out_probe_idx; returns number of emitted pairs
+ * (capacity out_cap; excess dropped). */
+int64_t oracle_join_bigint(int64_t n_build, const int64_t* build_keys,
+                           int64_t n_probe, const int64_t* probe_keys,
+                           int64_t* out_probe_idx, int64_t* out_build_idx,
+                           int64_t out_cap);
+
+#ifdef __cplusplus
+}
+#endif
+#endif

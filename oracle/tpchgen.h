@@ -1,0 +1,69 @@
+/* tpchgen.h — columns-only TPC-H data generator (dbgen restatement).
+ *
+ * TEST / BENCH INPUT INFRASTRUCTURE + ORACLE DATA SOURCE.
+ * This is NOT part of the product compute path: it synthesizes the input
+ * columns that both the CPU oracle (oracle.c) and the GPU kernels consume.
+ *
+ * The reference (prestodb/presto) generates TPC-H rows through the external
+ * dependency io.airlift.tpch:tpch:0.10 (pom.xml:1425-1429), called from
+ * presto-tpch/src/main/java/com/facebook/presto/tpch/TpchRecordSet.java:46-51.
+ * That dependency's source is NOT vendored in /root/reference, so this file
+ * restates the published TPC-H dbgen algorithm (TPC-H specification §4.2.3
+ * "Random number generation"; dbgen rnd.c / build.c / speed_seed.c semantics,
+ * of which io.airlift.tpch is a faithful port).  Parity of the restatement is
+ * pinned by the reference's own golden SF1 result vectors:
+ *   presto-product-tests/src/main/resources/sql-tests/testcases/hive_tpch/q01.result
+ *   presto-product-tests/src/main/resources/sql-tests/testcases/hive_tpch/q03.result
+ * (restated as committed fixtures in tests/golden/), which are exact-decimal
+ * aggregates over every generated row of lineitem/orders/customer at SF1.
+ *
+ * Dates are day indexes: idx 1 == 1992-01-01; epoch32 = 8035 + idx - 1
+ * (8035 = days from 1970-01-01 to 1992-01-01).
+ */
+#ifndef TPCHGEN_H
+#define TPCHGEN_H
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* row counts */
+int64_t tpch_customer_count(double sf); /* 150,000 * sf */
+int64_t tpch_orders_count(double sf);   /* 1,500,000 * sf */
+/* scans the line-count RNG stream; 6,001,215 at SF1 */
+int64_t tpch_lineitem_count(double sf);
+
+/* customer columns for rows [start, start+count) (0-based row index).
+ * custkey = row index + 1.  mktseg_id: 0=AUTOMOBILE 1=BUILDING 2=FURNITURE
+ * 3=MACHINERY 4=HOUSEHOLD (distribution order of dbgen dists.dss msegmnt). */
+void tpch_gen_customer(double sf, int64_t start, int64_t count,
+                       int64_t* custkey /*nullable*/, uint8_t* mktseg_id);
+
+/* orders columns for order rows [start, start+count) (0-based).
+ * shippriority is the constant 0 (dbgen mk_order) and is not emitted.
+ * lcnt: lineitems per order (1..7). Any output pointer may be NULL. */
+void tpch_gen_orders(double sf, int64_t start, int64_t count,
+                     int64_t* orderkey, int64_t* custkey,
+                     int32_t* orderdate_epoch, int32_t* lcnt);
+
+/* number of lineitem rows belonging to orders [0, ord_start) — prefix offset */
+int64_t tpch_lineitem_offset(double sf, int64_t ord_start);
+
+/* lineitem columns for all rows of orders [ord_start, ord_start+ord_count).
+ * Writes rows densely from output index 0; the caller uses
+ * tpch_lineitem_offset() to place chunks. Returns rows written.
+ * Money columns are DOUBLE-typed like the reference's presto-tpch mapping
+ * (TpchMetadata.java:537-553): quantity (integral 1..50), extendedprice
+ * (cents/100), discount (0.00..0.10), tax (0.00..0.08).
+ * returnflag in {'A','N','R'}, linestatus in {'F','O'} (ASCII u8). */
+int64_t tpch_gen_lineitem(double sf, int64_t ord_start, int64_t ord_count,
+                          int64_t* orderkey, double* quantity,
+                          double* extendedprice, double* discount, double* tax,
+                          int32_t* shipdate_epoch, uint8_t* returnflag,
+                          uint8_t* linestatus);
+
+#ifdef __cplusplus
+}
+#endif
+#endif

@@ -1,0 +1,126 @@
+/* fixed128.h — exact 64.64 fixed-point helpers shared by the GPU kernels and
+ * the CPU oracle (oracle/ includes this header; the product never includes
+ * anything under oracle/).
+ *
+ * Purpose: order-independent, bit-deterministic summation of non-negative
+ * f64 values (the revenue/charge sums of TPC-H Q1/Q3).  Every f64 value in
+ * the domain (0 <= p < 2^53, ulp(p) >= 2^-64) converts EXACTLY to 64.64
+ * fixed point, so the accumulated 128-bit sum is the exact real-number sum
+ * of the f64 addends, independent of addition order; converting the total
+ * back to f64 applies exactly one correctly-rounded (RNE) rounding.
+ * This meets and exceeds the <=1-ulp double SUM/AVG parity budget stated in
+ * BASELINE.json's north_star.  Compiles as both host C/C++ and HIP device
+ * code. */
+#ifndef PRESTO_AMD_FIXED128_H
+#define PRESTO_AMD_FIXED128_H
+#include <stdint.h>
+
+#ifdef __HIPCC__
+#define FX_HD __host__ __device__
+#else
+#define FX_HD
+#endif
+
+/* split non-negative p (< 2^63, ulp >= 2^-64) into 64.64 fixed point.
+ * hi = floor(p); lo = (p - floor(p)) * 2^64 — both steps exact in f64. */
+static inline FX_HD void fx128_from_f64(double p, uint64_t* hi, uint64_t* lo)
+{
+    uint64_t h = (uint64_t)p;
+    double frac = p - (double)h; /* exact: subtraction of floor */
+    *hi = h;
+    *lo = (uint64_t)(frac * 18446744073709551616.0 /* 2^64, exact scale */);
+}
+
+/* accumulate with carry: (ahi:alo) += (bhi:blo) */
+static inline FX_HD void fx128_add(uint64_t* ahi, uint64_t* alo, uint64_t bhi,
+                                   uint64_t blo)
+{
+    uint64_t lo = *alo + blo;
+    *ahi += bhi + (lo < blo ? 1u : 0u);
+    *alo = lo;
+}
+
+/* value = (hi + lo*2^-64) rounded to nearest-even f64.
+ * hi < 2^63 in all uses here. */
+static inline FX_HD double fx128_to_f64(uint64_t hi, uint64_t lo)
+{
+    if (hi == 0 && lo == 0) return 0.0;
+    /* normalize the 128-bit integer v = hi:lo; result = v * 2^-64 */
+    int shift; /* left shift to put msb at bit 127 */
+    uint64_t h = hi, l = lo;
+    if (h == 0) {
+        h = l;
+        l = 0;
+        shift = 64;
+    } else {
+        shift = 0;
+    }
+    /* count leading zeros of h (h != 0) */
+    int lz = 0;
+    for (uint64_t t = h; !(t & 0x8000000000000000ull); t <<= 1) lz++;
+    if (lz) {
+        h = (h << lz) | (l >> (64 - lz));
+        l <<= lz;
+    }
+    shift += lz;
+    /* h now has msb at bit 63; take top 54 bits: sig53 + round bit */
+    uint64_t top54 = h >> 10;
+    uint64_t sticky = (h & 0x3ffull) | (l ? 1ull : 0ull);
+    uint64_t sig = top54 >> 1;  /* 53-bit significand (implicit bit set) */
+    uint64_t round = top54 & 1; /* guard bit */
+    if (round && (sticky || (sig & 1))) {
+        sig++;
+        if (sig >> 53) { /* carry out of significand */
+            sig >>= 1;
+            shift--;
+        }
+    }
+    /* exponent: v = sig * 2^(127-52-shift); result = v * 2^-64 */
+    int e2 = 127 - 52 - shift - 64;
+    double r = (double)sig;
+    /* scale by 2^e2 via exact power-of-two multiply (|e2| < 1030) */
+    while (e2 > 0) {
+        int s = e2 > 63 ? 63 : e2;
+        r *= (double)(1ull << s);
+        e2 -= s;
+    }
+    while (e2 < 0) {
+        int s = (-e2) > 63 ? 63 : -e2;
+        r /= (double)(1ull << s);
+        e2 += s;
+    }
+    return r;
+}
+
+/* ---- hash primitives shared by kernels and oracle ---- */
+
+/* murmur3 64-bit finalizer — PagesHash.java:236-252,
+ * BigintGroupByHash.getHashPosition (same constants) */
+static inline FX_HD uint64_t pg_murmur3_finalize(uint64_t h)
+{
+    h ^= h >> 33;
+    h *= 0xff51afd7ed558ccdull;
+    h ^= h >> 33;
+    h *= 0xc4ceb9fe1a85ec53ull;
+    h ^= h >> 33;
+    return h;
+}
+
+/* bigint type hash — AbstractLongType.java:137-140 (xxhash64 mix):
+ * rotateLeft(v * 0xC2B2AE3D27D4EB4F, 31) * 0x9E3779B185EBCA87 */
+static inline FX_HD uint64_t pg_bigint_hash(int64_t v)
+{
+    uint64_t x = (uint64_t)v * 0xC2B2AE3D27D4EB4Full;
+    x = (x << 31) | (x >> 33);
+    return x * 0x9E3779B185EBCA87ull;
+}
+
+/* partition id — HashGenerator.java:22-29:
+ * (toUnsignedLong(Long.hashCode(rawHash)) * partitionCount) >> 32 */
+static inline FX_HD int32_t pg_partition(uint64_t raw_hash, int32_t n_part)
+{
+    uint32_t x = (uint32_t)(raw_hash ^ (raw_hash >> 32)); /* Long.hashCode */
+    return (int32_t)(((uint64_t)x * (uint64_t)n_part) >> 32);
+}
+
+#endif

@@ -1,0 +1,25 @@
+#!/usr/bin/env python3
+"""Regenerates the golden fixtures in this directory.
+
+The fixtures are the reference's own TPC-H SF1 known-answer vectors
+(exact-decimal result sets, produced by the reference's product-test suite):
+
+  q01_sf1.result <- /root/reference/presto-product-tests/src/main/resources/
+                    sql-tests/testcases/hive_tpch/q01.result
+  q03_sf1.result <- .../hive_tpch/q03.result
+
+They pin (a) the TPC-H dbgen restatement in oracle/tpchgen.c and (b) the
+decimal aggregate semantics of oracle/oracle.c — see tests/test_oracle.py.
+This script only runs in the build container (where /root/reference exists);
+the committed fixtures are what travels to the GPU box.
+"""
+import shutil
+import pathlib
+
+REF = pathlib.Path("/root/reference/presto-product-tests/src/main/resources/"
+                   "sql-tests/testcases/hive_tpch")
+HERE = pathlib.Path(__file__).parent
+
+for q in ("q01", "q03"):
+    shutil.copy(REF / f"{q}.result", HERE / f"{q}_sf1.result")
+    print(f"wrote {q}_sf1.result")

@@ -1,0 +1,139 @@
+"""ctypes binding for oracle/liboracle.so — TEST INFRASTRUCTURE ONLY.
+
+Only tests/, __graft_entry__.smoke() and bench.py's cpu_baseline leg use this
+module; the product package (presto_amd) never imports it.
+"""
+import ctypes as C
+
+import numpy as np
+
+
+class Q1Group(C.Structure):
+    _fields_ = [
+        ("returnflag", C.c_uint8), ("linestatus", C.c_uint8),
+        ("count_order", C.c_int64), ("sum_qty_units", C.c_int64),
+        ("sum_base_cents", C.c_int64), ("sum_disc_1e4", C.c_int64),
+        ("sum_charge_1e6_hi", C.c_int64), ("sum_charge_1e6_lo", C.c_uint64),
+        ("sum_disc_cents", C.c_int64),
+        ("f64_sum_qty", C.c_double), ("f64_sum_base", C.c_double),
+        ("f64_sum_disc_price", C.c_double), ("f64_sum_charge", C.c_double),
+        ("f64_sum_disc", C.c_double),
+    ]
+
+
+class Q3Row(C.Structure):
+    _fields_ = [
+        ("orderkey", C.c_int64), ("revenue_1e4", C.c_int64),
+        ("orderdate", C.c_int32), ("shippriority", C.c_int32),
+        ("f64_revenue", C.c_double),
+    ]
+
+
+def _p(a):
+    return a.ctypes.data_as(C.c_void_p)
+
+
+class OracleLib:
+    def __init__(self, path):
+        self.lib = C.CDLL(path)
+        L = self.lib
+        L.tpch_customer_count.restype = C.c_int64
+        L.tpch_customer_count.argtypes = [C.c_double]
+        L.tpch_orders_count.restype = C.c_int64
+        L.tpch_orders_count.argtypes = [C.c_double]
+        L.tpch_lineitem_count.restype = C.c_int64
+        L.tpch_lineitem_count.argtypes = [C.c_double]
+        L.tpch_lineitem_offset.restype = C.c_int64
+        L.tpch_lineitem_offset.argtypes = [C.c_double, C.c_int64]
+        L.tpch_gen_lineitem.restype = C.c_int64
+        L.oracle_q1.restype = C.c_int32
+        L.oracle_q3.restype = C.c_int32
+        L.oracle_murmur3_finalize.restype = C.c_uint64
+        L.oracle_murmur3_finalize.argtypes = [C.c_uint64]
+        L.oracle_bigint_hash.restype = C.c_uint64
+        L.oracle_bigint_hash.argtypes = [C.c_int64]
+        L.oracle_partition.restype = C.c_int32
+        L.oracle_partition.argtypes = [C.c_uint64, C.c_int32]
+        L.oracle_bigint_group_by.restype = C.c_int64
+        L.oracle_join_bigint.restype = C.c_int64
+
+    # ---- generator ----
+    def lineitem_count(self, sf):
+        return self.lib.tpch_lineitem_count(C.c_double(sf))
+
+    def gen_lineitem(self, sf):
+        n = self.lineitem_count(sf)
+        n_ord = self.lib.tpch_orders_count(C.c_double(sf))
+        cols = dict(
+            orderkey=np.empty(n, np.int64), quantity=np.empty(n, np.float64),
+            extendedprice=np.empty(n, np.float64),
+            discount=np.empty(n, np.float64), tax=np.empty(n, np.float64),
+            shipdate=np.empty(n, np.int32), returnflag=np.empty(n, np.uint8),
+            linestatus=np.empty(n, np.uint8))
+        w = self.lib.tpch_gen_lineitem(
+            C.c_double(sf), C.c_int64(0), C.c_int64(n_ord),
+            _p(cols["orderkey"]), _p(cols["quantity"]),
+            _p(cols["extendedprice"]), _p(cols["discount"]), _p(cols["tax"]),
+            _p(cols["shipdate"]), _p(cols["returnflag"]),
+            _p(cols["linestatus"]))
+        assert w == n
+        return cols
+
+    def gen_orders(self, sf):
+        n = self.lib.tpch_orders_count(C.c_double(sf))
+        cols = dict(orderkey=np.empty(n, np.int64),
+                    custkey=np.empty(n, np.int64),
+                    orderdate=np.empty(n, np.int32))
+        self.lib.tpch_gen_orders(C.c_double(sf), C.c_int64(0), C.c_int64(n),
+                                 _p(cols["orderkey"]), _p(cols["custkey"]),
+                                 _p(cols["orderdate"]), None)
+        return cols
+
+    def gen_customer(self, sf):
+        n = self.lib.tpch_customer_count(C.c_double(sf))
+        cols = dict(custkey=np.empty(n, np.int64),
+                    mktseg=np.empty(n, np.uint8))
+        self.lib.tpch_gen_customer(C.c_double(sf), C.c_int64(0), C.c_int64(n),
+                                   _p(cols["custkey"]), _p(cols["mktseg"]))
+        return cols
+
+    # ---- oracles ----
+    def q1(self, li):
+        groups = (Q1Group * 6)()
+        ng = self.lib.oracle_q1(
+            C.c_int64(len(li["quantity"])), _p(li["quantity"]),
+            _p(li["extendedprice"]), _p(li["discount"]), _p(li["tax"]),
+            _p(li["shipdate"]), _p(li["returnflag"]), _p(li["linestatus"]),
+            groups)
+        return [groups[i] for i in range(ng)]
+
+    def q3(self, cust, orders, li, limit=10):
+        rows = (Q3Row * limit)()
+        nr = self.lib.oracle_q3(
+            C.c_int64(len(cust["custkey"])), _p(cust["custkey"]),
+            _p(cust["mktseg"]), C.c_int64(len(orders["orderkey"])),
+            _p(orders["orderkey"]), _p(orders["custkey"]),
+            _p(orders["orderdate"]), C.c_int64(len(li["orderkey"])),
+            _p(li["orderkey"]), _p(li["extendedprice"]), _p(li["discount"]),
+            _p(li["shipdate"]), C.c_int32(limit), rows)
+        return [rows[i] for i in range(nr)]
+
+    def group_by(self, keys):
+        keys = np.ascontiguousarray(keys, np.int64)
+        gids = np.empty(len(keys), np.int32)
+        ng = self.lib.oracle_bigint_group_by(C.c_int64(len(keys)), _p(keys),
+                                             _p(gids))
+        return ng, gids
+
+    def join(self, build_keys, probe_keys, cap=None):
+        b = np.ascontiguousarray(build_keys, np.int64)
+        p = np.ascontiguousarray(probe_keys, np.int64)
+        if cap is None:
+            cap = 4 * (len(b) + len(p)) + 16
+        op = np.empty(cap, np.int64)
+        ob = np.empty(cap, np.int64)
+        n = self.lib.oracle_join_bigint(C.c_int64(len(b)), _p(b),
+                                        C.c_int64(len(p)), _p(p), _p(op),
+                                        _p(ob), C.c_int64(cap))
+        assert n <= cap
+        return op[:n], ob[:n]
