@@ -1,0 +1,246 @@
+/* presto_gpu.h — public C-ABI of the MI355X-native Presto hot-path library.
+ *
+ * This is the drop-in boundary of SURVEY.md §8b: a handle-based operator
+ * lifecycle mirroring the reference's Operator interface
+ * (presto-main-base/.../operator/Operator.java:20-102, constructed via
+ * OperatorFactory.createOperator and driven by Driver.processInternal:402),
+ * with Page data crossing the seam as flat column descriptors matching the
+ * reference's Block layouts (presto-common/.../block/LongArrayBlock.java:38-52,
+ * IntArrayBlock, ByteArrayBlock, DictionaryBlock.java:53-64).
+ *
+ * In production the Java side binds this ABI through a thin JNI veneer
+ * (see INTEGRATION.md for the binding a presto-main maintainer would add);
+ * in this repo the same ABI is driven by the Python harness in presto_amd/
+ * and by tests/ replicating OperatorAssertion.toPages
+ * (presto-main-base/src/test/.../OperatorAssertion.java:62-176).
+ *
+ * Contract (verified by the reference's Driver, restated here):
+ *  - one handle == one logical driver: all calls on a handle are
+ *    single-threaded (Driver.processInternal holds an exclusive lock);
+ *  - pg_op_add_input only when pg_op_needs_input returns 1
+ *    (Driver.java:446-458); input pages are borrowed for the call;
+ *  - pg_op_get_output may produce nothing (*out == NULL); output pages are
+ *    owned by the library until the next pg_op_get_output/pg_op_destroy;
+ *  - pg_op_finish is idempotent; cleanup via pg_op_destroy (close()).
+ *  - build->probe bridging uses table handles (mirrors
+ *    PartitionedLookupSourceFactory.java:149,181).
+ *
+ * Errors: negative return, message via pg_last_error().
+ * All compute requires an AMD gfx950 GPU; calls fail loudly without one
+ * (there is no CPU fallback in this library).
+ */
+#ifndef PRESTO_GPU_H
+#define PRESTO_GPU_H
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+#define PG_ABI_VERSION 1
+
+/* ---- status ---- */
+typedef int32_t pg_status; /* 0 ok, <0 error */
+#define PG_OK 0
+#define PG_ERR (-1)
+const char* pg_last_error(void);
+
+/* ---- device helpers (memory plumbing for the harness; tensors may also be
+ * allocated by the caller, e.g. torch, and passed as raw device pointers) */
+pg_status pg_device_count(int32_t* out);
+pg_status pg_device_sync(void);
+pg_status pg_device_malloc(int64_t bytes, void** out);
+pg_status pg_device_free(void* p);
+pg_status pg_memcpy_h2d(void* dst, const void* src, int64_t bytes);
+pg_status pg_memcpy_d2h(void* dst, const void* src, int64_t bytes);
+
+/* ---- Page / Block descriptors (SURVEY.md §8b struct) ----
+ * Concrete layouts follow the Java blocks: fixed-width values array +
+ * optional byte-per-position null mask (LongArrayBlock.java:38-52). */
+typedef enum {
+    PG_T_U8 = 0,  /* ByteArrayBlock / dictionary codes */
+    PG_T_I32 = 1, /* IntArrayBlock / DateType days */
+    PG_T_I64 = 2, /* LongArrayBlock / BigintType */
+    PG_T_F64 = 3, /* LongArrayBlock bits / DoubleType */
+} pg_type;
+
+typedef struct {
+    int32_t tag;            /* pg_type */
+    int32_t on_device;      /* 1: data is a device pointer */
+    void* data;             /* values array, n_rows elements */
+    const uint8_t* null_mask; /* optional, 1 byte/pos, 1 = null; may be NULL */
+} pg_col;
+
+typedef struct {
+    int64_t n_rows;
+    int32_t n_cols;
+    pg_col cols[16];
+} pg_page;
+
+/* ---- plan blobs ----
+ * The reference JIT-compiles per-query filter/projection/accumulator classes
+ * at runtime (sql/gen/PageFunctionCompiler.java:126, AccumulatorCompiler,
+ * JoinCompiler).  The MI355X-native analog is a set of ahead-of-time
+ * specialized HIP kernels selected by these plan descriptors. */
+
+typedef enum { PG_CMP_LT = 0, PG_CMP_LE, PG_CMP_GT, PG_CMP_GE, PG_CMP_EQ,
+               PG_CMP_NE } pg_cmp;
+
+typedef struct {
+    int32_t col;   /* input channel */
+    int32_t op;    /* pg_cmp */
+    int64_t ival;  /* compare value for integer columns */
+    double dval;   /* compare value for f64 columns */
+} pg_pred;
+
+/* projection expressions (PageProjection analogs) */
+typedef enum {
+    PG_PROJ_IDENT = 0,        /* column a */
+    PG_PROJ_DISC_PRICE = 1,   /* a * (1 - b)            (Q1/Q3 revenue) */
+    PG_PROJ_CHARGE = 2,       /* a * (1 - b) * (1 + c)  (Q1 charge) */
+} pg_proj_kind;
+
+typedef struct {
+    int32_t kind; /* pg_proj_kind */
+    int32_t a, b, c; /* input channels */
+} pg_proj;
+
+/* aggregate functions over a projection */
+typedef enum {
+    PG_AGG_COUNT = 0,    /* CountAggregation.java:34 */
+    PG_AGG_SUM_F64 = 1,  /* DoubleSumAggregation (deterministic schedule,
+                            DESIGN.md §determinism) */
+    PG_AGG_SUM_DEC = 2,  /* exact decimal ticks; scale from dec_scale
+                            (hive-decimal semantics of the golden vectors) */
+    PG_AGG_SUM_I64 = 3,  /* LongSumAggregation.java:33-37 */
+} pg_agg_func;
+
+typedef struct {
+    int32_t func;     /* pg_agg_func */
+    pg_proj proj;     /* input expression */
+    int32_t dec_scale; /* for SUM_DEC: ticks = round(value * 10^dec_scale) */
+} pg_agg;
+
+/* -------- operator kinds + plans -------- */
+typedef enum {
+    PG_OP_FILTER_PROJECT = 1, /* ScanFilterAndProjectOperator.java:67 +
+                                 PageProcessor.java:112,299-343 */
+    PG_OP_HASH_AGG_SMALL = 2, /* HashAggregationOperator.java:56 with
+                                 low-cardinality dict-u8 keys (Q1 shape) */
+    PG_OP_HASH_BUILD = 3,     /* HashBuilderOperator.java:55 */
+    PG_OP_LOOKUP_JOIN = 4,    /* LookupJoinOperator.java:481-604 (optionally
+                                 fused with grouped SUM into the table —
+                                 Q3's join+partial-agg pipeline) */
+    PG_OP_TOPN = 5,           /* TopNOperator.java:32,90-111 */
+    PG_OP_PARTITION = 6,      /* PartitionedOutputOperator.partitionPage:394 /
+                                 LocalExchange partition split */
+} pg_op_kind;
+
+#define PG_MAX_PRED 8
+#define PG_MAX_AGG 8
+#define PG_MAX_KEYVALS 8
+
+typedef struct {
+    int32_t n_preds;
+    pg_pred preds[PG_MAX_PRED]; /* conjunction */
+    int32_t n_proj;
+    pg_proj proj[16];
+} pg_plan_filter_project;
+
+typedef struct {
+    /* up to two u8 key channels with enumerated code values; group id =
+     * idx(key0)*n_vals1 + idx(key1); results emitted in that order,
+     * restricted to non-empty groups */
+    int32_t n_preds;
+    pg_pred preds[PG_MAX_PRED]; /* fused pre-filter (scan+filter+agg) */
+    int32_t n_keys;             /* 1 or 2 */
+    int32_t key_col[2];
+    int32_t n_vals[2];
+    uint8_t key_vals[2][PG_MAX_KEYVALS];
+    int32_t n_aggs;
+    pg_agg aggs[PG_MAX_AGG];
+} pg_plan_hash_agg_small;
+
+typedef struct {
+    int32_t n_preds;
+    pg_pred preds[PG_MAX_PRED]; /* fused pre-filter on build input */
+    int32_t key_col;            /* bigint key channel */
+    int64_t semijoin_table;     /* >=0: keep only rows whose column
+                                   semijoin_col matches that table's key set
+                                   (Q3: orders ⋉ building-customers) */
+    int32_t semijoin_col;
+    int32_t n_payload;
+    int32_t payload_col[4];     /* i32/i64 payload channels stored per row */
+    int64_t capacity_hint;      /* expected distinct build rows */
+    int32_t key_set_only;       /* 1: build a key SET (no payload slots) */
+} pg_plan_hash_build;
+
+typedef struct {
+    int64_t table;    /* handle from a finished PG_OP_HASH_BUILD */
+    int32_t n_preds;
+    pg_pred preds[PG_MAX_PRED]; /* fused pre-filter on probe input */
+    int32_t key_col;
+    /* mode 0: emit matched rows — output page =
+     *   probe columns emit_probe_cols[] + build payloads (join emit order:
+     *   probe rows ascending; within a probe row, chain head first —
+     *   LookupJoinPageBuilder.appendRow:75 + ArrayPositionLinks order)
+     * mode 1: fused grouped SUM into the build table (group = join key):
+     *   for each match, table.acc += proj(probe row) in exact decimal ticks
+     *   AND exact-f64 fixed-point (fixed128.h); get_output after finish
+     *   emits the groups page: key i64, payloads..., sum_dec i64,
+     *   sum_f64 f64 (rows = groups with >=1 match) */
+    int32_t mode;
+    int32_t n_emit;
+    int32_t emit_probe_cols[8];
+    pg_proj proj;
+    int32_t dec_scale;
+} pg_plan_lookup_join;
+
+typedef struct {
+    /* ORDER BY value DESC, date ASC, key ASC LIMIT n  (Q3 shape) */
+    int32_t limit;
+    int32_t val_col;  /* i64 (decimal ticks) or f64 */
+    int32_t date_col; /* i32 */
+    int32_t key_col;  /* i64 */
+} pg_plan_topn;
+
+typedef struct {
+    int32_t n_partitions;
+    int32_t key_col; /* bigint; rawHash = CombineHashFunction fold from
+                        INITIAL_HASH_VALUE=0 of AbstractLongType.hash
+                        (PlannerUtils.java:113, AbstractLongType.java:137-140);
+                        partition = HashGenerator.java:22-29 */
+    int32_t n_emit;
+    int32_t emit_cols[8];
+} pg_plan_partition;
+
+/* ---- operator lifecycle (Operator.java:20-102 analog) ---- */
+typedef int64_t pg_op;
+
+pg_status pg_op_create(int32_t kind, const void* plan, int64_t plan_bytes,
+                       pg_op* out);
+/* needsInput() */
+int32_t pg_op_needs_input(pg_op op);
+/* addInput(Page) — page borrowed for the call */
+pg_status pg_op_add_input(pg_op op, const pg_page* page);
+/* getOutput() — *out set to an internal page (device pointers) or NULL */
+pg_status pg_op_get_output(pg_op op, const pg_page** out);
+/* finish() */
+pg_status pg_op_finish(pg_op op);
+/* isFinished() */
+int32_t pg_op_is_finished(pg_op op);
+pg_status pg_op_destroy(pg_op op);
+
+/* table handle of a finished HASH_BUILD op
+ * (lendPartitionLookupSource analog, HashBuilderOperator.java:534) */
+pg_status pg_op_table(pg_op op, int64_t* out_table);
+/* per-partition row counts after a PARTITION op's get_output */
+pg_status pg_op_partition_counts(pg_op op, int64_t* counts, int32_t n);
+
+/* destroy a table explicitly (tables outlive their build op until freed) */
+pg_status pg_table_destroy(int64_t table);
+
+#ifdef __cplusplus
+}
+#endif
+#endif

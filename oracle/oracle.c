@@ -7,14 +7,29 @@
 #include <string.h>
 
 /* ---------------- fixed-tree f64 schedule (must match GPU kernels) ----
- * Virtual lanes: VL = NBLOCKS*NTHREADS.  Lane v owns row pairs
- * {2v, 2v+1} + k*2*VL.  Within a lane: sequential accumulation in row order.
- * Reduction: within block, pairwise stride-halving over the 256 lane
- * partials (for s=128..1: p[i] += p[i+s]); then the same stride-halving over
- * the NBLOCKS block partials.  Documented in DESIGN.md §determinism. */
+ * Mirrors k_agg_small / k_agg_small_finish in presto_amd/csrc/kernels.hip:
+ *  - virtual lane v = block*256 + thread owns row pairs {2v,2v+1} + k*2*VL,
+ *    accumulated sequentially in row order;
+ *  - wave reduction: xor-butterfly over each 64-lane group
+ *    (arr[i] += arr[i^s] simultaneously, s = 32,16,8,4,2,1; take index 0);
+ *  - the 4 wave sums reduce by butterfly s=2,1 (take index 0) -> block
+ *    partial (accumulated += across pages);
+ *  - grid reduction: lane l in 0..63 sums block partials l, l+64, ...
+ *    ascending, then one 64-wide butterfly.
+ * Documented in DESIGN.md §determinism. */
 #define FT_NBLOCKS 4096
 #define FT_NTHREADS 256
 #define FT_VL ((int64_t)FT_NBLOCKS * FT_NTHREADS)
+
+/* in-place xor-butterfly: a[i] = a[i] + a[i^s] for all i, s = n/2 .. 1 */
+static void bfly(double* a, int n)
+{
+    double tmp[64];
+    for (int s = n / 2; s >= 1; s >>= 1) {
+        for (int i = 0; i < n; i++) tmp[i] = a[i] + a[i ^ s];
+        for (int i = 0; i < n; i++) a[i] = tmp[i];
+    }
+}
 
 /* ---------------- Q1 ---------------- */
 
@@ -75,12 +90,10 @@ int32_t oracle_q1(int64_t n, const double* qty, const double* eprice,
         }
     }
 
-    /* ---- f64 mode: fixed-tree deterministic schedule ---- */
-    static double* blockp = 0; /* FT_NBLOCKS * NKEY * 5 */
+    /* ---- f64 mode: deterministic butterfly schedule (see header) ---- */
     double* bp = (double*)malloc((size_t)FT_NBLOCKS * NKEY * 5 * sizeof(double));
 #pragma omp parallel for schedule(static)
     for (int b = 0; b < FT_NBLOCKS; b++) {
-        /* lane partials for this block */
         double lane[FT_NTHREADS][NKEY][5];
         memset(lane, 0, sizeof(lane));
         for (int t = 0; t < FT_NTHREADS; t++) {
@@ -99,20 +112,35 @@ int32_t oracle_q1(int64_t n, const double* qty, const double* eprice,
                 }
             }
         }
-        for (int s = FT_NTHREADS / 2; s >= 1; s >>= 1)
-            for (int t = 0; t < s; t++)
-                for (int k = 0; k < NKEY; k++)
-                    for (int j = 0; j < 5; j++)
-                        lane[t][k][j] += lane[t + s][k][j];
-        memcpy(&bp[(size_t)b * NKEY * 5], &lane[0][0][0],
-               NKEY * 5 * sizeof(double));
+        for (int k = 0; k < NKEY; k++)
+            for (int j = 0; j < 5; j++) {
+                double wsum[4];
+                for (int w = 0; w < 4; w++) {
+                    double arr[64];
+                    for (int l = 0; l < 64; l++)
+                        arr[l] = lane[64 * w + l][k][j];
+                    bfly(arr, 64);
+                    wsum[w] = arr[0];
+                }
+                bfly(wsum, 4);
+                bp[((size_t)b * NKEY + k) * 5 + j] = wsum[0];
+            }
     }
-    /* stride-halving over block partials */
-    for (int s = FT_NBLOCKS / 2; s >= 1; s >>= 1)
-        for (int b = 0; b < s; b++)
-            for (int j = 0; j < NKEY * 5; j++)
-                bp[(size_t)b * NKEY * 5 + j] += bp[(size_t)(b + s) * NKEY * 5 + j];
-    (void)blockp;
+    /* grid reduce: lane l sums blocks l, l+64, ... then 64-wide butterfly */
+    double grid_out[NKEY * 5];
+    for (int k = 0; k < NKEY; k++)
+        for (int j = 0; j < 5; j++) {
+            double g[64];
+            for (int l = 0; l < 64; l++) {
+                double s = 0;
+                for (int m = 0; m < FT_NBLOCKS / 64; m++)
+                    s += bp[((size_t)(l + 64 * m) * NKEY + k) * 5 + j];
+                g[l] = s;
+            }
+            bfly(g, 64);
+            grid_out[k * 5 + j] = g[0];
+        }
+    memcpy(bp, grid_out, sizeof(grid_out));
 
     /* emit groups sorted by (returnflag, linestatus) == key order */
     static const uint8_t RF[3] = {'A', 'N', 'R'}, LS[2] = {'F', 'O'};

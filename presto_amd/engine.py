@@ -1,0 +1,258 @@
+"""ctypes binding of libpresto_gpu.so (the product C-ABI).
+
+Mirrors include/presto_gpu.h exactly; see that header for the contract and
+the reference citations.
+"""
+import ctypes as C
+import os
+import pathlib
+
+import numpy as np
+
+_HERE = pathlib.Path(__file__).resolve().parent
+_SO = _HERE / "libpresto_gpu.so"
+
+# ---- enums (presto_gpu.h) ----
+T_U8, T_I32, T_I64, T_F64 = 0, 1, 2, 3
+CMP_LT, CMP_LE, CMP_GT, CMP_GE, CMP_EQ, CMP_NE = range(6)
+PROJ_IDENT, PROJ_DISC_PRICE, PROJ_CHARGE = range(3)
+AGG_COUNT, AGG_SUM_F64, AGG_SUM_DEC, AGG_SUM_I64 = range(4)
+(OP_FILTER_PROJECT, OP_HASH_AGG_SMALL, OP_HASH_BUILD, OP_LOOKUP_JOIN,
+ OP_TOPN, OP_PARTITION) = range(1, 7)
+
+_NP_TAG = {np.dtype(np.uint8): T_U8, np.dtype(np.int32): T_I32,
+           np.dtype(np.int64): T_I64, np.dtype(np.float64): T_F64}
+_TAG_NP = {v: k for k, v in _NP_TAG.items()}
+
+
+class PgCol(C.Structure):
+    _fields_ = [("tag", C.c_int32), ("on_device", C.c_int32),
+                ("data", C.c_void_p), ("null_mask", C.c_void_p)]
+
+
+class PgPage(C.Structure):
+    _fields_ = [("n_rows", C.c_int64), ("n_cols", C.c_int32),
+                ("cols", PgCol * 16)]
+
+
+class Pred(C.Structure):
+    _fields_ = [("col", C.c_int32), ("op", C.c_int32), ("ival", C.c_int64),
+                ("dval", C.c_double)]
+
+
+class Proj(C.Structure):
+    _fields_ = [("kind", C.c_int32), ("a", C.c_int32), ("b", C.c_int32),
+                ("c", C.c_int32)]
+
+
+class Agg(C.Structure):
+    _fields_ = [("func", C.c_int32), ("proj", Proj), ("dec_scale", C.c_int32)]
+
+
+class PlanFilterProject(C.Structure):
+    _fields_ = [("n_preds", C.c_int32), ("preds", Pred * 8),
+                ("n_proj", C.c_int32), ("proj", Proj * 16)]
+
+
+class PlanHashAggSmall(C.Structure):
+    _fields_ = [("n_preds", C.c_int32), ("preds", Pred * 8),
+                ("n_keys", C.c_int32), ("key_col", C.c_int32 * 2),
+                ("n_vals", C.c_int32 * 2), ("key_vals", (C.c_uint8 * 8) * 2),
+                ("n_aggs", C.c_int32), ("aggs", Agg * 8)]
+
+
+class PlanHashBuild(C.Structure):
+    _fields_ = [("n_preds", C.c_int32), ("preds", Pred * 8),
+                ("key_col", C.c_int32), ("semijoin_table", C.c_int64),
+                ("semijoin_col", C.c_int32), ("n_payload", C.c_int32),
+                ("payload_col", C.c_int32 * 4), ("capacity_hint", C.c_int64),
+                ("key_set_only", C.c_int32)]
+
+
+class PlanLookupJoin(C.Structure):
+    _fields_ = [("table", C.c_int64), ("n_preds", C.c_int32),
+                ("preds", Pred * 8), ("key_col", C.c_int32),
+                ("mode", C.c_int32), ("n_emit", C.c_int32),
+                ("emit_probe_cols", C.c_int32 * 8), ("proj", Proj),
+                ("dec_scale", C.c_int32)]
+
+
+class PlanTopN(C.Structure):
+    _fields_ = [("limit", C.c_int32), ("val_col", C.c_int32),
+                ("date_col", C.c_int32), ("key_col", C.c_int32)]
+
+
+class PlanPartition(C.Structure):
+    _fields_ = [("n_partitions", C.c_int32), ("key_col", C.c_int32),
+                ("n_emit", C.c_int32), ("emit_cols", C.c_int32 * 8)]
+
+
+class _Lib:
+    def __init__(self):
+        if not _SO.exists():
+            raise RuntimeError(
+                f"{_SO} not built — run __graft_entry__.build() first "
+                "(the product HIP library is required; no CPU fallback)")
+        self.c = C.CDLL(str(_SO))
+        self.c.pg_last_error.restype = C.c_char_p
+        self.c.pg_op_create.argtypes = [C.c_int32, C.c_void_p, C.c_int64,
+                                        C.POINTER(C.c_int64)]
+        self.c.pg_op_add_input.argtypes = [C.c_int64, C.POINTER(PgPage)]
+        self.c.pg_op_get_output.argtypes = [C.c_int64,
+                                            C.POINTER(C.POINTER(PgPage))]
+        for f in ("pg_op_finish", "pg_op_destroy", "pg_op_needs_input",
+                  "pg_op_is_finished"):
+            getattr(self.c, f).argtypes = [C.c_int64]
+        self.c.pg_op_table.argtypes = [C.c_int64, C.POINTER(C.c_int64)]
+        self.c.pg_op_partition_counts.argtypes = [C.c_int64,
+                                                  C.POINTER(C.c_int64),
+                                                  C.c_int32]
+        self.c.pg_table_destroy.argtypes = [C.c_int64]
+        self.c.pg_memcpy_d2h.argtypes = [C.c_void_p, C.c_void_p, C.c_int64]
+        self.c.pg_device_sync.argtypes = []
+
+    def err(self):
+        return (self.c.pg_last_error() or b"").decode()
+
+    def check(self, st, what):
+        if st != 0:
+            raise RuntimeError(f"{what}: {self.err()}")
+
+
+_lib = None
+
+
+def lib():
+    global _lib
+    if _lib is None:
+        _lib = _Lib()
+    return _lib
+
+
+class Page:
+    """A Presto Page: named columns backed by numpy (host) or torch-cuda
+    (device) arrays.  Column order is the channel order."""
+
+    def __init__(self, cols, n_rows=None):
+        self.names = list(cols.keys())
+        self.cols = cols
+        first = next(iter(cols.values()))
+        self.n_rows = n_rows if n_rows is not None else len(first)
+
+    def channel(self, name):
+        return self.names.index(name)
+
+    def to_c(self):
+        pg = PgPage()
+        pg.n_rows = self.n_rows
+        pg.n_cols = len(self.names)
+        for i, name in enumerate(self.names):
+            a = self.cols[name]
+            col = PgCol()
+            if isinstance(a, np.ndarray):
+                col.tag = _NP_TAG[a.dtype]
+                col.on_device = 0
+                col.data = a.ctypes.data
+            else:  # torch tensor on cuda
+                import torch
+                assert isinstance(a, torch.Tensor) and a.is_cuda
+                tagmap = {torch.uint8: T_U8, torch.int32: T_I32,
+                          torch.int64: T_I64, torch.float64: T_F64}
+                col.tag = tagmap[a.dtype]
+                col.on_device = 1
+                col.data = a.data_ptr()
+            col.null_mask = None
+            pg.cols[i] = col
+        return pg
+
+
+def _read_output_page(cpage, names=None):
+    """Copy an output pg_page (host or device cols) into numpy arrays."""
+    L = lib()
+    n = cpage.n_rows
+    out = {}
+    for i in range(cpage.n_cols):
+        col = cpage.cols[i]
+        dt = _TAG_NP[col.tag]
+        a = np.empty(n, dt)
+        nbytes = n * dt.itemsize
+        if nbytes:
+            if col.on_device:
+                L.check(L.c.pg_memcpy_d2h(a.ctypes.data, col.data, nbytes),
+                        "d2h")
+            else:
+                C.memmove(a.ctypes.data, col.data, nbytes)
+        out[names[i] if names else f"c{i}"] = a
+    return out
+
+
+class Operator:
+    """Host mirror of operator/Operator.java:20-102 over the C-ABI."""
+
+    def __init__(self, kind, plan):
+        L = lib()
+        h = C.c_int64()
+        L.check(L.c.pg_op_create(kind, C.byref(plan), C.sizeof(plan),
+                                 C.byref(h)), "op_create")
+        self.h = h.value
+        self.kind = kind
+
+    def needs_input(self):
+        return bool(lib().c.pg_op_needs_input(self.h))
+
+    def add_input(self, page: Page):
+        L = lib()
+        cp = page.to_c()
+        L.check(L.c.pg_op_add_input(self.h, C.byref(cp)), "add_input")
+
+    def add_input_raw(self, cpage: PgPage):
+        """Feed another operator's raw output page (device pointers pass
+        through without copies — pages are borrowed for the call)."""
+        L = lib()
+        L.check(L.c.pg_op_add_input(self.h, C.byref(cpage)), "add_input")
+
+    def get_output(self, names=None):
+        """Returns dict of numpy arrays, or None. (Materializes device
+        outputs to host for harness use; device-to-device chaining uses
+        get_output_raw.)"""
+        L = lib()
+        pp = C.POINTER(PgPage)()
+        L.check(L.c.pg_op_get_output(self.h, C.byref(pp)), "get_output")
+        if not pp:
+            return None
+        return _read_output_page(pp.contents, names)
+
+    def get_output_raw(self):
+        """Returns the raw PgPage (device pointers stay on device) or None.
+        Valid until the next get_output*/destroy on this operator."""
+        L = lib()
+        pp = C.POINTER(PgPage)()
+        L.check(L.c.pg_op_get_output(self.h, C.byref(pp)), "get_output")
+        return pp.contents if pp else None
+
+    def finish(self):
+        lib().check(lib().c.pg_op_finish(self.h), "finish")
+
+    def is_finished(self):
+        return bool(lib().c.pg_op_is_finished(self.h))
+
+    def table(self):
+        t = C.c_int64()
+        lib().check(lib().c.pg_op_table(self.h, C.byref(t)), "table")
+        return t.value
+
+    def partition_counts(self, n):
+        arr = (C.c_int64 * n)()
+        lib().check(lib().c.pg_op_partition_counts(self.h, arr, n),
+                    "partition_counts")
+        return list(arr)
+
+    def destroy(self):
+        lib().c.pg_op_destroy(self.h)
+
+    def __del__(self):
+        try:
+            if _lib is not None:
+                _lib.c.pg_op_destroy(self.h)
+        except Exception:
+            pass

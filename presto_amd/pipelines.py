@@ -1,0 +1,168 @@
+"""Query pipelines assembled from the C-ABI operators — the host analog of
+LocalExecutionPlanner wiring OperatorFactories into a Driver
+(presto-main-base/.../sql/planner/LocalExecutionPlanner.java:1663,2552,3648
+for the ScanFilterAndProject / HashBuilder / HashAggregation factories) and
+of the hand-wired pipelines in
+presto-benchmark/.../HandTpchQuery1.java:30-110.
+
+Date constants: epoch-day literals from the benchmark SQL
+(presto-benchto-benchmarks/.../tpch/q01.sql, q03.sql):
+  Q1: shipdate <= DATE '1998-12-01' - 90 days = 1998-09-02 = 10471
+  Q3: DATE '1995-03-15' = 9204
+"""
+import ctypes as C
+
+from .engine import (
+    Operator, Page, PlanHashAggSmall, PlanHashBuild, PlanLookupJoin,
+    PlanTopN, PlanPartition, Pred, Proj, Agg,
+    CMP_LE, CMP_LT, CMP_GT, CMP_EQ,
+    PROJ_IDENT, PROJ_DISC_PRICE, PROJ_CHARGE,
+    AGG_COUNT, AGG_SUM_F64, AGG_SUM_DEC,
+    OP_HASH_AGG_SMALL, OP_HASH_BUILD, OP_LOOKUP_JOIN, OP_TOPN, OP_PARTITION,
+)
+
+Q1_SHIP_MAX = 10471
+Q3_DATE = 9204
+
+
+def q1_plan(page: Page, mode="f64"):
+    p = PlanHashAggSmall()
+    p.n_preds = 1
+    p.preds[0] = Pred(page.channel("shipdate"), CMP_LE, Q1_SHIP_MAX, 0.0)
+    p.n_keys = 2
+    p.key_col[0] = page.channel("returnflag")
+    p.key_col[1] = page.channel("linestatus")
+    p.n_vals[0] = 3
+    p.n_vals[1] = 2
+    for j, v in enumerate(b"ANR"):
+        p.key_vals[0][j] = v
+    for j, v in enumerate(b"FO"):
+        p.key_vals[1][j] = v
+    qty = page.channel("quantity")
+    ep = page.channel("extendedprice")
+    dc = page.channel("discount")
+    tx = page.channel("tax")
+    sums = [
+        (Proj(PROJ_IDENT, qty, 0, 0), 0),
+        (Proj(PROJ_IDENT, ep, 0, 0), 2),
+        (Proj(PROJ_DISC_PRICE, ep, dc, 0), 4),
+        (Proj(PROJ_CHARGE, ep, dc, tx), 6),
+        (Proj(PROJ_IDENT, dc, 0, 0), 2),
+    ]
+    func = AGG_SUM_DEC if mode == "dec" else AGG_SUM_F64
+    p.n_aggs = 6
+    for i, (proj, scale) in enumerate(sums):
+        p.aggs[i] = Agg(func, proj, scale)
+    p.aggs[5] = Agg(AGG_COUNT, Proj(PROJ_IDENT, 0, 0, 0), 0)
+    return p
+
+
+Q1_F64_NAMES = ["returnflag", "linestatus", "sum_qty", "sum_base",
+                "sum_disc_price", "sum_charge", "sum_disc", "count"]
+Q1_DEC_NAMES = ["returnflag", "linestatus",
+                "sum_qty_hi", "sum_qty_lo", "sum_base_hi", "sum_base_lo",
+                "sum_disc_price_hi", "sum_disc_price_lo",
+                "sum_charge_hi", "sum_charge_lo",
+                "sum_disc_hi", "sum_disc_lo", "count"]
+
+
+def q1(page: Page, mode="f64"):
+    """Full Q1 over one lineitem page. Returns dict of numpy group columns
+    (groups in (returnflag, linestatus) order)."""
+    op = Operator(OP_HASH_AGG_SMALL, q1_plan(page, mode))
+    try:
+        op.add_input(page)
+        op.finish()
+        return op.get_output(Q1_F64_NAMES if mode == "f64" else Q1_DEC_NAMES)
+    finally:
+        op.destroy()
+
+
+class Q3Pipeline:
+    """Q3 operator graph, reusable across inputs (tables freed on close).
+
+    customer(BUILDING) -> key set          [HashBuilderOperator analog]
+    orders(date<9204) semijoin set -> tbl  [HashBuilder w/ fused filter]
+    lineitem(date>9204) probe tbl          [LookupJoin + fused grouped SUM]
+    groups -> TopN 10 (rev desc, odate asc, okey asc)
+    """
+
+    def __init__(self, cust: Page, orders: Page, mode="dec", limit=10):
+        self.mode = mode
+        self.limit = limit
+        b1p = PlanHashBuild()
+        b1p.n_preds = 1
+        b1p.preds[0] = Pred(cust.channel("mktseg"), CMP_EQ, 1, 0.0)
+        b1p.key_col = cust.channel("custkey")
+        b1p.semijoin_table = -1
+        b1p.n_payload = 0
+        b1p.capacity_hint = max(cust.n_rows // 4, 16)
+        b1p.key_set_only = 1
+        self.b1 = Operator(OP_HASH_BUILD, b1p)
+        self.b1.add_input(cust)
+        self.b1.finish()
+        self.set_tbl = self.b1.table()
+
+        b2p = PlanHashBuild()
+        b2p.n_preds = 1
+        b2p.preds[0] = Pred(orders.channel("orderdate"), CMP_LT, Q3_DATE, 0.0)
+        b2p.key_col = orders.channel("orderkey")
+        b2p.semijoin_table = self.set_tbl
+        b2p.semijoin_col = orders.channel("custkey")
+        b2p.n_payload = 1
+        b2p.payload_col[0] = orders.channel("orderdate")
+        b2p.capacity_hint = max(orders.n_rows // 8, 16)
+        b2p.key_set_only = 0
+        self.b2 = Operator(OP_HASH_BUILD, b2p)
+        self.b2.add_input(orders)
+        self.b2.finish()
+        self.tbl = self.b2.table()
+
+    def run(self, lineitem: Page):
+        jp = PlanLookupJoin()
+        jp.table = self.tbl
+        jp.n_preds = 1
+        jp.preds[0] = Pred(lineitem.channel("shipdate"), CMP_GT, Q3_DATE, 0.0)
+        jp.key_col = lineitem.channel("orderkey")
+        jp.mode = 1
+        jp.proj = Proj(PROJ_DISC_PRICE, lineitem.channel("extendedprice"),
+                       lineitem.channel("discount"), 0)
+        jp.dec_scale = 4
+        j = Operator(OP_LOOKUP_JOIN, jp)
+        try:
+            j.add_input(lineitem)
+            j.finish()
+            groups = j.get_output_raw()
+            # groups page cols: [orderkey, orderdate, sum_dec, sum_f64, cnt]
+            tp = PlanTopN()
+            tp.limit = self.limit
+            tp.val_col = 2 if self.mode == "dec" else 3
+            tp.date_col = 1
+            tp.key_col = 0
+            t = Operator(OP_TOPN, tp)
+            try:
+                t.add_input_raw(groups)
+                t.finish()
+                names = ["orderkey",
+                         "revenue_1e4" if self.mode == "dec" else "revenue",
+                         "orderdate"]
+                return t.get_output(names)
+            finally:
+                t.destroy()
+        finally:
+            j.destroy()
+
+    def close(self):
+        from .engine import lib
+        lib().c.pg_table_destroy(self.set_tbl)
+        lib().c.pg_table_destroy(self.tbl)
+        self.b1.destroy()
+        self.b2.destroy()
+
+
+def q3(cust: Page, orders: Page, lineitem: Page, mode="dec", limit=10):
+    p = Q3Pipeline(cust, orders, mode=mode, limit=limit)
+    try:
+        return p.run(lineitem)
+    finally:
+        p.close()

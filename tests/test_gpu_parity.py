@@ -1,0 +1,303 @@
+"""GPU parity tests — every test compares HIP kernels (through the C-ABI)
+against the CPU oracle on identical inputs.  All marked gpu.
+
+Bar (BASELINE.json north_star): bit-exact for integer/hash/decimal results;
+double SUM/AVG bit-exact here because both sides run the same deterministic
+schedule (fixed butterfly tree for Q1, exact 64.64 fixed-point for Q3 —
+strictly tighter than the stated <=1 ulp budget).
+"""
+import numpy as np
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def P():
+    import presto_amd
+    return presto_amd
+
+
+@pytest.fixture(scope="session")
+def sf01(oracle_lib):
+    return dict(li=oracle_lib.gen_lineitem(0.1),
+                orders=oracle_lib.gen_orders(0.1),
+                cust=oracle_lib.gen_customer(0.1))
+
+
+def _li_page(P, li):
+    return P.Page({k: li[k] for k in
+                   ("quantity", "extendedprice", "discount", "tax",
+                    "shipdate", "returnflag", "linestatus", "orderkey")})
+
+
+def test_q1_decimal_exact(P, oracle_lib, sf01):
+    got = P.pipelines.q1(_li_page(P, sf01["li"]), mode="dec")
+    exp = oracle_lib.q1(sf01["li"])
+    assert len(exp) == len(got["returnflag"])
+    for i, g in enumerate(exp):
+        assert got["returnflag"][i] == g.returnflag
+        assert got["linestatus"][i] == g.linestatus
+        assert got["count"][i] == g.count_order
+        assert got["sum_qty_lo"][i] == g.sum_qty_units
+        assert got["sum_base_lo"][i] == g.sum_base_cents
+        assert got["sum_disc_price_lo"][i] == g.sum_disc_1e4
+        assert got["sum_charge_hi"][i] == g.sum_charge_1e6_hi
+        assert got["sum_charge_lo"][i].astype(np.uint64) == np.uint64(
+            g.sum_charge_1e6_lo)
+        assert got["sum_disc_lo"][i] == g.sum_disc_cents
+
+
+def test_q1_f64_bit_exact(P, oracle_lib, sf01):
+    got = P.pipelines.q1(_li_page(P, sf01["li"]), mode="f64")
+    exp = oracle_lib.q1(sf01["li"])
+    for i, g in enumerate(exp):
+        # bitwise equality of doubles
+        for name, val in (("sum_qty", g.f64_sum_qty),
+                          ("sum_base", g.f64_sum_base),
+                          ("sum_disc_price", g.f64_sum_disc_price),
+                          ("sum_charge", g.f64_sum_charge),
+                          ("sum_disc", g.f64_sum_disc)):
+            assert got[name][i].view(np.int64) == np.float64(val).view(
+                np.int64), (name, got[name][i], val)
+        assert got["count"][i] == g.count_order
+
+
+def test_q1_device_resident_input(P, oracle_lib, sf01):
+    """Columns already in HBM (torch tensors) — the bench path."""
+    import torch
+    li = sf01["li"]
+    cols = {k: torch.from_numpy(li[k]).cuda() for k in
+            ("quantity", "extendedprice", "discount", "tax", "shipdate",
+             "returnflag", "linestatus")}
+    got = P.pipelines.q1(P.Page(cols), mode="dec")
+    exp = oracle_lib.q1(li)
+    for i, g in enumerate(exp):
+        assert got["count"][i] == g.count_order
+        assert got["sum_disc_price_lo"][i] == g.sum_disc_1e4
+
+
+def test_q3_exact(P, oracle_lib, sf01):
+    cust, orders, li = sf01["cust"], sf01["orders"], sf01["li"]
+    got = P.pipelines.q3(
+        P.Page({"custkey": cust["custkey"], "mktseg": cust["mktseg"]}),
+        P.Page({k: orders[k] for k in ("orderkey", "custkey", "orderdate")}),
+        _li_page(P, li), mode="dec")
+    exp = oracle_lib.q3(cust, orders, li)
+    assert len(got["orderkey"]) == len(exp)
+    for i, r in enumerate(exp):
+        assert got["orderkey"][i] == r.orderkey
+        assert got["revenue_1e4"][i] == r.revenue_1e4
+        assert got["orderdate"][i] == r.orderdate
+
+
+def test_q3_f64_mode(P, oracle_lib, sf01):
+    """f64 revenue: exact fixed-point sum of f64 products — bit-equal to the
+    oracle's fx128 accumulation."""
+    cust, orders, li = sf01["cust"], sf01["orders"], sf01["li"]
+    got = P.pipelines.q3(
+        P.Page({"custkey": cust["custkey"], "mktseg": cust["mktseg"]}),
+        P.Page({k: orders[k] for k in ("orderkey", "custkey", "orderdate")}),
+        _li_page(P, li), mode="f64")
+    exp = oracle_lib.q3(cust, orders, li)
+    for i, r in enumerate(exp):
+        assert got["orderkey"][i] == r.orderkey
+        assert got["revenue"][i].view(np.int64) == np.float64(
+            r.f64_revenue).view(np.int64)
+
+
+# ---------------- operator-level parity ----------------
+
+def test_filter_project_stable(P):
+    rng = np.random.default_rng(3)
+    n = 100_000
+    a = rng.integers(0, 1000, n).astype(np.int64)
+    b = rng.random(n)
+    d = rng.integers(0, 100, n).astype(np.int32)
+    page = P.Page({"a": a, "b": b, "d": d})
+    plan = P.PlanFilterProject()
+    plan.n_preds = 2
+    plan.preds[0] = P.Pred(0, P.CMP_LT, 500, 0.0)
+    plan.preds[1] = P.Pred(2, P.CMP_GE, 10, 0.0)
+    plan.n_proj = 3
+    plan.proj[0] = P.Proj(P.PROJ_IDENT, 0, 0, 0)
+    plan.proj[1] = P.Proj(P.PROJ_IDENT, 1, 0, 0)
+    plan.proj[2] = P.Proj(P.PROJ_DISC_PRICE, 1, 1, 0)
+    op = P.Operator(P.OP_FILTER_PROJECT, plan)
+    op.add_input(page)
+    out = op.get_output(["a", "b", "dp"])
+    op.destroy()
+    sel = (a < 500) & (d >= 10)
+    # stable row-ascending selection (PageProcessor SelectedPositions order)
+    assert np.array_equal(out["a"], a[sel])
+    assert np.array_equal(out["b"], b[sel])
+    assert np.array_equal(out["dp"], (b * (1.0 - b))[sel])
+
+
+def test_filter_empty_and_full(P):
+    n = 10_000
+    a = np.arange(n, dtype=np.int64)
+    page = P.Page({"a": a})
+    for op_cmp, ival, expect in ((P.CMP_LT, 0, 0), (P.CMP_GE, 0, n)):
+        plan = P.PlanFilterProject()
+        plan.n_preds = 1
+        plan.preds[0] = P.Pred(0, op_cmp, ival, 0.0)
+        plan.n_proj = 1
+        plan.proj[0] = P.Proj(P.PROJ_IDENT, 0, 0, 0)
+        op = P.Operator(P.OP_FILTER_PROJECT, plan)
+        op.add_input(page)
+        out = op.get_output(["a"])
+        op.destroy()
+        assert len(out["a"]) == expect
+
+
+def test_join_emit_unique_keys(P, oracle_lib):
+    rng = np.random.default_rng(4)
+    bkeys = rng.permutation(50_000)[:20_000].astype(np.int64)
+    pkeys = rng.integers(0, 60_000, 100_000).astype(np.int64)
+    payload = (bkeys * 3).astype(np.int64)
+    bplan = P.PlanHashBuild()
+    bplan.key_col = 0
+    bplan.semijoin_table = -1
+    bplan.n_payload = 1
+    bplan.payload_col[0] = 1
+    bplan.capacity_hint = len(bkeys)
+    b = P.Operator(P.OP_HASH_BUILD, bplan)
+    b.add_input(P.Page({"k": bkeys, "p": payload}))
+    b.finish()
+    jplan = P.PlanLookupJoin()
+    jplan.table = b.table()
+    jplan.key_col = 0
+    jplan.mode = 0
+    jplan.n_emit = 2
+    jplan.emit_probe_cols[0] = 0
+    jplan.emit_probe_cols[1] = 1
+    j = P.Operator(P.OP_LOOKUP_JOIN, jplan)
+    j.add_input(P.Page({"k": pkeys, "i": np.arange(len(pkeys), dtype=np.int64)}))
+    out = j.get_output(["k", "i", "bp"])
+    j.destroy()
+    op_idx, ob_idx = oracle_lib.join(bkeys, pkeys)
+    assert np.array_equal(out["i"], op_idx)  # probe-ascending emit order
+    assert np.array_equal(out["k"], pkeys[op_idx])
+    assert np.array_equal(out["bp"], payload[ob_idx])
+    from presto_amd.engine import lib
+    lib().c.pg_table_destroy(jplan.table)
+    b.destroy()
+
+
+def test_join_emit_duplicate_keys(P, oracle_lib):
+    """Duplicate build keys: emitted pair SET must match the oracle
+    (chain order across parallel inserts is not deterministic — result-set
+    semantics)."""
+    rng = np.random.default_rng(5)
+    bkeys = rng.integers(0, 500, 5_000).astype(np.int64)
+    pkeys = rng.integers(0, 600, 8_000).astype(np.int64)
+    bplan = P.PlanHashBuild()
+    bplan.key_col = 0
+    bplan.semijoin_table = -1
+    bplan.n_payload = 1
+    bplan.payload_col[0] = 1
+    bplan.capacity_hint = len(bkeys)
+    rowid = np.arange(len(bkeys), dtype=np.int64)
+    b = P.Operator(P.OP_HASH_BUILD, bplan)
+    b.add_input(P.Page({"k": bkeys, "row": rowid}))
+    b.finish()
+    jplan = P.PlanLookupJoin()
+    jplan.table = b.table()
+    jplan.key_col = 0
+    jplan.mode = 0
+    jplan.n_emit = 1
+    jplan.emit_probe_cols[0] = 1
+    j = P.Operator(P.OP_LOOKUP_JOIN, jplan)
+    j.add_input(P.Page({"k": pkeys, "pi": np.arange(len(pkeys), dtype=np.int64)}))
+    out = j.get_output(["pi", "brow"])
+    j.destroy()
+    op_idx, ob_idx = oracle_lib.join(bkeys, pkeys)
+    got = set(zip(out["pi"].tolist(), out["brow"].tolist()))
+    exp = set(zip(op_idx.tolist(), ob_idx.tolist()))
+    assert got == exp
+    from presto_amd.engine import lib
+    lib().c.pg_table_destroy(jplan.table)
+    b.destroy()
+
+
+def test_partition_math_and_stability(P, oracle_lib):
+    rng = np.random.default_rng(6)
+    n = 200_000
+    keys = rng.integers(-10**12, 10**12, n).astype(np.int64)
+    vals = rng.random(n)
+    for nparts in (2, 8):
+        plan = P.PlanPartition()
+        plan.n_partitions = nparts
+        plan.key_col = 0
+        plan.n_emit = 2
+        plan.emit_cols[0] = 0
+        plan.emit_cols[1] = 1
+        op = P.Operator(P.OP_PARTITION, plan)
+        op.add_input(P.Page({"k": keys, "v": vals}))
+        counts = op.partition_counts(nparts)
+        # expected partition ids via the oracle's replicated reference math
+        pid = np.array([oracle_lib.lib.oracle_partition(
+            oracle_lib.lib.oracle_bigint_hash(int(k)), nparts)
+            for k in keys[:2000]])
+        pages = []
+        for p in range(nparts):
+            pages.append(op.get_output([f"k", f"v"]))
+        op.destroy()
+        assert sum(counts) == n
+        # full reference: partition ids via the oracle's replicated math,
+        # stable within partition (row-ascending) like the reference's
+        # per-partition position lists
+        h = np.array([oracle_lib.lib.oracle_bigint_hash(int(k))
+                      for k in keys.tolist()], dtype=np.uint64)
+        pid_all = np.array([oracle_lib.lib.oracle_partition(int(x), nparts)
+                            for x in h.tolist()])
+        for p in range(nparts):
+            sel = pid_all == p
+            assert len(pages[p]["k"]) == counts[p] == int(sel.sum())
+            assert np.array_equal(pages[p]["k"], keys[sel])
+            assert np.array_equal(pages[p]["v"], vals[sel])
+        assert pid[0] == pid_all[0]
+
+
+def test_topn_matches_numpy(P):
+    rng = np.random.default_rng(7)
+    n = 500_000
+    val = rng.integers(0, 10**9, n).astype(np.int64)
+    date = rng.integers(8000, 11000, n).astype(np.int32)
+    key = rng.permutation(n).astype(np.int64)
+    plan = P.PlanTopN()
+    plan.limit = 10
+    plan.val_col = 0
+    plan.date_col = 1
+    plan.key_col = 2
+    op = P.Operator(P.OP_TOPN, plan)
+    op.add_input(P.Page({"v": val, "d": date, "k": key}))
+    op.finish()
+    out = op.get_output(["k", "v", "d"])
+    op.destroy()
+    order = sorted(range(n), key=lambda i: (-val[i], date[i], key[i]))[:10]
+    assert out["k"].tolist() == [key[i] for i in order]
+    assert out["v"].tolist() == [val[i] for i in order]
+    assert out["d"].tolist() == [date[i] for i in order]
+
+
+def test_multi_page_decimal_agg(P, oracle_lib, sf01):
+    """Decimal sums are order/page-split independent: feeding the table in
+    4 pages must give identical results to one page."""
+    li = sf01["li"]
+    n = len(li["quantity"])
+    page_all = _li_page(P, li)
+    plan = P.pipelines.q1_plan(page_all, "dec")
+    op = P.Operator(P.OP_HASH_AGG_SMALL, plan)
+    cuts = [0, n // 4, n // 2, 3 * n // 4, n]
+    for a, b in zip(cuts, cuts[1:]):
+        sub = {k: v[a:b] for k, v in page_all.cols.items()}
+        op.add_input(P.Page(sub))
+    op.finish()
+    got = op.get_output(P.pipelines.Q1_DEC_NAMES)
+    op.destroy()
+    exp = oracle_lib.q1(li)
+    for i, g in enumerate(exp):
+        assert got["count"][i] == g.count_order
+        assert got["sum_disc_price_lo"][i] == g.sum_disc_1e4
