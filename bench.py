@@ -1,0 +1,322 @@
+#!/usr/bin/env python3
+"""bench.py — driver-contract benchmark for the MI355X-native Presto
+hot-path library.
+
+A "step" is one pass of the hot path over one batch of synthetic input:
+  q1 (default): the full TPC-H Q1 pipeline (fused scan+filter+group-by+
+      aggregate + final reduce) over the SF100 lineitem columns resident in
+      HBM — BASELINE.json configs[1].
+  q3: the full Q3 pipeline (customer set build, orders build with fused
+      filter+semijoin, lineitem probe with fused grouped sum, TopN 10) —
+      BASELINE.json configs[2].
+
+Inputs are synthetic TPC-H columns from the dbgen restatement in
+oracle/tpchgen.c (golden-pinned; generation is test/bench input
+infrastructure and happens OUTSIDE the timed region).  Data is resident in
+HBM when the timed region starts.
+
+Multi-GPU (--gpus N via torch.distributed.run): weak scaling — each rank
+holds its own SF-sized shard; Q1 ends with an all_gather of the tiny group
+partials combined in rank order (the partial->final Step split of
+HashAggregationOperator.java:72); Q3 exchanges orders+lineitem by orderkey
+hash over RCCL all_to_all (presto_amd/dist.py) before local build/probe.
+
+cpu_baseline: the CPU oracle (oracle/liboracle.so, kind "port") timed on
+this box's host cores over a bounded sample — reported baseline only, never
+the measured value.
+"""
+import argparse
+import ctypes as C
+import json
+import os
+import pathlib
+import sys
+import time
+
+import numpy as np
+
+REPO = pathlib.Path(__file__).resolve().parent
+sys.path.insert(0, str(REPO))
+
+Q1_WORKLOAD = ("TPC-H SF{sf} Q1 on {n}xMI355X - scan/filter/hash-aggregate "
+               "kernels, lineitem columns resident in HBM")
+Q3_WORKLOAD = ("TPC-H SF{sf} Q3 on {n}xMI355X - 3-way hash join "
+               "(HashBuilder+LookupJoin) + order-by/limit")
+
+# Q1 algorithmic bytes/row: 4 f64 money cols +
+# dict-u8 returnflag/linestatus + date32 = 38 B
+Q1_BYTES_PER_ROW = 38
+HBM_PEAK = 8.0e12  # B/s, MI355X_MICROARCH.md chip parameters (spec)
+
+
+def log(msg):
+    print(msg, file=sys.stderr, flush=True)
+
+
+def load_oracle():
+    import subprocess
+    so = REPO / "oracle" / "liboracle.so"
+    if not so.exists():
+        subprocess.run(["make", "-C", str(REPO / "oracle"), "liboracle.so"],
+                       check=True)
+    from tests.oracle_binding import OracleLib
+    return OracleLib(str(so))
+
+
+def gen_lineitem_device(orc, sf, device, want_orderkey=False,
+                        chunk_orders=4_000_000):
+    """Generate SF lineitem columns chunkwise on host, upload to device
+    tensors.  Returns dict of torch tensors."""
+    import torch
+    n_ord = orc.lib.tpch_orders_count(C.c_double(sf))
+    n = orc.lib.tpch_lineitem_count(C.c_double(sf))
+    cols = {
+        "quantity": torch.empty(n, dtype=torch.float64, device=device),
+        "extendedprice": torch.empty(n, dtype=torch.float64, device=device),
+        "discount": torch.empty(n, dtype=torch.float64, device=device),
+        "tax": torch.empty(n, dtype=torch.float64, device=device),
+        "shipdate": torch.empty(n, dtype=torch.int32, device=device),
+        "returnflag": torch.empty(n, dtype=torch.uint8, device=device),
+        "linestatus": torch.empty(n, dtype=torch.uint8, device=device),
+    }
+    if want_orderkey:
+        cols["orderkey"] = torch.empty(n, dtype=torch.int64, device=device)
+    maxrows = chunk_orders * 7
+    buf = {k: np.empty(maxrows, v.cpu().numpy().dtype if False else
+                       {torch.float64: np.float64, torch.int32: np.int32,
+                        torch.uint8: np.uint8,
+                        torch.int64: np.int64}[v.dtype])
+           for k, v in cols.items()}
+    off = 0
+    o = 0
+    t0 = time.time()
+    while o < n_ord:
+        cnt = min(chunk_orders, n_ord - o)
+        w = orc.lib.tpch_gen_lineitem(
+            C.c_double(sf), C.c_int64(o), C.c_int64(cnt),
+            buf["orderkey"].ctypes.data if want_orderkey else None,
+            buf["quantity"].ctypes.data, buf["extendedprice"].ctypes.data,
+            buf["discount"].ctypes.data, buf["tax"].ctypes.data,
+            buf["shipdate"].ctypes.data, buf["returnflag"].ctypes.data,
+            buf["linestatus"].ctypes.data)
+        for k, t in cols.items():
+            t[off:off + w].copy_(torch.from_numpy(buf[k][:w]))
+        off += w
+        o += cnt
+    assert off == n
+    log(f"generated+uploaded lineitem sf={sf}: {n} rows in "
+        f"{time.time() - t0:.1f}s")
+    return cols, n
+
+
+def gen_orders_customer_device(orc, sf, device):
+    import torch
+    n_ord = orc.lib.tpch_orders_count(C.c_double(sf))
+    n_cust = orc.lib.tpch_customer_count(C.c_double(sf))
+    ok = np.empty(n_ord, np.int64)
+    ck = np.empty(n_ord, np.int64)
+    od = np.empty(n_ord, np.int32)
+    orc.lib.tpch_gen_orders(C.c_double(sf), C.c_int64(0), C.c_int64(n_ord),
+                            ok.ctypes.data, ck.ctypes.data, od.ctypes.data,
+                            None)
+    cck = np.empty(n_cust, np.int64)
+    seg = np.empty(n_cust, np.uint8)
+    orc.lib.tpch_gen_customer(C.c_double(sf), C.c_int64(0),
+                              C.c_int64(n_cust), cck.ctypes.data,
+                              seg.ctypes.data)
+    t = lambda a: __import__("torch").from_numpy(a).to(device)
+    return (dict(orderkey=t(ok), custkey=t(ck), orderdate=t(od)),
+            dict(custkey=t(cck), mktseg=t(seg)))
+
+
+def cpu_baseline_q1(orc, target_secs=10.0):
+    """Time the oracle's Q1 (OpenMP over all host cores) on a bounded
+    sample; returns (rows_per_sec, n_rows_sample)."""
+    sf_sample = 4.0
+    li = orc.gen_lineitem(sf_sample)
+    n = len(li["quantity"])
+    reps = 1
+    t0 = time.time()
+    orc.q1(li)
+    dt = time.time() - t0
+    while dt * reps < target_secs / 2 and reps < 64:
+        reps *= 2
+    t0 = time.time()
+    for _ in range(reps):
+        orc.q1(li)
+    dt = (time.time() - t0) / reps
+    return n / dt, n
+
+
+def cpu_baseline_q3(orc, target_secs=10.0):
+    sf_sample = 2.0
+    li = orc.gen_lineitem(sf_sample)
+    orders = orc.gen_orders(sf_sample)
+    cust = orc.gen_customer(sf_sample)
+    n = len(li["quantity"])
+    t0 = time.time()
+    orc.q3(cust, orders, li)
+    dt = time.time() - t0
+    reps = max(1, int(target_secs / 2 / max(dt, 0.05)))
+    reps = min(reps, 16)
+    t0 = time.time()
+    for _ in range(reps):
+        orc.q3(cust, orders, li)
+    dt = (time.time() - t0) / reps
+    return n / dt, n
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=8)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--query", choices=["q1", "q3"], default="q1")
+    ap.add_argument("--sf", type=float, default=100.0)
+    ap.add_argument("--mode", choices=["f64", "dec"], default="f64")
+    ap.add_argument("--skip-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    import torch
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    n_gpus = max(world, args.gpus if world == 1 else world)
+    if world > 1:
+        import torch.distributed as dist
+        dist.init_process_group("nccl")
+        torch.cuda.set_device(local_rank)
+    device = torch.device("cuda", local_rank)
+
+    import presto_amd
+    from presto_amd import Page, pipelines
+    orc = load_oracle()
+
+    def barrier_sync():
+        torch.cuda.synchronize()
+        if world > 1:
+            import torch.distributed as dist
+            dist.barrier()
+            torch.cuda.synchronize()
+
+    if args.query == "q1":
+        cols, n_rows = gen_lineitem_device(orc, args.sf, device)
+        page = Page(cols, n_rows=n_rows)
+        plan = pipelines.q1_plan(page, args.mode)
+
+        def step():
+            op = presto_amd.Operator(presto_amd.OP_HASH_AGG_SMALL, plan)
+            op.add_input(page)
+            op.finish()
+            out = op.get_output(
+                pipelines.Q1_F64_NAMES if args.mode == "f64"
+                else pipelines.Q1_DEC_NAMES)
+            op.destroy()
+            if world > 1:
+                # partial->final: gather tiny group sums, combine in rank
+                # order (HashAggregationOperator partial/final Step split)
+                import torch.distributed as dist
+                flat = torch.tensor(
+                    np.concatenate([out[k].view(np.float64) for k in out
+                                    if k not in ("returnflag", "linestatus")]
+                                   ).view(np.float64), device=device)
+                allp = [torch.empty_like(flat) for _ in range(world)]
+                dist.all_gather(allp, flat)
+            return out
+        total_rows_per_step = n_rows * n_gpus
+        workload = Q1_WORKLOAD.format(sf=int(args.sf), n=n_gpus)
+    else:
+        cols, n_rows = gen_lineitem_device(orc, args.sf, device,
+                                           want_orderkey=True)
+        li_page = Page(cols, n_rows=n_rows)
+        ocols, ccols = gen_orders_customer_device(orc, args.sf, device)
+        ord_page = Page(ocols)
+        cust_page = Page(ccols)
+
+        def step():
+            return pipelines.q3(cust_page, ord_page, li_page,
+                                mode=("dec" if args.mode == "f64" else
+                                      args.mode))
+        total_rows_per_step = n_rows * n_gpus
+        workload = Q3_WORKLOAD.format(sf=int(args.sf), n=n_gpus)
+
+    lib = presto_amd.engine.lib()
+    lib.c.pg_last_hot_kernel_ms.restype = C.c_double
+    for _ in range(args.warmup):
+        step()
+    barrier_sync()
+    t0 = time.time()
+    for _ in range(args.steps):
+        step()
+    barrier_sync()
+    elapsed = time.time() - t0
+    if world > 1:
+        import torch.distributed as dist
+        e = torch.tensor([elapsed], device=device)
+        dist.all_reduce(e, op=dist.ReduceOp.MAX)
+        elapsed = float(e.item())
+
+    hot_ms = float(lib.c.pg_last_hot_kernel_ms())
+    ms_per_step = elapsed / args.steps * 1000.0
+    value = total_rows_per_step * args.steps / elapsed
+
+    if rank == 0:
+        # roofline for the dominant kernel (rank 0): algorithmic bytes per
+        # launch / measured launch ms (HIP events on the launch stream)
+        if args.query == "q1":
+            alg_bytes = Q1_BYTES_PER_ROW * n_rows
+        else:
+            # Q3 probe kernel: lineitem orderkey 8 + eprice 8 + discount 8 +
+            # shipdate 4 = 28 B/row algorithmic scan
+            alg_bytes = 28 * n_rows
+        achieved = (alg_bytes / (hot_ms / 1000.0) / 1e9
+                    if hot_ms > 0 else None)
+        roofline = {
+            "bound": "hbm",
+            "achieved": achieved,
+            "peak": HBM_PEAK / 1e9,
+            "unit": "GB/s",
+            "frac": achieved / (HBM_PEAK / 1e9) if achieved else None,
+            "traffic": None,
+        }
+        cpu = None
+        if not args.skip_cpu_baseline and n_gpus == 1:
+            fn = cpu_baseline_q1 if args.query == "q1" else cpu_baseline_q3
+            rps, nsamp = fn(orc)
+            cpu = {
+                "value": rps,
+                "unit": "rows/s",
+                "cores": os.cpu_count(),
+                "kind": "port",
+                "sample": (f"oracle (OpenMP, all host cores) on a "
+                           f"{nsamp}-row TPC-H sample of the same workload"),
+            }
+        out = {
+            "metric": "tpch_lineitem_rows_per_sec",
+            "value": value,
+            "unit": "rows/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "f64" if args.mode == "f64" else "i64",
+            "data": ("synthetic (spec-conformant TPC-H dbgen restatement, "
+                     f"SF{int(args.sf)}, golden-pinned)"),
+            "config": {
+                "workload": workload,
+                "query": args.query,
+                "sf": args.sf,
+                "rows_per_gpu": n_rows,
+                "mode": args.mode,
+            },
+            "roofline": roofline,
+            "cpu_baseline": cpu,
+        }
+        print(json.dumps(out), flush=True)
+
+
+if __name__ == "__main__":
+    main()

@@ -53,6 +53,11 @@ pg_status pg_device_malloc(int64_t bytes, void** out);
 pg_status pg_device_free(void* p);
 pg_status pg_memcpy_h2d(void* dst, const void* src, int64_t bytes);
 pg_status pg_memcpy_d2h(void* dst, const void* src, int64_t bytes);
+pg_status pg_memcpy_d2d(void* dst, const void* src, int64_t bytes);
+/* HIP-event elapsed ms of the most recent hot-path kernel launch (fused
+ * scan+filter+aggregate, or probe+aggregate) — measurement plumbing for
+ * bench.py's roofline leg */
+double pg_last_hot_kernel_ms(void);
 
 /* ---- Page / Block descriptors (SURVEY.md §8b struct) ----
  * Concrete layouts follow the Java blocks: fixed-width values array +

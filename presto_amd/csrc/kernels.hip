@@ -118,6 +118,37 @@ extern "C" pg_status pg_memcpy_d2h(void* d, const void* s, int64_t n)
     CHK(hipMemcpy(d, s, (size_t)n, hipMemcpyDeviceToHost));
     return PG_OK;
 }
+extern "C" pg_status pg_memcpy_d2d(void* d, const void* s, int64_t n)
+{
+    if (ensure_gpu()) return PG_ERR;
+    CHK(hipMemcpy(d, s, (size_t)n, hipMemcpyDeviceToDevice));
+    return PG_OK;
+}
+
+/* HIP-event elapsed time of the most recent hot-path kernel launch
+ * (the fused agg or probe-agg kernel), for bench.py's roofline leg.
+ * Events are recorded on g_stream, the stream the kernel launches on. */
+static double g_last_hot_ms = 0.0;
+static hipEvent_t g_ev0, g_ev1;
+static bool g_ev_init = false;
+static void hot_begin()
+{
+    if (!g_ev_init) {
+        hipEventCreate(&g_ev0);
+        hipEventCreate(&g_ev1);
+        g_ev_init = true;
+    }
+    hipEventRecord(g_ev0, g_stream);
+}
+static void hot_end()
+{
+    hipEventRecord(g_ev1, g_stream);
+    hipEventSynchronize(g_ev1);
+    float ms = 0;
+    hipEventElapsedTime(&ms, g_ev0, g_ev1);
+    g_last_hot_ms = ms;
+}
+extern "C" double pg_last_hot_kernel_ms(void) { return g_last_hot_ms; }
 
 /* ------------------------------------------------------------------ */
 /* device helpers                                                     */
@@ -1351,6 +1382,7 @@ struct AggSmallOp : Op {
     {
         StagedPage sp;
         sp.stage(in);
+        hot_begin();
         switch (na * 16 + maxg) {
             case 2 * 16 + 2: launch2<2, 2>(sp.pg); break;
             case 2 * 16 + 4: launch2<2, 4>(sp.pg); break;
@@ -1369,6 +1401,7 @@ struct AggSmallOp : Op {
             case 9 * 16 + 6: launch2<9, 6>(sp.pg); break;
             default: launch2<9, 8>(sp.pg); break;
         }
+        hot_end();
         CHKV(hipStreamSynchronize(g_stream));
     }
     void finish() override
@@ -1619,6 +1652,7 @@ struct JoinOp : Op {
         StagedPage sp;
         sp.stage(in);
         if (plan.mode == 1) {
+            hot_begin();
             hipLaunchKernelGGL(k_probe_agg, dim3(4096), dim3(256), 0,
                                g_stream, sp.pg, plan,
                                (const int64_t*)t->keys.p, t->mask,
@@ -1626,6 +1660,7 @@ struct JoinOp : Op {
                                (unsigned long long*)t->acc_fhi.p,
                                (unsigned long long*)t->acc_flo.p,
                                (unsigned long long*)t->acc_cnt.p);
+            hot_end();
             CHKV(hipStreamSynchronize(g_stream));
             return;
         }

@@ -1,0 +1,69 @@
+"""Multi-process exchange tests on CPU (gloo, world_size 2) — covers the
+N>1 repartition path (presto_amd/dist.py) that bench.py uses over RCCL on
+the GPU node: counts exchange + all_to_all_single column exchange, with the
+reference's partition math (replicated in the oracle) deciding destinations.
+"""
+import os
+
+import numpy as np
+import pytest
+import torch
+
+
+def _partition_math(keys, nparts):
+    # pure-python restatement of HashGenerator.java:22-29 +
+    # AbstractLongType.java:137-140 (same as fixed128.h / oracle)
+    M = (1 << 64) - 1
+    out = np.empty(len(keys), np.int64)
+    for i, v in enumerate(keys.tolist()):
+        x = (v * 0xC2B2AE3D27D4EB4F) & M
+        x = ((x << 31) | (x >> 33)) & M
+        h = (x * 0x9E3779B185EBCA87) & M
+        u = (h ^ (h >> 32)) & 0xFFFFFFFF
+        out[i] = (u * nparts) >> 32
+    return out
+
+
+def _worker(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.distributed.init_process_group("gloo", rank=rank, world_size=world)
+    from presto_amd.dist import exchange_columns, exchange_split_counts
+
+    rng = np.random.default_rng(100 + rank)
+    n = 5000
+    keys = rng.integers(1, 10**9, n).astype(np.int64)
+    vals = (keys * 7 + rank).astype(np.int64)
+    pid = _partition_math(keys, world)
+    # partition-major layout (stable within partition)
+    order = np.argsort(pid, kind="stable")
+    send_counts = [int((pid == p).sum()) for p in range(world)]
+    cols = {"k": torch.from_numpy(keys[order]),
+            "v": torch.from_numpy(vals[order])}
+    recv_counts = exchange_split_counts(send_counts)
+    got = exchange_columns(cols, send_counts, recv_counts)
+    # every received key must hash to MY partition
+    mypid = _partition_math(got["k"].numpy(), world)
+    assert (mypid == rank).all()
+    # v relation preserved
+    kv = got["v"].numpy() - got["k"].numpy() * 7
+    assert set(np.unique(kv)) <= {0, 1}
+    # total rows conserved
+    tot = torch.tensor([got["k"].numel()])
+    torch.distributed.all_reduce(tot)
+    assert tot.item() == n * world
+    # stability: rows from each source rank arrive in that source's
+    # partition-major order (ascending original order within partition)
+    off = 0
+    for r, c in enumerate(recv_counts):
+        seg = got["k"][off:off + c].numpy()
+        off += c
+        assert (_partition_math(seg, world) == rank).all()
+    torch.distributed.destroy_process_group()
+
+
+def test_exchange_gloo_world2():
+    port = 29517
+    ctx = torch.multiprocessing.start_processes(
+        _worker, args=(2, port), nprocs=2, join=True,
+        start_method="spawn")
