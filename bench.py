@@ -82,11 +82,11 @@ def gen_lineitem_device(orc, sf, device, want_orderkey=False,
     if want_orderkey:
         cols["orderkey"] = torch.empty(n, dtype=torch.int64, device=device)
     maxrows = chunk_orders * 7
-    buf = {k: np.empty(maxrows, v.cpu().numpy().dtype if False else
-                       {torch.float64: np.float64, torch.int32: np.int32,
-                        torch.uint8: np.uint8,
-                        torch.int64: np.int64}[v.dtype])
+    # pinned host staging: the generator writes straight into pinned
+    # memory, then one async h2d copy per column per chunk
+    buf = {k: torch.empty(maxrows, dtype=v.dtype, pin_memory=True)
            for k, v in cols.items()}
+    ptr = {k: C.c_void_p(t.data_ptr()) for k, t in buf.items()}
     off = 0
     o = 0
     t0 = time.time()
@@ -94,13 +94,13 @@ def gen_lineitem_device(orc, sf, device, want_orderkey=False,
         cnt = min(chunk_orders, n_ord - o)
         w = orc.lib.tpch_gen_lineitem(
             C.c_double(sf), C.c_int64(o), C.c_int64(cnt),
-            buf["orderkey"].ctypes.data if want_orderkey else None,
-            buf["quantity"].ctypes.data, buf["extendedprice"].ctypes.data,
-            buf["discount"].ctypes.data, buf["tax"].ctypes.data,
-            buf["shipdate"].ctypes.data, buf["returnflag"].ctypes.data,
-            buf["linestatus"].ctypes.data)
+            ptr["orderkey"] if want_orderkey else None,
+            ptr["quantity"], ptr["extendedprice"],
+            ptr["discount"], ptr["tax"],
+            ptr["shipdate"], ptr["returnflag"], ptr["linestatus"])
         for k, t in cols.items():
-            t[off:off + w].copy_(torch.from_numpy(buf[k][:w]))
+            t[off:off + w].copy_(buf[k][:w], non_blocking=True)
+        torch.cuda.synchronize()
         off += w
         o += cnt
     assert off == n
@@ -117,13 +117,15 @@ def gen_orders_customer_device(orc, sf, device):
     ck = np.empty(n_ord, np.int64)
     od = np.empty(n_ord, np.int32)
     orc.lib.tpch_gen_orders(C.c_double(sf), C.c_int64(0), C.c_int64(n_ord),
-                            ok.ctypes.data, ck.ctypes.data, od.ctypes.data,
-                            None)
+                            C.c_void_p(ok.ctypes.data),
+                            C.c_void_p(ck.ctypes.data),
+                            C.c_void_p(od.ctypes.data), None)
     cck = np.empty(n_cust, np.int64)
     seg = np.empty(n_cust, np.uint8)
     orc.lib.tpch_gen_customer(C.c_double(sf), C.c_int64(0),
-                              C.c_int64(n_cust), cck.ctypes.data,
-                              seg.ctypes.data)
+                              C.c_int64(n_cust),
+                              C.c_void_p(cck.ctypes.data),
+                              C.c_void_p(seg.ctypes.data))
     t = lambda a: __import__("torch").from_numpy(a).to(device)
     return (dict(orderkey=t(ok), custkey=t(ck), orderdate=t(od)),
             dict(custkey=t(cck), mktseg=t(seg)))

@@ -130,10 +130,12 @@ class OracleLib:
         p = np.ascontiguousarray(probe_keys, np.int64)
         if cap is None:
             cap = 4 * (len(b) + len(p)) + 16
-        op = np.empty(cap, np.int64)
-        ob = np.empty(cap, np.int64)
-        n = self.lib.oracle_join_bigint(C.c_int64(len(b)), _p(b),
-                                        C.c_int64(len(p)), _p(p), _p(op),
-                                        _p(ob), C.c_int64(cap))
-        assert n <= cap
-        return op[:n], ob[:n]
+        while True:
+            op = np.empty(cap, np.int64)
+            ob = np.empty(cap, np.int64)
+            n = self.lib.oracle_join_bigint(C.c_int64(len(b)), _p(b),
+                                            C.c_int64(len(p)), _p(p), _p(op),
+                                            _p(ob), C.c_int64(cap))
+            if n <= cap:
+                return op[:n], ob[:n]
+            cap = n  # function reports the true total; retry sized
