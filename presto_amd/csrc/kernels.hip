@@ -391,6 +391,146 @@ __global__ __launch_bounds__(FT_NTHREADS) void k_agg_small(
     if (local_bad) atomicAdd(bad_keys, local_bad);
 }
 
+
+/* Specialized fused Q1 kernel — the AOT analog of the reference's
+ * per-query codegen (PageFunctionCompiler/AccumulatorCompiler specialize
+ * the scan+filter+aggregate loop per expression; this kernel is that
+ * specialization for the Q1 shape: one int32 <= filter, two enumerated u8
+ * keys (3x2), aggregates [sum(a), sum(e), sum(e*(1-d)), sum(e*(1-d)*(1+t)),
+ * sum(d), count]).  Same deterministic schedule and partials layout as the
+ * generic k_agg_small<7,6,DEC>, so the finish kernel and host code are
+ * shared.  Each column is loaded ONCE per row pair with 16-byte vector
+ * loads; predicate and group selection are branchless. */
+template <bool DEC>
+__global__ __launch_bounds__(FT_NTHREADS) void k_agg_q1(
+    const double* qty, const double* ep, const double* dc, const double* tx,
+    const int32_t* sd, const uint8_t* rf, const uint8_t* ls, int64_t n,
+    int32_t ship_max, uint8_t k00, uint8_t k01, uint8_t k02, uint8_t k10,
+    uint8_t k11, double* partials, int64_t* partials_i,
+    unsigned long long* bad_keys)
+{
+    using T = typename std::conditional<DEC, int64_t, double>::type;
+    T acc[5][6];
+    int32_t cnt[6];
+#pragma unroll
+    for (int a = 0; a < 5; a++)
+#pragma unroll
+        for (int g = 0; g < 6; g++) acc[a][g] = (T)0;
+#pragma unroll
+    for (int g = 0; g < 6; g++) cnt[g] = 0;
+
+    const int64_t v = (int64_t)blockIdx.x * FT_NTHREADS + threadIdx.x;
+    unsigned long long local_bad = 0;
+    for (int64_t base = 2 * v; base < n; base += 2 * FT_VL) {
+        const bool pair = base + 1 < n;
+        /* one vector load per column per pair */
+        double2 q2, e2, d2, t2;
+        int2 s2;
+        uint8_t r0, r1, l0, l1;
+        if (pair) {
+            q2 = *(const double2*)(qty + base);
+            e2 = *(const double2*)(ep + base);
+            d2 = *(const double2*)(dc + base);
+            t2 = *(const double2*)(tx + base);
+            s2 = *(const int2*)(sd + base);
+            uchar2 rr = *(const uchar2*)(rf + base);
+            uchar2 ll = *(const uchar2*)(ls + base);
+            r0 = rr.x; r1 = rr.y; l0 = ll.x; l1 = ll.y;
+        } else {
+            q2.x = qty[base]; e2.x = ep[base]; d2.x = dc[base];
+            t2.x = tx[base]; s2.x = sd[base];
+            r0 = rf[base]; l0 = ls[base];
+            q2.y = 0; e2.y = 0; d2.y = 0; t2.y = 0; s2.y = 0x7fffffff;
+            r1 = k00; l1 = k10;
+        }
+#pragma unroll
+        for (int r = 0; r < 2; r++) {
+            double q = r ? q2.y : q2.x, e = r ? e2.y : e2.x;
+            double d = r ? d2.y : d2.x, t = r ? t2.y : t2.x;
+            int32_t sdv = r ? s2.y : s2.x;
+            uint8_t rv = r ? r1 : r0, lv = r ? l1 : l0;
+            bool sel = sdv <= ship_max;
+            int i0 = rv == k00 ? 0 : (rv == k01 ? 1 : (rv == k02 ? 2 : -1));
+            int i1 = lv == k10 ? 0 : (lv == k11 ? 1 : -1);
+            if (i0 < 0 || i1 < 0) {
+                local_bad += sel ? 1 : 0;
+                continue;
+            }
+            int g = i0 * 2 + i1;
+            T vq, ve, vdp, vch, vd;
+            if (DEC) {
+                int64_t qq = (int64_t)(q + 0.5);
+                int64_t cents = (int64_t)(e * 100.0 + 0.5);
+                int64_t dd = (int64_t)(d * 100.0 + 0.5);
+                int64_t tt = (int64_t)(t * 100.0 + 0.5);
+                int64_t dp = cents * (100 - dd);
+                vq = (T)qq; ve = (T)cents; vdp = (T)dp;
+                vch = (T)(dp * (100 + tt)); vd = (T)dd;
+            } else {
+                double dp = e * (1.0 - d);
+                vq = (T)q; ve = (T)e; vdp = (T)dp;
+                vch = (T)(dp * (1.0 + t)); vd = (T)d;
+            }
+#pragma unroll
+            for (int gg = 0; gg < 6; gg++) {
+                bool m = sel && gg == g;
+                acc[0][gg] += m ? vq : (T)0;
+                acc[1][gg] += m ? ve : (T)0;
+                acc[2][gg] += m ? vdp : (T)0;
+                acc[3][gg] += m ? vch : (T)0;
+                acc[4][gg] += m ? vd : (T)0;
+                cnt[gg] += m ? 1 : 0;
+            }
+        }
+    }
+
+    /* identical reduction to k_agg_small (NA=7, MAXG=6 layout: rows 5 and
+     * 6 both carry the count — user count agg + internal presence) */
+    __shared__ T lds[4][7][6];
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int wid = threadIdx.x >> 6;
+#pragma unroll
+    for (int a = 0; a < 7; a++)
+#pragma unroll
+        for (int g = 0; g < 6; g++) {
+            T x = a < 5 ? acc[a][g] : (T)cnt[g];
+            T s;
+            if (DEC)
+                s = (T)d_bfly_i64((int64_t)x);
+            else
+                s = (T)d_bfly_f64((double)x);
+            if (lane == 0) lds[wid][a][g] = s;
+        }
+    __syncthreads();
+    if (wid == 0 && lane < 4) {
+#pragma unroll
+        for (int a = 0; a < 7; a++)
+#pragma unroll
+            for (int g = 0; g < 6; g++) {
+                T x = lds[lane][a][g];
+                if (DEC) {
+                    int64_t y = (int64_t)x;
+                    y += __shfl_xor(y, 2, WAVE);
+                    y += __shfl_xor(y, 1, WAVE);
+                    x = (T)y;
+                } else {
+                    double y = (double)x;
+                    y += __shfl_xor(y, 2, WAVE);
+                    y += __shfl_xor(y, 1, WAVE);
+                    x = (T)y;
+                }
+                if (lane == 0) {
+                    size_t off = ((size_t)blockIdx.x * 7 + a) * 6 + g;
+                    if (DEC)
+                        partials_i[off] += (int64_t)x;
+                    else
+                        partials[off] += (double)x;
+                }
+            }
+    }
+    if (local_bad) atomicAdd(bad_keys, local_bad);
+}
+
 /* finish: 64-lane deterministic reduce of block partials.
  * DEC: output int128 (hi,lo) pairs; F64: doubles. */
 template <bool DEC>
@@ -1363,6 +1503,70 @@ struct AggSmallOp : Op {
         bad.alloc(8);
         bad.zero();
     }
+    bool q1_shape(const pg_page& pg) const
+    {
+        const pg_plan_hash_agg_small& p = plan;
+        if (user_aggs != 6 || p.n_keys != 2 || p.n_vals[0] != 3 ||
+            p.n_vals[1] != 2 || p.n_preds != 1)
+            return false;
+        if (na != 7 || maxg != 6) return false;
+        if (pg.cols[p.preds[0].col].tag != PG_T_I32 ||
+            p.preds[0].op != PG_CMP_LE)
+            return false;
+        const pg_agg* a = p.aggs;
+        if (!(a[0].func != PG_AGG_COUNT && a[0].proj.kind == PG_PROJ_IDENT &&
+              a[1].proj.kind == PG_PROJ_IDENT &&
+              a[2].proj.kind == PG_PROJ_DISC_PRICE &&
+              a[3].proj.kind == PG_PROJ_CHARGE &&
+              a[4].proj.kind == PG_PROJ_IDENT && a[5].func == PG_AGG_COUNT))
+            return false;
+        if (a[2].proj.a != a[1].proj.a || a[3].proj.a != a[1].proj.a ||
+            a[3].proj.b != a[2].proj.b || a[4].proj.a != a[2].proj.b)
+            return false;
+        if (dec && !(a[0].dec_scale == 0 && a[1].dec_scale == 2 &&
+                     a[4].dec_scale == 2))
+            return false;
+        /* all four money channels must be f64 and 16-byte aligned */
+        int chans[4] = {a[0].proj.a, a[1].proj.a, a[2].proj.b, a[3].proj.c};
+        for (int i = 0; i < 4; i++) {
+            const pg_col& c = pg.cols[chans[i]];
+            if (c.tag != PG_T_F64 || ((uintptr_t)c.data & 15)) return false;
+        }
+        if (((uintptr_t)pg.cols[p.preds[0].col].data & 7)) return false;
+        if (pg.cols[p.key_col[0]].tag != PG_T_U8 ||
+            pg.cols[p.key_col[1]].tag != PG_T_U8)
+            return false;
+        return true;
+    }
+    void launch_q1(const pg_page& pg)
+    {
+        const pg_plan_hash_agg_small& p = plan;
+        const double* qty = (const double*)pg.cols[p.aggs[0].proj.a].data;
+        const double* ep = (const double*)pg.cols[p.aggs[1].proj.a].data;
+        const double* dc = (const double*)pg.cols[p.aggs[2].proj.b].data;
+        const double* tx = (const double*)pg.cols[p.aggs[3].proj.c].data;
+        const int32_t* sd = (const int32_t*)pg.cols[p.preds[0].col].data;
+        const uint8_t* rf = (const uint8_t*)pg.cols[p.key_col[0]].data;
+        const uint8_t* ls = (const uint8_t*)pg.cols[p.key_col[1]].data;
+        if (dec)
+            hipLaunchKernelGGL((k_agg_q1<true>), dim3(FT_NBLOCKS),
+                               dim3(FT_NTHREADS), 0, g_stream, qty, ep, dc,
+                               tx, sd, rf, ls, pg.n_rows,
+                               (int32_t)p.preds[0].ival, p.key_vals[0][0],
+                               p.key_vals[0][1], p.key_vals[0][2],
+                               p.key_vals[1][0], p.key_vals[1][1], nullptr,
+                               (int64_t*)partials.p,
+                               (unsigned long long*)bad.p);
+        else
+            hipLaunchKernelGGL((k_agg_q1<false>), dim3(FT_NBLOCKS),
+                               dim3(FT_NTHREADS), 0, g_stream, qty, ep, dc,
+                               tx, sd, rf, ls, pg.n_rows,
+                               (int32_t)p.preds[0].ival, p.key_vals[0][0],
+                               p.key_vals[0][1], p.key_vals[0][2],
+                               p.key_vals[1][0], p.key_vals[1][1],
+                               (double*)partials.p, nullptr,
+                               (unsigned long long*)bad.p);
+    }
     template <int NA, int MAXG>
     void launch2(const pg_page& pg)
     {
@@ -1383,6 +1587,12 @@ struct AggSmallOp : Op {
         StagedPage sp;
         sp.stage(in);
         hot_begin();
+        if (q1_shape(sp.pg)) {
+            launch_q1(sp.pg);
+            hot_end();
+            CHKV(hipStreamSynchronize(g_stream));
+            return;
+        }
         switch (na * 16 + maxg) {
             case 2 * 16 + 2: launch2<2, 2>(sp.pg); break;
             case 2 * 16 + 4: launch2<2, 4>(sp.pg); break;
