@@ -1239,6 +1239,36 @@ __global__ __launch_bounds__(256) void k_part_emit(pg_page pg,
 
 namespace {
 
+/* size-bucketed device buffer pool: operator lifecycles alloc/free the
+ * same sizes every query, and hipMalloc/hipFree cost ~0.5-1 ms each —
+ * pooling keeps the per-query overhead at kernel time.  Exact-size reuse;
+ * bounded total. */
+struct BufPool {
+    std::mutex mu;
+    std::multimap<size_t, void*> free_bufs;
+    size_t pooled = 0;
+    static const size_t CAP = 8ull << 30;
+    void* get(size_t n)
+    {
+        std::lock_guard<std::mutex> lk(mu);
+        auto it = free_bufs.find(n);
+        if (it == free_bufs.end()) return nullptr;
+        void* p = it->second;
+        free_bufs.erase(it);
+        pooled -= n;
+        return p;
+    }
+    bool put(size_t n, void* p)
+    {
+        std::lock_guard<std::mutex> lk(mu);
+        if (pooled + n > CAP) return false;
+        free_bufs.emplace(n, p);
+        pooled += n;
+        return true;
+    }
+};
+static BufPool g_pool;
+
 struct DevBuf {
     void* p = nullptr;
     size_t sz = 0;
@@ -1262,13 +1292,15 @@ struct DevBuf {
     void alloc(size_t n)
     {
         free();
-        CHKV(hipMalloc(&p, n ? n : 1));
-        sz = n;
+        size_t want = n ? n : 1;
+        p = g_pool.get(want);
+        if (!p) CHKV(hipMalloc(&p, want));
+        sz = want;
     }
     void zero() { CHKV(hipMemsetAsync(p, 0, sz, g_stream)); }
     void free()
     {
-        if (p) hipFree(p);
+        if (p && !g_pool.put(sz, p)) (void)hipFree(p);
         p = nullptr;
         sz = 0;
     }
