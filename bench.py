@@ -281,6 +281,16 @@ def main():
             "frac": achieved / (HBM_PEAK / 1e9) if achieved else None,
             "traffic": None,
         }
+        # measured per-launch HBM traffic from committed rocprofv3 PMC
+        # passes (profiles/pmc_traffic.json; see profiles/r01_*_pmc.txt)
+        try:
+            tr = json.loads((REPO / "profiles" / "pmc_traffic.json"
+                             ).read_text())
+            key = f"{args.query}:{args.mode}:sf{int(args.sf)}"
+            if key in tr:
+                roofline["traffic"] = tr[key]
+        except Exception:
+            pass
         cpu = None
         if not args.skip_cpu_baseline and n_gpus == 1:
             fn = cpu_baseline_q1 if args.query == "q1" else cpu_baseline_q3

@@ -539,7 +539,9 @@ __global__ __launch_bounds__(WAVE) void k_agg_small_finish(
     double* out_f, int64_t* out_hi, uint64_t* out_lo)
 {
     int lane = threadIdx.x;
-    for (int j = 0; j < na_x_maxg; j++) {
+    {
+        int j = blockIdx.x; /* one block (one wave) per output field; the
+                               per-field reduction tree is unchanged */
         if (DEC) {
             /* lane l: sequential over blocks l, l+64, ... (i64 safe:
              * 64 block partials each bounded by per-block row counts) */
@@ -1654,13 +1656,14 @@ struct AggSmallOp : Op {
         out_hi.alloc(nm * 8);
         out_lo.alloc(nm * 8);
         if (dec)
-            hipLaunchKernelGGL(k_agg_small_finish<true>, dim3(1), dim3(WAVE),
-                               0, g_stream, nullptr, (int64_t*)partials.p, nm,
-                               nullptr, (int64_t*)out_hi.p,
-                               (uint64_t*)out_lo.p);
+            hipLaunchKernelGGL(k_agg_small_finish<true>, dim3(nm),
+                               dim3(WAVE), 0, g_stream, nullptr,
+                               (int64_t*)partials.p, nm, nullptr,
+                               (int64_t*)out_hi.p, (uint64_t*)out_lo.p);
         else
-            hipLaunchKernelGGL(k_agg_small_finish<false>, dim3(1), dim3(WAVE),
-                               0, g_stream, (double*)partials.p, nullptr, nm,
+            hipLaunchKernelGGL(k_agg_small_finish<false>, dim3(nm),
+                               dim3(WAVE), 0, g_stream,
+                               (double*)partials.p, nullptr, nm,
                                (double*)out_f.p, nullptr, nullptr);
         std::vector<double> hf(nm);
         std::vector<int64_t> hhi(nm);
