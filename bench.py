@@ -236,9 +236,11 @@ def main():
         cust_page = Page(ccols)
 
         def step():
-            return pipelines.q3(cust_page, ord_page, li_page,
-                                mode=("dec" if args.mode == "f64" else
-                                      args.mode))
+            if world > 1:
+                from presto_amd.dist import q3_distributed
+                return q3_distributed(cust_page, ord_page, li_page, world,
+                                      rank, device, mode="dec")
+            return pipelines.q3(cust_page, ord_page, li_page, mode="dec")
         total_rows_per_step = n_rows * n_gpus
         workload = Q3_WORKLOAD.format(sf=int(args.sf), n=n_gpus)
 

@@ -37,3 +37,219 @@ def exchange_columns(cols, send_counts, recv_counts=None, device=None):
                                input_split_sizes=list(send_counts))
         out[name] = r
     return out
+
+
+def _col_to_torch(col, n, device):
+    """Copy a library-owned device column (pg_col) into a torch tensor."""
+    import numpy as np
+    from .engine import lib, _TAG_NP
+    dtmap = {0: torch.uint8, 1: torch.int32, 2: torch.int64,
+             3: torch.float64}
+    t = torch.empty(n, dtype=dtmap[col.tag], device=device)
+    nbytes = n * t.element_size()
+    if nbytes:
+        L = lib()
+        if col.on_device:
+            L.check(L.c.pg_memcpy_d2d(t.data_ptr(), col.data, nbytes), "d2d")
+        else:
+            import ctypes
+            src = (ctypes.c_uint8 * nbytes).from_address(col.data)
+            t.copy_(torch.frombuffer(bytearray(src), dtype=t.dtype))
+    return t
+
+
+def _gather_padded(t, cap, device, pad_value=0):
+    """all_gather of a variable-length 1-D tensor, padded to cap."""
+    world = dist.get_world_size()
+    buf = torch.full((cap,), pad_value, dtype=t.dtype, device=device)
+    buf[:t.numel()] = t
+    outs = [torch.empty_like(buf) for _ in range(world)]
+    dist.all_gather(outs, buf)
+    return outs
+
+
+def q3_distributed(cust_page, ord_page, li_page, world, rank, device,
+                   limit=10, mode="dec"):
+    """Distributed Q3 (BASELINE config 4): the repartition seam over RCCL.
+
+    customer: local filter (mktsegment='BUILDING') -> broadcast key set
+      (all_gather of selected custkeys — the BroadcastOutputBuffer analog)
+    orders:   local filter (orderdate<9204) -> PARTITION by orderkey hash
+      -> all_to_all -> local HASH_BUILD with fused semijoin vs the set
+    lineitem: local filter (shipdate>9204) -> PARTITION by orderkey hash
+      -> all_to_all -> local LOOKUP_JOIN fused grouped sum -> local TopN
+      -> gather candidates -> final TopN on rank 0.
+
+    world==1 runs the same operator graph with the collectives as
+    identities (used by the N=1 parity test).
+    Returns dict of numpy arrays on rank 0, None elsewhere.
+    """
+    import numpy as np
+    import presto_amd as P
+    from .engine import lib
+
+    use_dist = world > 1
+
+    # ---- customer set ----
+    fp = P.PlanFilterProject()
+    fp.n_preds = 1
+    fp.preds[0] = P.Pred(cust_page.channel("mktseg"), P.CMP_EQ, 1, 0.0)
+    fp.n_proj = 1
+    fp.proj[0] = P.Proj(P.PROJ_IDENT, cust_page.channel("custkey"), 0, 0)
+    f = P.Operator(P.OP_FILTER_PROJECT, fp)
+    f.add_input(cust_page)
+    raw = f.get_output_raw()
+    ck = _col_to_torch(raw.cols[0], raw.n_rows, device)
+    f.destroy()
+    if use_dist:
+        cap = cust_page.n_rows  # selected <= local rows; same per rank
+        parts = _gather_padded(ck, cap, device)
+        ck = torch.cat(parts)
+    bp = P.PlanHashBuild()
+    bp.n_preds = 1
+    bp.preds[0] = P.Pred(0, P.CMP_GE, 1, 0.0)  # drop the 0 padding
+    bp.key_col = 0
+    bp.semijoin_table = -1
+    bp.n_payload = 0
+    bp.capacity_hint = max(int(ck.numel()), 16)
+    bp.key_set_only = 1
+    b1 = P.Operator(P.OP_HASH_BUILD, bp)
+    b1.add_input(P.Page({"custkey": ck}))
+    b1.finish()
+    set_tbl = b1.table()
+
+    def filter_partition_exchange(page, preds, emit_channels):
+        """filter -> partition by col0 of the emitted page -> exchange."""
+        fpl = P.PlanFilterProject()
+        fpl.n_preds = len(preds)
+        for i, pr in enumerate(preds):
+            fpl.preds[i] = pr
+        fpl.n_proj = len(emit_channels)
+        for i, ch in enumerate(emit_channels):
+            fpl.proj[i] = P.Proj(P.PROJ_IDENT, ch, 0, 0)
+        fo = P.Operator(P.OP_FILTER_PROJECT, fpl)
+        fo.add_input(page)
+        fraw = fo.get_output_raw()
+        pp = P.PlanPartition()
+        pp.n_partitions = world
+        pp.key_col = 0
+        pp.n_emit = len(emit_channels)
+        for i in range(len(emit_channels)):
+            pp.emit_cols[i] = i
+        po = P.Operator(P.OP_PARTITION, pp)
+        po.add_input_raw(fraw)
+        fo.destroy()
+        counts = po.partition_counts(world)
+        # concatenate partition slices into per-column send tensors
+        pages = [po.get_output_raw() for _ in range(world)]
+        total = sum(counts)
+        cols = {}
+        for c in range(len(emit_channels)):
+            col0 = pages[0].cols[c]
+            dtmap = {0: torch.uint8, 1: torch.int32, 2: torch.int64,
+                     3: torch.float64}
+            t = torch.empty(total, dtype=dtmap[col0.tag], device=device)
+            off = 0
+            for p in range(world):
+                nrow = pages[p].n_rows
+                if nrow:
+                    nb = nrow * t.element_size()
+                    L = lib()
+                    L.check(L.c.pg_memcpy_d2d(
+                        t.data_ptr() + off * t.element_size(),
+                        pages[p].cols[c].data, nb), "d2d")
+                off += nrow
+            cols[f"c{c}"] = t
+        po.destroy()
+        if use_dist:
+            return exchange_columns(cols, counts, device=device)
+        return cols
+
+    # ---- orders ----
+    ocols = filter_partition_exchange(
+        ord_page,
+        [P.Pred(ord_page.channel("orderdate"), P.CMP_LT, 9204, 0.0)],
+        [ord_page.channel("orderkey"), ord_page.channel("orderdate"),
+         ord_page.channel("custkey")])
+    b2p = P.PlanHashBuild()
+    b2p.n_preds = 0
+    b2p.key_col = 0
+    b2p.semijoin_table = set_tbl
+    b2p.semijoin_col = 2
+    b2p.n_payload = 1
+    b2p.payload_col[0] = 1
+    b2p.capacity_hint = max(int(ocols["c0"].numel()) // 4, 16)
+    b2 = P.Operator(P.OP_HASH_BUILD, b2p)
+    b2.add_input(P.Page({k: v for k, v in ocols.items()}))
+    b2.finish()
+    tbl = b2.table()
+
+    # ---- lineitem probe ----
+    lcols = filter_partition_exchange(
+        li_page,
+        [P.Pred(li_page.channel("shipdate"), P.CMP_GT, 9204, 0.0)],
+        [li_page.channel("orderkey"), li_page.channel("extendedprice"),
+         li_page.channel("discount")])
+    jp = P.PlanLookupJoin()
+    jp.table = tbl
+    jp.n_preds = 0
+    jp.key_col = 0
+    jp.mode = 1
+    jp.proj = P.Proj(P.PROJ_DISC_PRICE, 1, 2, 0)
+    jp.dec_scale = 4
+    j = P.Operator(P.OP_LOOKUP_JOIN, jp)
+    j.add_input(P.Page({k: v for k, v in lcols.items()}))
+    j.finish()
+    groups = j.get_output_raw()
+    tp = P.PlanTopN()
+    tp.limit = limit
+    tp.val_col = 2 if mode == "dec" else 3
+    tp.date_col = 1
+    tp.key_col = 0
+    t = P.Operator(P.OP_TOPN, tp)
+    t.add_input_raw(groups)
+    t.finish()
+    out = t.get_output(["orderkey", "rev", "orderdate"])
+    t.destroy()
+    j.destroy()
+    lib().c.pg_table_destroy(set_tbl)
+    lib().c.pg_table_destroy(tbl)
+    b1.destroy()
+    b2.destroy()
+
+    if not use_dist:
+        return out
+    # gather per-rank candidates, final TopN on rank 0
+    def pad(a, dt):
+        x = torch.full((limit,), -1, dtype=dt, device=device)
+        x[:len(a)] = torch.from_numpy(np.ascontiguousarray(a)).to(device)
+        return x
+    ks = _gather_padded(pad(out["orderkey"], torch.int64), limit, device, -1)
+    if mode == "dec":
+        vs = _gather_padded(pad(out["rev"], torch.int64), limit, device, -1)
+    else:
+        vs = _gather_padded(pad(out["rev"].view(np.int64), torch.int64),
+                            limit, device, -1)
+    ds = _gather_padded(pad(out["orderdate"], torch.int32), limit, device, -1)
+    if rank != 0:
+        return None
+    rows = []
+    for r in range(world):
+        k = ks[r].cpu().numpy()
+        v = vs[r].cpu().numpy()
+        d = ds[r].cpu().numpy()
+        for i in range(limit):
+            if k[i] >= 0:
+                rows.append((v[i], d[i], k[i]))
+    if mode == "dec":
+        rows.sort(key=lambda x: (-x[0], x[1], x[2]))
+    else:
+        rows.sort(key=lambda x: (-np.int64(x[0]).view(np.float64),
+                                 x[1], x[2]))
+    rows = rows[:limit]
+    res = dict(orderkey=np.array([r[2] for r in rows], np.int64),
+               rev=np.array([r[0] for r in rows], np.int64),
+               orderdate=np.array([r[1] for r in rows], np.int32))
+    if mode != "dec":
+        res["rev"] = res["rev"].view(np.float64)
+    return res

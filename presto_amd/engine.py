@@ -223,12 +223,18 @@ class Operator:
         return _read_output_page(pp.contents, names)
 
     def get_output_raw(self):
-        """Returns the raw PgPage (device pointers stay on device) or None.
-        Valid until the next get_output*/destroy on this operator."""
+        """Returns a snapshot of the raw PgPage descriptor (device pointers
+        stay on device) or None.  The column BUFFERS it points at are owned
+        by the library: valid until the next get_output on this operator
+        (PARTITION: until destroy — its pages alias op-owned buffers)."""
         L = lib()
         pp = C.POINTER(PgPage)()
         L.check(L.c.pg_op_get_output(self.h, C.byref(pp)), "get_output")
-        return pp.contents if pp else None
+        if not pp:
+            return None
+        snap = PgPage()
+        C.memmove(C.byref(snap), pp, C.sizeof(PgPage))
+        return snap
 
     def finish(self):
         lib().check(lib().c.pg_op_finish(self.h), "finish")

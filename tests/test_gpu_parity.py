@@ -301,3 +301,24 @@ def test_multi_page_decimal_agg(P, oracle_lib, sf01):
     for i, g in enumerate(exp):
         assert got["count"][i] == g.count_order
         assert got["sum_disc_price_lo"][i] == g.sum_disc_1e4
+
+
+def test_q3_distributed_graph_world1(P, oracle_lib, sf01):
+    """The distributed Q3 operator graph (filter->partition->exchange->
+    build->probe->topn) run with world==1 (collectives as identities) must
+    reproduce the oracle exactly — covers the repartition path end-to-end
+    on one GPU; the collectives themselves are covered by the gloo tests."""
+    import torch
+    from presto_amd.dist import q3_distributed
+    cust, orders, li = sf01["cust"], sf01["orders"], sf01["li"]
+    dev = torch.device("cuda", 0)
+    out = q3_distributed(
+        P.Page({"custkey": cust["custkey"], "mktseg": cust["mktseg"]}),
+        P.Page({k: orders[k] for k in ("orderkey", "custkey", "orderdate")}),
+        _li_page(P, li), world=1, rank=0, device=dev, mode="dec")
+    exp = oracle_lib.q3(cust, orders, li)
+    assert len(out["orderkey"]) == len(exp)
+    for i, r in enumerate(exp):
+        assert out["orderkey"][i] == r.orderkey
+        assert out["rev"][i] == r.revenue_1e4
+        assert out["orderdate"][i] == r.orderdate
