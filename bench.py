@@ -206,14 +206,26 @@ def main():
         page = Page(cols, n_rows=n_rows)
         plan = pipelines.q1_plan(page, args.mode)
 
-        def step():
+        phase_t = {}
+
+        def step(timing=False):
+            t = [time.time()]
             op = presto_amd.Operator(presto_amd.OP_HASH_AGG_SMALL, plan)
+            t.append(time.time())
             op.add_input(page)
+            t.append(time.time())
             op.finish()
+            t.append(time.time())
             out = op.get_output(
                 pipelines.Q1_F64_NAMES if args.mode == "f64"
                 else pipelines.Q1_DEC_NAMES)
+            t.append(time.time())
             op.destroy()
+            t.append(time.time())
+            if timing:
+                for i, nm in enumerate(("create", "add_input", "finish",
+                                        "get_output", "destroy")):
+                    phase_t[nm] = phase_t.get(nm, 0) + (t[i + 1] - t[i])
             if world > 1:
                 # partial->final: gather tiny group sums, combine in rank
                 # order (HashAggregationOperator partial/final Step split)
@@ -254,6 +266,11 @@ def main():
         step()
     barrier_sync()
     elapsed = time.time() - t0
+    if args.query == "q1" and rank == 0:
+        for _ in range(3):
+            step(timing=True)
+        log("q1 step phases (ms, avg of 3): " + ", ".join(
+            f"{k}={v / 3 * 1e3:.2f}" for k, v in phase_t.items()))
     if world > 1:
         import torch.distributed as dist
         e = torch.tensor([elapsed], device=device)
