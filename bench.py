@@ -64,12 +64,17 @@ def load_oracle():
 
 
 def gen_lineitem_device(orc, sf, device, want_orderkey=False,
-                        chunk_orders=4_000_000):
-    """Generate SF lineitem columns chunkwise on host, upload to device
-    tensors.  Returns dict of torch tensors."""
+                        chunk_orders=4_000_000, ord_start=0, ord_count=None):
+    """Generate lineitem columns for orders [ord_start, ord_start+ord_count)
+    of logical scale sf, chunkwise on host, uploaded to device tensors.
+    Returns dict of torch tensors."""
     import torch
-    n_ord = orc.lib.tpch_orders_count(C.c_double(sf))
-    n = orc.lib.tpch_lineitem_count(C.c_double(sf))
+    n_all = orc.lib.tpch_orders_count(C.c_double(sf))
+    if ord_count is None:
+        ord_count = n_all - ord_start
+    n = (orc.lib.tpch_lineitem_offset(C.c_double(sf),
+                                      C.c_int64(ord_start + ord_count)) -
+         orc.lib.tpch_lineitem_offset(C.c_double(sf), C.c_int64(ord_start)))
     cols = {
         "quantity": torch.empty(n, dtype=torch.float64, device=device),
         "extendedprice": torch.empty(n, dtype=torch.float64, device=device),
@@ -88,10 +93,11 @@ def gen_lineitem_device(orc, sf, device, want_orderkey=False,
            for k, v in cols.items()}
     ptr = {k: C.c_void_p(t.data_ptr()) for k, t in buf.items()}
     off = 0
-    o = 0
+    o = ord_start
+    end = ord_start + ord_count
     t0 = time.time()
-    while o < n_ord:
-        cnt = min(chunk_orders, n_ord - o)
+    while o < end:
+        cnt = min(chunk_orders, end - o)
         w = orc.lib.tpch_gen_lineitem(
             C.c_double(sf), C.c_int64(o), C.c_int64(cnt),
             ptr["orderkey"] if want_orderkey else None,
@@ -201,8 +207,13 @@ def main():
             dist.barrier()
             torch.cuda.synchronize()
 
+    sf_total = args.sf * world  # weak scaling: per-rank work fixed
+    n_ord_all = orc.lib.tpch_orders_count(C.c_double(sf_total))
+    o0 = n_ord_all * rank // world
+    o1 = n_ord_all * (rank + 1) // world
     if args.query == "q1":
-        cols, n_rows = gen_lineitem_device(orc, args.sf, device)
+        cols, n_rows = gen_lineitem_device(orc, sf_total, device,
+                                           ord_start=o0, ord_count=o1 - o0)
         page = Page(cols, n_rows=n_rows)
         plan = pipelines.q1_plan(page, args.mode)
 
@@ -238,12 +249,14 @@ def main():
                 dist.all_gather(allp, flat)
             return out
         total_rows_per_step = n_rows * n_gpus
-        workload = Q1_WORKLOAD.format(sf=int(args.sf), n=n_gpus)
+        workload = Q1_WORKLOAD.format(sf=int(sf_total), n=n_gpus)
     else:
-        cols, n_rows = gen_lineitem_device(orc, args.sf, device,
-                                           want_orderkey=True)
+        cols, n_rows = gen_lineitem_device(orc, sf_total, device,
+                                           want_orderkey=True,
+                                           ord_start=o0, ord_count=o1 - o0)
         li_page = Page(cols, n_rows=n_rows)
-        ocols, ccols = gen_orders_customer_device(orc, args.sf, device)
+        ocols, ccols = gen_orders_customer_device(orc, sf_total, device,
+                                                  rank, world)
         ord_page = Page(ocols)
         cust_page = Page(ccols)
 
@@ -254,8 +267,13 @@ def main():
                                       rank, device, mode="dec")
             return pipelines.q3(cust_page, ord_page, li_page, mode="dec")
         total_rows_per_step = n_rows * n_gpus
-        workload = Q3_WORKLOAD.format(sf=int(args.sf), n=n_gpus)
+        workload = Q3_WORKLOAD.format(sf=int(sf_total), n=n_gpus)
 
+    if world > 1:
+        import torch.distributed as dist
+        tr = torch.tensor([n_rows], dtype=torch.int64, device=device)
+        dist.all_reduce(tr)
+        total_rows_per_step = int(tr.item())
     lib = presto_amd.engine.lib()
     lib.c.pg_last_hot_kernel_ms.restype = C.c_double
     for _ in range(args.warmup):
