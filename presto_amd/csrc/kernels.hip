@@ -401,7 +401,7 @@ __global__ __launch_bounds__(FT_NTHREADS) void k_agg_small(
  * generic k_agg_small<7,6,DEC>, so the finish kernel and host code are
  * shared.  Each column is loaded ONCE per row pair with 16-byte vector
  * loads; predicate and group selection are branchless. */
-template <bool DEC>
+template <bool DEC, int VAR = 0>
 __global__ __launch_bounds__(FT_NTHREADS) void k_agg_q1(
     const double* qty, const double* ep, const double* dc, const double* tx,
     const int32_t* sd, const uint8_t* rf, const uint8_t* ls, int64_t n,
@@ -421,6 +421,7 @@ __global__ __launch_bounds__(FT_NTHREADS) void k_agg_q1(
 
     const int64_t v = (int64_t)blockIdx.x * FT_NTHREADS + threadIdx.x;
     unsigned long long local_bad = 0;
+#pragma unroll(VAR == 2 ? 2 : 1)
     for (int64_t base = 2 * v; base < n; base += 2 * FT_VL) {
         const bool pair = base + 1 < n;
         /* one vector load per column per pair */
@@ -428,6 +429,27 @@ __global__ __launch_bounds__(FT_NTHREADS) void k_agg_q1(
         int2 s2;
         uint8_t r0, r1, l0, l1;
         if (pair) {
+            if (VAR >= 1) {
+                /* nontemporal: streamed once, no L2 retention needed */
+                typedef double vd2 __attribute__((ext_vector_type(2)));
+                typedef int vi2 __attribute__((ext_vector_type(2)));
+                vd2 q_ = __builtin_nontemporal_load((const vd2*)(qty + base));
+                vd2 e_ = __builtin_nontemporal_load((const vd2*)(ep + base));
+                vd2 d_ = __builtin_nontemporal_load((const vd2*)(dc + base));
+                vd2 t_ = __builtin_nontemporal_load((const vd2*)(tx + base));
+                vi2 s_ = __builtin_nontemporal_load((const vi2*)(sd + base));
+                q2.x = q_[0]; q2.y = q_[1];
+                e2.x = e_[0]; e2.y = e_[1];
+                d2.x = d_[0]; d2.y = d_[1];
+                t2.x = t_[0]; t2.y = t_[1];
+                s2.x = s_[0]; s2.y = s_[1];
+                uint16_t rw = __builtin_nontemporal_load(
+                    (const uint16_t*)(rf + base));
+                uint16_t lw = __builtin_nontemporal_load(
+                    (const uint16_t*)(ls + base));
+                r0 = (uint8_t)rw; r1 = (uint8_t)(rw >> 8);
+                l0 = (uint8_t)lw; l1 = (uint8_t)(lw >> 8);
+            } else {
             q2 = *(const double2*)(qty + base);
             e2 = *(const double2*)(ep + base);
             d2 = *(const double2*)(dc + base);
@@ -436,6 +458,7 @@ __global__ __launch_bounds__(FT_NTHREADS) void k_agg_q1(
             uchar2 rr = *(const uchar2*)(rf + base);
             uchar2 ll = *(const uchar2*)(ls + base);
             r0 = rr.x; r1 = rr.y; l0 = ll.x; l1 = ll.y;
+            }
         } else {
             q2.x = qty[base]; e2.x = ep[base]; d2.x = dc[base];
             t2.x = tx[base]; s2.x = sd[base];
@@ -796,14 +819,22 @@ __device__ inline int64_t d_tbl_find(const int64_t* keys, int64_t mask,
     }
 }
 
+/* per-slot accumulators, interleaved so one probe hit touches ONE cache
+ * line (32 B of one 64-B line) instead of four separate arrays */
+struct slot_acc {
+    unsigned long long dec;  /* exact decimal ticks */
+    unsigned long long flo;  /* 64.64 fixed-point f64 sum, low word */
+    unsigned long long fhi;  /* high word (carries) */
+    unsigned long long cnt;
+};
+
 /* probe + fused grouped SUM into table accumulators (Q3's
  * LookupJoinOperator + HashAggregationOperator fused; revenue summed
  * exactly in decimal ticks AND in 64.64 fixed point — order-independent,
  * so atomics preserve bit-determinism) */
 __global__ __launch_bounds__(256) void k_probe_agg(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys, int64_t mask,
-    unsigned long long* acc_dec, unsigned long long* acc_fhi,
-    unsigned long long* acc_flo, unsigned long long* acc_cnt)
+    slot_acc* acc)
 {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -819,10 +850,10 @@ __global__ __launch_bounds__(256) void k_probe_agg(
         double p = d_eval_proj_f64(pg, plan.proj, i);
         uint64_t phi, plo;
         fx128_from_f64(p, &phi, &plo);
-        atomicAdd(&acc_dec[s], (unsigned long long)ticks);
-        unsigned long long old = atomicAdd(&acc_flo[s], plo);
-        atomicAdd(&acc_fhi[s], phi + (old > ~plo ? 1ull : 0ull));
-        atomicAdd(&acc_cnt[s], 1ull);
+        atomicAdd(&acc[s].dec, (unsigned long long)ticks);
+        unsigned long long old = atomicAdd(&acc[s].flo, plo);
+        atomicAdd(&acc[s].fhi, phi + (old > ~plo ? 1ull : 0ull));
+        atomicAdd(&acc[s].cnt, 1ull);
     }
 }
 
@@ -900,7 +931,7 @@ __global__ __launch_bounds__(256) void k_probe_emit(
 /* extract group rows after fused probe-agg: slots with count>0, emitted
  * slot-ascending (stable compaction over the slot array) */
 __global__ __launch_bounds__(256) void k_groups_count(
-    const unsigned long long* acc_cnt, int64_t cap, int64_t chunk,
+    const slot_acc* acc, int64_t cap, int64_t chunk,
     int64_t* block_counts)
 {
     const int64_t lo = (int64_t)blockIdx.x * chunk;
@@ -909,7 +940,7 @@ __global__ __launch_bounds__(256) void k_groups_count(
     int64_t cnt = 0;
     for (int64_t base = lo + 64 * wid; base < hi; base += 256) {
         int64_t i = base + lane;
-        bool sel = i < hi && acc_cnt[i] > 0;
+        bool sel = i < hi && acc[i].cnt > 0;
         uint64_t m = d_ballot(sel);
         if (lane == 0) cnt += __popcll(m);
     }
@@ -921,9 +952,8 @@ __global__ __launch_bounds__(256) void k_groups_count(
 }
 
 __global__ __launch_bounds__(256) void k_groups_emit(
-    const int64_t* keys, const int32_t* head, const unsigned long long* acc_dec,
-    const unsigned long long* acc_fhi, const unsigned long long* acc_flo,
-    const unsigned long long* acc_cnt, build_payloads bp, int64_t cap,
+    const int64_t* keys, const int32_t* head, const slot_acc* acc,
+    build_payloads bp, int64_t cap,
     int64_t chunk, const int64_t* block_offs, int64_t* out_key,
     emit_outs payload_outs, int64_t* out_dec, double* out_f64,
     int64_t* out_cnt)
@@ -937,7 +967,7 @@ __global__ __launch_bounds__(256) void k_groups_emit(
     __syncthreads();
     for (int64_t base = lo; base < hi; base += 256) {
         int64_t i = base + 64 * wid + lane;
-        bool sel = i < hi && acc_cnt[i] > 0;
+        bool sel = i < hi && acc[i].cnt > 0;
         uint64_t m = d_ballot(sel);
         int wsum = __popcll(m);
         if (lane == 0) wcnt[wid] = wsum;
@@ -963,9 +993,9 @@ __global__ __launch_bounds__(256) void k_groups_emit(
                             ((const double*)bp.ptr[o])[r];
                 }
             }
-            out_dec[pos] = (int64_t)acc_dec[i];
-            out_f64[pos] = fx128_to_f64(acc_fhi[i], acc_flo[i]);
-            out_cnt[pos] = (int64_t)acc_cnt[i];
+            out_dec[pos] = (int64_t)acc[i].dec;
+            out_f64[pos] = fx128_to_f64(acc[i].fhi, acc[i].flo);
+            out_cnt[pos] = (int64_t)acc[i].cnt;
         }
         __syncthreads();
         if (threadIdx.x == 0)
@@ -1360,7 +1390,7 @@ struct Table {
     int64_t cap = 0, mask = 0, n_rows = 0;
     bool key_set_only = false;
     DevBuf keys, head, next;
-    DevBuf acc_dec, acc_fhi, acc_flo, acc_cnt;
+    DevBuf acc; /* slot_acc[cap], interleaved */
     /* compacted build-row arrays: key + payloads */
     DevBuf key_rows;
     std::vector<DevBuf> payload;
@@ -1574,6 +1604,11 @@ struct AggSmallOp : Op {
     }
     void launch_q1(const pg_page& pg)
     {
+        static int var = -1;
+        if (var < 0) {
+            const char* e = getenv("PG_Q1_VARIANT");
+            var = e ? atoi(e) : 0;
+        }
         const pg_plan_hash_agg_small& p = plan;
         const double* qty = (const double*)pg.cols[p.aggs[0].proj.a].data;
         const double* ep = (const double*)pg.cols[p.aggs[1].proj.a].data;
@@ -1582,24 +1617,26 @@ struct AggSmallOp : Op {
         const int32_t* sd = (const int32_t*)pg.cols[p.preds[0].col].data;
         const uint8_t* rf = (const uint8_t*)pg.cols[p.key_col[0]].data;
         const uint8_t* ls = (const uint8_t*)pg.cols[p.key_col[1]].data;
-        if (dec)
-            hipLaunchKernelGGL((k_agg_q1<true>), dim3(FT_NBLOCKS),
-                               dim3(FT_NTHREADS), 0, g_stream, qty, ep, dc,
-                               tx, sd, rf, ls, pg.n_rows,
-                               (int32_t)p.preds[0].ival, p.key_vals[0][0],
-                               p.key_vals[0][1], p.key_vals[0][2],
-                               p.key_vals[1][0], p.key_vals[1][1], nullptr,
-                               (int64_t*)partials.p,
-                               (unsigned long long*)bad.p);
-        else
-            hipLaunchKernelGGL((k_agg_q1<false>), dim3(FT_NBLOCKS),
-                               dim3(FT_NTHREADS), 0, g_stream, qty, ep, dc,
-                               tx, sd, rf, ls, pg.n_rows,
-                               (int32_t)p.preds[0].ival, p.key_vals[0][0],
-                               p.key_vals[0][1], p.key_vals[0][2],
-                               p.key_vals[1][0], p.key_vals[1][1],
-                               (double*)partials.p, nullptr,
-                               (unsigned long long*)bad.p);
+#define LAUNCH_Q1(D, V)                                                  \
+    hipLaunchKernelGGL((k_agg_q1<D, V>), dim3(FT_NBLOCKS),               \
+                       dim3(FT_NTHREADS), 0, g_stream, qty, ep, dc, tx,  \
+                       sd, rf, ls, pg.n_rows, (int32_t)p.preds[0].ival,  \
+                       p.key_vals[0][0], p.key_vals[0][1],               \
+                       p.key_vals[0][2], p.key_vals[1][0],               \
+                       p.key_vals[1][1],                                 \
+                       D ? nullptr : (double*)partials.p,                \
+                       D ? (int64_t*)partials.p : nullptr,               \
+                       (unsigned long long*)bad.p)
+        if (dec) {
+            if (var == 1) LAUNCH_Q1(true, 1);
+            else if (var == 2) LAUNCH_Q1(true, 2);
+            else LAUNCH_Q1(true, 0);
+        } else {
+            if (var == 1) LAUNCH_Q1(false, 1);
+            else if (var == 2) LAUNCH_Q1(false, 2);
+            else LAUNCH_Q1(false, 0);
+        }
+#undef LAUNCH_Q1
     }
     template <int NA, int MAXG>
     void launch2(const pg_page& pg)
@@ -1843,14 +1880,8 @@ struct BuildOp : Op {
         if (!t->key_set_only) {
             t->head.alloc((size_t)cap * 4);
             t->next.alloc((size_t)(t->n_rows ? t->n_rows : 1) * 4);
-            t->acc_dec.alloc((size_t)cap * 8);
-            t->acc_fhi.alloc((size_t)cap * 8);
-            t->acc_flo.alloc((size_t)cap * 8);
-            t->acc_cnt.alloc((size_t)cap * 8);
-            t->acc_dec.zero();
-            t->acc_fhi.zero();
-            t->acc_flo.zero();
-            t->acc_cnt.zero();
+            t->acc.alloc((size_t)cap * sizeof(slot_acc));
+            t->acc.zero();
         }
         hipLaunchKernelGGL(k_tbl_init, dim3(1024), dim3(256), 0, g_stream,
                            (int64_t*)t->keys.p,
@@ -1901,10 +1932,7 @@ struct JoinOp : Op {
             hipLaunchKernelGGL(k_probe_agg, dim3(4096), dim3(256), 0,
                                g_stream, sp.pg, plan,
                                (const int64_t*)t->keys.p, t->mask,
-                               (unsigned long long*)t->acc_dec.p,
-                               (unsigned long long*)t->acc_fhi.p,
-                               (unsigned long long*)t->acc_flo.p,
-                               (unsigned long long*)t->acc_cnt.p);
+                               (slot_acc*)t->acc.p);
             hot_end();
             CHKV(hipStreamSynchronize(g_stream));
             return;
@@ -1994,7 +2022,7 @@ struct JoinOp : Op {
         DevBuf d_counts;
         d_counts.alloc(FLT_NB * 8);
         hipLaunchKernelGGL(k_groups_count, dim3(FLT_NB), dim3(256), 0,
-                           g_stream, (const unsigned long long*)t->acc_cnt.p,
+                           g_stream, (const slot_acc*)t->acc.p,
                            cap, chunk, (int64_t*)d_counts.p);
         std::vector<int64_t> h(FLT_NB);
         CHKV(hipMemcpyAsync(h.data(), d_counts.p, FLT_NB * 8,
@@ -2040,10 +2068,7 @@ struct JoinOp : Op {
         hipLaunchKernelGGL(k_groups_emit, dim3(FLT_NB), dim3(256), 0,
                            g_stream, (const int64_t*)t->keys.p,
                            (const int32_t*)t->head.p,
-                           (const unsigned long long*)t->acc_dec.p,
-                           (const unsigned long long*)t->acc_fhi.p,
-                           (const unsigned long long*)t->acc_flo.p,
-                           (const unsigned long long*)t->acc_cnt.p, bp, cap,
+                           (const slot_acc*)t->acc.p, bp, cap,
                            chunk, (const int64_t*)d_offs.p,
                            (int64_t*)op.pg.cols[c_key].data, pl_outs,
                            (int64_t*)op.pg.cols[c_dec].data,
