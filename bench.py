@@ -115,20 +115,25 @@ def gen_lineitem_device(orc, sf, device, want_orderkey=False,
     return cols, n
 
 
-def gen_orders_customer_device(orc, sf, device):
+def gen_orders_customer_device(orc, sf, device, rank=0, world=1):
+    """Rank shard of orders + customer at logical scale sf."""
     import torch
-    n_ord = orc.lib.tpch_orders_count(C.c_double(sf))
-    n_cust = orc.lib.tpch_customer_count(C.c_double(sf))
+    n_all = orc.lib.tpch_orders_count(C.c_double(sf))
+    o0, o1 = n_all * rank // world, n_all * (rank + 1) // world
+    n_ord = o1 - o0
+    nc_all = orc.lib.tpch_customer_count(C.c_double(sf))
+    c0, c1 = nc_all * rank // world, nc_all * (rank + 1) // world
+    n_cust = c1 - c0
     ok = np.empty(n_ord, np.int64)
     ck = np.empty(n_ord, np.int64)
     od = np.empty(n_ord, np.int32)
-    orc.lib.tpch_gen_orders(C.c_double(sf), C.c_int64(0), C.c_int64(n_ord),
+    orc.lib.tpch_gen_orders(C.c_double(sf), C.c_int64(o0), C.c_int64(n_ord),
                             C.c_void_p(ok.ctypes.data),
                             C.c_void_p(ck.ctypes.data),
                             C.c_void_p(od.ctypes.data), None)
     cck = np.empty(n_cust, np.int64)
     seg = np.empty(n_cust, np.uint8)
-    orc.lib.tpch_gen_customer(C.c_double(sf), C.c_int64(0),
+    orc.lib.tpch_gen_customer(C.c_double(sf), C.c_int64(c0),
                               C.c_int64(n_cust),
                               C.c_void_p(cck.ctypes.data),
                               C.c_void_p(seg.ctypes.data))

