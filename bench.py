@@ -188,6 +188,9 @@ def main():
     ap.add_argument("--sf", type=float, default=100.0)
     ap.add_argument("--mode", choices=["f64", "dec"], default="f64")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
+    ap.add_argument("--verify", action="store_true",
+                    help="after timing, check full-size results against the "
+                         "CPU oracle (exact decimal + bitwise f64)")
     args = ap.parse_args()
 
     import torch
@@ -300,6 +303,36 @@ def main():
         dist.all_reduce(e, op=dist.ReduceOp.MAX)
         elapsed = float(e.item())
 
+    if args.verify and rank == 0 and world == 1:
+        import numpy as _np
+        if args.query == "q1":
+            li_host = {k: v.cpu().numpy() for k, v in cols.items()}
+            exp = orc.q1(li_host)
+            got_d = pipelines.q1(page, mode="dec")
+            got_f = pipelines.q1(page, mode="f64")
+            for i, g in enumerate(exp):
+                assert got_d["count"][i] == g.count_order
+                assert got_d["sum_qty_lo"][i] == g.sum_qty_units
+                assert got_d["sum_base_lo"][i] == g.sum_base_cents
+                assert got_d["sum_disc_price_lo"][i] == g.sum_disc_1e4
+                assert got_d["sum_charge_hi"][i] == g.sum_charge_1e6_hi
+                assert _np.uint64(got_d["sum_charge_lo"][i].astype(
+                    _np.uint64)) == _np.uint64(g.sum_charge_1e6_lo)
+                assert got_f["sum_charge"][i].view(_np.int64) == _np.float64(
+                    g.f64_sum_charge).view(_np.int64)
+            log(f"verify q1 sf={args.sf}: exact decimal + bitwise f64 OK "
+                f"({len(exp)} groups, {n_rows} rows)")
+        else:
+            li_host = {k: v.cpu().numpy() for k, v in cols.items()}
+            oc = {k: v.cpu().numpy() for k, v in ocols.items()}
+            cc = {k: v.cpu().numpy() for k, v in ccols.items()}
+            exp3 = orc.q3(cc, oc, li_host)
+            got3 = step()
+            for i, r in enumerate(exp3):
+                assert got3["orderkey"][i] == r.orderkey
+                assert got3["revenue_1e4"][i] == r.revenue_1e4
+                assert got3["orderdate"][i] == r.orderdate
+            log(f"verify q3 sf={args.sf}: top-10 exact OK")
     hot_ms = float(lib.c.pg_last_hot_kernel_ms())
     ms_per_step = elapsed / args.steps * 1000.0
     value = total_rows_per_step * args.steps / elapsed
