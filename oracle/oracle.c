@@ -353,34 +353,63 @@ int32_t oracle_q3(int64_t n_cust, const int64_t* ck, const uint8_t* cseg,
     uint64_t* fhi = (uint64_t*)calloc(n_b, sizeof(uint64_t));
     uint64_t* flo = (uint64_t*)calloc(n_b, sizeof(uint64_t));
 
-    /* 3. lineitem filter (shipdate > 9204) + probe + grouped sum */
-#pragma omp parallel for schedule(static)
-    for (int64_t i = 0; i < n_li; i++) {
-        if (lsd[i] <= Q3_DATE) continue;
-        uint64_t h = pg_murmur3_finalize(pg_bigint_hash(lok[i]));
-        int64_t pos = (int64_t)(h & (cap - 1));
-        int64_t r = -1;
-        for (;;) {
-            int64_t s = slot[pos];
-            if (s == -1) break;
-            if (b_ok[s] == lok[i]) {
-                r = s;
-                break;
+    /* 3. lineitem filter (shipdate > 9204) + probe + grouped sum.
+     * Parallel: each thread accumulates into a private (rev,fhi,flo)
+     * shard for a contiguous row chunk, then shards merge serially —
+     * exact fixed-point addition is associative, so the result is
+     * identical to any order. */
+    {
+#ifdef _OPENMP
+        extern int omp_get_max_threads(void);
+        int nt = omp_get_max_threads();
+#else
+        int nt = 1;
+#endif
+        if (nt > 16) nt = 16; /* bound shard memory: nt * n_b * 24 B */
+        int64_t* trev = (int64_t*)calloc((size_t)nt * n_b, sizeof(int64_t));
+        uint64_t* thi = (uint64_t*)calloc((size_t)nt * n_b, sizeof(uint64_t));
+        uint64_t* tlo = (uint64_t*)calloc((size_t)nt * n_b, sizeof(uint64_t));
+#pragma omp parallel for schedule(static) num_threads(nt)
+        for (int64_t i = 0; i < n_li; i++) {
+#ifdef _OPENMP
+            extern int omp_get_thread_num(void);
+            int tid = omp_get_thread_num();
+#else
+            int tid = 0;
+#endif
+            if (lsd[i] <= Q3_DATE) continue;
+            uint64_t h = pg_murmur3_finalize(pg_bigint_hash(lok[i]));
+            int64_t pos = (int64_t)(h & (cap - 1));
+            int64_t r = -1;
+            for (;;) {
+                int64_t s = slot[pos];
+                if (s == -1) break;
+                if (b_ok[s] == lok[i]) {
+                    r = s;
+                    break;
+                }
+                pos = (pos + 1) & (cap - 1);
             }
-            pos = (pos + 1) & (cap - 1);
+            if (r == -1) continue;
+            int64_t cents = (int64_t)(lep[i] * 100.0 + 0.5);
+            int64_t d = (int64_t)(ldisc[i] * 100.0 + 0.5);
+            int64_t ticks = cents * (100 - d);
+            double p = lep[i] * (1.0 - ldisc[i]);
+            uint64_t phi, plo;
+            fx128_from_f64(p, &phi, &plo);
+            size_t o = (size_t)tid * n_b + r;
+            trev[o] += ticks;
+            fx128_add(&thi[o], &tlo[o], phi, plo);
         }
-        if (r == -1) continue;
-        int64_t cents = (int64_t)(lep[i] * 100.0 + 0.5);
-        int64_t d = (int64_t)(ldisc[i] * 100.0 + 0.5);
-        int64_t ticks = cents * (100 - d);
-        double p = lep[i] * (1.0 - ldisc[i]);
-        uint64_t phi, plo;
-        fx128_from_f64(p, &phi, &plo);
-#pragma omp critical
-        {
-            rev4[r] += ticks;
-            fx128_add(&fhi[r], &flo[r], phi, plo);
-        }
+        for (int t = 0; t < nt; t++)
+            for (int64_t r = 0; r < n_b; r++) {
+                size_t o = (size_t)t * n_b + r;
+                rev4[r] += trev[o];
+                fx128_add(&fhi[r], &flo[r], thi[o], tlo[o]);
+            }
+        free(trev);
+        free(thi);
+        free(tlo);
     }
 
     /* 4. TopN by (revenue desc, orderdate asc, orderkey asc) — insertion
