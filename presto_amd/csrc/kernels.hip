@@ -1939,6 +1939,18 @@ struct JoinOp : Op {
         }
         /* emit mode: count -> scan -> emit */
         int64_t n = sp.pg.n_rows;
+        if (n == 0) { /* empty probe page -> empty output page */
+            OutPage op;
+            op.pg.n_rows = 0;
+            op.pg.n_cols = plan.n_emit + (int32_t)t->payload.size();
+            for (int c = 0; c < op.pg.n_cols; c++) {
+                op.pg.cols[c].tag = PG_T_I64;
+                op.pg.cols[c].on_device = 1;
+                op.pg.cols[c].data = nullptr;
+            }
+            outq.push_back(std::move(op));
+            return;
+        }
         DevBuf counts, offs, ctot;
         counts.alloc((size_t)n * 4);
         offs.alloc((size_t)n * 8);
