@@ -193,8 +193,11 @@ def main():
                          "CPU oracle (exact decimal + bitwise f64)")
     args = ap.parse_args()
 
-    import torch
     world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world > 1 and "OMP_NUM_THREADS" not in os.environ:
+        os.environ["OMP_NUM_THREADS"] = str(
+            max(1, (os.cpu_count() or 8) // world))
+    import torch
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     n_gpus = max(world, args.gpus if world == 1 else world)
@@ -246,15 +249,21 @@ def main():
                                         "get_output", "destroy")):
                     phase_t[nm] = phase_t.get(nm, 0) + (t[i + 1] - t[i])
             if world > 1:
-                # partial->final: gather tiny group sums, combine in rank
-                # order (HashAggregationOperator partial/final Step split)
+                # partial->final: gather the tiny per-rank group partials
+                # (fixed 6x8 f64-bits buffer, zero-padded) and combine on
+                # every rank in rank order — the partial/final Step split
+                # of HashAggregationOperator.java:72
                 import torch.distributed as dist
-                flat = torch.tensor(
-                    np.concatenate([out[k].view(np.float64) for k in out
-                                    if k not in ("returnflag", "linestatus")]
-                                   ).view(np.float64), device=device)
-                allp = [torch.empty_like(flat) for _ in range(world)]
-                dist.all_gather(allp, flat)
+                buf = torch.zeros(6 * 8, dtype=torch.float64, device=device)
+                vals = [k for k in out
+                        if k not in ("returnflag", "linestatus")][:8]
+                ng = len(out["returnflag"])
+                for c, k in enumerate(vals):
+                    buf[c * 6:c * 6 + ng] = torch.from_numpy(
+                        np.ascontiguousarray(out[k]).view(np.float64)
+                    ).to(device)
+                allp = [torch.empty_like(buf) for _ in range(world)]
+                dist.all_gather(allp, buf)
             return out
         total_rows_per_step = n_rows * n_gpus
         workload = Q1_WORKLOAD.format(sf=int(sf_total), n=n_gpus)

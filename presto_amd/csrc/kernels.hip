@@ -159,6 +159,15 @@ extern "C" double pg_last_hot_kernel_ms(void) { return g_last_hot_ms; }
 #define FT_VL ((int64_t)FT_NBLOCKS * FT_NTHREADS)
 static const int64_t TBL_EMPTY = INT64_MIN; /* reserved key (documented) */
 
+/* one-byte tag per slot (the PagesHash.java:110 byte-tag idea, adapted):
+ * tag = high byte of the bucket hash, forced nonzero; 0 = empty slot.
+ * The tag array is 1/8 the key array and stays L3-resident, so the ~90%
+ * probe misses of Q3 are rejected without touching the HBM key lines. */
+__device__ __host__ inline uint8_t d_tbl_tag(uint64_t h)
+{
+    return (uint8_t)((h >> 56) | 1);
+}
+
 __device__ inline double d_load_f64(const pg_col& c, int64_t i)
 {
     switch (c.tag) {
@@ -782,6 +791,7 @@ __global__ __launch_bounds__(256) void k_set_insert(const int64_t* in_keys,
 /* insert build rows (already filtered/compacted) with chains */
 __global__ __launch_bounds__(256) void k_tbl_insert(const int64_t* in_keys,
                                                     int64_t n, int64_t* keys,
+                                                    uint8_t* tags,
                                                     int32_t* head,
                                                     int32_t* next,
                                                     int64_t mask)
@@ -797,6 +807,7 @@ __global__ __launch_bounds__(256) void k_tbl_insert(const int64_t* in_keys,
                                     (unsigned long long)TBL_EMPTY,
                                     (unsigned long long)key);
             if (old == TBL_EMPTY || old == key) {
+                tags[s] = d_tbl_tag(h); /* probes start after kernel sync */
                 int32_t prev = atomicExch(&head[s], (int32_t)i);
                 next[i] = prev;
                 break;
@@ -819,6 +830,22 @@ __device__ inline int64_t d_tbl_find(const int64_t* keys, int64_t mask,
     }
 }
 
+
+__device__ inline int64_t d_tbl_find_tagged(const int64_t* keys,
+                                            const uint8_t* tags,
+                                            int64_t mask, int64_t key)
+{
+    uint64_t h = pg_murmur3_finalize(pg_bigint_hash(key));
+    int64_t s = (int64_t)(h & (uint64_t)mask);
+    uint8_t tag = d_tbl_tag(h);
+    for (;;) {
+        uint8_t t = tags[s];
+        if (t == 0) return -1;
+        if (t == tag && keys[s] == key) return s;
+        s = (s + 1) & mask;
+    }
+}
+
 /* per-slot accumulators, interleaved so one probe hit touches ONE cache
  * line (32 B of one 64-B line) instead of four separate arrays */
 struct slot_acc {
@@ -833,15 +860,15 @@ struct slot_acc {
  * exactly in decimal ticks AND in 64.64 fixed point — order-independent,
  * so atomics preserve bit-determinism) */
 __global__ __launch_bounds__(256) void k_probe_agg(
-    pg_page pg, pg_plan_lookup_join plan, const int64_t* keys, int64_t mask,
-    slot_acc* acc)
+    pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
+    const uint8_t* tags, int64_t mask, slot_acc* acc)
 {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; i < pg.n_rows; i += stride) {
         if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
         int64_t key = d_load_i64(pg.cols[plan.key_col], i);
-        int64_t s = d_tbl_find(keys, mask, key);
+        int64_t s = d_tbl_find_tagged(keys, tags, mask, key);
         if (s < 0) continue;
         pg_agg ag;
         ag.proj = plan.proj;
@@ -860,7 +887,8 @@ __global__ __launch_bounds__(256) void k_probe_agg(
 /* probe match counting (emit mode, pass 1): count[i] = chain length */
 __global__ __launch_bounds__(256) void k_probe_count(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
-    const int32_t* head, const int32_t* next, int64_t mask, int32_t* counts)
+    const uint8_t* tags, const int32_t* head, const int32_t* next,
+    int64_t mask, int32_t* counts)
 {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -868,7 +896,7 @@ __global__ __launch_bounds__(256) void k_probe_count(
         int32_t c = 0;
         if (d_eval_preds(pg, plan.preds, plan.n_preds, i)) {
             int64_t key = d_load_i64(pg.cols[plan.key_col], i);
-            int64_t s = d_tbl_find(keys, mask, key);
+            int64_t s = d_tbl_find_tagged(keys, tags, mask, key);
             if (s >= 0)
                 for (int32_t r = head[s]; r >= 0; r = next[r]) c++;
         }
@@ -884,16 +912,16 @@ struct build_payloads {
 };
 __global__ __launch_bounds__(256) void k_probe_emit(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
-    const int32_t* head, const int32_t* next, int64_t mask,
-    const int64_t* offsets, emit_outs probe_outs, build_payloads bp,
-    emit_outs build_outs)
+    const uint8_t* tags, const int32_t* head, const int32_t* next,
+    int64_t mask, const int64_t* offsets, emit_outs probe_outs,
+    build_payloads bp, emit_outs build_outs)
 {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; i < pg.n_rows; i += stride) {
         if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
         int64_t key = d_load_i64(pg.cols[plan.key_col], i);
-        int64_t s = d_tbl_find(keys, mask, key);
+        int64_t s = d_tbl_find_tagged(keys, tags, mask, key);
         if (s < 0) continue;
         int64_t pos = offsets[i];
         for (int32_t r = head[s]; r >= 0; r = next[r]) {
@@ -1389,7 +1417,7 @@ struct OutPage {
 struct Table {
     int64_t cap = 0, mask = 0, n_rows = 0;
     bool key_set_only = false;
-    DevBuf keys, head, next;
+    DevBuf keys, head, next, tags;
     DevBuf acc; /* slot_acc[cap], interleaved */
     /* compacted build-row arrays: key + payloads */
     DevBuf key_rows;
@@ -1880,6 +1908,8 @@ struct BuildOp : Op {
         if (!t->key_set_only) {
             t->head.alloc((size_t)cap * 4);
             t->next.alloc((size_t)(t->n_rows ? t->n_rows : 1) * 4);
+            t->tags.alloc((size_t)cap);
+            t->tags.zero();
             t->acc.alloc((size_t)cap * sizeof(slot_acc));
             t->acc.zero();
         }
@@ -1896,6 +1926,7 @@ struct BuildOp : Op {
                 hipLaunchKernelGGL(k_tbl_insert, dim3(2048), dim3(256), 0,
                                    g_stream, (const int64_t*)t->key_rows.p,
                                    t->n_rows, (int64_t*)t->keys.p,
+                                   (uint8_t*)t->tags.p,
                                    (int32_t*)t->head.p, (int32_t*)t->next.p,
                                    t->mask);
         }
@@ -1931,7 +1962,8 @@ struct JoinOp : Op {
             hot_begin();
             hipLaunchKernelGGL(k_probe_agg, dim3(4096), dim3(256), 0,
                                g_stream, sp.pg, plan,
-                               (const int64_t*)t->keys.p, t->mask,
+                               (const int64_t*)t->keys.p,
+                               (const uint8_t*)t->tags.p, t->mask,
                                (slot_acc*)t->acc.p);
             hot_end();
             CHKV(hipStreamSynchronize(g_stream));
@@ -1958,6 +1990,7 @@ struct JoinOp : Op {
         ctot.alloc((size_t)nchunks * 8);
         hipLaunchKernelGGL(k_probe_count, dim3(4096), dim3(256), 0, g_stream,
                            sp.pg, plan, (const int64_t*)t->keys.p,
+                           (const uint8_t*)t->tags.p,
                            (const int32_t*)t->head.p,
                            (const int32_t*)t->next.p, t->mask,
                            (int32_t*)counts.p);
@@ -2017,6 +2050,7 @@ struct JoinOp : Op {
         op.pg.n_cols = nc;
         hipLaunchKernelGGL(k_probe_emit, dim3(4096), dim3(256), 0, g_stream,
                            sp.pg, plan, (const int64_t*)t->keys.p,
+                           (const uint8_t*)t->tags.p,
                            (const int32_t*)t->head.p,
                            (const int32_t*)t->next.p, t->mask,
                            (const int64_t*)offs.p, pouts, bp, bouts);
