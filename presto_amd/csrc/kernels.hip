@@ -807,7 +807,7 @@ __global__ __launch_bounds__(256) void k_tbl_insert(const int64_t* in_keys,
                                     (unsigned long long)TBL_EMPTY,
                                     (unsigned long long)key);
             if (old == TBL_EMPTY || old == key) {
-                tags[s] = d_tbl_tag(h); /* probes start after kernel sync */
+                if (tags) tags[s] = d_tbl_tag(h); /* probes run post-sync */
                 int32_t prev = atomicExch(&head[s], (int32_t)i);
                 next[i] = prev;
                 break;
@@ -835,6 +835,7 @@ __device__ inline int64_t d_tbl_find_tagged(const int64_t* keys,
                                             const uint8_t* tags,
                                             int64_t mask, int64_t key)
 {
+    if (!tags) return d_tbl_find(keys, mask, key);
     uint64_t h = pg_murmur3_finalize(pg_bigint_hash(key));
     int64_t s = (int64_t)(h & (uint64_t)mask);
     uint8_t tag = d_tbl_tag(h);
@@ -1908,8 +1909,14 @@ struct BuildOp : Op {
         if (!t->key_set_only) {
             t->head.alloc((size_t)cap * 4);
             t->next.alloc((size_t)(t->n_rows ? t->n_rows : 1) * 4);
-            t->tags.alloc((size_t)cap);
-            t->tags.zero();
+            /* byte tags pay off only when the key array is far beyond the
+             * 256 MiB L3 (measured: at SF100's 256 MB keys the insert-side
+             * extra random store costs more than the probe saves); enable
+             * for tables past 64M slots. */
+            if (cap >= (64ll << 20)) {
+                t->tags.alloc((size_t)cap);
+                t->tags.zero();
+            }
             t->acc.alloc((size_t)cap * sizeof(slot_acc));
             t->acc.zero();
         }
