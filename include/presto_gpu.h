@@ -178,6 +178,14 @@ typedef struct {
     int32_t payload_col[4];     /* i32/i64 payload channels stored per row */
     int64_t capacity_hint;      /* expected distinct build rows */
     int32_t key_set_only;       /* 1: build a key SET (no payload slots) */
+    int32_t agg_table;          /* 1: table feeds a fused-agg probe only
+                                   (LOOKUP_JOIN mode 1).  Rows are inserted
+                                   directly during addInput (single scan,
+                                   payloads stored per slot, keys must be
+                                   unique); join-emit probing (chains) is
+                                   rejected.  capacity_hint must be >= the
+                                   number of inserted rows (errors out
+                                   otherwise). */
 } pg_plan_hash_build;
 
 typedef struct {

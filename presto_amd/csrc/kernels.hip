@@ -817,6 +817,76 @@ __global__ __launch_bounds__(256) void k_tbl_insert(const int64_t* in_keys,
     }
 }
 
+/* direct insert for agg tables: filter+semijoin+insert in ONE scan of the
+ * build input; payloads stored per SLOT (keys unique).  count tracks
+ * inserted rows; full table -> give-up flag (bounded probe loop). */
+struct direct_payloads {
+    void* ptr[4];
+    int32_t tag[4];
+    int32_t src[4];
+    int32_t n;
+};
+__global__ __launch_bounds__(256) void k_tbl_insert_direct(
+    pg_page pg, pg_plan_hash_build plan, const int64_t* set_keys,
+    int64_t set_mask, int64_t* keys, uint8_t* tags, direct_payloads dp,
+    int64_t mask, unsigned long long* inserted,
+    unsigned long long* overflow)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < pg.n_rows; i += stride) {
+        if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
+        if (set_keys) {
+            int64_t sk = d_load_i64(pg.cols[plan.semijoin_col], i);
+            uint64_t h = pg_murmur3_finalize(pg_bigint_hash(sk));
+            int64_t s = (int64_t)(h & (uint64_t)set_mask);
+            bool found = false;
+            for (;;) {
+                int64_t k = set_keys[s];
+                if (k == sk) { found = true; break; }
+                if (k == TBL_EMPTY) break;
+                s = (s + 1) & set_mask;
+            }
+            if (!found) continue;
+        }
+        int64_t key = d_load_i64(pg.cols[plan.key_col], i);
+        uint64_t h = pg_murmur3_finalize(pg_bigint_hash(key));
+        int64_t s = (int64_t)(h & (uint64_t)mask);
+        int64_t tries = 0;
+        for (;;) {
+            int64_t old = atomicCAS((unsigned long long*)&keys[s],
+                                    (unsigned long long)TBL_EMPTY,
+                                    (unsigned long long)key);
+            if (old == TBL_EMPTY || old == key) {
+                if (old == TBL_EMPTY) {
+                    atomicAdd(inserted, 1ull);
+                    if (tags) tags[s] = d_tbl_tag(h);
+                }
+                for (int o = 0; o < dp.n; o++) {
+                    const pg_col& c = pg.cols[dp.src[o]];
+                    switch (dp.tag[o]) {
+                        case PG_T_I32:
+                            ((int32_t*)dp.ptr[o])[s] =
+                                (int32_t)d_load_i64(c, i);
+                            break;
+                        case PG_T_I64:
+                            ((int64_t*)dp.ptr[o])[s] = d_load_i64(c, i);
+                            break;
+                        default:
+                            ((double*)dp.ptr[o])[s] = d_load_f64(c, i);
+                    }
+                }
+                break;
+            }
+            if (++tries > mask) { /* table full: bounded give-up */
+                atomicAdd(overflow, 1ull);
+                break;
+            }
+            s = (s + 1) & mask;
+        }
+    }
+}
+
 __device__ inline int64_t d_tbl_find(const int64_t* keys, int64_t mask,
                                      int64_t key)
 {
@@ -910,6 +980,7 @@ struct build_payloads {
     const void* ptr[4];
     int32_t tag[4];
     int32_t n;
+    int32_t by_slot; /* 1: payload arrays indexed by slot, not build row */
 };
 __global__ __launch_bounds__(256) void k_probe_emit(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
@@ -1006,7 +1077,7 @@ __global__ __launch_bounds__(256) void k_groups_emit(
         if (sel) {
             int64_t pos = woff + __popcll(m & ((1ull << lane) - 1));
             out_key[pos] = keys[i];
-            int32_t r = head[i];
+            int64_t r = bp.by_slot ? i : (int64_t)head[i];
             for (int o = 0; o < bp.n; o++) {
                 switch (bp.tag[o]) {
                     case PG_T_I32:
@@ -1420,6 +1491,7 @@ struct Table {
     bool key_set_only = false;
     DevBuf keys, head, next, tags;
     DevBuf acc; /* slot_acc[cap], interleaved */
+    bool slot_payloads = false; /* payloads indexed by slot (agg tables) */
     /* compacted build-row arrays: key + payloads */
     DevBuf key_rows;
     std::vector<DevBuf> payload;
@@ -1817,10 +1889,36 @@ struct BuildOp : Op {
     int64_t tbl = -1;
     std::unique_ptr<Table> t;
     int64_t cap_rows = 0;
+    DevBuf counters; /* direct mode: [inserted, overflow] */
     void init()
     {
         t.reset(new Table());
         t->key_set_only = plan.key_set_only != 0;
+        if (plan.agg_table) {
+            /* direct mode: size the table now from the hint; payloads
+             * live per slot */
+            t->slot_payloads = true;
+            int64_t cap = next_pow2(plan.capacity_hint * 2 + 16);
+            t->cap = cap;
+            t->mask = cap - 1;
+            t->keys.alloc((size_t)cap * 8);
+            if (cap >= (64ll << 20)) {
+                t->tags.alloc((size_t)cap);
+                t->tags.zero();
+            }
+            t->acc.alloc((size_t)cap * sizeof(slot_acc));
+            t->acc.zero();
+            counters.alloc(16);
+            counters.zero();
+            hipLaunchKernelGGL(k_tbl_init, dim3(1024), dim3(256), 0,
+                               g_stream, (int64_t*)t->keys.p, nullptr, cap);
+            for (int i = 0; i < plan.n_payload; i++) {
+                t->payload.emplace_back();
+                t->ptag.push_back(-1);
+                t->payload.back().alloc((size_t)cap * 8);
+            }
+            return;
+        }
         cap_rows = plan.capacity_hint > 16 ? plan.capacity_hint : 16;
         t->key_rows.alloc((size_t)cap_rows * 8);
         for (int i = 0; i < plan.n_payload; i++) {
@@ -1852,6 +1950,15 @@ struct BuildOp : Op {
         }
         cap_rows = nc;
     }
+    const Table* semi_table()
+    {
+        if (plan.semijoin_table < 0) return nullptr;
+        std::lock_guard<std::mutex> lk(g_mu);
+        auto it = g_tables.find(plan.semijoin_table);
+        if (it == g_tables.end())
+            throw std::runtime_error("semijoin table not found");
+        return it->second.get();
+    }
     void add_input(const pg_page* in) override
     {
         StagedPage sp;
@@ -1860,6 +1967,27 @@ struct BuildOp : Op {
         for (int i = 0; i < plan.n_payload; i++)
             if (t->ptag[i] < 0)
                 t->ptag[i] = sp.pg.cols[plan.payload_col[i]].tag;
+        if (plan.agg_table) {
+            const Table* semi = semi_table();
+            direct_payloads dp{};
+            dp.n = plan.n_payload;
+            for (int o = 0; o < dp.n; o++) {
+                dp.ptr[o] = t->payload[o].p;
+                dp.tag[o] = t->ptag[o];
+                dp.src[o] = plan.payload_col[o];
+            }
+            hot_begin();
+            hipLaunchKernelGGL(k_tbl_insert_direct, dim3(4096), dim3(256),
+                               0, g_stream, sp.pg, plan,
+                               semi ? (const int64_t*)semi->keys.p : nullptr,
+                               semi ? semi->mask : 0, (int64_t*)t->keys.p,
+                               (uint8_t*)t->tags.p, dp, t->mask,
+                               (unsigned long long*)counters.p,
+                               (unsigned long long*)counters.p + 1);
+            hot_end();
+            CHKV(hipStreamSynchronize(g_stream));
+            return;
+        }
         pg_plan_filter_project fp{};
         fp.n_preds = plan.n_preds;
         memcpy(fp.preds, plan.preds, sizeof(fp.preds));
@@ -1902,6 +2030,21 @@ struct BuildOp : Op {
     }
     void finish() override
     {
+        if (plan.agg_table) {
+            unsigned long long c[2];
+            CHKV(hipMemcpy(c, counters.p, 16, hipMemcpyDeviceToHost));
+            if (c[1])
+                throw std::runtime_error(
+                    "agg table overflow: capacity_hint too small");
+            if ((int64_t)c[0] * 2 > t->cap)
+                throw std::runtime_error(
+                    "agg table fill exceeded 0.5: raise capacity_hint");
+            t->n_rows = (int64_t)c[0];
+            std::lock_guard<std::mutex> lk(g_mu);
+            tbl = g_next_table++;
+            g_tables[tbl] = std::move(t);
+            return;
+        }
         int64_t cap = next_pow2(t->n_rows * 2 + 16); /* fill <= 0.5 */
         t->cap = cap;
         t->mask = cap - 1;
@@ -1960,6 +2103,9 @@ struct JoinOp : Op {
         if (t->key_set_only)
             throw std::runtime_error(
                 "cannot probe a key-set-only table");
+        if (t->slot_payloads && plan.mode == 0)
+            throw std::runtime_error(
+                "agg_table supports fused-agg probing only (mode 1)");
     }
     void add_input(const pg_page* in) override
     {
@@ -2105,6 +2251,7 @@ struct JoinOp : Op {
         int c_key = add_dev_col(PG_T_I64);
         build_payloads bp{};
         bp.n = (int32_t)t->payload.size();
+        bp.by_slot = t->slot_payloads ? 1 : 0;
         emit_outs pl_outs{};
         pl_outs.n = bp.n;
         for (int o = 0; o < bp.n; o++) {

@@ -179,6 +179,7 @@ def q3_distributed(cust_page, ord_page, li_page, world, rank, device,
     b2p.n_payload = 1
     b2p.payload_col[0] = 1
     b2p.capacity_hint = max(int(ocols["c0"].numel()) // 4, 16)
+    b2p.agg_table = 1
     b2 = P.Operator(P.OP_HASH_BUILD, b2p)
     b2.add_input(P.Page({k: v for k, v in ocols.items()}))
     b2.finish()

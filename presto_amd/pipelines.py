@@ -113,6 +113,7 @@ class Q3Pipeline:
         b2p.payload_col[0] = orders.channel("orderdate")
         b2p.capacity_hint = max(orders.n_rows // 8, 16)
         b2p.key_set_only = 0
+        b2p.agg_table = 1  # direct single-scan insert (fused-agg probe only)
         self.b2 = Operator(OP_HASH_BUILD, b2p)
         self.b2.add_input(orders)
         self.b2.finish()
