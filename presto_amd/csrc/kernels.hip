@@ -834,6 +834,7 @@ __global__ __launch_bounds__(256) void k_tbl_insert_direct(
 {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    int64_t my_inserted = 0, my_overflow = 0;
     for (; i < pg.n_rows; i += stride) {
         if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
         if (set_keys) {
@@ -859,7 +860,7 @@ __global__ __launch_bounds__(256) void k_tbl_insert_direct(
                                     (unsigned long long)key);
             if (old == TBL_EMPTY || old == key) {
                 if (old == TBL_EMPTY) {
-                    atomicAdd(inserted, 1ull);
+                    my_inserted++;
                     if (tags) tags[s] = d_tbl_tag(h);
                 }
                 for (int o = 0; o < dp.n; o++) {
@@ -879,11 +880,19 @@ __global__ __launch_bounds__(256) void k_tbl_insert_direct(
                 break;
             }
             if (++tries > mask) { /* table full: bounded give-up */
-                atomicAdd(overflow, 1ull);
+                my_overflow++;
                 break;
             }
             s = (s + 1) & mask;
         }
+    }
+    /* one atomic per wave, not per insert (a single-address atomicAdd per
+     * row serializes the whole grid) */
+    my_inserted = d_bfly_i64(my_inserted);
+    my_overflow = d_bfly_i64(my_overflow);
+    if ((threadIdx.x & 63) == 0) {
+        if (my_inserted) atomicAdd(inserted, (unsigned long long)my_inserted);
+        if (my_overflow) atomicAdd(overflow, (unsigned long long)my_overflow);
     }
 }
 
