@@ -4,7 +4,9 @@
  *
  * Purpose: order-independent, bit-deterministic summation of non-negative
  * f64 values (the revenue/charge sums of TPC-H Q1/Q3).  Every f64 value in
- * the domain (0 <= p < 2^53, ulp(p) >= 2^-64) converts EXACTLY to 64.64
+ * the domain (0 <= p < 2^53, ulp(p) >= 2^-64 — i.e. p = 0 or p >= ~2^-12,
+ * which holds for all money sums here: Q3 revenue products are >= ~810)
+ * converts EXACTLY to 64.64
  * fixed point, so the accumulated 128-bit sum is the exact real-number sum
  * of the f64 addends, independent of addition order; converting the total
  * back to f64 applies exactly one correctly-rounded (RNE) rounding.

@@ -322,3 +322,50 @@ def test_q3_distributed_graph_world1(P, oracle_lib, sf01):
         assert out["orderkey"][i] == r.orderkey
         assert out["rev"][i] == r.revenue_1e4
         assert out["orderdate"][i] == r.orderdate
+
+
+def test_filter_null_mask(P):
+    """null_mask positions are excluded by predicates — PageProcessor
+    null-comparison semantics (a null comparison never selects the row)."""
+    import ctypes as C
+    import numpy as np
+    n = 10_000
+    a = np.arange(n, dtype=np.int64)
+    nulls = (np.arange(n) % 7 == 0).astype(np.uint8)
+    plan = P.PlanFilterProject()
+    plan.n_preds = 1
+    plan.preds[0] = P.Pred(0, P.CMP_GE, 0, 0.0)  # all rows pass, except nulls
+    plan.n_proj = 1
+    plan.proj[0] = P.Proj(P.PROJ_IDENT, 0, 0, 0)
+    op = P.Operator(P.OP_FILTER_PROJECT, plan)
+    # build the page by hand to attach the null mask
+    page = P.Page({"a": a})
+    cp = page.to_c()
+    cp.cols[0].null_mask = a.ctypes.data_as(C.c_void_p)  # placeholder
+    cp.cols[0].null_mask = nulls.ctypes.data_as(C.c_void_p).value
+    op.add_input_raw(cp)
+    out = op.get_output(["a"])
+    op.destroy()
+    expect = a[nulls == 0]
+    assert np.array_equal(out["a"], expect)
+
+
+def test_agg_table_overflow_errors(P):
+    """agg_table with a too-small capacity_hint must raise, not hang
+    (bounded probe give-up in k_tbl_insert_direct)."""
+    import numpy as np
+    keys = np.arange(1, 2001, dtype=np.int64)
+    pay = np.arange(2000, dtype=np.int32)
+    plan = P.PlanHashBuild()
+    plan.key_col = 0
+    plan.semijoin_table = -1
+    plan.n_payload = 1
+    plan.payload_col[0] = 1
+    plan.capacity_hint = 16  # cap 64 slots << 2000 keys
+    plan.agg_table = 1
+    b = P.Operator(P.OP_HASH_BUILD, plan)
+    b.add_input(P.Page({"k": keys, "p": pay}))
+    import pytest as _pytest
+    with _pytest.raises(RuntimeError):
+        b.finish()
+    b.destroy()

@@ -161,3 +161,49 @@ def test_join_random_vs_python(oracle_lib):
         for b in idx.get(k, []):
             expect.add((i, b))
     assert got == expect
+
+
+def test_fx128_exact_roundtrip(oracle_lib):
+    """fx128 (shared header, presto_amd/csrc/fixed128.h) must equal native
+    exact arithmetic: for values in the domain (ulp >= 2^-64, i.e.
+    |v| >= ~2^-12 or exactly-representable dyadics), fx128_to_f64(sum of
+    exact fixed-point reprs) == correctly-rounded sum via Fraction.
+    Q3 revenue products are >= ~810, far inside the domain."""
+    import ctypes as C
+    from fractions import Fraction
+    import subprocess, tempfile, textwrap, os
+    # drive the header through a tiny C harness compiled on the fly
+    src = textwrap.dedent("""
+        #include <stdio.h>
+        #include <stdint.h>
+        #include "presto_amd/csrc/fixed128.h"
+        int main(void) {
+            double vals[6] = {0.0625, 123456.78, 810.0, 99999.99, 0.01, 3.5};
+            uint64_t hi = 0, lo = 0;
+            for (int i = 0; i < 6; i++) {
+                uint64_t h, l;
+                fx128_from_f64(vals[i], &h, &l);
+                fx128_add(&hi, &lo, h, l);
+            }
+            printf("%llu %llu %.17g\\n", (unsigned long long)hi,
+                   (unsigned long long)lo, fx128_to_f64(hi, lo));
+            return 0;
+        }
+    """)
+    import pathlib
+    repo = pathlib.Path(__file__).resolve().parent.parent
+    with tempfile.TemporaryDirectory() as d:
+        cfile = os.path.join(d, "t.c")
+        open(cfile, "w").write(src)
+        exe = os.path.join(d, "t")
+        subprocess.run(["gcc", "-O2", "-I", str(repo), cfile, "-o", exe],
+                       check=True)
+        out = subprocess.run([exe], capture_output=True, text=True,
+                             check=True).stdout.split()
+    hi, lo, back = int(out[0]), int(out[1]), float(out[2])
+    vals = [0.0625, 123456.78, 810.0, 99999.99, 0.01, 3.5]
+    exact = sum(Fraction(v) for v in vals)  # Fraction(float) is exact
+    got = Fraction(hi) + Fraction(lo, 1 << 64)
+    assert got == exact  # conversion+sum is exact
+    # correctly rounded back-conversion matches python float of the exact sum
+    assert back == float(exact)
