@@ -102,8 +102,10 @@ def q3_distributed(cust_page, ord_page, li_page, world, rank, device,
     ck = _col_to_torch(raw.cols[0], raw.n_rows, device)
     f.destroy()
     if use_dist:
-        cap = cust_page.n_rows  # selected <= local rows; same per rank
-        parts = _gather_padded(ck, cap, device)
+        # cap must agree across ranks (shards differ by +-1 row)
+        capt = torch.tensor([int(cust_page.n_rows)], device=device)
+        dist.all_reduce(capt, op=dist.ReduceOp.MAX)
+        parts = _gather_padded(ck, int(capt.item()), device)
         ck = torch.cat(parts)
     bp = P.PlanHashBuild()
     bp.n_preds = 1
