@@ -203,8 +203,8 @@ def main():
     n_gpus = max(world, args.gpus if world == 1 else world)
     if world > 1:
         import torch.distributed as dist
-        dist.init_process_group("nccl")
         torch.cuda.set_device(local_rank)
+        dist.init_process_group("nccl")
     device = torch.device("cuda", local_rank)
 
     import presto_amd
