@@ -964,6 +964,105 @@ __global__ __launch_bounds__(256) void k_probe_agg(
     }
 }
 
+/* specialized fused probe+agg for the Q3 shape (the codegen-analog
+ * specialization, like k_agg_q1): optional single int32 predicate, i64
+ * key column, DISC_PRICE(a,b) projection over two f64 columns.  Streaming
+ * columns are read once per row PAIR with nontemporal 16-byte vector
+ * loads; the two rows' probe chains interleave for memory-level
+ * parallelism. */
+__global__ __launch_bounds__(256) void k_probe_agg_q3(
+    const int32_t* sd /* nullable pred col */, int32_t pred_op,
+    int32_t pred_val, const int64_t* okey, const double* ep,
+    const double* dc, int64_t n, const int64_t* keys, const uint8_t* tags,
+    int64_t mask, slot_acc* acc)
+{
+    typedef double vd2 __attribute__((ext_vector_type(2)));
+    typedef int vi2 __attribute__((ext_vector_type(2)));
+    typedef long vl2 __attribute__((ext_vector_type(2)));
+    int64_t base0 = 2 * ((int64_t)blockIdx.x * blockDim.x + threadIdx.x);
+    int64_t stride = 2 * (int64_t)gridDim.x * blockDim.x;
+    for (int64_t base = base0; base < n; base += stride) {
+        const bool pair = base + 1 < n;
+        int64_t k0, k1;
+        double e0, e1, d0, d1;
+        int32_t s0 = 0, s1 = 0;
+        if (pair) {
+            vl2 kk = __builtin_nontemporal_load((const vl2*)(okey + base));
+            vd2 ee = __builtin_nontemporal_load((const vd2*)(ep + base));
+            vd2 dd = __builtin_nontemporal_load((const vd2*)(dc + base));
+            k0 = kk[0]; k1 = kk[1];
+            e0 = ee[0]; e1 = ee[1];
+            d0 = dd[0]; d1 = dd[1];
+            if (sd) {
+                vi2 ss = __builtin_nontemporal_load((const vi2*)(sd + base));
+                s0 = ss[0]; s1 = ss[1];
+            }
+        } else {
+            k0 = okey[base]; e0 = ep[base]; d0 = dc[base];
+            if (sd) s0 = sd[base];
+            k1 = 0; e1 = 0; d1 = 0; s1 = pred_val; /* fails GT pred */
+        }
+        bool sel0 = true, sel1 = pair;
+        if (sd) {
+            sel0 = pred_op == PG_CMP_GT ? s0 > pred_val : s0 < pred_val;
+            sel1 = sel1 && (pred_op == PG_CMP_GT ? s1 > pred_val
+                                                 : s1 < pred_val);
+        }
+        /* interleaved probes: issue both hash chains */
+        int64_t slot0 = -1, slot1 = -1;
+        if (sel0) {
+            uint64_t h = pg_murmur3_finalize(pg_bigint_hash(k0));
+            int64_t p0 = (int64_t)(h & (uint64_t)mask);
+            uint8_t tg = tags ? d_tbl_tag(h) : 0;
+            for (;;) {
+                if (tags) {
+                    uint8_t t = tags[p0];
+                    if (t == 0) break;
+                    if (t == tg && keys[p0] == k0) { slot0 = p0; break; }
+                } else {
+                    int64_t k = keys[p0];
+                    if (k == k0) { slot0 = p0; break; }
+                    if (k == TBL_EMPTY) break;
+                }
+                p0 = (p0 + 1) & mask;
+            }
+        }
+        if (sel1) {
+            uint64_t h = pg_murmur3_finalize(pg_bigint_hash(k1));
+            int64_t p1 = (int64_t)(h & (uint64_t)mask);
+            uint8_t tg = tags ? d_tbl_tag(h) : 0;
+            for (;;) {
+                if (tags) {
+                    uint8_t t = tags[p1];
+                    if (t == 0) break;
+                    if (t == tg && keys[p1] == k1) { slot1 = p1; break; }
+                } else {
+                    int64_t k = keys[p1];
+                    if (k == k1) { slot1 = p1; break; }
+                    if (k == TBL_EMPTY) break;
+                }
+                p1 = (p1 + 1) & mask;
+            }
+        }
+#pragma unroll
+        for (int r = 0; r < 2; r++) {
+            int64_t sl = r ? slot1 : slot0;
+            if (sl < 0) continue;
+            double e = r ? e1 : e0, d = r ? d1 : d0;
+            int64_t cents = (int64_t)(e * 100.0 + 0.5);
+            int64_t di = (int64_t)(d * 100.0 + 0.5);
+            int64_t ticks = cents * (100 - di);
+            double pr = e * (1.0 - d);
+            uint64_t phi, plo;
+            fx128_from_f64(pr, &phi, &plo);
+            atomicAdd(&acc[sl].dec, (unsigned long long)ticks);
+            unsigned long long old = atomicAdd(&acc[sl].flo, plo);
+            atomicAdd(&acc[sl].fhi, phi + (old > ~plo ? 1ull : 0ull));
+            atomicAdd(&acc[sl].cnt, 1ull);
+        }
+    }
+}
+
 /* probe match counting (emit mode, pass 1): count[i] = chain length */
 __global__ __launch_bounds__(256) void k_probe_count(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
@@ -2121,12 +2220,49 @@ struct JoinOp : Op {
         StagedPage sp;
         sp.stage(in);
         if (plan.mode == 1) {
+            /* specialized fast path for the Q3 shape */
+            bool spec = plan.proj.kind == PG_PROJ_DISC_PRICE &&
+                        sp.pg.cols[plan.key_col].tag == PG_T_I64 &&
+                        sp.pg.cols[plan.proj.a].tag == PG_T_F64 &&
+                        sp.pg.cols[plan.proj.b].tag == PG_T_F64 &&
+                        plan.dec_scale == 4 && plan.n_preds <= 1 &&
+                        !((uintptr_t)sp.pg.cols[plan.key_col].data & 15) &&
+                        !((uintptr_t)sp.pg.cols[plan.proj.a].data & 15) &&
+                        !((uintptr_t)sp.pg.cols[plan.proj.b].data & 15);
+            if (plan.n_preds == 1) {
+                const pg_pred& pr = plan.preds[0];
+                spec = spec &&
+                       sp.pg.cols[pr.col].tag == PG_T_I32 &&
+                       (pr.op == PG_CMP_GT || pr.op == PG_CMP_LT) &&
+                       !((uintptr_t)sp.pg.cols[pr.col].data & 7) &&
+                       !sp.pg.cols[pr.col].null_mask;
+            }
+            spec = spec && !sp.pg.cols[plan.key_col].null_mask &&
+                   !sp.pg.cols[plan.proj.a].null_mask &&
+                   !sp.pg.cols[plan.proj.b].null_mask;
             hot_begin();
-            hipLaunchKernelGGL(k_probe_agg, dim3(4096), dim3(256), 0,
-                               g_stream, sp.pg, plan,
-                               (const int64_t*)t->keys.p,
-                               (const uint8_t*)t->tags.p, t->mask,
-                               (slot_acc*)t->acc.p);
+            if (spec) {
+                const pg_pred& pr = plan.preds[0];
+                hipLaunchKernelGGL(
+                    k_probe_agg_q3, dim3(4096), dim3(256), 0, g_stream,
+                    plan.n_preds
+                        ? (const int32_t*)sp.pg.cols[pr.col].data
+                        : nullptr,
+                    plan.n_preds ? pr.op : 0,
+                    plan.n_preds ? (int32_t)pr.ival : 0,
+                    (const int64_t*)sp.pg.cols[plan.key_col].data,
+                    (const double*)sp.pg.cols[plan.proj.a].data,
+                    (const double*)sp.pg.cols[plan.proj.b].data,
+                    sp.pg.n_rows, (const int64_t*)t->keys.p,
+                    (const uint8_t*)t->tags.p, t->mask,
+                    (slot_acc*)t->acc.p);
+            } else {
+                hipLaunchKernelGGL(k_probe_agg, dim3(4096), dim3(256), 0,
+                                   g_stream, sp.pg, plan,
+                                   (const int64_t*)t->keys.p,
+                                   (const uint8_t*)t->tags.p, t->mask,
+                                   (slot_acc*)t->acc.p);
+            }
             hot_end();
             CHKV(hipStreamSynchronize(g_stream));
             return;
