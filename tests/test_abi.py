@@ -25,7 +25,7 @@ def solib():
     if not SO.exists():
         import subprocess
         subprocess.run(["hipcc", "--offload-arch=gfx950", "-O3",
-                        "-std=c++17", "-fPIC", "-shared",
+                        "-std=c++17", "-fPIC", "-ffp-contract=off", "-shared",
                         str(REPO / "presto_amd/csrc/kernels.hip"),
                         "-o", str(SO)], check=True)
     return C.CDLL(str(SO))
