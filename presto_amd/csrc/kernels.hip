@@ -1730,7 +1730,7 @@ struct AggSmallOp : Op {
     int user_aggs = 0;
     bool dec = false;
     int n_groups = 0, na = 0, maxg = 0;
-    DevBuf partials, bad;
+    DevBuf partials; /* + trailing bad-key flag, see bad_ptr() */
     void init()
     {
         if (plan.n_aggs < 1 || plan.n_aggs >= PG_MAX_AGG)
@@ -1771,10 +1771,14 @@ struct AggSmallOp : Op {
             }
         if (!na || !maxg)
             throw std::runtime_error("agg plan exceeds kernel limits");
-        partials.alloc((size_t)FT_NBLOCKS * na * maxg * 8);
+        /* one allocation: [partials | bad flag] (fewer alloc/memset/launch
+         * round trips per operator lifecycle) */
+        partials.alloc((size_t)FT_NBLOCKS * na * maxg * 8 + 64);
         partials.zero();
-        bad.alloc(8);
-        bad.zero();
+    }
+    void* bad_ptr() const
+    {
+        return (int8_t*)partials.p + (size_t)FT_NBLOCKS * na * maxg * 8;
     }
     bool q1_shape(const pg_page& pg) const
     {
@@ -1835,7 +1839,7 @@ struct AggSmallOp : Op {
                        p.key_vals[1][1],                                 \
                        D ? nullptr : (double*)partials.p,                \
                        D ? (int64_t*)partials.p : nullptr,               \
-                       (unsigned long long*)bad.p)
+                       (unsigned long long*)bad_ptr())
         if (dec) {
             if (var == 1) LAUNCH_Q1(true, 1);
             else if (var == 2) LAUNCH_Q1(true, 2);
@@ -1855,7 +1859,7 @@ struct AggSmallOp : Op {
                                dim3(FT_NBLOCKS), dim3(FT_NTHREADS), 0,
                                g_stream, pg, plan, nullptr,
                                (int64_t*)partials.p,
-                               (unsigned long long*)bad.p);
+                               (unsigned long long*)bad_ptr());
         else
             hipLaunchKernelGGL((k_agg_small<NA, MAXG, false>),
                                dim3(FT_NBLOCKS), dim3(FT_NTHREADS), 0,
@@ -1891,8 +1895,8 @@ struct AggSmallOp : Op {
             case 9 * 16 + 6: launch2<9, 6>(sp.pg); break;
             default: launch2<9, 8>(sp.pg); break;
         }
-        hot_end();
-        CHKV(hipStreamSynchronize(g_stream));
+        hot_end(); /* waits for the kernel via its trailing event */
+        if (!sp.bufs.empty()) CHKV(hipStreamSynchronize(g_stream));
     }
     void finish() override
     {
@@ -1921,7 +1925,7 @@ struct AggSmallOp : Op {
                             hipMemcpyDeviceToHost, g_stream));
         CHKV(hipMemcpyAsync(hlo.data(), out_lo.p, nm * 8,
                             hipMemcpyDeviceToHost, g_stream));
-        CHKV(hipMemcpyAsync(&hbad, bad.p, 8, hipMemcpyDeviceToHost,
+        CHKV(hipMemcpyAsync(&hbad, bad_ptr(), 8, hipMemcpyDeviceToHost,
                             g_stream));
         CHKV(hipStreamSynchronize(g_stream));
         if (hbad)
