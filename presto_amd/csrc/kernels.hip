@@ -1864,7 +1864,7 @@ struct AggSmallOp : Op {
             hipLaunchKernelGGL((k_agg_small<NA, MAXG, false>),
                                dim3(FT_NBLOCKS), dim3(FT_NTHREADS), 0,
                                g_stream, pg, plan, (double*)partials.p,
-                               nullptr, (unsigned long long*)bad.p);
+                               nullptr, (unsigned long long*)bad_ptr());
     }
     void add_input(const pg_page* in) override
     {
