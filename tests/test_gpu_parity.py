@@ -369,3 +369,145 @@ def test_agg_table_overflow_errors(P):
     with _pytest.raises(RuntimeError):
         b.finish()
     b.destroy()
+
+
+def test_fuzz_operators_vs_oracle(P, oracle_lib):
+    """Randomized plans across the operator surface, each checked against
+    the oracle / a numpy model: filter (random typed predicates),
+    join build+probe (random key ranges/dup rates), partition (random
+    fan-out), small-key agg (random cardinalities, decimal + f64)."""
+    rng = np.random.default_rng(2024)
+    for case in range(10):
+        n = int(rng.integers(1, 50_000))
+        a = rng.integers(-1000, 1000, n).astype(np.int64)
+        b = rng.random(n)
+        d = rng.integers(0, 50, n).astype(np.int32)
+        u = rng.integers(0, 4, n).astype(np.uint8)
+        page = P.Page({"a": a, "b": b, "d": d, "u": u})
+
+        # ---- filter with 1-3 random predicates ----
+        npred = int(rng.integers(1, 4))
+        plan = P.PlanFilterProject()
+        plan.n_preds = npred
+        sel = np.ones(n, bool)
+        for j in range(npred):
+            col = int(rng.integers(0, 4))
+            op_ = int(rng.integers(0, 6))
+            if col == 1:
+                cv = float(rng.random())
+                plan.preds[j] = P.Pred(col, op_, 0, cv)
+                v = b
+            else:
+                cv = int(rng.integers(-500, 500))
+                plan.preds[j] = P.Pred(col, op_, cv, 0.0)
+                v = [a, None, d, u][col]
+            import operator as _op
+            f = [_op.lt, _op.le, _op.gt, _op.ge, _op.eq, _op.ne][op_]
+            sel &= f(v, cv)
+        plan.n_proj = 2
+        plan.proj[0] = P.Proj(P.PROJ_IDENT, 0, 0, 0)
+        plan.proj[1] = P.Proj(P.PROJ_IDENT, 1, 0, 0)
+        fo = P.Operator(P.OP_FILTER_PROJECT, plan)
+        fo.add_input(page)
+        out = fo.get_output(["a", "b"])
+        fo.destroy()
+        assert np.array_equal(out["a"], a[sel]), f"filter case {case}"
+        assert np.array_equal(out["b"], b[sel])
+
+        # ---- join with random dup rate ----
+        nb = int(rng.integers(1, 3000))
+        npb = int(rng.integers(1, 8000))
+        krange = int(rng.integers(2, 4000))
+        bkeys = rng.integers(0, krange, nb).astype(np.int64)
+        pkeys = rng.integers(0, krange + 50, npb).astype(np.int64)
+        bplan = P.PlanHashBuild()
+        bplan.key_col = 0
+        bplan.semijoin_table = -1
+        bplan.n_payload = 1
+        bplan.payload_col[0] = 1
+        bplan.capacity_hint = nb
+        rows = np.arange(nb, dtype=np.int64)
+        bo = P.Operator(P.OP_HASH_BUILD, bplan)
+        bo.add_input(P.Page({"k": bkeys, "r": rows}))
+        bo.finish()
+        jplan = P.PlanLookupJoin()
+        jplan.table = bo.table()
+        jplan.key_col = 0
+        jplan.mode = 0
+        jplan.n_emit = 1
+        jplan.emit_probe_cols[0] = 1
+        jo = P.Operator(P.OP_LOOKUP_JOIN, jplan)
+        jo.add_input(P.Page({"k": pkeys,
+                             "pi": np.arange(npb, dtype=np.int64)}))
+        jout = jo.get_output(["pi", "br"])
+        jo.destroy()
+        op_idx, ob_idx = oracle_lib.join(bkeys, pkeys)
+        assert set(zip(jout["pi"].tolist(), jout["br"].tolist())) == \
+            set(zip(op_idx.tolist(), ob_idx.tolist())), f"join case {case}"
+        from presto_amd.engine import lib as _lib
+        _lib().c.pg_table_destroy(jplan.table)
+        bo.destroy()
+
+        # ---- partition with random fan-out ----
+        nparts = int(rng.integers(1, 17))
+        pplan = P.PlanPartition()
+        pplan.n_partitions = nparts
+        pplan.key_col = 0
+        pplan.n_emit = 1
+        pplan.emit_cols[0] = 0
+        po = P.Operator(P.OP_PARTITION, pplan)
+        po.add_input(P.Page({"k": a}))
+        counts = po.partition_counts(nparts)
+        pages = [po.get_output(["k"]) for _ in range(nparts)]
+        po.destroy()
+        pid = np.array([oracle_lib.lib.oracle_partition(
+            oracle_lib.lib.oracle_bigint_hash(int(k)), nparts)
+            for k in a.tolist()])
+        for pp in range(nparts):
+            assert np.array_equal(pages[pp]["k"], a[pid == pp]), \
+                f"partition case {case} p{pp}"
+        assert sum(counts) == n
+
+
+def test_fuzz_small_agg_vs_numpy(P):
+    """Random small-key aggregations (1-2 u8 keys, decimal + f64 modes)
+    checked against numpy groupby."""
+    rng = np.random.default_rng(77)
+    for case in range(6):
+        n = int(rng.integers(100, 80_000))
+        nk0 = int(rng.integers(2, 5))
+        k0_vals = rng.choice(np.arange(1, 250, dtype=np.uint8), nk0,
+                             replace=False)
+        k0 = rng.choice(k0_vals, n)
+        x = (rng.integers(0, 10**6, n) / 100.0)  # cents-representable
+        sd = rng.integers(0, 100, n).astype(np.int32)
+        page = P.Page({"k0": k0, "x": x, "sd": sd})
+        plan = P.PlanHashAggSmall()
+        plan.n_preds = 1
+        thresh = int(rng.integers(20, 80))
+        plan.preds[0] = P.Pred(2, P.CMP_LT, thresh, 0.0)
+        plan.n_keys = 1
+        plan.key_col[0] = 0
+        plan.n_vals[0] = nk0
+        for j, v in enumerate(sorted(k0_vals.tolist())):
+            plan.key_vals[0][j] = v
+        plan.n_aggs = 2
+        plan.aggs[0] = P.Agg(P.AGG_SUM_DEC, P.Proj(P.PROJ_IDENT, 1, 0, 0), 2)
+        plan.aggs[1] = P.Agg(P.AGG_COUNT, P.Proj(P.PROJ_IDENT, 0, 0, 0), 0)
+        op = P.Operator(P.OP_HASH_AGG_SMALL, plan)
+        op.add_input(page)
+        op.finish()
+        out = op.get_output(["k", "s_hi", "s_lo", "cnt"])
+        op.destroy()
+        sel = sd < thresh
+        row = 0
+        for v in sorted(k0_vals.tolist()):
+            m = sel & (k0 == v)
+            if not m.any():
+                continue
+            exp_cents = np.round(x[m] * 100).astype(np.int64).sum()
+            assert out["k"][row] == v
+            assert out["s_lo"][row] == exp_cents, f"agg case {case} key {v}"
+            assert out["cnt"][row] == int(m.sum())
+            row += 1
+        assert row == len(out["k"])
