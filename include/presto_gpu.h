@@ -253,6 +253,27 @@ pg_status pg_op_partition_counts(pg_op op, int64_t* counts, int32_t n);
 /* destroy a table explicitly (tables outlive their build op until freed) */
 pg_status pg_table_destroy(int64_t table);
 
+/* ---- SerializedPage wire interop ----
+ * Presto's exchange wire format, restated from
+ * spi/page/PagesSerdeUtil.java:64-88 (metadata: positionCount i32, codec
+ * marker u8, uncompressedSize i32, size i32, checksum i64, then the page
+ * bytes; little-endian), checksum per computeSerializedPageChecksum:109-120
+ * (CRC32 over data + marker + positionCount + uncompressedSize bytes);
+ * raw page bytes per writeRawPage:45-51 (block count i32 then per block a
+ * length-prefixed encoding name, BlockEncodingManager.java:96-99) with
+ * encodings LONG_ARRAY / INT_ARRAY / BYTE_ARRAY
+ * (common/block/LongArrayBlockEncoding.java:26-48 etc.; nulls-as-bits per
+ * EncoderUtil.java:31-63; non-null values only).
+ * v1 scope: uncompressed, unencrypted (codec marker 0); fixed-width
+ * blocks (I64/F64 -> LONG_ARRAY, I32 -> INT_ARRAY, U8 -> BYTE_ARRAY).
+ * Host-side (the node-boundary seam stays on the host). */
+pg_status pg_page_serialize(const pg_page* page /* host cols */,
+                            void* out, int64_t cap, int64_t* out_len);
+/* parses and verifies; fills *out with malloc-backed host columns
+ * (free with pg_page_free). F64 consumers reinterpret LONG_ARRAY bits. */
+pg_status pg_page_deserialize(const void* buf, int64_t len, pg_page* out);
+pg_status pg_page_free(pg_page* page);
+
 #ifdef __cplusplus
 }
 #endif

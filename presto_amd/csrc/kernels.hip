@@ -2745,3 +2745,242 @@ extern "C" pg_status pg_table_destroy(int64_t t)
     g_tables.erase(t);
     return PG_OK;
 }
+
+/* ================================================================== */
+/* SerializedPage wire interop — citations in presto_gpu.h            */
+/* ================================================================== */
+namespace {
+
+/* standard CRC-32 (zlib polynomial), as java.util.zip.CRC32 */
+static uint32_t crc32_tab_[256];
+static bool crc32_init_ = false;
+static uint32_t pg_crc32(uint32_t crc, const uint8_t* p, size_t n)
+{
+    if (!crc32_init_) {
+        for (uint32_t i = 0; i < 256; i++) {
+            uint32_t c = i;
+            for (int k = 0; k < 8; k++)
+                c = (c & 1) ? 0xEDB88320u ^ (c >> 1) : c >> 1;
+            crc32_tab_[i] = c;
+        }
+        crc32_init_ = true;
+    }
+    crc = ~crc;
+    for (size_t i = 0; i < n; i++)
+        crc = crc32_tab_[(crc ^ p[i]) & 0xff] ^ (crc >> 8);
+    return ~crc;
+}
+
+struct ByteWriter {
+    uint8_t* p;
+    int64_t cap, off = 0;
+    bool ok = true;
+    void bytes(const void* src, int64_t n)
+    {
+        if (off + n > cap) { ok = false; return; }
+        memcpy(p + off, src, n);
+        off += n;
+    }
+    void u8(uint8_t v) { bytes(&v, 1); }
+    void i32(int32_t v) { bytes(&v, 4); } /* little-endian host */
+    void i64(int64_t v) { bytes(&v, 8); }
+};
+struct ByteReader {
+    const uint8_t* p;
+    int64_t len, off = 0;
+    bool ok = true;
+    const void* bytes(int64_t n)
+    {
+        if (off + n > len) { ok = false; return nullptr; }
+        const void* r = p + off;
+        off += n;
+        return r;
+    }
+    uint8_t u8() { auto* r = (const uint8_t*)bytes(1); return r ? *r : 0; }
+    int32_t i32()
+    {
+        int32_t v = 0;
+        auto* r = bytes(4);
+        if (r) memcpy(&v, r, 4);
+        return v;
+    }
+    int64_t i64()
+    {
+        int64_t v = 0;
+        auto* r = bytes(8);
+        if (r) memcpy(&v, r, 8);
+        return v;
+    }
+};
+
+static const char* enc_name(int tag)
+{
+    switch (tag) {
+        case PG_T_U8: return "BYTE_ARRAY";
+        case PG_T_I32: return "INT_ARRAY";
+        default: return "LONG_ARRAY"; /* I64 and F64 (bits) */
+    }
+}
+
+/* encodeNullsAsBits — EncoderUtil.java:31-63 (MSB-first per byte) */
+static void write_null_bits(ByteWriter& w, const uint8_t* mask, int64_t n)
+{
+    if (!mask) {
+        w.u8(0);
+        return;
+    }
+    bool any = false;
+    for (int64_t i = 0; i < n; i++) any = any || mask[i];
+    w.u8(any ? 1 : 0);
+    if (!any) return;
+    for (int64_t base = 0; base < n; base += 8) {
+        uint8_t v = 0;
+        for (int64_t j = base; j < base + 8 && j < n; j++)
+            if (mask[j]) v |= (uint8_t)(0x80u >> (j - base));
+        w.u8(v);
+    }
+}
+
+} /* namespace */
+
+extern "C" pg_status pg_page_serialize(const pg_page* page, void* out,
+                                       int64_t cap, int64_t* out_len)
+{
+    for (int c = 0; c < page->n_cols; c++)
+        if (page->cols[c].on_device)
+            return seterr("pg_page_serialize: host columns required");
+    ByteWriter w{(uint8_t*)out, cap};
+    /* metadata placeholder, filled after the body is written */
+    int64_t meta_at = w.off;
+    w.off += 4 + 1 + 4 + 4 + 8;
+    int64_t body_at = w.off;
+    w.i32(page->n_cols); /* writeRawPage: block count */
+    for (int c = 0; c < page->n_cols; c++) {
+        const pg_col& col = page->cols[c];
+        const char* name = enc_name(col.tag);
+        int32_t nl = (int32_t)strlen(name);
+        w.i32(nl);
+        w.bytes(name, nl);
+        w.i32((int32_t)page->n_rows);
+        write_null_bits(w, col.null_mask, page->n_rows);
+        size_t esz = type_size(col.tag);
+        if (!col.null_mask) {
+            w.bytes(col.data, page->n_rows * esz);
+        } else {
+            for (int64_t i = 0; i < page->n_rows; i++)
+                if (!col.null_mask[i])
+                    w.bytes((const uint8_t*)col.data + i * esz, esz);
+        }
+    }
+    if (!w.ok) return seterr("pg_page_serialize: buffer too small");
+    int32_t body = (int32_t)(w.off - body_at);
+    /* metadata: PagesSerdeUtil.writeSerializedPageMetadata:70-77 */
+    ByteWriter m{(uint8_t*)out + meta_at, 4 + 1 + 4 + 4 + 8};
+    m.i32((int32_t)page->n_rows);
+    m.u8(0); /* codec markers: uncompressed, unencrypted, no checksum bit —
+                the checksum field is still always present on the wire */
+    m.i32(body); /* uncompressed size */
+    m.i32(body); /* size */
+    /* computeSerializedPageChecksum:109-120: crc32(data) then marker byte
+     * then positionCount and uncompressedSize as 4 LE bytes each */
+    uint32_t crc = pg_crc32(0, (uint8_t*)out + body_at, body);
+    uint8_t tail[9];
+    tail[0] = 0;
+    int32_t pc = (int32_t)page->n_rows;
+    memcpy(tail + 1, &pc, 4);
+    memcpy(tail + 5, &body, 4);
+    crc = pg_crc32(crc, tail, 9);
+    m.i64((int64_t)(uint32_t)crc);
+    *out_len = w.off;
+    return PG_OK;
+}
+
+extern "C" pg_status pg_page_deserialize(const void* buf, int64_t len,
+                                         pg_page* out)
+{
+    ByteReader r{(const uint8_t*)buf, len};
+    int32_t pos_count = r.i32();
+    uint8_t marker = r.u8();
+    int32_t usize = r.i32();
+    int32_t size = r.i32();
+    int64_t checksum = r.i64();
+    if (!r.ok) return seterr("deserialize: truncated metadata");
+    if (marker != 0)
+        return seterr("deserialize: compressed/encrypted pages unsupported "
+                      "(codec marker != 0)");
+    if (size != usize || r.off + size > len)
+        return seterr("deserialize: bad sizes");
+    /* verify checksum */
+    uint32_t crc = pg_crc32(0, r.p + r.off, size);
+    uint8_t tail[9];
+    tail[0] = marker;
+    memcpy(tail + 1, &pos_count, 4);
+    memcpy(tail + 5, &usize, 4);
+    crc = pg_crc32(crc, tail, 9);
+    if ((int64_t)(uint32_t)crc != checksum)
+        return seterr("deserialize: checksum mismatch");
+    int32_t n_blocks = r.i32();
+    if (n_blocks < 0 || n_blocks > 16)
+        return seterr("deserialize: unsupported block count");
+    memset(out, 0, sizeof(*out));
+    out->n_rows = pos_count;
+    out->n_cols = n_blocks;
+    for (int c = 0; c < n_blocks; c++) {
+        int32_t nl = r.i32();
+        if (!r.ok || nl < 0 || nl > 64)
+            return seterr("deserialize: bad encoding name");
+        char name[65] = {0};
+        memcpy(name, r.bytes(nl), r.ok ? nl : 0);
+        int tag;
+        if (!strcmp(name, "LONG_ARRAY")) tag = PG_T_I64;
+        else if (!strcmp(name, "INT_ARRAY")) tag = PG_T_I32;
+        else if (!strcmp(name, "BYTE_ARRAY")) tag = PG_T_U8;
+        else return seterr("deserialize: unsupported block encoding");
+        int32_t n = r.i32();
+        if (n != pos_count)
+            return seterr("deserialize: block position count mismatch");
+        uint8_t may_null = r.u8();
+        uint8_t* mask = nullptr;
+        if (may_null) {
+            mask = (uint8_t*)calloc(n, 1);
+            for (int64_t base = 0; base < n; base += 8) {
+                uint8_t v = r.u8();
+                for (int64_t j = base; j < base + 8 && j < n; j++)
+                    mask[j] = (v >> (7 - (j - base))) & 1;
+            }
+        }
+        size_t esz = type_size(tag);
+        uint8_t* vals = (uint8_t*)calloc(n ? n : 1, esz);
+        if (!mask) {
+            const void* src = r.bytes((int64_t)n * esz);
+            if (src) memcpy(vals, src, (size_t)n * esz);
+        } else {
+            for (int64_t i = 0; i < n; i++)
+                if (!mask[i]) {
+                    const void* src = r.bytes(esz);
+                    if (src) memcpy(vals + i * esz, src, esz);
+                }
+        }
+        if (!r.ok) {
+            free(mask);
+            free(vals);
+            return seterr("deserialize: truncated block");
+        }
+        out->cols[c].tag = tag;
+        out->cols[c].on_device = 0;
+        out->cols[c].data = vals;
+        out->cols[c].null_mask = mask;
+    }
+    return PG_OK;
+}
+
+extern "C" pg_status pg_page_free(pg_page* page)
+{
+    for (int c = 0; c < page->n_cols; c++) {
+        free(page->cols[c].data);
+        free((void*)page->cols[c].null_mask);
+        page->cols[c].data = nullptr;
+        page->cols[c].null_mask = nullptr;
+    }
+    return PG_OK;
+}
