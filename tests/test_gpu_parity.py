@@ -511,3 +511,70 @@ def test_fuzz_small_agg_vs_numpy(P):
             assert out["cnt"][row] == int(m.sum())
             row += 1
         assert row == len(out["k"])
+
+
+def test_driver_loop_paged_q1(P, oracle_lib, sf01):
+    """Q1 driven page-at-a-time (8192-row pages, the reference's page size,
+    PageProcessor.java MAX_BATCH_SIZE:58) through the Driver-loop analog
+    (presto_amd/driver.py) — decimal results must equal the oracle exactly
+    (order-independent), exercising the streaming addInput path."""
+    from presto_amd.driver import run_chain
+    li = sf01["li"]
+    n = len(li["quantity"])
+    full = _li_page(P, li)
+    plan = P.pipelines.q1_plan(full, "dec")
+    op = P.Operator(P.OP_HASH_AGG_SMALL, plan)
+    pages = []
+    for a in range(0, n, 8192):
+        b = min(a + 8192, n)
+        pages.append(P.Page({k: v[a:b] for k, v in full.cols.items()}))
+    outs = []
+    run_chain(pages, [op], lambda raw: outs.append(
+        P.engine._read_output_page(raw, P.pipelines.Q1_DEC_NAMES)))
+    op.destroy()
+    assert len(outs) == 1
+    got = outs[0]
+    exp = oracle_lib.q1(li)
+    for i, g in enumerate(exp):
+        assert got["count"][i] == g.count_order
+        assert got["sum_disc_price_lo"][i] == g.sum_disc_1e4
+        assert got["sum_charge_hi"][i] == g.sum_charge_1e6_hi
+
+
+def test_driver_loop_filter_to_topn(P):
+    """A two-operator chain (filter -> topn) through the driver loop with
+    multiple input pages."""
+    from presto_amd.driver import run_chain
+    rng = np.random.default_rng(11)
+    n = 120_000
+    val = rng.integers(0, 10**9, n).astype(np.int64)
+    date = rng.integers(8000, 11000, n).astype(np.int32)
+    key = rng.permutation(n).astype(np.int64)
+    fp = P.PlanFilterProject()
+    fp.n_preds = 1
+    fp.preds[0] = P.Pred(0, P.CMP_GE, 5 * 10**8, 0.0)
+    fp.n_proj = 3
+    for i in range(3):
+        fp.proj[i] = P.Proj(P.PROJ_IDENT, i, 0, 0)
+    tp = P.PlanTopN()
+    tp.limit = 5
+    tp.val_col = 0
+    tp.date_col = 1
+    tp.key_col = 2
+    f = P.Operator(P.OP_FILTER_PROJECT, fp)
+    t = P.Operator(P.OP_TOPN, tp)
+    pages = []
+    for a in range(0, n, 30_000):
+        b = min(a + 30_000, n)
+        pages.append(P.Page({"v": val[a:b], "d": date[a:b], "k": key[a:b]}))
+    outs = []
+    run_chain(pages, [f, t], lambda raw: outs.append(
+        P.engine._read_output_page(raw, ["k", "v", "d"])))
+    f.destroy()
+    t.destroy()
+    assert len(outs) == 1
+    sel = val >= 5 * 10**8
+    vv, dd, kk = val[sel], date[sel], key[sel]
+    order = sorted(range(len(vv)),
+                   key=lambda i: (-vv[i], dd[i], kk[i]))[:5]
+    assert outs[0]["k"].tolist() == [kk[i] for i in order]
