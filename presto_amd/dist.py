@@ -41,8 +41,8 @@ def exchange_columns(cols, send_counts, recv_counts=None, device=None):
 
 def _col_to_torch(col, n, device):
     """Copy a library-owned device column (pg_col) into a torch tensor."""
-    import numpy as np
-    from .engine import lib, _TAG_NP
+    import numpy as np  # noqa: F401
+    from .engine import lib
     dtmap = {0: torch.uint8, 1: torch.int32, 2: torch.int64,
              3: torch.float64}
     t = torch.empty(n, dtype=dtmap[col.tag], device=device)

@@ -4,7 +4,6 @@ Mirrors include/presto_gpu.h exactly; see that header for the contract and
 the reference citations.
 """
 import ctypes as C
-import os
 import pathlib
 
 import numpy as np

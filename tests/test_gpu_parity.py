@@ -578,3 +578,68 @@ def test_driver_loop_filter_to_topn(P):
     order = sorted(range(len(vv)),
                    key=lambda i: (-vv[i], dd[i], kk[i]))[:5]
     assert outs[0]["k"].tolist() == [kk[i] for i in order]
+
+
+def test_generic_f64_agg_butterfly_model(P):
+    """The GENERIC small-key kernel's f64 path (non-Q1 shape: one u8 key)
+    must match a numpy replay of the deterministic butterfly schedule
+    bit-for-bit."""
+    rng = np.random.default_rng(31)
+    n = 150_000
+    k = rng.choice(np.array([5, 9], np.uint8), n)
+    x = rng.random(n) * 1000
+    sd = rng.integers(0, 100, n).astype(np.int32)
+    page = P.Page({"k": k, "x": x, "sd": sd})
+    plan = P.PlanHashAggSmall()
+    plan.n_preds = 1
+    plan.preds[0] = P.Pred(2, P.CMP_LT, 60, 0.0)
+    plan.n_keys = 1
+    plan.key_col[0] = 0
+    plan.n_vals[0] = 2
+    plan.key_vals[0][0] = 5
+    plan.key_vals[0][1] = 9
+    plan.n_aggs = 2
+    plan.aggs[0] = P.Agg(P.AGG_SUM_F64, P.Proj(P.PROJ_IDENT, 1, 0, 0), 0)
+    plan.aggs[1] = P.Agg(P.AGG_COUNT, P.Proj(P.PROJ_IDENT, 0, 0, 0), 0)
+    op = P.Operator(P.OP_HASH_AGG_SMALL, plan)
+    op.add_input(page)
+    op.finish()
+    out = op.get_output(["k", "s", "cnt"])
+    op.destroy()
+
+    VL = 4096 * 256
+    lanes_per_wave = 64
+
+    def model_sum(sel_vals):
+        # lane partials: rows 2v, 2v+1 (n < 2*VL so one pair per lane)
+        w = np.zeros(2 * VL)
+        w[:n] = sel_vals
+        lane = w[0::2] + w[1::2]  # VL lanes
+        # wave butterfly s=32..1 over each 64-lane group
+        a = lane.reshape(-1, lanes_per_wave)
+        idx = np.arange(lanes_per_wave)
+        for s in (32, 16, 8, 4, 2, 1):
+            a = a + a[:, idx ^ s]
+        wsum = a[:, 0].reshape(-1, 4)  # (4096 blocks, 4 waves)
+        for s in (2, 1):
+            wsum = wsum + wsum[:, np.arange(4) ^ s]
+        bp = wsum[:, 0]  # 4096 block partials
+        g = np.zeros(64)
+        for l in range(64):
+            acc = 0.0
+            for m in range(64):  # ascending, strict sequential
+                acc += bp[l + 64 * m]
+            g[l] = acc
+        g2 = g.copy()
+        for s in (32, 16, 8, 4, 2, 1):
+            g2 = g2 + g2[np.arange(64) ^ s]
+        return g2[0]
+
+    sel = sd < 60
+    for row, kv in enumerate([5, 9]):
+        vals = np.where(sel & (k == kv), x, 0.0)
+        exp = model_sum(vals)
+        assert out["k"][row] == kv
+        assert out["s"][row].view(np.int64) == np.float64(exp).view(
+            np.int64), (out["s"][row], exp)
+        assert out["cnt"][row] == int((sel & (k == kv)).sum())
