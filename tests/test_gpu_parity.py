@@ -643,3 +643,59 @@ def test_generic_f64_agg_butterfly_model(P):
         assert out["s"][row].view(np.int64) == np.float64(exp).view(
             np.int64), (out["s"][row], exp)
         assert out["cnt"][row] == int((sel & (k == kv)).sum())
+
+
+def test_bigint_groupby_sum_via_composition(P, oracle_lib):
+    """Arbitrary-cardinality bigint group-by (BigintGroupByHash.java:222-252
+    semantics at result level): composed as HASH_BUILD(agg_table, key=group
+    col) + LOOKUP_JOIN(mode 1) over the same pages — the recipe
+    LocalExecutionPlanner would emit for HashAggregation on a bigint key.
+    Grouped decimal sums must match a numpy groupby exactly; f64 sums are
+    the correctly-rounded exact sums (fx128)."""
+    rng = np.random.default_rng(42)
+    n = 300_000
+    ngroups = 5000
+    keys = rng.integers(1, ngroups + 1, n)
+    price = rng.integers(100, 10**7, n) / 100.0   # cents-representable
+    disc = rng.integers(0, 11, n) / 100.0
+    page = P.Page({"k": keys, "p": price, "d": disc})
+    bp = P.PlanHashBuild()
+    bp.key_col = 0
+    bp.semijoin_table = -1
+    bp.n_payload = 0
+    bp.capacity_hint = ngroups + 64
+    bp.agg_table = 1
+    b = P.Operator(P.OP_HASH_BUILD, bp)
+    b.add_input(page)
+    b.finish()
+    jp = P.PlanLookupJoin()
+    jp.table = b.table()
+    jp.key_col = 0
+    jp.mode = 1
+    jp.proj = P.Proj(P.PROJ_DISC_PRICE, 1, 2, 0)
+    jp.dec_scale = 4
+    j = P.Operator(P.OP_LOOKUP_JOIN, jp)
+    j.add_input(page)
+    j.finish()
+    out = j.get_output(["k", "sum_dec", "sum_f64", "cnt"])
+    j.destroy()
+    from presto_amd.engine import lib as _l
+    _l().c.pg_table_destroy(jp.table)
+    b.destroy()
+    # numpy reference (exact ticks)
+    cents = np.round(price * 100).astype(np.int64)
+    dd = np.round(disc * 100).astype(np.int64)
+    ticks = cents * (100 - dd)
+    order = np.argsort(out["k"])
+    got_k = out["k"][order]
+    got_s = out["sum_dec"][order]
+    got_c = out["cnt"][order]
+    uniq = np.unique(keys)
+    assert np.array_equal(got_k, uniq)
+    exp_s = np.zeros(len(uniq), np.int64)
+    exp_c = np.zeros(len(uniq), np.int64)
+    pos = np.searchsorted(uniq, keys)
+    np.add.at(exp_s, pos, ticks)
+    np.add.at(exp_c, pos, 1)
+    assert np.array_equal(got_s, exp_s)
+    assert np.array_equal(got_c, exp_c)
