@@ -67,13 +67,20 @@ typedef enum {
     PG_T_I32 = 1, /* IntArrayBlock / DateType days */
     PG_T_I64 = 2, /* LongArrayBlock / BigintType */
     PG_T_F64 = 3, /* LongArrayBlock bits / DoubleType */
+    PG_T_VARBIN = 4, /* VariableWidthBlock (VariableWidthBlock.java:48-61):
+                        data = bytes, offsets = int32[n_rows+1] (element i
+                        spans bytes [offsets[i], offsets[i+1])).  v1 scope:
+                        EQ/NE-const predicates and hashing/partitioning
+                        (XxHash64 per AbstractVariableWidthBlock.java:102-
+                        105); VARBIN projection/emit is rejected. */
 } pg_type;
 
 typedef struct {
     int32_t tag;            /* pg_type */
     int32_t on_device;      /* 1: data is a device pointer */
-    void* data;             /* values array, n_rows elements */
+    void* data;             /* values array (VARBIN: the byte buffer) */
     const uint8_t* null_mask; /* optional, 1 byte/pos, 1 = null; may be NULL */
+    const int32_t* offsets; /* VARBIN only: n_rows+1 offsets */
 } pg_col;
 
 typedef struct {
@@ -96,6 +103,8 @@ typedef struct {
     int32_t op;    /* pg_cmp */
     int64_t ival;  /* compare value for integer columns */
     double dval;   /* compare value for f64 columns */
+    char sval[16]; /* VARBIN: compare bytes (EQ/NE only) */
+    int32_t slen;
 } pg_pred;
 
 /* projection expressions (PageProjection analogs) */

@@ -172,6 +172,10 @@ int32_t oracle_q1(int64_t n, const double* qty, const double* eprice,
 uint64_t oracle_murmur3_finalize(uint64_t h) { return pg_murmur3_finalize(h); }
 uint64_t oracle_bigint_hash(int64_t v) { return pg_bigint_hash(v); }
 int32_t oracle_partition(uint64_t h, int32_t n) { return pg_partition(h, n); }
+uint64_t oracle_xxh64(const uint8_t* d, int64_t n)
+{
+    return pg_xxh64(d, (uint64_t)n);
+}
 
 /* next power of two >= ceil(x / 0.75), min 2 — fastutil arraySize semantics
  * used by BigintGroupByHash.java:49 / PagesHash.java:67 */

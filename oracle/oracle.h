@@ -103,6 +103,8 @@ uint64_t oracle_bigint_hash(int64_t v);
 /* partition id — HashGenerator.java:22-29:
  * (u32(Long.hashCode(rawHash)) * partitionCount) >> 32 */
 int32_t oracle_partition(uint64_t raw_hash, int32_t partition_count);
+/* varchar hash — AbstractVariableWidthBlock.java:102-105 (XxHash64 seed 0) */
+uint64_t oracle_xxh64(const uint8_t* data, int64_t len);
 
 /* Group-by over a single bigint key column, BigintGroupByHash.java:222-332:
  * open-address linear probe, fill 0.75, capacity = next pow2(ceil(hint/.75)),

@@ -117,6 +117,77 @@ static inline FX_HD uint64_t pg_bigint_hash(int64_t v)
     return x * 0x9E3779B185EBCA87ull;
 }
 
+/* XxHash64, seed 0 — the reference's varchar hash
+ * (AbstractVariableWidthBlock.java:102-105 -> io.airlift.slice.XxHash64,
+ * the standard XXH64 algorithm; pinned by known-answer vectors and an
+ * independent Python restatement in tests/test_oracle.py). */
+static inline FX_HD uint64_t pg_xxh64_rotl(uint64_t x, int r)
+{
+    return (x << r) | (x >> (64 - r));
+}
+static inline FX_HD uint64_t pg_xxh64(const uint8_t* p, uint64_t len)
+{
+    const uint64_t P1 = 0x9E3779B185EBCA87ull, P2 = 0xC2B2AE3D27D4EB4Full,
+                   P3 = 0x165667B19E3779F9ull, P4 = 0x85EBCA77C2B2AE63ull,
+                   P5 = 0x27D4EB2F165667C5ull;
+    const uint8_t* end = p + len;
+    uint64_t h;
+#define PG_XXH_RD64(q)                                                  \
+    (((uint64_t)(q)[0]) | ((uint64_t)(q)[1] << 8) |                     \
+     ((uint64_t)(q)[2] << 16) | ((uint64_t)(q)[3] << 24) |              \
+     ((uint64_t)(q)[4] << 32) | ((uint64_t)(q)[5] << 40) |              \
+     ((uint64_t)(q)[6] << 48) | ((uint64_t)(q)[7] << 56))
+#define PG_XXH_RD32(q)                                                  \
+    (((uint64_t)(q)[0]) | ((uint64_t)(q)[1] << 8) |                     \
+     ((uint64_t)(q)[2] << 16) | ((uint64_t)(q)[3] << 24))
+    if (len >= 32) {
+        uint64_t v1 = P1 + P2, v2 = P2, v3 = 0, v4 = (uint64_t)0 - P1;
+        do {
+            v1 = pg_xxh64_rotl(v1 + PG_XXH_RD64(p) * P2, 31) * P1;
+            p += 8;
+            v2 = pg_xxh64_rotl(v2 + PG_XXH_RD64(p) * P2, 31) * P1;
+            p += 8;
+            v3 = pg_xxh64_rotl(v3 + PG_XXH_RD64(p) * P2, 31) * P1;
+            p += 8;
+            v4 = pg_xxh64_rotl(v4 + PG_XXH_RD64(p) * P2, 31) * P1;
+            p += 8;
+        } while (p + 32 <= end);
+        h = pg_xxh64_rotl(v1, 1) + pg_xxh64_rotl(v2, 7) +
+            pg_xxh64_rotl(v3, 12) + pg_xxh64_rotl(v4, 18);
+        uint64_t vs[4] = {v1, v2, v3, v4};
+        for (int i = 0; i < 4; i++) {
+            h ^= pg_xxh64_rotl(vs[i] * P2, 31) * P1;
+            h = h * P1 + P4;
+        }
+    } else {
+        h = P5;
+    }
+    h += len;
+    while (p + 8 <= end) {
+        h ^= pg_xxh64_rotl(PG_XXH_RD64(p) * P2, 31) * P1;
+        h = pg_xxh64_rotl(h, 27) * P1 + P4;
+        p += 8;
+    }
+    if (p + 4 <= end) {
+        h ^= PG_XXH_RD32(p) * P1;
+        h = pg_xxh64_rotl(h, 23) * P2 + P3;
+        p += 4;
+    }
+    while (p < end) {
+        h ^= (uint64_t)(*p) * P5;
+        h = pg_xxh64_rotl(h, 11) * P1;
+        p++;
+    }
+    h ^= h >> 33;
+    h *= P2;
+    h ^= h >> 29;
+    h *= P3;
+    h ^= h >> 32;
+    return h;
+#undef PG_XXH_RD64
+#undef PG_XXH_RD32
+}
+
 /* partition id — HashGenerator.java:22-29:
  * (toUnsignedLong(Long.hashCode(rawHash)) * partitionCount) >> 32 */
 static inline FX_HD int32_t pg_partition(uint64_t raw_hash, int32_t n_part)

@@ -198,7 +198,16 @@ __device__ inline bool d_eval_preds(const pg_page& pg, const pg_pred* preds,
         const pg_col& c = pg.cols[pr.col];
         if (c.null_mask && c.null_mask[i]) return false;
         bool ok;
-        if (c.tag == PG_T_F64) {
+        if (c.tag == PG_T_VARBIN) {
+            /* EQ/NE against a constant (VariableWidthBlock bytesEqual,
+             * AbstractVariableWidthBlock.java:95-99) */
+            int32_t b0 = c.offsets[i], b1 = c.offsets[i + 1];
+            bool eq = (b1 - b0) == pr.slen;
+            const uint8_t* d = (const uint8_t*)c.data + b0;
+            for (int j = 0; eq && j < pr.slen; j++)
+                eq = d[j] == (uint8_t)pr.sval[j];
+            ok = pr.op == PG_CMP_EQ ? eq : !eq;
+        } else if (c.tag == PG_T_F64) {
             double v = ((const double*)c.data)[i];
             double x = pr.dval;
             switch (pr.op) {
@@ -1385,9 +1394,17 @@ __global__ __launch_bounds__(256) void k_topn(const void* vals, int is_f64,
 __device__ inline int d_part_id(const pg_page& pg, int key_col, int np,
                                 int64_t i)
 {
-    int64_t key = d_load_i64(pg.cols[key_col], i);
     /* rawHash: CombineHashFunction.java:28-30 fold from
-     * INITIAL_HASH_VALUE=0: h = 31*0 + bigint_hash(key) */
+     * INITIAL_HASH_VALUE=0: h = 31*0 + typeHash(key); bigint typeHash =
+     * AbstractLongType.java:137-140, varchar typeHash = XxHash64 over the
+     * bytes (AbstractVariableWidthBlock.java:102-105) */
+    const pg_col& c = pg.cols[key_col];
+    if (c.tag == PG_T_VARBIN) {
+        int32_t b0 = c.offsets[i], b1 = c.offsets[i + 1];
+        return pg_partition(
+            pg_xxh64((const uint8_t*)c.data + b0, (uint64_t)(b1 - b0)), np);
+    }
+    int64_t key = d_load_i64(c, i);
     return pg_partition(pg_bigint_hash(key), np);
 }
 
@@ -1566,6 +1583,32 @@ struct StagedPage {
         pg = *in;
         for (int c = 0; c < in->n_cols; c++) {
             if (!in->cols[c].data) continue;
+            if (in->cols[c].tag == PG_T_VARBIN && !in->cols[c].on_device) {
+                /* bytes buffer + offsets[n_rows+1] */
+                const int32_t* ho = in->cols[c].offsets;
+                size_t nb = (size_t)ho[in->n_rows];
+                bufs.emplace_back();
+                bufs.back().alloc(nb ? nb : 1);
+                CHKV(hipMemcpyAsync(bufs.back().p, in->cols[c].data, nb,
+                                    hipMemcpyHostToDevice, g_stream));
+                pg.cols[c].data = bufs.back().p;
+                bufs.emplace_back();
+                bufs.back().alloc(((size_t)in->n_rows + 1) * 4);
+                CHKV(hipMemcpyAsync(bufs.back().p, ho,
+                                    ((size_t)in->n_rows + 1) * 4,
+                                    hipMemcpyHostToDevice, g_stream));
+                pg.cols[c].offsets = (const int32_t*)bufs.back().p;
+                pg.cols[c].on_device = 1;
+                if (in->cols[c].null_mask) {
+                    bufs.emplace_back();
+                    bufs.back().alloc(in->n_rows);
+                    CHKV(hipMemcpyAsync(bufs.back().p, in->cols[c].null_mask,
+                                        in->n_rows, hipMemcpyHostToDevice,
+                                        g_stream));
+                    pg.cols[c].null_mask = (const uint8_t*)bufs.back().p;
+                }
+                continue;
+            }
             if (!in->cols[c].on_device) {
                 size_t n = (size_t)in->n_rows * type_size(in->cols[c].tag);
                 bufs.emplace_back();
@@ -1709,6 +1752,10 @@ struct FilterOp : Op {
             int tag = plan.proj[o].kind == PG_PROJ_IDENT
                           ? sp.pg.cols[plan.proj[o].a].tag
                           : PG_T_F64;
+            if (tag == PG_T_VARBIN)
+                throw std::runtime_error(
+                    "VARBIN projection/emit not supported in v1 "
+                    "(predicates and hashing only)");
             op.dev.emplace_back();
             op.dev.back().alloc((size_t)r.n * type_size(tag));
             op.pg.cols[o].tag = tag;

@@ -12,7 +12,7 @@ _HERE = pathlib.Path(__file__).resolve().parent
 _SO = _HERE / "libpresto_gpu.so"
 
 # ---- enums (presto_gpu.h) ----
-T_U8, T_I32, T_I64, T_F64 = 0, 1, 2, 3
+T_U8, T_I32, T_I64, T_F64, T_VARBIN = 0, 1, 2, 3, 4
 CMP_LT, CMP_LE, CMP_GT, CMP_GE, CMP_EQ, CMP_NE = range(6)
 PROJ_IDENT, PROJ_DISC_PRICE, PROJ_CHARGE = range(3)
 AGG_COUNT, AGG_SUM_F64, AGG_SUM_DEC, AGG_SUM_I64 = range(4)
@@ -26,7 +26,8 @@ _TAG_NP = {v: k for k, v in _NP_TAG.items()}
 
 class PgCol(C.Structure):
     _fields_ = [("tag", C.c_int32), ("on_device", C.c_int32),
-                ("data", C.c_void_p), ("null_mask", C.c_void_p)]
+                ("data", C.c_void_p), ("null_mask", C.c_void_p),
+                ("offsets", C.c_void_p)]
 
 
 class PgPage(C.Structure):
@@ -36,7 +37,8 @@ class PgPage(C.Structure):
 
 class Pred(C.Structure):
     _fields_ = [("col", C.c_int32), ("op", C.c_int32), ("ival", C.c_int64),
-                ("dval", C.c_double)]
+                ("dval", C.c_double), ("sval", C.c_char * 16),
+                ("slen", C.c_int32)]
 
 
 class Proj(C.Structure):
