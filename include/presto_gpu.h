@@ -105,6 +105,10 @@ typedef struct {
     double dval;   /* compare value for f64 columns */
     char sval[16]; /* VARBIN: compare bytes (EQ/NE only) */
     int32_t slen;
+    int32_t rhs_col; /* 0: compare against the constant; k>0: compare
+                        against integer channel k-1 (filter expression
+                        col OP col, e.g. Q5's c_nationkey = s_nationkey).
+                        Zero-initialized plans keep constant semantics. */
 } pg_pred;
 
 /* projection expressions (PageProjection analogs) */
@@ -173,6 +177,9 @@ typedef struct {
     uint8_t key_vals[2][PG_MAX_KEYVALS];
     int32_t n_aggs;
     pg_agg aggs[PG_MAX_AGG];
+    int32_t drop_unlisted_keys; /* 1: rows whose key is outside key_vals are
+                                   dropped (dictionary-filter pushdown, e.g.
+                                   Q5's region membership); 0: error */
 } pg_plan_hash_agg_small;
 
 typedef struct {

@@ -308,6 +308,101 @@ int64_t oracle_join_bigint(int64_t n_build, const int64_t* bkeys,
     return out;
 }
 
+/* ---------------- Q5 ---------------- */
+
+#define Q5_DATE_LO 8766 /* 1994-01-01 */
+#define Q5_DATE_HI 9131 /* 1995-01-01 */
+
+int32_t oracle_q5(int64_t n_cust, const int64_t* ck, const uint8_t* cnat,
+                  int64_t n_ord, const int64_t* ook, const int64_t* ock,
+                  const int32_t* odate, int64_t n_li, const int64_t* lok,
+                  const int64_t* lsk, const double* lep, const double* ldisc,
+                  int64_t n_supp, const uint8_t* snat, q5_row_t* out)
+{
+    extern int32_t tpch_nation_region(int32_t);
+    extern int32_t tpch_nation_name(int32_t, char*);
+    /* custkey -> nationkey (custkeys dense 1..n) */
+    int64_t max_ck = 0;
+    for (int64_t i = 0; i < n_cust; i++)
+        if (ck[i] > max_ck) max_ck = ck[i];
+    uint8_t* cn = (uint8_t*)calloc(max_ck + 1, 1);
+    for (int64_t i = 0; i < n_cust; i++) cn[ck[i]] = cnat[i];
+    /* orders filter (date) -> table orderkey -> customer nation */
+    int64_t* b_ok = (int64_t*)malloc(n_ord * sizeof(int64_t));
+    uint8_t* b_cn = (uint8_t*)malloc(n_ord);
+    int64_t n_b = 0;
+    for (int64_t i = 0; i < n_ord; i++) {
+        if (odate[i] >= Q5_DATE_LO && odate[i] < Q5_DATE_HI &&
+            ock[i] <= max_ck) {
+            b_ok[n_b] = ook[i];
+            b_cn[n_b] = cn[ock[i]];
+            n_b++;
+        }
+    }
+    int64_t cap = hash_capacity(n_b < 2 ? 2 : n_b);
+    int64_t* slot = (int64_t*)malloc(cap * sizeof(int64_t));
+    memset(slot, -1, cap * sizeof(int64_t));
+    for (int64_t i = 0; i < n_b; i++) {
+        uint64_t h = pg_murmur3_finalize(pg_bigint_hash(b_ok[i]));
+        int64_t pos = (int64_t)(h & (cap - 1));
+        while (slot[pos] != -1) pos = (pos + 1) & (cap - 1);
+        slot[pos] = i;
+    }
+    /* lineitem probe: match order, then local-supplier condition */
+    int64_t rev[25];
+    memset(rev, 0, sizeof(rev));
+#pragma omp parallel
+    {
+        int64_t loc[25];
+        memset(loc, 0, sizeof(loc));
+#pragma omp for schedule(static)
+        for (int64_t i = 0; i < n_li; i++) {
+            uint64_t h = pg_murmur3_finalize(pg_bigint_hash(lok[i]));
+            int64_t pos = (int64_t)(h & (cap - 1));
+            int64_t r = -1;
+            for (;;) {
+                int64_t s = slot[pos];
+                if (s == -1) break;
+                if (b_ok[s] == lok[i]) {
+                    r = s;
+                    break;
+                }
+                pos = (pos + 1) & (cap - 1);
+            }
+            if (r == -1) continue;
+            int64_t sk = lsk[i];
+            if (sk < 1 || sk > n_supp) continue;
+            uint8_t sn = snat[sk - 1];
+            if (sn != b_cn[r] || tpch_nation_region(sn) != 2) continue;
+            int64_t cents = (int64_t)(lep[i] * 100.0 + 0.5);
+            int64_t d = (int64_t)(ldisc[i] * 100.0 + 0.5);
+            loc[sn] += cents * (100 - d);
+        }
+#pragma omp critical
+        for (int k = 0; k < 25; k++) rev[k] += loc[k];
+    }
+    /* emit nations with revenue, sorted desc (ties: nationkey asc) */
+    int32_t n_out = 0;
+    for (int k = 0; k < 25; k++) {
+        if (!rev[k]) continue;
+        q5_row_t row;
+        row.nationkey = (uint8_t)k;
+        row.revenue_1e4 = rev[k];
+        tpch_nation_name(k, row.name);
+        int32_t pos = n_out;
+        while (pos > 0 && (out[pos - 1].revenue_1e4 < row.revenue_1e4))
+            pos--;
+        for (int32_t j = n_out; j > pos; j--) out[j] = out[j - 1];
+        out[pos] = row;
+        n_out++;
+    }
+    free(cn);
+    free(b_ok);
+    free(b_cn);
+    free(slot);
+    return n_out;
+}
+
 /* ---------------- Q3 ---------------- */
 
 #define Q3_DATE 9204 /* DATE '1995-03-15', epoch days */

@@ -93,6 +93,27 @@ int32_t oracle_q3(int64_t n_cust, const int64_t* c_custkey,
                   const double* l_discount, const int32_t* l_shipdate,
                   int32_t limit, q3_row_t* out);
 
+/* ---------------- TPC-H Q5 ----------------
+ * SQL: presto-benchto-benchmarks/.../tpch/q05.sql (region 'ASIA',
+ * orderdate in [1994-01-01, 1995-01-01) = [8766, 9131) epoch days).
+ * 6-way join: customer x orders x lineitem x supplier x nation x region
+ * with the local-supplier condition c_nationkey = s_nationkey; grouped
+ * revenue per nation, sorted desc.  Exact ticks of 1e-4. */
+typedef struct {
+    uint8_t nationkey;
+    char name[32];
+    int64_t revenue_1e4;
+} q5_row_t;
+
+int32_t oracle_q5(int64_t n_cust, const int64_t* c_custkey,
+                  const uint8_t* c_nationkey, int64_t n_ord,
+                  const int64_t* o_orderkey, const int64_t* o_custkey,
+                  const int32_t* o_orderdate, int64_t n_li,
+                  const int64_t* l_orderkey, const int64_t* l_suppkey,
+                  const double* l_extendedprice, const double* l_discount,
+                  int64_t n_supp, const uint8_t* s_nationkey,
+                  q5_row_t* out /* capacity 25 */);
+
 /* ---------------- operator-level primitives (parity targets) ---------- */
 
 /* murmur3 finalizer bucket — PagesHash.java:236-252 /

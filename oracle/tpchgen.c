@@ -42,6 +42,9 @@ static inline int64_t unif(int64_t* s, int64_t lo, int64_t hi)
 #define SEED_L_RDAYS   373135028LL /* receipt= ship + 1..30,  usage 7/order */
 #define SEED_L_RFLAG   717419739LL /* returnflag pick, usage 7/order */
 #define SEED_C_MSEG   1140279430LL /* mktsegment pick, usage 1/customer */
+#define SEED_C_NATION 1489529863LL /* customer nationkey, usage 1/customer */
+#define SEED_S_NATION  110356601LL /* supplier nationkey, usage 1/supplier */
+#define SEED_L_SUPPN  2095021727LL /* lineitem supplier number 0..3, 7/order */
 
 /* ---- calendar ----
  * day index 1 = 1992-01-01; order-date index in [1, 2406]
@@ -54,6 +57,62 @@ static inline int64_t unif(int64_t* s, int64_t lo, int64_t hi)
 #define CUSTOMER_MORTALITY 3
 
 int64_t tpch_customer_count(double sf) { return (int64_t)(150000.0 * sf + 0.5); }
+int64_t tpch_supplier_count(double sf) { return (int64_t)(10000.0 * sf + 0.5); }
+
+/* dbgen nation table order; region keys per dists.dss */
+static const int32_t NATION_REGION[25] = {
+    0, 1, 1, 1, 4, 0, 3, 3, 2, 2, 4, 4, 2, 4, 0, 0, 0, 1, 2, 3, 4, 2, 3, 3,
+    1};
+static const char* NATION_NAME[25] = {
+    "ALGERIA", "ARGENTINA", "BRAZIL", "CANADA", "EGYPT", "ETHIOPIA",
+    "FRANCE", "GERMANY", "INDIA", "INDONESIA", "IRAN", "IRAQ", "JAPAN",
+    "JORDAN", "KENYA", "MOROCCO", "MOZAMBIQUE", "PERU", "CHINA", "ROMANIA",
+    "SAUDI ARABIA", "VIETNAM", "RUSSIA", "UNITED KINGDOM", "UNITED STATES"};
+int32_t tpch_nation_region(int32_t nk)
+{
+    return nk >= 0 && nk < 25 ? NATION_REGION[nk] : -1;
+}
+int32_t tpch_nation_name(int32_t nk, char* buf)
+{
+    if (nk < 0 || nk >= 25) return -1;
+    const char* s = NATION_NAME[nk];
+    int32_t n = 0;
+    while (s[n]) {
+        buf[n] = s[n];
+        n++;
+    }
+    buf[n] = 0;
+    return n;
+}
+
+/* dbgen PART_SUPP bridge (build.c PART_SUPP_BRIDGE):
+ * supplier of (partkey, i) = (partkey + i*(S/4 + (partkey-1)/S)) % S + 1 */
+static inline int64_t part_supplier(int64_t partkey, int64_t i, int64_t S)
+{
+    return (partkey + i * (S / 4 + (partkey - 1) / S)) % S + 1;
+}
+
+void tpch_gen_supplier(double sf, int64_t start, int64_t count,
+                       int64_t* suppkey, uint8_t* nationkey)
+{
+#pragma omp parallel
+    {
+#ifdef _OPENMP
+        extern int omp_get_num_threads(void), omp_get_thread_num(void);
+        int nt = omp_get_num_threads(), tid = omp_get_thread_num();
+#else
+        int nt = 1, tid = 0;
+#endif
+        (void)sf;
+        int64_t lo = count * tid / nt, hi = count * (tid + 1) / nt;
+        int64_t s = rng_skip(SEED_S_NATION, (uint64_t)(start + lo));
+        for (int64_t i = lo; i < hi; i++) {
+            int64_t nk = unif(&s, 0, 24);
+            if (nationkey) nationkey[i] = (uint8_t)nk;
+            if (suppkey) suppkey[i] = start + i + 1;
+        }
+    }
+}
 int64_t tpch_orders_count(double sf)   { return (int64_t)(1500000.0 * sf + 0.5); }
 
 /* dbgen mk_sparse (build.c): keep low 3 bits, shift the rest up by 2.
@@ -104,6 +163,32 @@ int64_t tpch_lineitem_offset(double sf, int64_t ord_start)
         for (int64_t i = lo; i < hi; i++) total += unif(&s, 1, 7);
     }
     return total;
+}
+
+void tpch_gen_customer2(double sf, int64_t start, int64_t count,
+                        int64_t* custkey, uint8_t* mktseg_id,
+                        uint8_t* nationkey)
+{
+#pragma omp parallel
+    {
+#ifdef _OPENMP
+        extern int omp_get_num_threads(void), omp_get_thread_num(void);
+        int nt = omp_get_num_threads(), tid = omp_get_thread_num();
+#else
+        int nt = 1, tid = 0;
+#endif
+        (void)sf;
+        int64_t lo = count * tid / nt, hi = count * (tid + 1) / nt;
+        int64_t s = rng_skip(SEED_C_MSEG, (uint64_t)(start + lo));
+        int64_t sn = rng_skip(SEED_C_NATION, (uint64_t)(start + lo));
+        for (int64_t i = lo; i < hi; i++) {
+            int64_t j = unif(&s, 1, 5);
+            int64_t nk = unif(&sn, 0, 24);
+            if (mktseg_id) mktseg_id[i] = (uint8_t)(j - 1);
+            if (nationkey) nationkey[i] = (uint8_t)nk;
+            if (custkey) custkey[i] = start + i + 1;
+        }
+    }
 }
 
 void tpch_gen_customer(double sf, int64_t start, int64_t count,
@@ -185,9 +270,10 @@ static int64_t gen_lineitem_chunk(double sf, int64_t ord_start,
                                   double* quantity, double* extendedprice,
                                   double* discount, double* tax,
                                   int32_t* shipdate_epoch, uint8_t* returnflag,
-                                  uint8_t* linestatus)
+                                  uint8_t* linestatus, int64_t* suppkey)
 {
     int64_t max_pkey = (int64_t)(200000.0 * sf + 0.5);
+    int64_t n_supp = tpch_supplier_count(sf);
     int64_t s_lcnt = rng_skip(SEED_O_LCNT, (uint64_t)ord_start);
     int64_t s_odate = rng_skip(SEED_O_ODATE, (uint64_t)ord_start);
     uint64_t l7 = (uint64_t)ord_start * 7;
@@ -199,6 +285,7 @@ static int64_t gen_lineitem_chunk(double sf, int64_t ord_start,
     int64_t s_cdays = rng_skip(SEED_L_CDAYS, l7);
     int64_t s_rdays = rng_skip(SEED_L_RDAYS, l7);
     int64_t s_rflag = rng_skip(SEED_L_RFLAG, l7);
+    int64_t s_suppn = rng_skip(SEED_L_SUPPN, l7);
     int64_t out = 0;
     for (int64_t o = 0; o < ord_count; o++) {
         int64_t idx1 = ord_start + o + 1;
@@ -211,6 +298,7 @@ static int64_t gen_lineitem_chunk(double sf, int64_t ord_start,
             int64_t d = unif(&s_disc, 0, 10);
             int64_t t = unif(&s_tax, 0, 8);
             int64_t pk = unif(&s_pkey, 1, max_pkey);
+            int64_t sn = unif(&s_suppn, 0, 3);
             int64_t sd = unif(&s_sdays, 1, 121);
             (void)unif(&s_cdays, 30, 90); /* commitdate: stream consumed */
             int64_t rd = unif(&s_rdays, 1, 30);
@@ -236,6 +324,7 @@ static int64_t gen_lineitem_chunk(double sf, int64_t ord_start,
                 shipdate_epoch[out] = (int32_t)(EPOCH_1992 + ship - 1);
             if (returnflag) returnflag[out] = rf;
             if (linestatus) linestatus[out] = (ship <= CURRENT_IDX) ? 'F' : 'O';
+            if (suppkey) suppkey[out] = part_supplier(pk, sn, n_supp);
             out++;
         }
         /* rowFinished: advance every per-line stream to usage 7/order
@@ -249,6 +338,7 @@ static int64_t gen_lineitem_chunk(double sf, int64_t ord_start,
         s_cdays = rng_skip(s_cdays, rest);
         s_rdays = rng_skip(s_rdays, rest);
         s_rflag = rng_skip(s_rflag, (uint64_t)(7 - rflag_draws));
+        s_suppn = rng_skip(s_suppn, rest);
     }
     return out;
 }
@@ -282,7 +372,40 @@ int64_t tpch_gen_lineitem(double sf, int64_t ord_start, int64_t ord_count,
             discount ? discount + off : 0, tax ? tax + off : 0,
             shipdate_epoch ? shipdate_epoch + off : 0,
             returnflag ? returnflag + off : 0,
-            linestatus ? linestatus + off : 0);
+            linestatus ? linestatus + off : 0, 0);
+    }
+    return written;
+}
+
+int64_t tpch_gen_lineitem2(double sf, int64_t ord_start, int64_t ord_count,
+                           int64_t* orderkey, double* quantity,
+                           double* extendedprice, double* discount,
+                           double* tax, int32_t* shipdate_epoch,
+                           uint8_t* returnflag, uint8_t* linestatus,
+                           int64_t* suppkey)
+{
+    int64_t written = 0;
+#pragma omp parallel reduction(+ : written)
+    {
+#ifdef _OPENMP
+        extern int omp_get_num_threads(void), omp_get_thread_num(void);
+        int nt = omp_get_num_threads(), tid = omp_get_thread_num();
+#else
+        int nt = 1, tid = 0;
+#endif
+        int64_t lo = ord_count * tid / nt, hi = ord_count * (tid + 1) / nt;
+        int64_t s = rng_skip(SEED_O_LCNT, (uint64_t)ord_start);
+        int64_t off = 0;
+        for (int64_t i = 0; i < lo; i++) off += unif(&s, 1, 7);
+        written += gen_lineitem_chunk(
+            sf, ord_start + lo, hi - lo, orderkey ? orderkey + off : 0,
+            quantity ? quantity + off : 0,
+            extendedprice ? extendedprice + off : 0,
+            discount ? discount + off : 0, tax ? tax + off : 0,
+            shipdate_epoch ? shipdate_epoch + off : 0,
+            returnflag ? returnflag + off : 0,
+            linestatus ? linestatus + off : 0,
+            suppkey ? suppkey + off : 0);
     }
     return written;
 }

@@ -36,9 +36,34 @@ int64_t tpch_lineitem_count(double sf);
 
 /* customer columns for rows [start, start+count) (0-based row index).
  * custkey = row index + 1.  mktseg_id: 0=AUTOMOBILE 1=BUILDING 2=FURNITURE
- * 3=MACHINERY 4=HOUSEHOLD (distribution order of dbgen dists.dss msegmnt). */
+ * 3=MACHINERY 4=HOUSEHOLD (distribution order of dbgen dists.dss msegmnt).
+ * nationkey: uniform 0..24 (dbgen nation table order). */
 void tpch_gen_customer(double sf, int64_t start, int64_t count,
                        int64_t* custkey /*nullable*/, uint8_t* mktseg_id);
+void tpch_gen_customer2(double sf, int64_t start, int64_t count,
+                        int64_t* custkey, uint8_t* mktseg_id,
+                        uint8_t* nationkey);
+
+/* supplier: 10,000 * sf rows; suppkey = index + 1 */
+int64_t tpch_supplier_count(double sf);
+void tpch_gen_supplier(double sf, int64_t start, int64_t count,
+                       int64_t* suppkey, uint8_t* nationkey);
+
+/* dbgen nation table (25 rows): returns region key of nation 0..24
+ * (0 AFRICA, 1 AMERICA, 2 ASIA, 3 EUROPE, 4 MIDDLE EAST) */
+int32_t tpch_nation_region(int32_t nationkey);
+/* nation name (dbgen order) into buf (<=32B incl NUL); returns length */
+int32_t tpch_nation_name(int32_t nationkey, char* buf);
+
+/* lineitem with suppkey: same as tpch_gen_lineitem plus l_suppkey
+ * (dbgen PART_SUPP bridge: selectPartSupplier(partkey, supplierNumber,
+ * supplierCount), supplierNumber stream 0..3) */
+int64_t tpch_gen_lineitem2(double sf, int64_t ord_start, int64_t ord_count,
+                           int64_t* orderkey, double* quantity,
+                           double* extendedprice, double* discount,
+                           double* tax, int32_t* shipdate_epoch,
+                           uint8_t* returnflag, uint8_t* linestatus,
+                           int64_t* suppkey);
 
 /* orders columns for order rows [start, start+count) (0-based).
  * shippriority is the constant 0 (dbgen mk_order) and is not emitted.

@@ -220,7 +220,14 @@ __device__ inline bool d_eval_preds(const pg_page& pg, const pg_pred* preds,
             }
         } else {
             int64_t v = d_load_i64(c, i);
-            int64_t x = pr.ival;
+            int64_t x;
+            if (pr.rhs_col > 0) {
+                const pg_col& rc = pg.cols[pr.rhs_col - 1];
+                if (rc.null_mask && rc.null_mask[i]) return false;
+                x = d_load_i64(rc, i);
+            } else {
+                x = pr.ival;
+            }
             switch (pr.op) {
                 case PG_CMP_LT: ok = v < x; break;
                 case PG_CMP_LE: ok = v <= x; break;
@@ -332,7 +339,7 @@ __global__ __launch_bounds__(FT_NTHREADS) void k_agg_small(
                         idx1 = j;
             }
             if (idx0 < 0 || idx1 < 0) {
-                local_bad++;
+                if (!plan.drop_unlisted_keys) local_bad++;
                 continue;
             }
             int g = idx0 * n1 + idx1;
@@ -1835,7 +1842,7 @@ struct AggSmallOp : Op {
             return false;
         if (na != 7 || maxg != 6) return false;
         if (pg.cols[p.preds[0].col].tag != PG_T_I32 ||
-            p.preds[0].op != PG_CMP_LE)
+            p.preds[0].op != PG_CMP_LE || p.preds[0].rhs_col != 0)
             return false;
         const pg_agg* a = p.aggs;
         if (!(a[0].func != PG_AGG_COUNT && a[0].proj.kind == PG_PROJ_IDENT &&
@@ -2282,7 +2289,7 @@ struct JoinOp : Op {
                         !((uintptr_t)sp.pg.cols[plan.proj.b].data & 15);
             if (plan.n_preds == 1) {
                 const pg_pred& pr = plan.preds[0];
-                spec = spec &&
+                spec = spec && pr.rhs_col == 0 &&
                        sp.pg.cols[pr.col].tag == PG_T_I32 &&
                        (pr.op == PG_CMP_GT || pr.op == PG_CMP_LT) &&
                        !((uintptr_t)sp.pg.cols[pr.col].data & 7) &&

@@ -38,7 +38,7 @@ class PgPage(C.Structure):
 class Pred(C.Structure):
     _fields_ = [("col", C.c_int32), ("op", C.c_int32), ("ival", C.c_int64),
                 ("dval", C.c_double), ("sval", C.c_char * 16),
-                ("slen", C.c_int32)]
+                ("slen", C.c_int32), ("rhs_col", C.c_int32)]
 
 
 class Proj(C.Structure):
@@ -59,7 +59,8 @@ class PlanHashAggSmall(C.Structure):
     _fields_ = [("n_preds", C.c_int32), ("preds", Pred * 8),
                 ("n_keys", C.c_int32), ("key_col", C.c_int32 * 2),
                 ("n_vals", C.c_int32 * 2), ("key_vals", (C.c_uint8 * 8) * 2),
-                ("n_aggs", C.c_int32), ("aggs", Agg * 8)]
+                ("n_aggs", C.c_int32), ("aggs", Agg * 8),
+                ("drop_unlisted_keys", C.c_int32)]
 
 
 class PlanHashBuild(C.Structure):

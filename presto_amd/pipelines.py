@@ -167,3 +167,129 @@ def q3(cust: Page, orders: Page, lineitem: Page, mode="dec", limit=10):
         return p.run(lineitem)
     finally:
         p.close()
+
+
+class Q5Pipeline:
+    """Q5 operator graph (q05.sql): region 'ASIA', orderdate in
+    [1994-01-01, 1995-01-01), local-supplier join c_nationkey=s_nationkey,
+    revenue grouped per nation.
+
+    customer -> custkey->nationkey table
+    orders(date) probe it (emit orderkey + cust nation) -> orderkey table
+    supplier -> suppkey->nationkey table
+    lineitem probe orders-table (emit suppkey, price, disc + cust nation)
+             probe supplier-table (emit ... + supplier nation)
+    small-key agg: key = supplier nation restricted to ASIA nations
+    (drop_unlisted = the region membership), pred cnat == snat (col-col).
+    """
+    Q5_LO, Q5_HI = 8766, 9131
+    ASIA = (8, 9, 12, 18, 21)
+
+    def __init__(self, cust: Page, orders: Page, supp: Page):
+        b1 = PlanHashBuild()
+        b1.key_col = cust.channel("custkey")
+        b1.semijoin_table = -1
+        b1.n_payload = 1
+        b1.payload_col[0] = cust.channel("nationkey")
+        b1.capacity_hint = cust.n_rows
+        self.b1 = Operator(OP_HASH_BUILD, b1)
+        self.b1.add_input(cust)
+        self.b1.finish()
+
+        jo = PlanLookupJoin()
+        jo.table = self.b1.table()
+        jo.n_preds = 2
+        jo.preds[0] = Pred(orders.channel("orderdate"), CMP_GE, self.Q5_LO,
+                           0.0)
+        jo.preds[1] = Pred(orders.channel("orderdate"), CMP_LT, self.Q5_HI,
+                           0.0)
+        jo.key_col = orders.channel("custkey")
+        jo.mode = 0
+        jo.n_emit = 1
+        jo.emit_probe_cols[0] = orders.channel("orderkey")
+        j = Operator(OP_LOOKUP_JOIN, jo)
+        j.add_input(orders)
+        opage = j.get_output_raw()  # [orderkey, cust_nationkey]
+
+        b2 = PlanHashBuild()
+        b2.key_col = 0
+        b2.semijoin_table = -1
+        b2.n_payload = 1
+        b2.payload_col[0] = 1
+        b2.capacity_hint = max(opage.n_rows, 16)
+        self.b2 = Operator(OP_HASH_BUILD, b2)
+        self.b2.add_input_raw(opage)
+        self.b2.finish()
+        j.destroy()
+
+        b3 = PlanHashBuild()
+        b3.key_col = supp.channel("suppkey")
+        b3.semijoin_table = -1
+        b3.n_payload = 1
+        b3.payload_col[0] = supp.channel("nationkey")
+        b3.capacity_hint = supp.n_rows
+        self.b3 = Operator(OP_HASH_BUILD, b3)
+        self.b3.add_input(supp)
+        self.b3.finish()
+
+    def run(self, li: Page):
+        j1 = PlanLookupJoin()
+        j1.table = self.b2.table()
+        j1.key_col = li.channel("orderkey")
+        j1.mode = 0
+        j1.n_emit = 3
+        j1.emit_probe_cols[0] = li.channel("suppkey")
+        j1.emit_probe_cols[1] = li.channel("extendedprice")
+        j1.emit_probe_cols[2] = li.channel("discount")
+        ja = Operator(OP_LOOKUP_JOIN, j1)
+        ja.add_input(li)
+        pa = ja.get_output_raw()  # [suppkey, ep, dc, cnat]
+
+        j2 = PlanLookupJoin()
+        j2.table = self.b3.table()
+        j2.key_col = 0
+        j2.mode = 0
+        j2.n_emit = 3
+        j2.emit_probe_cols[0] = 1
+        j2.emit_probe_cols[1] = 2
+        j2.emit_probe_cols[2] = 3
+        jb = Operator(OP_LOOKUP_JOIN, j2)
+        jb.add_input_raw(pa)
+        pb = jb.get_output_raw()  # [ep, dc, cnat, snat]
+        ja.destroy()
+
+        agg = PlanHashAggSmall()
+        agg.n_preds = 1
+        p = Pred(2, CMP_EQ, 0, 0.0)
+        p.rhs_col = 3 + 1  # compare channel 2 (cnat) == channel 3 (snat)
+        agg.preds[0] = p
+        agg.n_keys = 1
+        agg.key_col[0] = 3
+        agg.n_vals[0] = len(self.ASIA)
+        for i, v in enumerate(self.ASIA):
+            agg.key_vals[0][i] = v
+        agg.n_aggs = 2
+        agg.aggs[0] = Agg(AGG_SUM_DEC, Proj(PROJ_DISC_PRICE, 0, 1, 0), 4)
+        agg.aggs[1] = Agg(AGG_COUNT, Proj(PROJ_IDENT, 0, 0, 0), 0)
+        agg.drop_unlisted_keys = 1
+        ao = Operator(OP_HASH_AGG_SMALL, agg)
+        ao.add_input_raw(pb)
+        ao.finish()
+        out = ao.get_output(["nationkey", "rev_hi", "rev_lo", "count"])
+        ao.destroy()
+        jb.destroy()
+        return out
+
+    def close(self):
+        from .engine import lib
+        for b in (self.b1, self.b2, self.b3):
+            lib().c.pg_table_destroy(b.table())
+            b.destroy()
+
+
+def q5(cust: Page, orders: Page, supp: Page, li: Page):
+    p = Q5Pipeline(cust, orders, supp)
+    try:
+        return p.run(li)
+    finally:
+        p.close()

@@ -139,3 +139,71 @@ class OracleLib:
             if n <= cap:
                 return op[:n], ob[:n]
             cap = n  # function reports the true total; retry sized
+
+
+class Q5Row(C.Structure):
+    _fields_ = [("nationkey", C.c_uint8), ("name", C.c_char * 32),
+                ("revenue_1e4", C.c_int64)]
+
+
+def _bind_q5(self):
+    self.lib.oracle_q5.restype = C.c_int32
+
+
+def gen_lineitem2(self, sf):
+    n = self.lineitem_count(sf)
+    n_ord = self.lib.tpch_orders_count(C.c_double(sf))
+    cols = dict(
+        orderkey=np.empty(n, np.int64), quantity=np.empty(n, np.float64),
+        extendedprice=np.empty(n, np.float64),
+        discount=np.empty(n, np.float64), tax=np.empty(n, np.float64),
+        shipdate=np.empty(n, np.int32), returnflag=np.empty(n, np.uint8),
+        linestatus=np.empty(n, np.uint8), suppkey=np.empty(n, np.int64))
+    w = self.lib.tpch_gen_lineitem2(
+        C.c_double(sf), C.c_int64(0), C.c_int64(n_ord),
+        _p(cols["orderkey"]), _p(cols["quantity"]),
+        _p(cols["extendedprice"]), _p(cols["discount"]), _p(cols["tax"]),
+        _p(cols["shipdate"]), _p(cols["returnflag"]), _p(cols["linestatus"]),
+        _p(cols["suppkey"]))
+    assert w == n
+    return cols
+
+
+def gen_customer2(self, sf):
+    n = self.lib.tpch_customer_count(C.c_double(sf))
+    cols = dict(custkey=np.empty(n, np.int64), mktseg=np.empty(n, np.uint8),
+                nationkey=np.empty(n, np.uint8))
+    self.lib.tpch_gen_customer2(C.c_double(sf), C.c_int64(0), C.c_int64(n),
+                                _p(cols["custkey"]), _p(cols["mktseg"]),
+                                _p(cols["nationkey"]))
+    return cols
+
+
+def gen_supplier(self, sf):
+    n = self.lib.tpch_supplier_count(C.c_double(sf))
+    cols = dict(suppkey=np.empty(n, np.int64),
+                nationkey=np.empty(n, np.uint8))
+    self.lib.tpch_gen_supplier(C.c_double(sf), C.c_int64(0), C.c_int64(n),
+                               _p(cols["suppkey"]), _p(cols["nationkey"]))
+    return cols
+
+
+def q5(self, cust, orders, li, supp):
+    self.lib.tpch_gen_lineitem2.restype = C.c_int64
+    rows = (Q5Row * 25)()
+    self.lib.oracle_q5.restype = C.c_int32
+    nr = self.lib.oracle_q5(
+        C.c_int64(len(cust["custkey"])), _p(cust["custkey"]),
+        _p(cust["nationkey"]), C.c_int64(len(orders["orderkey"])),
+        _p(orders["orderkey"]), _p(orders["custkey"]),
+        _p(orders["orderdate"]), C.c_int64(len(li["orderkey"])),
+        _p(li["orderkey"]), _p(li["suppkey"]), _p(li["extendedprice"]),
+        _p(li["discount"]), C.c_int64(len(supp["suppkey"])),
+        _p(supp["nationkey"]), rows)
+    return [rows[i] for i in range(nr)]
+
+
+OracleLib.gen_lineitem2 = gen_lineitem2
+OracleLib.gen_customer2 = gen_customer2
+OracleLib.gen_supplier = gen_supplier
+OracleLib.q5 = q5

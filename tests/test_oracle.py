@@ -272,3 +272,18 @@ def test_xxh64_restatement(oracle_lib):
         data = bytes(rng.integers(0, 256, ln, dtype=np.uint8))
         assert L.oracle_xxh64(data, ln) == _xxh64_py(data), ln
     assert L.oracle_xxh64(b"BUILDING", 8) == _xxh64_py(b"BUILDING")
+
+
+def test_q5_sf1_golden(oracle_lib):
+    """Q5 golden pin: also pins the customer/supplier nationkey streams and
+    the dbgen PART_SUPP supplier bridge formula."""
+    li = oracle_lib.gen_lineitem2(1.0)
+    orders = oracle_lib.gen_orders(1.0)
+    cust = oracle_lib.gen_customer2(1.0)
+    supp = oracle_lib.gen_supplier(1.0)
+    rows = oracle_lib.q5(cust, orders, li, supp)
+    golden = _parse_golden("q05_sf1.result")
+    assert len(rows) == len(golden) == 5
+    for r, g in zip(rows, golden):
+        assert r.name.decode() == g[0]
+        assert Decimal(r.revenue_1e4) / 10**4 == Decimal(g[1])
