@@ -42,6 +42,9 @@ Q1_WORKLOAD = ("TPC-H SF{sf} Q1 on {n}xMI355X - scan/filter/hash-aggregate "
                "kernels, lineitem columns resident in HBM")
 Q3_WORKLOAD = ("TPC-H SF{sf} Q3 on {n}xMI355X - 3-way hash join "
                "(HashBuilder+LookupJoin) + order-by/limit")
+Q5_WORKLOAD = ("TPC-H SF{sf} Q5 on {n}xMI355X - 6-way join "
+               "(customer/orders/lineitem/supplier/nation/region), "
+               "local-supplier condition, per-nation revenue")
 
 # Q1 algorithmic bytes/row: 4 f64 money cols +
 # dict-u8 returnflag/linestatus + date32 = 38 B
@@ -64,6 +67,7 @@ def load_oracle():
 
 
 def gen_lineitem_device(orc, sf, device, want_orderkey=False,
+                        want_suppkey=False,
                         chunk_orders=4_000_000, ord_start=0, ord_count=None):
     """Generate lineitem columns for orders [ord_start, ord_start+ord_count)
     of logical scale sf, chunkwise on host, uploaded to device tensors.
@@ -86,6 +90,8 @@ def gen_lineitem_device(orc, sf, device, want_orderkey=False,
     }
     if want_orderkey:
         cols["orderkey"] = torch.empty(n, dtype=torch.int64, device=device)
+    if want_suppkey:
+        cols["suppkey"] = torch.empty(n, dtype=torch.int64, device=device)
     maxrows = chunk_orders * 7
     # pinned host staging: the generator writes straight into pinned
     # memory, then one async h2d copy per column per chunk
@@ -98,12 +104,19 @@ def gen_lineitem_device(orc, sf, device, want_orderkey=False,
     t0 = time.time()
     while o < end:
         cnt = min(chunk_orders, end - o)
-        w = orc.lib.tpch_gen_lineitem(
-            C.c_double(sf), C.c_int64(o), C.c_int64(cnt),
-            ptr["orderkey"] if want_orderkey else None,
-            ptr["quantity"], ptr["extendedprice"],
-            ptr["discount"], ptr["tax"],
-            ptr["shipdate"], ptr["returnflag"], ptr["linestatus"])
+        if want_suppkey:
+            w = orc.lib.tpch_gen_lineitem2(
+                C.c_double(sf), C.c_int64(o), C.c_int64(cnt),
+                ptr["orderkey"], ptr["quantity"], ptr["extendedprice"],
+                ptr["discount"], ptr["tax"], ptr["shipdate"],
+                ptr["returnflag"], ptr["linestatus"], ptr["suppkey"])
+        else:
+            w = orc.lib.tpch_gen_lineitem(
+                C.c_double(sf), C.c_int64(o), C.c_int64(cnt),
+                ptr["orderkey"] if want_orderkey else None,
+                ptr["quantity"], ptr["extendedprice"],
+                ptr["discount"], ptr["tax"],
+                ptr["shipdate"], ptr["returnflag"], ptr["linestatus"])
         for k, t in cols.items():
             t[off:off + w].copy_(buf[k][:w], non_blocking=True)
         torch.cuda.synchronize()
@@ -184,7 +197,7 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=8)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--query", choices=["q1", "q3"], default="q1")
+    ap.add_argument("--query", choices=["q1", "q3", "q5"], default="q1")
     ap.add_argument("--sf", type=float, default=100.0)
     ap.add_argument("--mode", choices=["f64", "dec"], default="f64")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
@@ -267,7 +280,7 @@ def main():
             return out
         total_rows_per_step = n_rows * n_gpus
         workload = Q1_WORKLOAD.format(sf=int(sf_total), n=n_gpus)
-    else:
+    elif args.query == "q3":
         cols, n_rows = gen_lineitem_device(orc, sf_total, device,
                                            want_orderkey=True,
                                            ord_start=o0, ord_count=o1 - o0)
@@ -285,6 +298,49 @@ def main():
             return pipelines.q3(cust_page, ord_page, li_page, mode="dec")
         total_rows_per_step = n_rows * n_gpus
         workload = Q3_WORKLOAD.format(sf=int(sf_total), n=n_gpus)
+    else:  # q5 (single-GPU round-1 path)
+        if world > 1:
+            raise SystemExit("q5 multi-GPU lands in round 2")
+        cols, n_rows = gen_lineitem_device(orc, sf_total, device,
+                                           want_orderkey=True,
+                                           want_suppkey=True,
+                                           ord_start=o0, ord_count=o1 - o0)
+        li_page = Page(cols, n_rows=n_rows)
+        n_cust = orc.lib.tpch_customer_count(C.c_double(sf_total))
+        cck = np.empty(n_cust, np.int64)
+        cnat = np.empty(n_cust, np.uint8)
+        orc.lib.tpch_gen_customer2(C.c_double(sf_total), C.c_int64(0),
+                                   C.c_int64(n_cust),
+                                   C.c_void_p(cck.ctypes.data), None,
+                                   C.c_void_p(cnat.ctypes.data))
+        n_supp = orc.lib.tpch_supplier_count(C.c_double(sf_total))
+        ssk = np.empty(n_supp, np.int64)
+        snat = np.empty(n_supp, np.uint8)
+        orc.lib.tpch_gen_supplier(C.c_double(sf_total), C.c_int64(0),
+                                  C.c_int64(n_supp),
+                                  C.c_void_p(ssk.ctypes.data),
+                                  C.c_void_p(snat.ctypes.data))
+        n_ordx = orc.lib.tpch_orders_count(C.c_double(sf_total))
+        ook = np.empty(n_ordx, np.int64)
+        ock = np.empty(n_ordx, np.int64)
+        ood = np.empty(n_ordx, np.int32)
+        orc.lib.tpch_gen_orders(C.c_double(sf_total), C.c_int64(0),
+                                C.c_int64(n_ordx),
+                                C.c_void_p(ook.ctypes.data),
+                                C.c_void_p(ock.ctypes.data),
+                                C.c_void_p(ood.ctypes.data), None)
+        t = lambda a: torch.from_numpy(a).to(device)
+        cust_page = Page({"custkey": t(cck), "nationkey": t(cnat)})
+        ord_page = Page({"orderkey": t(ook), "custkey": t(ock),
+                         "orderdate": t(ood)})
+        supp_page = Page({"suppkey": t(ssk), "nationkey": t(snat)})
+        li5 = Page({k: cols[k] for k in ("orderkey", "suppkey",
+                                         "extendedprice", "discount")})
+
+        def step():
+            return pipelines.q5(cust_page, ord_page, supp_page, li5)
+        total_rows_per_step = n_rows * n_gpus
+        workload = Q5_WORKLOAD.format(sf=int(sf_total), n=n_gpus)
 
     if world > 1:
         import torch.distributed as dist
