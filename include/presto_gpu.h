@@ -58,6 +58,10 @@ pg_status pg_memcpy_d2d(void* dst, const void* src, int64_t bytes);
  * scan+filter+aggregate, or probe+aggregate) — measurement plumbing for
  * bench.py's roofline leg */
 double pg_last_hot_kernel_ms(void);
+/* max hot-region ms since pg_hot_reset (for pipelines with several hot
+ * launches per step: the max is the dominant one) */
+double pg_hot_max_ms(void);
+void pg_hot_reset(void);
 
 /* ---- Page / Block descriptors (SURVEY.md §8b struct) ----
  * Concrete layouts follow the Java blocks: fixed-width values array +

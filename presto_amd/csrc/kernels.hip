@@ -129,6 +129,7 @@ extern "C" pg_status pg_memcpy_d2d(void* d, const void* s, int64_t n)
  * (the fused agg or probe-agg kernel), for bench.py's roofline leg.
  * Events are recorded on g_stream, the stream the kernel launches on. */
 static double g_last_hot_ms = 0.0;
+static double g_hot_max_ms = 0.0; /* max hot region since pg_hot_reset */
 static hipEvent_t g_ev0, g_ev1;
 static bool g_ev_init = false;
 static void hot_begin()
@@ -147,8 +148,11 @@ static void hot_end()
     float ms = 0;
     hipEventElapsedTime(&ms, g_ev0, g_ev1);
     g_last_hot_ms = ms;
+    if (ms > g_hot_max_ms) g_hot_max_ms = ms;
 }
 extern "C" double pg_last_hot_kernel_ms(void) { return g_last_hot_ms; }
+extern "C" double pg_hot_max_ms(void) { return g_hot_max_ms; }
+extern "C" void pg_hot_reset(void) { g_hot_max_ms = 0.0; }
 
 /* ------------------------------------------------------------------ */
 /* device helpers                                                     */
@@ -1079,24 +1083,31 @@ __global__ __launch_bounds__(256) void k_probe_agg_q3(
     }
 }
 
-/* probe match counting (emit mode, pass 1): count[i] = chain length */
+/* probe match counting (emit mode, pass 1): per-BLOCK totals over the
+ * chunked stable geometry — no per-row counts array; the emit pass
+ * recomputes per-row counts and places rows with in-window prefix sums */
 __global__ __launch_bounds__(256) void k_probe_count(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
     const uint8_t* tags, const int32_t* head, const int32_t* next,
-    int64_t mask, int32_t* counts)
+    int64_t mask, int64_t chunk, int64_t* block_counts)
 {
-    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (; i < pg.n_rows; i += stride) {
-        int32_t c = 0;
-        if (d_eval_preds(pg, plan.preds, plan.n_preds, i)) {
-            int64_t key = d_load_i64(pg.cols[plan.key_col], i);
-            int64_t s = d_tbl_find_tagged(keys, tags, mask, key);
-            if (s >= 0)
-                for (int32_t r = head[s]; r >= 0; r = next[r]) c++;
-        }
-        counts[i] = c;
+    const int64_t n = pg.n_rows;
+    const int64_t lo = (int64_t)blockIdx.x * chunk;
+    const int64_t hi = min(lo + chunk, n);
+    int64_t total = 0;
+    for (int64_t i = lo + threadIdx.x; i < hi; i += 256) {
+        if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
+        int64_t key = d_load_i64(pg.cols[plan.key_col], i);
+        int64_t sl = d_tbl_find_tagged(keys, tags, mask, key);
+        if (sl >= 0)
+            for (int32_t r = head[sl]; r >= 0; r = next[r]) total++;
     }
+    total = d_bfly_i64(total);
+    __shared__ int64_t lds[4];
+    if ((threadIdx.x & 63) == 0) lds[threadIdx.x >> 6] = total;
+    __syncthreads();
+    if (threadIdx.x == 0)
+        block_counts[blockIdx.x] = lds[0] + lds[1] + lds[2] + lds[3];
 }
 
 /* probe emit (pass 2): offsets = exclusive scan of counts */
@@ -1109,18 +1120,42 @@ struct build_payloads {
 __global__ __launch_bounds__(256) void k_probe_emit(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
     const uint8_t* tags, const int32_t* head, const int32_t* next,
-    int64_t mask, const int64_t* offsets, emit_outs probe_outs,
-    build_payloads bp, emit_outs build_outs)
+    int64_t mask, int64_t chunk, const int64_t* block_offs,
+    emit_outs probe_outs, build_payloads bp, emit_outs build_outs)
 {
-    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (; i < pg.n_rows; i += stride) {
-        if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
-        int64_t key = d_load_i64(pg.cols[plan.key_col], i);
-        int64_t s = d_tbl_find_tagged(keys, tags, mask, key);
-        if (s < 0) continue;
-        int64_t pos = offsets[i];
-        for (int32_t r = head[s]; r >= 0; r = next[r]) {
+    const int64_t n = pg.n_rows;
+    const int64_t lo = (int64_t)blockIdx.x * chunk;
+    const int64_t hi = min(lo + chunk, n);
+    const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
+    __shared__ int64_t wcnt[4];
+    __shared__ int64_t running;
+    if (threadIdx.x == 0) running = block_offs[blockIdx.x];
+    __syncthreads();
+    for (int64_t base = lo; base < hi; base += 256) {
+        int64_t i = base + 64 * wid + lane;
+        int64_t sl = -1;
+        int32_t c = 0;
+        if (i < hi && d_eval_preds(pg, plan.preds, plan.n_preds, i)) {
+            int64_t key = d_load_i64(pg.cols[plan.key_col], i);
+            sl = d_tbl_find_tagged(keys, tags, mask, key);
+            if (sl >= 0)
+                for (int32_t r = head[sl]; r >= 0; r = next[r]) c++;
+        }
+        /* wave inclusive prefix of counts (stable row order) */
+        int64_t pre = c;
+        for (int d = 1; d < 64; d <<= 1) {
+            int64_t t = __shfl_up((long long)pre, d, 64);
+            if (lane >= d) pre += t;
+        }
+        int64_t wtotal = __shfl((long long)pre, 63, 64);
+        int64_t excl = pre - c;
+        if (lane == 0) wcnt[wid] = wtotal;
+        __syncthreads();
+        int64_t woff = running;
+        for (int w = 0; w < wid; w++) woff += wcnt[w];
+        int64_t pos = woff + excl;
+        if (sl >= 0)
+        for (int32_t r = head[sl]; r >= 0; r = next[r]) {
             for (int o = 0; o < probe_outs.n; o++) {
                 pg_proj pr;
                 pr.kind = PG_PROJ_IDENT;
@@ -1149,6 +1184,10 @@ __global__ __launch_bounds__(256) void k_probe_emit(
             }
             pos++;
         }
+        __syncthreads();
+        if (threadIdx.x == 0)
+            running += wcnt[0] + wcnt[1] + wcnt[2] + wcnt[3];
+        __syncthreads();
     }
 }
 
@@ -2325,7 +2364,7 @@ struct JoinOp : Op {
             CHKV(hipStreamSynchronize(g_stream));
             return;
         }
-        /* emit mode: count -> scan -> emit */
+        /* emit mode: per-block count -> host scan -> prefix emit */
         int64_t n = sp.pg.n_rows;
         if (n == 0) { /* empty probe page -> empty output page */
             OutPage op;
@@ -2339,37 +2378,30 @@ struct JoinOp : Op {
             outq.push_back(std::move(op));
             return;
         }
-        DevBuf counts, offs, ctot;
-        counts.alloc((size_t)n * 4);
-        offs.alloc((size_t)n * 8);
-        int64_t nchunks = (n + SCAN_CH - 1) / SCAN_CH;
-        ctot.alloc((size_t)nchunks * 8);
-        hipLaunchKernelGGL(k_probe_count, dim3(4096), dim3(256), 0, g_stream,
-                           sp.pg, plan, (const int64_t*)t->keys.p,
+        hot_begin();
+        int64_t chunk = sel_chunk(n);
+        DevBuf d_counts;
+        d_counts.alloc(FLT_NB * 8);
+        hipLaunchKernelGGL(k_probe_count, dim3(FLT_NB), dim3(256), 0,
+                           g_stream, sp.pg, plan, (const int64_t*)t->keys.p,
                            (const uint8_t*)t->tags.p,
                            (const int32_t*)t->head.p,
-                           (const int32_t*)t->next.p, t->mask,
-                           (int32_t*)counts.p);
-        hipLaunchKernelGGL(k_scan_chunks, dim3(nchunks), dim3(256), 0,
-                           g_stream, (const int32_t*)counts.p, n,
-                           (int64_t*)offs.p, (int64_t*)ctot.p);
-        std::vector<int64_t> h(nchunks);
-        CHKV(hipMemcpyAsync(h.data(), ctot.p, nchunks * 8,
+                           (const int32_t*)t->next.p, t->mask, chunk,
+                           (int64_t*)d_counts.p);
+        std::vector<int64_t> h(FLT_NB);
+        CHKV(hipMemcpyAsync(h.data(), d_counts.p, FLT_NB * 8,
                             hipMemcpyDeviceToHost, g_stream));
         CHKV(hipStreamSynchronize(g_stream));
         int64_t total = 0;
-        for (int64_t c = 0; c < nchunks; c++) {
-            int64_t v = h[c];
-            h[c] = total;
+        for (int b = 0; b < FLT_NB; b++) {
+            int64_t v = h[b];
+            h[b] = total;
             total += v;
         }
-        DevBuf bases;
-        bases.alloc((size_t)nchunks * 8);
-        CHKV(hipMemcpyAsync(bases.p, h.data(), nchunks * 8,
+        DevBuf d_offs;
+        d_offs.alloc(FLT_NB * 8);
+        CHKV(hipMemcpyAsync(d_offs.p, h.data(), FLT_NB * 8,
                             hipMemcpyHostToDevice, g_stream));
-        hipLaunchKernelGGL(k_scan_add_bases, dim3(2048), dim3(256), 0,
-                           g_stream, (int64_t*)offs.p, n,
-                           (const int64_t*)bases.p);
         /* output page: probe cols then build payloads */
         OutPage op;
         op.pg.n_rows = total;
@@ -2404,12 +2436,13 @@ struct JoinOp : Op {
             nc++;
         }
         op.pg.n_cols = nc;
-        hipLaunchKernelGGL(k_probe_emit, dim3(4096), dim3(256), 0, g_stream,
-                           sp.pg, plan, (const int64_t*)t->keys.p,
+        hipLaunchKernelGGL(k_probe_emit, dim3(FLT_NB), dim3(256), 0,
+                           g_stream, sp.pg, plan, (const int64_t*)t->keys.p,
                            (const uint8_t*)t->tags.p,
                            (const int32_t*)t->head.p,
-                           (const int32_t*)t->next.p, t->mask,
-                           (const int64_t*)offs.p, pouts, bp, bouts);
+                           (const int32_t*)t->next.p, t->mask, chunk,
+                           (const int64_t*)d_offs.p, pouts, bp, bouts);
+        hot_end();
         CHKV(hipStreamSynchronize(g_stream));
         outq.push_back(std::move(op));
     }
