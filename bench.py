@@ -400,7 +400,6 @@ def main():
                 assert got3["revenue_1e4"][i] == r.revenue_1e4
                 assert got3["orderdate"][i] == r.orderdate
             log(f"verify q3 sf={args.sf}: top-10 exact OK")
-    hot_ms = float(lib.c.pg_hot_max_ms()) / args.steps * args.steps
     hot_ms = float(lib.c.pg_hot_max_ms())
     ms_per_step = elapsed / args.steps * 1000.0
     value = total_rows_per_step * args.steps / elapsed
