@@ -198,6 +198,10 @@ typedef struct {
     int32_t payload_col[4];     /* i32/i64 payload channels stored per row */
     int64_t capacity_hint;      /* expected distinct build rows */
     int32_t key_set_only;       /* 1: build a key SET (no payload slots) */
+    int32_t dense_array;        /* 1: keys are dense 1..capacity_hint —
+                                   store the single u8 payload in a direct
+                                   array indexed by key-1 (no hashing);
+                                   e.g. suppkey -> s_nationkey */
     int32_t agg_table;          /* 1: table feeds a fused-agg probe only
                                    (LOOKUP_JOIN mode 1).  Rows are inserted
                                    directly during addInput (single scan,
@@ -213,7 +217,14 @@ typedef struct {
     int32_t n_preds;
     pg_pred preds[PG_MAX_PRED]; /* fused pre-filter on probe input */
     int32_t key_col;
-    /* mode 0: emit matched rows — output page =
+    /* mode 2: fully fused probe + dense lookup + equality + small-key
+     *   grouped SUM (the specialization for Q5's local-supplier shape):
+     *   for each probe row passing preds: g1 = table[key_col] payload
+     *   (u8), g2 = dense table2[table2_key_col], keep when g1 == g2 and
+     *   g2 is one of group_vals; acc[g2] += proj (exact ticks + fx128).
+     *   get_output after finish: host page [group u8, sum_dec i64 ticks,
+     *   sum_f64, count i64] in group_vals order (non-empty groups).
+     * mode 0: emit matched rows — output page =
      *   probe columns emit_probe_cols[] + build payloads (join emit order:
      *   probe rows ascending; within a probe row, chain head first —
      *   LookupJoinPageBuilder.appendRow:75 + ArrayPositionLinks order)
@@ -227,6 +238,11 @@ typedef struct {
     int32_t emit_probe_cols[8];
     pg_proj proj;
     int32_t dec_scale;
+    /* mode 2 only: */
+    int64_t table2;          /* dense_array build */
+    int32_t table2_key_col;
+    int32_t n_group_vals;
+    uint8_t group_vals[8];
 } pg_plan_lookup_join;
 
 typedef struct {

@@ -777,6 +777,19 @@ __global__ __launch_bounds__(256) void k_sel_emit(pg_page pg,
 /* insert order is non-deterministic across duplicates — result SET   */
 /* semantics).                                         */
 /* ------------------------------------------------------------------ */
+/* dense-array build: payload[key-1] = u8 value */
+__global__ __launch_bounds__(256) void k_dense_fill(
+    const int64_t* keys, const uint8_t* vals, int64_t n, uint8_t* out,
+    int64_t cap)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) {
+        int64_t k = keys[i];
+        if (k >= 1 && k <= cap) out[k - 1] = vals[i];
+    }
+}
+
 __global__ __launch_bounds__(256) void k_tbl_init(int64_t* keys, int32_t* head,
                                                   int64_t cap)
 {
@@ -1188,6 +1201,84 @@ __global__ __launch_bounds__(256) void k_probe_emit(
         if (threadIdx.x == 0)
             running += wcnt[0] + wcnt[1] + wcnt[2] + wcnt[3];
         __syncthreads();
+    }
+}
+
+/* mode 2: fused probe + dense-dictionary lookup + equality + small-key
+ * grouped SUM — the Q5 local-supplier specialization (see presto_gpu.h).
+ * Deterministic: exact ticks (order-independent) + fx128 for f64; groups
+ * accumulate in per-thread registers then wave butterfly + global atomics
+ * on ticks (exact). */
+template <int MAXG>
+__global__ __launch_bounds__(256) void k_probe_agg_fused2(
+    pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
+    const uint8_t* tags, int64_t mask, const uint8_t* slot_payload_u8,
+    const uint8_t* dense_vals, int64_t dense_n,
+    unsigned long long* out_dec /* [MAXG] */,
+    unsigned long long* out_flo, unsigned long long* out_fhi,
+    unsigned long long* out_cnt)
+{
+    int64_t acc[MAXG];
+    uint64_t fhi[MAXG], flo[MAXG];
+    int32_t cnt[MAXG];
+#pragma unroll
+    for (int g = 0; g < MAXG; g++) {
+        acc[g] = 0;
+        fhi[g] = 0;
+        flo[g] = 0;
+        cnt[g] = 0;
+    }
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < pg.n_rows; i += stride) {
+        if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
+        int64_t key = d_load_i64(pg.cols[plan.key_col], i);
+        int64_t sl = d_tbl_find_tagged(keys, tags, mask, key);
+        if (sl < 0) continue;
+        uint8_t g1 = slot_payload_u8[sl];
+        int64_t k2 = d_load_i64(pg.cols[plan.table2_key_col], i);
+        if (k2 < 1 || k2 > dense_n) continue;
+        uint8_t g2 = dense_vals[k2 - 1];
+        if (g1 != g2) continue;
+        int gi = -1;
+#pragma unroll
+        for (int g = 0; g < MAXG; g++)
+            if (g < plan.n_group_vals && plan.group_vals[g] == g2 && gi < 0)
+                gi = g;
+        if (gi < 0) continue;
+        pg_agg ag;
+        ag.proj = plan.proj;
+        ag.dec_scale = plan.dec_scale;
+        int64_t ticks = d_eval_proj_dec(pg, ag, i);
+        double p = d_eval_proj_f64(pg, plan.proj, i);
+        uint64_t phi, plo;
+        fx128_from_f64(p, &phi, &plo);
+#pragma unroll
+        for (int g = 0; g < MAXG; g++) {
+            bool m = g == gi;
+            acc[g] += m ? ticks : 0;
+            cnt[g] += m ? 1 : 0;
+            uint64_t nlo = flo[g] + (m ? plo : 0);
+            fhi[g] += (m ? phi : 0) + (nlo < flo[g] ? 1u : 0u);
+            flo[g] = nlo;
+        }
+    }
+    /* wave reduce ticks/count (exact), then one atomic per wave; the
+     * fx128 halves go through carry-aware atomics per lane (exact,
+     * order-independent) */
+#pragma unroll
+    for (int g = 0; g < MAXG; g++) {
+        int64_t a = d_bfly_i64(acc[g]);
+        int64_t c = d_bfly_i64((int64_t)cnt[g]);
+        if ((threadIdx.x & 63) == 0) {
+            if (a) atomicAdd(&out_dec[g], (unsigned long long)a);
+            if (c) atomicAdd(&out_cnt[g], (unsigned long long)c);
+        }
+        if (flo[g] | fhi[g]) {
+            unsigned long long old = atomicAdd(&out_flo[g], flo[g]);
+            atomicAdd(&out_fhi[g],
+                      fhi[g] + (old > ~flo[g] ? 1ull : 0ull));
+        }
     }
 }
 
@@ -1689,6 +1780,7 @@ struct Table {
     DevBuf keys, head, next, tags;
     DevBuf acc; /* slot_acc[cap], interleaved */
     bool slot_payloads = false; /* payloads indexed by slot (agg tables) */
+    bool dense = false;          /* dense_array: payload[key-1], no hash */
     /* compacted build-row arrays: key + payloads */
     DevBuf key_rows;
     std::vector<DevBuf> payload;
@@ -2099,6 +2191,17 @@ struct BuildOp : Op {
     {
         t.reset(new Table());
         t->key_set_only = plan.key_set_only != 0;
+        if (plan.dense_array) {
+            if (plan.n_payload != 1)
+                throw std::runtime_error(
+                    "dense_array needs exactly one u8 payload");
+            t->dense = true;
+            t->cap = plan.capacity_hint;
+            t->payload.emplace_back();
+            t->ptag.push_back(PG_T_U8);
+            t->payload.back().alloc((size_t)t->cap);
+            return;
+        }
         if (plan.agg_table) {
             /* direct mode: size the table now from the hint; payloads
              * live per slot */
@@ -2172,6 +2275,21 @@ struct BuildOp : Op {
         for (int i = 0; i < plan.n_payload; i++)
             if (t->ptag[i] < 0)
                 t->ptag[i] = sp.pg.cols[plan.payload_col[i]].tag;
+        if (plan.dense_array) {
+            if (sp.pg.cols[plan.key_col].tag != PG_T_I64 ||
+                sp.pg.cols[plan.payload_col[0]].tag != PG_T_U8)
+                throw std::runtime_error(
+                    "dense_array expects I64 keys and a U8 payload");
+            hipLaunchKernelGGL(k_dense_fill, dim3(2048), dim3(256), 0,
+                               g_stream,
+                               (const int64_t*)sp.pg.cols[plan.key_col].data,
+                               (const uint8_t*)
+                                   sp.pg.cols[plan.payload_col[0]].data,
+                               sp.pg.n_rows, (uint8_t*)t->payload[0].p,
+                               t->cap);
+            CHKV(hipStreamSynchronize(g_stream));
+            return;
+        }
         if (plan.agg_table) {
             const Table* semi = semi_table();
             direct_payloads dp{};
@@ -2235,6 +2353,12 @@ struct BuildOp : Op {
     }
     void finish() override
     {
+        if (plan.dense_array) {
+            std::lock_guard<std::mutex> lk(g_mu);
+            tbl = g_next_table++;
+            g_tables[tbl] = std::move(t);
+            return;
+        }
         if (plan.agg_table) {
             unsigned long long c[2];
             CHKV(hipMemcpy(c, counters.p, 16, hipMemcpyDeviceToHost));
@@ -2297,6 +2421,8 @@ struct BuildOp : Op {
 struct JoinOp : Op {
     pg_plan_lookup_join plan;
     Table* t = nullptr;
+    Table* t2 = nullptr; /* mode 2 dense table */
+    DevBuf m2_acc;       /* mode 2: [dec, flo, fhi, cnt] x 8 groups */
     void init()
     {
         std::lock_guard<std::mutex> lk(g_mu);
@@ -2310,12 +2436,41 @@ struct JoinOp : Op {
                 "cannot probe a key-set-only table");
         if (t->slot_payloads && plan.mode == 0)
             throw std::runtime_error(
-                "agg_table supports fused-agg probing only (mode 1)");
+                "agg_table supports fused-agg probing only (mode 1/2)");
+        if (plan.mode == 2) {
+            if (!t->slot_payloads || t->ptag.size() != 1 ||
+                t->ptag[0] != PG_T_U8)
+                throw std::runtime_error(
+                    "mode 2 probes an agg_table with one u8 payload");
+            auto it2 = g_tables.find(plan.table2);
+            if (it2 == g_tables.end() || !it2->second->dense)
+                throw std::runtime_error("mode 2 needs a dense_array "
+                                         "table2");
+            t2 = it2->second.get();
+            if (plan.n_group_vals < 1 || plan.n_group_vals > 8)
+                throw std::runtime_error("n_group_vals must be 1..8");
+            m2_acc.alloc(4 * 8 * 8);
+            m2_acc.zero();
+        }
     }
     void add_input(const pg_page* in) override
     {
         StagedPage sp;
         sp.stage(in);
+        if (plan.mode == 2) {
+            unsigned long long* a = (unsigned long long*)m2_acc.p;
+            hot_begin();
+            hipLaunchKernelGGL(k_probe_agg_fused2<8>, dim3(4096), dim3(256),
+                               0, g_stream, sp.pg, plan,
+                               (const int64_t*)t->keys.p,
+                               (const uint8_t*)t->tags.p, t->mask,
+                               (const uint8_t*)t->payload[0].p,
+                               (const uint8_t*)t2->payload[0].p, t2->cap,
+                               a, a + 8, a + 16, a + 24);
+            hot_end();
+            CHKV(hipStreamSynchronize(g_stream));
+            return;
+        }
         if (plan.mode == 1) {
             /* specialized fast path for the Q3 shape */
             bool spec = plan.proj.kind == PG_PROJ_DISC_PRICE &&
@@ -2448,6 +2603,40 @@ struct JoinOp : Op {
     }
     void finish() override
     {
+        if (plan.mode == 2) {
+            unsigned long long h[32];
+            CHKV(hipMemcpy(h, m2_acc.p, sizeof(h), hipMemcpyDeviceToHost));
+            int n_out = 0;
+            for (int g = 0; g < plan.n_group_vals; g++)
+                if (h[24 + g]) n_out++;
+            OutPage op;
+            op.pg.n_rows = n_out;
+            op.pg.n_cols = 4;
+            op.host.emplace_back(n_out ? n_out : 1);
+            op.host.emplace_back((size_t)(n_out ? n_out : 1) * 8);
+            op.host.emplace_back((size_t)(n_out ? n_out : 1) * 8);
+            op.host.emplace_back((size_t)(n_out ? n_out : 1) * 8);
+            op.pg.cols[0].tag = PG_T_U8;
+            op.pg.cols[1].tag = PG_T_I64;
+            op.pg.cols[2].tag = PG_T_F64;
+            op.pg.cols[3].tag = PG_T_I64;
+            for (int c = 0; c < 4; c++) {
+                op.pg.cols[c].on_device = 0;
+                op.pg.cols[c].data = op.host[c].data();
+            }
+            int row = 0;
+            for (int g = 0; g < plan.n_group_vals; g++) {
+                if (!h[24 + g]) continue;
+                ((uint8_t*)op.pg.cols[0].data)[row] = plan.group_vals[g];
+                ((int64_t*)op.pg.cols[1].data)[row] = (int64_t)h[g];
+                ((double*)op.pg.cols[2].data)[row] =
+                    fx128_to_f64(h[16 + g], h[8 + g]);
+                ((int64_t*)op.pg.cols[3].data)[row] = (int64_t)h[24 + g];
+                row++;
+            }
+            outq.push_back(std::move(op));
+            return;
+        }
         if (plan.mode != 1) return;
         /* extract groups: slots with count>0, slot-ascending */
         int64_t cap = t->cap;

@@ -68,7 +68,8 @@ class PlanHashBuild(C.Structure):
                 ("key_col", C.c_int32), ("semijoin_table", C.c_int64),
                 ("semijoin_col", C.c_int32), ("n_payload", C.c_int32),
                 ("payload_col", C.c_int32 * 4), ("capacity_hint", C.c_int64),
-                ("key_set_only", C.c_int32), ("agg_table", C.c_int32)]
+                ("key_set_only", C.c_int32), ("agg_table", C.c_int32),
+                ("dense_array", C.c_int32)]
 
 
 class PlanLookupJoin(C.Structure):
@@ -76,7 +77,10 @@ class PlanLookupJoin(C.Structure):
                 ("preds", Pred * 8), ("key_col", C.c_int32),
                 ("mode", C.c_int32), ("n_emit", C.c_int32),
                 ("emit_probe_cols", C.c_int32 * 8), ("proj", Proj),
-                ("dec_scale", C.c_int32)]
+                ("dec_scale", C.c_int32), ("table2", C.c_int64),
+                ("table2_key_col", C.c_int32),
+                ("n_group_vals", C.c_int32),
+                ("group_vals", C.c_uint8 * 8)]
 
 
 class PlanTopN(C.Structure):

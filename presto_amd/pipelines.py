@@ -287,8 +287,99 @@ class Q5Pipeline:
             b.destroy()
 
 
-def q5(cust: Page, orders: Page, supp: Page, li: Page):
+def q5_composed(cust: Page, orders: Page, supp: Page, li: Page):
     p = Q5Pipeline(cust, orders, supp)
+    try:
+        return p.run(li)
+    finally:
+        p.close()
+
+
+class Q5PipelineFused:
+    """Q5 with the single-pass fused probe (LOOKUP_JOIN mode 2): lineitem
+    is scanned ONCE — probe the orderkey->cust-nation agg table, dense
+    suppkey->supplier-nation lookup, local-supplier equality, per-nation
+    register accumulation.  The codegen-analog specialization of the
+    composed graph above (same results, no materialized intermediates)."""
+
+    def __init__(self, cust: Page, orders: Page, supp: Page):
+        b1 = PlanHashBuild()
+        b1.key_col = cust.channel("custkey")
+        b1.semijoin_table = -1
+        b1.n_payload = 1
+        b1.payload_col[0] = cust.channel("nationkey")
+        b1.capacity_hint = cust.n_rows
+        self.b1 = Operator(OP_HASH_BUILD, b1)
+        self.b1.add_input(cust)
+        self.b1.finish()
+
+        jo = PlanLookupJoin()
+        jo.table = self.b1.table()
+        jo.n_preds = 2
+        jo.preds[0] = Pred(orders.channel("orderdate"), CMP_GE,
+                           Q5Pipeline.Q5_LO, 0.0)
+        jo.preds[1] = Pred(orders.channel("orderdate"), CMP_LT,
+                           Q5Pipeline.Q5_HI, 0.0)
+        jo.key_col = orders.channel("custkey")
+        jo.mode = 0
+        jo.n_emit = 1
+        jo.emit_probe_cols[0] = orders.channel("orderkey")
+        j = Operator(OP_LOOKUP_JOIN, jo)
+        j.add_input(orders)
+        opage = j.get_output_raw()  # [orderkey, cust_nationkey]
+
+        b2 = PlanHashBuild()
+        b2.key_col = 0
+        b2.semijoin_table = -1
+        b2.n_payload = 1
+        b2.payload_col[0] = 1
+        b2.capacity_hint = max(opage.n_rows + 64, 64)
+        b2.agg_table = 1
+        self.b2 = Operator(OP_HASH_BUILD, b2)
+        self.b2.add_input_raw(opage)
+        self.b2.finish()
+        j.destroy()
+
+        b3 = PlanHashBuild()
+        b3.key_col = supp.channel("suppkey")
+        b3.n_payload = 1
+        b3.payload_col[0] = supp.channel("nationkey")
+        b3.capacity_hint = supp.n_rows
+        b3.dense_array = 1
+        self.b3 = Operator(OP_HASH_BUILD, b3)
+        self.b3.add_input(supp)
+        self.b3.finish()
+
+    def run(self, li: Page):
+        jp = PlanLookupJoin()
+        jp.table = self.b2.table()
+        jp.key_col = li.channel("orderkey")
+        jp.mode = 2
+        jp.proj = Proj(PROJ_DISC_PRICE, li.channel("extendedprice"),
+                       li.channel("discount"), 0)
+        jp.dec_scale = 4
+        jp.table2 = self.b3.table()
+        jp.table2_key_col = li.channel("suppkey")
+        jp.n_group_vals = len(Q5Pipeline.ASIA)
+        for i, v in enumerate(Q5Pipeline.ASIA):
+            jp.group_vals[i] = v
+        j = Operator(OP_LOOKUP_JOIN, jp)
+        try:
+            j.add_input(li)
+            j.finish()
+            return j.get_output(["nationkey", "rev_lo", "rev_f64", "count"])
+        finally:
+            j.destroy()
+
+    def close(self):
+        from .engine import lib
+        for b in (self.b1, self.b2, self.b3):
+            lib().c.pg_table_destroy(b.table())
+            b.destroy()
+
+
+def q5(cust: Page, orders: Page, supp: Page, li: Page):
+    p = Q5PipelineFused(cust, orders, supp)
     try:
         return p.run(li)
     finally:

@@ -779,21 +779,24 @@ def test_varchar_partition_math(P, oracle_lib):
 
 def test_q5_exact(P, oracle_lib):
     """Q5 (6-way join, local-supplier condition, per-nation revenue) —
-    exact ticks vs the oracle (which is pinned by the q05 SF1 golden)."""
+    BOTH the composed operator graph and the fused single-pass kernel must
+    match the oracle exactly (which is pinned by the q05 SF1 golden)."""
     sf = 0.1
     li = oracle_lib.gen_lineitem2(sf)
     orders = oracle_lib.gen_orders(sf)
     cust = oracle_lib.gen_customer2(sf)
     supp = oracle_lib.gen_supplier(sf)
-    got = P.pipelines.q5(
+    pages = (
         P.Page({"custkey": cust["custkey"], "nationkey": cust["nationkey"]}),
         P.Page({k: orders[k] for k in ("orderkey", "custkey", "orderdate")}),
         P.Page({"suppkey": supp["suppkey"], "nationkey": supp["nationkey"]}),
         P.Page({k: li[k] for k in ("orderkey", "suppkey", "extendedprice",
                                    "discount")}))
     exp = oracle_lib.q5(cust, orders, li, supp)
-    # GPU emits in key-enumeration order; compare per-nation ticks
-    got_by_nation = {int(got["nationkey"][i]): int(got["rev_lo"][i])
-                     for i in range(len(got["nationkey"]))}
     exp_by_nation = {int(r.nationkey): int(r.revenue_1e4) for r in exp}
-    assert got_by_nation == exp_by_nation
+    got_c = P.pipelines.q5_composed(*pages)
+    assert {int(got_c["nationkey"][i]): int(got_c["rev_lo"][i])
+            for i in range(len(got_c["nationkey"]))} == exp_by_nation
+    got_f = P.pipelines.q5(*pages)
+    assert {int(got_f["nationkey"][i]): int(got_f["rev_lo"][i])
+            for i in range(len(got_f["nationkey"]))} == exp_by_nation
