@@ -68,8 +68,8 @@ class PlanHashBuild(C.Structure):
                 ("key_col", C.c_int32), ("semijoin_table", C.c_int64),
                 ("semijoin_col", C.c_int32), ("n_payload", C.c_int32),
                 ("payload_col", C.c_int32 * 4), ("capacity_hint", C.c_int64),
-                ("key_set_only", C.c_int32), ("agg_table", C.c_int32),
-                ("dense_array", C.c_int32)]
+                ("key_set_only", C.c_int32), ("dense_array", C.c_int32),
+                ("agg_table", C.c_int32)]
 
 
 class PlanLookupJoin(C.Structure):
