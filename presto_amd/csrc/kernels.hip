@@ -899,6 +899,10 @@ __global__ __launch_bounds__(256) void k_tbl_insert_direct(
                 for (int o = 0; o < dp.n; o++) {
                     const pg_col& c = pg.cols[dp.src[o]];
                     switch (dp.tag[o]) {
+                        case PG_T_U8:
+                            ((uint8_t*)dp.ptr[o])[s] =
+                                (uint8_t)d_load_i64(c, i);
+                            break;
                         case PG_T_I32:
                             ((int32_t*)dp.ptr[o])[s] =
                                 (int32_t)d_load_i64(c, i);
@@ -1334,6 +1338,10 @@ __global__ __launch_bounds__(256) void k_groups_emit(
             int64_t r = bp.by_slot ? i : (int64_t)head[i];
             for (int o = 0; o < bp.n; o++) {
                 switch (bp.tag[o]) {
+                    case PG_T_U8:
+                        ((uint8_t*)payload_outs.ptr[o])[pos] =
+                            ((const uint8_t*)bp.ptr[o])[r];
+                        break;
                     case PG_T_I32:
                         ((int32_t*)payload_outs.ptr[o])[pos] =
                             ((const int32_t*)bp.ptr[o])[r];
