@@ -202,6 +202,12 @@ typedef struct {
                                    store the single u8 payload in a direct
                                    array indexed by key-1 (no hashing);
                                    e.g. suppkey -> s_nationkey */
+    /* agg_table only: fetch the u8 payload THROUGH another agg table
+     * (star-schema dimension join fused into the build: e.g. Q5's orders
+     * build stores customer nation = cust_table[o_custkey].payload; rows
+     * whose lookup key misses are dropped).  0 = unused. */
+    int64_t payload_lookup_table;
+    int32_t payload_lookup_key_col;
     int32_t agg_table;          /* 1: table feeds a fused-agg probe only
                                    (LOOKUP_JOIN mode 1).  Rows are inserted
                                    directly during addInput (single scan,

@@ -850,6 +850,10 @@ __global__ __launch_bounds__(256) void k_tbl_insert(const int64_t* in_keys,
     }
 }
 
+__device__ inline int64_t d_tbl_find_tagged(const int64_t* keys,
+                                            const uint8_t* tags,
+                                            int64_t mask, int64_t key);
+
 /* direct insert for agg tables: filter+semijoin+insert in ONE scan of the
  * build input; payloads stored per SLOT (keys unique).  count tracks
  * inserted rows; full table -> give-up flag (bounded probe loop). */
@@ -861,15 +865,26 @@ struct direct_payloads {
 };
 __global__ __launch_bounds__(256) void k_tbl_insert_direct(
     pg_page pg, pg_plan_hash_build plan, const int64_t* set_keys,
-    int64_t set_mask, int64_t* keys, uint8_t* tags, direct_payloads dp,
-    int64_t mask, unsigned long long* inserted,
-    unsigned long long* overflow)
+    int64_t set_mask, const int64_t* lu_keys, const uint8_t* lu_tags,
+    int64_t lu_mask, const uint8_t* lu_payload, int64_t* keys,
+    uint8_t* tags, direct_payloads dp, int64_t mask,
+    unsigned long long* inserted, unsigned long long* overflow)
 {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     int64_t my_inserted = 0, my_overflow = 0;
     for (; i < pg.n_rows; i += stride) {
         if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
+        uint8_t plv = 0;
+        if (lu_keys) {
+            /* fused dimension join: fetch u8 payload through the lookup
+             * table; rows that miss are dropped (inner-join semantics) */
+            int64_t k2 =
+                d_load_i64(pg.cols[plan.payload_lookup_key_col], i);
+            int64_t sl2 = d_tbl_find_tagged(lu_keys, lu_tags, lu_mask, k2);
+            if (sl2 < 0) continue;
+            plv = lu_payload[sl2];
+        }
         if (set_keys) {
             int64_t sk = d_load_i64(pg.cols[plan.semijoin_col], i);
             uint64_t h = pg_murmur3_finalize(pg_bigint_hash(sk));
@@ -896,7 +911,9 @@ __global__ __launch_bounds__(256) void k_tbl_insert_direct(
                     my_inserted++;
                     if (tags) tags[s] = d_tbl_tag(h);
                 }
-                for (int o = 0; o < dp.n; o++) {
+                if (lu_keys && dp.n >= 1)
+                    ((uint8_t*)dp.ptr[0])[s] = plv;
+                for (int o = lu_keys ? 1 : 0; o < dp.n; o++) {
                     const pg_col& c = pg.cols[dp.src[o]];
                     switch (dp.tag[o]) {
                         case PG_T_U8:
@@ -2300,18 +2317,36 @@ struct BuildOp : Op {
         }
         if (plan.agg_table) {
             const Table* semi = semi_table();
+            const Table* lu = nullptr;
+            if (plan.payload_lookup_table > 0) {
+                std::lock_guard<std::mutex> lk(g_mu);
+                auto it = g_tables.find(plan.payload_lookup_table);
+                if (it == g_tables.end() || !it->second->slot_payloads ||
+                    it->second->ptag.empty() ||
+                    it->second->ptag[0] != PG_T_U8)
+                    throw std::runtime_error(
+                        "payload_lookup_table must be an agg_table with a "
+                        "u8 payload");
+                lu = it->second.get();
+            }
             direct_payloads dp{};
             dp.n = plan.n_payload;
             for (int o = 0; o < dp.n; o++) {
                 dp.ptr[o] = t->payload[o].p;
-                dp.tag[o] = t->ptag[o];
+                dp.tag[o] = lu && o == 0 ? PG_T_U8 : t->ptag[o];
                 dp.src[o] = plan.payload_col[o];
             }
             hot_begin();
             hipLaunchKernelGGL(k_tbl_insert_direct, dim3(4096), dim3(256),
                                0, g_stream, sp.pg, plan,
                                semi ? (const int64_t*)semi->keys.p : nullptr,
-                               semi ? semi->mask : 0, (int64_t*)t->keys.p,
+                               semi ? semi->mask : 0,
+                               lu ? (const int64_t*)lu->keys.p : nullptr,
+                               lu ? (const uint8_t*)lu->tags.p : nullptr,
+                               lu ? lu->mask : 0,
+                               lu ? (const uint8_t*)lu->payload[0].p
+                                  : nullptr,
+                               (int64_t*)t->keys.p,
                                (uint8_t*)t->tags.p, dp, t->mask,
                                (unsigned long long*)counters.p,
                                (unsigned long long*)counters.p + 1);

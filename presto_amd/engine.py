@@ -69,6 +69,8 @@ class PlanHashBuild(C.Structure):
                 ("semijoin_col", C.c_int32), ("n_payload", C.c_int32),
                 ("payload_col", C.c_int32 * 4), ("capacity_hint", C.c_int64),
                 ("key_set_only", C.c_int32), ("dense_array", C.c_int32),
+                ("payload_lookup_table", C.c_int64),
+                ("payload_lookup_key_col", C.c_int32),
                 ("agg_table", C.c_int32)]
 
 

@@ -303,42 +303,37 @@ class Q5PipelineFused:
     composed graph above (same results, no materialized intermediates)."""
 
     def __init__(self, cust: Page, orders: Page, supp: Page):
+        # customer dimension as an agg table (slot payload = nationkey)
         b1 = PlanHashBuild()
         b1.key_col = cust.channel("custkey")
         b1.semijoin_table = -1
         b1.n_payload = 1
         b1.payload_col[0] = cust.channel("nationkey")
-        b1.capacity_hint = cust.n_rows
+        b1.capacity_hint = cust.n_rows + 64
+        b1.agg_table = 1
         self.b1 = Operator(OP_HASH_BUILD, b1)
         self.b1.add_input(cust)
         self.b1.finish()
 
-        jo = PlanLookupJoin()
-        jo.table = self.b1.table()
-        jo.n_preds = 2
-        jo.preds[0] = Pred(orders.channel("orderdate"), CMP_GE,
-                           Q5Pipeline.Q5_LO, 0.0)
-        jo.preds[1] = Pred(orders.channel("orderdate"), CMP_LT,
-                           Q5Pipeline.Q5_HI, 0.0)
-        jo.key_col = orders.channel("custkey")
-        jo.mode = 0
-        jo.n_emit = 1
-        jo.emit_probe_cols[0] = orders.channel("orderkey")
-        j = Operator(OP_LOOKUP_JOIN, jo)
-        j.add_input(orders)
-        opage = j.get_output_raw()  # [orderkey, cust_nationkey]
-
+        # orders build fused with the customer dimension join: one scan of
+        # orders, payload = cust_table[o_custkey].nationkey
         b2 = PlanHashBuild()
-        b2.key_col = 0
+        b2.n_preds = 2
+        b2.preds[0] = Pred(orders.channel("orderdate"), CMP_GE,
+                           Q5Pipeline.Q5_LO, 0.0)
+        b2.preds[1] = Pred(orders.channel("orderdate"), CMP_LT,
+                           Q5Pipeline.Q5_HI, 0.0)
+        b2.key_col = orders.channel("orderkey")
         b2.semijoin_table = -1
         b2.n_payload = 1
-        b2.payload_col[0] = 1
-        b2.capacity_hint = max(opage.n_rows + 64, 64)
+        b2.payload_col[0] = 0  # sourced through the lookup instead
+        b2.payload_lookup_table = self.b1.table()
+        b2.payload_lookup_key_col = orders.channel("custkey")
+        b2.capacity_hint = max(orders.n_rows // 4, 64)
         b2.agg_table = 1
         self.b2 = Operator(OP_HASH_BUILD, b2)
-        self.b2.add_input_raw(opage)
+        self.b2.add_input(orders)
         self.b2.finish()
-        j.destroy()
 
         b3 = PlanHashBuild()
         b3.key_col = supp.channel("suppkey")
