@@ -2296,10 +2296,14 @@ struct BuildOp : Op {
     {
         StagedPage sp;
         sp.stage(in);
-        /* resolve payload tags on first page */
+        /* resolve payload tags on first page (payload 0 is u8 when it
+         * is sourced through a fused dimension lookup) */
         for (int i = 0; i < plan.n_payload; i++)
             if (t->ptag[i] < 0)
-                t->ptag[i] = sp.pg.cols[plan.payload_col[i]].tag;
+                t->ptag[i] = (plan.agg_table &&
+                              plan.payload_lookup_table > 0 && i == 0)
+                                 ? PG_T_U8
+                                 : sp.pg.cols[plan.payload_col[i]].tag;
         if (plan.dense_array) {
             if (sp.pg.cols[plan.key_col].tag != PG_T_I64 ||
                 sp.pg.cols[plan.payload_col[0]].tag != PG_T_U8)
