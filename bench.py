@@ -174,6 +174,24 @@ def cpu_baseline_q1(orc, target_secs=10.0):
     return n / dt, n
 
 
+def cpu_baseline_q5(orc, target_secs=10.0):
+    sf_sample = 2.0
+    li = orc.gen_lineitem2(sf_sample)
+    orders = orc.gen_orders(sf_sample)
+    cust = orc.gen_customer2(sf_sample)
+    supp = orc.gen_supplier(sf_sample)
+    n = len(li["quantity"])
+    t0 = time.time()
+    orc.q5(cust, orders, li, supp)
+    dt = time.time() - t0
+    reps = min(max(1, int(target_secs / 2 / max(dt, 0.05))), 16)
+    t0 = time.time()
+    for _ in range(reps):
+        orc.q5(cust, orders, li, supp)
+    dt = (time.time() - t0) / reps
+    return n / dt, n
+
+
 def cpu_baseline_q3(orc, target_secs=10.0):
     sf_sample = 2.0
     li = orc.gen_lineitem(sf_sample)
@@ -439,7 +457,8 @@ def main():
             pass
         cpu = None
         if not args.skip_cpu_baseline and n_gpus == 1:
-            fn = cpu_baseline_q1 if args.query == "q1" else cpu_baseline_q3
+            fn = {"q1": cpu_baseline_q1, "q3": cpu_baseline_q3,
+                  "q5": cpu_baseline_q5}[args.query]
             rps, nsamp = fn(orc)
             cpu = {
                 "value": rps,
