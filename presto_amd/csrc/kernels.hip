@@ -134,19 +134,23 @@ static hipEvent_t g_ev0, g_ev1;
 static bool g_ev_init = false;
 static void hot_begin()
 {
+    hipError_t e;
     if (!g_ev_init) {
-        hipEventCreate(&g_ev0);
-        hipEventCreate(&g_ev1);
+        e = hipEventCreate(&g_ev0);
+        e = hipEventCreate(&g_ev1);
         g_ev_init = true;
     }
-    hipEventRecord(g_ev0, g_stream);
+    e = hipEventRecord(g_ev0, g_stream);
+    (void)e;
 }
 static void hot_end()
 {
-    hipEventRecord(g_ev1, g_stream);
-    hipEventSynchronize(g_ev1);
+    hipError_t e;
+    e = hipEventRecord(g_ev1, g_stream);
+    e = hipEventSynchronize(g_ev1);
     float ms = 0;
-    hipEventElapsedTime(&ms, g_ev0, g_ev1);
+    e = hipEventElapsedTime(&ms, g_ev0, g_ev1);
+    (void)e;
     g_last_hot_ms = ms;
     if (ms > g_hot_max_ms) g_hot_max_ms = ms;
 }
@@ -349,7 +353,7 @@ __global__ __launch_bounds__(FT_NTHREADS) void k_agg_small(
             int g = idx0 * n1 + idx1;
 #pragma unroll
             for (int a = 0; a < NA; a++) {
-                if (a >= plan.n_aggs) break;
+                if (a >= plan.n_aggs) continue;
                 T val;
                 if (DEC) {
                     int64_t t;
@@ -450,7 +454,7 @@ __global__ __launch_bounds__(FT_NTHREADS) void k_agg_q1(
 
     const int64_t v = (int64_t)blockIdx.x * FT_NTHREADS + threadIdx.x;
     unsigned long long local_bad = 0;
-#pragma unroll(VAR == 2 ? 2 : 1)
+
     for (int64_t base = 2 * v; base < n; base += 2 * FT_VL) {
         const bool pair = base + 1 < n;
         /* one vector load per column per pair */
@@ -1719,7 +1723,10 @@ struct DevBuf {
     void zero() { CHKV(hipMemsetAsync(p, 0, sz, g_stream)); }
     void free()
     {
-        if (p && !g_pool.put(sz, p)) (void)hipFree(p);
+        if (p && !g_pool.put(sz, p)) {
+            hipError_t e = hipFree(p);
+            (void)e;
+        }
         p = nullptr;
         sz = 0;
     }
