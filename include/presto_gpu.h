@@ -120,6 +120,7 @@ typedef enum {
     PG_PROJ_IDENT = 0,        /* column a */
     PG_PROJ_DISC_PRICE = 1,   /* a * (1 - b)            (Q1/Q3 revenue) */
     PG_PROJ_CHARGE = 2,       /* a * (1 - b) * (1 + c)  (Q1 charge) */
+    PG_PROJ_MUL = 3,          /* a * b                  (Q6 revenue) */
 } pg_proj_kind;
 
 typedef struct {
@@ -175,7 +176,7 @@ typedef struct {
      * restricted to non-empty groups */
     int32_t n_preds;
     pg_pred preds[PG_MAX_PRED]; /* fused pre-filter (scan+filter+agg) */
-    int32_t n_keys;             /* 1 or 2 */
+    int32_t n_keys;             /* 0 (single global group), 1 or 2 */
     int32_t key_col[2];
     int32_t n_vals[2];
     uint8_t key_vals[2][PG_MAX_KEYVALS];

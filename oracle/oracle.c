@@ -308,6 +308,27 @@ int64_t oracle_join_bigint(int64_t n_build, const int64_t* bkeys,
     return out;
 }
 
+/* ---------------- Q6 ---------------- */
+
+void oracle_q6(int64_t n, const double* qty, const double* ep,
+               const double* disc, const int32_t* sd, int64_t* revenue_1e4,
+               int64_t* count)
+{
+    int64_t rev = 0, cnt = 0;
+#pragma omp parallel for schedule(static) reduction(+ : rev, cnt)
+    for (int64_t i = 0; i < n; i++) {
+        if (sd[i] < 8766 || sd[i] >= 9131) continue;
+        if (!(disc[i] >= 0.05 && disc[i] <= 0.07)) continue;
+        if (!(qty[i] < 24.0)) continue;
+        int64_t cents = (int64_t)(ep[i] * 100.0 + 0.5);
+        int64_t d = (int64_t)(disc[i] * 100.0 + 0.5);
+        rev += cents * d;
+        cnt++;
+    }
+    *revenue_1e4 = rev;
+    *count = cnt;
+}
+
 /* ---------------- Q5 ---------------- */
 
 #define Q5_DATE_LO 8766 /* 1994-01-01 */

@@ -114,6 +114,16 @@ int32_t oracle_q5(int64_t n_cust, const int64_t* c_custkey,
                   int64_t n_supp, const uint8_t* s_nationkey,
                   q5_row_t* out /* capacity 25 */);
 
+/* ---------------- TPC-H Q6 ----------------
+ * SQL: q06.sql — scalar aggregate: sum(extendedprice*discount) over
+ * shipdate in [1994-01-01, 1995-01-01), discount in [0.05, 0.07] (f64
+ * compare; the generated hundredths are exactly representable), quantity
+ * < 24.  Returns exact 1e-4 ticks (= cents * hundredths) and the count. */
+void oracle_q6(int64_t n, const double* quantity,
+               const double* extendedprice, const double* discount,
+               const int32_t* shipdate, int64_t* revenue_1e4,
+               int64_t* count);
+
 /* ---------------- operator-level primitives (parity targets) ---------- */
 
 /* murmur3 finalizer bucket — PagesHash.java:236-252 /

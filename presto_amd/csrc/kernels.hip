@@ -256,6 +256,7 @@ __device__ inline double d_eval_proj_f64(const pg_page& pg, const pg_proj& p,
     double a = d_load_f64(pg.cols[p.a], i);
     if (p.kind == PG_PROJ_IDENT) return a;
     double b = d_load_f64(pg.cols[p.b], i);
+    if (p.kind == PG_PROJ_MUL) return a * b;
     double v = a * (1.0 - b);
     if (p.kind == PG_PROJ_DISC_PRICE) return v;
     double c = d_load_f64(pg.cols[p.c], i);
@@ -276,6 +277,7 @@ __device__ inline int64_t d_eval_proj_dec(const pg_page& pg, const pg_agg& ag,
     }
     int64_t cents = (int64_t)(d_load_f64(pg.cols[p.a], i) * 100.0 + 0.5);
     int64_t d = (int64_t)(d_load_f64(pg.cols[p.b], i) * 100.0 + 0.5);
+    if (p.kind == PG_PROJ_MUL) return cents * d; /* scale 4 ticks */
     int64_t v = cents * (100 - d);
     if (p.kind == PG_PROJ_DISC_PRICE) return v;
     int64_t t = (int64_t)(d_load_f64(pg.cols[p.c], i) * 100.0 + 0.5);
@@ -327,9 +329,13 @@ __global__ __launch_bounds__(FT_NTHREADS) void k_agg_small(
         int64_t lim = base + 2 < n ? base + 2 : n;
         for (int64_t i = base; i < lim; i++) {
             if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
-            /* group id: enumerated u8 key values */
+            /* group id: enumerated u8 key values (n_keys==0: one group) */
             int idx0 = -1, idx1 = 0, n1 = 1;
-            uint8_t k0 = ((const uint8_t*)pg.cols[plan.key_col[0]].data)[i];
+            uint8_t k0 = plan.n_keys
+                             ? ((const uint8_t*)
+                                    pg.cols[plan.key_col[0]].data)[i]
+                             : 0;
+            if (plan.n_keys == 0) idx0 = 0;
 #pragma unroll
             for (int j = 0; j < PG_MAX_KEYVALS; j++)
                 if (j < plan.n_vals[0] && k0 == plan.key_vals[0][j] &&
@@ -1973,7 +1979,10 @@ struct AggSmallOp : Op {
                 throw std::runtime_error(
                     "mixed decimal/f64 aggregates in one op unsupported");
         }
-        n_groups = plan.n_vals[0] * (plan.n_keys == 2 ? plan.n_vals[1] : 1);
+        n_groups = plan.n_keys == 0
+                       ? 1
+                       : plan.n_vals[0] *
+                             (plan.n_keys == 2 ? plan.n_vals[1] : 1);
         static const int NAS[] = {2, 4, 7, 9};
         static const int MGS[] = {2, 4, 6, 8};
         for (int x : NAS)
@@ -2169,7 +2178,7 @@ struct AggSmallOp : Op {
             op.pg.cols[nc].null_mask = nullptr;
             return nc++;
         };
-        int c_k0 = add_host_col(PG_T_U8);
+        int c_k0 = plan.n_keys >= 1 ? add_host_col(PG_T_U8) : -1;
         int c_k1 = plan.n_keys == 2 ? add_host_col(PG_T_U8) : -1;
         std::vector<int> agg_col(user_aggs);
         std::vector<int> agg_col2(user_aggs, -1);
@@ -2189,7 +2198,8 @@ struct AggSmallOp : Op {
             if (cnt_of(g) <= 0) continue;
             int i0 = plan.n_keys == 2 ? g / plan.n_vals[1] : g;
             int i1 = plan.n_keys == 2 ? g % plan.n_vals[1] : 0;
-            ((uint8_t*)op.pg.cols[c_k0].data)[row] = plan.key_vals[0][i0];
+            if (c_k0 >= 0)
+                ((uint8_t*)op.pg.cols[c_k0].data)[row] = plan.key_vals[0][i0];
             if (c_k1 >= 0)
                 ((uint8_t*)op.pg.cols[c_k1].data)[row] =
                     plan.key_vals[1][i1];

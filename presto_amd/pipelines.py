@@ -379,3 +379,28 @@ def q5(cust: Page, orders: Page, supp: Page, li: Page):
         return p.run(li)
     finally:
         p.close()
+
+
+def q6(li: Page):
+    """Q6 scalar aggregate (q06.sql): keyless HASH_AGG_SMALL with the
+    BETWEEN predicates fused; exact decimal ticks (scale 4) + count."""
+    p = PlanHashAggSmall()
+    p.n_preds = 5
+    p.preds[0] = Pred(li.channel("shipdate"), CMP_GE, 8766, 0.0)
+    p.preds[1] = Pred(li.channel("shipdate"), CMP_LT, 9131, 0.0)
+    p.preds[2] = Pred(li.channel("discount"), CMP_GE, 0, 0.05)
+    p.preds[3] = Pred(li.channel("discount"), CMP_LE, 0, 0.07)
+    p.preds[4] = Pred(li.channel("quantity"), CMP_LT, 0, 24.0)
+    p.n_keys = 0
+    p.n_aggs = 2
+    p.aggs[0] = Agg(AGG_SUM_DEC,
+                    Proj(PROJ_MUL, li.channel("extendedprice"),
+                         li.channel("discount"), 0), 4)
+    p.aggs[1] = Agg(AGG_COUNT, Proj(PROJ_IDENT, 0, 0, 0), 0)
+    op = Operator(OP_HASH_AGG_SMALL, p)
+    try:
+        op.add_input(li)
+        op.finish()
+        return op.get_output(["rev_hi", "rev_lo", "count"])
+    finally:
+        op.destroy()

@@ -207,3 +207,15 @@ OracleLib.gen_lineitem2 = gen_lineitem2
 OracleLib.gen_customer2 = gen_customer2
 OracleLib.gen_supplier = gen_supplier
 OracleLib.q5 = q5
+
+
+def q6(self, li):
+    rev = C.c_int64()
+    cnt = C.c_int64()
+    self.lib.oracle_q6(C.c_int64(len(li["quantity"])), _p(li["quantity"]),
+                       _p(li["extendedprice"]), _p(li["discount"]),
+                       _p(li["shipdate"]), C.byref(rev), C.byref(cnt))
+    return rev.value, cnt.value
+
+
+OracleLib.q6 = q6

@@ -800,3 +800,14 @@ def test_q5_exact(P, oracle_lib):
     got_f = P.pipelines.q5(*pages)
     assert {int(got_f["nationkey"][i]): int(got_f["rev_lo"][i])
             for i in range(len(got_f["nationkey"]))} == exp_by_nation
+
+
+def test_q6_exact(P, oracle_lib, sf01):
+    """Q6 scalar aggregate (keyless path, f64 BETWEEN predicates, a*b
+    decimal projection) — exact vs the golden-pinned oracle."""
+    li = sf01["li"]
+    got = P.pipelines.q6(_li_page(P, li))
+    rev, cnt = oracle_lib.q6(li)
+    assert len(got["rev_lo"]) == 1
+    assert int(got["rev_lo"][0]) == rev
+    assert int(got["count"][0]) == cnt

@@ -287,3 +287,9 @@ def test_q5_sf1_golden(oracle_lib):
     for r, g in zip(rows, golden):
         assert r.name.decode() == g[0]
         assert Decimal(r.revenue_1e4) / 10**4 == Decimal(g[1])
+
+
+def test_q6_sf1_golden(oracle_lib, sf1):
+    rev, cnt = oracle_lib.q6(sf1["li"])
+    golden = _parse_golden("q06_sf1.result")
+    assert Decimal(rev) / 10**4 == Decimal(golden[0][0])
