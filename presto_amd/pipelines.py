@@ -16,7 +16,7 @@ from .engine import (
     Operator, Page, PlanHashAggSmall, PlanHashBuild, PlanLookupJoin,
     PlanTopN, PlanPartition, Pred, Proj, Agg,
     CMP_LE, CMP_LT, CMP_GT, CMP_GE, CMP_EQ,
-    PROJ_IDENT, PROJ_DISC_PRICE, PROJ_CHARGE,
+    PROJ_IDENT, PROJ_DISC_PRICE, PROJ_CHARGE, PROJ_MUL,
     AGG_COUNT, AGG_SUM_F64, AGG_SUM_DEC,
     OP_HASH_AGG_SMALL, OP_HASH_BUILD, OP_LOOKUP_JOIN, OP_TOPN, OP_PARTITION,
 )
