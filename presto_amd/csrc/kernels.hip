@@ -1394,52 +1394,6 @@ __global__ __launch_bounds__(256) void k_groups_emit(
 }
 
 /* ------------------------------------------------------------------ */
-/* exclusive scan over int32 counts -> int64 offsets (two-level)      */
-/* ------------------------------------------------------------------ */
-#define SCAN_CH 16384 /* elements per block: 256 threads x 64 */
-
-__global__ __launch_bounds__(256) void k_scan_chunks(const int32_t* counts,
-                                                     int64_t n,
-                                                     int64_t* offsets,
-                                                     int64_t* chunk_totals)
-{
-    __shared__ int64_t tsum[256];
-    const int64_t lo = (int64_t)blockIdx.x * SCAN_CH;
-    const int t = threadIdx.x;
-    int64_t s = 0;
-    for (int j = 0; j < 64; j++) {
-        int64_t i = lo + (int64_t)t * 64 + j;
-        if (i < n) s += counts[i];
-    }
-    tsum[t] = s;
-    __syncthreads();
-    /* Hillis-Steele inclusive scan over 256 thread sums */
-    for (int d = 1; d < 256; d <<= 1) {
-        int64_t v = t >= d ? tsum[t - d] : 0;
-        __syncthreads();
-        tsum[t] += v;
-        __syncthreads();
-    }
-    int64_t base = t > 0 ? tsum[t - 1] : 0;
-    for (int j = 0; j < 64; j++) {
-        int64_t i = lo + (int64_t)t * 64 + j;
-        if (i < n) {
-            offsets[i] = base;
-            base += counts[i];
-        }
-    }
-    if (t == 255) chunk_totals[blockIdx.x] = tsum[255];
-}
-
-__global__ __launch_bounds__(256) void k_scan_add_bases(
-    int64_t* offsets, int64_t n, const int64_t* chunk_bases)
-{
-    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (; i < n; i += stride) offsets[i] += chunk_bases[i / SCAN_CH];
-}
-
-/* ------------------------------------------------------------------ */
 /* TopN — TopNOperator.java:32,90-111 / InMemoryGroupedTopNBuilder:   */
 /* ORDER BY val DESC, date ASC, key ASC LIMIT L.  Per-thread local    */
 /* top-L, then a block tournament in LDS; block winners merged on the */
