@@ -270,7 +270,11 @@ __device__ inline int64_t d_eval_proj_dec(const pg_page& pg, const pg_agg& ag,
 {
     const pg_proj& p = ag.proj;
     if (p.kind == PG_PROJ_IDENT) {
-        double a = d_load_f64(pg.cols[p.a], i);
+        const pg_col& c = pg.cols[p.a];
+        if (c.tag != PG_T_F64 && ag.dec_scale == 0)
+            return d_load_i64(c, i); /* integer column: exact, no f64 trip
+                                        (LongSumAggregation semantics) */
+        double a = d_load_f64(c, i);
         double s = 1.0;
         for (int k = 0; k < ag.dec_scale; k++) s *= 10.0;
         return (int64_t)(a * s + 0.5);
