@@ -890,12 +890,18 @@ __global__ __launch_bounds__(256) void k_tbl_insert_direct(
     for (; i < pg.n_rows; i += stride) {
         if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
         uint8_t plv = 0;
-        if (lu_keys) {
+        if (lu_payload) {
             /* fused dimension join: fetch u8 payload through the lookup
-             * table; rows that miss are dropped (inner-join semantics) */
+             * table (hash or dense); rows that miss are dropped
+             * (inner-join semantics) */
             int64_t k2 =
                 d_load_i64(pg.cols[plan.payload_lookup_key_col], i);
-            int64_t sl2 = d_tbl_find_tagged(lu_keys, lu_tags, lu_mask, k2);
+            int64_t sl2;
+            if (lu_keys) {
+                sl2 = d_tbl_find_tagged(lu_keys, lu_tags, lu_mask, k2);
+            } else { /* dense: payload[key-1], lu_mask = capacity */
+                sl2 = (k2 >= 1 && k2 <= lu_mask) ? k2 - 1 : -1;
+            }
             if (sl2 < 0) continue;
             plv = lu_payload[sl2];
         }
@@ -925,9 +931,9 @@ __global__ __launch_bounds__(256) void k_tbl_insert_direct(
                     my_inserted++;
                     if (tags) tags[s] = d_tbl_tag(h);
                 }
-                if (lu_keys && dp.n >= 1)
+                if (lu_payload && dp.n >= 1)
                     ((uint8_t*)dp.ptr[0])[s] = plv;
-                for (int o = lu_keys ? 1 : 0; o < dp.n; o++) {
+                for (int o = lu_payload ? 1 : 0; o < dp.n; o++) {
                     const pg_col& c = pg.cols[dp.src[o]];
                     switch (dp.tag[o]) {
                         case PG_T_U8:
@@ -2214,8 +2220,9 @@ struct BuildOp : Op {
                 t->tags.alloc((size_t)cap);
                 t->tags.zero();
             }
-            t->acc.alloc((size_t)cap * sizeof(slot_acc));
-            t->acc.zero();
+            /* acc is allocated lazily by the first mode-1 probe (mode-2
+             * and emit probes never touch it — at SF300 the slot_acc
+             * array is 8 GB of alloc+memset otherwise) */
             counters.alloc(16);
             counters.zero();
             hipLaunchKernelGGL(k_tbl_init, dim3(1024), dim3(256), 0,
@@ -2300,12 +2307,13 @@ struct BuildOp : Op {
             if (plan.payload_lookup_table > 0) {
                 std::lock_guard<std::mutex> lk(g_mu);
                 auto it = g_tables.find(plan.payload_lookup_table);
-                if (it == g_tables.end() || !it->second->slot_payloads ||
+                if (it == g_tables.end() ||
+                    !(it->second->slot_payloads || it->second->dense) ||
                     it->second->ptag.empty() ||
                     it->second->ptag[0] != PG_T_U8)
                     throw std::runtime_error(
-                        "payload_lookup_table must be an agg_table with a "
-                        "u8 payload");
+                        "payload_lookup_table must be an agg_table or "
+                        "dense_array with a u8 payload");
                 lu = it->second.get();
             }
             direct_payloads dp{};
@@ -2320,9 +2328,13 @@ struct BuildOp : Op {
                                0, g_stream, sp.pg, plan,
                                semi ? (const int64_t*)semi->keys.p : nullptr,
                                semi ? semi->mask : 0,
-                               lu ? (const int64_t*)lu->keys.p : nullptr,
-                               lu ? (const uint8_t*)lu->tags.p : nullptr,
-                               lu ? lu->mask : 0,
+                               lu && !lu->dense
+                                   ? (const int64_t*)lu->keys.p
+                                   : nullptr,
+                               lu && !lu->dense
+                                   ? (const uint8_t*)lu->tags.p
+                                   : nullptr,
+                               lu ? (lu->dense ? lu->cap : lu->mask) : 0,
                                lu ? (const uint8_t*)lu->payload[0].p
                                   : nullptr,
                                (int64_t*)t->keys.p,
@@ -2411,8 +2423,6 @@ struct BuildOp : Op {
                 t->tags.alloc((size_t)cap);
                 t->tags.zero();
             }
-            t->acc.alloc((size_t)cap * sizeof(slot_acc));
-            t->acc.zero();
         }
         hipLaunchKernelGGL(k_tbl_init, dim3(1024), dim3(256), 0, g_stream,
                            (int64_t*)t->keys.p,
@@ -2459,6 +2469,11 @@ struct JoinOp : Op {
         if (t->slot_payloads && plan.mode == 0)
             throw std::runtime_error(
                 "agg_table supports fused-agg probing only (mode 1/2)");
+        if (plan.mode == 1 && !t->acc.p) {
+            t->acc.alloc((size_t)t->cap * sizeof(slot_acc));
+            t->acc.zero();
+            CHKV(hipStreamSynchronize(g_stream));
+        }
         if (plan.mode == 2) {
             if (!t->slot_payloads || t->ptag.size() != 1 ||
                 t->ptag[0] != PG_T_U8)

@@ -309,8 +309,8 @@ class Q5PipelineFused:
         b1.semijoin_table = -1
         b1.n_payload = 1
         b1.payload_col[0] = cust.channel("nationkey")
-        b1.capacity_hint = cust.n_rows + 64
-        b1.agg_table = 1
+        b1.capacity_hint = cust.n_rows  # custkeys dense 1..n
+        b1.dense_array = 1
         self.b1 = Operator(OP_HASH_BUILD, b1)
         self.b1.add_input(cust)
         self.b1.finish()
