@@ -811,3 +811,40 @@ def test_q6_exact(P, oracle_lib, sf01):
     assert len(got["rev_lo"]) == 1
     assert int(got["rev_lo"][0]) == rev
     assert int(got["count"][0]) == cnt
+
+
+def test_multi_page_build_grow(P, oracle_lib):
+    """Join build fed in many pages (exercises the build-row growth /
+    realloc path), then emit-probe — set-equal to the oracle."""
+    rng = np.random.default_rng(90)
+    bkeys = rng.integers(0, 5000, 40_000).astype(np.int64)
+    pay = (bkeys * 11).astype(np.int64)
+    bplan = P.PlanHashBuild()
+    bplan.key_col = 0
+    bplan.semijoin_table = -1
+    bplan.n_payload = 1
+    bplan.payload_col[0] = 1
+    bplan.capacity_hint = 64  # tiny: forces repeated growth
+    b = P.Operator(P.OP_HASH_BUILD, bplan)
+    for a in range(0, 40_000, 3_000):
+        z = min(a + 3_000, 40_000)
+        b.add_input(P.Page({"k": bkeys[a:z], "p": pay[a:z]}))
+    b.finish()
+    pkeys = rng.integers(0, 6000, 30_000).astype(np.int64)
+    jplan = P.PlanLookupJoin()
+    jplan.table = b.table()
+    jplan.key_col = 0
+    jplan.mode = 0
+    jplan.n_emit = 1
+    jplan.emit_probe_cols[0] = 1
+    j = P.Operator(P.OP_LOOKUP_JOIN, jplan)
+    j.add_input(P.Page({"k": pkeys, "pi": np.arange(30_000, dtype=np.int64)}))
+    out = j.get_output(["pi", "bp"])
+    j.destroy()
+    op_idx, ob_idx = oracle_lib.join(bkeys, pkeys)
+    got = sorted(zip(out["pi"].tolist(), out["bp"].tolist()))
+    exp = sorted(zip(op_idx.tolist(), (bkeys[ob_idx] * 11).tolist()))
+    assert got == exp
+    from presto_amd.engine import lib as _l
+    _l().c.pg_table_destroy(jplan.table)
+    b.destroy()
