@@ -308,6 +308,86 @@ int64_t oracle_join_bigint(int64_t n_build, const int64_t* bkeys,
     return out;
 }
 
+/* ---------------- Q7 ---------------- */
+
+int32_t oracle_q7(int64_t n_cust, const int64_t* ck, const uint8_t* cnat,
+                  int64_t n_ord, const int64_t* ook, const int64_t* ock,
+                  int64_t n_li, const int64_t* lok, const int64_t* lsk,
+                  const double* lep, const double* ldisc,
+                  const int32_t* lsd, int64_t n_supp, const uint8_t* snat,
+                  q7_row_t* out)
+{
+    int64_t max_ck = 0;
+    for (int64_t i = 0; i < n_cust; i++)
+        if (ck[i] > max_ck) max_ck = ck[i];
+    uint8_t* cn = (uint8_t*)calloc(max_ck + 1, 1);
+    for (int64_t i = 0; i < n_cust; i++) cn[ck[i]] = cnat[i];
+    /* orderkey -> cust nation (all orders; no date filter in Q7) */
+    int64_t cap = hash_capacity(n_ord < 2 ? 2 : n_ord);
+    int64_t* slot = (int64_t*)malloc(cap * sizeof(int64_t));
+    memset(slot, -1, cap * sizeof(int64_t));
+    for (int64_t i = 0; i < n_ord; i++) {
+        uint64_t h = pg_murmur3_finalize(pg_bigint_hash(ook[i]));
+        int64_t pos = (int64_t)(h & (cap - 1));
+        while (slot[pos] != -1) pos = (pos + 1) & (cap - 1);
+        slot[pos] = i;
+    }
+    /* rev[dir][year]: dir 0 = FR supp -> DE cust, 1 = DE -> FR */
+    int64_t rev[2][2];
+    memset(rev, 0, sizeof(rev));
+#pragma omp parallel
+    {
+        int64_t loc[2][2];
+        memset(loc, 0, sizeof(loc));
+#pragma omp for schedule(static)
+        for (int64_t i = 0; i < n_li; i++) {
+            if (lsd[i] < 9131 || lsd[i] > 9861) continue;
+            int64_t sk = lsk[i];
+            if (sk < 1 || sk > n_supp) continue;
+            uint8_t sn = snat[sk - 1];
+            if (sn != 6 && sn != 7) continue;
+            uint64_t h = pg_murmur3_finalize(pg_bigint_hash(lok[i]));
+            int64_t pos = (int64_t)(h & (cap - 1));
+            int64_t r = -1;
+            for (;;) {
+                int64_t sI = slot[pos];
+                if (sI == -1) break;
+                if (ook[sI] == lok[i]) {
+                    r = sI;
+                    break;
+                }
+                pos = (pos + 1) & (cap - 1);
+            }
+            if (r == -1) continue;
+            uint8_t cnk = cn[ock[r]];
+            int dir;
+            if (sn == 6 && cnk == 7) dir = 0;
+            else if (sn == 7 && cnk == 6) dir = 1;
+            else continue;
+            int yr = lsd[i] <= 9495 ? 0 : 1;
+            int64_t cents = (int64_t)(lep[i] * 100.0 + 0.5);
+            int64_t d = (int64_t)(ldisc[i] * 100.0 + 0.5);
+            loc[dir][yr] += cents * (100 - d);
+        }
+#pragma omp critical
+        for (int a = 0; a < 2; a++)
+            for (int b = 0; b < 2; b++) rev[a][b] += loc[a][b];
+    }
+    free(cn);
+    free(slot);
+    int32_t n_out = 0;
+    for (int dir = 0; dir < 2; dir++)
+        for (int yr = 0; yr < 2; yr++) {
+            if (!rev[dir][yr]) continue;
+            out[n_out].supp_nation = dir == 0 ? 6 : 7;
+            out[n_out].cust_nation = dir == 0 ? 7 : 6;
+            out[n_out].year = 1995 + yr;
+            out[n_out].revenue_1e4 = rev[dir][yr];
+            n_out++;
+        }
+    return n_out;
+}
+
 /* ---------------- Q6 ---------------- */
 
 void oracle_q6(int64_t n, const double* qty, const double* ep,

@@ -114,6 +114,28 @@ int32_t oracle_q5(int64_t n_cust, const int64_t* c_custkey,
                   int64_t n_supp, const uint8_t* s_nationkey,
                   q5_row_t* out /* capacity 25 */);
 
+/* ---------------- TPC-H Q7 ----------------
+ * SQL: q07.sql — volume shipping between FRANCE(6) and GERMANY(7):
+ * s_nationkey/c_nationkey in {(6,7),(7,6)}, shipdate in
+ * [1995-01-01, 1996-12-31] (epoch [9131, 9861]), volume =
+ * extendedprice*(1-discount) grouped by (supp_nation, cust_nation,
+ * year).  rows out in (supp_nation, cust_nation, year) order.
+ * revenue in exact 1e-4 ticks. */
+typedef struct {
+    uint8_t supp_nation, cust_nation;
+    int32_t year;
+    int64_t revenue_1e4;
+} q7_row_t;
+
+int32_t oracle_q7(int64_t n_cust, const int64_t* c_custkey,
+                  const uint8_t* c_nationkey, int64_t n_ord,
+                  const int64_t* o_orderkey, const int64_t* o_custkey,
+                  int64_t n_li, const int64_t* l_orderkey,
+                  const int64_t* l_suppkey, const double* l_extendedprice,
+                  const double* l_discount, const int32_t* l_shipdate,
+                  int64_t n_supp, const uint8_t* s_nationkey,
+                  q7_row_t* out /* capacity 4 */);
+
 /* ---------------- TPC-H Q6 ----------------
  * SQL: q06.sql — scalar aggregate: sum(extendedprice*discount) over
  * shipdate in [1994-01-01, 1995-01-01), discount in [0.05, 0.07] (f64

@@ -404,3 +404,103 @@ def q6(li: Page):
         return op.get_output(["rev_hi", "rev_lo", "count"])
     finally:
         op.destroy()
+
+
+def q7(cust: Page, orders: Page, supp: Page, li: Page):
+    """Q7 volume shipping (q07.sql): FRANCE(6)<->GERMANY(7) pairs, shipdate
+    in [1995-01-01, 1996-12-31], volume grouped by (supp_nation,
+    cust_nation, year).  Composed: dense customer dimension, orders
+    agg-table with fused dimension lookup (no date filter), supplier hash
+    table, two emit joins, then four keyless aggregations (the OR over
+    nation pairs and the year split decompose into conjunctive plans).
+    Returns list of (supp_nation, cust_nation, year, revenue_1e4)."""
+    b1 = PlanHashBuild()
+    b1.key_col = cust.channel("custkey")
+    b1.semijoin_table = -1
+    b1.n_payload = 1
+    b1.payload_col[0] = cust.channel("nationkey")
+    b1.capacity_hint = cust.n_rows
+    b1.dense_array = 1
+    o1 = Operator(OP_HASH_BUILD, b1)
+    o1.add_input(cust)
+    o1.finish()
+
+    b2 = PlanHashBuild()
+    b2.key_col = orders.channel("orderkey")
+    b2.semijoin_table = -1
+    b2.n_payload = 1
+    b2.payload_col[0] = 0
+    b2.payload_lookup_table = o1.table()
+    b2.payload_lookup_key_col = orders.channel("custkey")
+    b2.capacity_hint = orders.n_rows + 64
+    b2.agg_table = 1
+    o2 = Operator(OP_HASH_BUILD, b2)
+    o2.add_input(orders)
+    o2.finish()
+
+    b3 = PlanHashBuild()
+    b3.key_col = supp.channel("suppkey")
+    b3.semijoin_table = -1
+    b3.n_payload = 1
+    b3.payload_col[0] = supp.channel("nationkey")
+    b3.capacity_hint = supp.n_rows
+    o3 = Operator(OP_HASH_BUILD, b3)
+    o3.add_input(supp)
+    o3.finish()
+
+    j1 = PlanLookupJoin()
+    j1.table = o2.table()
+    j1.n_preds = 2
+    j1.preds[0] = Pred(li.channel("shipdate"), CMP_GE, 9131, 0.0)
+    j1.preds[1] = Pred(li.channel("shipdate"), CMP_LE, 9861, 0.0)
+    j1.key_col = li.channel("orderkey")
+    j1.mode = 0
+    j1.n_emit = 3
+    j1.emit_probe_cols[0] = li.channel("suppkey")
+    j1.emit_probe_cols[1] = li.channel("extendedprice")
+    j1.emit_probe_cols[2] = li.channel("discount")
+    # NOTE: shipdate needed downstream -> emit it too
+    j1.n_emit = 4
+    j1.emit_probe_cols[3] = li.channel("shipdate")
+    ja = Operator(OP_LOOKUP_JOIN, j1)
+    ja.add_input(li)
+    pa = ja.get_output_raw()  # [suppkey, ep, dc, sdate, cnat]
+
+    j2 = PlanLookupJoin()
+    j2.table = o3.table()
+    j2.key_col = 0
+    j2.mode = 0
+    j2.n_emit = 4
+    for i, c in enumerate((1, 2, 3, 4)):
+        j2.emit_probe_cols[i] = c
+    jb = Operator(OP_LOOKUP_JOIN, j2)
+    jb.add_input_raw(pa)
+    pb = jb.get_output_raw()  # [ep, dc, sdate, cnat, snat]
+    ja.destroy()
+
+    out = []
+    for sn, cn in ((6, 7), (7, 6)):
+        for ylo, yhi, yr in ((9131, 9495, 1995), (9496, 9861, 1996)):
+            p = PlanHashAggSmall()
+            p.n_preds = 4
+            p.preds[0] = Pred(4, CMP_EQ, sn, 0.0)
+            p.preds[1] = Pred(3, CMP_EQ, cn, 0.0)
+            p.preds[2] = Pred(2, CMP_GE, ylo, 0.0)
+            p.preds[3] = Pred(2, CMP_LE, yhi, 0.0)
+            p.n_keys = 0
+            p.n_aggs = 2
+            p.aggs[0] = Agg(AGG_SUM_DEC, Proj(PROJ_DISC_PRICE, 0, 1, 0), 4)
+            p.aggs[1] = Agg(AGG_COUNT, Proj(PROJ_IDENT, 0, 0, 0), 0)
+            ao = Operator(OP_HASH_AGG_SMALL, p)
+            ao.add_input_raw(pb)
+            ao.finish()
+            res = ao.get_output(["hi", "lo", "cnt"])
+            ao.destroy()
+            if len(res["lo"]) and res["cnt"][0] > 0:
+                out.append((sn, cn, yr, int(res["lo"][0])))
+    jb.destroy()
+    from .engine import lib
+    for o in (o1, o2, o3):
+        lib().c.pg_table_destroy(o.table())
+        o.destroy()
+    return out

@@ -219,3 +219,25 @@ def q6(self, li):
 
 
 OracleLib.q6 = q6
+
+
+class Q7Row(C.Structure):
+    _fields_ = [("supp_nation", C.c_uint8), ("cust_nation", C.c_uint8),
+                ("year", C.c_int32), ("revenue_1e4", C.c_int64)]
+
+
+def q7(self, cust, orders, li, supp):
+    rows = (Q7Row * 4)()
+    self.lib.oracle_q7.restype = C.c_int32
+    nr = self.lib.oracle_q7(
+        C.c_int64(len(cust["custkey"])), _p(cust["custkey"]),
+        _p(cust["nationkey"]), C.c_int64(len(orders["orderkey"])),
+        _p(orders["orderkey"]), _p(orders["custkey"]),
+        C.c_int64(len(li["orderkey"])), _p(li["orderkey"]),
+        _p(li["suppkey"]), _p(li["extendedprice"]), _p(li["discount"]),
+        _p(li["shipdate"]), C.c_int64(len(supp["suppkey"])),
+        _p(supp["nationkey"]), rows)
+    return [rows[i] for i in range(nr)]
+
+
+OracleLib.q7 = q7

@@ -848,3 +848,23 @@ def test_multi_page_build_grow(P, oracle_lib):
     from presto_amd.engine import lib as _l
     _l().c.pg_table_destroy(jplan.table)
     b.destroy()
+
+
+def test_q7_exact(P, oracle_lib):
+    """Q7 (nation-pair volume by year) — composed pipeline vs the
+    golden-pinned oracle, exact ticks."""
+    sf = 0.1
+    li = oracle_lib.gen_lineitem2(sf)
+    orders = oracle_lib.gen_orders(sf)
+    cust = oracle_lib.gen_customer2(sf)
+    supp = oracle_lib.gen_supplier(sf)
+    got = P.pipelines.q7(
+        P.Page({"custkey": cust["custkey"], "nationkey": cust["nationkey"]}),
+        P.Page({k: orders[k] for k in ("orderkey", "custkey")}),
+        P.Page({"suppkey": supp["suppkey"], "nationkey": supp["nationkey"]}),
+        P.Page({k: li[k] for k in ("orderkey", "suppkey", "extendedprice",
+                                   "discount", "shipdate")}))
+    exp = oracle_lib.q7(cust, orders, li, supp)
+    exp_t = [(r.supp_nation, r.cust_nation, r.year, r.revenue_1e4)
+             for r in exp]
+    assert sorted(got) == sorted(exp_t)

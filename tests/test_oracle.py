@@ -293,3 +293,19 @@ def test_q6_sf1_golden(oracle_lib, sf1):
     rev, cnt = oracle_lib.q6(sf1["li"])
     golden = _parse_golden("q06_sf1.result")
     assert Decimal(rev) / 10**4 == Decimal(golden[0][0])
+
+
+def test_q7_sf1_golden(oracle_lib):
+    li = oracle_lib.gen_lineitem2(1.0)
+    orders = oracle_lib.gen_orders(1.0)
+    cust = oracle_lib.gen_customer2(1.0)
+    supp = oracle_lib.gen_supplier(1.0)
+    rows = oracle_lib.q7(cust, orders, li, supp)
+    golden = _parse_golden("q07_sf1.result")
+    assert len(rows) == len(golden) == 4
+    names = {6: "FRANCE", 7: "GERMANY"}
+    for r, g in zip(rows, golden):
+        assert names[r.supp_nation] == g[0]
+        assert names[r.cust_nation] == g[1]
+        assert r.year == int(g[2])
+        assert Decimal(r.revenue_1e4) / 10**4 == Decimal(g[3])
