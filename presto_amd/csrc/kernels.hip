@@ -1153,8 +1153,12 @@ __global__ __launch_bounds__(256) void k_probe_count(
         if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
         int64_t key = d_load_i64(pg.cols[plan.key_col], i);
         int64_t sl = d_tbl_find_tagged(keys, tags, mask, key);
-        if (sl >= 0)
-            for (int32_t r = head[sl]; r >= 0; r = next[r]) total++;
+        if (sl >= 0) {
+            if (head)
+                for (int32_t r = head[sl]; r >= 0; r = next[r]) total++;
+            else
+                total++; /* slot-payload table: unique keys */
+        }
     }
     total = d_bfly_i64(total);
     __shared__ int64_t lds[4];
@@ -1192,8 +1196,12 @@ __global__ __launch_bounds__(256) void k_probe_emit(
         if (i < hi && d_eval_preds(pg, plan.preds, plan.n_preds, i)) {
             int64_t key = d_load_i64(pg.cols[plan.key_col], i);
             sl = d_tbl_find_tagged(keys, tags, mask, key);
-            if (sl >= 0)
-                for (int32_t r = head[sl]; r >= 0; r = next[r]) c++;
+            if (sl >= 0) {
+                if (head)
+                    for (int32_t r = head[sl]; r >= 0; r = next[r]) c++;
+                else
+                    c = 1;
+            }
         }
         /* wave inclusive prefix of counts (stable row order) */
         int64_t pre = c;
@@ -1209,7 +1217,8 @@ __global__ __launch_bounds__(256) void k_probe_emit(
         for (int w = 0; w < wid; w++) woff += wcnt[w];
         int64_t pos = woff + excl;
         if (sl >= 0)
-        for (int32_t r = head[sl]; r >= 0; r = next[r]) {
+        for (int64_t r = head ? (int64_t)head[sl] : sl; r >= 0;
+             r = head ? (int64_t)next[r] : -1) {
             for (int o = 0; o < probe_outs.n; o++) {
                 pg_proj pr;
                 pr.kind = PG_PROJ_IDENT;
@@ -2466,9 +2475,8 @@ struct JoinOp : Op {
         if (t->key_set_only)
             throw std::runtime_error(
                 "cannot probe a key-set-only table");
-        if (t->slot_payloads && plan.mode == 0)
-            throw std::runtime_error(
-                "agg_table supports fused-agg probing only (mode 1/2)");
+        /* mode 0 emit over a slot-payload table: unique keys, payloads
+         * indexed by slot (no chains) */
         if (plan.mode == 1 && !t->acc.p) {
             t->acc.alloc((size_t)t->cap * sizeof(slot_acc));
             t->acc.zero();
@@ -2613,6 +2621,7 @@ struct JoinOp : Op {
         }
         build_payloads bp{};
         bp.n = (int32_t)t->payload.size();
+        bp.by_slot = t->slot_payloads ? 1 : 0;
         emit_outs bouts{};
         bouts.n = bp.n;
         for (int o = 0; o < bp.n; o++) {
