@@ -168,6 +168,10 @@ typedef struct {
     pg_pred preds[PG_MAX_PRED]; /* conjunction */
     int32_t n_proj;
     pg_proj proj[16];
+    int64_t semijoin_table; /* >0: keep only rows whose semijoin_col key is
+                               in that key-set table (EXISTS pushdown, e.g.
+                               Q4); 0 = unused */
+    int32_t semijoin_col;
 } pg_plan_filter_project;
 
 typedef struct {

@@ -308,6 +308,59 @@ int64_t oracle_join_bigint(int64_t n_build, const int64_t* bkeys,
     return out;
 }
 
+/* ---------------- Q4 ---------------- */
+
+void oracle_q4(int64_t n_ord, const int64_t* ook, const int32_t* od,
+               const uint8_t* opri, int64_t n_li, const int64_t* lok,
+               const int32_t* lcd, const int32_t* lrd, int64_t* out_counts)
+{
+    /* set of orderkeys having a late lineitem (commit < receipt) */
+    int64_t n_late = 0;
+    int64_t* late = (int64_t*)malloc(n_li * sizeof(int64_t));
+    for (int64_t i = 0; i < n_li; i++)
+        if (lcd[i] < lrd[i]) late[n_late++] = lok[i];
+    int64_t cap = hash_capacity(n_late < 2 ? 2 : n_late);
+    int64_t* slot = (int64_t*)malloc(cap * sizeof(int64_t));
+    memset(slot, -1, cap * sizeof(int64_t));
+    for (int64_t i = 0; i < n_late; i++) {
+        uint64_t h = pg_murmur3_finalize(pg_bigint_hash(late[i]));
+        int64_t pos = (int64_t)(h & (cap - 1));
+        for (;;) {
+            if (slot[pos] == -1) {
+                slot[pos] = late[i];
+                break;
+            }
+            if (slot[pos] == late[i]) break;
+            pos = (pos + 1) & (cap - 1);
+        }
+    }
+    for (int k = 0; k < 5; k++) out_counts[k] = 0;
+#pragma omp parallel
+    {
+        int64_t loc[5] = {0, 0, 0, 0, 0};
+#pragma omp for schedule(static)
+        for (int64_t i = 0; i < n_ord; i++) {
+            if (od[i] < 8582 || od[i] >= 8674) continue;
+            uint64_t h = pg_murmur3_finalize(pg_bigint_hash(ook[i]));
+            int64_t pos = (int64_t)(h & (cap - 1));
+            int found = 0;
+            for (;;) {
+                if (slot[pos] == -1) break;
+                if (slot[pos] == ook[i]) {
+                    found = 1;
+                    break;
+                }
+                pos = (pos + 1) & (cap - 1);
+            }
+            if (found) loc[opri[i]]++;
+        }
+#pragma omp critical
+        for (int k = 0; k < 5; k++) out_counts[k] += loc[k];
+    }
+    free(late);
+    free(slot);
+}
+
 /* ---------------- Q7 ---------------- */
 
 int32_t oracle_q7(int64_t n_cust, const int64_t* ck, const uint8_t* cnat,

@@ -114,6 +114,16 @@ int32_t oracle_q5(int64_t n_cust, const int64_t* c_custkey,
                   int64_t n_supp, const uint8_t* s_nationkey,
                   q5_row_t* out /* capacity 25 */);
 
+/* ---------------- TPC-H Q4 ----------------
+ * SQL: q04.sql — order priority checking: orders with orderdate in
+ * [1993-07-01, 1993-10-01) = [8582, 8674) having EXISTS a lineitem with
+ * commitdate < receiptdate; count per priority (0..4).  out_counts[5]. */
+void oracle_q4(int64_t n_ord, const int64_t* o_orderkey,
+               const int32_t* o_orderdate, const uint8_t* o_priority,
+               int64_t n_li, const int64_t* l_orderkey,
+               const int32_t* l_commitdate, const int32_t* l_receiptdate,
+               int64_t* out_counts);
+
 /* ---------------- TPC-H Q7 ----------------
  * SQL: q07.sql — volume shipping between FRANCE(6) and GERMANY(7):
  * s_nationkey/c_nationkey in {(6,7),(7,6)}, shipdate in

@@ -45,6 +45,7 @@ static inline int64_t unif(int64_t* s, int64_t lo, int64_t hi)
 #define SEED_C_NATION 1489529863LL /* customer nationkey, usage 1/customer */
 #define SEED_S_NATION  110356601LL /* supplier nationkey, usage 1/supplier */
 #define SEED_L_SUPPN  2095021727LL /* lineitem supplier number 0..3, 7/order */
+#define SEED_O_PRIO    591449447LL /* order priority pick 1..5, 1/order */
 
 /* ---- calendar ----
  * day index 1 = 1992-01-01; order-date index in [1, 2406]
@@ -240,6 +241,77 @@ static void gen_orders_chunk(double sf, int64_t start, int64_t count,
         if (orderdate_epoch) orderdate_epoch[i] = (int32_t)(EPOCH_1992 + od - 1);
         if (lcnt) lcnt[i] = (int32_t)lc;
     }
+}
+
+void tpch_gen_orders_priority(double sf, int64_t start, int64_t count,
+                              uint8_t* priority)
+{
+#pragma omp parallel
+    {
+#ifdef _OPENMP
+        extern int omp_get_num_threads(void), omp_get_thread_num(void);
+        int nt = omp_get_num_threads(), tid = omp_get_thread_num();
+#else
+        int nt = 1, tid = 0;
+#endif
+        (void)sf;
+        int64_t lo = count * tid / nt, hi = count * (tid + 1) / nt;
+        int64_t s = rng_skip(SEED_O_PRIO, (uint64_t)(start + lo));
+        for (int64_t i = lo; i < hi; i++)
+            priority[i] = (uint8_t)(unif(&s, 1, 5) - 1);
+    }
+}
+
+int64_t tpch_gen_lineitem_dates(double sf, int64_t ord_start,
+                                int64_t ord_count, int64_t* orderkey,
+                                int32_t* commitd, int32_t* receiptd)
+{
+    (void)sf;
+    int64_t written = 0;
+#pragma omp parallel reduction(+ : written)
+    {
+#ifdef _OPENMP
+        extern int omp_get_num_threads(void), omp_get_thread_num(void);
+        int nt = omp_get_num_threads(), tid = omp_get_thread_num();
+#else
+        int nt = 1, tid = 0;
+#endif
+        int64_t tlo = ord_count * tid / nt, thi = ord_count * (tid + 1) / nt;
+        int64_t s = rng_skip(SEED_O_LCNT, (uint64_t)ord_start);
+        int64_t off = 0;
+        for (int64_t i = 0; i < tlo; i++) off += unif(&s, 1, 7);
+        int64_t s_lcnt = rng_skip(SEED_O_LCNT, (uint64_t)(ord_start + tlo));
+        int64_t s_odate = rng_skip(SEED_O_ODATE, (uint64_t)(ord_start + tlo));
+        uint64_t l7 = (uint64_t)(ord_start + tlo) * 7;
+        int64_t s_sdays = rng_skip(SEED_L_SDAYS, l7);
+        int64_t s_cdays = rng_skip(SEED_L_CDAYS, l7);
+        int64_t s_rdays = rng_skip(SEED_L_RDAYS, l7);
+        int64_t out = off;
+        for (int64_t o = tlo; o < thi; o++) {
+            int64_t idx1 = ord_start + o + 1;
+            int64_t ok = make_orderkey(idx1);
+            int64_t od = unif(&s_odate, ODATE_MIN, ODATE_MAX);
+            int64_t lc = unif(&s_lcnt, 1, 7);
+            for (int64_t l = 0; l < lc; l++) {
+                int64_t sd = unif(&s_sdays, 1, 121);
+                int64_t cd = unif(&s_cdays, 30, 90);
+                int64_t rd = unif(&s_rdays, 1, 30);
+                if (orderkey) orderkey[out] = ok;
+                if (commitd)
+                    commitd[out] = (int32_t)(EPOCH_1992 + od + cd - 1);
+                if (receiptd)
+                    receiptd[out] =
+                        (int32_t)(EPOCH_1992 + od + sd + rd - 1);
+                out++;
+            }
+            uint64_t rest = (uint64_t)(7 - lc);
+            s_sdays = rng_skip(s_sdays, rest);
+            s_cdays = rng_skip(s_cdays, rest);
+            s_rdays = rng_skip(s_rdays, rest);
+            written += lc;
+        }
+    }
+    return written;
 }
 
 void tpch_gen_orders(double sf, int64_t start, int64_t count,

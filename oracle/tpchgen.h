@@ -65,6 +65,15 @@ int64_t tpch_gen_lineitem2(double sf, int64_t ord_start, int64_t ord_count,
                            uint8_t* returnflag, uint8_t* linestatus,
                            int64_t* suppkey);
 
+/* order priority (0..4 = 1-URGENT, 2-HIGH, 3-MEDIUM, 4-NOT SPECIFIED,
+ * 5-LOW) and lineitem commit/receipt dates for Q4 */
+void tpch_gen_orders_priority(double sf, int64_t start, int64_t count,
+                              uint8_t* priority);
+int64_t tpch_gen_lineitem_dates(double sf, int64_t ord_start,
+                                int64_t ord_count, int64_t* orderkey,
+                                int32_t* commitdate_epoch,
+                                int32_t* receiptdate_epoch);
+
 /* orders columns for order rows [start, start+count) (0-based).
  * shippriority is the constant 0 (dbgen mk_order) and is not emitted.
  * lcnt: lineitems per order (1..7). Any output pointer may be NULL. */

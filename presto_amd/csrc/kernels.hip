@@ -1890,8 +1890,16 @@ struct FilterOp : Op {
     {
         StagedPage sp;
         sp.stage(in);
+        const Table* semi = nullptr;
+        if (plan.semijoin_table > 0) {
+            std::lock_guard<std::mutex> lk(g_mu);
+            auto it = g_tables.find(plan.semijoin_table);
+            if (it == g_tables.end())
+                throw std::runtime_error("semijoin table not found");
+            semi = it->second.get();
+        }
         int64_t chunk = sel_chunk(sp.pg.n_rows);
-        SelResult r = sel_count(sp.pg, plan, nullptr, 0, chunk);
+        SelResult r = sel_count(sp.pg, plan, semi, plan.semijoin_col, chunk);
         OutPage op;
         op.pg.n_rows = r.n;
         op.pg.n_cols = plan.n_proj;
@@ -1914,7 +1922,7 @@ struct FilterOp : Op {
             outs.ptr[o] = op.dev.back().p;
             outs.tag[o] = tag;
         }
-        sel_emit(sp.pg, plan, nullptr, 0, chunk, r, outs);
+        sel_emit(sp.pg, plan, semi, plan.semijoin_col, chunk, r, outs);
         outq.push_back(std::move(op));
     }
     void finish() override {}

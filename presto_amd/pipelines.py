@@ -504,3 +504,56 @@ def q7(cust: Page, orders: Page, supp: Page, li: Page):
         lib().c.pg_table_destroy(o.table())
         o.destroy()
     return out
+
+
+def q4(orders: Page, li_dates: Page):
+    """Q4 order-priority checking (q04.sql): EXISTS(lineitem with
+    commitdate < receiptdate) as a key-set build with a col-vs-col
+    predicate, then orders filter (date range + semijoin) and a 5-value
+    priority COUNT aggregation.  Returns counts per priority 0..4."""
+    bs = PlanHashBuild()
+    bs.n_preds = 1
+    p = Pred(li_dates.channel("commitdate"), CMP_LT, 0, 0.0)
+    p.rhs_col = li_dates.channel("receiptdate") + 1
+    bs.preds[0] = p
+    bs.key_col = li_dates.channel("orderkey")
+    bs.semijoin_table = -1
+    bs.capacity_hint = max(li_dates.n_rows // 2, 16)
+    bs.key_set_only = 1
+    b = Operator(OP_HASH_BUILD, bs)
+    b.add_input(li_dates)
+    b.finish()
+
+    fp = PlanFilterProject()
+    fp.n_preds = 2
+    fp.preds[0] = Pred(orders.channel("orderdate"), CMP_GE, 8582, 0.0)
+    fp.preds[1] = Pred(orders.channel("orderdate"), CMP_LT, 8674, 0.0)
+    fp.n_proj = 1
+    fp.proj[0] = Proj(PROJ_IDENT, orders.channel("priority"), 0, 0)
+    fp.semijoin_table = b.table()
+    fp.semijoin_col = orders.channel("orderkey")
+    f = Operator(OP_FILTER_PROJECT, fp)
+    f.add_input(orders)
+    fpage = f.get_output_raw()
+
+    ag = PlanHashAggSmall()
+    ag.n_keys = 1
+    ag.key_col[0] = 0
+    ag.n_vals[0] = 5
+    for i in range(5):
+        ag.key_vals[0][i] = i
+    ag.n_aggs = 1
+    ag.aggs[0] = Agg(AGG_COUNT, Proj(PROJ_IDENT, 0, 0, 0), 0)
+    a = Operator(OP_HASH_AGG_SMALL, ag)
+    a.add_input_raw(fpage)
+    a.finish()
+    out = a.get_output(["priority", "count"])
+    a.destroy()
+    f.destroy()
+    from .engine import lib
+    lib().c.pg_table_destroy(bs.key_col and b.table() or b.table())
+    b.destroy()
+    counts = [0] * 5
+    for i in range(len(out["priority"])):
+        counts[int(out["priority"][i])] = int(out["count"][i])
+    return counts

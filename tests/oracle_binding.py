@@ -241,3 +241,40 @@ def q7(self, cust, orders, li, supp):
 
 
 OracleLib.q7 = q7
+
+
+def gen_orders_priority(self, sf):
+    n = self.lib.tpch_orders_count(C.c_double(sf))
+    pri = np.empty(n, np.uint8)
+    self.lib.tpch_gen_orders_priority(C.c_double(sf), C.c_int64(0),
+                                      C.c_int64(n), _p(pri))
+    return pri
+
+
+def gen_lineitem_dates(self, sf):
+    n = self.lineitem_count(sf)
+    n_ord = self.lib.tpch_orders_count(C.c_double(sf))
+    ok = np.empty(n, np.int64)
+    cd = np.empty(n, np.int32)
+    rd = np.empty(n, np.int32)
+    self.lib.tpch_gen_lineitem_dates.restype = C.c_int64
+    w = self.lib.tpch_gen_lineitem_dates(C.c_double(sf), C.c_int64(0),
+                                         C.c_int64(n_ord), _p(ok), _p(cd),
+                                         _p(rd))
+    assert w == n
+    return dict(orderkey=ok, commitdate=cd, receiptdate=rd)
+
+
+def q4(self, orders, pri, lid):
+    counts = (C.c_int64 * 5)()
+    self.lib.oracle_q4(
+        C.c_int64(len(orders["orderkey"])), _p(orders["orderkey"]),
+        _p(orders["orderdate"]), _p(pri),
+        C.c_int64(len(lid["orderkey"])), _p(lid["orderkey"]),
+        _p(lid["commitdate"]), _p(lid["receiptdate"]), counts)
+    return list(counts)
+
+
+OracleLib.gen_orders_priority = gen_orders_priority
+OracleLib.gen_lineitem_dates = gen_lineitem_dates
+OracleLib.q4 = q4

@@ -868,3 +868,19 @@ def test_q7_exact(P, oracle_lib):
     exp_t = [(r.supp_nation, r.cust_nation, r.year, r.revenue_1e4)
              for r in exp]
     assert sorted(got) == sorted(exp_t)
+
+
+def test_q4_exact(P, oracle_lib, sf01):
+    """Q4 (EXISTS semijoin with a col-vs-col predicate + priority counts)
+    vs the golden-pinned oracle."""
+    sf = 0.1
+    orders = sf01["orders"]
+    pri = oracle_lib.gen_orders_priority(sf)
+    lid = oracle_lib.gen_lineitem_dates(sf)
+    got = P.pipelines.q4(
+        P.Page({"orderkey": orders["orderkey"],
+                "orderdate": orders["orderdate"], "priority": pri}),
+        P.Page({k: lid[k] for k in ("orderkey", "commitdate",
+                                    "receiptdate")}))
+    exp = oracle_lib.q4(orders, pri, lid)
+    assert got == exp

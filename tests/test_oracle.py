@@ -309,3 +309,15 @@ def test_q7_sf1_golden(oracle_lib):
         assert names[r.cust_nation] == g[1]
         assert r.year == int(g[2])
         assert Decimal(r.revenue_1e4) / 10**4 == Decimal(g[3])
+
+
+def test_q4_sf1_golden(oracle_lib, sf1):
+    pri = oracle_lib.gen_orders_priority(1.0)
+    lid = oracle_lib.gen_lineitem_dates(1.0)
+    counts = oracle_lib.q4(sf1["orders"], pri, lid)
+    golden = _parse_golden("q04_sf1.result")
+    names = ["1-URGENT", "2-HIGH", "3-MEDIUM", "4-NOT SPECIFIED", "5-LOW"]
+    assert len(golden) == 5
+    for k, g in enumerate(golden):
+        assert names[k] == g[0]
+        assert counts[k] == int(g[1])
