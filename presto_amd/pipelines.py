@@ -551,7 +551,7 @@ def q4(orders: Page, li_dates: Page):
     a.destroy()
     f.destroy()
     from .engine import lib
-    lib().c.pg_table_destroy(bs.key_col and b.table() or b.table())
+    lib().c.pg_table_destroy(b.table())
     b.destroy()
     counts = [0] * 5
     for i in range(len(out["priority"])):
