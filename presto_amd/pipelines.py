@@ -13,8 +13,9 @@ Date constants: epoch-day literals from the benchmark SQL
 import ctypes as C
 
 from .engine import (
-    Operator, Page, PlanHashAggSmall, PlanHashBuild, PlanLookupJoin,
-    PlanTopN, PlanPartition, Pred, Proj, Agg,
+    Operator, Page, PlanFilterProject, PlanHashAggSmall, PlanHashBuild,
+    PlanLookupJoin, PlanTopN, PlanPartition, Pred, Proj, Agg,
+    OP_FILTER_PROJECT,
     CMP_LE, CMP_LT, CMP_GT, CMP_GE, CMP_EQ,
     PROJ_IDENT, PROJ_DISC_PRICE, PROJ_CHARGE, PROJ_MUL,
     AGG_COUNT, AGG_SUM_F64, AGG_SUM_DEC,
