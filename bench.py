@@ -316,46 +316,54 @@ def main():
             return pipelines.q3(cust_page, ord_page, li_page, mode="dec")
         total_rows_per_step = n_rows * n_gpus
         workload = Q3_WORKLOAD.format(sf=int(sf_total), n=n_gpus)
-    else:  # q5 (single-GPU round-1 path)
-        if world > 1:
-            raise SystemExit("q5 multi-GPU lands in round 2")
+    else:  # q5
         cols, n_rows = gen_lineitem_device(orc, sf_total, device,
                                            want_orderkey=True,
                                            want_suppkey=True,
                                            ord_start=o0, ord_count=o1 - o0)
         li_page = Page(cols, n_rows=n_rows)
-        n_cust = orc.lib.tpch_customer_count(C.c_double(sf_total))
-        cck = np.empty(n_cust, np.int64)
-        cnat = np.empty(n_cust, np.uint8)
-        orc.lib.tpch_gen_customer2(C.c_double(sf_total), C.c_int64(0),
-                                   C.c_int64(n_cust),
+        # rank shards of the dimension tables
+        n_cust_all = orc.lib.tpch_customer_count(C.c_double(sf_total))
+        c0 = n_cust_all * rank // world
+        c1 = n_cust_all * (rank + 1) // world
+        cck = np.empty(c1 - c0, np.int64)
+        cnat = np.empty(c1 - c0, np.uint8)
+        orc.lib.tpch_gen_customer2(C.c_double(sf_total), C.c_int64(c0),
+                                   C.c_int64(c1 - c0),
                                    C.c_void_p(cck.ctypes.data), None,
                                    C.c_void_p(cnat.ctypes.data))
-        n_supp = orc.lib.tpch_supplier_count(C.c_double(sf_total))
-        ssk = np.empty(n_supp, np.int64)
-        snat = np.empty(n_supp, np.uint8)
-        orc.lib.tpch_gen_supplier(C.c_double(sf_total), C.c_int64(0),
-                                  C.c_int64(n_supp),
+        n_supp_all = orc.lib.tpch_supplier_count(C.c_double(sf_total))
+        s0 = n_supp_all * rank // world
+        s1 = n_supp_all * (rank + 1) // world
+        ssk = np.empty(s1 - s0, np.int64)
+        snat = np.empty(s1 - s0, np.uint8)
+        orc.lib.tpch_gen_supplier(C.c_double(sf_total), C.c_int64(s0),
+                                  C.c_int64(s1 - s0),
                                   C.c_void_p(ssk.ctypes.data),
                                   C.c_void_p(snat.ctypes.data))
-        n_ordx = orc.lib.tpch_orders_count(C.c_double(sf_total))
-        ook = np.empty(n_ordx, np.int64)
-        ock = np.empty(n_ordx, np.int64)
-        ood = np.empty(n_ordx, np.int32)
-        orc.lib.tpch_gen_orders(C.c_double(sf_total), C.c_int64(0),
-                                C.c_int64(n_ordx),
+        ook = np.empty(o1 - o0, np.int64)
+        ock = np.empty(o1 - o0, np.int64)
+        ood = np.empty(o1 - o0, np.int32)
+        orc.lib.tpch_gen_orders(C.c_double(sf_total), C.c_int64(o0),
+                                C.c_int64(o1 - o0),
                                 C.c_void_p(ook.ctypes.data),
                                 C.c_void_p(ock.ctypes.data),
                                 C.c_void_p(ood.ctypes.data), None)
         t = lambda a: torch.from_numpy(a).to(device)
         cust_page = Page({"custkey": t(cck), "nationkey": t(cnat)})
+        cust_page.n_total = n_cust_all
         ord_page = Page({"orderkey": t(ook), "custkey": t(ock),
                          "orderdate": t(ood)})
         supp_page = Page({"suppkey": t(ssk), "nationkey": t(snat)})
+        supp_page.n_total = n_supp_all
         li5 = Page({k: cols[k] for k in ("orderkey", "suppkey",
                                          "extendedprice", "discount")})
 
         def step():
+            if world > 1:
+                from presto_amd.dist import q5_distributed
+                return q5_distributed(cust_page, ord_page, supp_page, li5,
+                                      world, rank, device)
             return pipelines.q5(cust_page, ord_page, supp_page, li5)
         total_rows_per_step = n_rows * n_gpus
         workload = Q5_WORKLOAD.format(sf=int(sf_total), n=n_gpus)

@@ -256,3 +256,161 @@ def q3_distributed(cust_page, ord_page, li_page, world, rank, device,
     if mode != "dec":
         res["rev"] = res["rev"].view(np.float64)
     return res
+
+
+def q5_distributed(cust_page, ord_page, supp_page, li_page, world, rank,
+                   device, sf_hint=None):
+    """Distributed Q5 (BASELINE config 4): the nation dimensions are
+    replicated (all_reduce of the dense custkey/suppkey -> nationkey
+    arrays — the broadcast-join analog), orders and lineitem are
+    repartitioned by orderkey hash over RCCL all_to_all, then each rank
+    runs the fused single-pass probe on its partition; the five per-nation
+    tick sums combine exactly with one small all_reduce.
+    Returns dict {nationkey: revenue_1e4 ticks} on rank 0 (None elsewhere);
+    world == 1 runs the same graph with collectives as identities."""
+    import numpy as np
+    import presto_amd as P
+    from .engine import lib
+    from .pipelines import Q5Pipeline
+
+    use_dist = world > 1
+
+    def global_dense(page, key_name, val_name, n_all):
+        """u8 dense array key->val across all ranks (keys are dense 1..n,
+        each rank holds a disjoint shard)."""
+        full = torch.zeros(int(n_all), dtype=torch.uint8, device=device)
+        k = page.cols[key_name]
+        v = page.cols[val_name]
+        full[k - 1] = v
+        if use_dist:
+            dist.all_reduce(full)  # disjoint shards -> sum == union
+        return full
+
+    # callers set .n_total (global table cardinality) on the dimension
+    # pages; default = local rows x world (exact when shards are even)
+    n_cust_all = getattr(cust_page, "n_total", cust_page.n_rows * world)
+    n_supp_all = getattr(supp_page, "n_total", supp_page.n_rows * world)
+    cnat = global_dense(cust_page, "custkey", "nationkey", n_cust_all)
+    snat = global_dense(supp_page, "suppkey", "nationkey", n_supp_all)
+
+    # dense tables from the replicated arrays
+    def dense_table(vals):
+        n = vals.numel()
+        keys = torch.arange(1, n + 1, dtype=torch.int64, device=device)
+        bp = P.PlanHashBuild()
+        bp.key_col = 0
+        bp.semijoin_table = -1
+        bp.n_payload = 1
+        bp.payload_col[0] = 1
+        bp.capacity_hint = n
+        bp.dense_array = 1
+        b = P.Operator(P.OP_HASH_BUILD, bp)
+        b.add_input(P.Page({"k": keys, "v": vals}))
+        b.finish()
+        return b
+    bc = dense_table(cnat)
+    bs = dense_table(snat)
+
+    # orders: filter (date) -> partition by orderkey -> exchange
+    fp = P.PlanFilterProject()
+    fp.n_preds = 2
+    fp.preds[0] = P.Pred(ord_page.channel("orderdate"), P.CMP_GE,
+                         Q5Pipeline.Q5_LO, 0.0)
+    fp.preds[1] = P.Pred(ord_page.channel("orderdate"), P.CMP_LT,
+                         Q5Pipeline.Q5_HI, 0.0)
+    fp.n_proj = 2
+    fp.proj[0] = P.Proj(P.PROJ_IDENT, ord_page.channel("orderkey"), 0, 0)
+    fp.proj[1] = P.Proj(P.PROJ_IDENT, ord_page.channel("custkey"), 0, 0)
+    fo = P.Operator(P.OP_FILTER_PROJECT, fp)
+    fo.add_input(ord_page)
+    fraw = fo.get_output_raw()
+
+    def partition_exchange_raw(raw, ncols):
+        pp = P.PlanPartition()
+        pp.n_partitions = world
+        pp.key_col = 0
+        pp.n_emit = ncols
+        for i in range(ncols):
+            pp.emit_cols[i] = i
+        po = P.Operator(P.OP_PARTITION, pp)
+        po.add_input_raw(raw)
+        counts = po.partition_counts(world)
+        pages = [po.get_output_raw() for _ in range(world)]
+        total = sum(counts)
+        dtmap = {0: torch.uint8, 1: torch.int32, 2: torch.int64,
+                 3: torch.float64}
+        cols = {}
+        for c in range(ncols):
+            t = torch.empty(total, dtype=dtmap[pages[0].cols[c].tag],
+                            device=device)
+            off = 0
+            for p in range(world):
+                nrow = pages[p].n_rows
+                if nrow:
+                    lib().check(lib().c.pg_memcpy_d2d(
+                        t.data_ptr() + off * t.element_size(),
+                        pages[p].cols[c].data,
+                        nrow * t.element_size()), "d2d")
+                off += nrow
+            cols[f"c{c}"] = t
+        po.destroy()
+        if use_dist:
+            return exchange_columns(cols, counts, device=device)
+        return cols
+
+    ocols = partition_exchange_raw(fraw, 2)
+    fo.destroy()
+
+    b2p = P.PlanHashBuild()
+    b2p.key_col = 0
+    b2p.semijoin_table = -1
+    b2p.n_payload = 1
+    b2p.payload_col[0] = 0
+    b2p.payload_lookup_table = bc.table()
+    b2p.payload_lookup_key_col = 1
+    b2p.capacity_hint = max(int(ocols["c0"].numel()) + 64, 64)
+    b2p.agg_table = 1
+    b2 = P.Operator(P.OP_HASH_BUILD, b2p)
+    b2.add_input(P.Page(ocols))
+    b2.finish()
+
+    # lineitem: partition (no filter in Q5) -> exchange -> fused probe
+    lcols_in = {"c0": li_page.cols["orderkey"],
+                "c1": li_page.cols["suppkey"],
+                "c2": li_page.cols["extendedprice"],
+                "c3": li_page.cols["discount"]}
+    lp = P.Page(lcols_in)
+    lraw = lp.to_c()
+    lcols = partition_exchange_raw(lraw, 4)
+
+    jp = P.PlanLookupJoin()
+    jp.table = b2.table()
+    jp.key_col = 0
+    jp.mode = 2
+    jp.proj = P.Proj(P.PROJ_DISC_PRICE, 2, 3, 0)
+    jp.dec_scale = 4
+    jp.table2 = bs.table()
+    jp.table2_key_col = 1
+    jp.n_group_vals = len(Q5Pipeline.ASIA)
+    for i, v in enumerate(Q5Pipeline.ASIA):
+        jp.group_vals[i] = v
+    j = P.Operator(P.OP_LOOKUP_JOIN, jp)
+    j.add_input(P.Page(lcols))
+    j.finish()
+    out = j.get_output(["nationkey", "rev_lo", "rev_f64", "count"])
+    j.destroy()
+    for b in (bc, bs, b2):
+        lib().c.pg_table_destroy(b.table())
+        b.destroy()
+    # exact final combine: per-nation tick sums + counts
+    ticks = torch.zeros(len(Q5Pipeline.ASIA), dtype=torch.int64,
+                        device=device)
+    for i in range(len(out["nationkey"])):
+        ticks[Q5Pipeline.ASIA.index(int(out["nationkey"][i]))] = \
+            int(out["rev_lo"][i])
+    if use_dist:
+        dist.all_reduce(ticks)
+    if rank != 0:
+        return None
+    t = ticks.cpu().numpy()
+    return {n: int(t[i]) for i, n in enumerate(Q5Pipeline.ASIA) if t[i]}

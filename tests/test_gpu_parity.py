@@ -884,3 +884,32 @@ def test_q4_exact(P, oracle_lib, sf01):
                                     "receiptdate")}))
     exp = oracle_lib.q4(orders, pri, lid)
     assert got == exp
+
+
+def test_q5_distributed_graph_world1(P, oracle_lib):
+    """The distributed Q5 graph at world==1 (replicated dimensions +
+    partition/exchange identities + fused probe + exact tick combine)
+    must match the oracle."""
+    import torch
+    from presto_amd.dist import q5_distributed
+    sf = 0.1
+    li = oracle_lib.gen_lineitem2(sf)
+    orders = oracle_lib.gen_orders(sf)
+    cust = oracle_lib.gen_customer2(sf)
+    supp = oracle_lib.gen_supplier(sf)
+    dev = torch.device("cuda", 0)
+    t = lambda a: torch.from_numpy(a).to(dev)
+    cp = P.Page({"custkey": t(cust["custkey"]),
+                 "nationkey": t(cust["nationkey"])})
+    cp.n_total = len(cust["custkey"])
+    sp = P.Page({"suppkey": t(supp["suppkey"]),
+                 "nationkey": t(supp["nationkey"])})
+    sp.n_total = len(supp["suppkey"])
+    op = P.Page({k: t(orders[k]) for k in ("orderkey", "custkey",
+                                           "orderdate")})
+    lp = P.Page({k: t(li[k]) for k in ("orderkey", "suppkey",
+                                       "extendedprice", "discount")})
+    got = q5_distributed(cp, op, sp, lp, world=1, rank=0, device=dev)
+    exp = {int(r.nationkey): int(r.revenue_1e4)
+           for r in oracle_lib.q5(cust, orders, li, supp)}
+    assert got == exp
