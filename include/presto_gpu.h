@@ -136,6 +136,9 @@ typedef enum {
     PG_AGG_SUM_DEC = 2,  /* exact decimal ticks; scale from dec_scale
                             (hive-decimal semantics of the golden vectors) */
     PG_AGG_SUM_I64 = 3,  /* LongSumAggregation.java:33-37 */
+    PG_AGG_MIN = 4,      /* MinAggregationFunction: decimal-ticks/i64 min
+                            in decimal mode, f64 min in f64 mode */
+    PG_AGG_MAX = 5,
 } pg_agg_func;
 
 typedef struct {

@@ -321,9 +321,17 @@ __global__ __launch_bounds__(FT_NTHREADS) void k_agg_small(
     using T = typename std::conditional<DEC, int64_t, double>::type;
     T acc[NA][MAXG];
 #pragma unroll
-    for (int a = 0; a < NA; a++)
+    for (int a = 0; a < NA; a++) {
+        T id = (T)0;
+        if (a < plan.n_aggs) {
+            if (plan.aggs[a].func == PG_AGG_MIN)
+                id = DEC ? (T)INT64_MAX : (T)INFINITY;
+            else if (plan.aggs[a].func == PG_AGG_MAX)
+                id = DEC ? (T)INT64_MIN : (T)-INFINITY;
+        }
 #pragma unroll
-        for (int g = 0; g < MAXG; g++) acc[a][g] = (T)0;
+        for (int g = 0; g < MAXG; g++) acc[a][g] = id;
+    }
 
     const int64_t n = pg.n_rows;
     const int64_t v = (int64_t)blockIdx.x * FT_NTHREADS + threadIdx.x;
@@ -380,56 +388,96 @@ __global__ __launch_bounds__(FT_NTHREADS) void k_agg_small(
                         t = d_eval_proj_f64(pg, plan.aggs[a].proj, i);
                     val = (T)t;
                 }
+                int32_t fn = plan.aggs[a].func;
 #pragma unroll
-                for (int gg = 0; gg < MAXG; gg++)
-                    acc[a][gg] += (gg == g) ? val : (T)0;
+                for (int gg = 0; gg < MAXG; gg++) {
+                    bool m = gg == g;
+                    if (fn == PG_AGG_MIN)
+                        acc[a][gg] =
+                            m && val < acc[a][gg] ? val : acc[a][gg];
+                    else if (fn == PG_AGG_MAX)
+                        acc[a][gg] =
+                            m && val > acc[a][gg] ? val : acc[a][gg];
+                    else
+                        acc[a][gg] += m ? val : (T)0;
+                }
             }
         }
     }
 
-    /* wave butterfly, then 4-wave combine in LDS, then block partial += */
+    /* wave butterfly, then 4-wave combine in LDS, then block partial
+     * merge (sum, or min/max per the aggregate) */
     __shared__ T lds[4][NA][MAXG];
     const int lane = threadIdx.x & (WAVE - 1);
     const int wid = threadIdx.x >> 6;
 #pragma unroll
-    for (int a = 0; a < NA; a++)
+    for (int a = 0; a < NA; a++) {
+        int32_t fn = a < plan.n_aggs ? plan.aggs[a].func : PG_AGG_COUNT;
 #pragma unroll
         for (int g = 0; g < MAXG; g++) {
-            T s;
-            if (DEC)
-                s = (T)d_bfly_i64((int64_t)acc[a][g]);
-            else
-                s = (T)d_bfly_f64((double)acc[a][g]);
+            T s = acc[a][g];
+#pragma unroll
+            for (int w = 32; w >= 1; w >>= 1) {
+                T o;
+                if (DEC)
+                    o = (T)__shfl_xor((int64_t)s, w, WAVE);
+                else
+                    o = (T)__shfl_xor((double)s, w, WAVE);
+                if (fn == PG_AGG_MIN)
+                    s = o < s ? o : s;
+                else if (fn == PG_AGG_MAX)
+                    s = o > s ? o : s;
+                else
+                    s = s + o;
+            }
             if (lane == 0) lds[wid][a][g] = s;
         }
+    }
     __syncthreads();
     if (wid == 0 && lane < 4) {
 #pragma unroll
-        for (int a = 0; a < NA; a++)
+        for (int a = 0; a < NA; a++) {
+            int32_t fn = a < plan.n_aggs ? plan.aggs[a].func : PG_AGG_COUNT;
 #pragma unroll
             for (int g = 0; g < MAXG; g++) {
                 T x = lds[lane][a][g];
-                /* butterfly over 4 lanes: s=2, s=1 */
-                if (DEC) {
-                    int64_t y = (int64_t)x;
-                    y += __shfl_xor(y, 2, WAVE);
-                    y += __shfl_xor(y, 1, WAVE);
-                    x = (T)y;
-                } else {
-                    double y = (double)x;
-                    y += __shfl_xor(y, 2, WAVE);
-                    y += __shfl_xor(y, 1, WAVE);
-                    x = (T)y;
+#pragma unroll
+                for (int w = 2; w >= 1; w >>= 1) {
+                    T o;
+                    if (DEC)
+                        o = (T)__shfl_xor((int64_t)x, w, WAVE);
+                    else
+                        o = (T)__shfl_xor((double)x, w, WAVE);
+                    if (fn == PG_AGG_MIN)
+                        x = o < x ? o : x;
+                    else if (fn == PG_AGG_MAX)
+                        x = o > x ? o : x;
+                    else
+                        x = x + o;
                 }
                 if (lane == 0) {
                     size_t off =
                         ((size_t)blockIdx.x * NA + a) * MAXG + g;
-                    if (DEC)
-                        partials_i[off] += (int64_t)x;
-                    else
-                        partials[off] += (double)x;
+                    if (DEC) {
+                        int64_t* d = &partials_i[off];
+                        if (fn == PG_AGG_MIN)
+                            *d = (int64_t)x < *d ? (int64_t)x : *d;
+                        else if (fn == PG_AGG_MAX)
+                            *d = (int64_t)x > *d ? (int64_t)x : *d;
+                        else
+                            *d += (int64_t)x;
+                    } else {
+                        double* d = &partials[off];
+                        if (fn == PG_AGG_MIN)
+                            *d = (double)x < *d ? (double)x : *d;
+                        else if (fn == PG_AGG_MAX)
+                            *d = (double)x > *d ? (double)x : *d;
+                        else
+                            *d += (double)x;
+                    }
                 }
             }
+        }
     }
     if (local_bad) atomicAdd(bad_keys, local_bad);
 }
@@ -597,18 +645,66 @@ __global__ __launch_bounds__(FT_NTHREADS) void k_agg_q1(
     if (local_bad) atomicAdd(bad_keys, local_bad);
 }
 
+struct agg_funcs {
+    int32_t f[12];
+    int32_t maxg;
+};
+
+/* initialize block partials to each aggregate's identity (0 for sums and
+ * counts; +/-extremes for MIN/MAX) */
+template <bool DEC>
+__global__ __launch_bounds__(256) void k_agg_partials_init(
+    double* partials, int64_t* partials_i, agg_funcs fns, int64_t n_total,
+    int na)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n_total; i += stride) {
+        int a = (int)((i / fns.maxg) % na);
+        int32_t fn = fns.f[a];
+        if (DEC) {
+            int64_t id = fn == PG_AGG_MIN
+                             ? INT64_MAX
+                             : (fn == PG_AGG_MAX ? INT64_MIN : 0);
+            partials_i[i] = id;
+        } else {
+            double id = fn == PG_AGG_MIN
+                            ? INFINITY
+                            : (fn == PG_AGG_MAX ? -INFINITY : 0.0);
+            partials[i] = id;
+        }
+    }
+}
+
 /* finish: 64-lane deterministic reduce of block partials.
- * DEC: output int128 (hi,lo) pairs; F64: doubles. */
+ * DEC: output int128 (hi,lo) pairs (plain i64 for MIN/MAX); F64: doubles. */
 template <bool DEC>
 __global__ __launch_bounds__(WAVE) void k_agg_small_finish(
     const double* partials, const int64_t* partials_i, int na_x_maxg,
-    double* out_f, int64_t* out_hi, uint64_t* out_lo)
+    double* out_f, int64_t* out_hi, uint64_t* out_lo, agg_funcs fns)
 {
     int lane = threadIdx.x;
     {
         int j = blockIdx.x; /* one block (one wave) per output field; the
                                per-field reduction tree is unchanged */
-        if (DEC) {
+        int32_t fn = fns.f[j / fns.maxg];
+        if (DEC && (fn == PG_AGG_MIN || fn == PG_AGG_MAX)) {
+            int64_t v = fn == PG_AGG_MIN ? INT64_MAX : INT64_MIN;
+            for (int k = 0; k < FT_NBLOCKS / WAVE; k++) {
+                int64_t x = partials_i[(size_t)(lane + k * WAVE) *
+                                           na_x_maxg + j];
+                v = fn == PG_AGG_MIN ? (x < v ? x : v) : (x > v ? x : v);
+            }
+#pragma unroll
+            for (int w = 32; w >= 1; w >>= 1) {
+                int64_t o = __shfl_xor(v, w, WAVE);
+                v = fn == PG_AGG_MIN ? (o < v ? o : v) : (o > v ? o : v);
+            }
+            if (lane == 0) {
+                out_hi[j] = v < 0 ? -1 : 0;
+                out_lo[j] = (uint64_t)v;
+            }
+        } else if (DEC) {
             /* lane l: sequential over blocks l, l+64, ... (i64 safe:
              * 64 block partials each bounded by per-block row counts) */
             int64_t hi = 0;
@@ -637,10 +733,29 @@ __global__ __launch_bounds__(WAVE) void k_agg_small_finish(
                 out_lo[j] = lo;
             }
         } else {
-            double v = 0;
-            for (int k = 0; k < FT_NBLOCKS / WAVE; k++)
-                v += partials[(size_t)(lane + k * WAVE) * na_x_maxg + j];
-            v = d_bfly_f64(v);
+            double v = fn == PG_AGG_MIN
+                           ? INFINITY
+                           : (fn == PG_AGG_MAX ? -INFINITY : 0.0);
+            for (int k = 0; k < FT_NBLOCKS / WAVE; k++) {
+                double x =
+                    partials[(size_t)(lane + k * WAVE) * na_x_maxg + j];
+                if (fn == PG_AGG_MIN)
+                    v = x < v ? x : v;
+                else if (fn == PG_AGG_MAX)
+                    v = x > v ? x : v;
+                else
+                    v += x;
+            }
+#pragma unroll
+            for (int w = 32; w >= 1; w >>= 1) {
+                double o = __shfl_xor(v, w, WAVE);
+                if (fn == PG_AGG_MIN)
+                    v = o < v ? o : v;
+                else if (fn == PG_AGG_MAX)
+                    v = o > v ? o : v;
+                else
+                    v += o;
+            }
             if (lane == 0) out_f[j] = v;
         }
     }
@@ -1951,11 +2066,12 @@ struct AggSmallOp : Op {
                 dec = true;
         }
         for (int a = 0; a < user_aggs; a++) {
-            bool adec = plan.aggs[a].func == PG_AGG_SUM_DEC ||
-                        plan.aggs[a].func == PG_AGG_SUM_I64 ||
-                        plan.aggs[a].func == PG_AGG_COUNT;
-            bool af64 = plan.aggs[a].func == PG_AGG_SUM_F64 ||
-                        plan.aggs[a].func == PG_AGG_COUNT;
+            int32_t fn = plan.aggs[a].func;
+            bool both = fn == PG_AGG_COUNT || fn == PG_AGG_MIN ||
+                        fn == PG_AGG_MAX;
+            bool adec = both || fn == PG_AGG_SUM_DEC ||
+                        fn == PG_AGG_SUM_I64;
+            bool af64 = both || fn == PG_AGG_SUM_F64;
             if (dec ? !adec : !af64)
                 throw std::runtime_error(
                     "mixed decimal/f64 aggregates in one op unsupported");
@@ -1981,7 +2097,32 @@ struct AggSmallOp : Op {
         /* one allocation: [partials | bad flag] (fewer alloc/memset/launch
          * round trips per operator lifecycle) */
         partials.alloc((size_t)FT_NBLOCKS * na * maxg * 8 + 64);
-        partials.zero();
+        partials.zero(); /* zero covers sums/counts + the bad flag */
+        bool has_mm = false;
+        for (int a = 0; a < plan.n_aggs; a++)
+            has_mm = has_mm || plan.aggs[a].func == PG_AGG_MIN ||
+                     plan.aggs[a].func == PG_AGG_MAX;
+        if (has_mm) {
+            agg_funcs fns = funcs();
+            if (dec)
+                hipLaunchKernelGGL(k_agg_partials_init<true>, dim3(512),
+                                   dim3(256), 0, g_stream, nullptr,
+                                   (int64_t*)partials.p, fns,
+                                   (int64_t)FT_NBLOCKS * na * maxg, na);
+            else
+                hipLaunchKernelGGL(k_agg_partials_init<false>, dim3(512),
+                                   dim3(256), 0, g_stream,
+                                   (double*)partials.p, nullptr, fns,
+                                   (int64_t)FT_NBLOCKS * na * maxg, na);
+        }
+    }
+    agg_funcs funcs() const
+    {
+        agg_funcs f{};
+        f.maxg = maxg;
+        for (int a = 0; a < 12; a++)
+            f.f[a] = a < plan.n_aggs ? plan.aggs[a].func : PG_AGG_COUNT;
+        return f;
     }
     void* bad_ptr() const
     {
@@ -2116,12 +2257,14 @@ struct AggSmallOp : Op {
             hipLaunchKernelGGL(k_agg_small_finish<true>, dim3(nm),
                                dim3(WAVE), 0, g_stream, nullptr,
                                (int64_t*)partials.p, nm, nullptr,
-                               (int64_t*)out_hi.p, (uint64_t*)out_lo.p);
+                               (int64_t*)out_hi.p, (uint64_t*)out_lo.p,
+                               funcs());
         else
             hipLaunchKernelGGL(k_agg_small_finish<false>, dim3(nm),
                                dim3(WAVE), 0, g_stream,
                                (double*)partials.p, nullptr, nm,
-                               (double*)out_f.p, nullptr, nullptr);
+                               (double*)out_f.p, nullptr, nullptr,
+                               funcs());
         std::vector<double> hf(nm);
         std::vector<int64_t> hhi(nm);
         std::vector<uint64_t> hlo(nm);

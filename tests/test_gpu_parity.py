@@ -913,3 +913,59 @@ def test_q5_distributed_graph_world1(P, oracle_lib):
     exp = {int(r.nationkey): int(r.revenue_1e4)
            for r in oracle_lib.q5(cust, orders, li, supp)}
     assert got == exp
+
+
+def test_min_max_aggregates(P):
+    """MIN/MAX aggregates (MinAggregationFunction analogs) in both modes,
+    grouped and keyless, vs numpy."""
+    rng = np.random.default_rng(123)
+    n = 200_000
+    k = rng.choice(np.array([3, 9, 200], np.uint8), n)
+    vi = rng.integers(-10**12, 10**12, n)
+    vf = rng.standard_normal(n) * 1e6
+    sd = rng.integers(0, 100, n).astype(np.int32)
+    sel = sd < 70
+    page = P.Page({"k": k, "vi": vi, "vf": vf, "sd": sd})
+    # decimal/i64 mode: MIN, MAX, COUNT grouped by k
+    plan = P.PlanHashAggSmall()
+    plan.n_preds = 1
+    plan.preds[0] = P.Pred(3, P.CMP_LT, 70, 0.0)
+    plan.n_keys = 1
+    plan.key_col[0] = 0
+    plan.n_vals[0] = 3
+    for j, v in enumerate((3, 9, 200)):
+        plan.key_vals[0][j] = v
+    plan.n_aggs = 3
+    plan.aggs[0] = P.Agg(P.AGG_MIN, P.Proj(P.PROJ_IDENT, 1, 0, 0), 0)
+    plan.aggs[1] = P.Agg(P.AGG_MAX, P.Proj(P.PROJ_IDENT, 1, 0, 0), 0)
+    plan.aggs[2] = P.Agg(P.AGG_SUM_DEC, P.Proj(P.PROJ_IDENT, 1, 0, 0), 0)
+    op = P.Operator(P.OP_HASH_AGG_SMALL, plan)
+    op.add_input(page)
+    op.finish()
+    out = op.get_output(["k", "min_hi", "min_lo", "max_hi", "max_lo",
+                         "s_hi", "s_lo", "cnt"])
+    op.destroy()
+    for row, kv in enumerate((3, 9, 200)):
+        m = sel & (k == kv)
+        assert out["min_lo"][row] == vi[m].min()
+        assert np.int64(out["min_hi"][row]) == (-1 if vi[m].min() < 0 else 0)
+        assert out["max_lo"][row] == vi[m].max()
+        assert out["s_lo"][row] == vi[m].sum()
+        assert out["cnt"][row] == int(m.sum())
+    # f64 mode: keyless MIN/MAX/COUNT
+    plan2 = P.PlanHashAggSmall()
+    plan2.n_preds = 1
+    plan2.preds[0] = P.Pred(3, P.CMP_LT, 70, 0.0)
+    plan2.n_keys = 0
+    plan2.n_aggs = 3
+    plan2.aggs[0] = P.Agg(P.AGG_MIN, P.Proj(P.PROJ_IDENT, 2, 0, 0), 0)
+    plan2.aggs[1] = P.Agg(P.AGG_MAX, P.Proj(P.PROJ_IDENT, 2, 0, 0), 0)
+    plan2.aggs[2] = P.Agg(P.AGG_COUNT, P.Proj(P.PROJ_IDENT, 0, 0, 0), 0)
+    op = P.Operator(P.OP_HASH_AGG_SMALL, plan2)
+    op.add_input(page)
+    op.finish()
+    out2 = op.get_output(["mn", "mx", "cnt"])
+    op.destroy()
+    assert out2["mn"][0] == vf[sel].min()
+    assert out2["mx"][0] == vf[sel].max()
+    assert out2["cnt"][0] == int(sel.sum())
