@@ -935,10 +935,11 @@ def test_min_max_aggregates(P):
     plan.n_vals[0] = 3
     for j, v in enumerate((3, 9, 200)):
         plan.key_vals[0][j] = v
-    plan.n_aggs = 3
+    plan.n_aggs = 4
     plan.aggs[0] = P.Agg(P.AGG_MIN, P.Proj(P.PROJ_IDENT, 1, 0, 0), 0)
     plan.aggs[1] = P.Agg(P.AGG_MAX, P.Proj(P.PROJ_IDENT, 1, 0, 0), 0)
     plan.aggs[2] = P.Agg(P.AGG_SUM_DEC, P.Proj(P.PROJ_IDENT, 1, 0, 0), 0)
+    plan.aggs[3] = P.Agg(P.AGG_COUNT, P.Proj(P.PROJ_IDENT, 0, 0, 0), 0)
     op = P.Operator(P.OP_HASH_AGG_SMALL, plan)
     op.add_input(page)
     op.finish()
