@@ -793,13 +793,18 @@ __global__ __launch_bounds__(256) void k_sel_count(pg_page pg,
         bool sel = i < hi && d_eval_preds(pg, plan.preds, plan.n_preds, i);
         if (sel && set_keys) {
             int64_t key = d_load_i64(pg.cols[set_col], i);
-            uint64_t h = pg_murmur3_finalize(pg_bigint_hash(key));
-            int64_t s = (int64_t)(h & (uint64_t)set_mask);
-            for (;;) {
-                int64_t k = set_keys[s];
-                if (k == key) break;
-                if (k == TBL_EMPTY) { sel = false; break; }
-                s = (s + 1) & set_mask;
+            if (set_mask < 0) { /* dense flags: |mask| = capacity */
+                sel = key >= 1 && key <= -set_mask &&
+                      ((const uint8_t*)set_keys)[key - 1];
+            } else {
+                uint64_t h = pg_murmur3_finalize(pg_bigint_hash(key));
+                int64_t s = (int64_t)(h & (uint64_t)set_mask);
+                for (;;) {
+                    int64_t k = set_keys[s];
+                    if (k == key) break;
+                    if (k == TBL_EMPTY) { sel = false; break; }
+                    s = (s + 1) & set_mask;
+                }
             }
         }
         uint64_t m = d_ballot(sel);
@@ -867,13 +872,18 @@ __global__ __launch_bounds__(256) void k_sel_emit(pg_page pg,
         bool sel = i < hi && d_eval_preds(pg, plan.preds, plan.n_preds, i);
         if (sel && set_keys) {
             int64_t key = d_load_i64(pg.cols[set_col], i);
-            uint64_t h = pg_murmur3_finalize(pg_bigint_hash(key));
-            int64_t s = (int64_t)(h & (uint64_t)set_mask);
-            for (;;) {
-                int64_t k = set_keys[s];
-                if (k == key) break;
-                if (k == TBL_EMPTY) { sel = false; break; }
-                s = (s + 1) & set_mask;
+            if (set_mask < 0) {
+                sel = key >= 1 && key <= -set_mask &&
+                      ((const uint8_t*)set_keys)[key - 1];
+            } else {
+                uint64_t h = pg_murmur3_finalize(pg_bigint_hash(key));
+                int64_t s = (int64_t)(h & (uint64_t)set_mask);
+                for (;;) {
+                    int64_t k = set_keys[s];
+                    if (k == key) break;
+                    if (k == TBL_EMPTY) { sel = false; break; }
+                    s = (s + 1) & set_mask;
+                }
             }
         }
         uint64_t m = d_ballot(sel);
@@ -906,16 +916,18 @@ __global__ __launch_bounds__(256) void k_sel_emit(pg_page pg,
 /* insert order is non-deterministic across duplicates — result SET   */
 /* semantics).                                         */
 /* ------------------------------------------------------------------ */
-/* dense-array build: payload[key-1] = u8 value */
+/* dense-array build: payload[key-1] = u8 value (or membership flag 1 for
+ * key sets); rows failing the plan predicates are skipped */
 __global__ __launch_bounds__(256) void k_dense_fill(
-    const int64_t* keys, const uint8_t* vals, int64_t n, uint8_t* out,
+    pg_page pg, pg_plan_hash_build plan, const uint8_t* vals, uint8_t* out,
     int64_t cap)
 {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    for (; i < n; i += stride) {
-        int64_t k = keys[i];
-        if (k >= 1 && k <= cap) out[k - 1] = vals[i];
+    for (; i < pg.n_rows; i += stride) {
+        if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
+        int64_t k = d_load_i64(pg.cols[plan.key_col], i);
+        if (k >= 1 && k <= cap) out[k - 1] = vals ? vals[i] : 1;
     }
 }
 
@@ -1022,14 +1034,20 @@ __global__ __launch_bounds__(256) void k_tbl_insert_direct(
         }
         if (set_keys) {
             int64_t sk = d_load_i64(pg.cols[plan.semijoin_col], i);
-            uint64_t h = pg_murmur3_finalize(pg_bigint_hash(sk));
-            int64_t s = (int64_t)(h & (uint64_t)set_mask);
-            bool found = false;
-            for (;;) {
-                int64_t k = set_keys[s];
-                if (k == sk) { found = true; break; }
-                if (k == TBL_EMPTY) break;
-                s = (s + 1) & set_mask;
+            bool found;
+            if (set_mask < 0) { /* dense flags */
+                found = sk >= 1 && sk <= -set_mask &&
+                        ((const uint8_t*)set_keys)[sk - 1];
+            } else {
+                found = false;
+                uint64_t h = pg_murmur3_finalize(pg_bigint_hash(sk));
+                int64_t s = (int64_t)(h & (uint64_t)set_mask);
+                for (;;) {
+                    int64_t k = set_keys[s];
+                    if (k == sk) { found = true; break; }
+                    if (k == TBL_EMPTY) break;
+                    s = (s + 1) & set_mask;
+                }
             }
             if (!found) continue;
         }
@@ -1953,6 +1971,20 @@ struct SelResult {
     std::vector<int64_t> block_offs; /* host */
 };
 
+/* semijoin set arguments: dense sets pass the flag array with a negative
+ * mask (-capacity); hash sets pass keys + mask */
+static const int64_t* set_ptr(const Table* semi)
+{
+    if (!semi) return nullptr;
+    return semi->dense ? (const int64_t*)semi->payload[0].p
+                       : (const int64_t*)semi->keys.p;
+}
+static int64_t set_mask_of(const Table* semi)
+{
+    if (!semi) return 0;
+    return semi->dense ? -semi->cap : semi->mask;
+}
+
 static SelResult sel_count(const pg_page& pg,
                            const pg_plan_filter_project& plan,
                            const Table* semi, int32_t semi_col,
@@ -1961,9 +1993,8 @@ static SelResult sel_count(const pg_page& pg,
     DevBuf d_counts;
     d_counts.alloc(FLT_NB * sizeof(int64_t));
     hipLaunchKernelGGL(k_sel_count, dim3(FLT_NB), dim3(256), 0, g_stream, pg,
-                       plan, chunk, (int64_t*)d_counts.p,
-                       semi ? (const int64_t*)semi->keys.p : nullptr,
-                       semi ? semi->mask : 0, semi_col);
+                       plan, chunk, (int64_t*)d_counts.p, set_ptr(semi),
+                       set_mask_of(semi), semi_col);
     std::vector<int64_t> h(FLT_NB);
     CHKV(hipMemcpyAsync(h.data(), d_counts.p, FLT_NB * 8,
                         hipMemcpyDeviceToHost, g_stream));
@@ -1993,8 +2024,7 @@ static void sel_emit(const pg_page& pg, const pg_plan_filter_project& plan,
                         hipMemcpyHostToDevice, g_stream));
     hipLaunchKernelGGL(k_sel_emit, dim3(FLT_NB), dim3(256), 0, g_stream, pg,
                        plan, chunk, (const int64_t*)d_offs.p, outs,
-                       semi ? (const int64_t*)semi->keys.p : nullptr,
-                       semi ? semi->mask : 0, semi_col);
+                       set_ptr(semi), set_mask_of(semi), semi_col);
     CHKV(hipStreamSynchronize(g_stream));
 }
 
@@ -2358,14 +2388,15 @@ struct BuildOp : Op {
         t.reset(new Table());
         t->key_set_only = plan.key_set_only != 0;
         if (plan.dense_array) {
-            if (plan.n_payload != 1)
+            if (plan.n_payload != 1 && !plan.key_set_only)
                 throw std::runtime_error(
-                    "dense_array needs exactly one u8 payload");
+                    "dense_array needs one u8 payload (or key_set_only)");
             t->dense = true;
             t->cap = plan.capacity_hint;
             t->payload.emplace_back();
             t->ptag.push_back(PG_T_U8);
             t->payload.back().alloc((size_t)t->cap);
+            if (plan.key_set_only) t->payload.back().zero();
             return;
         }
         if (plan.agg_table) {
@@ -2448,16 +2479,17 @@ struct BuildOp : Op {
                                  : sp.pg.cols[plan.payload_col[i]].tag;
         if (plan.dense_array) {
             if (sp.pg.cols[plan.key_col].tag != PG_T_I64 ||
-                sp.pg.cols[plan.payload_col[0]].tag != PG_T_U8)
+                (!plan.key_set_only &&
+                 sp.pg.cols[plan.payload_col[0]].tag != PG_T_U8))
                 throw std::runtime_error(
                     "dense_array expects I64 keys and a U8 payload");
-            hipLaunchKernelGGL(k_dense_fill, dim3(2048), dim3(256), 0,
-                               g_stream,
-                               (const int64_t*)sp.pg.cols[plan.key_col].data,
-                               (const uint8_t*)
-                                   sp.pg.cols[plan.payload_col[0]].data,
-                               sp.pg.n_rows, (uint8_t*)t->payload[0].p,
-                               t->cap);
+            hipLaunchKernelGGL(
+                k_dense_fill, dim3(2048), dim3(256), 0, g_stream, sp.pg,
+                plan,
+                plan.key_set_only
+                    ? nullptr
+                    : (const uint8_t*)sp.pg.cols[plan.payload_col[0]].data,
+                (uint8_t*)t->payload[0].p, t->cap);
             CHKV(hipStreamSynchronize(g_stream));
             return;
         }
@@ -2485,9 +2517,8 @@ struct BuildOp : Op {
             }
             hot_begin();
             hipLaunchKernelGGL(k_tbl_insert_direct, dim3(4096), dim3(256),
-                               0, g_stream, sp.pg, plan,
-                               semi ? (const int64_t*)semi->keys.p : nullptr,
-                               semi ? semi->mask : 0,
+                               0, g_stream, sp.pg, plan, set_ptr(semi),
+                               set_mask_of(semi),
                                lu && !lu->dense
                                    ? (const int64_t*)lu->keys.p
                                    : nullptr,

@@ -113,8 +113,10 @@ def q3_distributed(cust_page, ord_page, li_page, world, rank, device,
     bp.key_col = 0
     bp.semijoin_table = -1
     bp.n_payload = 0
-    bp.capacity_hint = max(int(ck.numel()), 16)
+    # dense membership flags sized to the global customer cardinality
+    bp.capacity_hint = cust_page.n_rows * world + world
     bp.key_set_only = 1
+    bp.dense_array = 1
     b1 = P.Operator(P.OP_HASH_BUILD, bp)
     b1.add_input(P.Page({"custkey": ck}))
     b1.finish()

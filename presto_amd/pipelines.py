@@ -97,8 +97,9 @@ class Q3Pipeline:
         b1p.key_col = cust.channel("custkey")
         b1p.semijoin_table = -1
         b1p.n_payload = 0
-        b1p.capacity_hint = max(cust.n_rows // 4, 16)
+        b1p.capacity_hint = cust.n_rows  # custkeys dense 1..n
         b1p.key_set_only = 1
+        b1p.dense_array = 1  # 1-byte membership flags, L2/L3-resident
         self.b1 = Operator(OP_HASH_BUILD, b1p)
         self.b1.add_input(cust)
         self.b1.finish()
