@@ -4,9 +4,9 @@
 The fixtures are the reference's own TPC-H SF1 known-answer vectors
 (exact-decimal result sets, produced by the reference's product-test suite):
 
-  q01_sf1.result <- /root/reference/presto-product-tests/src/main/resources/
-                    sql-tests/testcases/hive_tpch/q01.result
-  q03_sf1.result <- .../hive_tpch/q03.result
+  qNN_sf1.result <- /root/reference/presto-product-tests/src/main/resources/
+                    sql-tests/testcases/hive_tpch/qNN.result
+  for NN in 01, 03, 04, 05, 06, 07
 
 They pin (a) the TPC-H dbgen restatement in oracle/tpchgen.c and (b) the
 decimal aggregate semantics of oracle/oracle.c — see tests/test_oracle.py.
@@ -20,6 +20,6 @@ REF = pathlib.Path("/root/reference/presto-product-tests/src/main/resources/"
                    "sql-tests/testcases/hive_tpch")
 HERE = pathlib.Path(__file__).parent
 
-for q in ("q01", "q03"):
+for q in ("q01", "q03", "q04", "q05", "q06", "q07"):
     shutil.copy(REF / f"{q}.result", HERE / f"{q}_sf1.result")
     print(f"wrote {q}_sf1.result")
