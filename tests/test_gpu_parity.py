@@ -970,3 +970,22 @@ def test_min_max_aggregates(P):
     assert out2["mn"][0] == vf[sel].min()
     assert out2["mx"][0] == vf[sel].max()
     assert out2["cnt"][0] == int(sel.sum())
+
+
+def test_cross_run_bit_determinism(P, oracle_lib, sf01):
+    """Two independent executions must produce bit-identical results —
+    the deterministic-schedule (Q1 f64) and exact-arithmetic (Q3 fx128 via
+    atomics) guarantees of DESIGN.md §determinism."""
+    li, orders, cust = sf01["li"], sf01["orders"], sf01["cust"]
+    page = _li_page(P, li)
+    a = P.pipelines.q1(page, mode="f64")
+    b = P.pipelines.q1(page, mode="f64")
+    for k in a:
+        assert np.array_equal(a[k].view(np.int64), b[k].view(np.int64)), k
+    cp = P.Page({"custkey": cust["custkey"], "mktseg": cust["mktseg"]})
+    op = P.Page({k: orders[k] for k in ("orderkey", "custkey", "orderdate")})
+    r1 = P.pipelines.q3(cp, op, page, mode="f64")
+    r2 = P.pipelines.q3(cp, op, page, mode="f64")
+    assert np.array_equal(r1["revenue"].view(np.int64),
+                          r2["revenue"].view(np.int64))
+    assert np.array_equal(r1["orderkey"], r2["orderkey"])
