@@ -308,6 +308,89 @@ int64_t oracle_join_bigint(int64_t n_build, const int64_t* bkeys,
     return out;
 }
 
+/* ---------------- Q8 ---------------- */
+
+void oracle_q8(int64_t n_cust, const int64_t* ck, const uint8_t* cnat,
+               int64_t n_ord, const int64_t* ook, const int64_t* ock,
+               const int32_t* od, int64_t n_li, const int64_t* lok,
+               const int64_t* lsk, const int64_t* lpk, const double* lep,
+               const double* ldisc, int64_t n_supp, const uint8_t* snat,
+               int64_t n_part, const uint8_t* ptype, int64_t* brazil,
+               int64_t* total)
+{
+    extern int32_t tpch_nation_region(int32_t);
+    int64_t max_ck = 0;
+    for (int64_t i = 0; i < n_cust; i++)
+        if (ck[i] > max_ck) max_ck = ck[i];
+    uint8_t* cn = (uint8_t*)calloc(max_ck + 1, 1);
+    for (int64_t i = 0; i < n_cust; i++) cn[ck[i]] = cnat[i];
+    /* orders with date in [1995-01-01, 1996-12-31] and AMERICA customer */
+    int64_t* b_ok = (int64_t*)malloc(n_ord * sizeof(int64_t));
+    int32_t* b_od = (int32_t*)malloc(n_ord * sizeof(int32_t));
+    int64_t n_b = 0;
+    for (int64_t i = 0; i < n_ord; i++) {
+        if (od[i] < 9131 || od[i] > 9861) continue;
+        if (ock[i] > max_ck || tpch_nation_region(cn[ock[i]]) != 1)
+            continue;
+        b_ok[n_b] = ook[i];
+        b_od[n_b] = od[i];
+        n_b++;
+    }
+    int64_t cap = hash_capacity(n_b < 2 ? 2 : n_b);
+    int64_t* slot = (int64_t*)malloc(cap * sizeof(int64_t));
+    memset(slot, -1, cap * sizeof(int64_t));
+    for (int64_t i = 0; i < n_b; i++) {
+        uint64_t h = pg_murmur3_finalize(pg_bigint_hash(b_ok[i]));
+        int64_t pos = (int64_t)(h & (cap - 1));
+        while (slot[pos] != -1) pos = (pos + 1) & (cap - 1);
+        slot[pos] = i;
+    }
+    int64_t br[2] = {0, 0}, tt[2] = {0, 0};
+#pragma omp parallel
+    {
+        int64_t lbr[2] = {0, 0}, ltt[2] = {0, 0};
+#pragma omp for schedule(static)
+        for (int64_t i = 0; i < n_li; i++) {
+            int64_t pk = lpk[i];
+            if (pk < 1 || pk > n_part || ptype[pk - 1] != 103) continue;
+            uint64_t h = pg_murmur3_finalize(pg_bigint_hash(lok[i]));
+            int64_t pos = (int64_t)(h & (cap - 1));
+            int64_t r = -1;
+            for (;;) {
+                int64_t sI = slot[pos];
+                if (sI == -1) break;
+                if (b_ok[sI] == lok[i]) {
+                    r = sI;
+                    break;
+                }
+                pos = (pos + 1) & (cap - 1);
+            }
+            if (r == -1) continue;
+            int yr = b_od[r] <= 9495 ? 0 : 1;
+            int64_t cents = (int64_t)(lep[i] * 100.0 + 0.5);
+            int64_t d = (int64_t)(ldisc[i] * 100.0 + 0.5);
+            int64_t t = cents * (100 - d);
+            ltt[yr] += t;
+            int64_t sk = lsk[i];
+            if (sk >= 1 && sk <= n_supp && snat[sk - 1] == 2)
+                lbr[yr] += t;
+        }
+#pragma omp critical
+        for (int y = 0; y < 2; y++) {
+            br[y] += lbr[y];
+            tt[y] += ltt[y];
+        }
+    }
+    free(cn);
+    free(b_ok);
+    free(b_od);
+    free(slot);
+    for (int y = 0; y < 2; y++) {
+        brazil[y] = br[y];
+        total[y] = tt[y];
+    }
+}
+
 /* ---------------- Q4 ---------------- */
 
 void oracle_q4(int64_t n_ord, const int64_t* ook, const int32_t* od,

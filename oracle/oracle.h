@@ -114,6 +114,23 @@ int32_t oracle_q5(int64_t n_cust, const int64_t* c_custkey,
                   int64_t n_supp, const uint8_t* s_nationkey,
                   q5_row_t* out /* capacity 25 */);
 
+/* ---------------- TPC-H Q8 ----------------
+ * SQL: q08.sql — national market share: volume of 'ECONOMY ANODIZED
+ * STEEL' (type id 103) parts sold to AMERICA-region customers in
+ * 1995/1996, share of BRAZIL(2)-nation suppliers.  Returns exact tick
+ * sums: brazil[2] and total[2] for years 1995, 1996 (share =
+ * brazil/total). */
+void oracle_q8(int64_t n_cust, const int64_t* c_custkey,
+               const uint8_t* c_nationkey, int64_t n_ord,
+               const int64_t* o_orderkey, const int64_t* o_custkey,
+               const int32_t* o_orderdate, int64_t n_li,
+               const int64_t* l_orderkey, const int64_t* l_suppkey,
+               const int64_t* l_partkey, const double* l_extendedprice,
+               const double* l_discount, int64_t n_supp,
+               const uint8_t* s_nationkey, int64_t n_part,
+               const uint8_t* p_type, int64_t* brazil_1e4,
+               int64_t* total_1e4);
+
 /* ---------------- TPC-H Q4 ----------------
  * SQL: q04.sql — order priority checking: orders with orderdate in
  * [1993-07-01, 1993-10-01) = [8582, 8674) having EXISTS a lineitem with

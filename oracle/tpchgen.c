@@ -46,6 +46,10 @@ static inline int64_t unif(int64_t* s, int64_t lo, int64_t hi)
 #define SEED_S_NATION  110356601LL /* supplier nationkey, usage 1/supplier */
 #define SEED_L_SUPPN  2095021727LL /* lineitem supplier number 0..3, 7/order */
 #define SEED_O_PRIO    591449447LL /* order priority pick 1..5, 1/order */
+#define SEED_P_TYPE   1841581359LL /* part type pick 1..150, 1/part
+                                    * (dbgen rnd.h P_TYPE_SD; pinned by the
+                                    * q08 golden: id 103 'ECONOMY ANODIZED
+                                    * STEEL' shares 0.0344/0.0415 match) */
 
 /* ---- calendar ----
  * day index 1 = 1992-01-01; order-date index in [1, 2406]
@@ -241,6 +245,57 @@ static void gen_orders_chunk(double sf, int64_t start, int64_t count,
         if (orderdate_epoch) orderdate_epoch[i] = (int32_t)(EPOCH_1992 + od - 1);
         if (lcnt) lcnt[i] = (int32_t)lc;
     }
+}
+
+void tpch_gen_part_type(double sf, int64_t start, int64_t count,
+                        uint8_t* type_id)
+{
+#pragma omp parallel
+    {
+#ifdef _OPENMP
+        extern int omp_get_num_threads(void), omp_get_thread_num(void);
+        int nt = omp_get_num_threads(), tid = omp_get_thread_num();
+#else
+        int nt = 1, tid = 0;
+#endif
+        (void)sf;
+        int64_t lo = count * tid / nt, hi = count * (tid + 1) / nt;
+        int64_t s = rng_skip(SEED_P_TYPE, (uint64_t)(start + lo));
+        for (int64_t i = lo; i < hi; i++)
+            type_id[i] = (uint8_t)(unif(&s, 1, 150) - 1);
+    }
+}
+
+int64_t tpch_gen_lineitem_partkey(double sf, int64_t ord_start,
+                                  int64_t ord_count, int64_t* partkey)
+{
+    int64_t max_pkey = (int64_t)(200000.0 * sf + 0.5);
+    int64_t written = 0;
+#pragma omp parallel reduction(+ : written)
+    {
+#ifdef _OPENMP
+        extern int omp_get_num_threads(void), omp_get_thread_num(void);
+        int nt = omp_get_num_threads(), tid = omp_get_thread_num();
+#else
+        int nt = 1, tid = 0;
+#endif
+        int64_t tlo = ord_count * tid / nt, thi = ord_count * (tid + 1) / nt;
+        int64_t s = rng_skip(SEED_O_LCNT, (uint64_t)ord_start);
+        int64_t off = 0;
+        for (int64_t i = 0; i < tlo; i++) off += unif(&s, 1, 7);
+        int64_t s_lcnt = rng_skip(SEED_O_LCNT, (uint64_t)(ord_start + tlo));
+        int64_t s_pkey =
+            rng_skip(SEED_L_PKEY, (uint64_t)(ord_start + tlo) * 7);
+        int64_t out = off;
+        for (int64_t o = tlo; o < thi; o++) {
+            int64_t lc = unif(&s_lcnt, 1, 7);
+            for (int64_t l = 0; l < lc; l++)
+                partkey[out++] = unif(&s_pkey, 1, max_pkey);
+            s_pkey = rng_skip(s_pkey, (uint64_t)(7 - lc));
+            written += lc;
+        }
+    }
+    return written;
 }
 
 void tpch_gen_orders_priority(double sf, int64_t start, int64_t count,

@@ -74,6 +74,15 @@ int64_t tpch_gen_lineitem_dates(double sf, int64_t ord_start,
                                 int32_t* commitdate_epoch,
                                 int32_t* receiptdate_epoch);
 
+/* part type ids for part rows [start, start+count): id 0..149 indexes the
+ * dbgen p_type 3-word combinations (Types1 x Types2 x Types3 nested);
+ * 'ECONOMY ANODIZED STEEL' = 103. */
+void tpch_gen_part_type(double sf, int64_t start, int64_t count,
+                        uint8_t* type_id);
+/* lineitem partkeys (the L_PKEY stream replayed standalone) */
+int64_t tpch_gen_lineitem_partkey(double sf, int64_t ord_start,
+                                  int64_t ord_count, int64_t* partkey);
+
 /* orders columns for order rows [start, start+count) (0-based).
  * shippriority is the constant 0 (dbgen mk_order) and is not emitted.
  * lcnt: lineitems per order (1..7). Any output pointer may be NULL. */

@@ -278,3 +278,42 @@ def q4(self, orders, pri, lid):
 OracleLib.gen_orders_priority = gen_orders_priority
 OracleLib.gen_lineitem_dates = gen_lineitem_dates
 OracleLib.q4 = q4
+
+
+def gen_part_type(self, sf):
+    n = int(200000 * sf)
+    t = np.empty(n, np.uint8)
+    self.lib.tpch_gen_part_type(C.c_double(sf), C.c_int64(0), C.c_int64(n),
+                                _p(t))
+    return t
+
+
+def gen_lineitem_partkey(self, sf):
+    n = self.lineitem_count(sf)
+    n_ord = self.lib.tpch_orders_count(C.c_double(sf))
+    pk = np.empty(n, np.int64)
+    self.lib.tpch_gen_lineitem_partkey.restype = C.c_int64
+    w = self.lib.tpch_gen_lineitem_partkey(C.c_double(sf), C.c_int64(0),
+                                           C.c_int64(n_ord), _p(pk))
+    assert w == n
+    return pk
+
+
+def q8(self, cust, orders, li, lpk, supp, ptype):
+    br = (C.c_int64 * 2)()
+    tt = (C.c_int64 * 2)()
+    self.lib.oracle_q8(
+        C.c_int64(len(cust["custkey"])), _p(cust["custkey"]),
+        _p(cust["nationkey"]), C.c_int64(len(orders["orderkey"])),
+        _p(orders["orderkey"]), _p(orders["custkey"]),
+        _p(orders["orderdate"]), C.c_int64(len(li["orderkey"])),
+        _p(li["orderkey"]), _p(li["suppkey"]), _p(lpk),
+        _p(li["extendedprice"]), _p(li["discount"]),
+        C.c_int64(len(supp["suppkey"])), _p(supp["nationkey"]),
+        C.c_int64(len(ptype)), _p(ptype), br, tt)
+    return list(br), list(tt)
+
+
+OracleLib.gen_part_type = gen_part_type
+OracleLib.gen_lineitem_partkey = gen_lineitem_partkey
+OracleLib.q8 = q8

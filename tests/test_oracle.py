@@ -321,3 +321,21 @@ def test_q4_sf1_golden(oracle_lib, sf1):
     for k, g in enumerate(golden):
         assert names[k] == g[0]
         assert counts[k] == int(g[1])
+
+
+def test_q8_sf1_golden(oracle_lib):
+    """Q8 market share — pins the part-type stream + partkey replay.
+    Golden prints share rounded to 4 decimals (HALF_UP)."""
+    li = oracle_lib.gen_lineitem2(1.0)
+    orders = oracle_lib.gen_orders(1.0)
+    cust = oracle_lib.gen_customer2(1.0)
+    supp = oracle_lib.gen_supplier(1.0)
+    ptype = oracle_lib.gen_part_type(1.0)
+    lpk = oracle_lib.gen_lineitem_partkey(1.0)
+    br, tt = oracle_lib.q8(cust, orders, li, lpk, supp, ptype)
+    golden = _parse_golden("q08_sf1.result")
+    for y, g in enumerate(golden):
+        assert int(g[0]) == 1995 + y
+        share = (Decimal(br[y]) / Decimal(tt[y])).quantize(
+            Decimal("0.0001"))
+        assert share == Decimal(g[1]), (br[y], tt[y])
