@@ -508,6 +508,171 @@ def q7(cust: Page, orders: Page, supp: Page, li: Page):
     return out
 
 
+AMERICA_NATIONS = (1, 2, 3, 17, 24)  # tpch_nation_region(k) == 1 (AMERICA)
+BRAZIL = 2
+
+
+def q8(cust: Page, orders: Page, supp: Page, part: Page, li: Page):
+    """Q8 national market share (q08.sql): revenue of 'ECONOMY ANODIZED
+    STEEL' (type id 103) parts sold to AMERICA-region customers with
+    orderdate in 1995..1996, split into BRAZIL(2)-supplier vs total per
+    order year.  Composed from the §8 operators only:
+
+      nation keys -> key set; customer semijoin filter -> custkey flag set
+      orders chained build (date range + customer semijoin, payload
+        orderdate)                      [HashBuilderOperator analog]
+      part(type==103) -> partkey flag set; supplier(BRAZIL) -> flag set
+      lineitem semijoin(part) filter -> emit join (payload orderdate)
+        -> brazil semijoin filter + 4 keyless year aggregations
+    Returns (brazil_1e4[2], total_1e4[2]) exact ticks for (1995, 1996);
+    share = brazil/total (golden q08_sf1.result: 0.0344 / 0.0415)."""
+    import numpy as np
+    from .engine import lib
+
+    # American nation keys as a tiny key set
+    natp = Page({"nationkey": np.asarray(AMERICA_NATIONS, dtype=np.int64)})
+    bn = PlanHashBuild()
+    bn.key_col = 0
+    bn.semijoin_table = -1
+    bn.capacity_hint = 32
+    bn.key_set_only = 1
+    on = Operator(OP_HASH_BUILD, bn)
+    on.add_input(natp)
+    on.finish()
+
+    # AMERICA customers -> dense custkey flag set
+    fc = PlanFilterProject()
+    fc.n_proj = 1
+    fc.proj[0] = Proj(PROJ_IDENT, cust.channel("custkey"), 0, 0)
+    fc.semijoin_table = on.table()
+    fc.semijoin_col = cust.channel("nationkey")
+    f1 = Operator(OP_FILTER_PROJECT, fc)
+    f1.add_input(cust)
+    cpage = f1.get_output_raw()
+
+    bc = PlanHashBuild()
+    bc.key_col = 0
+    bc.semijoin_table = -1
+    bc.capacity_hint = cust.n_rows  # custkeys dense 1..n
+    bc.key_set_only = 1
+    bc.dense_array = 1
+    oc = Operator(OP_HASH_BUILD, bc)
+    oc.add_input_raw(cpage)
+    oc.finish()
+
+    # qualifying orders: chained table keyed by orderkey, payload orderdate
+    bo = PlanHashBuild()
+    bo.n_preds = 2
+    bo.preds[0] = Pred(orders.channel("orderdate"), CMP_GE, 9131, 0.0)
+    bo.preds[1] = Pred(orders.channel("orderdate"), CMP_LE, 9861, 0.0)
+    bo.key_col = orders.channel("orderkey")
+    bo.semijoin_table = oc.table()
+    bo.semijoin_col = orders.channel("custkey")
+    bo.n_payload = 1
+    bo.payload_col[0] = orders.channel("orderdate")
+    bo.capacity_hint = max(orders.n_rows // 8, 16)
+    oo = Operator(OP_HASH_BUILD, bo)
+    oo.add_input(orders)
+    oo.finish()
+
+    # part type 103 -> dense partkey flag set
+    bp = PlanHashBuild()
+    bp.n_preds = 1
+    bp.preds[0] = Pred(part.channel("type_id"), CMP_EQ, 103, 0.0)
+    bp.key_col = part.channel("partkey")
+    bp.semijoin_table = -1
+    bp.capacity_hint = part.n_rows
+    bp.key_set_only = 1
+    bp.dense_array = 1
+    opart = Operator(OP_HASH_BUILD, bp)
+    opart.add_input(part)
+    opart.finish()
+
+    # BRAZIL suppliers -> dense suppkey flag set
+    bb = PlanHashBuild()
+    bb.n_preds = 1
+    bb.preds[0] = Pred(supp.channel("nationkey"), CMP_EQ, BRAZIL, 0.0)
+    bb.key_col = supp.channel("suppkey")
+    bb.semijoin_table = -1
+    bb.capacity_hint = supp.n_rows
+    bb.key_set_only = 1
+    bb.dense_array = 1
+    ob = Operator(OP_HASH_BUILD, bb)
+    ob.add_input(supp)
+    ob.finish()
+
+    # lineitem: keep type-103 parts, project the join/agg columns
+    fl = PlanFilterProject()
+    fl.n_proj = 4
+    fl.proj[0] = Proj(PROJ_IDENT, li.channel("orderkey"), 0, 0)
+    fl.proj[1] = Proj(PROJ_IDENT, li.channel("suppkey"), 0, 0)
+    fl.proj[2] = Proj(PROJ_IDENT, li.channel("extendedprice"), 0, 0)
+    fl.proj[3] = Proj(PROJ_IDENT, li.channel("discount"), 0, 0)
+    fl.semijoin_table = opart.table()
+    fl.semijoin_col = li.channel("partkey")
+    f2 = Operator(OP_FILTER_PROJECT, fl)
+    f2.add_input(li)
+    lpage = f2.get_output_raw()  # [orderkey, suppkey, ep, dc]
+
+    # emit join against qualifying orders; payload appends orderdate
+    jp = PlanLookupJoin()
+    jp.table = oo.table()
+    jp.key_col = 0
+    jp.mode = 0
+    jp.n_emit = 3
+    jp.emit_probe_cols[0] = 1
+    jp.emit_probe_cols[1] = 2
+    jp.emit_probe_cols[2] = 3
+    jo = Operator(OP_LOOKUP_JOIN, jp)
+    jo.add_input_raw(lpage)
+    jpage = jo.get_output_raw()  # [suppkey, ep, dc, orderdate]
+
+    # brazil-supplier subset
+    fb = PlanFilterProject()
+    fb.n_proj = 3
+    fb.proj[0] = Proj(PROJ_IDENT, 1, 0, 0)
+    fb.proj[1] = Proj(PROJ_IDENT, 2, 0, 0)
+    fb.proj[2] = Proj(PROJ_IDENT, 3, 0, 0)
+    fb.semijoin_table = ob.table()
+    fb.semijoin_col = 0
+    f3 = Operator(OP_FILTER_PROJECT, fb)
+    f3.add_input_raw(jpage)
+    bpage = f3.get_output_raw()  # [ep, dc, orderdate]
+
+    def rev(page, odcol, epcol, dccol, lo, hi):
+        p = PlanHashAggSmall()
+        p.n_preds = 2
+        p.preds[0] = Pred(odcol, CMP_GE, lo, 0.0)
+        p.preds[1] = Pred(odcol, CMP_LE, hi, 0.0)
+        p.n_keys = 0
+        p.n_aggs = 1
+        p.aggs[0] = Agg(AGG_SUM_DEC,
+                        Proj(PROJ_DISC_PRICE, epcol, dccol, 0), 4)
+        a = Operator(OP_HASH_AGG_SMALL, p)
+        try:
+            a.add_input_raw(page)
+            a.finish()
+            r = a.get_output(["hi", "lo"])
+            if not len(r["lo"]):
+                return 0
+            return (int(r["hi"][0]) << 64) | int(np.uint64(r["lo"][0]))
+        finally:
+            a.destroy()
+
+    YEARS = ((9131, 9495), (9496, 9861))  # 1995, 1996 epoch-day ranges
+    total = [rev(jpage, 3, 1, 2, lo, hi) for lo, hi in YEARS]
+    brazil = [rev(bpage, 2, 0, 1, lo, hi) for lo, hi in YEARS]
+
+    f3.destroy()
+    jo.destroy()
+    f2.destroy()
+    f1.destroy()
+    for o in (on, oc, oo, opart, ob):
+        lib().c.pg_table_destroy(o.table())
+        o.destroy()
+    return brazil, total
+
+
 def q4(orders: Page, li_dates: Page):
     """Q4 order-priority checking (q04.sql): EXISTS(lineitem with
     commitdate < receiptdate) as a key-set build with a col-vs-col

@@ -886,6 +886,33 @@ def test_q4_exact(P, oracle_lib, sf01):
     assert got == exp
 
 
+def test_q8_exact(P, oracle_lib):
+    """Q8 (national market share: two-year split, part-type + region +
+    supplier-nation constraints) — composed pipeline vs the golden-pinned
+    oracle, exact ticks for both the BRAZIL and total legs."""
+    import numpy as np
+    sf = 0.1
+    li = oracle_lib.gen_lineitem2(sf)
+    lpk = oracle_lib.gen_lineitem_partkey(sf)
+    orders = oracle_lib.gen_orders(sf)
+    cust = oracle_lib.gen_customer2(sf)
+    supp = oracle_lib.gen_supplier(sf)
+    ptype = oracle_lib.gen_part_type(sf)
+    got_br, got_tt = P.pipelines.q8(
+        P.Page({"custkey": cust["custkey"], "nationkey": cust["nationkey"]}),
+        P.Page({k: orders[k] for k in ("orderkey", "custkey", "orderdate")}),
+        P.Page({"suppkey": supp["suppkey"], "nationkey": supp["nationkey"]}),
+        P.Page({"partkey": np.arange(1, len(ptype) + 1, dtype=np.int64),
+                "type_id": ptype}),
+        P.Page({"orderkey": li["orderkey"], "suppkey": li["suppkey"],
+                "partkey": lpk, "extendedprice": li["extendedprice"],
+                "discount": li["discount"]}))
+    exp_br, exp_tt = oracle_lib.q8(cust, orders, li, lpk, supp, ptype)
+    assert got_br == exp_br
+    assert got_tt == exp_tt
+    assert all(t > 0 for t in got_tt)
+
+
 def test_q5_distributed_graph_world1(P, oracle_lib):
     """The distributed Q5 graph at world==1 (replicated dimensions +
     partition/exchange identities + fused probe + exact tick combine)
