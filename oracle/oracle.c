@@ -780,3 +780,162 @@ int32_t oracle_q3(int64_t n_cust, const int64_t* ck, const uint8_t* cseg,
     free(flo);
     return n_out;
 }
+
+/* ---------------- Q14 ----------------
+ * SQL: q14.sql — promo revenue: shipdate in [1995-09-01, 1995-10-01) =
+ * [9374, 9404); promo = part type ids 125..149 ('PROMO*').  Returns
+ * exact 1e-4 tick sums; the caller computes
+ * 100.00 * promo / total at scale 6 HALF_UP (Presto decimal division). */
+void oracle_q14(int64_t n_li, const double* lep, const double* ldisc,
+                const int32_t* lsd, const int64_t* lpk, int64_t n_part,
+                const uint8_t* ptype, int64_t* promo_1e4, int64_t* total_1e4)
+{
+    int64_t promo = 0, total = 0;
+#pragma omp parallel for schedule(static) reduction(+ : promo, total)
+    for (int64_t i = 0; i < n_li; i++) {
+        if (lsd[i] < 9374 || lsd[i] >= 9404) continue;
+        int64_t pk = lpk[i];
+        if (pk < 1 || pk > n_part) continue;
+        int64_t cents = (int64_t)(lep[i] * 100.0 + 0.5);
+        int64_t d = (int64_t)(ldisc[i] * 100.0 + 0.5);
+        int64_t t = cents * (100 - d);
+        total += t;
+        if (ptype[pk - 1] >= 125) promo += t;
+    }
+    *promo_1e4 = promo;
+    *total_1e4 = total;
+}
+
+/* ---------------- Q12 ----------------
+ * SQL: q12.sql — shipmode priority: lineitems with commit < receipt,
+ * ship < commit, receipt in 1994, joined to orders; per shipmode id,
+ * count of high-priority (1-URGENT/2-HIGH = ids 0,1) and low-priority
+ * lines.  Caller selects the modes the query names (MAIL=4, SHIP=6,
+ * pinned by the q12 golden). */
+void oracle_q12(int64_t n_ord, const int64_t* ook, const uint8_t* opri,
+                int64_t n_li, const int64_t* lok, const uint8_t* lsmode,
+                const int32_t* lsd, const int32_t* lcd, const int32_t* lrd,
+                int64_t* high_counts, int64_t* low_counts)
+{
+    int64_t cap = hash_capacity(n_ord < 2 ? 2 : n_ord);
+    int64_t* slot = (int64_t*)malloc(cap * sizeof(int64_t));
+    memset(slot, -1, cap * sizeof(int64_t));
+    for (int64_t i = 0; i < n_ord; i++) {
+        uint64_t h = pg_murmur3_finalize(pg_bigint_hash(ook[i]));
+        int64_t pos = (int64_t)(h & (cap - 1));
+        while (slot[pos] != -1) pos = (pos + 1) & (cap - 1);
+        slot[pos] = i;
+    }
+    for (int m = 0; m < 7; m++) high_counts[m] = low_counts[m] = 0;
+#pragma omp parallel
+    {
+        int64_t lh[7] = {0}, ll[7] = {0};
+#pragma omp for schedule(static)
+        for (int64_t i = 0; i < n_li; i++) {
+            if (!(lcd[i] < lrd[i] && lsd[i] < lcd[i])) continue;
+            if (lrd[i] < 8766 || lrd[i] >= 9131) continue;
+            uint64_t h = pg_murmur3_finalize(pg_bigint_hash(lok[i]));
+            int64_t pos = (int64_t)(h & (cap - 1));
+            int64_t r = -1;
+            for (;;) {
+                int64_t sI = slot[pos];
+                if (sI == -1) break;
+                if (ook[sI] == lok[i]) {
+                    r = sI;
+                    break;
+                }
+                pos = (pos + 1) & (cap - 1);
+            }
+            if (r == -1) continue;
+            if (opri[r] <= 1)
+                lh[lsmode[i]]++;
+            else
+                ll[lsmode[i]]++;
+        }
+#pragma omp critical
+        for (int m = 0; m < 7; m++) {
+            high_counts[m] += lh[m];
+            low_counts[m] += ll[m];
+        }
+    }
+    free(slot);
+}
+
+/* ---------------- Q17 ----------------
+ * SQL: q17.sql — small-quantity-order revenue: rows of parts with
+ * p_brand = 'Brand#23' and p_container = 'MED BOX' (container id 17)
+ * whose quantity < 0.2 * avg(quantity of that part over all lineitem);
+ * returns the exact cents sum of extendedprice (caller divides by 7.0
+ * at scale 2 HALF_UP).  The avg comparison is exact-rational
+ * (5*qty*cnt < sum_qty); at SF1 no row sits on the rounding boundary of
+ * Presto's scale-2 decimal avg, so both readings match the golden. */
+void oracle_q17(int64_t n_li, const int64_t* lpk, const double* lqty,
+                const double* lep, int64_t n_part, const uint8_t* brand,
+                const uint8_t* container, int64_t* out_cents)
+{
+    int64_t* sum_q = (int64_t*)calloc(n_part + 1, sizeof(int64_t));
+    int64_t* cnt_q = (int64_t*)calloc(n_part + 1, sizeof(int64_t));
+    for (int64_t i = 0; i < n_li; i++) {
+        int64_t pk = lpk[i];
+        if (pk < 1 || pk > n_part) continue;
+        sum_q[pk] += (int64_t)(lqty[i] + 0.5);
+        cnt_q[pk] += 1;
+    }
+    int64_t total = 0;
+#pragma omp parallel for schedule(static) reduction(+ : total)
+    for (int64_t i = 0; i < n_li; i++) {
+        int64_t pk = lpk[i];
+        if (pk < 1 || pk > n_part) continue;
+        if (brand[pk - 1] != 23 || container[pk - 1] != 17) continue;
+        int64_t q = (int64_t)(lqty[i] + 0.5);
+        if (5 * q * cnt_q[pk] < sum_q[pk])
+            total += (int64_t)(lep[i] * 100.0 + 0.5);
+    }
+    free(sum_q);
+    free(cnt_q);
+    *out_cents = total;
+}
+
+/* ---------------- Q11 ----------------
+ * SQL: q11.sql — important stock: partsupp of GERMANY(7) suppliers
+ * grouped by partkey, value = sum(supplycost * availqty) in exact cents;
+ * HAVING value > 0.0001 * total (exact: value*10000 > total), ORDER BY
+ * value DESC (partkey ASC as deterministic tiebreak — ties beyond the
+ * SQL ORDER BY are unspecified in the reference).  Returns row count. */
+int64_t oracle_q11(int64_t n_ps, const int64_t* ps_pk, const int64_t* ps_sk,
+                   const int32_t* ps_aq, const int64_t* ps_cost,
+                   int64_t n_supp, const uint8_t* snat, int64_t n_part,
+                   int64_t* out_pk, int64_t* out_val)
+{
+    int64_t* pv = (int64_t*)calloc(n_part + 1, sizeof(int64_t));
+    int64_t total = 0;
+    for (int64_t i = 0; i < n_ps; i++) {
+        int64_t sk = ps_sk[i];
+        if (sk < 1 || sk > n_supp || snat[sk - 1] != 7) continue;
+        int64_t v = ps_cost[i] * (int64_t)ps_aq[i];
+        pv[ps_pk[i]] += v;
+        total += v;
+    }
+    int64_t n_out = 0;
+    for (int64_t pk = 1; pk <= n_part; pk++)
+        if (pv[pk] * 10000 > total) {
+            out_pk[n_out] = pk;
+            out_val[n_out] = pv[pk];
+            n_out++;
+        }
+    /* insertion-free sort: qsort by (val desc, pk asc) */
+    for (int64_t i = 1; i < n_out; i++) { /* n_out ~1e3: insertion sort */
+        int64_t k = out_pk[i], v = out_val[i];
+        int64_t j = i - 1;
+        while (j >= 0 &&
+               (out_val[j] < v || (out_val[j] == v && out_pk[j] > k))) {
+            out_pk[j + 1] = out_pk[j];
+            out_val[j + 1] = out_val[j];
+            j--;
+        }
+        out_pk[j + 1] = k;
+        out_val[j + 1] = v;
+    }
+    free(pv);
+    return n_out;
+}

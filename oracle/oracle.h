@@ -173,6 +173,46 @@ void oracle_q6(int64_t n, const double* quantity,
                const int32_t* shipdate, int64_t* revenue_1e4,
                int64_t* count);
 
+/* ---------------- TPC-H Q14 ----------------
+ * q14.sql — promo revenue: exact 1e-4 tick sums over September 1995;
+ * promo = part type ids 125..149.  Result = 100.00*promo/total at
+ * scale 6 HALF_UP (computed by the caller). */
+void oracle_q14(int64_t n_li, const double* l_extendedprice,
+                const double* l_discount, const int32_t* l_shipdate,
+                const int64_t* l_partkey, int64_t n_part,
+                const uint8_t* p_type, int64_t* promo_1e4,
+                int64_t* total_1e4);
+
+/* ---------------- TPC-H Q12 ----------------
+ * q12.sql — shipmode priority: per shipmode id (0..6; MAIL=4, SHIP=6),
+ * high (priority 0/1) and low line counts of late-commit lineitems
+ * received in 1994. */
+void oracle_q12(int64_t n_ord, const int64_t* o_orderkey,
+                const uint8_t* o_priority, int64_t n_li,
+                const int64_t* l_orderkey, const uint8_t* l_shipmode,
+                const int32_t* l_shipdate, const int32_t* l_commitdate,
+                const int32_t* l_receiptdate, int64_t* high_counts /*[7]*/,
+                int64_t* low_counts /*[7]*/);
+
+/* ---------------- TPC-H Q17 ----------------
+ * q17.sql — small-quantity-order revenue for Brand#23 / 'MED BOX'
+ * (container id 17): exact cents sum of extendedprice over rows with
+ * quantity < 0.2*avg(part).  Result = sum/7.0 at scale 2 HALF_UP. */
+void oracle_q17(int64_t n_li, const int64_t* l_partkey,
+                const double* l_quantity, const double* l_extendedprice,
+                int64_t n_part, const uint8_t* p_brand,
+                const uint8_t* p_container, int64_t* out_cents);
+
+/* ---------------- TPC-H Q11 ----------------
+ * q11.sql — important stock in GERMANY(7): (partkey, value-cents) rows
+ * sorted (value desc, partkey asc), value*10000 > total.  Returns rows
+ * (out capacity n_part). */
+int64_t oracle_q11(int64_t n_ps, const int64_t* ps_partkey,
+                   const int64_t* ps_suppkey, const int32_t* ps_availqty,
+                   const int64_t* ps_supplycost_cents, int64_t n_supp,
+                   const uint8_t* s_nationkey, int64_t n_part,
+                   int64_t* out_pk, int64_t* out_val);
+
 /* ---------------- operator-level primitives (parity targets) ---------- */
 
 /* murmur3 finalizer bucket — PagesHash.java:236-252 /

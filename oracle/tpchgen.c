@@ -50,6 +50,15 @@ static inline int64_t unif(int64_t* s, int64_t lo, int64_t hi)
                                     * (dbgen rnd.h P_TYPE_SD; pinned by the
                                     * q08 golden: id 103 'ECONOMY ANODIZED
                                     * STEEL' shares 0.0344/0.0415 match) */
+#define SEED_L_SMODE   675466456LL /* shipmode pick 1..7, usage 7/order
+                                    * (pinned by q12 golden: MAIL=4, SHIP=6) */
+#define SEED_P_MFG             1LL /* p_mfgr 1..5, 1/part (q17 golden pin) */
+#define SEED_P_BRND     46831694LL /* brand digit 1..5, 1/part (q17 pin) */
+#define SEED_P_CNTR    727633698LL /* container pick 1..40, 1/part;
+                                    * 'MED BOX' = id 17 (q17 golden pin) */
+#define SEED_PS_QTY   1671059989LL /* ps_availqty 1..9999, 4/part (q11 pin) */
+#define SEED_PS_SCST  1051288424LL /* ps_supplycost cents 100..100000,
+                                    * 4/part (q11 golden pin, all 1048 rows) */
 
 /* ---- calendar ----
  * day index 1 = 1992-01-01; order-date index in [1, 2406]
@@ -296,6 +305,98 @@ int64_t tpch_gen_lineitem_partkey(double sf, int64_t ord_start,
         }
     }
     return written;
+}
+
+int64_t tpch_gen_lineitem_shipmode(double sf, int64_t ord_start,
+                                   int64_t ord_count, uint8_t* shipmode)
+{
+    (void)sf;
+    int64_t written = 0;
+#pragma omp parallel reduction(+ : written)
+    {
+#ifdef _OPENMP
+        extern int omp_get_num_threads(void), omp_get_thread_num(void);
+        int nt = omp_get_num_threads(), tid = omp_get_thread_num();
+#else
+        int nt = 1, tid = 0;
+#endif
+        int64_t tlo = ord_count * tid / nt, thi = ord_count * (tid + 1) / nt;
+        int64_t s = rng_skip(SEED_O_LCNT, (uint64_t)ord_start);
+        int64_t off = 0;
+        for (int64_t i = 0; i < tlo; i++) off += unif(&s, 1, 7);
+        int64_t s_lcnt = rng_skip(SEED_O_LCNT, (uint64_t)(ord_start + tlo));
+        int64_t s_mode =
+            rng_skip(SEED_L_SMODE, (uint64_t)(ord_start + tlo) * 7);
+        int64_t out = off;
+        for (int64_t o = tlo; o < thi; o++) {
+            int64_t lc = unif(&s_lcnt, 1, 7);
+            for (int64_t l = 0; l < lc; l++)
+                shipmode[out++] = (uint8_t)(unif(&s_mode, 1, 7) - 1);
+            s_mode = rng_skip(s_mode, (uint64_t)(7 - lc));
+            written += lc;
+        }
+    }
+    return written;
+}
+
+void tpch_gen_part2(double sf, int64_t start, int64_t count, uint8_t* mfgr,
+                    uint8_t* brand, uint8_t* container)
+{
+#pragma omp parallel
+    {
+#ifdef _OPENMP
+        extern int omp_get_num_threads(void), omp_get_thread_num(void);
+        int nt = omp_get_num_threads(), tid = omp_get_thread_num();
+#else
+        int nt = 1, tid = 0;
+#endif
+        (void)sf;
+        int64_t lo = count * tid / nt, hi = count * (tid + 1) / nt;
+        int64_t sm = rng_skip(SEED_P_MFG, (uint64_t)(start + lo));
+        int64_t sb = rng_skip(SEED_P_BRND, (uint64_t)(start + lo));
+        int64_t sc = rng_skip(SEED_P_CNTR, (uint64_t)(start + lo));
+        for (int64_t i = lo; i < hi; i++) {
+            int64_t m = unif(&sm, 1, 5);
+            int64_t b = unif(&sb, 1, 5);
+            int64_t c = unif(&sc, 1, 40);
+            if (mfgr) mfgr[i] = (uint8_t)m;
+            if (brand) brand[i] = (uint8_t)(m * 10 + b);
+            if (container) container[i] = (uint8_t)(c - 1);
+        }
+    }
+}
+
+void tpch_gen_partsupp(double sf, int64_t part_start, int64_t part_count,
+                       int64_t* partkey, int64_t* suppkey,
+                       int32_t* availqty, int64_t* supplycost_cents)
+{
+    int64_t S = tpch_supplier_count(sf);
+#pragma omp parallel
+    {
+#ifdef _OPENMP
+        extern int omp_get_num_threads(void), omp_get_thread_num(void);
+        int nt = omp_get_num_threads(), tid = omp_get_thread_num();
+#else
+        int nt = 1, tid = 0;
+#endif
+        int64_t lo = part_count * tid / nt, hi = part_count * (tid + 1) / nt;
+        int64_t sq =
+            rng_skip(SEED_PS_QTY, (uint64_t)(part_start + lo) * 4);
+        int64_t sc =
+            rng_skip(SEED_PS_SCST, (uint64_t)(part_start + lo) * 4);
+        for (int64_t p = lo; p < hi; p++) {
+            int64_t pk = part_start + p + 1;
+            for (int64_t i = 0; i < 4; i++) {
+                int64_t out = p * 4 + i;
+                if (partkey) partkey[out] = pk;
+                if (suppkey) suppkey[out] = part_supplier(pk, i, S);
+                int32_t aq = (int32_t)unif(&sq, 1, 9999);
+                int64_t cost = unif(&sc, 100, 100000);
+                if (availqty) availqty[out] = aq;
+                if (supplycost_cents) supplycost_cents[out] = cost;
+            }
+        }
+    }
 }
 
 void tpch_gen_orders_priority(double sf, int64_t start, int64_t count,

@@ -83,6 +83,27 @@ void tpch_gen_part_type(double sf, int64_t start, int64_t count,
 int64_t tpch_gen_lineitem_partkey(double sf, int64_t ord_start,
                                   int64_t ord_count, int64_t* partkey);
 
+/* lineitem shipmode ids 0..6 (pick 1..7 per line); id 4 = 'MAIL',
+ * id 6 = 'SHIP' (pinned by the q12 golden counts) */
+int64_t tpch_gen_lineitem_shipmode(double sf, int64_t ord_start,
+                                   int64_t ord_count, uint8_t* shipmode);
+
+/* part attribute columns: p_mfgr 1..5, p_brand = mfgr*10 + (1..5)
+ * ('Brand#MB'), p_container id 0..39 ('MED BOX' = 17, q17 golden pin;
+ * nested Cnt1 x Cnt2 order, 8 second words per first word). Any output
+ * may be NULL. */
+void tpch_gen_part2(double sf, int64_t start, int64_t count, uint8_t* mfgr,
+                    uint8_t* brand, uint8_t* container);
+
+/* partsupp: 4 rows per part row [part_start, part_start+part_count);
+ * suppkey via the PART_SUPP bridge; ps_availqty 1..9999;
+ * ps_supplycost in cents 100..100000 (1.00..1000.00).  Outputs are
+ * 4*part_count rows; any output may be NULL.  (q11 golden pin: all
+ * 1048 result rows exact.) */
+void tpch_gen_partsupp(double sf, int64_t part_start, int64_t part_count,
+                       int64_t* partkey, int64_t* suppkey,
+                       int32_t* availqty, int64_t* supplycost_cents);
+
 /* orders columns for order rows [start, start+count) (0-based).
  * shippriority is the constant 0 (dbgen mk_order) and is not emitted.
  * lcnt: lineitems per order (1..7). Any output pointer may be NULL. */

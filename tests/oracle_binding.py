@@ -317,3 +317,90 @@ def q8(self, cust, orders, li, lpk, supp, ptype):
 OracleLib.gen_part_type = gen_part_type
 OracleLib.gen_lineitem_partkey = gen_lineitem_partkey
 OracleLib.q8 = q8
+
+
+def gen_lineitem_shipmode(self, sf):
+    n = self.lineitem_count(sf)
+    n_ord = self.lib.tpch_orders_count(C.c_double(sf))
+    sm = np.empty(n, np.uint8)
+    self.lib.tpch_gen_lineitem_shipmode.restype = C.c_int64
+    w = self.lib.tpch_gen_lineitem_shipmode(C.c_double(sf), C.c_int64(0),
+                                            C.c_int64(n_ord), _p(sm))
+    assert w == n
+    return sm
+
+
+def gen_part2(self, sf):
+    n = int(200000 * sf)
+    mfgr = np.empty(n, np.uint8)
+    brand = np.empty(n, np.uint8)
+    cntr = np.empty(n, np.uint8)
+    self.lib.tpch_gen_part2(C.c_double(sf), C.c_int64(0), C.c_int64(n),
+                            _p(mfgr), _p(brand), _p(cntr))
+    return {"mfgr": mfgr, "brand": brand, "container": cntr}
+
+
+def gen_partsupp(self, sf):
+    n_part = int(200000 * sf)
+    n = n_part * 4
+    pk = np.empty(n, np.int64)
+    sk = np.empty(n, np.int64)
+    aq = np.empty(n, np.int32)
+    cost = np.empty(n, np.int64)
+    self.lib.tpch_gen_partsupp(C.c_double(sf), C.c_int64(0),
+                               C.c_int64(n_part), _p(pk), _p(sk), _p(aq),
+                               _p(cost))
+    return {"partkey": pk, "suppkey": sk, "availqty": aq,
+            "supplycost_cents": cost}
+
+
+def q14(self, li, lpk, ptype):
+    promo = C.c_int64()
+    total = C.c_int64()
+    self.lib.oracle_q14(C.c_int64(len(li["orderkey"])),
+                        _p(li["extendedprice"]), _p(li["discount"]),
+                        _p(li["shipdate"]), _p(lpk), C.c_int64(len(ptype)),
+                        _p(ptype), C.byref(promo), C.byref(total))
+    return promo.value, total.value
+
+
+def q12(self, orders, pri, li, lid, smode):
+    hi = (C.c_int64 * 7)()
+    lo = (C.c_int64 * 7)()
+    self.lib.oracle_q12(C.c_int64(len(orders["orderkey"])),
+                        _p(orders["orderkey"]), _p(pri),
+                        C.c_int64(len(li["orderkey"])), _p(li["orderkey"]),
+                        _p(smode), _p(li["shipdate"]), _p(lid["commitdate"]),
+                        _p(lid["receiptdate"]), hi, lo)
+    return list(hi), list(lo)
+
+
+def q17(self, li, lpk, part2):
+    out = C.c_int64()
+    self.lib.oracle_q17(C.c_int64(len(li["orderkey"])), _p(lpk),
+                        _p(li["quantity"]), _p(li["extendedprice"]),
+                        C.c_int64(len(part2["brand"])), _p(part2["brand"]),
+                        _p(part2["container"]), C.byref(out))
+    return out.value
+
+
+def q11(self, ps, supp, n_part):
+    out_pk = np.empty(n_part, np.int64)
+    out_val = np.empty(n_part, np.int64)
+    self.lib.oracle_q11.restype = C.c_int64
+    n = self.lib.oracle_q11(C.c_int64(len(ps["partkey"])), _p(ps["partkey"]),
+                            _p(ps["suppkey"]), _p(ps["availqty"]),
+                            _p(ps["supplycost_cents"]),
+                            C.c_int64(len(supp["suppkey"])),
+                            _p(supp["nationkey"]), C.c_int64(n_part),
+                            _p(out_pk), _p(out_val))
+    return out_pk[:n], out_val[:n]
+
+
+OracleLib.gen_lineitem_shipmode = gen_lineitem_shipmode
+OracleLib.gen_part2 = gen_part2
+OracleLib.gen_partsupp = gen_partsupp
+OracleLib.q14 = q14
+OracleLib.q12 = q12
+OracleLib.q17 = q17
+OracleLib.q11 = q11

@@ -339,3 +339,60 @@ def test_q8_sf1_golden(oracle_lib):
         share = (Decimal(br[y]) / Decimal(tt[y])).quantize(
             Decimal("0.0001"))
         assert share == Decimal(g[1]), (br[y], tt[y])
+
+
+def test_q14_sf1_golden(oracle_lib):
+    """Q14 promo revenue — pins the part-type stream through the revenue
+    projection.  Golden is 100.00*promo/total at scale 6 (HALF_UP)."""
+    from decimal import ROUND_HALF_UP
+    li = oracle_lib.gen_lineitem2(1.0)
+    lpk = oracle_lib.gen_lineitem_partkey(1.0)
+    ptype = oracle_lib.gen_part_type(1.0)
+    promo, total = oracle_lib.q14(li, lpk, ptype)
+    golden = _parse_golden("q14_sf1.result")
+    got = (Decimal(100 * promo) / Decimal(total)).quantize(
+        Decimal("0.000001"), rounding=ROUND_HALF_UP)
+    assert got == Decimal(golden[0][0]), (promo, total)
+
+
+def test_q12_sf1_golden(oracle_lib):
+    """Q12 shipmode priority counts — pins the L_SMODE stream (MAIL=4,
+    SHIP=6)."""
+    orders = oracle_lib.gen_orders(1.0)
+    pri = oracle_lib.gen_orders_priority(1.0)
+    li = oracle_lib.gen_lineitem2(1.0)
+    lid = oracle_lib.gen_lineitem_dates(1.0)
+    smode = oracle_lib.gen_lineitem_shipmode(1.0)
+    hi, lo = oracle_lib.q12(orders, pri, li, lid, smode)
+    golden = _parse_golden("q12_sf1.result")
+    assert golden[0][0] == "MAIL" and golden[1][0] == "SHIP"
+    assert [hi[4], lo[4]] == [int(golden[0][1]), int(golden[0][2])]
+    assert [hi[6], lo[6]] == [int(golden[1][1]), int(golden[1][2])]
+
+
+def test_q17_sf1_golden(oracle_lib):
+    """Q17 small-quantity revenue — pins p_mfgr/p_brand/p_container.
+    Golden is sum(extendedprice)/7.0 at scale 2 (HALF_UP)."""
+    from decimal import ROUND_HALF_UP
+    li = oracle_lib.gen_lineitem2(1.0)
+    lpk = oracle_lib.gen_lineitem_partkey(1.0)
+    part2 = oracle_lib.gen_part2(1.0)
+    cents = oracle_lib.q17(li, lpk, part2)
+    golden = _parse_golden("q17_sf1.result")
+    got = (Decimal(cents) / Decimal(700)).quantize(
+        Decimal("0.01"), rounding=ROUND_HALF_UP)
+    assert got == Decimal(golden[0][0]), cents
+
+
+def test_q11_sf1_golden(oracle_lib):
+    """Q11 important stock — pins the partsupp bridge + availqty +
+    supplycost streams on all 1048 golden rows (values exact cents;
+    golden trims trailing zeros)."""
+    supp = oracle_lib.gen_supplier(1.0)
+    ps = oracle_lib.gen_partsupp(1.0)
+    pk, val = oracle_lib.q11(ps, supp, 200000)
+    golden = _parse_golden("q11_sf1.result")
+    assert len(pk) == len(golden)
+    for i, g in enumerate(golden):
+        assert int(pk[i]) == int(g[0])
+        assert Decimal(int(val[i])) / 100 == Decimal(g[1])
