@@ -673,6 +673,356 @@ def q8(cust: Page, orders: Page, supp: Page, part: Page, li: Page):
     return brazil, total
 
 
+def q14(part: Page, li: Page):
+    """Q14 promo revenue (q14.sql): shipdate in [1995-09-01, 1995-10-01) =
+    [9374, 9404); promo parts = type ids 125..149 ('PROMO*') as a dense
+    partkey flag set.  Returns (promo_1e4, total_1e4) exact ticks; the
+    result is 100.00*promo/total at scale 6 HALF_UP."""
+    import numpy as np
+    from .engine import lib
+
+    bp = PlanHashBuild()
+    bp.n_preds = 1
+    bp.preds[0] = Pred(part.channel("type_id"), CMP_GE, 125, 0.0)
+    bp.key_col = part.channel("partkey")
+    bp.semijoin_table = -1
+    bp.capacity_hint = part.n_rows
+    bp.key_set_only = 1
+    bp.dense_array = 1
+    ob = Operator(OP_HASH_BUILD, bp)
+    ob.add_input(part)
+    ob.finish()
+
+    def rev(page, semijoin, epc, dcc, sdc, pkc):
+        p = PlanHashAggSmall()
+        p.n_preds = 2
+        p.preds[0] = Pred(sdc, CMP_GE, 9374, 0.0)
+        p.preds[1] = Pred(sdc, CMP_LT, 9404, 0.0)
+        p.n_keys = 0
+        p.n_aggs = 1
+        p.aggs[0] = Agg(AGG_SUM_DEC, Proj(PROJ_DISC_PRICE, epc, dcc, 0), 4)
+        if semijoin:
+            fp = PlanFilterProject()
+            fp.n_preds = 2
+            fp.preds[0] = Pred(sdc, CMP_GE, 9374, 0.0)
+            fp.preds[1] = Pred(sdc, CMP_LT, 9404, 0.0)
+            fp.n_proj = 2
+            fp.proj[0] = Proj(PROJ_IDENT, epc, 0, 0)
+            fp.proj[1] = Proj(PROJ_IDENT, dcc, 0, 0)
+            fp.semijoin_table = ob.table()
+            fp.semijoin_col = pkc
+            f = Operator(OP_FILTER_PROJECT, fp)
+            f.add_input(page)
+            fpage = f.get_output_raw()
+            p.n_preds = 0
+            p.aggs[0] = Agg(AGG_SUM_DEC, Proj(PROJ_DISC_PRICE, 0, 1, 0), 4)
+            a = Operator(OP_HASH_AGG_SMALL, p)
+            a.add_input_raw(fpage)
+            a.finish()
+            r = a.get_output(["hi", "lo"])
+            a.destroy()
+            f.destroy()
+        else:
+            a = Operator(OP_HASH_AGG_SMALL, p)
+            a.add_input(page)
+            a.finish()
+            r = a.get_output(["hi", "lo"])
+            a.destroy()
+        if not len(r["lo"]):
+            return 0
+        return (int(r["hi"][0]) << 64) | int(np.uint64(r["lo"][0]))
+
+    epc, dcc = li.channel("extendedprice"), li.channel("discount")
+    sdc, pkc = li.channel("shipdate"), li.channel("partkey")
+    total = rev(li, False, epc, dcc, sdc, pkc)
+    promo = rev(li, True, epc, dcc, sdc, pkc)
+    lib().c.pg_table_destroy(ob.table())
+    ob.destroy()
+    return promo, total
+
+
+def q12(orders: Page, li: Page):
+    """Q12 shipmode priority (q12.sql): late-commit lineitems received in
+    1994 joined to orders; counts per (shipmode in {MAIL=4, SHIP=6},
+    priority class).  Returns {mode_id: (high, low)}."""
+    bo = PlanHashBuild()
+    bo.key_col = orders.channel("orderkey")
+    bo.semijoin_table = -1
+    bo.n_payload = 1
+    bo.payload_col[0] = orders.channel("priority")
+    bo.capacity_hint = orders.n_rows
+    oo = Operator(OP_HASH_BUILD, bo)
+    oo.add_input(orders)
+    oo.finish()
+
+    cdc, rdc = li.channel("commitdate"), li.channel("receiptdate")
+    sdc = li.channel("shipdate")
+    fl = PlanFilterProject()
+    fl.n_preds = 4
+    p0 = Pred(cdc, CMP_LT, 0, 0.0)
+    p0.rhs_col = rdc + 1
+    fl.preds[0] = p0
+    p1 = Pred(sdc, CMP_LT, 0, 0.0)
+    p1.rhs_col = cdc + 1
+    fl.preds[1] = p1
+    fl.preds[2] = Pred(rdc, CMP_GE, 8766, 0.0)
+    fl.preds[3] = Pred(rdc, CMP_LT, 9131, 0.0)
+    fl.n_proj = 2
+    fl.proj[0] = Proj(PROJ_IDENT, li.channel("orderkey"), 0, 0)
+    fl.proj[1] = Proj(PROJ_IDENT, li.channel("shipmode"), 0, 0)
+    f = Operator(OP_FILTER_PROJECT, fl)
+    f.add_input(li)
+    fpage = f.get_output_raw()  # [orderkey, shipmode]
+
+    jp = PlanLookupJoin()
+    jp.table = oo.table()
+    jp.key_col = 0
+    jp.mode = 0
+    jp.n_emit = 1
+    jp.emit_probe_cols[0] = 1
+    jo = Operator(OP_LOOKUP_JOIN, jp)
+    jo.add_input_raw(fpage)
+    jpage = jo.get_output_raw()  # [shipmode, priority]
+
+    ag = PlanHashAggSmall()
+    ag.n_keys = 2
+    ag.key_col[0] = 0
+    ag.key_col[1] = 1
+    ag.n_vals[0] = 2
+    ag.key_vals[0][0] = 4  # MAIL
+    ag.key_vals[0][1] = 6  # SHIP
+    ag.n_vals[1] = 5
+    for i in range(5):
+        ag.key_vals[1][i] = i
+    ag.drop_unlisted_keys = 1
+    ag.n_aggs = 1
+    ag.aggs[0] = Agg(AGG_COUNT, Proj(PROJ_IDENT, 0, 0, 0), 0)
+    a = Operator(OP_HASH_AGG_SMALL, ag)
+    a.add_input_raw(jpage)
+    a.finish()
+    out = a.get_output(["shipmode", "priority", "count"])
+    a.destroy()
+    jo.destroy()
+    f.destroy()
+    from .engine import lib
+    lib().c.pg_table_destroy(oo.table())
+    oo.destroy()
+    res = {4: [0, 0], 6: [0, 0]}
+    for i in range(len(out["shipmode"])):
+        m = int(out["shipmode"][i])
+        cls = 0 if int(out["priority"][i]) <= 1 else 1
+        res[m][cls] += int(out["count"][i])
+    return {m: tuple(v) for m, v in res.items()}
+
+
+def q17(part: Page, li: Page):
+    """Q17 small-quantity-order revenue (q17.sql): Brand#23 'MED BOX'
+    (container id 17) parts; rows with quantity < 0.2*avg(part quantity).
+    The correlated avg is a fused-agg probe (sum_qty + count per target
+    part); the tiny per-part cutoffs (0.2*avg <= 10) come back to the
+    host (coordinator-side scalar-subquery plan constants, ~200 rows) and
+    drive per-cutoff-class flag-set semijoins.  Returns the exact cents
+    sum of extendedprice (result = cents/7.0 at scale 2 HALF_UP)."""
+    import numpy as np
+    from .engine import lib
+    pkc = part.channel("partkey")
+    tables = []
+
+    def target_preds(plan):
+        plan.n_preds = 2
+        plan.preds[0] = Pred(part.channel("brand"), CMP_EQ, 23, 0.0)
+        plan.preds[1] = Pred(part.channel("container"), CMP_EQ, 17, 0.0)
+
+    bt = PlanHashBuild()
+    target_preds(bt)
+    bt.key_col = pkc
+    bt.semijoin_table = -1
+    bt.capacity_hint = max(part.n_rows // 16, 4096)
+    bt.agg_table = 1
+    ot = Operator(OP_HASH_BUILD, bt)
+    ot.add_input(part)
+    ot.finish()
+    tables.append(ot)
+
+    bs = PlanHashBuild()
+    target_preds(bs)
+    bs.key_col = pkc
+    bs.semijoin_table = -1
+    bs.capacity_hint = part.n_rows
+    bs.key_set_only = 1
+    bs.dense_array = 1
+    os_ = Operator(OP_HASH_BUILD, bs)
+    os_.add_input(part)
+    os_.finish()
+    tables.append(os_)
+
+    # per-target-part sum(quantity) + count over ALL lineitems
+    jp = PlanLookupJoin()
+    jp.table = ot.table()
+    jp.key_col = li.channel("partkey")
+    jp.mode = 1
+    jp.proj = Proj(PROJ_IDENT, li.channel("quantity"), 0, 0)
+    jp.dec_scale = 0
+    jo = Operator(OP_LOOKUP_JOIN, jp)
+    jo.add_input(li)
+    jo.finish()
+    g = jo.get_output(["partkey", "sum_qty", "sum_f64", "cnt"])
+    jo.destroy()
+
+    # cutoff class per part: qty < sum/(5*cnt)  =>  qty <= c_p
+    classes = {}
+    for i in range(len(g["partkey"])):
+        s, c = int(g["sum_qty"][i]), int(g["cnt"][i])
+        cp = (s + 5 * c - 1) // (5 * c) - 1  # max qty with 5*qty*c < s
+        if cp >= 1:
+            classes.setdefault(cp, []).append(int(g["partkey"][i]))
+
+    # li restricted to target parts (small page)
+    fl = PlanFilterProject()
+    fl.n_proj = 3
+    fl.proj[0] = Proj(PROJ_IDENT, li.channel("partkey"), 0, 0)
+    fl.proj[1] = Proj(PROJ_IDENT, li.channel("quantity"), 0, 0)
+    fl.proj[2] = Proj(PROJ_IDENT, li.channel("extendedprice"), 0, 0)
+    fl.semijoin_table = os_.table()
+    fl.semijoin_col = li.channel("partkey")
+    f = Operator(OP_FILTER_PROJECT, fl)
+    f.add_input(li)
+    fpage = f.get_output_raw()  # [partkey, qty, ep]
+
+    total = 0
+    for cp, pks in sorted(classes.items()):
+        cpage = Page({"partkey": np.asarray(pks, dtype=np.int64)})
+        bc = PlanHashBuild()
+        bc.key_col = 0
+        bc.semijoin_table = -1
+        bc.capacity_hint = part.n_rows
+        bc.key_set_only = 1
+        bc.dense_array = 1
+        oc = Operator(OP_HASH_BUILD, bc)
+        oc.add_input(cpage)
+        oc.finish()
+
+        fc = PlanFilterProject()
+        fc.n_preds = 1
+        fc.preds[0] = Pred(1, CMP_LE, 0, float(cp))
+        fc.n_proj = 1
+        fc.proj[0] = Proj(PROJ_IDENT, 2, 0, 0)
+        fc.semijoin_table = oc.table()
+        fc.semijoin_col = 0
+        ff = Operator(OP_FILTER_PROJECT, fc)
+        ff.add_input_raw(fpage)
+        cpage2 = ff.get_output_raw()
+
+        ap = PlanHashAggSmall()
+        ap.n_keys = 0
+        ap.n_aggs = 1
+        ap.aggs[0] = Agg(AGG_SUM_DEC, Proj(PROJ_IDENT, 0, 0, 0), 2)
+        a = Operator(OP_HASH_AGG_SMALL, ap)
+        a.add_input_raw(cpage2)
+        a.finish()
+        r = a.get_output(["hi", "lo"])
+        if len(r["lo"]):
+            total += (int(r["hi"][0]) << 64) | int(np.uint64(r["lo"][0]))
+        a.destroy()
+        ff.destroy()
+        lib().c.pg_table_destroy(oc.table())
+        oc.destroy()
+
+    f.destroy()
+    for o in tables:
+        lib().c.pg_table_destroy(o.table())
+        o.destroy()
+    return total
+
+
+def q11(supp: Page, ps: Page, n_part: int):
+    """Q11 important stock (q11.sql): GERMANY(7) partsupp grouped by
+    partkey, value = sum(supplycost*availqty); HAVING value >
+    0.0001*total; ORDER BY value DESC (partkey ASC tiebreak).  The MUL
+    decimal projection yields 1e-4 ticks (cents x hundredth-encoded
+    qty); the strict HAVING threshold total/10000 becomes a plan
+    constant.  Returns (partkeys, value_cents) host arrays; the ~1k-row
+    final ORDER BY runs host-side (output-stage sort)."""
+    import numpy as np
+    from .engine import lib
+
+    bg = PlanHashBuild()
+    bg.n_preds = 1
+    bg.preds[0] = Pred(supp.channel("nationkey"), CMP_EQ, 7, 0.0)
+    bg.key_col = supp.channel("suppkey")
+    bg.semijoin_table = -1
+    bg.capacity_hint = supp.n_rows
+    bg.key_set_only = 1
+    bg.dense_array = 1
+    og = Operator(OP_HASH_BUILD, bg)
+    og.add_input(supp)
+    og.finish()
+
+    fl = PlanFilterProject()
+    fl.n_proj = 3
+    fl.proj[0] = Proj(PROJ_IDENT, ps.channel("partkey"), 0, 0)
+    fl.proj[1] = Proj(PROJ_IDENT, ps.channel("supplycost"), 0, 0)
+    fl.proj[2] = Proj(PROJ_IDENT, ps.channel("availqty"), 0, 0)
+    fl.semijoin_table = og.table()
+    fl.semijoin_col = ps.channel("suppkey")
+    f = Operator(OP_FILTER_PROJECT, fl)
+    f.add_input(ps)
+    gpage = f.get_output_raw()  # [partkey, cost, qty]
+
+    ta = PlanHashAggSmall()
+    ta.n_keys = 0
+    ta.n_aggs = 1
+    ta.aggs[0] = Agg(AGG_SUM_DEC, Proj(PROJ_MUL, 1, 2, 0), 4)
+    a = Operator(OP_HASH_AGG_SMALL, ta)
+    a.add_input_raw(gpage)
+    a.finish()
+    r = a.get_output(["hi", "lo"])
+    total_1e4 = (int(r["hi"][0]) << 64) | int(np.uint64(r["lo"][0]))
+    a.destroy()
+
+    keys = Page({"partkey": np.arange(1, n_part + 1, dtype=np.int64)})
+    bk = PlanHashBuild()
+    bk.key_col = 0
+    bk.semijoin_table = -1
+    bk.capacity_hint = n_part
+    bk.agg_table = 1
+    ok = Operator(OP_HASH_BUILD, bk)
+    ok.add_input(keys)
+    ok.finish()
+
+    jp = PlanLookupJoin()
+    jp.table = ok.table()
+    jp.key_col = 0
+    jp.mode = 1
+    jp.proj = Proj(PROJ_MUL, 1, 2, 0)
+    jp.dec_scale = 4
+    jo = Operator(OP_LOOKUP_JOIN, jp)
+    jo.add_input_raw(gpage)
+    jo.finish()
+    groups = jo.get_output_raw()  # [partkey, sum_1e4, f64, cnt]
+
+    ft = PlanFilterProject()
+    ft.n_preds = 1
+    ft.preds[0] = Pred(1, CMP_GT, total_1e4 // 10000, 0.0)
+    ft.n_proj = 2
+    ft.proj[0] = Proj(PROJ_IDENT, 0, 0, 0)
+    ft.proj[1] = Proj(PROJ_IDENT, 1, 0, 0)
+    ft.semijoin_table = 0
+    fo = Operator(OP_FILTER_PROJECT, ft)
+    fo.add_input_raw(groups)
+    out = fo.get_output(["partkey", "value_1e4"])
+    fo.destroy()
+    jo.destroy()
+    f.destroy()
+    for o in (og, ok):
+        lib().c.pg_table_destroy(o.table())
+        o.destroy()
+    pk = out["partkey"]
+    val = out["value_1e4"] // 100  # exact: ticks are cents*100
+    order = np.lexsort((pk, -val))
+    return pk[order], val[order]
+
+
 def q4(orders: Page, li_dates: Page):
     """Q4 order-priority checking (q04.sql): EXISTS(lineitem with
     commitdate < receiptdate) as a key-set build with a col-vs-col

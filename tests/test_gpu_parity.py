@@ -913,6 +913,83 @@ def test_q8_exact(P, oracle_lib):
     assert all(t > 0 for t in got_tt)
 
 
+def test_q14_exact(P, oracle_lib):
+    """Q14 promo revenue — dense type-flag semijoin + keyless aggs vs
+    the golden-pinned oracle, exact ticks both legs."""
+    import numpy as np
+    sf = 0.1
+    li = oracle_lib.gen_lineitem2(sf)
+    lpk = oracle_lib.gen_lineitem_partkey(sf)
+    ptype = oracle_lib.gen_part_type(sf)
+    got_p, got_t = P.pipelines.q14(
+        P.Page({"partkey": np.arange(1, len(ptype) + 1, dtype=np.int64),
+                "type_id": ptype}),
+        P.Page({"partkey": lpk, "extendedprice": li["extendedprice"],
+                "discount": li["discount"], "shipdate": li["shipdate"]}))
+    exp_p, exp_t = oracle_lib.q14(li, lpk, ptype)
+    assert (got_p, got_t) == (exp_p, exp_t)
+    assert got_t > 0
+
+
+def test_q12_exact(P, oracle_lib):
+    """Q12 shipmode priority counts — col-vs-col filter + payload emit
+    join + 2-key aggregation vs the golden-pinned oracle."""
+    sf = 0.1
+    orders = oracle_lib.gen_orders(sf)
+    pri = oracle_lib.gen_orders_priority(sf)
+    li = oracle_lib.gen_lineitem2(sf)
+    lid = oracle_lib.gen_lineitem_dates(sf)
+    smode = oracle_lib.gen_lineitem_shipmode(sf)
+    got = P.pipelines.q12(
+        P.Page({"orderkey": orders["orderkey"], "priority": pri}),
+        P.Page({"orderkey": li["orderkey"], "shipmode": smode,
+                "shipdate": li["shipdate"], "commitdate": lid["commitdate"],
+                "receiptdate": lid["receiptdate"]}))
+    hi, lo = oracle_lib.q12(orders, pri, li, lid, smode)
+    assert got == {4: (hi[4], lo[4]), 6: (hi[6], lo[6])}
+    assert sum(got[4]) > 0
+
+
+def test_q17_exact(P, oracle_lib):
+    """Q17 correlated-avg revenue — fused-agg probe for the per-part avg
+    + cutoff-class flag-set semijoins vs the golden-pinned oracle."""
+    import numpy as np
+    sf = 0.1
+    li = oracle_lib.gen_lineitem2(sf)
+    lpk = oracle_lib.gen_lineitem_partkey(sf)
+    part2 = oracle_lib.gen_part2(sf)
+    n = len(part2["brand"])
+    got = P.pipelines.q17(
+        P.Page({"partkey": np.arange(1, n + 1, dtype=np.int64),
+                "brand": part2["brand"], "container": part2["container"]}),
+        P.Page({"partkey": lpk, "quantity": li["quantity"],
+                "extendedprice": li["extendedprice"]}))
+    exp = oracle_lib.q17(li, lpk, part2)
+    assert got == exp
+    assert got > 0
+
+
+def test_q11_exact(P, oracle_lib):
+    """Q11 important stock — MUL decimal projection, fused group-by
+    probe, strict HAVING threshold vs the golden-pinned oracle, all
+    rows + order."""
+    import numpy as np
+    sf = 0.1
+    supp = oracle_lib.gen_supplier(sf)
+    ps = oracle_lib.gen_partsupp(sf)
+    n_part = int(200000 * sf)
+    got_pk, got_val = P.pipelines.q11(
+        P.Page({"suppkey": supp["suppkey"], "nationkey": supp["nationkey"]}),
+        P.Page({"partkey": ps["partkey"], "suppkey": ps["suppkey"],
+                "supplycost": ps["supplycost_cents"] / 100.0,
+                "availqty": ps["availqty"].astype(np.float64)}),
+        n_part)
+    exp_pk, exp_val = oracle_lib.q11(ps, supp, n_part)
+    assert list(got_pk) == list(exp_pk)
+    assert [v * 100 for v in got_val.tolist()] == list(exp_val)
+    assert len(got_pk) > 0
+
+
 def test_q5_distributed_graph_world1(P, oracle_lib):
     """The distributed Q5 graph at world==1 (replicated dimensions +
     partition/exchange identities + fused probe + exact tick combine)
