@@ -939,3 +939,79 @@ int64_t oracle_q11(int64_t n_ps, const int64_t* ps_pk, const int64_t* ps_sk,
     free(pv);
     return n_out;
 }
+
+/* ---------------- Q18 ----------------
+ * SQL: q18.sql — large-volume customers: orders whose lineitem quantity
+ * sum exceeds 300; emits (custkey, orderkey, orderdate,
+ * totalprice_cents, sum_qty) sorted by (totalprice desc, orderdate asc,
+ * orderkey asc — deterministic final tiebreak) LIMIT limit.  c_name is
+ * the deterministic 'Customer#%09d' of custkey (formatted by callers).
+ * Returns rows written. */
+int64_t oracle_q18(int64_t n_ord, const int64_t* ook, const int64_t* ock,
+                   const int32_t* od, const int64_t* otp, int64_t n_li,
+                   const int64_t* lok, const double* lqty, int32_t limit,
+                   int64_t* out_ck, int64_t* out_ok, int32_t* out_od,
+                   int64_t* out_tp, int64_t* out_qty)
+{
+    int64_t cap = hash_capacity(n_ord < 2 ? 2 : n_ord);
+    int64_t* slot = (int64_t*)malloc(cap * sizeof(int64_t));
+    memset(slot, -1, cap * sizeof(int64_t));
+    for (int64_t i = 0; i < n_ord; i++) {
+        uint64_t h = pg_murmur3_finalize(pg_bigint_hash(ook[i]));
+        int64_t pos = (int64_t)(h & (cap - 1));
+        while (slot[pos] != -1) pos = (pos + 1) & (cap - 1);
+        slot[pos] = i;
+    }
+    int64_t* qsum = (int64_t*)calloc(n_ord, sizeof(int64_t));
+    for (int64_t i = 0; i < n_li; i++) {
+        uint64_t h = pg_murmur3_finalize(pg_bigint_hash(lok[i]));
+        int64_t pos = (int64_t)(h & (cap - 1));
+        for (;;) {
+            int64_t sI = slot[pos];
+            if (sI == -1) break;
+            if (ook[sI] == lok[i]) {
+                qsum[sI] += (int64_t)(lqty[i] + 0.5);
+                break;
+            }
+            pos = (pos + 1) & (cap - 1);
+        }
+    }
+    int64_t n_out = 0;
+    for (int64_t i = 0; i < n_ord; i++) {
+        if (qsum[i] <= 300) continue;
+        /* insertion into the bounded (tp desc, od asc, ok asc) list */
+        int64_t j;
+        if (n_out == limit) {
+            /* only enter when strictly better than the current last */
+            if (!(otp[i] > out_tp[limit - 1] ||
+                  (otp[i] == out_tp[limit - 1] &&
+                   (od[i] < out_od[limit - 1] ||
+                    (od[i] == out_od[limit - 1] &&
+                     ook[i] < out_ok[limit - 1])))))
+                continue;
+            j = limit - 1;
+        } else {
+            j = n_out++;
+        }
+        while (j > 0 &&
+               (out_tp[j - 1] < otp[i] ||
+                (out_tp[j - 1] == otp[i] &&
+                 (out_od[j - 1] > od[i] ||
+                  (out_od[j - 1] == od[i] && out_ok[j - 1] > ook[i]))))) {
+            out_ck[j] = out_ck[j - 1];
+            out_ok[j] = out_ok[j - 1];
+            out_od[j] = out_od[j - 1];
+            out_tp[j] = out_tp[j - 1];
+            out_qty[j] = out_qty[j - 1];
+            j--;
+        }
+        out_ck[j] = ock[i];
+        out_ok[j] = ook[i];
+        out_od[j] = od[i];
+        out_tp[j] = otp[i];
+        out_qty[j] = qsum[i];
+    }
+    free(slot);
+    free(qsum);
+    return n_out;
+}

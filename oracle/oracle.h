@@ -213,6 +213,18 @@ int64_t oracle_q11(int64_t n_ps, const int64_t* ps_partkey,
                    const uint8_t* s_nationkey, int64_t n_part,
                    int64_t* out_pk, int64_t* out_val);
 
+/* ---------------- TPC-H Q18 ----------------
+ * q18.sql — large-volume customers: orders with sum(l_quantity) > 300;
+ * rows (custkey, orderkey, orderdate, totalprice_cents, sum_qty) sorted
+ * (totalprice desc, orderdate asc, orderkey asc) LIMIT limit.  Returns
+ * rows written. */
+int64_t oracle_q18(int64_t n_ord, const int64_t* o_orderkey,
+                   const int64_t* o_custkey, const int32_t* o_orderdate,
+                   const int64_t* o_totalprice_cents, int64_t n_li,
+                   const int64_t* l_orderkey, const double* l_quantity,
+                   int32_t limit, int64_t* out_ck, int64_t* out_ok,
+                   int32_t* out_od, int64_t* out_tp, int64_t* out_qty);
+
 /* ---------------- operator-level primitives (parity targets) ---------- */
 
 /* murmur3 finalizer bucket — PagesHash.java:236-252 /

@@ -571,6 +571,49 @@ static int64_t gen_lineitem_chunk(double sf, int64_t ord_start,
     return out;
 }
 
+void tpch_gen_orders_totalprice(double sf, int64_t ord_start,
+                                int64_t ord_count, int64_t* totalprice_cents)
+{
+    /* dbgen mk_order: totalprice = sum over lines of
+     *   eprice*(100-disc)/100*(100+tax)/100   (integer floor divisions)
+     * (pinned by the q18 golden o_totalprice column) */
+    int64_t max_pkey = (int64_t)(200000.0 * sf + 0.5);
+#pragma omp parallel
+    {
+#ifdef _OPENMP
+        extern int omp_get_num_threads(void), omp_get_thread_num(void);
+        int nt = omp_get_num_threads(), tid = omp_get_thread_num();
+#else
+        int nt = 1, tid = 0;
+#endif
+        int64_t lo = ord_count * tid / nt, hi = ord_count * (tid + 1) / nt;
+        int64_t s_lcnt = rng_skip(SEED_O_LCNT, (uint64_t)(ord_start + lo));
+        uint64_t l7 = (uint64_t)(ord_start + lo) * 7;
+        int64_t s_qty = rng_skip(SEED_L_QTY, l7);
+        int64_t s_disc = rng_skip(SEED_L_DISC, l7);
+        int64_t s_tax = rng_skip(SEED_L_TAX, l7);
+        int64_t s_pkey = rng_skip(SEED_L_PKEY, l7);
+        for (int64_t o = lo; o < hi; o++) {
+            int64_t lc = unif(&s_lcnt, 1, 7);
+            int64_t tp = 0;
+            for (int64_t l = 0; l < lc; l++) {
+                int64_t qty = unif(&s_qty, 1, 50);
+                int64_t d = unif(&s_disc, 0, 10);
+                int64_t t = unif(&s_tax, 0, 8);
+                int64_t pk = unif(&s_pkey, 1, max_pkey);
+                int64_t cents = qty * part_price_cents(pk);
+                tp += cents * (100 - d) / 100 * (100 + t) / 100;
+            }
+            totalprice_cents[o] = tp;
+            uint64_t rest = (uint64_t)(7 - lc);
+            s_qty = rng_skip(s_qty, rest);
+            s_disc = rng_skip(s_disc, rest);
+            s_tax = rng_skip(s_tax, rest);
+            s_pkey = rng_skip(s_pkey, rest);
+        }
+    }
+}
+
 int64_t tpch_gen_lineitem(double sf, int64_t ord_start, int64_t ord_count,
                           int64_t* orderkey, double* quantity,
                           double* extendedprice, double* discount, double* tax,

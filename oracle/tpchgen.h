@@ -104,6 +104,12 @@ void tpch_gen_partsupp(double sf, int64_t part_start, int64_t part_count,
                        int64_t* partkey, int64_t* suppkey,
                        int32_t* availqty, int64_t* supplycost_cents);
 
+/* o_totalprice in exact cents (dbgen mk_order floor-div chain over the
+ * order's lineitems; pinned by the q18 golden) */
+void tpch_gen_orders_totalprice(double sf, int64_t ord_start,
+                                int64_t ord_count,
+                                int64_t* totalprice_cents);
+
 /* orders columns for order rows [start, start+count) (0-based).
  * shippriority is the constant 0 (dbgen mk_order) and is not emitted.
  * lcnt: lineitems per order (1..7). Any output pointer may be NULL. */

@@ -784,34 +784,31 @@ def q12(orders: Page, li: Page):
     jo.add_input_raw(fpage)
     jpage = jo.get_output_raw()  # [shipmode, priority]
 
-    ag = PlanHashAggSmall()
-    ag.n_keys = 2
-    ag.key_col[0] = 0
-    ag.key_col[1] = 1
-    ag.n_vals[0] = 2
-    ag.key_vals[0][0] = 4  # MAIL
-    ag.key_vals[0][1] = 6  # SHIP
-    ag.n_vals[1] = 5
-    for i in range(5):
-        ag.key_vals[1][i] = i
-    ag.drop_unlisted_keys = 1
-    ag.n_aggs = 1
-    ag.aggs[0] = Agg(AGG_COUNT, Proj(PROJ_IDENT, 0, 0, 0), 0)
-    a = Operator(OP_HASH_AGG_SMALL, ag)
-    a.add_input_raw(jpage)
-    a.finish()
-    out = a.get_output(["shipmode", "priority", "count"])
-    a.destroy()
+    res = {4: [0, 0], 6: [0, 0]}
+    for cls, (op_, val) in enumerate(((CMP_LE, 1), (CMP_GE, 2))):
+        ag = PlanHashAggSmall()
+        ag.n_preds = 1
+        ag.preds[0] = Pred(1, op_, val, 0.0)  # priority class split
+        ag.n_keys = 1
+        ag.key_col[0] = 0
+        ag.n_vals[0] = 2
+        ag.key_vals[0][0] = 4  # MAIL
+        ag.key_vals[0][1] = 6  # SHIP
+        ag.drop_unlisted_keys = 1
+        ag.n_aggs = 1
+        ag.aggs[0] = Agg(AGG_COUNT, Proj(PROJ_IDENT, 0, 0, 0), 0)
+        a = Operator(OP_HASH_AGG_SMALL, ag)
+        a.add_input_raw(jpage)
+        a.finish()
+        out = a.get_output(["shipmode", "count"])
+        a.destroy()
+        for i in range(len(out["shipmode"])):
+            res[int(out["shipmode"][i])][cls] += int(out["count"][i])
     jo.destroy()
     f.destroy()
     from .engine import lib
     lib().c.pg_table_destroy(oo.table())
     oo.destroy()
-    res = {4: [0, 0], 6: [0, 0]}
-    for i in range(len(out["shipmode"])):
-        m = int(out["shipmode"][i])
-        cls = 0 if int(out["priority"][i]) <= 1 else 1
-        res[m][cls] += int(out["count"][i])
     return {m: tuple(v) for m, v in res.items()}
 
 

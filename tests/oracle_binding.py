@@ -404,3 +404,33 @@ OracleLib.q14 = q14
 OracleLib.q12 = q12
 OracleLib.q17 = q17
 OracleLib.q11 = q11
+
+
+def gen_orders_totalprice(self, sf):
+    n = self.lib.tpch_orders_count(C.c_double(sf))
+    tp = np.empty(n, np.int64)
+    self.lib.tpch_gen_orders_totalprice(C.c_double(sf), C.c_int64(0),
+                                        C.c_int64(n), _p(tp))
+    return tp
+
+
+def q18(self, orders, tp, li, limit=100):
+    n_ord = len(orders["orderkey"])
+    ck = np.empty(limit, np.int64)
+    ok = np.empty(limit, np.int64)
+    od = np.empty(limit, np.int32)
+    otp = np.empty(limit, np.int64)
+    qt = np.empty(limit, np.int64)
+    self.lib.oracle_q18.restype = C.c_int64
+    n = self.lib.oracle_q18(C.c_int64(n_ord), _p(orders["orderkey"]),
+                            _p(orders["custkey"]), _p(orders["orderdate"]),
+                            _p(tp), C.c_int64(len(li["orderkey"])),
+                            _p(li["orderkey"]), _p(li["quantity"]),
+                            C.c_int32(limit), _p(ck), _p(ok), _p(od),
+                            _p(otp), _p(qt))
+    return [(int(ck[i]), int(ok[i]), int(od[i]), int(otp[i]), int(qt[i]))
+            for i in range(n)]
+
+
+OracleLib.gen_orders_totalprice = gen_orders_totalprice
+OracleLib.q18 = q18

@@ -396,3 +396,24 @@ def test_q11_sf1_golden(oracle_lib):
     for i, g in enumerate(golden):
         assert int(pk[i]) == int(g[0])
         assert Decimal(int(val[i])) / 100 == Decimal(g[1])
+
+
+def test_q18_sf1_golden(oracle_lib):
+    """Q18 large-volume customers — pins the o_totalprice floor-div
+    derivation over all golden rows (names are the deterministic
+    Customer#%09d of custkey)."""
+    import datetime
+    orders = oracle_lib.gen_orders(1.0)
+    tp = oracle_lib.gen_orders_totalprice(1.0)
+    li = oracle_lib.gen_lineitem(1.0)
+    rows = oracle_lib.q18(orders, tp, li)
+    golden = _parse_golden("q18_sf1.result")
+    assert len(rows) == len(golden)
+    epoch = datetime.date(1970, 1, 1)
+    for r, g in enumerate(golden):
+        ck, ok, od, otp, qt = rows[r]
+        assert f"Customer#{ck:09d}" == g[0]
+        assert (ck, ok) == (int(g[1]), int(g[2]))
+        assert (epoch + datetime.timedelta(days=od)).isoformat() == g[3]
+        assert Decimal(otp) / 100 == Decimal(g[4])
+        assert qt == int(g[5])
