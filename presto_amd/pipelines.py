@@ -1020,6 +1020,83 @@ def q11(supp: Page, ps: Page, n_part: int):
     return pk[order], val[order]
 
 
+def q18(orders: Page, li: Page, limit=100):
+    """Q18 large-volume customers (q18.sql): per-order quantity sums via
+    a fused-agg probe over the full orders key table; HAVING sum > 300
+    as a plan-constant filter; the ~1e-5-selectivity survivors join back
+    to the orders columns (emit join) and the bounded final ORDER BY
+    runs host-side.  Returns [(custkey, orderkey, orderdate,
+    totalprice_cents, sum_qty)] sorted (totalprice desc, orderdate asc,
+    orderkey asc) LIMIT limit."""
+    import numpy as np
+    from .engine import lib
+
+    bo = PlanHashBuild()
+    bo.key_col = orders.channel("orderkey")
+    bo.semijoin_table = -1
+    bo.capacity_hint = orders.n_rows + 64
+    bo.agg_table = 1
+    oo = Operator(OP_HASH_BUILD, bo)
+    oo.add_input(orders)
+    oo.finish()
+
+    jp = PlanLookupJoin()
+    jp.table = oo.table()
+    jp.key_col = li.channel("orderkey")
+    jp.mode = 1
+    jp.proj = Proj(PROJ_IDENT, li.channel("quantity"), 0, 0)
+    jp.dec_scale = 0
+    jo = Operator(OP_LOOKUP_JOIN, jp)
+    jo.add_input(li)
+    jo.finish()
+    groups = jo.get_output_raw()  # [orderkey, sum_qty, f64, cnt]
+
+    ft = PlanFilterProject()
+    ft.n_preds = 1
+    ft.preds[0] = Pred(1, CMP_GT, 300, 0.0)
+    ft.n_proj = 2
+    ft.proj[0] = Proj(PROJ_IDENT, 0, 0, 0)
+    ft.proj[1] = Proj(PROJ_IDENT, 1, 0, 0)
+    fo = Operator(OP_FILTER_PROJECT, ft)
+    fo.add_input_raw(groups)
+    big = fo.get_output_raw()  # [orderkey, sum_qty]
+
+    bb = PlanHashBuild()
+    bb.key_col = 0
+    bb.semijoin_table = -1
+    bb.n_payload = 1
+    bb.payload_col[0] = 1
+    bb.capacity_hint = max(big.n_rows, 16)
+    ob = Operator(OP_HASH_BUILD, bb)
+    ob.add_input_raw(big)
+    ob.finish()
+
+    je = PlanLookupJoin()
+    je.table = ob.table()
+    je.key_col = orders.channel("orderkey")
+    je.mode = 0
+    je.n_emit = 4
+    je.emit_probe_cols[0] = orders.channel("orderkey")
+    je.emit_probe_cols[1] = orders.channel("custkey")
+    je.emit_probe_cols[2] = orders.channel("orderdate")
+    je.emit_probe_cols[3] = orders.channel("totalprice")
+    js = Operator(OP_LOOKUP_JOIN, je)
+    js.add_input(orders)
+    out = js.get_output(["orderkey", "custkey", "orderdate", "totalprice",
+                         "sum_qty"])
+    js.destroy()
+    fo.destroy()
+    jo.destroy()
+    for o in (oo, ob):
+        lib().c.pg_table_destroy(o.table())
+        o.destroy()
+    rows = [(int(out["custkey"][i]), int(out["orderkey"][i]),
+             int(out["orderdate"][i]), int(out["totalprice"][i]),
+             int(out["sum_qty"][i])) for i in range(len(out["orderkey"]))]
+    rows.sort(key=lambda r: (-r[3], r[2], r[1]))
+    return rows[:limit]
+
+
 def q4(orders: Page, li_dates: Page):
     """Q4 order-priority checking (q04.sql): EXISTS(lineitem with
     commitdate < receiptdate) as a key-set build with a col-vs-col

@@ -990,6 +990,24 @@ def test_q11_exact(P, oracle_lib):
     assert len(got_pk) > 0
 
 
+def test_q18_exact(P, oracle_lib):
+    """Q18 large-volume customers — fused-agg quantity sums + HAVING
+    constant + emit join vs the golden-pinned oracle (threshold lowered
+    to keep the small-SF result non-empty is NOT done: at sf0.1 a few
+    orders still exceed 300)."""
+    sf = 0.1
+    orders = oracle_lib.gen_orders(sf)
+    tp = oracle_lib.gen_orders_totalprice(sf)
+    li = oracle_lib.gen_lineitem(sf)
+    got = P.pipelines.q18(
+        P.Page({"orderkey": orders["orderkey"], "custkey": orders["custkey"],
+                "orderdate": orders["orderdate"], "totalprice": tp}),
+        P.Page({"orderkey": li["orderkey"], "quantity": li["quantity"]}))
+    exp = oracle_lib.q18(orders, tp, li)
+    assert got == exp
+    assert len(got) > 0
+
+
 def test_q5_distributed_graph_world1(P, oracle_lib):
     """The distributed Q5 graph at world==1 (replicated dimensions +
     partition/exchange identities + fused probe + exact tick combine)
