@@ -1015,3 +1015,111 @@ int64_t oracle_q18(int64_t n_ord, const int64_t* ook, const int64_t* ock,
     free(qsum);
     return n_out;
 }
+
+/* ---------------- Q21 ----------------
+ * SQL: q21.sql — suppliers who kept orders waiting: SAUDI ARABIA(20)
+ * suppliers' late lines (receipt > commit) in multi-supplier 'F'-status
+ * orders where theirs is the ONLY late supplier.  o_orderstatus is
+ * derived like dbgen mk_order: 'F' iff every line's linestatus is 'F'.
+ * Lineitem arrays must be grouped by orderkey (generator order).
+ * Returns rows (suppkey, numwait) sorted (numwait desc, suppkey asc =
+ * s_name asc, names being 'Supplier#%09d') LIMIT limit. */
+int64_t oracle_q21(int64_t n_supp, const uint8_t* snat, int64_t n_li,
+                   const int64_t* lok, const int64_t* lsk,
+                   const uint8_t* lls, const int32_t* lcd,
+                   const int32_t* lrd, int32_t limit, int64_t* out_sk,
+                   int64_t* out_cnt)
+{
+    int64_t* wait = (int64_t*)calloc(n_supp + 1, sizeof(int64_t));
+    int64_t a = 0;
+    while (a < n_li) {
+        int64_t b = a;
+        while (b < n_li && lok[b] == lok[a]) b++;
+        /* one order's lines: [a, b) (at most 7) */
+        int all_f = 1;
+        for (int64_t i = a; i < b && all_f; i++)
+            if (lls[i] != 'F') all_f = 0;
+        if (all_f) {
+            int64_t first = lsk[a];
+            int multi = 0;
+            for (int64_t i = a + 1; i < b; i++)
+                if (lsk[i] != first) multi = 1;
+            if (multi) {
+                int64_t late_supp = -1;
+                int64_t late_rows = 0;
+                int single = 1;
+                for (int64_t i = a; i < b; i++) {
+                    if (lrd[i] <= lcd[i]) continue;
+                    if (late_supp == -1) late_supp = lsk[i];
+                    else if (lsk[i] != late_supp) single = 0;
+                    late_rows++;
+                }
+                if (late_supp != -1 && single && late_supp >= 1 &&
+                    late_supp <= n_supp && snat[late_supp - 1] == 20)
+                    wait[late_supp] += late_rows;
+            }
+        }
+        a = b;
+    }
+    int64_t n_out = 0;
+    for (int64_t s = 1; s <= n_supp; s++) {
+        if (!wait[s]) continue;
+        int64_t j;
+        if (n_out == limit) {
+            if (!(wait[s] > out_cnt[limit - 1])) continue; /* sk asc ties */
+            j = limit - 1;
+        } else {
+            j = n_out++;
+        }
+        while (j > 0 && out_cnt[j - 1] < wait[s]) {
+            out_sk[j] = out_sk[j - 1];
+            out_cnt[j] = out_cnt[j - 1];
+            j--;
+        }
+        out_sk[j] = s;
+        out_cnt[j] = wait[s];
+    }
+    free(wait);
+    return n_out;
+}
+
+/* ---------------- Q22 ----------------
+ * SQL: q22.sql — global sales opportunity: customers of the 7 named
+ * phone country codes (code = nationkey + 10), acctbal above the
+ * average positive acctbal of that population (exact-rational compare;
+ * no golden-boundary sensitivity at SF1), with no orders.  Returns per
+ * nationkey (caller orders by code string): count + exact cents sum. */
+void oracle_q22(int64_t n_cust, const int64_t* ck, const uint8_t* cnat,
+                const int64_t* abal_cents, int64_t n_ord,
+                const int64_t* ock, int32_t n_codes,
+                const uint8_t* code_nations, int64_t* out_cnt,
+                int64_t* out_sum)
+{
+    int64_t max_ck = 0;
+    for (int64_t i = 0; i < n_cust; i++)
+        if (ck[i] > max_ck) max_ck = ck[i];
+    uint8_t* has_ord = (uint8_t*)calloc(max_ck + 1, 1);
+    for (int64_t i = 0; i < n_ord; i++)
+        if (ock[i] >= 0 && ock[i] <= max_ck) has_ord[ock[i]] = 1;
+    uint8_t in_codes[32] = {0};
+    for (int32_t c = 0; c < n_codes; c++) in_codes[code_nations[c]] = 1;
+    __int128 sum_pos = 0;
+    int64_t cnt_pos = 0;
+    for (int64_t i = 0; i < n_cust; i++)
+        if (in_codes[cnat[i]] && abal_cents[i] > 0) {
+            sum_pos += abal_cents[i];
+            cnt_pos++;
+        }
+    for (int32_t c = 0; c < n_codes; c++) out_cnt[c] = out_sum[c] = 0;
+    for (int64_t i = 0; i < n_cust; i++) {
+        if (!in_codes[cnat[i]] || has_ord[ck[i]]) continue;
+        if ((__int128)abal_cents[i] * cnt_pos <= sum_pos) continue;
+        for (int32_t c = 0; c < n_codes; c++)
+            if (code_nations[c] == cnat[i]) {
+                out_cnt[c]++;
+                out_sum[c] += abal_cents[i];
+                break;
+            }
+    }
+    free(has_ord);
+}

@@ -225,6 +225,27 @@ int64_t oracle_q18(int64_t n_ord, const int64_t* o_orderkey,
                    int32_t limit, int64_t* out_ck, int64_t* out_ok,
                    int32_t* out_od, int64_t* out_tp, int64_t* out_qty);
 
+/* ---------------- TPC-H Q21 ----------------
+ * q21.sql — suppliers who kept orders waiting (SAUDI ARABIA=20): late
+ * lines in multi-supplier all-'F' orders whose only late supplier is
+ * theirs.  Lineitem arrays must be grouped by orderkey.  Rows
+ * (suppkey, numwait) sorted (numwait desc, suppkey asc) LIMIT limit. */
+int64_t oracle_q21(int64_t n_supp, const uint8_t* s_nationkey, int64_t n_li,
+                   const int64_t* l_orderkey, const int64_t* l_suppkey,
+                   const uint8_t* l_linestatus, const int32_t* l_commitdate,
+                   const int32_t* l_receiptdate, int32_t limit,
+                   int64_t* out_sk, int64_t* out_cnt);
+
+/* ---------------- TPC-H Q22 ----------------
+ * q22.sql — global sales opportunity: per phone country code
+ * (= nationkey+10; the 7 codes the query names), count + exact cents
+ * sum of above-average-balance customers with no orders. */
+void oracle_q22(int64_t n_cust, const int64_t* c_custkey,
+                const uint8_t* c_nationkey, const int64_t* c_acctbal_cents,
+                int64_t n_ord, const int64_t* o_custkey, int32_t n_codes,
+                const uint8_t* code_nations, int64_t* out_cnt,
+                int64_t* out_sum);
+
 /* ---------------- operator-level primitives (parity targets) ---------- */
 
 /* murmur3 finalizer bucket — PagesHash.java:236-252 /

@@ -59,6 +59,8 @@ static inline int64_t unif(int64_t* s, int64_t lo, int64_t hi)
 #define SEED_PS_QTY   1671059989LL /* ps_availqty 1..9999, 4/part (q11 pin) */
 #define SEED_PS_SCST  1051288424LL /* ps_supplycost cents 100..100000,
                                     * 4/part (q11 golden pin, all 1048 rows) */
+#define SEED_C_ABAL    298370230LL /* c_acctbal cents -99999..999999,
+                                    * 1/customer (q22 golden pin) */
 
 /* ---- calendar ----
  * day index 1 = 1992-01-01; order-date index in [1, 2406]
@@ -396,6 +398,25 @@ void tpch_gen_partsupp(double sf, int64_t part_start, int64_t part_count,
                 if (supplycost_cents) supplycost_cents[out] = cost;
             }
         }
+    }
+}
+
+void tpch_gen_customer_acctbal(double sf, int64_t start, int64_t count,
+                               int64_t* acctbal_cents)
+{
+#pragma omp parallel
+    {
+#ifdef _OPENMP
+        extern int omp_get_num_threads(void), omp_get_thread_num(void);
+        int nt = omp_get_num_threads(), tid = omp_get_thread_num();
+#else
+        int nt = 1, tid = 0;
+#endif
+        (void)sf;
+        int64_t lo = count * tid / nt, hi = count * (tid + 1) / nt;
+        int64_t s = rng_skip(SEED_C_ABAL, (uint64_t)(start + lo));
+        for (int64_t i = lo; i < hi; i++)
+            acctbal_cents[i] = unif(&s, -99999, 999999);
     }
 }
 

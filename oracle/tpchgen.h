@@ -104,6 +104,10 @@ void tpch_gen_partsupp(double sf, int64_t part_start, int64_t part_count,
                        int64_t* partkey, int64_t* suppkey,
                        int32_t* availqty, int64_t* supplycost_cents);
 
+/* c_acctbal in exact cents, -99999..999999 (q22 golden pin) */
+void tpch_gen_customer_acctbal(double sf, int64_t start, int64_t count,
+                               int64_t* acctbal_cents);
+
 /* o_totalprice in exact cents (dbgen mk_order floor-div chain over the
  * order's lineitems; pinned by the q18 golden) */
 void tpch_gen_orders_totalprice(double sf, int64_t ord_start,

@@ -434,3 +434,46 @@ def q18(self, orders, tp, li, limit=100):
 
 OracleLib.gen_orders_totalprice = gen_orders_totalprice
 OracleLib.q18 = q18
+
+
+def gen_customer_acctbal(self, sf):
+    n = self.lib.tpch_customer_count(C.c_double(sf))
+    ab = np.empty(n, np.int64)
+    self.lib.tpch_gen_customer_acctbal(C.c_double(sf), C.c_int64(0),
+                                       C.c_int64(n), _p(ab))
+    return ab
+
+
+def q21(self, supp, li, lid, limit=100):
+    n_supp = len(supp["suppkey"])
+    out_sk = np.empty(limit, np.int64)
+    out_cnt = np.empty(limit, np.int64)
+    self.lib.oracle_q21.restype = C.c_int64
+    n = self.lib.oracle_q21(C.c_int64(n_supp), _p(supp["nationkey"]),
+                            C.c_int64(len(li["orderkey"])),
+                            _p(li["orderkey"]), _p(li["suppkey"]),
+                            _p(li["linestatus"]), _p(lid["commitdate"]),
+                            _p(lid["receiptdate"]), C.c_int32(limit),
+                            _p(out_sk), _p(out_cnt))
+    return [(int(out_sk[i]), int(out_cnt[i])) for i in range(n)]
+
+
+Q22_CODE_NATIONS = np.array([3, 7, 8, 13, 19, 20, 21], np.uint8)  # codes
+# '13','17','18','23','29','30','31' ascending
+
+
+def q22(self, cust, abal, orders):
+    cn = Q22_CODE_NATIONS
+    out_cnt = np.empty(len(cn), np.int64)
+    out_sum = np.empty(len(cn), np.int64)
+    self.lib.oracle_q22(C.c_int64(len(cust["custkey"])), _p(cust["custkey"]),
+                        _p(cust["nationkey"]), _p(abal),
+                        C.c_int64(len(orders["orderkey"])),
+                        _p(orders["custkey"]), C.c_int32(len(cn)), _p(cn),
+                        _p(out_cnt), _p(out_sum))
+    return out_cnt.tolist(), out_sum.tolist()
+
+
+OracleLib.gen_customer_acctbal = gen_customer_acctbal
+OracleLib.q21 = q21
+OracleLib.q22 = q22

@@ -417,3 +417,33 @@ def test_q18_sf1_golden(oracle_lib):
         assert (epoch + datetime.timedelta(days=od)).isoformat() == g[3]
         assert Decimal(otp) / 100 == Decimal(g[4])
         assert qt == int(g[5])
+
+
+def test_q21_sf1_golden(oracle_lib):
+    """Q21 waiting suppliers — pins the derived o_orderstatus and the
+    only-late-supplier EXISTS/NOT-EXISTS pair on all 100 golden rows."""
+    supp = oracle_lib.gen_supplier(1.0)
+    li = oracle_lib.gen_lineitem2(1.0)
+    lid = oracle_lib.gen_lineitem_dates(1.0)
+    rows = oracle_lib.q21(supp, li, lid)
+    golden = _parse_golden("q21_sf1.result")
+    assert len(rows) == len(golden) == 100
+    for (sk, cnt), g in zip(rows, golden):
+        assert f"Supplier#{sk:09d}" == g[0]
+        assert cnt == int(g[1])
+
+
+def test_q22_sf1_golden(oracle_lib):
+    """Q22 global sales opportunity — pins the c_acctbal stream and the
+    phone-country-code mapping (code = nationkey + 10)."""
+    from tests.oracle_binding import Q22_CODE_NATIONS
+    cust = oracle_lib.gen_customer2(1.0)
+    abal = oracle_lib.gen_customer_acctbal(1.0)
+    orders = oracle_lib.gen_orders(1.0)
+    cnt, tot = oracle_lib.q22(cust, abal, orders)
+    golden = _parse_golden("q22_sf1.result")
+    assert len(golden) == len(Q22_CODE_NATIONS)
+    for i, g in enumerate(golden):
+        assert int(Q22_CODE_NATIONS[i]) + 10 == int(g[0])
+        assert cnt[i] == int(g[1])
+        assert Decimal(tot[i]) / 100 == Decimal(g[2])
