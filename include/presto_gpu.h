@@ -175,6 +175,10 @@ typedef struct {
                                in that key-set table (EXISTS pushdown, e.g.
                                Q4); 0 = unused */
     int32_t semijoin_col;
+    int32_t semijoin_anti;  /* 1: keep rows whose key is NOT in the set
+                               (NOT-EXISTS pushdown, e.g. Q22's customers
+                               without orders — LookupJoinOperator's
+                               probe-side anti join) */
 } pg_plan_filter_project;
 
 typedef struct {

@@ -775,6 +775,31 @@ __device__ inline uint64_t d_ballot(bool p)
     return __ballot(p);
 }
 
+/* key-set membership test (dense flags when set_mask < 0, else hash set);
+ * anti inverts it (NOT-EXISTS pushdown) */
+__device__ inline bool d_semi_ok(const pg_page& pg, int32_t set_col,
+                                 int64_t i, const int64_t* set_keys,
+                                 int64_t set_mask, int32_t anti)
+{
+    int64_t key = d_load_i64(pg.cols[set_col], i);
+    bool in;
+    if (set_mask < 0) { /* dense flags: |mask| = capacity */
+        in = key >= 1 && key <= -set_mask &&
+             ((const uint8_t*)set_keys)[key - 1];
+    } else {
+        in = false;
+        uint64_t h = pg_murmur3_finalize(pg_bigint_hash(key));
+        int64_t s = (int64_t)(h & (uint64_t)set_mask);
+        for (;;) {
+            int64_t k = set_keys[s];
+            if (k == key) { in = true; break; }
+            if (k == TBL_EMPTY) break;
+            s = (s + 1) & set_mask;
+        }
+    }
+    return in != (bool)anti;
+}
+
 __global__ __launch_bounds__(256) void k_sel_count(pg_page pg,
                                                    pg_plan_filter_project plan,
                                                    int64_t chunk,
@@ -791,22 +816,9 @@ __global__ __launch_bounds__(256) void k_sel_count(pg_page pg,
     for (int64_t base = lo + 64 * wid; base < hi; base += 256) {
         int64_t i = base + lane;
         bool sel = i < hi && d_eval_preds(pg, plan.preds, plan.n_preds, i);
-        if (sel && set_keys) {
-            int64_t key = d_load_i64(pg.cols[set_col], i);
-            if (set_mask < 0) { /* dense flags: |mask| = capacity */
-                sel = key >= 1 && key <= -set_mask &&
-                      ((const uint8_t*)set_keys)[key - 1];
-            } else {
-                uint64_t h = pg_murmur3_finalize(pg_bigint_hash(key));
-                int64_t s = (int64_t)(h & (uint64_t)set_mask);
-                for (;;) {
-                    int64_t k = set_keys[s];
-                    if (k == key) break;
-                    if (k == TBL_EMPTY) { sel = false; break; }
-                    s = (s + 1) & set_mask;
-                }
-            }
-        }
+        if (sel && set_keys)
+            sel = d_semi_ok(pg, set_col, i, set_keys, set_mask,
+                            plan.semijoin_anti);
         uint64_t m = d_ballot(sel);
         if (lane == 0) cnt += __popcll(m);
     }
@@ -870,22 +882,9 @@ __global__ __launch_bounds__(256) void k_sel_emit(pg_page pg,
     for (int64_t base = lo; base < hi; base += 256) {
         int64_t i = base + 64 * wid + lane;
         bool sel = i < hi && d_eval_preds(pg, plan.preds, plan.n_preds, i);
-        if (sel && set_keys) {
-            int64_t key = d_load_i64(pg.cols[set_col], i);
-            if (set_mask < 0) {
-                sel = key >= 1 && key <= -set_mask &&
-                      ((const uint8_t*)set_keys)[key - 1];
-            } else {
-                uint64_t h = pg_murmur3_finalize(pg_bigint_hash(key));
-                int64_t s = (int64_t)(h & (uint64_t)set_mask);
-                for (;;) {
-                    int64_t k = set_keys[s];
-                    if (k == key) break;
-                    if (k == TBL_EMPTY) { sel = false; break; }
-                    s = (s + 1) & set_mask;
-                }
-            }
-        }
+        if (sel && set_keys)
+            sel = d_semi_ok(pg, set_col, i, set_keys, set_mask,
+                            plan.semijoin_anti);
         uint64_t m = d_ballot(sel);
         int wsum = __popcll(m);
         if (lane == 0) wcnt[wid] = wsum;

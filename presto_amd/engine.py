@@ -54,7 +54,8 @@ class Agg(C.Structure):
 class PlanFilterProject(C.Structure):
     _fields_ = [("n_preds", C.c_int32), ("preds", Pred * 8),
                 ("n_proj", C.c_int32), ("proj", Proj * 16),
-                ("semijoin_table", C.c_int64), ("semijoin_col", C.c_int32)]
+                ("semijoin_table", C.c_int64), ("semijoin_col", C.c_int32),
+                ("semijoin_anti", C.c_int32)]
 
 
 class PlanHashAggSmall(C.Structure):

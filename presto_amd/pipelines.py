@@ -1097,6 +1097,116 @@ def q18(orders: Page, li: Page, limit=100):
     return rows[:limit]
 
 
+Q22_CODE_NATIONS = (3, 7, 8, 13, 19, 20, 21)  # codes '13'..'31' ascending
+
+
+def q22(cust: Page, orders: Page):
+    """Q22 global sales opportunity (q22.sql): customers of the 7 phone
+    country codes (code = nationkey+10) with above-average positive
+    balance and no orders (anti-semijoin NOT-EXISTS pushdown).  The
+    population average becomes a strict plan constant
+    floor(sum/cnt).  Returns (counts[7], sums_cents[7]) in code order."""
+    import numpy as np
+    from .engine import lib
+
+    natp = Page({"nationkey": np.asarray(Q22_CODE_NATIONS, dtype=np.int64)})
+    bn = PlanHashBuild()
+    bn.key_col = 0
+    bn.semijoin_table = -1
+    bn.capacity_hint = 32
+    bn.key_set_only = 1
+    on = Operator(OP_HASH_BUILD, bn)
+    on.add_input(natp)
+    on.finish()
+
+    abc = cust.channel("acctbal")
+    # positive-balance population aggregate (sum, count)
+    fp = PlanFilterProject()
+    fp.n_preds = 1
+    fp.preds[0] = Pred(abc, CMP_GT, 0, 0.0)
+    fp.n_proj = 1
+    fp.proj[0] = Proj(PROJ_IDENT, abc, 0, 0)
+    fp.semijoin_table = on.table()
+    fp.semijoin_col = cust.channel("nationkey")
+    f0 = Operator(OP_FILTER_PROJECT, fp)
+    f0.add_input(cust)
+    ppage = f0.get_output_raw()
+
+    ap = PlanHashAggSmall()
+    ap.n_keys = 0
+    ap.n_aggs = 2
+    ap.aggs[0] = Agg(AGG_SUM_DEC, Proj(PROJ_IDENT, 0, 0, 0), 0)
+    ap.aggs[1] = Agg(AGG_COUNT, Proj(PROJ_IDENT, 0, 0, 0), 0)
+    a0 = Operator(OP_HASH_AGG_SMALL, ap)
+    a0.add_input_raw(ppage)
+    a0.finish()
+    r = a0.get_output(["hi", "lo", "cnt"])
+    sum_pos = (int(r["hi"][0]) << 64) | int(np.uint64(r["lo"][0]))
+    cnt_pos = int(r["cnt"][0])
+    a0.destroy()
+    f0.destroy()
+
+    bo = PlanHashBuild()
+    bo.key_col = orders.channel("custkey")
+    bo.semijoin_table = -1
+    bo.capacity_hint = orders.n_rows
+    bo.key_set_only = 1
+    oo = Operator(OP_HASH_BUILD, bo)
+    oo.add_input(orders)
+    oo.finish()
+
+    f1p = PlanFilterProject()
+    f1p.n_preds = 1
+    f1p.preds[0] = Pred(abc, CMP_GT, sum_pos // cnt_pos, 0.0)
+    f1p.n_proj = 3
+    f1p.proj[0] = Proj(PROJ_IDENT, cust.channel("custkey"), 0, 0)
+    f1p.proj[1] = Proj(PROJ_IDENT, cust.channel("nationkey"), 0, 0)
+    f1p.proj[2] = Proj(PROJ_IDENT, abc, 0, 0)
+    f1p.semijoin_table = on.table()
+    f1p.semijoin_col = cust.channel("nationkey")
+    f1 = Operator(OP_FILTER_PROJECT, f1p)
+    f1.add_input(cust)
+    page1 = f1.get_output_raw()
+
+    f2p = PlanFilterProject()
+    f2p.n_proj = 2
+    f2p.proj[0] = Proj(PROJ_IDENT, 1, 0, 0)
+    f2p.proj[1] = Proj(PROJ_IDENT, 2, 0, 0)
+    f2p.semijoin_table = oo.table()
+    f2p.semijoin_col = 0
+    f2p.semijoin_anti = 1
+    f2 = Operator(OP_FILTER_PROJECT, f2p)
+    f2.add_input_raw(page1)
+    page2 = f2.get_output_raw()  # [nationkey, acctbal]
+
+    ag = PlanHashAggSmall()
+    ag.n_keys = 1
+    ag.key_col[0] = 0
+    ag.n_vals[0] = len(Q22_CODE_NATIONS)
+    for i, v in enumerate(Q22_CODE_NATIONS):
+        ag.key_vals[0][i] = v
+    ag.n_aggs = 2
+    ag.aggs[0] = Agg(AGG_SUM_DEC, Proj(PROJ_IDENT, 1, 0, 0), 0)
+    ag.aggs[1] = Agg(AGG_COUNT, Proj(PROJ_IDENT, 0, 0, 0), 0)
+    a = Operator(OP_HASH_AGG_SMALL, ag)
+    a.add_input_raw(page2)
+    a.finish()
+    out = a.get_output(["nationkey", "hi", "lo", "cnt"])
+    a.destroy()
+    f2.destroy()
+    f1.destroy()
+    for o in (on, oo):
+        lib().c.pg_table_destroy(o.table())
+        o.destroy()
+    cnts = [0] * len(Q22_CODE_NATIONS)
+    sums = [0] * len(Q22_CODE_NATIONS)
+    for i in range(len(out["nationkey"])):
+        j = Q22_CODE_NATIONS.index(int(out["nationkey"][i]))
+        cnts[j] = int(out["cnt"][i])
+        sums[j] = (int(out["hi"][i]) << 64) | int(np.uint64(out["lo"][i]))
+    return cnts, sums
+
+
 def q4(orders: Page, li_dates: Page):
     """Q4 order-priority checking (q04.sql): EXISTS(lineitem with
     commitdate < receiptdate) as a key-set build with a col-vs-col

@@ -1008,6 +1008,23 @@ def test_q18_exact(P, oracle_lib):
     assert len(got) > 0
 
 
+def test_q22_exact(P, oracle_lib):
+    """Q22 global sales opportunity — anti-semijoin (NOT-EXISTS pushdown)
+    + plan-constant average threshold vs the golden-pinned oracle."""
+    sf = 0.1
+    cust = oracle_lib.gen_customer2(sf)
+    abal = oracle_lib.gen_customer_acctbal(sf)
+    orders = oracle_lib.gen_orders(sf)
+    got_cnt, got_sum = P.pipelines.q22(
+        P.Page({"custkey": cust["custkey"], "nationkey": cust["nationkey"],
+                "acctbal": abal}),
+        P.Page({"custkey": orders["custkey"]}, n_rows=len(orders["custkey"])))
+    exp_cnt, exp_sum = oracle_lib.q22(cust, abal, orders)
+    assert got_cnt == exp_cnt
+    assert got_sum == exp_sum
+    assert sum(got_cnt) > 0
+
+
 def test_q5_distributed_graph_world1(P, oracle_lib):
     """The distributed Q5 graph at world==1 (replicated dimensions +
     partition/exchange identities + fused probe + exact tick combine)
