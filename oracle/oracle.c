@@ -1123,3 +1123,55 @@ void oracle_q22(int64_t n_cust, const int64_t* ck, const uint8_t* cnat,
     }
     free(has_ord);
 }
+
+/* ---------------- Q19 ----------------
+ * SQL: q19.sql — discounted revenue, three brand/container/size/qty
+ * disjuncts over 'AIR'-shipped (id 1; the query's literal 'AIR REG'
+ * matches no generated value — the data has 'REG AIR'), DELIVER IN
+ * PERSON (id 0) lines.  Container ids per the nested Cnt1 x Cnt2 order
+ * pinned by q17/q19 goldens.  Exact 1e-4 tick sum. */
+void oracle_q19(int64_t n_li, const int64_t* lpk, const double* lqty,
+                const double* lep, const double* ldisc,
+                const uint8_t* lsmode, const uint8_t* lsinst,
+                int64_t n_part, const uint8_t* brand,
+                const uint8_t* container, const uint8_t* size,
+                int64_t* revenue_1e4)
+{
+    /* per-part disjunct class: 1,2,3 or 0 (none) */
+    uint8_t* cls = (uint8_t*)calloc(n_part, 1);
+    static const uint8_t SM[4] = {0, 1, 5, 4};    /* SM CASE/BOX/PACK/PKG */
+    static const uint8_t MED[4] = {18, 17, 21, 20}; /* MED BAG/BOX/PACK/PKG */
+    static const uint8_t LG[4] = {8, 9, 13, 12};  /* LG CASE/BOX/PACK/PKG */
+    for (int64_t p = 0; p < n_part; p++) {
+        uint8_t b = brand[p], c = container[p], z = size[p];
+        if (z < 1) continue;
+        if (b == 12 && z <= 5) {
+            for (int k = 0; k < 4; k++)
+                if (c == SM[k]) cls[p] = 1;
+        } else if (b == 23 && z <= 10) {
+            for (int k = 0; k < 4; k++)
+                if (c == MED[k]) cls[p] = 2;
+        } else if (b == 34 && z <= 15) {
+            for (int k = 0; k < 4; k++)
+                if (c == LG[k]) cls[p] = 3;
+        }
+    }
+    static const int64_t QLO[4] = {0, 1, 10, 20};
+    static const int64_t QHI[4] = {-1, 11, 20, 30};
+    int64_t rev = 0;
+#pragma omp parallel for schedule(static) reduction(+ : rev)
+    for (int64_t i = 0; i < n_li; i++) {
+        if (lsmode[i] != 1 || lsinst[i] != 0) continue;
+        int64_t pk = lpk[i];
+        if (pk < 1 || pk > n_part) continue;
+        uint8_t k = cls[pk - 1];
+        if (!k) continue;
+        int64_t q = (int64_t)(lqty[i] + 0.5);
+        if (q < QLO[k] || q > QHI[k]) continue;
+        int64_t cents = (int64_t)(lep[i] * 100.0 + 0.5);
+        int64_t d = (int64_t)(ldisc[i] * 100.0 + 0.5);
+        rev += cents * (100 - d);
+    }
+    free(cls);
+    *revenue_1e4 = rev;
+}

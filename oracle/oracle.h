@@ -246,6 +246,17 @@ void oracle_q22(int64_t n_cust, const int64_t* c_custkey,
                 const uint8_t* code_nations, int64_t* out_cnt,
                 int64_t* out_sum);
 
+/* ---------------- TPC-H Q19 ----------------
+ * q19.sql — discounted revenue over three brand/container/size/quantity
+ * disjuncts ('AIR' shipmode id 1, 'DELIVER IN PERSON' instruct id 0).
+ * Exact 1e-4 ticks. */
+void oracle_q19(int64_t n_li, const int64_t* l_partkey,
+                const double* l_quantity, const double* l_extendedprice,
+                const double* l_discount, const uint8_t* l_shipmode,
+                const uint8_t* l_shipinstruct, int64_t n_part,
+                const uint8_t* p_brand, const uint8_t* p_container,
+                const uint8_t* p_size, int64_t* revenue_1e4);
+
 /* ---------------- operator-level primitives (parity targets) ---------- */
 
 /* murmur3 finalizer bucket — PagesHash.java:236-252 /

@@ -61,6 +61,10 @@ static inline int64_t unif(int64_t* s, int64_t lo, int64_t hi)
                                     * 4/part (q11 golden pin, all 1048 rows) */
 #define SEED_C_ABAL    298370230LL /* c_acctbal cents -99999..999999,
                                     * 1/customer (q22 golden pin) */
+#define SEED_P_NAME    922618556LL /* p_name permutation, 92/part (q9 pin) */
+#define SEED_L_SINST  1371272478LL /* shipinstruct pick 1..4, 7/order
+                                    * ('DELIVER IN PERSON' = id 0; q19 pin) */
+#define SEED_P_SIZE   1193163244LL /* p_size 1..50, 1/part (q19 pin) */
 
 /* ---- calendar ----
  * day index 1 = 1992-01-01; order-date index in [1, 2406]
@@ -309,6 +313,39 @@ int64_t tpch_gen_lineitem_partkey(double sf, int64_t ord_start,
     return written;
 }
 
+int64_t tpch_gen_lineitem_shipinstruct(double sf, int64_t ord_start,
+                                       int64_t ord_count,
+                                       uint8_t* shipinstruct)
+{
+    (void)sf;
+    int64_t written = 0;
+#pragma omp parallel reduction(+ : written)
+    {
+#ifdef _OPENMP
+        extern int omp_get_num_threads(void), omp_get_thread_num(void);
+        int nt = omp_get_num_threads(), tid = omp_get_thread_num();
+#else
+        int nt = 1, tid = 0;
+#endif
+        int64_t tlo = ord_count * tid / nt, thi = ord_count * (tid + 1) / nt;
+        int64_t s = rng_skip(SEED_O_LCNT, (uint64_t)ord_start);
+        int64_t off = 0;
+        for (int64_t i = 0; i < tlo; i++) off += unif(&s, 1, 7);
+        int64_t s_lcnt = rng_skip(SEED_O_LCNT, (uint64_t)(ord_start + tlo));
+        int64_t s_inst =
+            rng_skip(SEED_L_SINST, (uint64_t)(ord_start + tlo) * 7);
+        int64_t out = off;
+        for (int64_t o = tlo; o < thi; o++) {
+            int64_t lc = unif(&s_lcnt, 1, 7);
+            for (int64_t l = 0; l < lc; l++)
+                shipinstruct[out++] = (uint8_t)(unif(&s_inst, 1, 4) - 1);
+            s_inst = rng_skip(s_inst, (uint64_t)(7 - lc));
+            written += lc;
+        }
+    }
+    return written;
+}
+
 int64_t tpch_gen_lineitem_shipmode(double sf, int64_t ord_start,
                                    int64_t ord_count, uint8_t* shipmode)
 {
@@ -344,6 +381,12 @@ int64_t tpch_gen_lineitem_shipmode(double sf, int64_t ord_start,
 void tpch_gen_part2(double sf, int64_t start, int64_t count, uint8_t* mfgr,
                     uint8_t* brand, uint8_t* container)
 {
+    tpch_gen_part3(sf, start, count, mfgr, brand, container, 0);
+}
+
+void tpch_gen_part3(double sf, int64_t start, int64_t count, uint8_t* mfgr,
+                    uint8_t* brand, uint8_t* container, uint8_t* size)
+{
 #pragma omp parallel
     {
 #ifdef _OPENMP
@@ -357,13 +400,16 @@ void tpch_gen_part2(double sf, int64_t start, int64_t count, uint8_t* mfgr,
         int64_t sm = rng_skip(SEED_P_MFG, (uint64_t)(start + lo));
         int64_t sb = rng_skip(SEED_P_BRND, (uint64_t)(start + lo));
         int64_t sc = rng_skip(SEED_P_CNTR, (uint64_t)(start + lo));
+        int64_t sz = rng_skip(SEED_P_SIZE, (uint64_t)(start + lo));
         for (int64_t i = lo; i < hi; i++) {
             int64_t m = unif(&sm, 1, 5);
             int64_t b = unif(&sb, 1, 5);
             int64_t c = unif(&sc, 1, 40);
+            int64_t z = unif(&sz, 1, 50);
             if (mfgr) mfgr[i] = (uint8_t)m;
             if (brand) brand[i] = (uint8_t)(m * 10 + b);
             if (container) container[i] = (uint8_t)(c - 1);
+            if (size) size[i] = (uint8_t)z;
         }
     }
 }
@@ -398,6 +444,59 @@ void tpch_gen_partsupp(double sf, int64_t part_start, int64_t part_count,
                 if (supplycost_cents) supplycost_cents[out] = cost;
             }
         }
+    }
+}
+
+/* dbgen dists.dss "colors" (92 words, file order) — p_name draws 5 of
+ * these via the persistent permutation stream (dbgen rnd.c permute /
+ * build.c agg_str; airlift RandomStringSequence is the same algorithm) */
+static const char* P_COLORS[92] = {
+    "almond", "antique", "aquamarine", "azure", "beige", "bisque", "black",
+    "blanched", "blue", "blush", "brown", "burlywood", "burnished",
+    "chartreuse", "chiffon", "chocolate", "coral", "cornflower", "cornsilk",
+    "cream", "cyan", "dark", "deep", "dim", "dodger", "drab", "firebrick",
+    "floral", "forest", "frosted", "gainsboro", "ghost", "goldenrod",
+    "green", "grey", "honeydew", "hot", "indian", "ivory", "khaki", "lace",
+    "lavender", "lawn", "lemon", "light", "lime", "linen", "magenta",
+    "maroon", "medium", "metallic", "midnight", "mint", "misty", "moccasin",
+    "navajo", "navy", "olive", "orange", "orchid", "pale", "papaya",
+    "peach", "peru", "pink", "plum", "powder", "puff", "purple", "red",
+    "rose", "rosy", "royal", "saddle", "salmon", "sandy", "seashell",
+    "sienna", "sky", "slate", "smoke", "snow", "spring", "steel", "tan",
+    "thistle", "tomato", "turquoise", "violet", "wheat", "white", "yellow"};
+
+int32_t tpch_color_id(const char* word)
+{
+    for (int32_t i = 0; i < 92; i++) {
+        const char* a = P_COLORS[i];
+        const char* b = word;
+        while (*a && *a == *b) {
+            a++;
+            b++;
+        }
+        if (!*a && !*b) return i;
+    }
+    return -1;
+}
+
+/* word ids (0..91 into the colors list) of the 5 p_name words per part,
+ * rows [0, count) from part row 0 (the permutation state is sequential
+ * from the first part, so generation always starts at part 1).
+ * words[p*5 + j] = j-th name word of part p+1. */
+void tpch_gen_part_name_words(double sf, int64_t count, uint8_t* words)
+{
+    (void)sf;
+    uint8_t perm[92];
+    for (int i = 0; i < 92; i++) perm[i] = (uint8_t)i;
+    int64_t s = SEED_P_NAME;
+    for (int64_t p = 0; p < count; p++) {
+        for (int i = 0; i < 92; i++) {
+            int64_t j = unif(&s, i, 91);
+            uint8_t t = perm[j];
+            perm[j] = perm[i];
+            perm[i] = t;
+        }
+        for (int j = 0; j < 5; j++) words[p * 5 + j] = perm[j];
     }
 }
 

@@ -94,6 +94,14 @@ int64_t tpch_gen_lineitem_shipmode(double sf, int64_t ord_start,
  * may be NULL. */
 void tpch_gen_part2(double sf, int64_t start, int64_t count, uint8_t* mfgr,
                     uint8_t* brand, uint8_t* container);
+/* as tpch_gen_part2 plus p_size 1..50 (q19 golden pin) */
+void tpch_gen_part3(double sf, int64_t start, int64_t count, uint8_t* mfgr,
+                    uint8_t* brand, uint8_t* container, uint8_t* size);
+
+/* lineitem shipinstruct ids 0..3 ('DELIVER IN PERSON' = 0, q19 pin) */
+int64_t tpch_gen_lineitem_shipinstruct(double sf, int64_t ord_start,
+                                       int64_t ord_count,
+                                       uint8_t* shipinstruct);
 
 /* partsupp: 4 rows per part row [part_start, part_start+part_count);
  * suppkey via the PART_SUPP bridge; ps_availqty 1..9999;
@@ -103,6 +111,13 @@ void tpch_gen_part2(double sf, int64_t start, int64_t count, uint8_t* mfgr,
 void tpch_gen_partsupp(double sf, int64_t part_start, int64_t part_count,
                        int64_t* partkey, int64_t* suppkey,
                        int32_t* availqty, int64_t* supplycost_cents);
+
+/* p_name word ids: 5 words per part drawn from the 92-color list via the
+ * persistent permutation stream (dbgen agg_str).  words has 5*count
+ * entries; generation always starts at part 1 (sequential state). */
+void tpch_gen_part_name_words(double sf, int64_t count, uint8_t* words);
+/* index of a color word in the dists order, -1 if absent */
+int32_t tpch_color_id(const char* word);
 
 /* c_acctbal in exact cents, -99999..999999 (q22 golden pin) */
 void tpch_gen_customer_acctbal(double sf, int64_t start, int64_t count,

@@ -477,3 +477,41 @@ def q22(self, cust, abal, orders):
 OracleLib.gen_customer_acctbal = gen_customer_acctbal
 OracleLib.q21 = q21
 OracleLib.q22 = q22
+
+
+def gen_part3(self, sf):
+    n = int(200000 * sf)
+    mfgr = np.empty(n, np.uint8)
+    brand = np.empty(n, np.uint8)
+    cntr = np.empty(n, np.uint8)
+    size = np.empty(n, np.uint8)
+    self.lib.tpch_gen_part3(C.c_double(sf), C.c_int64(0), C.c_int64(n),
+                            _p(mfgr), _p(brand), _p(cntr), _p(size))
+    return {"mfgr": mfgr, "brand": brand, "container": cntr, "size": size}
+
+
+def gen_lineitem_shipinstruct(self, sf):
+    n = self.lineitem_count(sf)
+    n_ord = self.lib.tpch_orders_count(C.c_double(sf))
+    si = np.empty(n, np.uint8)
+    self.lib.tpch_gen_lineitem_shipinstruct.restype = C.c_int64
+    w = self.lib.tpch_gen_lineitem_shipinstruct(C.c_double(sf), C.c_int64(0),
+                                                C.c_int64(n_ord), _p(si))
+    assert w == n
+    return si
+
+
+def q19(self, li, lpk, smode, sinst, part3):
+    out = C.c_int64()
+    self.lib.oracle_q19(C.c_int64(len(li["orderkey"])), _p(lpk),
+                        _p(li["quantity"]), _p(li["extendedprice"]),
+                        _p(li["discount"]), _p(smode), _p(sinst),
+                        C.c_int64(len(part3["brand"])), _p(part3["brand"]),
+                        _p(part3["container"]), _p(part3["size"]),
+                        C.byref(out))
+    return out.value
+
+
+OracleLib.gen_part3 = gen_part3
+OracleLib.gen_lineitem_shipinstruct = gen_lineitem_shipinstruct
+OracleLib.q19 = q19

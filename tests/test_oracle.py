@@ -447,3 +447,17 @@ def test_q22_sf1_golden(oracle_lib):
         assert int(Q22_CODE_NATIONS[i]) + 10 == int(g[0])
         assert cnt[i] == int(g[1])
         assert Decimal(tot[i]) / 100 == Decimal(g[2])
+
+
+def test_q19_sf1_golden(oracle_lib):
+    """Q19 discounted revenue — pins p_size, shipinstruct and the 'AIR'
+    shipmode id (the query's 'AIR REG' literal matches no generated
+    value)."""
+    li = oracle_lib.gen_lineitem2(1.0)
+    lpk = oracle_lib.gen_lineitem_partkey(1.0)
+    smode = oracle_lib.gen_lineitem_shipmode(1.0)
+    sinst = oracle_lib.gen_lineitem_shipinstruct(1.0)
+    part3 = oracle_lib.gen_part3(1.0)
+    rev = oracle_lib.q19(li, lpk, smode, sinst, part3)
+    golden = _parse_golden("q19_sf1.result")
+    assert Decimal(rev) / 10**4 == Decimal(golden[0][0]), rev
