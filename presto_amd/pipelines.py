@@ -1097,6 +1097,73 @@ def q18(orders: Page, li: Page, limit=100):
     return rows[:limit]
 
 
+def q19(part: Page, li: Page):
+    """Q19 discounted revenue (q19.sql): part attributes join (chained
+    table, 3 u8 payloads) fused with the shipmode/shipinstruct/quantity
+    pre-filter, then the three brand/container/size/qty disjuncts as
+    twelve conjunctive keyless aggregations over the joined page (the
+    OR decomposes per container id).  Exact 1e-4 ticks."""
+    import numpy as np
+    from .engine import lib
+
+    bp = PlanHashBuild()
+    bp.key_col = part.channel("partkey")
+    bp.semijoin_table = -1
+    bp.n_payload = 3
+    bp.payload_col[0] = part.channel("brand")
+    bp.payload_col[1] = part.channel("container")
+    bp.payload_col[2] = part.channel("size")
+    bp.capacity_hint = part.n_rows
+    ob = Operator(OP_HASH_BUILD, bp)
+    ob.add_input(part)
+    ob.finish()
+
+    jp = PlanLookupJoin()
+    jp.table = ob.table()
+    jp.n_preds = 3
+    jp.preds[0] = Pred(li.channel("shipmode"), CMP_EQ, 1, 0.0)   # AIR
+    jp.preds[1] = Pred(li.channel("shipinstruct"), CMP_EQ, 0, 0.0)
+    jp.preds[2] = Pred(li.channel("quantity"), CMP_LE, 0, 30.0)
+    jp.key_col = li.channel("partkey")
+    jp.mode = 0
+    jp.n_emit = 3
+    jp.emit_probe_cols[0] = li.channel("quantity")
+    jp.emit_probe_cols[1] = li.channel("extendedprice")
+    jp.emit_probe_cols[2] = li.channel("discount")
+    jo = Operator(OP_LOOKUP_JOIN, jp)
+    jo.add_input(li)
+    jpage = jo.get_output_raw()  # [qty, ep, dc, brand, container, size]
+
+    DISJ = ((12, (0, 1, 5, 4), 1, 11, 5),
+            (23, (18, 17, 21, 20), 10, 20, 10),
+            (34, (8, 9, 13, 12), 20, 30, 15))
+    total = 0
+    for bnum, cset, qlo, qhi, szhi in DISJ:
+        for c in cset:
+            p = PlanHashAggSmall()
+            p.n_preds = 6
+            p.preds[0] = Pred(3, CMP_EQ, bnum, 0.0)
+            p.preds[1] = Pred(4, CMP_EQ, c, 0.0)
+            p.preds[2] = Pred(5, CMP_GE, 1, 0.0)
+            p.preds[3] = Pred(5, CMP_LE, szhi, 0.0)
+            p.preds[4] = Pred(0, CMP_GE, 0, float(qlo))
+            p.preds[5] = Pred(0, CMP_LE, 0, float(qhi))
+            p.n_keys = 0
+            p.n_aggs = 1
+            p.aggs[0] = Agg(AGG_SUM_DEC, Proj(PROJ_DISC_PRICE, 1, 2, 0), 4)
+            a = Operator(OP_HASH_AGG_SMALL, p)
+            a.add_input_raw(jpage)
+            a.finish()
+            r = a.get_output(["hi", "lo"])
+            if len(r["lo"]):
+                total += (int(r["hi"][0]) << 64) | int(np.uint64(r["lo"][0]))
+            a.destroy()
+    jo.destroy()
+    lib().c.pg_table_destroy(ob.table())
+    ob.destroy()
+    return total
+
+
 Q22_CODE_NATIONS = (3, 7, 8, 13, 19, 20, 21)  # codes '13'..'31' ascending
 
 

@@ -1008,6 +1008,30 @@ def test_q18_exact(P, oracle_lib):
     assert len(got) > 0
 
 
+def test_q19_exact(P, oracle_lib):
+    """Q19 disjunctive revenue — fused-filter attribute join + twelve
+    conjunctive keyless aggs vs the golden-pinned oracle, exact ticks."""
+    import numpy as np
+    sf = 0.1
+    li = oracle_lib.gen_lineitem2(sf)
+    lpk = oracle_lib.gen_lineitem_partkey(sf)
+    smode = oracle_lib.gen_lineitem_shipmode(sf)
+    sinst = oracle_lib.gen_lineitem_shipinstruct(sf)
+    part3 = oracle_lib.gen_part3(sf)
+    n = len(part3["brand"])
+    got = P.pipelines.q19(
+        P.Page({"partkey": np.arange(1, n + 1, dtype=np.int64),
+                "brand": part3["brand"], "container": part3["container"],
+                "size": part3["size"]}),
+        P.Page({"partkey": lpk, "quantity": li["quantity"],
+                "extendedprice": li["extendedprice"],
+                "discount": li["discount"], "shipmode": smode,
+                "shipinstruct": sinst}))
+    exp = oracle_lib.q19(li, lpk, smode, sinst, part3)
+    assert got == exp
+    assert got > 0
+
+
 def test_q22_exact(P, oracle_lib):
     """Q22 global sales opportunity — anti-semijoin (NOT-EXISTS pushdown)
     + plan-constant average threshold vs the golden-pinned oracle."""
