@@ -1175,3 +1175,73 @@ void oracle_q19(int64_t n_li, const int64_t* lpk, const double* lqty,
     free(cls);
     *revenue_1e4 = rev;
 }
+
+/* ---------------- Q9 ----------------
+ * SQL: q09.sql — product-type profit: amount = extendedprice*(1-disc)
+ * - supplycost*quantity over parts whose p_name contains a given word
+ * (precomputed p_match flags), grouped by (supplier nation, order
+ * year).  partsupp arrays are the generator's layout: 4 rows per part
+ * starting at part 1 (supplycost lookup scans the part's 4 suppliers).
+ * Fills profit_1e4[nation*7 + (year-1992)] exact ticks. */
+void oracle_q9(int64_t n_li, const int64_t* lpk, const int64_t* lsk,
+               const double* lqty, const double* lep, const double* ldisc,
+               const int64_t* lok, int64_t n_ord, const int64_t* ook,
+               const int32_t* odate, int64_t n_supp, const uint8_t* snat,
+               int64_t n_part, const uint8_t* p_match,
+               const int64_t* ps_suppkey, const int64_t* ps_cost,
+               int64_t* profit_1e4)
+{
+    /* orderkey -> orderdate */
+    int64_t cap = hash_capacity(n_ord < 2 ? 2 : n_ord);
+    int64_t* slot = (int64_t*)malloc(cap * sizeof(int64_t));
+    memset(slot, -1, cap * sizeof(int64_t));
+    for (int64_t i = 0; i < n_ord; i++) {
+        uint64_t h = pg_murmur3_finalize(pg_bigint_hash(ook[i]));
+        int64_t pos = (int64_t)(h & (cap - 1));
+        while (slot[pos] != -1) pos = (pos + 1) & (cap - 1);
+        slot[pos] = i;
+    }
+    /* year boundaries 1992..1999 as epoch days */
+    static const int32_t YB[8] = {8035, 8401, 8766, 9131, 9496, 9862,
+                                  10227, 10592};
+    for (int i = 0; i < 25 * 7; i++) profit_1e4[i] = 0;
+#pragma omp parallel
+    {
+        int64_t loc[25 * 7];
+        memset(loc, 0, sizeof(loc));
+#pragma omp for schedule(static)
+        for (int64_t i = 0; i < n_li; i++) {
+            int64_t pk = lpk[i];
+            if (pk < 1 || pk > n_part || !p_match[pk - 1]) continue;
+            uint64_t h = pg_murmur3_finalize(pg_bigint_hash(lok[i]));
+            int64_t pos = (int64_t)(h & (cap - 1));
+            int64_t r = -1;
+            for (;;) {
+                int64_t sI = slot[pos];
+                if (sI == -1) break;
+                if (ook[sI] == lok[i]) {
+                    r = sI;
+                    break;
+                }
+                pos = (pos + 1) & (cap - 1);
+            }
+            if (r == -1) continue;
+            int y = 0;
+            while (y < 7 && odate[r] >= YB[y + 1]) y++;
+            int64_t sk = lsk[i];
+            if (sk < 1 || sk > n_supp) continue;
+            int64_t cost = -1;
+            for (int j = 0; j < 4; j++)
+                if (ps_suppkey[(pk - 1) * 4 + j] == sk)
+                    cost = ps_cost[(pk - 1) * 4 + j];
+            if (cost < 0) continue;
+            int64_t q = (int64_t)(lqty[i] + 0.5);
+            int64_t cents = (int64_t)(lep[i] * 100.0 + 0.5);
+            int64_t d = (int64_t)(ldisc[i] * 100.0 + 0.5);
+            loc[snat[sk - 1] * 7 + y] += cents * (100 - d) - cost * q * 100;
+        }
+#pragma omp critical
+        for (int i = 0; i < 25 * 7; i++) profit_1e4[i] += loc[i];
+    }
+    free(slot);
+}

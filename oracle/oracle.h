@@ -257,6 +257,19 @@ void oracle_q19(int64_t n_li, const int64_t* l_partkey,
                 const uint8_t* p_brand, const uint8_t* p_container,
                 const uint8_t* p_size, int64_t* revenue_1e4);
 
+/* ---------------- TPC-H Q9 ----------------
+ * q09.sql — product-type profit by (supplier nation, order year);
+ * p_match flags mark parts whose name contains the query's word.
+ * profit_1e4[nation*7 + (year-1992)] exact ticks. */
+void oracle_q9(int64_t n_li, const int64_t* l_partkey,
+               const int64_t* l_suppkey, const double* l_quantity,
+               const double* l_extendedprice, const double* l_discount,
+               const int64_t* l_orderkey, int64_t n_ord,
+               const int64_t* o_orderkey, const int32_t* o_orderdate,
+               int64_t n_supp, const uint8_t* s_nationkey, int64_t n_part,
+               const uint8_t* p_match, const int64_t* ps_suppkey,
+               const int64_t* ps_supplycost_cents, int64_t* profit_1e4);
+
 /* ---------------- operator-level primitives (parity targets) ---------- */
 
 /* murmur3 finalizer bucket — PagesHash.java:236-252 /

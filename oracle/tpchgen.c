@@ -61,7 +61,9 @@ static inline int64_t unif(int64_t* s, int64_t lo, int64_t hi)
                                     * 4/part (q11 golden pin, all 1048 rows) */
 #define SEED_C_ABAL    298370230LL /* c_acctbal cents -99999..999999,
                                     * 1/customer (q22 golden pin) */
-#define SEED_P_NAME    922618556LL /* p_name permutation, 92/part (q9 pin) */
+#define SEED_P_NAME    709314158LL /* p_name permutation, 92/part; fresh
+                                    * identity re-permuted per part (the
+                                    * airlift port's semantics — q9 pin) */
 #define SEED_L_SINST  1371272478LL /* shipinstruct pick 1..4, 7/order
                                     * ('DELIVER IN PERSON' = id 0; q19 pin) */
 #define SEED_P_SIZE   1193163244LL /* p_size 1..50, 1/part (q19 pin) */
@@ -480,16 +482,16 @@ int32_t tpch_color_id(const char* word)
 }
 
 /* word ids (0..91 into the colors list) of the 5 p_name words per part,
- * rows [0, count) from part row 0 (the permutation state is sequential
- * from the first part, so generation always starts at part 1).
+ * rows [0, count): per part the identity permutation is re-permuted with
+ * 92 draws (j in [i,91] swap pass) and the first 5 entries are the name.
  * words[p*5 + j] = j-th name word of part p+1. */
 void tpch_gen_part_name_words(double sf, int64_t count, uint8_t* words)
 {
     (void)sf;
     uint8_t perm[92];
-    for (int i = 0; i < 92; i++) perm[i] = (uint8_t)i;
     int64_t s = SEED_P_NAME;
     for (int64_t p = 0; p < count; p++) {
+        for (int i = 0; i < 92; i++) perm[i] = (uint8_t)i;
         for (int i = 0; i < 92; i++) {
             int64_t j = unif(&s, i, 91);
             uint8_t t = perm[j];

@@ -515,3 +515,40 @@ def q19(self, li, lpk, smode, sinst, part3):
 OracleLib.gen_part3 = gen_part3
 OracleLib.gen_lineitem_shipinstruct = gen_lineitem_shipinstruct
 OracleLib.q19 = q19
+
+
+def gen_part_name_words(self, sf):
+    n = int(200000 * sf)
+    w = np.empty(n * 5, np.uint8)
+    self.lib.tpch_gen_part_name_words(C.c_double(sf), C.c_int64(n), _p(w))
+    return w.reshape(-1, 5)
+
+
+def color_id(self, word):
+    return int(self.lib.tpch_color_id(word.encode()))
+
+
+def q9(self, li, lpk, orders, supp, ps, p_match):
+    out = np.empty(25 * 7, np.int64)
+    self.lib.oracle_q9(C.c_int64(len(li["orderkey"])), _p(lpk),
+                       _p(li["suppkey"]), _p(li["quantity"]),
+                       _p(li["extendedprice"]), _p(li["discount"]),
+                       _p(li["orderkey"]), C.c_int64(len(orders["orderkey"])),
+                       _p(orders["orderkey"]), _p(orders["orderdate"]),
+                       C.c_int64(len(supp["suppkey"])), _p(supp["nationkey"]),
+                       C.c_int64(len(p_match)), _p(p_match),
+                       _p(ps["suppkey"]), _p(ps["supplycost_cents"]),
+                       _p(out))
+    return out.reshape(25, 7)
+
+
+def nation_name(self, nk):
+    buf = C.create_string_buffer(32)
+    n = self.lib.tpch_nation_name(C.c_int32(nk), buf)
+    return buf.value.decode()
+
+
+OracleLib.gen_part_name_words = gen_part_name_words
+OracleLib.color_id = color_id
+OracleLib.q9 = q9
+OracleLib.nation_name = nation_name

@@ -461,3 +461,29 @@ def test_q19_sf1_golden(oracle_lib):
     rev = oracle_lib.q19(li, lpk, smode, sinst, part3)
     golden = _parse_golden("q19_sf1.result")
     assert Decimal(rev) / 10**4 == Decimal(golden[0][0]), rev
+
+
+def test_q9_sf1_golden(oracle_lib):
+    """Q9 product-type profit — pins the p_name permutation stream on
+    all 175 golden rows (nation x year, ordered by nation NAME asc,
+    year desc)."""
+    import numpy as np
+    li = oracle_lib.gen_lineitem2(1.0)
+    lpk = oracle_lib.gen_lineitem_partkey(1.0)
+    orders = oracle_lib.gen_orders(1.0)
+    supp = oracle_lib.gen_supplier(1.0)
+    ps = oracle_lib.gen_partsupp(1.0)
+    words = oracle_lib.gen_part_name_words(1.0)
+    gid = oracle_lib.color_id("green")
+    p_match = (words == gid).any(axis=1).astype(np.uint8)
+    prof = oracle_lib.q9(li, lpk, orders, supp, ps, p_match)
+    golden = _parse_golden("q09_sf1.result")
+    names = {oracle_lib.nation_name(k): k for k in range(25)}
+    rows = []
+    for name in sorted(names):
+        for y in range(1998, 1991, -1):
+            rows.append((name, y, int(prof[names[name], y - 1992])))
+    assert len(rows) == len(golden) == 175
+    for (name, y, ticks), g in zip(rows, golden):
+        assert name == g[0] and y == int(g[1])
+        assert Decimal(ticks) / 10**4 == Decimal(g[2]), (name, y)
