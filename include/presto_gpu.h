@@ -100,14 +100,18 @@ typedef struct {
  * specialized HIP kernels selected by these plan descriptors. */
 
 typedef enum { PG_CMP_LT = 0, PG_CMP_LE, PG_CMP_GT, PG_CMP_GE, PG_CMP_EQ,
-               PG_CMP_NE } pg_cmp;
+               PG_CMP_NE,
+               /* VARBIN only — the LIKE '%w%' / 'w%' pushdowns
+                * (LikeFunctions.java:64-77 likeVarchar for patterns
+                * without '_' reduce to substring/prefix search) */
+               PG_CMP_CONTAINS, PG_CMP_PREFIX } pg_cmp;
 
 typedef struct {
     int32_t col;   /* input channel */
     int32_t op;    /* pg_cmp */
     int64_t ival;  /* compare value for integer columns */
     double dval;   /* compare value for f64 columns */
-    char sval[16]; /* VARBIN: compare bytes (EQ/NE only) */
+    char sval[16]; /* VARBIN: compare bytes (EQ/NE/CONTAINS/PREFIX) */
     int32_t slen;
     int32_t rhs_col; /* 0: compare against the constant; k>0: compare
                         against integer channel k-1 (filter expression

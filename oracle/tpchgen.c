@@ -467,6 +467,19 @@ static const char* P_COLORS[92] = {
     "sienna", "sky", "slate", "smoke", "snow", "spring", "steel", "tan",
     "thistle", "tomato", "turquoise", "violet", "wheat", "white", "yellow"};
 
+int32_t tpch_color_name(int32_t id, char* buf)
+{
+    if (id < 0 || id >= 92) return -1;
+    const char* s = P_COLORS[id];
+    int32_t n = 0;
+    while (s[n]) {
+        buf[n] = s[n];
+        n++;
+    }
+    buf[n] = 0;
+    return n;
+}
+
 int32_t tpch_color_id(const char* word)
 {
     for (int32_t i = 0; i < 92; i++) {

@@ -112,12 +112,15 @@ void tpch_gen_partsupp(double sf, int64_t part_start, int64_t part_count,
                        int64_t* partkey, int64_t* suppkey,
                        int32_t* availqty, int64_t* supplycost_cents);
 
-/* p_name word ids: 5 words per part drawn from the 92-color list via the
- * persistent permutation stream (dbgen agg_str).  words has 5*count
- * entries; generation always starts at part 1 (sequential state). */
+/* p_name word ids: 5 words per part drawn from the 92-color list; per
+ * part the identity permutation is re-permuted with a 92-draw swap pass
+ * (the airlift port's agg_str semantics, q9 golden pin).  words has
+ * 5*count entries, rows from part 1. */
 void tpch_gen_part_name_words(double sf, int64_t count, uint8_t* words);
 /* index of a color word in the dists order, -1 if absent */
 int32_t tpch_color_id(const char* word);
+/* color word of an id into buf (<=16B incl NUL); returns length */
+int32_t tpch_color_name(int32_t id, char* buf);
 
 /* c_acctbal in exact cents, -99999..999999 (q22 golden pin) */
 void tpch_gen_customer_acctbal(double sf, int64_t start, int64_t count,

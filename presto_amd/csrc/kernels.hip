@@ -207,14 +207,29 @@ __device__ inline bool d_eval_preds(const pg_page& pg, const pg_pred* preds,
         if (c.null_mask && c.null_mask[i]) return false;
         bool ok;
         if (c.tag == PG_T_VARBIN) {
-            /* EQ/NE against a constant (VariableWidthBlock bytesEqual,
-             * AbstractVariableWidthBlock.java:95-99) */
+            /* EQ/NE/CONTAINS/PREFIX against a constant
+             * (VariableWidthBlock bytesEqual,
+             * AbstractVariableWidthBlock.java:95-99; LikeFunctions.java
+             * likeVarchar for wildcard-free %w% / w% patterns) */
             int32_t b0 = c.offsets[i], b1 = c.offsets[i + 1];
-            bool eq = (b1 - b0) == pr.slen;
             const uint8_t* d = (const uint8_t*)c.data + b0;
-            for (int j = 0; eq && j < pr.slen; j++)
-                eq = d[j] == (uint8_t)pr.sval[j];
-            ok = pr.op == PG_CMP_EQ ? eq : !eq;
+            int32_t n = b1 - b0;
+            if (pr.op == PG_CMP_CONTAINS || pr.op == PG_CMP_PREFIX) {
+                bool m = false;
+                int32_t last = pr.op == PG_CMP_PREFIX ? 0 : n - pr.slen;
+                for (int32_t s = 0; !m && s <= last; s++) {
+                    bool e = s + pr.slen <= n;
+                    for (int j = 0; e && j < pr.slen; j++)
+                        e = d[s + j] == (uint8_t)pr.sval[j];
+                    m = e;
+                }
+                ok = m;
+            } else {
+                bool eq = n == pr.slen;
+                for (int j = 0; eq && j < pr.slen; j++)
+                    eq = d[j] == (uint8_t)pr.sval[j];
+                ok = pr.op == PG_CMP_EQ ? eq : !eq;
+            }
         } else if (c.tag == PG_T_F64) {
             double v = ((const double*)c.data)[i];
             double x = pr.dval;

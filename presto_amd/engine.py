@@ -13,7 +13,8 @@ _SO = _HERE / "libpresto_gpu.so"
 
 # ---- enums (presto_gpu.h) ----
 T_U8, T_I32, T_I64, T_F64, T_VARBIN = 0, 1, 2, 3, 4
-CMP_LT, CMP_LE, CMP_GT, CMP_GE, CMP_EQ, CMP_NE = range(6)
+CMP_LT, CMP_LE, CMP_GT, CMP_GE, CMP_EQ, CMP_NE, CMP_CONTAINS, \
+    CMP_PREFIX = range(8)
 PROJ_IDENT, PROJ_DISC_PRICE, PROJ_CHARGE, PROJ_MUL = range(4)
 (AGG_COUNT, AGG_SUM_F64, AGG_SUM_DEC, AGG_SUM_I64,
  AGG_MIN, AGG_MAX) = range(6)
@@ -142,9 +143,27 @@ def lib():
     return _lib
 
 
+class Varbin:
+    """A host VariableWidthBlock column (VariableWidthBlock.java:48-61):
+    concatenated bytes + n+1 int32 offsets."""
+
+    def __init__(self, strings):
+        self.offsets = np.zeros(len(strings) + 1, np.int32)
+        for i, b in enumerate(strings):
+            self.offsets[i + 1] = self.offsets[i] + len(b)
+        joined = b"".join(strings)
+        self.data = np.frombuffer(joined, np.uint8).copy() if joined \
+            else np.empty(0, np.uint8)
+        self.n = len(strings)
+
+    def __len__(self):
+        return self.n
+
+
 class Page:
     """A Presto Page: named columns backed by numpy (host) or torch-cuda
-    (device) arrays.  Column order is the channel order."""
+    (device) arrays (Varbin for variable-width columns).  Column order is
+    the channel order."""
 
     def __init__(self, cols, n_rows=None):
         self.names = list(cols.keys())
@@ -162,7 +181,12 @@ class Page:
         for i, name in enumerate(self.names):
             a = self.cols[name]
             col = PgCol()
-            if isinstance(a, np.ndarray):
+            if isinstance(a, Varbin):
+                col.tag = T_VARBIN
+                col.on_device = 0
+                col.data = a.data.ctypes.data
+                col.offsets = a.offsets.ctypes.data
+            elif isinstance(a, np.ndarray):
                 col.tag = _NP_TAG[a.dtype]
                 col.on_device = 0
                 col.data = a.ctypes.data

@@ -1164,6 +1164,156 @@ def q19(part: Page, li: Page):
     return total
 
 
+Q9_YEAR_BOUNDS = (8035, 8401, 8766, 9131, 9496, 9862, 10227, 10592)
+
+
+def q9(part: Page, supp: Page, orders: Page, ps: Page, li: Page):
+    """Q9 product-type profit (q09.sql): LIKE '%green%' as a VARBIN
+    CONTAINS predicate building a dense part flag set; the partsupp
+    composite-key (partkey, suppkey) lookup is an emit join on partkey
+    followed by the suppkey equality (4 candidate suppliers per part);
+    supplier-nation and order-year attach by payload emit joins; the
+    (nation, year) grouping runs as per-group conjunctive aggregations
+    of the two exact sums (revenue - supplycost*qty).  Returns a 25x7
+    list of exact 1e-4 tick profits ([nation][year-1992])."""
+    import numpy as np
+    from .engine import lib
+
+    bg = PlanHashBuild()
+    bg.n_preds = 1
+    pgreen = Pred(part.channel("name"), CMP_CONTAINS, 0, 0.0)
+    pgreen.sval = b"green"
+    pgreen.slen = 5
+    bg.preds[0] = pgreen
+    bg.key_col = part.channel("partkey")
+    bg.semijoin_table = -1
+    bg.capacity_hint = part.n_rows
+    bg.key_set_only = 1
+    bg.dense_array = 1
+    og = Operator(OP_HASH_BUILD, bg)
+    og.add_input(part)
+    og.finish()
+
+    fl = PlanFilterProject()
+    fl.n_proj = 6
+    for i, name in enumerate(("partkey", "suppkey", "orderkey", "quantity",
+                              "extendedprice", "discount")):
+        fl.proj[i] = Proj(PROJ_IDENT, li.channel(name), 0, 0)
+    fl.semijoin_table = og.table()
+    fl.semijoin_col = li.channel("partkey")
+    f = Operator(OP_FILTER_PROJECT, fl)
+    f.add_input(li)
+    gli = f.get_output_raw()  # [pk, sk, ok, qty, ep, dc]
+
+    bp = PlanHashBuild()
+    bp.key_col = ps.channel("partkey")
+    bp.semijoin_table = -1
+    bp.n_payload = 2
+    bp.payload_col[0] = ps.channel("suppkey")
+    bp.payload_col[1] = ps.channel("supplycost")
+    bp.capacity_hint = ps.n_rows
+    ops_ = Operator(OP_HASH_BUILD, bp)
+    ops_.add_input(ps)
+    ops_.finish()
+
+    j1 = PlanLookupJoin()
+    j1.table = ops_.table()
+    j1.key_col = 0
+    j1.mode = 0
+    j1.n_emit = 6
+    for i in range(6):
+        j1.emit_probe_cols[i] = i
+    ja = Operator(OP_LOOKUP_JOIN, j1)
+    ja.add_input_raw(gli)
+    pa = ja.get_output_raw()  # [pk, sk, ok, qty, ep, dc, ps_sk, cost]
+
+    fe = PlanFilterProject()
+    fe.n_preds = 1
+    pe = Pred(1, CMP_EQ, 0, 0.0)
+    pe.rhs_col = 6 + 1
+    fe.preds[0] = pe
+    fe.n_proj = 6
+    for i, c in enumerate((1, 2, 3, 4, 5, 7)):
+        fe.proj[i] = Proj(PROJ_IDENT, c, 0, 0)
+    f2 = Operator(OP_FILTER_PROJECT, fe)
+    f2.add_input_raw(pa)
+    pb = f2.get_output_raw()  # [sk, ok, qty, ep, dc, cost]
+
+    bs = PlanHashBuild()
+    bs.key_col = supp.channel("suppkey")
+    bs.semijoin_table = -1
+    bs.n_payload = 1
+    bs.payload_col[0] = supp.channel("nationkey")
+    bs.capacity_hint = supp.n_rows
+    os_ = Operator(OP_HASH_BUILD, bs)
+    os_.add_input(supp)
+    os_.finish()
+
+    j2 = PlanLookupJoin()
+    j2.table = os_.table()
+    j2.key_col = 0
+    j2.mode = 0
+    j2.n_emit = 5
+    for i, c in enumerate((1, 2, 3, 4, 5)):
+        j2.emit_probe_cols[i] = c
+    jb = Operator(OP_LOOKUP_JOIN, j2)
+    jb.add_input_raw(pb)
+    pc = jb.get_output_raw()  # [ok, qty, ep, dc, cost, nat]
+
+    bo = PlanHashBuild()
+    bo.key_col = orders.channel("orderkey")
+    bo.semijoin_table = -1
+    bo.n_payload = 1
+    bo.payload_col[0] = orders.channel("orderdate")
+    bo.capacity_hint = orders.n_rows
+    oo = Operator(OP_HASH_BUILD, bo)
+    oo.add_input(orders)
+    oo.finish()
+
+    j3 = PlanLookupJoin()
+    j3.table = oo.table()
+    j3.key_col = 0
+    j3.mode = 0
+    j3.n_emit = 5
+    for i, c in enumerate((1, 2, 3, 4, 5)):
+        j3.emit_probe_cols[i] = c
+    jc = Operator(OP_LOOKUP_JOIN, j3)
+    jc.add_input_raw(pc)
+    pd = jc.get_output_raw()  # [qty, ep, dc, cost, nat, odate]
+
+    profit = [[0] * 7 for _ in range(25)]
+    for nat in range(25):
+        for y in range(7):
+            p = PlanHashAggSmall()
+            p.n_preds = 3
+            p.preds[0] = Pred(4, CMP_EQ, nat, 0.0)
+            p.preds[1] = Pred(5, CMP_GE, Q9_YEAR_BOUNDS[y], 0.0)
+            p.preds[2] = Pred(5, CMP_LT, Q9_YEAR_BOUNDS[y + 1], 0.0)
+            p.n_keys = 0
+            p.n_aggs = 2
+            p.aggs[0] = Agg(AGG_SUM_DEC, Proj(PROJ_DISC_PRICE, 1, 2, 0), 4)
+            p.aggs[1] = Agg(AGG_SUM_DEC, Proj(PROJ_MUL, 3, 0, 0), 4)
+            a = Operator(OP_HASH_AGG_SMALL, p)
+            a.add_input_raw(pd)
+            a.finish()
+            r = a.get_output(["rhi", "rlo", "chi", "clo"])
+            if len(r["rlo"]):
+                rev = (int(r["rhi"][0]) << 64) | int(np.uint64(r["rlo"][0]))
+                cst = (int(r["chi"][0]) << 64) | int(np.uint64(r["clo"][0]))
+                profit[nat][y] = rev - cst
+            a.destroy()
+
+    jc.destroy()
+    jb.destroy()
+    f2.destroy()
+    ja.destroy()
+    f.destroy()
+    for o in (og, ops_, os_, oo):
+        lib().c.pg_table_destroy(o.table())
+        o.destroy()
+    return profit
+
+
 Q22_CODE_NATIONS = (3, 7, 8, 13, 19, 20, 21)  # codes '13'..'31' ascending
 
 

@@ -552,3 +552,12 @@ OracleLib.gen_part_name_words = gen_part_name_words
 OracleLib.color_id = color_id
 OracleLib.q9 = q9
 OracleLib.nation_name = nation_name
+
+
+def color_name(self, cid):
+    buf = C.create_string_buffer(16)
+    self.lib.tpch_color_name(C.c_int32(cid), buf)
+    return buf.value.decode()
+
+
+OracleLib.color_name = color_name

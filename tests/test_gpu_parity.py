@@ -1032,6 +1032,67 @@ def test_q19_exact(P, oracle_lib):
     assert got > 0
 
 
+def test_q9_exact(P, oracle_lib):
+    """Q9 product-type profit — VARBIN CONTAINS ('%green%') flag set,
+    composite partsupp lookup via chain-emit + equality, payload joins,
+    (nation, year) aggregation grid vs the golden-pinned oracle."""
+    import numpy as np
+    sf = 0.1
+    li = oracle_lib.gen_lineitem2(sf)
+    lpk = oracle_lib.gen_lineitem_partkey(sf)
+    orders = oracle_lib.gen_orders(sf)
+    supp = oracle_lib.gen_supplier(sf)
+    ps = oracle_lib.gen_partsupp(sf)
+    words = oracle_lib.gen_part_name_words(sf)
+    names = [oracle_lib.color_name(i) for i in range(92)]
+    strings = [" ".join(names[w] for w in row).encode() for row in words]
+    n = len(strings)
+    got = P.pipelines.q9(
+        P.Page({"partkey": np.arange(1, n + 1, dtype=np.int64),
+                "name": P.Varbin(strings)}),
+        P.Page({"suppkey": supp["suppkey"], "nationkey": supp["nationkey"]}),
+        P.Page({"orderkey": orders["orderkey"],
+                "orderdate": orders["orderdate"]}),
+        P.Page({"partkey": ps["partkey"], "suppkey": ps["suppkey"],
+                "supplycost": ps["supplycost_cents"] / 100.0}),
+        P.Page({"partkey": lpk, "suppkey": li["suppkey"],
+                "orderkey": li["orderkey"], "quantity": li["quantity"],
+                "extendedprice": li["extendedprice"],
+                "discount": li["discount"]}))
+    gid = oracle_lib.color_id("green")
+    p_match = (words == gid).any(axis=1).astype(np.uint8)
+    exp = oracle_lib.q9(li, lpk, orders, supp, ps, p_match)
+    got_a = np.array(got, dtype=np.int64)
+    assert np.array_equal(got_a, exp)
+    assert np.abs(got_a).sum() > 0
+
+
+def test_varchar_contains_prefix(P):
+    """VARBIN CONTAINS / PREFIX predicates (LIKE '%w%' / 'w%' pushdown,
+    LikeFunctions.java likeVarchar) vs python substring semantics."""
+    rng = np.random.default_rng(62)
+    words = [b"forest", b"green", b"greenish", b"ivory", b"f", b"",
+             b"evergreen", b"gre", b"xgreeny", b"fores"]
+    strings = [b" ".join(rng.choice(words, rng.integers(1, 4)))
+               for _ in range(5000)]
+    ids = np.arange(len(strings), dtype=np.int64)
+    pg, keep = _varbin_page(P, {"id": ids}, strings)
+    for op_, pat, pyfn in (
+            (P.CMP_CONTAINS, b"green", lambda s: b"green" in s),
+            (P.CMP_PREFIX, b"forest", lambda s: s.startswith(b"forest"))):
+        plan = P.PlanFilterProject()
+        plan.n_preds = 1
+        plan.preds[0] = P.Pred(0, op_, 0, 0.0, pat, len(pat))
+        plan.n_proj = 1
+        plan.proj[0] = P.Proj(P.PROJ_IDENT, 1, 0, 0)
+        op = P.Operator(P.OP_FILTER_PROJECT, plan)
+        op.add_input_raw(pg)
+        out = op.get_output(["id"])
+        op.destroy()
+        expect = np.array([pyfn(s) for s in strings])
+        assert np.array_equal(out["id"], ids[expect])
+
+
 def test_q22_exact(P, oracle_lib):
     """Q22 global sales opportunity — anti-semijoin (NOT-EXISTS pushdown)
     + plan-constant average threshold vs the golden-pinned oracle."""
