@@ -125,6 +125,10 @@ typedef enum {
     PG_PROJ_DISC_PRICE = 1,   /* a * (1 - b)            (Q1/Q3 revenue) */
     PG_PROJ_CHARGE = 2,       /* a * (1 - b) * (1 + c)  (Q1 charge) */
     PG_PROJ_MUL = 3,          /* a * b                  (Q6 revenue) */
+    PG_PROJ_DIV = 4,          /* a / b (f64 emit only — e.g. a grouped
+                                 sum/count mean materialized for a
+                                 downstream compare, Q21's only-late
+                                 supplier) */
 } pg_proj_kind;
 
 typedef struct {

@@ -232,7 +232,14 @@ __device__ inline bool d_eval_preds(const pg_page& pg, const pg_pred* preds,
             }
         } else if (c.tag == PG_T_F64) {
             double v = ((const double*)c.data)[i];
-            double x = pr.dval;
+            double x;
+            if (pr.rhs_col > 0) {
+                const pg_col& rc = pg.cols[pr.rhs_col - 1];
+                if (rc.null_mask && rc.null_mask[i]) return false;
+                x = d_load_f64(rc, i);
+            } else {
+                x = pr.dval;
+            }
             switch (pr.op) {
                 case PG_CMP_LT: ok = v < x; break;
                 case PG_CMP_LE: ok = v <= x; break;
@@ -272,6 +279,7 @@ __device__ inline double d_eval_proj_f64(const pg_page& pg, const pg_proj& p,
     if (p.kind == PG_PROJ_IDENT) return a;
     double b = d_load_f64(pg.cols[p.b], i);
     if (p.kind == PG_PROJ_MUL) return a * b;
+    if (p.kind == PG_PROJ_DIV) return a / b;
     double v = a * (1.0 - b);
     if (p.kind == PG_PROJ_DISC_PRICE) return v;
     double c = d_load_f64(pg.cols[p.c], i);

@@ -16,8 +16,8 @@ from .engine import (
     Operator, Page, PlanFilterProject, PlanHashAggSmall, PlanHashBuild,
     PlanLookupJoin, PlanTopN, PlanPartition, Pred, Proj, Agg,
     OP_FILTER_PROJECT,
-    CMP_LE, CMP_LT, CMP_GT, CMP_GE, CMP_EQ,
-    PROJ_IDENT, PROJ_DISC_PRICE, PROJ_CHARGE, PROJ_MUL,
+    CMP_LE, CMP_LT, CMP_GT, CMP_GE, CMP_EQ, CMP_NE, CMP_CONTAINS, CMP_PREFIX,
+    PROJ_IDENT, PROJ_DISC_PRICE, PROJ_CHARGE, PROJ_MUL, PROJ_DIV,
     AGG_COUNT, AGG_SUM_F64, AGG_SUM_DEC,
     OP_HASH_AGG_SMALL, OP_HASH_BUILD, OP_LOOKUP_JOIN, OP_TOPN, OP_PARTITION,
 )
@@ -1162,6 +1162,174 @@ def q19(part: Page, li: Page):
     lib().c.pg_table_destroy(ob.table())
     ob.destroy()
     return total
+
+
+def q21(supp: Page, orders: Page, li: Page, limit=100):
+    """Q21 suppliers who kept orders waiting (q21.sql).  The correlated
+    EXISTS / NOT-EXISTS pair decomposes into per-order aggregates over
+    one fused-agg probe each: (sum suppkey, count) over all lines, the
+    'F'-line count, and (sum, sum*1e4, sum-of-squares, count) over LATE
+    lines.  An order qualifies when every line is 'F', some line has a
+    different supplier, and all late lines share one supplier — checked
+    with exact integer identities (cnt*sum_sq == sum*sum_ticks for the
+    zero-variance test; s* = sum/cnt is that supplier).  numwait(s) then
+    sums cnt_late over qualifying orders with s* = s (SAUDI ARABIA
+    semijoin).  Returns [(suppkey, numwait)] sorted (numwait desc,
+    suppkey asc) LIMIT limit."""
+    import numpy as np
+    from .engine import lib
+    tables = []
+
+    def agg_probe(preds, proj, scale):
+        b = PlanHashBuild()
+        b.key_col = orders.channel("orderkey")
+        b.semijoin_table = -1
+        b.capacity_hint = orders.n_rows + 64
+        b.agg_table = 1
+        ob = Operator(OP_HASH_BUILD, b)
+        ob.add_input(orders)
+        ob.finish()
+        tables.append(ob)
+        jp = PlanLookupJoin()
+        jp.table = ob.table()
+        jp.n_preds = len(preds)
+        for i, pr in enumerate(preds):
+            jp.preds[i] = pr
+        jp.key_col = li.channel("orderkey")
+        jp.mode = 1
+        jp.proj = proj
+        jp.dec_scale = scale
+        jo = Operator(OP_LOOKUP_JOIN, jp)
+        jo.add_input(li)
+        jo.finish()
+        g = jo.get_output_raw()  # [orderkey, sum_dec, sum_f64, cnt]
+        return jo, g
+
+    skc = li.channel("suppkey")
+    late = Pred(li.channel("receiptdate"), CMP_GT, 0, 0.0)
+    late.rhs_col = li.channel("commitdate") + 1
+    j_all, g_all = agg_probe([], Proj(PROJ_IDENT, skc, 0, 0), 0)
+    j_f, g_f = agg_probe([Pred(li.channel("linestatus"), CMP_EQ, ord("F"),
+                               0.0)], Proj(PROJ_IDENT, skc, 0, 0), 0)
+    j_lu, g_lu = agg_probe([late], Proj(PROJ_IDENT, skc, 0, 0), 0)
+    j_lt, g_lt = agg_probe([late], Proj(PROJ_IDENT, skc, 0, 0), 4)
+    j_sq, g_sq = agg_probe([late], Proj(PROJ_MUL, skc, skc, 0), 4)
+
+    def chain_join(probe_page, emit_cols, build_page, payload_cols):
+        b = PlanHashBuild()
+        b.key_col = 0
+        b.semijoin_table = -1
+        b.n_payload = len(payload_cols)
+        for i, c in enumerate(payload_cols):
+            b.payload_col[i] = c
+        b.capacity_hint = max(build_page.n_rows, 16)
+        ob = Operator(OP_HASH_BUILD, b)
+        ob.add_input_raw(build_page)
+        ob.finish()
+        tables.append(ob)
+        jp = PlanLookupJoin()
+        jp.table = ob.table()
+        jp.key_col = 0
+        jp.mode = 0
+        jp.n_emit = len(emit_cols)
+        for i, c in enumerate(emit_cols):
+            jp.emit_probe_cols[i] = c
+        jo = Operator(OP_LOOKUP_JOIN, jp)
+        jo.add_input_raw(probe_page)
+        return jo, jo.get_output_raw()
+
+    # combined per-order page
+    ja, pa = chain_join(g_all, (0, 1, 3), g_f, (3,))
+    # pa: [ok, sum_all, cnt_all, cnt_F]
+    jb, pb = chain_join(pa, (0, 1, 2, 3), g_lu, (1, 3))
+    # pb: [ok, sum_all, cnt_all, cnt_F, sum_lu, cnt_l]
+    jc, pc = chain_join(pb, (0, 1, 2, 3, 4, 5), g_lt, (1,))
+    # pc: [ok, sum_all, cnt_all, cnt_F, sum_lu, cnt_l, sum_lt]
+    jd, pd = chain_join(pc, (1, 2, 3, 4, 5, 6), g_sq, (1,))
+    # pd: [sum_all, cnt_all, cnt_F, sum_lu, cnt_l, sum_lt, sum_sq]
+
+    f1p = PlanFilterProject()
+    f1p.n_preds = 1
+    pf = Pred(2, CMP_EQ, 0, 0.0)  # cnt_F == cnt_all (all lines 'F')
+    pf.rhs_col = 1 + 1
+    f1p.preds[0] = pf
+    f1p.n_proj = 6
+    f1p.proj[0] = Proj(PROJ_DIV, 3, 4, 0)   # s* = sum_lu / cnt_l
+    f1p.proj[1] = Proj(PROJ_IDENT, 4, 0, 0)  # cnt_l
+    f1p.proj[2] = Proj(PROJ_MUL, 4, 6, 0)   # m1 = cnt_l * sum_sq
+    f1p.proj[3] = Proj(PROJ_MUL, 3, 5, 0)   # m2 = sum_lu * sum_lt
+    f1p.proj[4] = Proj(PROJ_MUL, 3, 1, 0)   # m3' = sum_lu * cnt_all
+    f1p.proj[5] = Proj(PROJ_MUL, 0, 4, 0)   # m4' = sum_all * cnt_l
+    f1 = Operator(OP_FILTER_PROJECT, f1p)
+    f1.add_input_raw(pd)
+    pe = f1.get_output_raw()  # [s*, cnt_l, m1, m2, m3', m4']
+    # multi-supplier: sum_all != s* * cnt_all  <=>  m4' != m3' (x cnt_l)
+
+    bsa = PlanHashBuild()
+    bsa.n_preds = 1
+    bsa.preds[0] = Pred(supp.channel("nationkey"), CMP_EQ, 20, 0.0)
+    bsa.key_col = supp.channel("suppkey")
+    bsa.semijoin_table = -1
+    bsa.capacity_hint = supp.n_rows
+    bsa.key_set_only = 1
+    bsa.dense_array = 1
+    osa = Operator(OP_HASH_BUILD, bsa)
+    osa.add_input(supp)
+    osa.finish()
+    tables.append(osa)
+
+    f2p = PlanFilterProject()
+    f2p.n_preds = 2
+    p1 = Pred(2, CMP_EQ, 0, 0.0)  # zero variance: m1 == m2
+    p1.rhs_col = 3 + 1
+    f2p.preds[0] = p1
+    p2 = Pred(4, CMP_NE, 0, 0.0)  # some different supplier: m3' != m4'
+    p2.rhs_col = 5 + 1
+    f2p.preds[1] = p2
+    f2p.n_proj = 2
+    f2p.proj[0] = Proj(PROJ_IDENT, 0, 0, 0)
+    f2p.proj[1] = Proj(PROJ_IDENT, 1, 0, 0)
+    f2p.semijoin_table = osa.table()
+    f2p.semijoin_col = 0
+    f2 = Operator(OP_FILTER_PROJECT, f2p)
+    f2.add_input_raw(pe)
+    pf2 = f2.get_output_raw()  # [s*, cnt_l] qualifying orders
+
+    bw = PlanHashBuild()
+    bw.n_preds = 1
+    bw.preds[0] = Pred(supp.channel("nationkey"), CMP_EQ, 20, 0.0)
+    bw.key_col = supp.channel("suppkey")
+    bw.semijoin_table = -1
+    bw.capacity_hint = supp.n_rows + 64
+    bw.agg_table = 1
+    ow = Operator(OP_HASH_BUILD, bw)
+    ow.add_input(supp)
+    ow.finish()
+    tables.append(ow)
+
+    jw = PlanLookupJoin()
+    jw.table = ow.table()
+    jw.key_col = 0
+    jw.mode = 1
+    jw.proj = Proj(PROJ_IDENT, 1, 0, 0)
+    jw.dec_scale = 0
+    jo = Operator(OP_LOOKUP_JOIN, jw)
+    jo.add_input_raw(pf2)
+    jo.finish()
+    out = jo.get_output(["suppkey", "numwait", "f64", "cnt"])
+    jo.destroy()
+    f2.destroy()
+    f1.destroy()
+    for j in (jd, jc, jb, ja, j_sq, j_lt, j_lu, j_f, j_all):
+        j.destroy()
+    for o in tables:
+        lib().c.pg_table_destroy(o.table())
+        o.destroy()
+    rows = sorted(
+        ((int(out["suppkey"][i]), int(out["numwait"][i]))
+         for i in range(len(out["suppkey"]))),
+        key=lambda r: (-r[1], r[0]))
+    return rows[:limit]
 
 
 Q9_YEAR_BOUNDS = (8035, 8401, 8766, 9131, 9496, 9862, 10227, 10592)

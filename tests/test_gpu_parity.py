@@ -1093,6 +1093,28 @@ def test_varchar_contains_prefix(P):
         assert np.array_equal(out["id"], ids[expect])
 
 
+def test_q21_exact(P, oracle_lib):
+    """Q21 waiting suppliers — per-order fused aggregates + exact
+    zero-variance / different-supplier identities + SAUDI semijoin vs
+    the golden-pinned oracle."""
+    sf = 0.1
+    supp = oracle_lib.gen_supplier(sf)
+    orders = oracle_lib.gen_orders(sf)
+    li = oracle_lib.gen_lineitem2(sf)
+    lid = oracle_lib.gen_lineitem_dates(sf)
+    got = P.pipelines.q21(
+        P.Page({"suppkey": supp["suppkey"], "nationkey": supp["nationkey"]}),
+        P.Page({"orderkey": orders["orderkey"]},
+               n_rows=len(orders["orderkey"])),
+        P.Page({"orderkey": li["orderkey"], "suppkey": li["suppkey"],
+                "linestatus": li["linestatus"],
+                "commitdate": lid["commitdate"],
+                "receiptdate": lid["receiptdate"]}))
+    exp = oracle_lib.q21(supp, li, lid)
+    assert got == exp
+    assert len(got) > 0
+
+
 def test_q22_exact(P, oracle_lib):
     """Q22 global sales opportunity — anti-semijoin (NOT-EXISTS pushdown)
     + plan-constant average threshold vs the golden-pinned oracle."""
