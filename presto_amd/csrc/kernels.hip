@@ -852,11 +852,18 @@ __global__ __launch_bounds__(256) void k_sel_count(pg_page pg,
         block_counts[blockIdx.x] = lds[0] + lds[1] + lds[2] + lds[3];
 }
 
+#define PG_PROJ_ROWID 100 /* internal: emit the source row index (used by
+                             the VARBIN two-pass emit) */
+
 /* emit one output column value for a selected row */
 __device__ inline void d_emit_val(const pg_page& pg, const pg_proj& p,
                                   int64_t i, void* out, int out_tag,
                                   int64_t pos)
 {
+    if (p.kind == PG_PROJ_ROWID) {
+        ((int64_t*)out)[pos] = i;
+        return;
+    }
     if (p.kind == PG_PROJ_IDENT) {
         const pg_col& c = pg.cols[p.a];
         switch (c.tag) {
@@ -927,6 +934,38 @@ __global__ __launch_bounds__(256) void k_sel_emit(pg_page pg,
         if (threadIdx.x == 0)
             running += wcnt[0] + wcnt[1] + wcnt[2] + wcnt[3];
         __syncthreads();
+    }
+}
+
+/* VARBIN emit pass 2: gather lengths then bytes through the selected
+ * source row indexes (two-pass VariableWidthBlockBuilder analog) */
+__global__ __launch_bounds__(256) void k_varbin_len(const int64_t* rowid,
+                                                    int64_t n,
+                                                    const int32_t* src_offs,
+                                                    int32_t* out_len)
+{
+    int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; j < n; j += stride) {
+        int64_t i = rowid[j];
+        out_len[j] = src_offs[i + 1] - src_offs[i];
+    }
+}
+
+__global__ __launch_bounds__(256) void k_varbin_gather(
+    const int64_t* rowid, int64_t n, const uint8_t* src_bytes,
+    const int32_t* src_offs, const int32_t* dst_offs, uint8_t* dst_bytes)
+{
+    /* one 64-lane wave per row: coalesced byte copies for short strings */
+    int64_t w = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    int lane = threadIdx.x & 63;
+    int64_t stride = ((int64_t)gridDim.x * blockDim.x) >> 6;
+    for (; w < n; w += stride) {
+        int64_t i = rowid[w];
+        int32_t s0 = src_offs[i], len = src_offs[i + 1] - s0;
+        int32_t d0 = dst_offs[w];
+        for (int32_t b = lane; b < len; b += 64)
+            dst_bytes[d0 + b] = src_bytes[s0 + b];
     }
 }
 
@@ -2072,14 +2111,24 @@ struct FilterOp : Op {
         op.pg.n_cols = plan.n_proj;
         emit_outs outs{};
         outs.n = plan.n_proj;
+        pg_plan_filter_project ep = plan; /* VARBIN slots -> ROWID emit */
+        std::vector<int> vb_cols;
         for (int o = 0; o < plan.n_proj; o++) {
             int tag = plan.proj[o].kind == PG_PROJ_IDENT
                           ? sp.pg.cols[plan.proj[o].a].tag
                           : PG_T_F64;
-            if (tag == PG_T_VARBIN)
-                throw std::runtime_error(
-                    "VARBIN projection/emit not supported in v1 "
-                    "(predicates and hashing only)");
+            if (tag == PG_T_VARBIN) {
+                /* pass 1 emits the source row index; pass 2 gathers the
+                 * variable-width bytes (see k_varbin_len/gather) */
+                vb_cols.push_back(o);
+                ep.proj[o].kind = PG_PROJ_ROWID;
+                op.dev.emplace_back();
+                op.dev.back().alloc((size_t)r.n * 8 + 8);
+                outs.ptr[o] = op.dev.back().p;
+                outs.tag[o] = PG_T_I64;
+                op.pg.cols[o].tag = PG_T_VARBIN;
+                continue;
+            }
             op.dev.emplace_back();
             op.dev.back().alloc((size_t)r.n * type_size(tag));
             op.pg.cols[o].tag = tag;
@@ -2089,7 +2138,43 @@ struct FilterOp : Op {
             outs.ptr[o] = op.dev.back().p;
             outs.tag[o] = tag;
         }
-        sel_emit(sp.pg, plan, semi, plan.semijoin_col, chunk, r, outs);
+        sel_emit(sp.pg, ep, semi, plan.semijoin_col, chunk, r, outs);
+        for (int o : vb_cols) {
+            const pg_col& src = sp.pg.cols[plan.proj[o].a];
+            const int64_t* rowid = (const int64_t*)outs.ptr[o];
+            DevBuf d_len;
+            d_len.alloc((size_t)r.n * 4 + 4);
+            hipLaunchKernelGGL(k_varbin_len, dim3(2048), dim3(256), 0,
+                               g_stream, rowid, r.n, src.offsets,
+                               (int32_t*)d_len.p);
+            std::vector<int32_t> h_len(r.n);
+            if (r.n)
+                CHKV(hipMemcpyAsync(h_len.data(), d_len.p, (size_t)r.n * 4,
+                                    hipMemcpyDeviceToHost, g_stream));
+            CHKV(hipStreamSynchronize(g_stream));
+            std::vector<int32_t> h_offs(r.n + 1);
+            h_offs[0] = 0;
+            for (int64_t j = 0; j < r.n; j++)
+                h_offs[j + 1] = h_offs[j] + h_len[j];
+            op.dev.emplace_back(); /* offsets */
+            op.dev.back().alloc(((size_t)r.n + 1) * 4);
+            void* offs_p = op.dev.back().p;
+            CHKV(hipMemcpyAsync(offs_p, h_offs.data(),
+                                ((size_t)r.n + 1) * 4, hipMemcpyHostToDevice,
+                                g_stream));
+            op.dev.emplace_back(); /* bytes */
+            op.dev.back().alloc((size_t)h_offs[r.n] + 1);
+            void* bytes_p = op.dev.back().p;
+            hipLaunchKernelGGL(k_varbin_gather, dim3(2048), dim3(256), 0,
+                               g_stream, rowid, r.n,
+                               (const uint8_t*)src.data, src.offsets,
+                               (const int32_t*)offs_p, (uint8_t*)bytes_p);
+            CHKV(hipStreamSynchronize(g_stream));
+            op.pg.cols[o].on_device = 1;
+            op.pg.cols[o].data = bytes_p;
+            op.pg.cols[o].offsets = (const int32_t*)offs_p;
+            op.pg.cols[o].null_mask = nullptr;
+        }
         outq.push_back(std::move(op));
     }
     void finish() override {}

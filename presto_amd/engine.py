@@ -159,6 +159,10 @@ class Varbin:
     def __len__(self):
         return self.n
 
+    def tolist(self):
+        return [self.data[self.offsets[i]:self.offsets[i + 1]].tobytes()
+                for i in range(self.n)]
+
 
 class Page:
     """A Presto Page: named columns backed by numpy (host) or torch-cuda
@@ -210,6 +214,25 @@ def _read_output_page(cpage, names=None):
     out = {}
     for i in range(cpage.n_cols):
         col = cpage.cols[i]
+        if col.tag == T_VARBIN:
+            offs = np.empty(n + 1, np.int32)
+            if col.on_device:
+                L.check(L.c.pg_memcpy_d2h(offs.ctypes.data, col.offsets,
+                                          (n + 1) * 4), "d2h")
+            else:
+                C.memmove(offs.ctypes.data, col.offsets, (n + 1) * 4)
+            nb = int(offs[n])
+            data = np.empty(nb, np.uint8)
+            if nb:
+                if col.on_device:
+                    L.check(L.c.pg_memcpy_d2h(data.ctypes.data, col.data,
+                                              nb), "d2h")
+                else:
+                    C.memmove(data.ctypes.data, col.data, nb)
+            v = Varbin.__new__(Varbin)
+            v.offsets, v.data, v.n = offs, data, n
+            out[names[i] if names else f"c{i}"] = v
+            continue
         dt = _TAG_NP[col.tag]
         a = np.empty(n, dt)
         nbytes = n * dt.itemsize

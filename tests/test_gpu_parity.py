@@ -91,6 +91,21 @@ def test_q3_exact(P, oracle_lib, sf01):
         assert got["orderdate"][i] == r.orderdate
 
 
+def test_q3_grace_spill(P, oracle_lib, sf01):
+    """HBM-overflow partitioned join (grace): orders+lineitem hash-
+    partitioned by orderkey and spilled to host, per-partition resident
+    joins, bounded TopN merge — identical to the resident pipeline and
+    the oracle."""
+    from presto_amd.spill import q3_grace
+    cust, orders, li = sf01["cust"], sf01["orders"], sf01["li"]
+    got = q3_grace(
+        P.Page({"custkey": cust["custkey"], "mktseg": cust["mktseg"]}),
+        P.Page({k: orders[k] for k in ("orderkey", "custkey", "orderdate")}),
+        _li_page(P, li), n_parts=8, mode="dec")
+    exp = oracle_lib.q3(cust, orders, li)
+    assert got == [(r.orderkey, r.revenue_1e4, r.orderdate) for r in exp]
+
+
 def test_q3_f64_mode(P, oracle_lib, sf01):
     """f64 revenue: exact fixed-point sum of f64 products — bit-equal to the
     oracle's fx128 accumulation."""
@@ -1091,6 +1106,34 @@ def test_varchar_contains_prefix(P):
         op.destroy()
         expect = np.array([pyfn(s) for s in strings])
         assert np.array_equal(out["id"], ids[expect])
+
+
+def test_varchar_projection_emit(P):
+    """VARBIN projection through ScanFilterAndProject: the two-pass
+    (rowid then length-prefix gather) emit must reproduce the selected
+    strings byte-exact, composed with preds + semijoin."""
+    import numpy as np
+    rng = np.random.default_rng(63)
+    strings = [bytes(rng.integers(97, 123, rng.integers(0, 40),
+                                  dtype=np.uint8)) for _ in range(30_000)]
+    ids = np.arange(len(strings), dtype=np.int64)
+    vals = rng.integers(0, 100, len(strings))
+    page = P.Page({"name": P.Varbin(strings), "id": ids,
+                   "val": vals.astype(np.int64)})
+    plan = P.PlanFilterProject()
+    plan.n_preds = 1
+    plan.preds[0] = P.Pred(2, P.CMP_LT, 50, 0.0)
+    plan.n_proj = 2
+    plan.proj[0] = P.Proj(P.PROJ_IDENT, 0, 0, 0)  # VARBIN emit
+    plan.proj[1] = P.Proj(P.PROJ_IDENT, 1, 0, 0)
+    op = P.Operator(P.OP_FILTER_PROJECT, plan)
+    op.add_input(page)
+    out = op.get_output(["name", "id"])
+    op.destroy()
+    mask = vals < 50
+    exp = [s for s, m in zip(strings, mask) if m]
+    assert out["name"].tolist() == exp
+    assert np.array_equal(out["id"], ids[mask])
 
 
 def test_q21_exact(P, oracle_lib):
