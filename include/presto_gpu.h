@@ -73,10 +73,11 @@ typedef enum {
     PG_T_F64 = 3, /* LongArrayBlock bits / DoubleType */
     PG_T_VARBIN = 4, /* VariableWidthBlock (VariableWidthBlock.java:48-61):
                         data = bytes, offsets = int32[n_rows+1] (element i
-                        spans bytes [offsets[i], offsets[i+1])).  v1 scope:
-                        EQ/NE-const predicates and hashing/partitioning
-                        (XxHash64 per AbstractVariableWidthBlock.java:102-
-                        105); VARBIN projection/emit is rejected. */
+                        spans bytes [offsets[i], offsets[i+1])).  Supports
+                        EQ/NE/CONTAINS/PREFIX predicates, hashing/
+                        partitioning (XxHash64 per
+                        AbstractVariableWidthBlock.java:102-105) and
+                        IDENT projection/emit (two-pass gather). */
 } pg_type;
 
 typedef struct {
@@ -84,7 +85,15 @@ typedef struct {
     int32_t on_device;      /* 1: data is a device pointer */
     void* data;             /* values array (VARBIN: the byte buffer) */
     const uint8_t* null_mask; /* optional, 1 byte/pos, 1 = null; may be NULL */
-    const int32_t* offsets; /* VARBIN only: n_rows+1 offsets */
+    const int32_t* offsets; /* VARBIN only: offsets (n_rows+1, or dict_n+1
+                               for dictionary columns) */
+    /* dictionary encoding (DictionaryBlock.java:60-86): when dict_ids is
+     * non-NULL the column is a dictionary VARBIN block — data/offsets
+     * describe the dict_n dictionary entries and dict_ids[n_rows] maps
+     * each position to an entry.  Predicates, hashing/partitioning and
+     * emit read through the ids. */
+    const int32_t* dict_ids;
+    int32_t dict_n;
 } pg_col;
 
 typedef struct {

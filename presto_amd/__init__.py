@@ -14,7 +14,8 @@ without a GPU) but creating any operator requires an AMD gfx950 GPU and
 fails loudly otherwise.
 """
 from .engine import (  # noqa: F401
-    lib, Operator, Page, Varbin, PlanFilterProject, PlanHashAggSmall,
+    lib, Operator, Page, Varbin, DictVarbin, PlanFilterProject,
+    PlanHashAggSmall,
     PlanHashBuild, PlanLookupJoin, PlanTopN, PlanPartition, Pred, Proj, Agg,
     CMP_LT, CMP_LE, CMP_GT, CMP_GE, CMP_EQ, CMP_NE, CMP_CONTAINS, CMP_PREFIX,
     PROJ_IDENT, PROJ_DISC_PRICE, PROJ_CHARGE, PROJ_MUL, PROJ_DIV,

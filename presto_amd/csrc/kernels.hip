@@ -210,8 +210,10 @@ __device__ inline bool d_eval_preds(const pg_page& pg, const pg_pred* preds,
             /* EQ/NE/CONTAINS/PREFIX against a constant
              * (VariableWidthBlock bytesEqual,
              * AbstractVariableWidthBlock.java:95-99; LikeFunctions.java
-             * likeVarchar for wildcard-free %w% / w% patterns) */
-            int32_t b0 = c.offsets[i], b1 = c.offsets[i + 1];
+             * likeVarchar for wildcard-free %w% / w% patterns).
+             * Dictionary blocks read through the ids. */
+            int64_t e = c.dict_ids ? (int64_t)c.dict_ids[i] : i;
+            int32_t b0 = c.offsets[e], b1 = c.offsets[e + 1];
             const uint8_t* d = (const uint8_t*)c.data + b0;
             int32_t n = b1 - b0;
             if (pr.op == PG_CMP_CONTAINS || pr.op == PG_CMP_PREFIX) {
@@ -942,19 +944,22 @@ __global__ __launch_bounds__(256) void k_sel_emit(pg_page pg,
 __global__ __launch_bounds__(256) void k_varbin_len(const int64_t* rowid,
                                                     int64_t n,
                                                     const int32_t* src_offs,
+                                                    const int32_t* dict_ids,
                                                     int32_t* out_len)
 {
     int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; j < n; j += stride) {
         int64_t i = rowid[j];
-        out_len[j] = src_offs[i + 1] - src_offs[i];
+        int64_t e = dict_ids ? (int64_t)dict_ids[i] : i;
+        out_len[j] = src_offs[e + 1] - src_offs[e];
     }
 }
 
 __global__ __launch_bounds__(256) void k_varbin_gather(
     const int64_t* rowid, int64_t n, const uint8_t* src_bytes,
-    const int32_t* src_offs, const int32_t* dst_offs, uint8_t* dst_bytes)
+    const int32_t* src_offs, const int32_t* dict_ids,
+    const int32_t* dst_offs, uint8_t* dst_bytes)
 {
     /* one 64-lane wave per row: coalesced byte copies for short strings */
     int64_t w = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
@@ -962,7 +967,8 @@ __global__ __launch_bounds__(256) void k_varbin_gather(
     int64_t stride = ((int64_t)gridDim.x * blockDim.x) >> 6;
     for (; w < n; w += stride) {
         int64_t i = rowid[w];
-        int32_t s0 = src_offs[i], len = src_offs[i + 1] - s0;
+        int64_t e = dict_ids ? (int64_t)dict_ids[i] : i;
+        int32_t s0 = src_offs[e], len = src_offs[e + 1] - s0;
         int32_t d0 = dst_offs[w];
         for (int32_t b = lane; b < len; b += 64)
             dst_bytes[d0 + b] = src_bytes[s0 + b];
@@ -1739,7 +1745,8 @@ __device__ inline int d_part_id(const pg_page& pg, int key_col, int np,
      * bytes (AbstractVariableWidthBlock.java:102-105) */
     const pg_col& c = pg.cols[key_col];
     if (c.tag == PG_T_VARBIN) {
-        int32_t b0 = c.offsets[i], b1 = c.offsets[i + 1];
+        int64_t e = c.dict_ids ? (int64_t)c.dict_ids[i] : i;
+        int32_t b0 = c.offsets[e], b1 = c.offsets[e + 1];
         return pg_partition(
             pg_xxh64((const uint8_t*)c.data + b0, (uint64_t)(b1 - b0)), np);
     }
@@ -1926,20 +1933,31 @@ struct StagedPage {
         for (int c = 0; c < in->n_cols; c++) {
             if (!in->cols[c].data) continue;
             if (in->cols[c].tag == PG_T_VARBIN && !in->cols[c].on_device) {
-                /* bytes buffer + offsets[n_rows+1] */
+                /* bytes buffer + offsets (n_rows+1, or dict_n+1 for
+                 * dictionary blocks) + optional dict ids */
                 const int32_t* ho = in->cols[c].offsets;
-                size_t nb = (size_t)ho[in->n_rows];
+                int64_t n_elem = in->cols[c].dict_ids ? in->cols[c].dict_n
+                                                      : in->n_rows;
+                size_t nb = (size_t)ho[n_elem];
                 bufs.emplace_back();
                 bufs.back().alloc(nb ? nb : 1);
                 CHKV(hipMemcpyAsync(bufs.back().p, in->cols[c].data, nb,
                                     hipMemcpyHostToDevice, g_stream));
                 pg.cols[c].data = bufs.back().p;
                 bufs.emplace_back();
-                bufs.back().alloc(((size_t)in->n_rows + 1) * 4);
+                bufs.back().alloc(((size_t)n_elem + 1) * 4);
                 CHKV(hipMemcpyAsync(bufs.back().p, ho,
-                                    ((size_t)in->n_rows + 1) * 4,
+                                    ((size_t)n_elem + 1) * 4,
                                     hipMemcpyHostToDevice, g_stream));
                 pg.cols[c].offsets = (const int32_t*)bufs.back().p;
+                if (in->cols[c].dict_ids) {
+                    bufs.emplace_back();
+                    bufs.back().alloc((size_t)in->n_rows * 4);
+                    CHKV(hipMemcpyAsync(bufs.back().p, in->cols[c].dict_ids,
+                                        (size_t)in->n_rows * 4,
+                                        hipMemcpyHostToDevice, g_stream));
+                    pg.cols[c].dict_ids = (const int32_t*)bufs.back().p;
+                }
                 pg.cols[c].on_device = 1;
                 if (in->cols[c].null_mask) {
                     bufs.emplace_back();
@@ -2146,7 +2164,7 @@ struct FilterOp : Op {
             d_len.alloc((size_t)r.n * 4 + 4);
             hipLaunchKernelGGL(k_varbin_len, dim3(2048), dim3(256), 0,
                                g_stream, rowid, r.n, src.offsets,
-                               (int32_t*)d_len.p);
+                               src.dict_ids, (int32_t*)d_len.p);
             std::vector<int32_t> h_len(r.n);
             if (r.n)
                 CHKV(hipMemcpyAsync(h_len.data(), d_len.p, (size_t)r.n * 4,
@@ -2168,7 +2186,8 @@ struct FilterOp : Op {
             hipLaunchKernelGGL(k_varbin_gather, dim3(2048), dim3(256), 0,
                                g_stream, rowid, r.n,
                                (const uint8_t*)src.data, src.offsets,
-                               (const int32_t*)offs_p, (uint8_t*)bytes_p);
+                               src.dict_ids, (const int32_t*)offs_p,
+                               (uint8_t*)bytes_p);
             CHKV(hipStreamSynchronize(g_stream));
             op.pg.cols[o].on_device = 1;
             op.pg.cols[o].data = bytes_p;
@@ -3457,9 +3476,14 @@ static void write_null_bits(ByteWriter& w, const uint8_t* mask, int64_t n)
 extern "C" pg_status pg_page_serialize(const pg_page* page, void* out,
                                        int64_t cap, int64_t* out_len)
 {
-    for (int c = 0; c < page->n_cols; c++)
+    for (int c = 0; c < page->n_cols; c++) {
         if (page->cols[c].on_device)
             return seterr("pg_page_serialize: host columns required");
+        if (page->cols[c].dict_ids)
+            return seterr(
+                "pg_page_serialize: dictionary blocks must be expanded "
+                "before serialization (v1 wire scope)");
+    }
     ByteWriter w{(uint8_t*)out, cap};
     /* metadata placeholder, filled after the body is written */
     int64_t meta_at = w.off;

@@ -29,7 +29,8 @@ _TAG_NP = {v: k for k, v in _NP_TAG.items()}
 class PgCol(C.Structure):
     _fields_ = [("tag", C.c_int32), ("on_device", C.c_int32),
                 ("data", C.c_void_p), ("null_mask", C.c_void_p),
-                ("offsets", C.c_void_p)]
+                ("offsets", C.c_void_p), ("dict_ids", C.c_void_p),
+                ("dict_n", C.c_int32)]
 
 
 class PgPage(C.Structure):
@@ -164,6 +165,19 @@ class Varbin:
                 for i in range(self.n)]
 
 
+class DictVarbin:
+    """A host DictionaryBlock over a Varbin dictionary
+    (DictionaryBlock.java:60-86): per-position int32 ids into the
+    dictionary entries."""
+
+    def __init__(self, dictionary_strings, ids):
+        self.dictionary = Varbin(dictionary_strings)
+        self.ids = np.ascontiguousarray(ids, np.int32)
+
+    def __len__(self):
+        return len(self.ids)
+
+
 class Page:
     """A Presto Page: named columns backed by numpy (host) or torch-cuda
     (device) arrays (Varbin for variable-width columns).  Column order is
@@ -190,6 +204,13 @@ class Page:
                 col.on_device = 0
                 col.data = a.data.ctypes.data
                 col.offsets = a.offsets.ctypes.data
+            elif isinstance(a, DictVarbin):
+                col.tag = T_VARBIN
+                col.on_device = 0
+                col.data = a.dictionary.data.ctypes.data
+                col.offsets = a.dictionary.offsets.ctypes.data
+                col.dict_ids = a.ids.ctypes.data
+                col.dict_n = a.dictionary.n
             elif isinstance(a, np.ndarray):
                 col.tag = _NP_TAG[a.dtype]
                 col.on_device = 0

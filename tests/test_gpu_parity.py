@@ -1136,6 +1136,60 @@ def test_varchar_projection_emit(P):
     assert np.array_equal(out["id"], ids[mask])
 
 
+def test_dictionary_block_path(P, oracle_lib):
+    """DictionaryBlock over a VARBIN dictionary (DictionaryBlock.java:
+    60-86): predicates, projection-emit expansion and partition math
+    must all read through the ids."""
+    import ctypes as CT
+    import numpy as np
+    rng = np.random.default_rng(64)
+    segs = [b"AUTOMOBILE", b"BUILDING", b"FURNITURE", b"MACHINERY",
+            b"HOUSEHOLD"]
+    ids_ = rng.integers(0, 5, 20_000).astype(np.int32)
+    rows = np.arange(20_000, dtype=np.int64)
+    page = P.Page({"seg": P.DictVarbin(segs, ids_), "row": rows})
+    # EQ predicate + VARBIN emit (expanded)
+    plan = P.PlanFilterProject()
+    plan.n_preds = 1
+    plan.preds[0] = P.Pred(0, P.CMP_EQ, 0, 0.0, b"BUILDING", 8)
+    plan.n_proj = 2
+    plan.proj[0] = P.Proj(P.PROJ_IDENT, 1, 0, 0)
+    plan.proj[1] = P.Proj(P.PROJ_IDENT, 0, 0, 0)
+    op = P.Operator(P.OP_FILTER_PROJECT, plan)
+    op.add_input(page)
+    out = op.get_output(["row", "seg"])
+    op.destroy()
+    mask = ids_ == 1
+    assert np.array_equal(out["row"], rows[mask])
+    assert out["seg"].tolist() == [b"BUILDING"] * int(mask.sum())
+    # CONTAINS through the dictionary
+    plan.preds[0] = P.Pred(0, P.CMP_CONTAINS, 0, 0.0, b"UR", 2)
+    op = P.Operator(P.OP_FILTER_PROJECT, plan)
+    op.add_input(page)
+    out = op.get_output(["row", "seg"])
+    op.destroy()
+    m2 = np.isin(ids_, [i for i, s in enumerate(segs) if b"UR" in s])
+    assert np.array_equal(out["row"], rows[m2])
+    # partition math == pg_partition(xxh64(expanded bytes))
+    L = oracle_lib.lib
+    L.oracle_xxh64.restype = CT.c_uint64
+    L.oracle_xxh64.argtypes = [CT.c_char_p, CT.c_int64]
+    nparts = 8
+    pp = P.PlanPartition()
+    pp.n_partitions = nparts
+    pp.key_col = 0
+    pp.n_emit = 1
+    pp.emit_cols[0] = 1
+    op = P.Operator(P.OP_PARTITION, pp)
+    op.add_input(page)
+    pages = [op.get_output(["row"]) for _ in range(nparts)]
+    op.destroy()
+    pid = np.array([oracle_lib.lib.oracle_partition(
+        L.oracle_xxh64(segs[i], len(segs[i])), nparts) for i in ids_])
+    for k in range(nparts):
+        assert np.array_equal(pages[k]["row"], rows[pid == k])
+
+
 def test_q21_exact(P, oracle_lib):
     """Q21 waiting suppliers — per-order fused aggregates + exact
     zero-variance / different-supplier identities + SAUDI semijoin vs
