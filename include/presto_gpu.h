@@ -323,6 +323,11 @@ pg_status pg_op_partition_counts(pg_op op, int64_t* counts, int32_t n);
 
 /* destroy a table explicitly (tables outlive their build op until freed) */
 pg_status pg_table_destroy(int64_t table);
+/* zero an agg table's accumulators so the table (keys + chains) can be
+ * reused by another fused-agg probe — the analog of reusing a lookup
+ * source across probe factories (HashBuilderOperator.java:534
+ * lendPartitionLookupSource is multi-consumer) */
+pg_status pg_table_reset_acc(int64_t table);
 
 /* ---- SerializedPage wire interop ----
  * Presto's exchange wire format, restated from

@@ -3376,6 +3376,26 @@ extern "C" pg_status pg_table_destroy(int64_t t)
     return PG_OK;
 }
 
+extern "C" pg_status pg_table_reset_acc(int64_t t)
+{
+    try {
+        Table* tbl;
+        {
+            std::lock_guard<std::mutex> lk(g_mu);
+            auto it = g_tables.find(t);
+            if (it == g_tables.end())
+                return seterr("pg_table_reset_acc: table not found");
+            tbl = it->second.get();
+        }
+        if (!tbl->acc.p) return PG_OK; /* lazily allocated: nothing yet */
+        CHK(hipMemsetAsync(tbl->acc.p, 0, tbl->acc.sz, g_stream));
+        CHK(hipStreamSynchronize(g_stream));
+        return PG_OK;
+    } catch (const std::exception& e) {
+        return seterr(e.what());
+    }
+}
+
 /* ================================================================== */
 /* SerializedPage wire interop — citations in presto_gpu.h            */
 /* ================================================================== */

@@ -121,6 +121,7 @@ class _Lib:
                                                   C.POINTER(C.c_int64),
                                                   C.c_int32]
         self.c.pg_table_destroy.argtypes = [C.c_int64]
+        self.c.pg_table_reset_acc.argtypes = [C.c_int64]
         self.c.pg_memcpy_d2h.argtypes = [C.c_void_p, C.c_void_p, C.c_int64]
         self.c.pg_memcpy_d2d.argtypes = [C.c_void_p, C.c_void_p, C.c_int64]
         self.c.pg_memcpy_h2d.argtypes = [C.c_void_p, C.c_void_p, C.c_int64]

@@ -1180,18 +1180,22 @@ def q21(supp: Page, orders: Page, li: Page, limit=100):
     from .engine import lib
     tables = []
 
+    # ONE orders key table reused by all five fused-agg probes
+    # (lendPartitionLookupSource semantics; accumulators reset between)
+    b = PlanHashBuild()
+    b.key_col = orders.channel("orderkey")
+    b.semijoin_table = -1
+    b.capacity_hint = orders.n_rows + 64
+    b.agg_table = 1
+    otbl = Operator(OP_HASH_BUILD, b)
+    otbl.add_input(orders)
+    otbl.finish()
+    tables.append(otbl)
+
     def agg_probe(preds, proj, scale):
-        b = PlanHashBuild()
-        b.key_col = orders.channel("orderkey")
-        b.semijoin_table = -1
-        b.capacity_hint = orders.n_rows + 64
-        b.agg_table = 1
-        ob = Operator(OP_HASH_BUILD, b)
-        ob.add_input(orders)
-        ob.finish()
-        tables.append(ob)
+        lib().c.pg_table_reset_acc(otbl.table())
         jp = PlanLookupJoin()
-        jp.table = ob.table()
+        jp.table = otbl.table()
         jp.n_preds = len(preds)
         for i, pr in enumerate(preds):
             jp.preds[i] = pr
@@ -1451,18 +1455,29 @@ def q9(part: Page, supp: Page, orders: Page, ps: Page, li: Page):
 
     profit = [[0] * 7 for _ in range(25)]
     for nat in range(25):
+        # one nation-split scan, then the 7 year aggregations run over
+        # the ~1/25-sized slice (175 full-page scans would dominate the
+        # whole pipeline — see profiles/r01_q9_sf10_kernel_stats.txt)
+        fn = PlanFilterProject()
+        fn.n_preds = 1
+        fn.preds[0] = Pred(4, CMP_EQ, nat, 0.0)
+        fn.n_proj = 5
+        for i, c in enumerate((0, 1, 2, 3, 5)):
+            fn.proj[i] = Proj(PROJ_IDENT, c, 0, 0)
+        fnat = Operator(OP_FILTER_PROJECT, fn)
+        fnat.add_input_raw(pd)
+        pnat = fnat.get_output_raw()  # [qty, ep, dc, cost, odate]
         for y in range(7):
             p = PlanHashAggSmall()
-            p.n_preds = 3
-            p.preds[0] = Pred(4, CMP_EQ, nat, 0.0)
-            p.preds[1] = Pred(5, CMP_GE, Q9_YEAR_BOUNDS[y], 0.0)
-            p.preds[2] = Pred(5, CMP_LT, Q9_YEAR_BOUNDS[y + 1], 0.0)
+            p.n_preds = 2
+            p.preds[0] = Pred(4, CMP_GE, Q9_YEAR_BOUNDS[y], 0.0)
+            p.preds[1] = Pred(4, CMP_LT, Q9_YEAR_BOUNDS[y + 1], 0.0)
             p.n_keys = 0
             p.n_aggs = 2
             p.aggs[0] = Agg(AGG_SUM_DEC, Proj(PROJ_DISC_PRICE, 1, 2, 0), 4)
             p.aggs[1] = Agg(AGG_SUM_DEC, Proj(PROJ_MUL, 3, 0, 0), 4)
             a = Operator(OP_HASH_AGG_SMALL, p)
-            a.add_input_raw(pd)
+            a.add_input_raw(pnat)
             a.finish()
             r = a.get_output(["rhi", "rlo", "chi", "clo"])
             if len(r["rlo"]):
@@ -1470,6 +1485,7 @@ def q9(part: Page, supp: Page, orders: Page, ps: Page, li: Page):
                 cst = (int(r["chi"][0]) << 64) | int(np.uint64(r["clo"][0]))
                 profit[nat][y] = rev - cst
             a.destroy()
+        fnat.destroy()
 
     jc.destroy()
     jb.destroy()
