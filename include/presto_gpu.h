@@ -278,6 +278,11 @@ typedef struct {
     int32_t table2_key_col;
     int32_t n_group_vals;
     uint8_t group_vals[8];
+    /* mode 1 only: skip the exact-f64 (fx128) accumulator legs when the
+     * consumer reads only the decimal sum + count (integer aggregates
+     * such as Q17/Q18/Q21 quantity sums — halves the atomic traffic of
+     * all-match probes) */
+    int32_t dec_only;
 } pg_plan_lookup_join;
 
 typedef struct {

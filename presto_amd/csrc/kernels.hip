@@ -1228,12 +1228,14 @@ __global__ __launch_bounds__(256) void k_probe_agg(
         ag.proj = plan.proj;
         ag.dec_scale = plan.dec_scale;
         int64_t ticks = d_eval_proj_dec(pg, ag, i);
-        double p = d_eval_proj_f64(pg, plan.proj, i);
-        uint64_t phi, plo;
-        fx128_from_f64(p, &phi, &plo);
         atomicAdd(&acc[s].dec, (unsigned long long)ticks);
-        unsigned long long old = atomicAdd(&acc[s].flo, plo);
-        atomicAdd(&acc[s].fhi, phi + (old > ~plo ? 1ull : 0ull));
+        if (!plan.dec_only) {
+            double p = d_eval_proj_f64(pg, plan.proj, i);
+            uint64_t phi, plo;
+            fx128_from_f64(p, &phi, &plo);
+            unsigned long long old = atomicAdd(&acc[s].flo, plo);
+            atomicAdd(&acc[s].fhi, phi + (old > ~plo ? 1ull : 0ull));
+        }
         atomicAdd(&acc[s].cnt, 1ull);
     }
 }

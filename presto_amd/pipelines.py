@@ -860,6 +860,7 @@ def q17(part: Page, li: Page):
     jp.mode = 1
     jp.proj = Proj(PROJ_IDENT, li.channel("quantity"), 0, 0)
     jp.dec_scale = 0
+    jp.dec_only = 1
     jo = Operator(OP_LOOKUP_JOIN, jp)
     jo.add_input(li)
     jo.finish()
@@ -993,6 +994,7 @@ def q11(supp: Page, ps: Page, n_part: int):
     jp.mode = 1
     jp.proj = Proj(PROJ_MUL, 1, 2, 0)
     jp.dec_scale = 4
+    jp.dec_only = 1
     jo = Operator(OP_LOOKUP_JOIN, jp)
     jo.add_input_raw(gpage)
     jo.finish()
@@ -1046,6 +1048,7 @@ def q18(orders: Page, li: Page, limit=100):
     jp.mode = 1
     jp.proj = Proj(PROJ_IDENT, li.channel("quantity"), 0, 0)
     jp.dec_scale = 0
+    jp.dec_only = 1
     jo = Operator(OP_LOOKUP_JOIN, jp)
     jo.add_input(li)
     jo.finish()
@@ -1192,32 +1195,51 @@ def q21(supp: Page, orders: Page, li: Page, limit=100):
     otbl.finish()
     tables.append(otbl)
 
-    def agg_probe(preds, proj, scale):
+    def agg_probe(page, key_col, proj, scale):
         lib().c.pg_table_reset_acc(otbl.table())
         jp = PlanLookupJoin()
         jp.table = otbl.table()
-        jp.n_preds = len(preds)
-        for i, pr in enumerate(preds):
-            jp.preds[i] = pr
-        jp.key_col = li.channel("orderkey")
+        jp.key_col = key_col
         jp.mode = 1
         jp.proj = proj
         jp.dec_scale = scale
+        jp.dec_only = 1  # integer sums only: skip the fx128 legs
         jo = Operator(OP_LOOKUP_JOIN, jp)
-        jo.add_input(li)
+        if isinstance(page, Page):
+            jo.add_input(page)
+        else:
+            jo.add_input_raw(page)
         jo.finish()
         g = jo.get_output_raw()  # [orderkey, sum_dec, sum_f64, cnt]
         return jo, g
 
+    # prefilter the late and 'F' rows once; the per-order probes then
+    # scan the narrow subsets instead of full lineitem three more times
+    def prefilter(preds, cols):
+        fp = PlanFilterProject()
+        fp.n_preds = len(preds)
+        for i, pr in enumerate(preds):
+            fp.preds[i] = pr
+        fp.n_proj = len(cols)
+        for i, c in enumerate(cols):
+            fp.proj[i] = Proj(PROJ_IDENT, c, 0, 0)
+        f = Operator(OP_FILTER_PROJECT, fp)
+        f.add_input(li)
+        return f, f.get_output_raw()
+
     skc = li.channel("suppkey")
     late = Pred(li.channel("receiptdate"), CMP_GT, 0, 0.0)
     late.rhs_col = li.channel("commitdate") + 1
-    j_all, g_all = agg_probe([], Proj(PROJ_IDENT, skc, 0, 0), 0)
-    j_f, g_f = agg_probe([Pred(li.channel("linestatus"), CMP_EQ, ord("F"),
-                               0.0)], Proj(PROJ_IDENT, skc, 0, 0), 0)
-    j_lu, g_lu = agg_probe([late], Proj(PROJ_IDENT, skc, 0, 0), 0)
-    j_lt, g_lt = agg_probe([late], Proj(PROJ_IDENT, skc, 0, 0), 4)
-    j_sq, g_sq = agg_probe([late], Proj(PROJ_MUL, skc, skc, 0), 4)
+    f_late, p_late = prefilter([late], (li.channel("orderkey"), skc))
+    f_stat, p_stat = prefilter(
+        [Pred(li.channel("linestatus"), CMP_EQ, ord("F"), 0.0)],
+        (li.channel("orderkey"),))
+    j_all, g_all = agg_probe(li, li.channel("orderkey"),
+                             Proj(PROJ_IDENT, skc, 0, 0), 0)
+    j_f, g_f = agg_probe(p_stat, 0, Proj(PROJ_IDENT, 0, 0, 0), 0)
+    j_lu, g_lu = agg_probe(p_late, 0, Proj(PROJ_IDENT, 1, 0, 0), 0)
+    j_lt, g_lt = agg_probe(p_late, 0, Proj(PROJ_IDENT, 1, 0, 0), 4)
+    j_sq, g_sq = agg_probe(p_late, 0, Proj(PROJ_MUL, 1, 1, 0), 4)
 
     def chain_join(probe_page, emit_cols, build_page, payload_cols):
         b = PlanHashBuild()
@@ -1317,6 +1339,7 @@ def q21(supp: Page, orders: Page, li: Page, limit=100):
     jw.mode = 1
     jw.proj = Proj(PROJ_IDENT, 1, 0, 0)
     jw.dec_scale = 0
+    jw.dec_only = 1
     jo = Operator(OP_LOOKUP_JOIN, jw)
     jo.add_input_raw(pf2)
     jo.finish()
@@ -1324,7 +1347,7 @@ def q21(supp: Page, orders: Page, li: Page, limit=100):
     jo.destroy()
     f2.destroy()
     f1.destroy()
-    for j in (jd, jc, jb, ja, j_sq, j_lt, j_lu, j_f, j_all):
+    for j in (jd, jc, jb, ja, j_sq, j_lt, j_lu, j_f, j_all, f_late, f_stat):
         j.destroy()
     for o in tables:
         lib().c.pg_table_destroy(o.table())
