@@ -1852,7 +1852,11 @@ struct BufPool {
     std::mutex mu;
     std::multimap<size_t, void*> free_bufs;
     size_t pooled = 0;
-    static const size_t CAP = 8ull << 30;
+    /* 288 GB HBM3E per GPU: cache aggressively — GB-scale hipMalloc
+     * costs ~100 ms (page-table setup) and an 8 GB cap thrashed the
+     * multi-probe pipelines (q21 at SF30: 109 ms of kernels inside a
+     * 990 ms wall — see profiles/r01_q21_sf10_kernel_stats_before.txt) */
+    static const size_t CAP = 96ull << 30;
     void* get(size_t n)
     {
         std::lock_guard<std::mutex> lk(mu);
