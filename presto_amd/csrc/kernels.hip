@@ -3474,6 +3474,7 @@ static const char* enc_name(int tag)
     switch (tag) {
         case PG_T_U8: return "BYTE_ARRAY";
         case PG_T_I32: return "INT_ARRAY";
+        case PG_T_VARBIN: return "VARIABLE_WIDTH";
         default: return "LONG_ARRAY"; /* I64 and F64 (bits) */
     }
 }
@@ -3523,6 +3524,18 @@ extern "C" pg_status pg_page_serialize(const pg_page* page, void* out,
         w.i32(nl);
         w.bytes(name, nl);
         w.i32((int32_t)page->n_rows);
+        if (col.tag == PG_T_VARBIN) {
+            /* VariableWidthBlockEncoding.writeBlock:37-58: cumulative
+             * lengths for every position, null bits, total length, then
+             * the raw byte region */
+            for (int64_t i = 0; i < page->n_rows; i++)
+                w.i32(col.offsets[i + 1]);
+            write_null_bits(w, col.null_mask, page->n_rows);
+            int32_t total = col.offsets[page->n_rows];
+            w.i32(total);
+            w.bytes(col.data, total);
+            continue;
+        }
         write_null_bits(w, col.null_mask, page->n_rows);
         size_t esz = type_size(col.tag);
         if (!col.null_mask) {
@@ -3596,10 +3609,42 @@ extern "C" pg_status pg_page_deserialize(const void* buf, int64_t len,
         if (!strcmp(name, "LONG_ARRAY")) tag = PG_T_I64;
         else if (!strcmp(name, "INT_ARRAY")) tag = PG_T_I32;
         else if (!strcmp(name, "BYTE_ARRAY")) tag = PG_T_U8;
+        else if (!strcmp(name, "VARIABLE_WIDTH")) tag = PG_T_VARBIN;
         else return seterr("deserialize: unsupported block encoding");
         int32_t n = r.i32();
         if (n != pos_count)
             return seterr("deserialize: block position count mismatch");
+        if (tag == PG_T_VARBIN) {
+            /* VariableWidthBlockEncoding.readBlock:62-76 */
+            int32_t* offs = (int32_t*)calloc((size_t)n + 1, 4);
+            for (int64_t i = 0; i < n; i++) offs[i + 1] = r.i32();
+            uint8_t vb_null = r.u8();
+            uint8_t* vmask = nullptr;
+            if (vb_null) {
+                vmask = (uint8_t*)calloc(n, 1);
+                for (int64_t base = 0; base < n; base += 8) {
+                    uint8_t v = r.u8();
+                    for (int64_t j = base; j < base + 8 && j < n; j++)
+                        vmask[j] = (v >> (7 - (j - base))) & 1;
+                }
+            }
+            int32_t nb = r.i32();
+            uint8_t* vb = (uint8_t*)calloc(nb ? nb : 1, 1);
+            const void* src = r.bytes(nb);
+            if (src) memcpy(vb, src, (size_t)nb);
+            if (!r.ok || nb != offs[n]) {
+                free(offs);
+                free(vmask);
+                free(vb);
+                return seterr("deserialize: bad VARIABLE_WIDTH block");
+            }
+            out->cols[c].tag = PG_T_VARBIN;
+            out->cols[c].on_device = 0;
+            out->cols[c].data = vb;
+            out->cols[c].offsets = offs;
+            out->cols[c].null_mask = vmask;
+            continue;
+        }
         uint8_t may_null = r.u8();
         uint8_t* mask = nullptr;
         if (may_null) {
@@ -3640,8 +3685,10 @@ extern "C" pg_status pg_page_free(pg_page* page)
     for (int c = 0; c < page->n_cols; c++) {
         free(page->cols[c].data);
         free((void*)page->cols[c].null_mask);
+        free((void*)page->cols[c].offsets);
         page->cols[c].data = nullptr;
         page->cols[c].null_mask = nullptr;
+        page->cols[c].offsets = nullptr;
     }
     return PG_OK;
 }
