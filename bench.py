@@ -415,7 +415,7 @@ def main():
                     g.f64_sum_charge).view(_np.int64)
             log(f"verify q1 sf={args.sf}: exact decimal + bitwise f64 OK "
                 f"({len(exp)} groups, {n_rows} rows)")
-        else:
+        elif args.query == "q3":
             li_host = {k: v.cpu().numpy() for k, v in cols.items()}
             oc = {k: v.cpu().numpy() for k, v in ocols.items()}
             cc = {k: v.cpu().numpy() for k, v in ccols.items()}
@@ -426,6 +426,20 @@ def main():
                 assert got3["revenue_1e4"][i] == r.revenue_1e4
                 assert got3["orderdate"][i] == r.orderdate
             log(f"verify q3 sf={args.sf}: top-10 exact OK")
+        else:  # q5
+            li_host = {k: v.cpu().numpy() for k, v in cols.items()}
+            exp5 = orc.q5({"custkey": cck, "nationkey": cnat},
+                          {"orderkey": ook, "custkey": ock,
+                           "orderdate": ood}, li_host,
+                          {"suppkey": ssk, "nationkey": snat})
+            got5 = step()
+            got_rows = sorted(
+                ((int(got5["nationkey"][i]), int(got5["rev_lo"][i]))
+                 for i in range(len(got5["nationkey"]))),
+                key=lambda r: (-r[1], r[0]))
+            exp_rows = [(r.nationkey, r.revenue_1e4) for r in exp5]
+            assert got_rows == exp_rows, (got_rows, exp_rows)
+            log(f"verify q5 sf={args.sf}: per-nation revenue exact OK")
     hot_ms = float(lib.c.pg_hot_max_ms())
     ms_per_step = elapsed / args.steps * 1000.0
     value = total_rows_per_step * args.steps / elapsed
