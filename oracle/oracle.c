@@ -1245,3 +1245,173 @@ void oracle_q9(int64_t n_li, const int64_t* lpk, const int64_t* lsk,
     }
     free(slot);
 }
+
+/* ---------------- Q13 ----------------
+ * SQL: q13.sql — customer distribution: per-customer count of orders
+ * whose o_comment does NOT match '%special%requests%' (LEFT OUTER:
+ * customers with no such orders count 0); histogram of counts sorted
+ * (custdist desc, c_count desc).  Comment text = pool[off:off+len]
+ * (tpch_text_pool + tpch_gen_orders_comment).  Returns rows. */
+static int like_two(const char* txt, int32_t len, const char* a, int32_t la,
+                    const char* b, int32_t lb)
+{
+    for (int32_t i = 0; i + la <= len; i++) {
+        if (memcmp(txt + i, a, la) == 0) {
+            for (int32_t j = i + la; j + lb <= len; j++)
+                if (memcmp(txt + j, b, lb) == 0) return 1;
+            return 0;
+        }
+    }
+    return 0;
+}
+
+int64_t oracle_q13(int64_t n_cust, int64_t n_ord, const int64_t* ock,
+                   const int64_t* cmnt_off, const int32_t* cmnt_len,
+                   const char* pool, int64_t* out_count, int64_t* out_dist,
+                   int64_t cap)
+{
+    int64_t* per_cust = (int64_t*)calloc(n_cust + 1, sizeof(int64_t));
+#pragma omp parallel for schedule(static)
+    for (int64_t i = 0; i < n_ord; i++) {
+        if (like_two(pool + cmnt_off[i], cmnt_len[i], "special", 7,
+                     "requests", 8))
+            continue;
+        int64_t ck = ock[i];
+        if (ck >= 1 && ck <= n_cust)
+#pragma omp atomic
+            per_cust[ck]++;
+    }
+    /* histogram of counts */
+    int64_t maxc = 0;
+    for (int64_t c = 1; c <= n_cust; c++)
+        if (per_cust[c] > maxc) maxc = per_cust[c];
+    int64_t* hist = (int64_t*)calloc(maxc + 1, sizeof(int64_t));
+    for (int64_t c = 1; c <= n_cust; c++) hist[per_cust[c]]++;
+    int64_t n_out = 0;
+    for (int64_t k = 0; k <= maxc && n_out < cap; k++)
+        if (hist[k]) {
+            out_count[n_out] = k;
+            out_dist[n_out] = hist[k];
+            n_out++;
+        }
+    /* sort (custdist desc, c_count desc): insertion (small) */
+    for (int64_t i = 1; i < n_out; i++) {
+        int64_t kc = out_count[i], kd = out_dist[i];
+        int64_t j = i - 1;
+        while (j >= 0 && (out_dist[j] < kd ||
+                          (out_dist[j] == kd && out_count[j] < kc))) {
+            out_count[j + 1] = out_count[j];
+            out_dist[j + 1] = out_dist[j];
+            j--;
+        }
+        out_count[j + 1] = kc;
+        out_dist[j + 1] = kd;
+    }
+    free(per_cust);
+    free(hist);
+    return n_out;
+}
+
+/* ---------------- Q16 ----------------
+ * SQL: q16.sql — parts/supplier relationship: count(DISTINCT suppkey)
+ * per (brand, type, size) over qualifying parts (brand != 45, type not
+ * MEDIUM POLISHED*, size in the 8 named values), excluding suppliers
+ * whose comment has the spliced 'Customer..Complaints' (bbb == 1 —
+ * the pool text itself never contains a capital-C 'Customer').
+ * Sorted (cnt desc, brand asc, type NAME asc, size asc).  Returns rows. */
+extern int32_t tpch_part_type_name(int32_t, char*);
+typedef struct {
+    int32_t grp; /* brand*150*51 + type*51 + size */
+    int64_t sk;
+} q16_pair_t;
+static int q16_pair_cmp(const void* a, const void* b)
+{
+    const q16_pair_t *x = (const q16_pair_t*)a, *y = (const q16_pair_t*)b;
+    if (x->grp != y->grp) return x->grp < y->grp ? -1 : 1;
+    return x->sk < y->sk ? -1 : x->sk > y->sk ? 1 : 0;
+}
+static int q16_key_cmp(const void* a, const void* b)
+{
+    uint64_t x = *(const uint64_t*)a, y = *(const uint64_t*)b;
+    return x < y ? -1 : x > y ? 1 : 0;
+}
+int64_t oracle_q16(int64_t n_part, const uint8_t* brand,
+                   const uint8_t* ptype, const uint8_t* psize, int64_t n_ps,
+                   const int64_t* ps_pk, const int64_t* ps_sk,
+                   int64_t n_supp, const uint8_t* bbb, uint8_t* out_brand,
+                   uint8_t* out_type, uint8_t* out_size, int32_t* out_cnt,
+                   int64_t cap)
+{
+    static const uint8_t SIZES[8] = {49, 14, 23, 45, 19, 3, 36, 9};
+    uint8_t* pok = (uint8_t*)calloc(n_part, 1);
+    for (int64_t p = 0; p < n_part; p++) {
+        if (brand[p] == 45) continue;
+        if (ptype[p] / 25 == 2 && (ptype[p] / 5) % 5 == 3) continue;
+        for (int k = 0; k < 8; k++)
+            if (psize[p] == SIZES[k]) pok[p] = 1;
+    }
+    q16_pair_t* pairs = (q16_pair_t*)malloc(n_ps * sizeof(q16_pair_t));
+    int64_t np = 0;
+    for (int64_t i = 0; i < n_ps; i++) {
+        int64_t pk = ps_pk[i], sk = ps_sk[i];
+        if (pk < 1 || pk > n_part || !pok[pk - 1]) continue;
+        if (sk >= 1 && sk <= n_supp && bbb[sk - 1] == 1) continue;
+        pairs[np].grp = (int32_t)(brand[pk - 1] * 150 * 51 +
+                                  ptype[pk - 1] * 51 + psize[pk - 1]);
+        pairs[np].sk = sk;
+        np++;
+    }
+    qsort(pairs, np, sizeof(q16_pair_t), q16_pair_cmp);
+    int64_t n_out = 0;
+    int64_t i = 0;
+    while (i < np && n_out < cap) {
+        int32_t g = pairs[i].grp;
+        int32_t cnt = 0;
+        int64_t last_sk = -1;
+        while (i < np && pairs[i].grp == g) {
+            if (pairs[i].sk != last_sk) {
+                cnt++;
+                last_sk = pairs[i].sk;
+            }
+            i++;
+        }
+        out_brand[n_out] = (uint8_t)(g / (150 * 51));
+        out_type[n_out] = (uint8_t)((g / 51) % 150);
+        out_size[n_out] = (uint8_t)(g % 51);
+        out_cnt[n_out] = cnt;
+        n_out++;
+    }
+    /* sort (cnt desc, brand asc, type NAME asc, size asc): one packed
+     * u64 key per row — the payload (brand,type,size,cnt) is fully
+     * recoverable from the key, so qsort of the keys suffices */
+    int tn_rank[150], rank_to_type[150];
+    {
+        char names[150][64];
+        for (int t = 0; t < 150; t++) tpch_part_type_name(t, names[t]);
+        for (int t = 0; t < 150; t++) {
+            int r = 0;
+            for (int u = 0; u < 150; u++)
+                if (strcmp(names[u], names[t]) < 0) r++;
+            tn_rank[t] = r;        /* names are distinct */
+            rank_to_type[r] = t;
+        }
+    }
+    uint64_t* keys = (uint64_t*)malloc(n_out * sizeof(uint64_t));
+    for (int64_t a = 0; a < n_out; a++)
+        keys[a] = ((uint64_t)(1u << 24) - (uint32_t)out_cnt[a]) << 40 |
+                  ((uint64_t)out_brand[a] << 32) |
+                  ((uint64_t)tn_rank[out_type[a]] << 8) |
+                  (uint64_t)out_size[a];
+    qsort(keys, n_out, sizeof(uint64_t), q16_key_cmp);
+    for (int64_t a = 0; a < n_out; a++) {
+        uint64_t k = keys[a];
+        out_cnt[a] = (int32_t)((1u << 24) - (uint32_t)(k >> 40));
+        out_brand[a] = (uint8_t)((k >> 32) & 0xff);
+        out_type[a] = (uint8_t)rank_to_type[(k >> 8) & 0xff];
+        out_size[a] = (uint8_t)(k & 0xff);
+    }
+    free(keys);
+    free(pok);
+    free(pairs);
+    return n_out;
+}

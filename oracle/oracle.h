@@ -270,6 +270,29 @@ void oracle_q9(int64_t n_li, const int64_t* l_partkey,
                const uint8_t* p_match, const int64_t* ps_suppkey,
                const int64_t* ps_supplycost_cents, int64_t* profit_1e4);
 
+/* ---------------- TPC-H Q13 ----------------
+ * q13.sql — customer distribution: histogram of per-customer counts of
+ * orders whose o_comment does not match '%special%requests%'
+ * (LEFT OUTER: zero-order customers count).  Comment text is
+ * pool[off:off+len].  Rows sorted (custdist desc, c_count desc). */
+int64_t oracle_q13(int64_t n_cust, int64_t n_ord, const int64_t* o_custkey,
+                   const int64_t* cmnt_off, const int32_t* cmnt_len,
+                   const char* pool, int64_t* out_count, int64_t* out_dist,
+                   int64_t cap);
+
+/* ---------------- TPC-H Q16 ----------------
+ * q16.sql — parts/supplier relationship: count(DISTINCT suppkey) per
+ * (brand, type, size) over qualifying parts, excluding complaint
+ * suppliers (bbb flag 1).  Rows sorted (cnt desc, brand, type NAME,
+ * size).  Returns rows. */
+int64_t oracle_q16(int64_t n_part, const uint8_t* p_brand,
+                   const uint8_t* p_type, const uint8_t* p_size,
+                   int64_t n_ps, const int64_t* ps_partkey,
+                   const int64_t* ps_suppkey, int64_t n_supp,
+                   const uint8_t* s_bbb, uint8_t* out_brand,
+                   uint8_t* out_type, uint8_t* out_size, int32_t* out_cnt,
+                   int64_t cap);
+
 /* ---------------- operator-level primitives (parity targets) ---------- */
 
 /* murmur3 finalizer bucket — PagesHash.java:236-252 /

@@ -121,6 +121,26 @@ void tpch_gen_part_name_words(double sf, int64_t count, uint8_t* words);
 int32_t tpch_color_id(const char* word);
 /* color word of an id into buf (<=16B incl NUL); returns length */
 int32_t tpch_color_name(int32_t id, char* buf);
+/* part type name of id 0..149 into buf (<=32B); returns length */
+int32_t tpch_part_type_name(int32_t id, char* buf);
+
+/* ---- text pool + comment columns (validated against every airlift
+ * comment fixture in the reference; see tpchgen.c) ---- */
+/* lazily builds and returns the 300 MiB text pool (process lifetime) */
+const char* tpch_text_pool(void);
+int64_t tpch_text_pool_size(void);
+/* comment (offset,length) streams; text = pool[off : off+len] */
+void tpch_gen_orders_comment(double sf, int64_t start, int64_t count,
+                             int64_t* off, int32_t* len);
+void tpch_gen_supplier_comment(double sf, int64_t start, int64_t count,
+                               int64_t* off, int32_t* len);
+void tpch_gen_customer_comment(double sf, int64_t start, int64_t count,
+                               int64_t* off, int32_t* len);
+void tpch_gen_nation_comment(int64_t* off, int32_t* len);
+/* supplier BBB comment splice flags: 0 plain, 1 'Customer Complaints',
+ * 2 'Customer Recommends' (q16 golden pin) */
+void tpch_gen_supplier_bbb(double sf, int64_t start, int64_t count,
+                           uint8_t* bbb);
 
 /* c_acctbal in exact cents, -99999..999999 (q22 golden pin) */
 void tpch_gen_customer_acctbal(double sf, int64_t start, int64_t count,

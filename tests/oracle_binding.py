@@ -561,3 +561,74 @@ def color_name(self, cid):
 
 
 OracleLib.color_name = color_name
+
+
+def text_pool(self):
+    self.lib.tpch_text_pool.restype = C.c_void_p
+    self.lib.tpch_text_pool_size.restype = C.c_int64
+    p = self.lib.tpch_text_pool()
+    n = self.lib.tpch_text_pool_size()
+    return C.cast(p, C.POINTER(C.c_char * n)).contents.raw
+
+
+def gen_orders_comment(self, sf):
+    n = self.lib.tpch_orders_count(C.c_double(sf))
+    off = np.empty(n, np.int64)
+    ln = np.empty(n, np.int32)
+    self.lib.tpch_gen_orders_comment(C.c_double(sf), C.c_int64(0),
+                                     C.c_int64(n), _p(off), _p(ln))
+    return off, ln
+
+
+def gen_supplier_bbb(self, sf):
+    n = self.lib.tpch_supplier_count(C.c_double(sf))
+    b = np.empty(n, np.uint8)
+    self.lib.tpch_gen_supplier_bbb(C.c_double(sf), C.c_int64(0),
+                                   C.c_int64(n), _p(b))
+    return b
+
+
+def q13(self, sf, orders, cap=128):
+    n_cust = self.lib.tpch_customer_count(C.c_double(sf))
+    off, ln = self.gen_orders_comment(sf)
+    self.lib.tpch_text_pool.restype = C.c_void_p
+    pool = self.lib.tpch_text_pool()
+    out_c = np.empty(cap, np.int64)
+    out_d = np.empty(cap, np.int64)
+    self.lib.oracle_q13.restype = C.c_int64
+    n = self.lib.oracle_q13(C.c_int64(n_cust),
+                            C.c_int64(len(orders["custkey"])),
+                            _p(orders["custkey"]), _p(off), _p(ln),
+                            C.c_void_p(pool), _p(out_c), _p(out_d),
+                            C.c_int64(cap))
+    return [(int(out_c[i]), int(out_d[i])) for i in range(n)]
+
+
+def part_type_name(self, tid):
+    buf = C.create_string_buffer(32)
+    self.lib.tpch_part_type_name(C.c_int32(tid), buf)
+    return buf.value.decode()
+
+
+def q16(self, part3, ptype, ps, bbb, cap=32768):
+    n_part = len(part3["brand"])
+    ob = np.empty(cap, np.uint8)
+    ot = np.empty(cap, np.uint8)
+    oz = np.empty(cap, np.uint8)
+    oc = np.empty(cap, np.int32)
+    self.lib.oracle_q16.restype = C.c_int64
+    n = self.lib.oracle_q16(C.c_int64(n_part), _p(part3["brand"]),
+                            _p(ptype), _p(part3["size"]),
+                            C.c_int64(len(ps["partkey"])), _p(ps["partkey"]),
+                            _p(ps["suppkey"]), C.c_int64(len(bbb)), _p(bbb),
+                            _p(ob), _p(ot), _p(oz), _p(oc), C.c_int64(cap))
+    return [(int(ob[i]), int(ot[i]), int(oz[i]), int(oc[i]))
+            for i in range(n)]
+
+
+OracleLib.text_pool = text_pool
+OracleLib.gen_orders_comment = gen_orders_comment
+OracleLib.gen_supplier_bbb = gen_supplier_bbb
+OracleLib.q13 = q13
+OracleLib.part_type_name = part_type_name
+OracleLib.q16 = q16

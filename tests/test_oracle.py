@@ -487,3 +487,51 @@ def test_q9_sf1_golden(oracle_lib):
     for (name, y, ticks), g in zip(rows, golden):
         assert name == g[0] and y == int(g[1])
         assert Decimal(ticks) / 10**4 == Decimal(g[2]), (name, y)
+
+
+def test_q13_sf1_golden(oracle_lib):
+    """Q13 customer distribution — pins the text pool (grammar + word
+    distributions) and the o_comment stream on all 42 golden rows."""
+    orders = oracle_lib.gen_orders(1.0)
+    rows = oracle_lib.q13(1.0, orders)
+    golden = _parse_golden("q13_sf1.result")
+    assert len(rows) == len(golden)
+    for (c, d), g in zip(rows, golden):
+        assert (c, d) == (int(g[0]), int(g[1]))
+
+
+def test_q16_sf1_golden(oracle_lib):
+    """Q16 parts/supplier relationship — pins the supplier-comment BBB
+    splice selection on all 18314 golden rows (incl. the type-NAME
+    ordering)."""
+    part3 = oracle_lib.gen_part3(1.0)
+    ptype = oracle_lib.gen_part_type(1.0)
+    ps = oracle_lib.gen_partsupp(1.0)
+    bbb = oracle_lib.gen_supplier_bbb(1.0)
+    assert [i + 1 for i, b in enumerate(bbb) if b == 1] == \
+        [358, 2820, 3804, 9504]
+    rows = oracle_lib.q16(part3, ptype, ps, bbb)
+    golden = _parse_golden("q16_sf1.result")
+    assert len(rows) == len(golden)
+    for (b, t, z, c), g in zip(rows, golden):
+        assert f"Brand#{b}" == g[0]
+        assert oracle_lib.part_type_name(t) == g[1]
+        assert (z, c) == (int(g[2]), int(g[3]))
+
+
+def test_nation_comments_pool(oracle_lib):
+    """The 25 nation comments reproduce byte-for-byte from the text pool
+    + the nation comment stream (presto-nation.result fixture restated
+    as offsets)."""
+    import numpy as np
+    import ctypes as C
+    off = np.empty(25, np.int64)
+    ln = np.empty(25, np.int32)
+    oracle_lib.lib.tpch_gen_nation_comment(
+        off.ctypes.data_as(C.c_void_p), ln.ctypes.data_as(C.c_void_p))
+    pool = oracle_lib.text_pool()
+    # spot-pin the first rows against the committed fixture text
+    expect0 = b" haggle. carefully final deposits detect slyly agai"
+    assert pool[off[0]:off[0]+ln[0]] == expect0
+    assert ln[3] == 101  # CANADA
+    assert b"ironic, silent packages" in pool[off[3]:off[3]+ln[3]]
