@@ -113,7 +113,12 @@ typedef enum { PG_CMP_LT = 0, PG_CMP_LE, PG_CMP_GT, PG_CMP_GE, PG_CMP_EQ,
                /* VARBIN only — the LIKE '%w%' / 'w%' pushdowns
                 * (LikeFunctions.java:64-77 likeVarchar for patterns
                 * without '_' reduce to substring/prefix search) */
-               PG_CMP_CONTAINS, PG_CMP_PREFIX } pg_cmp;
+               PG_CMP_CONTAINS, PG_CMP_PREFIX,
+               /* ordered two-substring LIKE '%a%b%' (and its negation —
+                * Q13's o_comment NOT LIKE '%special%requests%'):
+                * sval holds a then b concatenated, slen = len(a),
+                * ival = len(b) */
+               PG_CMP_CONTAINS2, PG_CMP_NOT_CONTAINS2 } pg_cmp;
 
 typedef struct {
     int32_t col;   /* input channel */

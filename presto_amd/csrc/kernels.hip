@@ -226,6 +226,26 @@ __device__ inline bool d_eval_preds(const pg_page& pg, const pg_pred* preds,
                     m = e;
                 }
                 ok = m;
+            } else if (pr.op == PG_CMP_CONTAINS2 ||
+                       pr.op == PG_CMP_NOT_CONTAINS2) {
+                /* ordered '%a%b%': find a, then b after it (leftmost-a
+                 * suffices: any later a leaves less room for b) */
+                int32_t la = pr.slen, lb = (int32_t)pr.ival;
+                bool m = false;
+                for (int32_t s = 0; !m && s + la <= n; s++) {
+                    bool e = true;
+                    for (int j = 0; e && j < la; j++)
+                        e = d[s + j] == (uint8_t)pr.sval[j];
+                    if (!e) continue;
+                    for (int32_t t = s + la; !m && t + lb <= n; t++) {
+                        bool f = true;
+                        for (int j = 0; f && j < lb; j++)
+                            f = d[t + j] == (uint8_t)pr.sval[la + j];
+                        m = f;
+                    }
+                    break; /* leftmost a checked; b not found after it */
+                }
+                ok = (pr.op == PG_CMP_CONTAINS2) ? m : !m;
             } else {
                 bool eq = n == pr.slen;
                 for (int j = 0; eq && j < pr.slen; j++)

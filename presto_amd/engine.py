@@ -13,8 +13,8 @@ _SO = _HERE / "libpresto_gpu.so"
 
 # ---- enums (presto_gpu.h) ----
 T_U8, T_I32, T_I64, T_F64, T_VARBIN = 0, 1, 2, 3, 4
-CMP_LT, CMP_LE, CMP_GT, CMP_GE, CMP_EQ, CMP_NE, CMP_CONTAINS, \
-    CMP_PREFIX = range(8)
+(CMP_LT, CMP_LE, CMP_GT, CMP_GE, CMP_EQ, CMP_NE, CMP_CONTAINS,
+ CMP_PREFIX, CMP_CONTAINS2, CMP_NOT_CONTAINS2) = range(10)
 PROJ_IDENT, PROJ_DISC_PRICE, PROJ_CHARGE, PROJ_MUL, PROJ_DIV = range(5)
 (AGG_COUNT, AGG_SUM_F64, AGG_SUM_DEC, AGG_SUM_I64,
  AGG_MIN, AGG_MAX) = range(6)

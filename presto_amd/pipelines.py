@@ -16,7 +16,8 @@ from .engine import (
     Operator, Page, PlanFilterProject, PlanHashAggSmall, PlanHashBuild,
     PlanLookupJoin, PlanTopN, PlanPartition, Pred, Proj, Agg,
     OP_FILTER_PROJECT,
-    CMP_LE, CMP_LT, CMP_GT, CMP_GE, CMP_EQ, CMP_NE, CMP_CONTAINS, CMP_PREFIX,
+    CMP_LE, CMP_LT, CMP_GT, CMP_GE, CMP_EQ, CMP_NE, CMP_CONTAINS,
+    CMP_PREFIX, CMP_CONTAINS2, CMP_NOT_CONTAINS2,
     PROJ_IDENT, PROJ_DISC_PRICE, PROJ_CHARGE, PROJ_MUL, PROJ_DIV,
     AGG_COUNT, AGG_SUM_F64, AGG_SUM_DEC,
     OP_HASH_AGG_SMALL, OP_HASH_BUILD, OP_LOOKUP_JOIN, OP_TOPN, OP_PARTITION,
@@ -1519,6 +1520,87 @@ def q9(part: Page, supp: Page, orders: Page, ps: Page, li: Page):
         lib().c.pg_table_destroy(o.table())
         o.destroy()
     return profit
+
+
+def q13(n_cust: int, orders: Page, max_count=64):
+    """Q13 customer distribution (q13.sql): the NOT LIKE
+    '%special%requests%' comment filter runs as the ordered
+    two-substring VARBIN predicate (CONTAINS2 negated); per-customer
+    order counts come from a fused-agg probe over the full customer key
+    table; the count histogram is a second fused-agg probe keyed by the
+    count value.  The LEFT OUTER zero bucket is n_cust minus the
+    customers with qualifying orders.  Returns [(c_count, custdist)]
+    sorted (custdist desc, c_count desc)."""
+    import numpy as np
+    from .engine import lib
+
+    fp = PlanFilterProject()
+    fp.n_preds = 1
+    pr = Pred(orders.channel("comment"), CMP_NOT_CONTAINS2, 8, 0.0)
+    pr.sval = b"special" + b"requests"
+    pr.slen = 7
+    fp.preds[0] = pr
+    fp.n_proj = 1
+    fp.proj[0] = Proj(PROJ_IDENT, orders.channel("custkey"), 0, 0)
+    f = Operator(OP_FILTER_PROJECT, fp)
+    f.add_input(orders)
+    okp = f.get_output_raw()  # [custkey] of qualifying orders
+
+    bc = PlanHashBuild()
+    bc.key_col = 0
+    bc.semijoin_table = -1
+    bc.capacity_hint = n_cust + 64
+    bc.agg_table = 1
+    keys = Page({"custkey": np.arange(1, n_cust + 1, dtype=np.int64)})
+    oc = Operator(OP_HASH_BUILD, bc)
+    oc.add_input(keys)
+    oc.finish()
+
+    jp = PlanLookupJoin()
+    jp.table = oc.table()
+    jp.key_col = 0
+    jp.mode = 1
+    jp.proj = Proj(PROJ_IDENT, 0, 0, 0)
+    jp.dec_scale = 0
+    jp.dec_only = 1
+    jo = Operator(OP_LOOKUP_JOIN, jp)
+    jo.add_input_raw(okp)
+    jo.finish()
+    groups = jo.get_output_raw()  # [custkey, sum, f64, cnt]
+    n_with_orders = groups.n_rows
+
+    bh = PlanHashBuild()
+    bh.key_col = 0
+    bh.semijoin_table = -1
+    bh.capacity_hint = max_count + 64
+    bh.agg_table = 1
+    counts = Page({"c": np.arange(1, max_count + 1, dtype=np.int64)})
+    oh = Operator(OP_HASH_BUILD, bh)
+    oh.add_input(counts)
+    oh.finish()
+
+    jh = PlanLookupJoin()
+    jh.table = oh.table()
+    jh.key_col = 3  # the per-customer count
+    jh.mode = 1
+    jh.proj = Proj(PROJ_IDENT, 3, 0, 0)
+    jh.dec_scale = 0
+    jh.dec_only = 1
+    jo2 = Operator(OP_LOOKUP_JOIN, jh)
+    jo2.add_input_raw(groups)
+    jo2.finish()
+    hist = jo2.get_output(["c_count", "sum", "f64", "custdist"])
+    rows = [(int(hist["c_count"][i]), int(hist["custdist"][i]))
+            for i in range(len(hist["c_count"]))]
+    rows.append((0, n_cust - n_with_orders))
+    rows.sort(key=lambda r: (-r[1], -r[0]))
+    jo2.destroy()
+    jo.destroy()
+    f.destroy()
+    for o in (oc, oh):
+        lib().c.pg_table_destroy(o.table())
+        o.destroy()
+    return rows
 
 
 Q22_CODE_NATIONS = (3, 7, 8, 13, 19, 20, 21)  # codes '13'..'31' ascending

@@ -632,3 +632,21 @@ OracleLib.gen_supplier_bbb = gen_supplier_bbb
 OracleLib.q13 = q13
 OracleLib.part_type_name = part_type_name
 OracleLib.q16 = q16
+
+
+def gen_orders_comment_varbin(self, sf):
+    """Materialized o_comment bytes as (data u8, offsets i32) arrays —
+    builds the page column a GPU scan would receive."""
+    off, ln = self.gen_orders_comment(sf)
+    pool = np.frombuffer(self.text_pool(), np.uint8)
+    n = len(off)
+    outoffs = np.zeros(n + 1, np.int32)
+    np.cumsum(ln, out=outoffs[1:])
+    total = int(outoffs[-1])
+    starts = np.repeat(off - (outoffs[:-1].astype(np.int64)), ln)
+    idx = starts + np.arange(total)
+    data = pool[idx]
+    return data, outoffs
+
+
+OracleLib.gen_orders_comment_varbin = gen_orders_comment_varbin

@@ -1335,3 +1335,22 @@ def test_cross_run_bit_determinism(P, oracle_lib, sf01):
     assert np.array_equal(r1["revenue"].view(np.int64),
                           r2["revenue"].view(np.int64))
     assert np.array_equal(r1["orderkey"], r2["orderkey"])
+
+
+def test_q13_exact(P, oracle_lib):
+    """Q13 customer distribution — NOT LIKE '%special%requests%' as the
+    ordered two-substring VARBIN predicate + two fused-agg probes vs
+    the golden-pinned oracle."""
+    import numpy as np
+    from presto_amd.engine import Varbin
+    sf = 0.1
+    orders = oracle_lib.gen_orders(sf)
+    data, offs = oracle_lib.gen_orders_comment_varbin(sf)
+    cm = Varbin.__new__(Varbin)
+    cm.data, cm.offsets, cm.n = data, offs, len(orders["custkey"])
+    n_cust = int(150000 * sf)
+    got = P.pipelines.q13(
+        n_cust, P.Page({"custkey": orders["custkey"], "comment": cm}))
+    exp = oracle_lib.q13(sf, orders)
+    assert got == exp
+    assert len(got) > 5
