@@ -125,7 +125,8 @@ typedef struct {
     int32_t op;    /* pg_cmp */
     int64_t ival;  /* compare value for integer columns */
     double dval;   /* compare value for f64 columns */
-    char sval[16]; /* VARBIN: compare bytes (EQ/NE/CONTAINS/PREFIX) */
+    char sval[24]; /* VARBIN: compare bytes (EQ/NE/CONTAINS/PREFIX;
+                      CONTAINS2 packs both patterns) */
     int32_t slen;
     int32_t rhs_col; /* 0: compare against the constant; k>0: compare
                         against integer channel k-1 (filter expression
@@ -143,6 +144,12 @@ typedef enum {
                                  sum/count mean materialized for a
                                  downstream compare, Q21's only-late
                                  supplier) */
+    PG_PROJ_KEYSHL = 5,       /* (a << c) | b — composite grouping keys
+                                 (c = shift constant, i64 emit; the
+                                 codegen analog of CombineHashFunction
+                                 key packing for multi-channel group-bys,
+                                 e.g. Q16's (brand,type,size,suppkey)) */
+    PG_PROJ_SHR = 6,          /* a >> c — composite key extraction */
 } pg_proj_kind;
 
 typedef struct {

@@ -886,6 +886,16 @@ __device__ inline void d_emit_val(const pg_page& pg, const pg_proj& p,
         ((int64_t*)out)[pos] = i;
         return;
     }
+    if (p.kind == PG_PROJ_KEYSHL) {
+        int64_t va = d_load_i64(pg.cols[p.a], i);
+        int64_t vb = d_load_i64(pg.cols[p.b], i);
+        ((int64_t*)out)[pos] = (va << p.c) | vb;
+        return;
+    }
+    if (p.kind == PG_PROJ_SHR) {
+        ((int64_t*)out)[pos] = d_load_i64(pg.cols[p.a], i) >> p.c;
+        return;
+    }
     if (p.kind == PG_PROJ_IDENT) {
         const pg_col& c = pg.cols[p.a];
         switch (c.tag) {
@@ -2160,6 +2170,9 @@ struct FilterOp : Op {
         for (int o = 0; o < plan.n_proj; o++) {
             int tag = plan.proj[o].kind == PG_PROJ_IDENT
                           ? sp.pg.cols[plan.proj[o].a].tag
+                      : (plan.proj[o].kind == PG_PROJ_KEYSHL ||
+                         plan.proj[o].kind == PG_PROJ_SHR)
+                          ? PG_T_I64
                           : PG_T_F64;
             if (tag == PG_T_VARBIN) {
                 /* pass 1 emits the source row index; pass 2 gathers the

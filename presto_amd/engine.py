@@ -15,7 +15,8 @@ _SO = _HERE / "libpresto_gpu.so"
 T_U8, T_I32, T_I64, T_F64, T_VARBIN = 0, 1, 2, 3, 4
 (CMP_LT, CMP_LE, CMP_GT, CMP_GE, CMP_EQ, CMP_NE, CMP_CONTAINS,
  CMP_PREFIX, CMP_CONTAINS2, CMP_NOT_CONTAINS2) = range(10)
-PROJ_IDENT, PROJ_DISC_PRICE, PROJ_CHARGE, PROJ_MUL, PROJ_DIV = range(5)
+(PROJ_IDENT, PROJ_DISC_PRICE, PROJ_CHARGE, PROJ_MUL, PROJ_DIV,
+ PROJ_KEYSHL, PROJ_SHR) = range(7)
 (AGG_COUNT, AGG_SUM_F64, AGG_SUM_DEC, AGG_SUM_I64,
  AGG_MIN, AGG_MAX) = range(6)
 (OP_FILTER_PROJECT, OP_HASH_AGG_SMALL, OP_HASH_BUILD, OP_LOOKUP_JOIN,
@@ -40,7 +41,7 @@ class PgPage(C.Structure):
 
 class Pred(C.Structure):
     _fields_ = [("col", C.c_int32), ("op", C.c_int32), ("ival", C.c_int64),
-                ("dval", C.c_double), ("sval", C.c_char * 16),
+                ("dval", C.c_double), ("sval", C.c_char * 24),
                 ("slen", C.c_int32), ("rhs_col", C.c_int32)]
 
 

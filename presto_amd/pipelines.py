@@ -19,6 +19,7 @@ from .engine import (
     CMP_LE, CMP_LT, CMP_GT, CMP_GE, CMP_EQ, CMP_NE, CMP_CONTAINS,
     CMP_PREFIX, CMP_CONTAINS2, CMP_NOT_CONTAINS2,
     PROJ_IDENT, PROJ_DISC_PRICE, PROJ_CHARGE, PROJ_MUL, PROJ_DIV,
+    PROJ_KEYSHL, PROJ_SHR,
     AGG_COUNT, AGG_SUM_F64, AGG_SUM_DEC,
     OP_HASH_AGG_SMALL, OP_HASH_BUILD, OP_LOOKUP_JOIN, OP_TOPN, OP_PARTITION,
 )
@@ -1598,6 +1599,186 @@ def q13(n_cust: int, orders: Page, max_count=64):
     jo.destroy()
     f.destroy()
     for o in (oc, oh):
+        lib().c.pg_table_destroy(o.table())
+        o.destroy()
+    return rows
+
+
+def q16(part: Page, ps: Page, supp: Page, type_name):
+    """Q16 parts/supplier relationship (q16.sql).  The disjunctive part
+    qualifiers (size IN 8 values, type NOT LIKE 'MEDIUM POLISHED%')
+    build one dense flag set through repeated conjunctive fill passes;
+    complaint suppliers come from the CONTAINS2 'Customer..Complaints'
+    comment predicate (anti-semijoin); count(DISTINCT suppkey) per
+    (brand, type, size) runs as composite-key (KEYSHL) fused-agg
+    dedup + a second grouping probe.  type_name maps a type id to its
+    display string for the final ORDER BY.  Returns
+    [(brand, type_id, size, supplier_cnt)] in golden order."""
+    import numpy as np
+    from .engine import lib
+
+    # qualifying-part flag set: the disjunction (8 sizes x NOT a type
+    # range) decomposes into 16 conjunctive filter passes whose partkey
+    # outputs feed ONE dense flag-set build (flags OR across add_input)
+    SIZES = (49, 14, 23, 45, 19, 3, 36, 9)
+    brandc = part.channel("brand")
+    typec = part.channel("type_id")
+    sizec = part.channel("size")
+    qual_pages = []
+    fops = []
+    for v in SIZES:
+        for trange in ((CMP_LE, 64), (CMP_GE, 70)):
+            fp = PlanFilterProject()
+            fp.n_preds = 3
+            fp.preds[0] = Pred(sizec, CMP_EQ, v, 0.0)
+            fp.preds[1] = Pred(brandc, CMP_NE, 45, 0.0)
+            fp.preds[2] = Pred(typec, trange[0], trange[1], 0.0)
+            fp.n_proj = 1
+            fp.proj[0] = Proj(PROJ_IDENT, part.channel("partkey"), 0, 0)
+            f = Operator(OP_FILTER_PROJECT, fp)
+            f.add_input(part)
+            qual_pages.append(f.get_output_raw())
+            fops.append(f)
+    bqual = PlanHashBuild()
+    bqual.key_col = 0
+    bqual.semijoin_table = -1
+    bqual.capacity_hint = part.n_rows
+    bqual.key_set_only = 1
+    bqual.dense_array = 1
+    oqual = Operator(OP_HASH_BUILD, bqual)
+    for pg_ in qual_pages:
+        oqual.add_input_raw(pg_)
+    oqual.finish()
+    for f in fops:
+        f.destroy()
+
+    # complaint suppliers: CONTAINS2('Customer','Complaints') flag set
+    bc = PlanHashBuild()
+    bc.n_preds = 1
+    pc = Pred(supp.channel("comment"), CMP_CONTAINS2, 10, 0.0)
+    pc.sval = b"Customer" + b"Complaints"
+    pc.slen = 8
+    bc.preds[0] = pc
+    bc.key_col = supp.channel("suppkey")
+    bc.semijoin_table = -1
+    bc.capacity_hint = supp.n_rows
+    bc.key_set_only = 1
+    bc.dense_array = 1
+    ocompl = Operator(OP_HASH_BUILD, bc)
+    ocompl.add_input(supp)
+    ocompl.finish()
+
+    # part attributes keyed by partkey (qualifying parts only)
+    ba = PlanHashBuild()
+    ba.key_col = part.channel("partkey")
+    ba.semijoin_table = oqual.table()
+    ba.semijoin_col = part.channel("partkey")
+    ba.n_payload = 3
+    ba.payload_col[0] = brandc
+    ba.payload_col[1] = typec
+    ba.payload_col[2] = sizec
+    ba.capacity_hint = part.n_rows
+    oattr = Operator(OP_HASH_BUILD, ba)
+    oattr.add_input(part)
+    oattr.finish()
+
+    # ps -> attributes emit join, then composite keys
+    j1 = PlanLookupJoin()
+    j1.table = oattr.table()
+    j1.key_col = ps.channel("partkey")
+    j1.mode = 0
+    j1.n_emit = 1
+    j1.emit_probe_cols[0] = ps.channel("suppkey")
+    ja = Operator(OP_LOOKUP_JOIN, j1)
+    ja.add_input(ps)
+    pa = ja.get_output_raw()  # [sk, brand, type, size]
+
+    f1p = PlanFilterProject()
+    f1p.n_proj = 3
+    f1p.proj[0] = Proj(PROJ_KEYSHL, 1, 2, 8)   # bt = brand<<8|type
+    f1p.proj[1] = Proj(PROJ_IDENT, 3, 0, 0)    # size
+    f1p.proj[2] = Proj(PROJ_IDENT, 0, 0, 0)    # sk
+    f1p.semijoin_table = ocompl.table()
+    f1p.semijoin_col = 0
+    f1p.semijoin_anti = 1
+    f1 = Operator(OP_FILTER_PROJECT, f1p)
+    f1.add_input_raw(pa)
+    pb = f1.get_output_raw()  # [bt, size, sk]
+
+    f2p = PlanFilterProject()
+    f2p.n_proj = 2
+    f2p.proj[0] = Proj(PROJ_KEYSHL, 0, 1, 8)   # g = bt<<8|size
+    f2p.proj[1] = Proj(PROJ_IDENT, 2, 0, 0)
+    f2 = Operator(OP_FILTER_PROJECT, f2p)
+    f2.add_input_raw(pb)
+    pc2 = f2.get_output_raw()  # [g, sk]
+
+    f3p = PlanFilterProject()
+    f3p.n_proj = 1
+    f3p.proj[0] = Proj(PROJ_KEYSHL, 0, 1, 32)  # gk = g<<32|sk
+    f3 = Operator(OP_FILTER_PROJECT, f3p)
+    f3.add_input_raw(pc2)
+    pd = f3.get_output_raw()  # [gk]
+
+    # distinct (group, supplier) pairs: a chained table keyed by the
+    # composite (duplicates share a slot) + a fused-agg probe — the
+    # groups ARE the distinct pairs
+    bgk = PlanHashBuild()
+    bgk.key_col = 0
+    bgk.semijoin_table = -1
+    bgk.capacity_hint = max(pd.n_rows, 1024)
+    ogk = Operator(OP_HASH_BUILD, bgk)
+    ogk.add_input_raw(pd)
+    ogk.finish()
+
+    jgk = PlanLookupJoin()
+    jgk.table = ogk.table()
+    jgk.key_col = 0
+    jgk.mode = 1
+    jgk.proj = Proj(PROJ_IDENT, 0, 0, 0)
+    jgk.dec_scale = 0
+    jgk.dec_only = 1
+    jo = Operator(OP_LOOKUP_JOIN, jgk)
+    jo.add_input_raw(pd)
+    jo.finish()
+    gk_groups = jo.get_output_raw()  # [gk, sum, f64, cnt] distinct gk rows
+
+    f4p = PlanFilterProject()
+    f4p.n_proj = 1
+    f4p.proj[0] = Proj(PROJ_SHR, 0, 0, 32)  # g
+    f4 = Operator(OP_FILTER_PROJECT, f4p)
+    f4.add_input_raw(gk_groups)
+    pe = f4.get_output_raw()  # [g] one row per distinct (g, sk)
+
+    bg2 = PlanHashBuild()
+    bg2.key_col = 0
+    bg2.semijoin_table = -1
+    bg2.capacity_hint = max(pe.n_rows, 1024)
+    og2 = Operator(OP_HASH_BUILD, bg2)
+    og2.add_input_raw(pe)
+    og2.finish()
+
+    jg2 = PlanLookupJoin()
+    jg2.table = og2.table()
+    jg2.key_col = 0
+    jg2.mode = 1
+    jg2.proj = Proj(PROJ_IDENT, 0, 0, 0)
+    jg2.dec_scale = 0
+    jg2.dec_only = 1
+    jo2 = Operator(OP_LOOKUP_JOIN, jg2)
+    jo2.add_input_raw(pe)
+    jo2.finish()
+    out = jo2.get_output(["g", "sum", "f64", "cnt"])
+
+    rows = []
+    for i in range(len(out["g"])):
+        g = int(out["g"][i])
+        rows.append((g >> 16, (g >> 8) & 0xFF, g & 0xFF,
+                     int(out["cnt"][i])))
+    rows.sort(key=lambda r: (-r[3], r[0], type_name(r[1]), r[2]))
+    for op_ in (jo2, f4, jo, f3, f2, f1, ja):
+        op_.destroy()
+    for o in (oqual, ocompl, oattr, ogk, og2):
         lib().c.pg_table_destroy(o.table())
         o.destroy()
     return rows

@@ -1354,3 +1354,46 @@ def test_q13_exact(P, oracle_lib):
     exp = oracle_lib.q13(sf, orders)
     assert got == exp
     assert len(got) > 5
+
+
+def test_q16_exact(P, oracle_lib):
+    """Q16 parts/supplier relationship — disjunctive flag-set fills,
+    CONTAINS2 complaint anti-semijoin, composite-key (KEYSHL) distinct
+    dedup vs the golden-pinned oracle, including the type-NAME order."""
+    import numpy as np
+    from presto_amd.engine import Varbin
+    sf = 0.1
+    part3 = oracle_lib.gen_part3(sf)
+    ptype = oracle_lib.gen_part_type(sf)
+    ps = oracle_lib.gen_partsupp(sf)
+    bbb = oracle_lib.gen_supplier_bbb(sf)
+    supp = oracle_lib.gen_supplier(sf)
+    pool = oracle_lib.text_pool()
+    # supplier comment text: pool substring, with the BBB splice text
+    # embedded for flagged rows (the reference's planted rows)
+    soff = np.empty(len(bbb), np.int64)
+    sln = np.empty(len(bbb), np.int32)
+    import ctypes as CT
+    oracle_lib.lib.tpch_gen_supplier_comment(
+        CT.c_double(sf), CT.c_int64(0), CT.c_int64(len(bbb)),
+        soff.ctypes.data_as(CT.c_void_p), sln.ctypes.data_as(CT.c_void_p))
+    strings = []
+    for i in range(len(bbb)):
+        t = pool[soff[i]:soff[i] + sln[i]]
+        if bbb[i] == 1:
+            t = t[:5] + b"Customer recommends against Complaints" + t[5:]
+        elif bbb[i] == 2:
+            t = t[:5] + b"Customer Recommends" + t[5:]
+        strings.append(t)
+    n_part = len(ptype)
+    got = P.pipelines.q16(
+        P.Page({"partkey": np.arange(1, n_part + 1, dtype=np.int64),
+                "brand": part3["brand"], "type_id": ptype,
+                "size": part3["size"]}),
+        P.Page({"partkey": ps["partkey"], "suppkey": ps["suppkey"]}),
+        P.Page({"suppkey": supp["suppkey"],
+                "comment": P.Varbin(strings)}),
+        oracle_lib.part_type_name)
+    exp = oracle_lib.q16(part3, ptype, ps, bbb)
+    assert got == exp
+    assert len(got) > 100
