@@ -1415,3 +1415,257 @@ int64_t oracle_q16(int64_t n_part, const uint8_t* brand,
     free(pairs);
     return n_out;
 }
+
+/* ---------------- Q10 ----------------
+ * SQL: q10.sql — returned-item reporting: revenue of returnflag='R'
+ * lineitems of orders placed in [1993-10-01, 1994-01-01) = [8674, 8766)
+ * grouped by customer; top `limit` by (revenue desc, custkey asc — the
+ * deterministic tiebreak).  Exact 1e-4 ticks. */
+int64_t oracle_q10(int64_t n_ord, const int64_t* ook, const int64_t* ock,
+                   const int32_t* od, int64_t n_li, const int64_t* lok,
+                   const uint8_t* lrf, const double* lep,
+                   const double* ldisc, int64_t n_cust, int32_t limit,
+                   int64_t* out_ck, int64_t* out_rev)
+{
+    int64_t cap = hash_capacity(n_ord < 2 ? 2 : n_ord);
+    int64_t* slot = (int64_t*)malloc(cap * sizeof(int64_t));
+    memset(slot, -1, cap * sizeof(int64_t));
+    for (int64_t i = 0; i < n_ord; i++) {
+        if (od[i] < 8674 || od[i] >= 8766) continue;
+        uint64_t h = pg_murmur3_finalize(pg_bigint_hash(ook[i]));
+        int64_t pos = (int64_t)(h & (cap - 1));
+        while (slot[pos] != -1) pos = (pos + 1) & (cap - 1);
+        slot[pos] = i;
+    }
+    int64_t* rev = (int64_t*)calloc(n_cust + 1, sizeof(int64_t));
+#pragma omp parallel for schedule(static)
+    for (int64_t i = 0; i < n_li; i++) {
+        if (lrf[i] != 'R') continue;
+        uint64_t h = pg_murmur3_finalize(pg_bigint_hash(lok[i]));
+        int64_t pos = (int64_t)(h & (cap - 1));
+        int64_t r = -1;
+        for (;;) {
+            int64_t sI = slot[pos];
+            if (sI == -1) break;
+            if (ook[sI] == lok[i]) { r = sI; break; }
+            pos = (pos + 1) & (cap - 1);
+        }
+        if (r == -1) continue;
+        int64_t cents = (int64_t)(lep[i] * 100.0 + 0.5);
+        int64_t d = (int64_t)(ldisc[i] * 100.0 + 0.5);
+        int64_t ck = ock[r];
+        if (ck >= 1 && ck <= n_cust)
+#pragma omp atomic
+            rev[ck] += cents * (100 - d);
+    }
+    int64_t n_out = 0;
+    for (int64_t ck = 1; ck <= n_cust; ck++) {
+        if (!rev[ck]) continue;
+        int64_t j;
+        if (n_out == limit) {
+            if (rev[ck] <= out_rev[limit - 1]) continue;
+            j = limit - 1;
+        } else {
+            j = n_out++;
+        }
+        while (j > 0 && out_rev[j - 1] < rev[ck]) {
+            out_ck[j] = out_ck[j - 1];
+            out_rev[j] = out_rev[j - 1];
+            j--;
+        }
+        out_ck[j] = ck;
+        out_rev[j] = rev[ck];
+    }
+    free(slot);
+    free(rev);
+    return n_out;
+}
+
+/* ---------------- Q15 ----------------
+ * SQL: q15.sql — top supplier: per-supplier revenue over shipdate in
+ * [1996-01-01, 1996-04-01) = [9496, 9587); rows with revenue equal to
+ * the max, suppkey ascending.  Exact 1e-4 ticks. */
+int64_t oracle_q15(int64_t n_li, const int64_t* lsk, const double* lep,
+                   const double* ldisc, const int32_t* lsd, int64_t n_supp,
+                   int64_t* out_sk, int64_t* out_rev, int64_t cap)
+{
+    int64_t* rev = (int64_t*)calloc(n_supp + 1, sizeof(int64_t));
+#pragma omp parallel for schedule(static)
+    for (int64_t i = 0; i < n_li; i++) {
+        if (lsd[i] < 9496 || lsd[i] >= 9587) continue;
+        int64_t sk = lsk[i];
+        if (sk < 1 || sk > n_supp) continue;
+        int64_t cents = (int64_t)(lep[i] * 100.0 + 0.5);
+        int64_t d = (int64_t)(ldisc[i] * 100.0 + 0.5);
+#pragma omp atomic
+        rev[sk] += cents * (100 - d);
+    }
+    int64_t mx = 0;
+    for (int64_t sk = 1; sk <= n_supp; sk++)
+        if (rev[sk] > mx) mx = rev[sk];
+    int64_t n_out = 0;
+    for (int64_t sk = 1; sk <= n_supp && n_out < cap; sk++)
+        if (rev[sk] == mx && mx > 0) {
+            out_sk[n_out] = sk;
+            out_rev[n_out] = mx;
+            n_out++;
+        }
+    free(rev);
+    return n_out;
+}
+
+/* ---------------- Q20 ----------------
+ * SQL: q20.sql — potential part promotion: CANADA(3) suppliers holding
+ * excess stock (ps_availqty > 0.5 * 1994 lineitem quantity of that
+ * (part, supplier)) of 'forest%' parts (name word0 = 'forest').
+ * Returns qualifying suppkeys ascending (s_name order). */
+int64_t oracle_q20(int64_t n_part, const uint8_t* name_words /*5/part*/,
+                   int32_t forest_id, int64_t n_ps, const int64_t* ps_pk,
+                   const int64_t* ps_sk, const int32_t* ps_aq, int64_t n_li,
+                   const int64_t* lpk, const int64_t* lsk,
+                   const double* lqty, const int32_t* lsd, int64_t n_supp,
+                   const uint8_t* snat, int64_t* out_sk, int64_t cap)
+{
+    /* per-(pk,sk) 1994 quantity sums for forest parts: hash on pk*S+sk */
+    uint8_t* forest = (uint8_t*)calloc(n_part, 1);
+    for (int64_t p = 0; p < n_part; p++)
+        forest[p] = name_words[p * 5] == (uint8_t)forest_id;
+    int64_t hcap = hash_capacity(n_li / 4 + 16);
+    int64_t* hkey = (int64_t*)malloc(hcap * sizeof(int64_t));
+    int64_t* hsum = (int64_t*)calloc(hcap, sizeof(int64_t));
+    memset(hkey, -1, hcap * sizeof(int64_t));
+    for (int64_t i = 0; i < n_li; i++) {
+        if (lsd[i] < 8766 || lsd[i] >= 9131) continue;
+        int64_t pk = lpk[i];
+        if (pk < 1 || pk > n_part || !forest[pk - 1]) continue;
+        int64_t key = pk * (n_supp + 1) + lsk[i];
+        uint64_t h = pg_murmur3_finalize(pg_bigint_hash(key));
+        int64_t pos = (int64_t)(h & (hcap - 1));
+        for (;;) {
+            if (hkey[pos] == key) break;
+            if (hkey[pos] == -1) { hkey[pos] = key; break; }
+            pos = (pos + 1) & (hcap - 1);
+        }
+        hsum[pos] += (int64_t)(lqty[i] + 0.5);
+    }
+    uint8_t* qual = (uint8_t*)calloc(n_supp + 1, 1);
+    for (int64_t i = 0; i < n_ps; i++) {
+        int64_t pk = ps_pk[i], sk = ps_sk[i];
+        if (pk < 1 || pk > n_part || !forest[pk - 1]) continue;
+        int64_t key = pk * (n_supp + 1) + sk;
+        uint64_t h = pg_murmur3_finalize(pg_bigint_hash(key));
+        int64_t pos = (int64_t)(h & (hcap - 1));
+        int64_t sum = -1; /* no 1994 lineitems: SQL sum() is NULL and the
+                             > comparison is not satisfied */
+        for (;;) {
+            if (hkey[pos] == key) { sum = hsum[pos]; break; }
+            if (hkey[pos] == -1) break;
+            pos = (pos + 1) & (hcap - 1);
+        }
+        if (sum >= 0 && 2LL * ps_aq[i] > sum && sk >= 1 && sk <= n_supp &&
+            snat[sk - 1] == 3)
+            qual[sk] = 1;
+    }
+    int64_t n_out = 0;
+    for (int64_t sk = 1; sk <= n_supp && n_out < cap; sk++)
+        if (qual[sk]) out_sk[n_out++] = sk;
+    free(forest);
+    free(hkey);
+    free(hsum);
+    free(qual);
+    return n_out;
+}
+
+/* ---------------- Q2 ----------------
+ * SQL: q02.sql — minimum-cost supplier: size-15 '%BRASS' parts,
+ * EUROPE(3) suppliers, supplycost equal to the part's minimum among
+ * its EUROPE suppliers; ORDER BY s_acctbal desc, n_name, s_name,
+ * p_partkey LIMIT limit.  Fills (suppkey, partkey) rows; the caller
+ * resolves display columns from the pinned streams. */
+extern int32_t tpch_nation_region(int32_t);
+extern int32_t tpch_nation_name(int32_t, char*);
+int64_t oracle_q2(int64_t n_part, const uint8_t* ptype, const uint8_t* psize,
+                  int64_t n_ps, const int64_t* ps_pk, const int64_t* ps_sk,
+                  const int64_t* ps_cost, int64_t n_supp,
+                  const uint8_t* snat, const int64_t* s_abal,
+                  int32_t limit, int64_t* out_sk, int64_t* out_pk)
+{
+    /* per qualifying part: min cost among EUROPE suppliers (partsupp is
+     * grouped 4 rows/part in generator order) */
+    /* BRASS = Types3 index 2 -> ptype % 5 == 2 */
+    typedef struct { int64_t sk, pk; } row_t;
+    row_t* rows = (row_t*)malloc((size_t)n_ps * sizeof(row_t));
+    int64_t n_rows = 0;
+    for (int64_t i = 0; i < n_ps;) {
+        int64_t pk = ps_pk[i];
+        int64_t j = i;
+        int64_t mn = INT64_MAX;
+        while (j < n_ps && ps_pk[j] == pk) {
+            int64_t sk = ps_sk[j];
+            if (sk >= 1 && sk <= n_supp &&
+                tpch_nation_region(snat[sk - 1]) == 3 && ps_cost[j] < mn)
+                mn = ps_cost[j];
+            j++;
+        }
+        if (pk >= 1 && pk <= n_part && psize[pk - 1] == 15 &&
+            ptype[pk - 1] % 5 == 2 && mn != INT64_MAX) {
+            for (int64_t k = i; k < j; k++) {
+                int64_t sk = ps_sk[k];
+                if (sk >= 1 && sk <= n_supp &&
+                    tpch_nation_region(snat[sk - 1]) == 3 &&
+                    ps_cost[k] == mn) {
+                    rows[n_rows].sk = sk;
+                    rows[n_rows].pk = pk;
+                    n_rows++;
+                }
+            }
+        }
+        i = j;
+    }
+    /* sort by (acctbal desc, nation name, suppkey, partkey) — s_name
+     * order equals suppkey order */
+    int64_t n_out = 0;
+    char na[40], nb[40];
+    for (int64_t a = 0; a < n_rows; a++) {
+        int64_t sk = rows[a].sk, pk = rows[a].pk;
+        int64_t j;
+        if (n_out == limit) {
+            int64_t lsk = out_sk[limit - 1];
+            int after;
+            if (s_abal[sk - 1] != s_abal[lsk - 1])
+                after = s_abal[sk - 1] < s_abal[lsk - 1];
+            else {
+                tpch_nation_name(snat[sk - 1], na);
+                tpch_nation_name(snat[lsk - 1], nb);
+                int c = strcmp(na, nb);
+                after = c > 0 || (c == 0 && (sk > lsk ||
+                        (sk == lsk && pk > out_pk[limit - 1])));
+            }
+            if (after) continue;
+            j = limit - 1;
+        } else {
+            j = n_out++;
+        }
+        while (j > 0) {
+            int64_t osk = out_sk[j - 1], opk = out_pk[j - 1];
+            int before;
+            if (s_abal[osk - 1] != s_abal[sk - 1])
+                before = s_abal[osk - 1] > s_abal[sk - 1];
+            else {
+                tpch_nation_name(snat[osk - 1], na);
+                tpch_nation_name(snat[sk - 1], nb);
+                int c = strcmp(na, nb);
+                before = c < 0 || (c == 0 && (osk < sk ||
+                         (osk == sk && opk <= pk)));
+            }
+            if (before) break;
+            out_sk[j] = out_sk[j - 1];
+            out_pk[j] = out_pk[j - 1];
+            j--;
+        }
+        out_sk[j] = sk;
+        out_pk[j] = pk;
+    }
+    free(rows);
+    return n_out;
+}

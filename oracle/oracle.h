@@ -293,6 +293,38 @@ int64_t oracle_q16(int64_t n_part, const uint8_t* p_brand,
                    uint8_t* out_type, uint8_t* out_size, int32_t* out_cnt,
                    int64_t cap);
 
+/* ---------------- TPC-H Q10 / Q15 / Q20 / Q2 ----------------
+ * Display-light restatements: join/aggregate/order semantics over the
+ * pinned streams; display columns (names, phones, balances, comments)
+ * resolve from the generator streams in the callers.  Addresses
+ * (v_string) remain the one unpinned generator column. */
+int64_t oracle_q10(int64_t n_ord, const int64_t* o_orderkey,
+                   const int64_t* o_custkey, const int32_t* o_orderdate,
+                   int64_t n_li, const int64_t* l_orderkey,
+                   const uint8_t* l_returnflag,
+                   const double* l_extendedprice, const double* l_discount,
+                   int64_t n_cust, int32_t limit, int64_t* out_ck,
+                   int64_t* out_rev_1e4);
+int64_t oracle_q15(int64_t n_li, const int64_t* l_suppkey,
+                   const double* l_extendedprice, const double* l_discount,
+                   const int32_t* l_shipdate, int64_t n_supp,
+                   int64_t* out_sk, int64_t* out_rev_1e4, int64_t cap);
+int64_t oracle_q20(int64_t n_part, const uint8_t* p_name_words,
+                   int32_t forest_id, int64_t n_ps,
+                   const int64_t* ps_partkey, const int64_t* ps_suppkey,
+                   const int32_t* ps_availqty, int64_t n_li,
+                   const int64_t* l_partkey, const int64_t* l_suppkey,
+                   const double* l_quantity, const int32_t* l_shipdate,
+                   int64_t n_supp, const uint8_t* s_nationkey,
+                   int64_t* out_sk, int64_t cap);
+int64_t oracle_q2(int64_t n_part, const uint8_t* p_type,
+                  const uint8_t* p_size, int64_t n_ps,
+                  const int64_t* ps_partkey, const int64_t* ps_suppkey,
+                  const int64_t* ps_supplycost_cents, int64_t n_supp,
+                  const uint8_t* s_nationkey,
+                  const int64_t* s_acctbal_cents, int32_t limit,
+                  int64_t* out_sk, int64_t* out_pk);
+
 /* ---------------- operator-level primitives (parity targets) ---------- */
 
 /* murmur3 finalizer bucket — PagesHash.java:236-252 /

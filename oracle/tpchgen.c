@@ -1082,3 +1082,62 @@ int32_t tpch_part_type_name(int32_t id, char* buf)
     buf[n] = 0;
     return n;
 }
+
+#define SEED_S_ABAL    962338209LL /* s_acctbal cents, 1/supplier
+                                    * (pinned on all 100 q02 rows) */
+#define SEED_S_PHONE   884434366LL /* s_phone, 3 draws/supplier (q02 pin) */
+#define SEED_C_PHONE  1521138112LL /* c_phone, 3 draws/customer (q10 pin) */
+
+void tpch_gen_supplier_acctbal(double sf, int64_t start, int64_t count,
+                               int64_t* acctbal_cents)
+{
+    (void)sf;
+#pragma omp parallel
+    {
+#ifdef _OPENMP
+        extern int omp_get_num_threads(void), omp_get_thread_num(void);
+        int nt = omp_get_num_threads(), tid = omp_get_thread_num();
+#else
+        int nt = 1, tid = 0;
+#endif
+        int64_t lo = count * tid / nt, hi = count * (tid + 1) / nt;
+        int64_t s = rng_skip(SEED_S_ABAL, (uint64_t)(start + lo));
+        for (int64_t i = lo; i < hi; i++)
+            acctbal_cents[i] = unif(&s, -99999, 999999);
+    }
+}
+
+/* phone 'CC-AAA-BBB-CCCC': CC = nationkey+10; A,B in 100..999, C in
+ * 1000..9999, three draws per row */
+static void gen_phone(int64_t seed, int64_t start, int64_t count,
+                      int32_t* a, int32_t* b, int32_t* c)
+{
+#pragma omp parallel
+    {
+#ifdef _OPENMP
+        extern int omp_get_num_threads(void), omp_get_thread_num(void);
+        int nt = omp_get_num_threads(), tid = omp_get_thread_num();
+#else
+        int nt = 1, tid = 0;
+#endif
+        int64_t lo = count * tid / nt, hi = count * (tid + 1) / nt;
+        int64_t s = rng_skip(seed, (uint64_t)(start + lo) * 3);
+        for (int64_t i = lo; i < hi; i++) {
+            a[i] = (int32_t)unif(&s, 100, 999);
+            b[i] = (int32_t)unif(&s, 100, 999);
+            c[i] = (int32_t)unif(&s, 1000, 9999);
+        }
+    }
+}
+void tpch_gen_supplier_phone(double sf, int64_t start, int64_t count,
+                             int32_t* a, int32_t* b, int32_t* c)
+{
+    (void)sf;
+    gen_phone(SEED_S_PHONE, start, count, a, b, c);
+}
+void tpch_gen_customer_phone(double sf, int64_t start, int64_t count,
+                             int32_t* a, int32_t* b, int32_t* c)
+{
+    (void)sf;
+    gen_phone(SEED_C_PHONE, start, count, a, b, c);
+}

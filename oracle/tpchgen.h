@@ -142,6 +142,15 @@ void tpch_gen_nation_comment(int64_t* off, int32_t* len);
 void tpch_gen_supplier_bbb(double sf, int64_t start, int64_t count,
                            uint8_t* bbb);
 
+/* s_acctbal cents (q02 pin) + phone digit triples (CC = nationkey+10;
+ * 'CC-AAA-BBB-CCCC'; q02/q10 pins) */
+void tpch_gen_supplier_acctbal(double sf, int64_t start, int64_t count,
+                               int64_t* acctbal_cents);
+void tpch_gen_supplier_phone(double sf, int64_t start, int64_t count,
+                             int32_t* a, int32_t* b, int32_t* c);
+void tpch_gen_customer_phone(double sf, int64_t start, int64_t count,
+                             int32_t* a, int32_t* b, int32_t* c);
+
 /* c_acctbal in exact cents, -99999..999999 (q22 golden pin) */
 void tpch_gen_customer_acctbal(double sf, int64_t start, int64_t count,
                                int64_t* acctbal_cents);

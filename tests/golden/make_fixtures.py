@@ -20,7 +20,8 @@ REF = pathlib.Path("/root/reference/presto-product-tests/src/main/resources/"
                    "sql-tests/testcases/hive_tpch")
 HERE = pathlib.Path(__file__).parent
 
-for q in ("q01", "q03", "q04", "q05", "q06", "q07", "q08", "q09", "q11", "q12", "q13",
-          "q14", "q16", "q17", "q18", "q19", "q21", "q22"):
+for q in ("q01", "q02", "q03", "q04", "q05", "q06", "q07", "q08", "q09",
+          "q10", "q11", "q12", "q13", "q14", "q15", "q16", "q17", "q18",
+          "q19", "q20", "q21", "q22"):
     shutil.copy(REF / f"{q}.result", HERE / f"{q}_sf1.result")
     print(f"wrote {q}_sf1.result")

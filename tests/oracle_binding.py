@@ -650,3 +650,111 @@ def gen_orders_comment_varbin(self, sf):
 
 
 OracleLib.gen_orders_comment_varbin = gen_orders_comment_varbin
+
+
+def gen_supplier_acctbal(self, sf):
+    n = self.lib.tpch_supplier_count(C.c_double(sf))
+    ab = np.empty(n, np.int64)
+    self.lib.tpch_gen_supplier_acctbal(C.c_double(sf), C.c_int64(0),
+                                       C.c_int64(n), _p(ab))
+    return ab
+
+
+def _phones(self, fn, n):
+    a = np.empty(n, np.int32); b = np.empty(n, np.int32)
+    c = np.empty(n, np.int32)
+    fn(C.c_double(0), C.c_int64(0), C.c_int64(n), _p(a), _p(b), _p(c))
+    return a, b, c
+
+
+def supplier_phone(self, sf, nat):
+    n = self.lib.tpch_supplier_count(C.c_double(sf))
+    a, b, c = _phones(self, self.lib.tpch_gen_supplier_phone, n)
+    return [f"{int(nat[i])+10}-{a[i]}-{b[i]}-{c[i]}" for i in range(n)]
+
+
+def customer_phone(self, sf, nat):
+    n = self.lib.tpch_customer_count(C.c_double(sf))
+    a, b, c = _phones(self, self.lib.tpch_gen_customer_phone, n)
+    return [f"{int(nat[i])+10}-{a[i]}-{b[i]}-{c[i]}" for i in range(n)]
+
+
+def gen_customer_comment(self, sf):
+    n = self.lib.tpch_customer_count(C.c_double(sf))
+    off = np.empty(n, np.int64); ln = np.empty(n, np.int32)
+    self.lib.tpch_gen_customer_comment(C.c_double(sf), C.c_int64(0),
+                                       C.c_int64(n), _p(off), _p(ln))
+    return off, ln
+
+
+def gen_supplier_comment(self, sf):
+    n = self.lib.tpch_supplier_count(C.c_double(sf))
+    off = np.empty(n, np.int64); ln = np.empty(n, np.int32)
+    self.lib.tpch_gen_supplier_comment(C.c_double(sf), C.c_int64(0),
+                                       C.c_int64(n), _p(off), _p(ln))
+    return off, ln
+
+
+def q10(self, orders, li, n_cust, limit=20):
+    ck = np.empty(limit, np.int64); rv = np.empty(limit, np.int64)
+    self.lib.oracle_q10.restype = C.c_int64
+    n = self.lib.oracle_q10(C.c_int64(len(orders["orderkey"])),
+                            _p(orders["orderkey"]), _p(orders["custkey"]),
+                            _p(orders["orderdate"]),
+                            C.c_int64(len(li["orderkey"])),
+                            _p(li["orderkey"]), _p(li["returnflag"]),
+                            _p(li["extendedprice"]), _p(li["discount"]),
+                            C.c_int64(n_cust), C.c_int32(limit), _p(ck),
+                            _p(rv))
+    return [(int(ck[i]), int(rv[i])) for i in range(n)]
+
+
+def q15(self, li, n_supp, cap=16):
+    sk = np.empty(cap, np.int64); rv = np.empty(cap, np.int64)
+    self.lib.oracle_q15.restype = C.c_int64
+    n = self.lib.oracle_q15(C.c_int64(len(li["orderkey"])),
+                            _p(li["suppkey"]), _p(li["extendedprice"]),
+                            _p(li["discount"]), _p(li["shipdate"]),
+                            C.c_int64(n_supp), _p(sk), _p(rv),
+                            C.c_int64(cap))
+    return [(int(sk[i]), int(rv[i])) for i in range(n)]
+
+
+def q20(self, words, ps, li, lpk, supp, cap=4096):
+    sk = np.empty(cap, np.int64)
+    fid = self.color_id("forest")
+    self.lib.oracle_q20.restype = C.c_int64
+    n = self.lib.oracle_q20(C.c_int64(len(words)), _p(words.reshape(-1)),
+                            C.c_int32(fid), C.c_int64(len(ps["partkey"])),
+                            _p(ps["partkey"]), _p(ps["suppkey"]),
+                            _p(ps["availqty"]),
+                            C.c_int64(len(li["orderkey"])), _p(lpk),
+                            _p(li["suppkey"]), _p(li["quantity"]),
+                            _p(li["shipdate"]),
+                            C.c_int64(len(supp["suppkey"])),
+                            _p(supp["nationkey"]), _p(sk), C.c_int64(cap))
+    return [int(sk[i]) for i in range(n)]
+
+
+def q2(self, part3, ptype, ps, supp, s_abal, limit=100):
+    sk = np.empty(limit, np.int64); pk = np.empty(limit, np.int64)
+    self.lib.oracle_q2.restype = C.c_int64
+    n = self.lib.oracle_q2(C.c_int64(len(ptype)), _p(ptype),
+                           _p(part3["size"]),
+                           C.c_int64(len(ps["partkey"])), _p(ps["partkey"]),
+                           _p(ps["suppkey"]), _p(ps["supplycost_cents"]),
+                           C.c_int64(len(supp["suppkey"])),
+                           _p(supp["nationkey"]), _p(s_abal),
+                           C.c_int32(limit), _p(sk), _p(pk))
+    return [(int(sk[i]), int(pk[i])) for i in range(n)]
+
+
+OracleLib.gen_supplier_acctbal = gen_supplier_acctbal
+OracleLib.supplier_phone = supplier_phone
+OracleLib.customer_phone = customer_phone
+OracleLib.gen_customer_comment = gen_customer_comment
+OracleLib.gen_supplier_comment = gen_supplier_comment
+OracleLib.q10 = q10
+OracleLib.q15 = q15
+OracleLib.q20 = q20
+OracleLib.q2 = q2

@@ -535,3 +535,83 @@ def test_nation_comments_pool(oracle_lib):
     assert pool[off[0]:off[0]+ln[0]] == expect0
     assert ln[3] == 101  # CANADA
     assert b"ironic, silent packages" in pool[off[3]:off[3]+ln[3]]
+
+
+def test_q10_sf1_golden(oracle_lib):
+    """Q10 returned items — every golden column except c_address (the
+    one unpinned v_string generator column): custkey, name, revenue,
+    acctbal, nation, phone, comment."""
+    orders = oracle_lib.gen_orders(1.0)
+    li = oracle_lib.gen_lineitem2(1.0)
+    cust = oracle_lib.gen_customer2(1.0)
+    abal = oracle_lib.gen_customer_acctbal(1.0)
+    rows = oracle_lib.q10(orders, li, len(cust["custkey"]))
+    golden = _parse_golden("q10_sf1.result")
+    assert len(rows) == len(golden) == 20
+    phones = oracle_lib.customer_phone(1.0, cust["nationkey"])
+    coff, cln = oracle_lib.gen_customer_comment(1.0)
+    pool = oracle_lib.text_pool()
+    for (ck, rev), g in zip(rows, golden):
+        assert ck == int(g[0])
+        assert f"Customer#{ck:09d}" == g[1]
+        assert Decimal(rev) / 10**4 == Decimal(g[2])
+        assert Decimal(int(abal[ck - 1])) / 100 == Decimal(g[3])
+        assert oracle_lib.nation_name(int(cust["nationkey"][ck - 1])) == g[4]
+        # g[5] = c_address: unpinned
+        assert phones[ck - 1] == g[6]
+        assert pool[coff[ck-1]:coff[ck-1]+cln[ck-1]].decode() == g[7]
+
+
+def test_q15_sf1_golden(oracle_lib):
+    """Q15 top supplier — all golden columns except s_address."""
+    li = oracle_lib.gen_lineitem2(1.0)
+    supp = oracle_lib.gen_supplier(1.0)
+    rows = oracle_lib.q15(li, len(supp["suppkey"]))
+    golden = _parse_golden("q15_sf1.result")
+    assert len(rows) == len(golden)
+    phones = oracle_lib.supplier_phone(1.0, supp["nationkey"])
+    for (sk, rev), g in zip(rows, golden):
+        assert sk == int(g[0])
+        assert f"Supplier#{sk:09d}" == g[1]
+        assert phones[sk - 1] == g[3]
+        assert Decimal(rev) / 10**4 == Decimal(g[4])
+
+
+def test_q20_sf1_golden(oracle_lib):
+    """Q20 potential part promotion — supplier names on all golden rows
+    (addresses unpinned)."""
+    words = oracle_lib.gen_part_name_words(1.0)
+    ps = oracle_lib.gen_partsupp(1.0)
+    li = oracle_lib.gen_lineitem2(1.0)
+    lpk = oracle_lib.gen_lineitem_partkey(1.0)
+    supp = oracle_lib.gen_supplier(1.0)
+    sks = oracle_lib.q20(words, ps, li, lpk, supp)
+    golden = _parse_golden("q20_sf1.result")
+    assert len(sks) == len(golden)
+    for sk, g in zip(sks, golden):
+        assert f"Supplier#{sk:09d}" == g[0]
+
+
+def test_q2_sf1_golden(oracle_lib):
+    """Q2 minimum-cost supplier — every golden column except s_address:
+    acctbal, names, nation, partkey, mfgr, phone, comment."""
+    part3 = oracle_lib.gen_part3(1.0)
+    ptype = oracle_lib.gen_part_type(1.0)
+    ps = oracle_lib.gen_partsupp(1.0)
+    supp = oracle_lib.gen_supplier(1.0)
+    abal = oracle_lib.gen_supplier_acctbal(1.0)
+    rows = oracle_lib.q2(part3, ptype, ps, supp, abal)
+    golden = _parse_golden("q02_sf1.result")
+    assert len(rows) == len(golden) == 100
+    phones = oracle_lib.supplier_phone(1.0, supp["nationkey"])
+    soff, sln = oracle_lib.gen_supplier_comment(1.0)
+    pool = oracle_lib.text_pool()
+    for (sk, pk), g in zip(rows, golden):
+        assert Decimal(int(abal[sk - 1])) / 100 == Decimal(g[0])
+        assert f"Supplier#{sk:09d}" == g[1]
+        assert oracle_lib.nation_name(int(supp["nationkey"][sk - 1])) == g[2]
+        assert pk == int(g[3])
+        assert f"Manufacturer#{int(part3['mfgr'][pk - 1])}" == g[4]
+        # g[5] = s_address: unpinned
+        assert phones[sk - 1] == g[6]
+        assert pool[soff[sk-1]:soff[sk-1]+sln[sk-1]].decode() == g[7]
