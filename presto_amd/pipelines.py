@@ -1784,6 +1784,271 @@ def q16(part: Page, ps: Page, supp: Page, type_name):
     return rows
 
 
+def q10(cust_n: int, orders: Page, li: Page, limit=20):
+    """Q10 returned items (q10.sql): orders date filter fused into a
+    chained build (payload custkey); returned lineitems emit-join to
+    their customer; per-customer revenue by fused-agg probe; bounded
+    top-`limit` (revenue desc, custkey asc) host-merged from the group
+    page.  Returns [(custkey, revenue_1e4)]."""
+    import numpy as np
+    from .engine import lib
+
+    bo = PlanHashBuild()
+    bo.n_preds = 2
+    bo.preds[0] = Pred(orders.channel("orderdate"), CMP_GE, 8674, 0.0)
+    bo.preds[1] = Pred(orders.channel("orderdate"), CMP_LT, 8766, 0.0)
+    bo.key_col = orders.channel("orderkey")
+    bo.semijoin_table = -1
+    bo.n_payload = 1
+    bo.payload_col[0] = orders.channel("custkey")
+    bo.capacity_hint = max(orders.n_rows // 8, 16)
+    oo = Operator(OP_HASH_BUILD, bo)
+    oo.add_input(orders)
+    oo.finish()
+
+    jp = PlanLookupJoin()
+    jp.table = oo.table()
+    jp.n_preds = 1
+    jp.preds[0] = Pred(li.channel("returnflag"), CMP_EQ, ord("R"), 0.0)
+    jp.key_col = li.channel("orderkey")
+    jp.mode = 0
+    jp.n_emit = 2
+    jp.emit_probe_cols[0] = li.channel("extendedprice")
+    jp.emit_probe_cols[1] = li.channel("discount")
+    jo = Operator(OP_LOOKUP_JOIN, jp)
+    jo.add_input(li)
+    pa = jo.get_output_raw()  # [ep, dc, custkey]
+
+    bc = PlanHashBuild()
+    bc.key_col = 0
+    bc.semijoin_table = -1
+    bc.capacity_hint = cust_n + 64
+    bc.agg_table = 1
+    keys = Page({"custkey": np.arange(1, cust_n + 1, dtype=np.int64)})
+    oc = Operator(OP_HASH_BUILD, bc)
+    oc.add_input(keys)
+    oc.finish()
+
+    ja = PlanLookupJoin()
+    ja.table = oc.table()
+    ja.key_col = 2
+    ja.mode = 1
+    ja.proj = Proj(PROJ_DISC_PRICE, 0, 1, 0)
+    ja.dec_scale = 4
+    ja.dec_only = 1
+    j2 = Operator(OP_LOOKUP_JOIN, ja)
+    j2.add_input_raw(pa)
+    j2.finish()
+    g = j2.get_output(["custkey", "rev", "f64", "cnt"])
+    rows = sorted(((int(g["custkey"][i]), int(g["rev"][i]))
+                   for i in range(len(g["custkey"])) if g["rev"][i] > 0),
+                  key=lambda r: (-r[1], r[0]))[:limit]
+    j2.destroy()
+    jo.destroy()
+    for o in (oo, oc):
+        lib().c.pg_table_destroy(o.table())
+        o.destroy()
+    return rows
+
+
+def q15(supp: Page, li: Page):
+    """Q15 top supplier (q15.sql): per-supplier revenue over the 1996-Q1
+    window by fused-agg probe; the scalar max subquery resolves on the
+    group page (output stage).  Returns [(suppkey, revenue_1e4)] of the
+    max-revenue supplier(s), suppkey ascending."""
+    from .engine import lib
+    bs = PlanHashBuild()
+    bs.key_col = supp.channel("suppkey")
+    bs.semijoin_table = -1
+    bs.capacity_hint = supp.n_rows + 64
+    bs.agg_table = 1
+    os_ = Operator(OP_HASH_BUILD, bs)
+    os_.add_input(supp)
+    os_.finish()
+
+    jp = PlanLookupJoin()
+    jp.table = os_.table()
+    jp.n_preds = 2
+    jp.preds[0] = Pred(li.channel("shipdate"), CMP_GE, 9496, 0.0)
+    jp.preds[1] = Pred(li.channel("shipdate"), CMP_LT, 9587, 0.0)
+    jp.key_col = li.channel("suppkey")
+    jp.mode = 1
+    jp.proj = Proj(PROJ_DISC_PRICE, li.channel("extendedprice"),
+                   li.channel("discount"), 0)
+    jp.dec_scale = 4
+    jp.dec_only = 1
+    jo = Operator(OP_LOOKUP_JOIN, jp)
+    jo.add_input(li)
+    jo.finish()
+    g = jo.get_output(["suppkey", "rev", "f64", "cnt"])
+    mx = max((int(v) for v in g["rev"]), default=0)
+    rows = sorted((int(g["suppkey"][i]), int(g["rev"][i]))
+                  for i in range(len(g["suppkey"]))
+                  if int(g["rev"][i]) == mx and mx > 0)
+    jo.destroy()
+    lib().c.pg_table_destroy(os_.table())
+    os_.destroy()
+    return rows
+
+
+def q20(part: Page, ps: Page, supp: Page, li: Page):
+    """Q20 potential part promotion (q20.sql): 'forest%' as a VARBIN
+    PREFIX flag set; per-(part,supplier) 1994 quantities via a
+    composite-key (KEYSHL) fused-agg probe; the correlated
+    availqty > 0.5*sum compare runs as 2*availqty > sum using a KEYSHL
+    doubling against the group page's zero column (missing sums drop at
+    the inner join — SQL's NULL comparison); CANADA(3) semijoin;
+    distinct suppliers via a final fused-agg probe.  Returns qualifying
+    suppkeys ascending."""
+    from .engine import lib
+
+    bf = PlanHashBuild()
+    bf.n_preds = 1
+    pf = Pred(part.channel("name"), CMP_PREFIX, 0, 0.0)
+    pf.sval = b"forest"
+    pf.slen = 6
+    bf.preds[0] = pf
+    bf.key_col = part.channel("partkey")
+    bf.semijoin_table = -1
+    bf.capacity_hint = part.n_rows
+    bf.key_set_only = 1
+    bf.dense_array = 1
+    of = Operator(OP_HASH_BUILD, bf)
+    of.add_input(part)
+    of.finish()
+
+    # forest 1994 lineitems -> composite (pk<<32|sk, qty)
+    fl = PlanFilterProject()
+    fl.n_preds = 2
+    fl.preds[0] = Pred(li.channel("shipdate"), CMP_GE, 8766, 0.0)
+    fl.preds[1] = Pred(li.channel("shipdate"), CMP_LT, 9131, 0.0)
+    fl.n_proj = 2
+    fl.proj[0] = Proj(PROJ_KEYSHL, li.channel("partkey"),
+                      li.channel("suppkey"), 32)
+    fl.proj[1] = Proj(PROJ_IDENT, li.channel("quantity"), 0, 0)
+    fl.semijoin_table = of.table()
+    fl.semijoin_col = li.channel("partkey")
+    f1 = Operator(OP_FILTER_PROJECT, fl)
+    f1.add_input(li)
+    lq = f1.get_output_raw()  # [key, qty]
+
+    bk = PlanHashBuild()
+    bk.key_col = 0
+    bk.semijoin_table = -1
+    bk.capacity_hint = max(lq.n_rows, 1024)
+    ok_ = Operator(OP_HASH_BUILD, bk)
+    ok_.add_input_raw(lq)
+    ok_.finish()
+
+    jq = PlanLookupJoin()
+    jq.table = ok_.table()
+    jq.key_col = 0
+    jq.mode = 1
+    jq.proj = Proj(PROJ_IDENT, 1, 0, 0)
+    jq.dec_scale = 0
+    jq.dec_only = 1
+    jo = Operator(OP_LOOKUP_JOIN, jq)
+    jo.add_input_raw(lq)
+    jo.finish()
+    sums = jo.get_output_raw()  # [key, sum_qty, f64(zeros), cnt]
+
+    # forest partsupp rows -> [key, availqty, suppkey]
+    fp = PlanFilterProject()
+    fp.n_proj = 3
+    fp.proj[0] = Proj(PROJ_KEYSHL, ps.channel("partkey"),
+                      ps.channel("suppkey"), 32)
+    fp.proj[1] = Proj(PROJ_IDENT, ps.channel("availqty"), 0, 0)
+    fp.proj[2] = Proj(PROJ_IDENT, ps.channel("suppkey"), 0, 0)
+    fp.semijoin_table = of.table()
+    fp.semijoin_col = ps.channel("partkey")
+    f2 = Operator(OP_FILTER_PROJECT, fp)
+    f2.add_input(ps)
+    psq = f2.get_output_raw()
+
+    bs = PlanHashBuild()
+    bs.key_col = 0
+    bs.semijoin_table = -1
+    bs.n_payload = 2
+    bs.payload_col[0] = 1  # sum_qty
+    bs.payload_col[1] = 2  # the zero column (for the KEYSHL doubling)
+    bs.capacity_hint = max(sums.n_rows, 1024)
+    osum = Operator(OP_HASH_BUILD, bs)
+    osum.add_input_raw(sums)
+    osum.finish()
+
+    je = PlanLookupJoin()
+    je.table = osum.table()
+    je.key_col = 0
+    je.mode = 0
+    je.n_emit = 2
+    je.emit_probe_cols[0] = 1  # availqty
+    je.emit_probe_cols[1] = 2  # suppkey
+    j3 = Operator(OP_LOOKUP_JOIN, je)
+    j3.add_input_raw(psq)
+    pj = j3.get_output_raw()  # [aq, sk, sum, zeros]
+
+    bn = PlanHashBuild()
+    bn.n_preds = 1
+    bn.preds[0] = Pred(supp.channel("nationkey"), CMP_EQ, 3, 0.0)
+    bn.key_col = supp.channel("suppkey")
+    bn.semijoin_table = -1
+    bn.capacity_hint = supp.n_rows
+    bn.key_set_only = 1
+    bn.dense_array = 1
+    ocan = Operator(OP_HASH_BUILD, bn)
+    ocan.add_input(supp)
+    ocan.finish()
+
+    f3p = PlanFilterProject()
+    f3p.n_proj = 3
+    f3p.proj[0] = Proj(PROJ_KEYSHL, 0, 3, 1)  # 2*availqty (zeros low bit)
+    f3p.proj[1] = Proj(PROJ_IDENT, 1, 0, 0)   # suppkey
+    f3p.proj[2] = Proj(PROJ_IDENT, 2, 0, 0)   # sum
+    f3p.semijoin_table = ocan.table()
+    f3p.semijoin_col = 1
+    f3 = Operator(OP_FILTER_PROJECT, f3p)
+    f3.add_input_raw(pj)
+    pk2 = f3.get_output_raw()  # [2aq, sk, sum]
+
+    f4p = PlanFilterProject()
+    f4p.n_preds = 1
+    p4 = Pred(0, CMP_GT, 0, 0.0)
+    p4.rhs_col = 2 + 1
+    f4p.preds[0] = p4
+    f4p.n_proj = 1
+    f4p.proj[0] = Proj(PROJ_IDENT, 1, 0, 0)
+    f4 = Operator(OP_FILTER_PROJECT, f4p)
+    f4.add_input_raw(pk2)
+    excess = f4.get_output_raw()  # [sk] with dupes
+
+    bq = PlanHashBuild()
+    bq.key_col = supp.channel("suppkey")
+    bq.semijoin_table = -1
+    bq.capacity_hint = supp.n_rows + 64
+    bq.agg_table = 1
+    oq = Operator(OP_HASH_BUILD, bq)
+    oq.add_input(supp)
+    oq.finish()
+    jd = PlanLookupJoin()
+    jd.table = oq.table()
+    jd.key_col = 0
+    jd.mode = 1
+    jd.proj = Proj(PROJ_IDENT, 0, 0, 0)
+    jd.dec_scale = 0
+    jd.dec_only = 1
+    j5 = Operator(OP_LOOKUP_JOIN, jd)
+    j5.add_input_raw(excess)
+    j5.finish()
+    out = j5.get_output(["suppkey", "s", "f", "c"])
+    sks = sorted(int(v) for v in out["suppkey"])
+    for o in (j5, f4, f3, j3, f2, jo, f1):
+        o.destroy()
+    for o in (of, ok_, osum, ocan, oq):
+        lib().c.pg_table_destroy(o.table())
+        o.destroy()
+    return sks
+
+
 Q22_CODE_NATIONS = (3, 7, 8, 13, 19, 20, 21)  # codes '13'..'31' ascending
 
 

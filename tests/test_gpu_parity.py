@@ -1397,3 +1397,64 @@ def test_q16_exact(P, oracle_lib):
     exp = oracle_lib.q16(part3, ptype, ps, bbb)
     assert got == exp
     assert len(got) > 100
+
+
+def test_q10_exact(P, oracle_lib):
+    """Q10 returned items — date-fused build + emit join + per-customer
+    fused-agg vs the golden-pinned oracle."""
+    sf = 0.1
+    orders = oracle_lib.gen_orders(sf)
+    li = oracle_lib.gen_lineitem2(sf)
+    n_cust = int(150000 * sf)
+    got = P.pipelines.q10(
+        n_cust,
+        P.Page({k: orders[k] for k in ("orderkey", "custkey", "orderdate")}),
+        P.Page({"orderkey": li["orderkey"], "returnflag": li["returnflag"],
+                "extendedprice": li["extendedprice"],
+                "discount": li["discount"]}))
+    exp = oracle_lib.q10(orders, li, n_cust)
+    assert got == exp
+    assert len(got) == 20
+
+
+def test_q15_exact(P, oracle_lib):
+    """Q15 top supplier — fused-agg revenue + scalar-max output stage vs
+    the golden-pinned oracle."""
+    sf = 0.1
+    li = oracle_lib.gen_lineitem2(sf)
+    supp = oracle_lib.gen_supplier(sf)
+    got = P.pipelines.q15(
+        P.Page({"suppkey": supp["suppkey"], "nationkey": supp["nationkey"]}),
+        P.Page({"suppkey": li["suppkey"], "shipdate": li["shipdate"],
+                "extendedprice": li["extendedprice"],
+                "discount": li["discount"]}))
+    exp = oracle_lib.q15(li, len(supp["suppkey"]))
+    assert got == exp
+    assert len(got) >= 1
+
+
+def test_q20_exact(P, oracle_lib):
+    """Q20 potential part promotion — PREFIX flag set, composite-key
+    quantity sums, NULL-dropping inner join, KEYSHL doubling compare,
+    CANADA semijoin vs the golden-pinned oracle."""
+    import numpy as np
+    from presto_amd.engine import Varbin
+    sf = 0.1
+    words = oracle_lib.gen_part_name_words(sf)
+    names = [oracle_lib.color_name(i) for i in range(92)]
+    strings = [" ".join(names[w] for w in row).encode() for row in words]
+    ps = oracle_lib.gen_partsupp(sf)
+    li = oracle_lib.gen_lineitem2(sf)
+    lpk = oracle_lib.gen_lineitem_partkey(sf)
+    supp = oracle_lib.gen_supplier(sf)
+    got = P.pipelines.q20(
+        P.Page({"partkey": np.arange(1, len(strings) + 1, dtype=np.int64),
+                "name": P.Varbin(strings)}),
+        P.Page({"partkey": ps["partkey"], "suppkey": ps["suppkey"],
+                "availqty": ps["availqty"]}),
+        P.Page({"suppkey": supp["suppkey"], "nationkey": supp["nationkey"]}),
+        P.Page({"partkey": lpk, "suppkey": li["suppkey"],
+                "quantity": li["quantity"], "shipdate": li["shipdate"]}))
+    exp = oracle_lib.q20(words, ps, li, lpk, supp)
+    assert got == exp
+    assert len(got) > 0
