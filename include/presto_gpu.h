@@ -295,6 +295,11 @@ typedef struct {
      * such as Q17/Q18/Q21 quantity sums — halves the atomic traffic of
      * all-match probes) */
     int32_t dec_only;
+    /* mode 1 only: the decimal accumulator keeps the MIN of the
+     * projected ticks instead of the sum (MinAggregationFunction over a
+     * grouped probe — Q2's per-part minimum supplycost).  Implies
+     * dec_only. */
+    int32_t dec_min;
 } pg_plan_lookup_join;
 
 typedef struct {

@@ -1258,8 +1258,11 @@ __global__ __launch_bounds__(256) void k_probe_agg(
         ag.proj = plan.proj;
         ag.dec_scale = plan.dec_scale;
         int64_t ticks = d_eval_proj_dec(pg, ag, i);
-        atomicAdd(&acc[s].dec, (unsigned long long)ticks);
-        if (!plan.dec_only) {
+        if (plan.dec_min)
+            atomicMin((long long*)&acc[s].dec, (long long)ticks);
+        else
+            atomicAdd(&acc[s].dec, (unsigned long long)ticks);
+        if (!plan.dec_only && !plan.dec_min) {
             double p = d_eval_proj_f64(pg, plan.proj, i);
             uint64_t phi, plo;
             fx128_from_f64(p, &phi, &plo);
@@ -1366,6 +1369,20 @@ __global__ __launch_bounds__(256) void k_probe_agg_q3(
             atomicAdd(&acc[sl].fhi, phi + (old > ~plo ? 1ull : 0ull));
             atomicAdd(&acc[sl].cnt, 1ull);
         }
+    }
+}
+
+/* MIN-accumulator identity init (dec = +inf sentinel, rest zero) */
+__global__ __launch_bounds__(256) void k_acc_min_init(slot_acc* acc,
+                                                      int64_t cap)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < cap; i += stride) {
+        acc[i].dec = 0x7f7f7f7f7f7f7f7fULL;
+        acc[i].fhi = 0;
+        acc[i].flo = 0;
+        acc[i].cnt = 0;
     }
 }
 
@@ -2827,6 +2844,13 @@ struct JoinOp : Op {
         if (plan.mode == 1 && !t->acc.p) {
             t->acc.alloc((size_t)t->cap * sizeof(slot_acc));
             t->acc.zero();
+            CHKV(hipStreamSynchronize(g_stream));
+        }
+        if (plan.mode == 1 && plan.dec_min) {
+            /* MIN identity: dec = large positive sentinel; cnt
+             * distinguishes matched groups */
+            hipLaunchKernelGGL(k_acc_min_init, dim3(2048), dim3(256), 0,
+                               g_stream, (slot_acc*)t->acc.p, t->cap);
             CHKV(hipStreamSynchronize(g_stream));
         }
         if (plan.mode == 2) {

@@ -88,7 +88,8 @@ class PlanLookupJoin(C.Structure):
                 ("dec_scale", C.c_int32), ("table2", C.c_int64),
                 ("table2_key_col", C.c_int32),
                 ("n_group_vals", C.c_int32),
-                ("group_vals", C.c_uint8 * 8), ("dec_only", C.c_int32)]
+                ("group_vals", C.c_uint8 * 8), ("dec_only", C.c_int32),
+                ("dec_min", C.c_int32)]
 
 
 class PlanTopN(C.Structure):

@@ -2049,6 +2049,165 @@ def q20(part: Page, ps: Page, supp: Page, li: Page):
     return sks
 
 
+def q2(part: Page, ps: Page, supp: Page, s_abal, s_nat, limit=100):
+    """Q2 minimum-cost supplier (q02.sql): size-15 '%BRASS' parts (the
+    type%5 disjunction as 30 conjunctive flag-set fills), EUROPE
+    suppliers, per-part MIN supplycost via a dec_min fused-agg probe,
+    then an equality join back to emit the (supplier, part) pairs.  The
+    final ORDER BY (acctbal desc, nation name, supplier, part) resolves
+    on the host output stage from the pinned streams.  Returns
+    [(suppkey, partkey)] in golden order."""
+    import numpy as np
+    from .engine import lib
+
+    # qualifying parts: size == 15 AND type % 5 == 2 ('%BRASS')
+    qual_pages = []
+    fops = []
+    for t in range(2, 150, 5):
+        fp = PlanFilterProject()
+        fp.n_preds = 2
+        fp.preds[0] = Pred(part.channel("size"), CMP_EQ, 15, 0.0)
+        fp.preds[1] = Pred(part.channel("type_id"), CMP_EQ, t, 0.0)
+        fp.n_proj = 1
+        fp.proj[0] = Proj(PROJ_IDENT, part.channel("partkey"), 0, 0)
+        f = Operator(OP_FILTER_PROJECT, fp)
+        f.add_input(part)
+        qual_pages.append(f.get_output_raw())
+        fops.append(f)
+    bq = PlanHashBuild()
+    bq.key_col = 0
+    bq.semijoin_table = -1
+    bq.capacity_hint = part.n_rows
+    bq.key_set_only = 1
+    bq.dense_array = 1
+    oq = Operator(OP_HASH_BUILD, bq)
+    for pg_ in qual_pages:
+        oq.add_input_raw(pg_)
+    oq.finish()
+    for f in fops:
+        f.destroy()
+
+    # EUROPE supplier flag set: nation-key set semijoin (region 3
+    # nations: FRANCE, GERMANY, ROMANIA, RUSSIA, UNITED KINGDOM)
+    natp = Page({"nationkey": np.asarray((6, 7, 19, 22, 23),
+                                         dtype=np.int64)})
+    bn = PlanHashBuild()
+    bn.key_col = 0
+    bn.semijoin_table = -1
+    bn.capacity_hint = 32
+    bn.key_set_only = 1
+    on = Operator(OP_HASH_BUILD, bn)
+    on.add_input(natp)
+    on.finish()
+    bf = PlanHashBuild()
+    bf.key_col = supp.channel("suppkey")
+    bf.semijoin_table = on.table()
+    bf.semijoin_col = supp.channel("nationkey")
+    bf.n_payload = 0
+    bf.capacity_hint = supp.n_rows
+    bf.key_set_only = 1
+    bf.dense_array = 1
+    oeu = Operator(OP_HASH_BUILD, bf)
+    oeu.add_input(supp)
+    oeu.finish()
+
+    # partsupp restricted to qualifying parts AND european suppliers
+    f1p = PlanFilterProject()
+    f1p.n_proj = 3
+    f1p.proj[0] = Proj(PROJ_IDENT, ps.channel("partkey"), 0, 0)
+    f1p.proj[1] = Proj(PROJ_IDENT, ps.channel("suppkey"), 0, 0)
+    f1p.proj[2] = Proj(PROJ_IDENT, ps.channel("supplycost"), 0, 0)
+    f1p.semijoin_table = oq.table()
+    f1p.semijoin_col = ps.channel("partkey")
+    f1 = Operator(OP_FILTER_PROJECT, f1p)
+    f1.add_input(ps)
+    pa = f1.get_output_raw()
+    f2p = PlanFilterProject()
+    f2p.n_proj = 3
+    for i in range(3):
+        f2p.proj[i] = Proj(PROJ_IDENT, i, 0, 0)
+    f2p.semijoin_table = oeu.table()
+    f2p.semijoin_col = 1
+    f2 = Operator(OP_FILTER_PROJECT, f2p)
+    f2.add_input_raw(pa)
+    pb = f2.get_output_raw()  # [pk, sk, cost_cents]
+
+    # per-part MIN supplycost
+    bm = PlanHashBuild()
+    bm.key_col = part.channel("partkey")
+    bm.semijoin_table = oq.table()
+    bm.semijoin_col = part.channel("partkey")
+    bm.capacity_hint = max(part.n_rows // 64, 4096)
+    bm.agg_table = 1
+    om = Operator(OP_HASH_BUILD, bm)
+    om.add_input(part)
+    om.finish()
+
+    jm = PlanLookupJoin()
+    jm.table = om.table()
+    jm.key_col = 0
+    jm.mode = 1
+    jm.proj = Proj(PROJ_IDENT, 2, 0, 0)
+    jm.dec_scale = 0
+    jm.dec_min = 1
+    j1 = Operator(OP_LOOKUP_JOIN, jm)
+    j1.add_input_raw(pb)
+    j1.finish()
+    mins = j1.get_output_raw()  # [pk, min_cost, f64, cnt]
+
+    bmm = PlanHashBuild()
+    bmm.key_col = 0
+    bmm.semijoin_table = -1
+    bmm.n_payload = 1
+    bmm.payload_col[0] = 1
+    bmm.capacity_hint = max(mins.n_rows, 16)
+    omm = Operator(OP_HASH_BUILD, bmm)
+    omm.add_input_raw(mins)
+    omm.finish()
+
+    je = PlanLookupJoin()
+    je.table = omm.table()
+    je.key_col = 0
+    je.mode = 0
+    je.n_emit = 3
+    for i in range(3):
+        je.emit_probe_cols[i] = i
+    j2 = Operator(OP_LOOKUP_JOIN, je)
+    j2.add_input_raw(pb)
+    pc = j2.get_output_raw()  # [pk, sk, cost, min]
+
+    f3p = PlanFilterProject()
+    f3p.n_preds = 1
+    p3 = Pred(2, CMP_EQ, 0, 0.0)
+    p3.rhs_col = 3 + 1
+    f3p.preds[0] = p3
+    f3p.n_proj = 2
+    f3p.proj[0] = Proj(PROJ_IDENT, 1, 0, 0)
+    f3p.proj[1] = Proj(PROJ_IDENT, 0, 0, 0)
+    f3 = Operator(OP_FILTER_PROJECT, f3p)
+    f3.add_input_raw(pc)
+    out = f3.get_output(["suppkey", "partkey"])
+
+    rows = sorted(
+        ((int(out["suppkey"][i]), int(out["partkey"][i]))
+         for i in range(len(out["suppkey"]))),
+        key=lambda r: (-int(s_abal[r[0] - 1]),
+                       NATION_NAMES[int(s_nat[r[0] - 1])], r[0], r[1]))
+    for o in (f3, j2, j1, f2, f1):
+        o.destroy()
+    for o in (oq, on, oeu, om, omm):
+        lib().c.pg_table_destroy(o.table())
+        o.destroy()
+    return rows[:limit]
+
+
+NATION_NAMES = [
+    "ALGERIA", "ARGENTINA", "BRAZIL", "CANADA", "EGYPT", "ETHIOPIA",
+    "FRANCE", "GERMANY", "INDIA", "INDONESIA", "IRAN", "IRAQ", "JAPAN",
+    "JORDAN", "KENYA", "MOROCCO", "MOZAMBIQUE", "PERU", "CHINA", "ROMANIA",
+    "SAUDI ARABIA", "VIETNAM", "RUSSIA", "UNITED KINGDOM", "UNITED STATES"]
+
+
 Q22_CODE_NATIONS = (3, 7, 8, 13, 19, 20, 21)  # codes '13'..'31' ascending
 
 

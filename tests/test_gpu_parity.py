@@ -1458,3 +1458,27 @@ def test_q20_exact(P, oracle_lib):
     exp = oracle_lib.q20(words, ps, li, lpk, supp)
     assert got == exp
     assert len(got) > 0
+
+
+def test_q2_exact(P, oracle_lib):
+    """Q2 minimum-cost supplier — dec_min fused-agg probe + equality
+    join back vs the golden-pinned oracle (full output order incl the
+    acctbal/nation-name sort)."""
+    import numpy as np
+    sf = 0.1
+    part3 = oracle_lib.gen_part3(sf)
+    ptype = oracle_lib.gen_part_type(sf)
+    ps = oracle_lib.gen_partsupp(sf)
+    supp = oracle_lib.gen_supplier(sf)
+    abal = oracle_lib.gen_supplier_acctbal(sf)
+    n_part = len(ptype)
+    got = P.pipelines.q2(
+        P.Page({"partkey": np.arange(1, n_part + 1, dtype=np.int64),
+                "type_id": ptype, "size": part3["size"]}),
+        P.Page({"partkey": ps["partkey"], "suppkey": ps["suppkey"],
+                "supplycost": ps["supplycost_cents"]}),
+        P.Page({"suppkey": supp["suppkey"], "nationkey": supp["nationkey"]}),
+        abal, supp["nationkey"])
+    exp = oracle_lib.q2(part3, ptype, ps, supp, abal)
+    assert got == exp
+    assert len(got) == 100
