@@ -2660,6 +2660,10 @@ struct BuildOp : Op {
                                  ? PG_T_U8
                                  : sp.pg.cols[plan.payload_col[i]].tag;
         if (plan.dense_array) {
+            if (plan.semijoin_table >= 0)
+                throw std::runtime_error(
+                    "dense_array builds do not evaluate semijoins — "
+                    "pre-filter with a FILTER_PROJECT semijoin instead");
             if (sp.pg.cols[plan.key_col].tag != PG_T_I64 ||
                 (!plan.key_set_only &&
                  sp.pg.cols[plan.payload_col[0]].tag != PG_T_U8))

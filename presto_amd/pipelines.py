@@ -2099,17 +2099,24 @@ def q2(part: Page, ps: Page, supp: Page, s_abal, s_nat, limit=100):
     on = Operator(OP_HASH_BUILD, bn)
     on.add_input(natp)
     on.finish()
+    feu = PlanFilterProject()
+    feu.n_proj = 1
+    feu.proj[0] = Proj(PROJ_IDENT, supp.channel("suppkey"), 0, 0)
+    feu.semijoin_table = on.table()
+    feu.semijoin_col = supp.channel("nationkey")
+    fe = Operator(OP_FILTER_PROJECT, feu)
+    fe.add_input(supp)
+    eup = fe.get_output_raw()
     bf = PlanHashBuild()
-    bf.key_col = supp.channel("suppkey")
-    bf.semijoin_table = on.table()
-    bf.semijoin_col = supp.channel("nationkey")
-    bf.n_payload = 0
+    bf.key_col = 0
+    bf.semijoin_table = -1
     bf.capacity_hint = supp.n_rows
     bf.key_set_only = 1
     bf.dense_array = 1
     oeu = Operator(OP_HASH_BUILD, bf)
-    oeu.add_input(supp)
+    oeu.add_input_raw(eup)
     oeu.finish()
+    fe.destroy()
 
     # partsupp restricted to qualifying parts AND european suppliers
     f1p = PlanFilterProject()
