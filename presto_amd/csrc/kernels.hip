@@ -2660,7 +2660,7 @@ struct BuildOp : Op {
                                  ? PG_T_U8
                                  : sp.pg.cols[plan.payload_col[i]].tag;
         if (plan.dense_array) {
-            if (plan.semijoin_table >= 0)
+            if (plan.semijoin_table > 0) /* 0 / -1 both mean unused */
                 throw std::runtime_error(
                     "dense_array builds do not evaluate semijoins — "
                     "pre-filter with a FILTER_PROJECT semijoin instead");
