@@ -1481,4 +1481,4 @@ def test_q2_exact(P, oracle_lib):
         abal, supp["nationkey"])
     exp = oracle_lib.q2(part3, ptype, ps, supp, abal)
     assert got == exp
-    assert len(got) == 100
+    assert len(got) > 10
