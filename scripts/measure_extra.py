@@ -57,7 +57,9 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--sf", type=float, default=30.0)
     ap.add_argument("--reps", type=int, default=3)
-    ap.add_argument("--queries", default="q8,q9,q12,q14,q17,q18,q19,q21,q22")
+    ap.add_argument("--queries",
+                    default="q2,q8,q9,q10,q12,q13,q14,q15,q16,q17,q18,"
+                            "q19,q20,q21,q22")
     args = ap.parse_args()
     sf = args.sf
     orc = OracleLib(str(REPO / "oracle" / "liboracle.so"))
@@ -80,6 +82,117 @@ def main():
             r.update(extra)
         results.append(r)
         print(json.dumps(r), flush=True)
+
+    if "q2" in want:
+        part3 = orc.gen_part3(sf)
+        ptype = orc.gen_part_type(sf)
+        ps = orc.gen_partsupp(sf)
+        abal = orc.gen_supplier_acctbal(sf)
+        n_part = len(ptype)
+        pages = (page({"partkey": np.arange(1, n_part + 1, dtype=np.int64),
+                       "type_id": ptype, "size": part3["size"]}),
+                 page({"partkey": ps["partkey"], "suppkey": ps["suppkey"],
+                       "supplycost": ps["supplycost_cents"]}),
+                 page({"suppkey": supp["suppkey"],
+                       "nationkey": supp["nationkey"]}),
+                 abal, supp["nationkey"])
+        got, secs = run("q2", lambda: P.pipelines.q2(*pages), args.reps)
+        exp = orc.q2(part3, ptype, ps, supp, abal)
+        record("q2", secs, got == exp)
+        del pages, ps
+
+    if "q10" in want:
+        n_cust = len(cust["custkey"])
+        pages = (page({k: orders[k] for k in ("orderkey", "custkey",
+                                              "orderdate")}),
+                 page({"orderkey": li["orderkey"],
+                       "returnflag": li["returnflag"],
+                       "extendedprice": li["extendedprice"],
+                       "discount": li["discount"]}))
+        got, secs = run("q10", lambda: P.pipelines.q10(n_cust, *pages),
+                        args.reps)
+        exp = orc.q10(orders, li, n_cust)
+        record("q10", secs, got == exp)
+        del pages
+
+    if "q13" in want:
+        from presto_amd.engine import Varbin
+        data, offs = orc.gen_orders_comment_varbin(sf)
+        cm = Varbin.__new__(Varbin)
+        cm.data, cm.offsets, cm.n = data, offs, len(orders["custkey"])
+        n_cust = len(cust["custkey"])
+        pages = (P.Page({"custkey": dev(orders["custkey"]),
+                         "comment": cm}),)
+        got, secs = run("q13", lambda: P.pipelines.q13(n_cust, *pages),
+                        args.reps)
+        exp = orc.q13(sf, orders)
+        record("q13", secs, got == exp)
+        del pages, data, offs
+
+    if "q15" in want:
+        pages = (page({"suppkey": supp["suppkey"],
+                       "nationkey": supp["nationkey"]}),
+                 page({"suppkey": li["suppkey"], "shipdate": li["shipdate"],
+                       "extendedprice": li["extendedprice"],
+                       "discount": li["discount"]}))
+        got, secs = run("q15", lambda: P.pipelines.q15(*pages), args.reps)
+        exp = orc.q15(li, len(supp["suppkey"]))
+        record("q15", secs, got == exp)
+        del pages
+
+    if "q16" in want:
+        part3 = orc.gen_part3(sf)
+        ptype = orc.gen_part_type(sf)
+        ps = orc.gen_partsupp(sf)
+        bbb = orc.gen_supplier_bbb(sf)
+        pool = orc.text_pool()
+        soff = np.empty(len(bbb), np.int64)
+        sln = np.empty(len(bbb), np.int32)
+        import ctypes as CT
+        orc.lib.tpch_gen_supplier_comment(
+            CT.c_double(sf), CT.c_int64(0), CT.c_int64(len(bbb)),
+            soff.ctypes.data_as(CT.c_void_p),
+            sln.ctypes.data_as(CT.c_void_p))
+        strings = []
+        for i in range(len(bbb)):
+            t = pool[soff[i]:soff[i] + sln[i]]
+            if bbb[i] == 1:
+                t = t[:5] + b"Customer criticizes Complaints" + t[5:]
+            elif bbb[i] == 2:
+                t = t[:5] + b"Customer Recommends" + t[5:]
+            strings.append(t)
+        n_part = len(ptype)
+        pages = (page({"partkey": np.arange(1, n_part + 1, dtype=np.int64),
+                       "brand": part3["brand"], "type_id": ptype,
+                       "size": part3["size"]}),
+                 page({"partkey": ps["partkey"], "suppkey": ps["suppkey"]}),
+                 P.Page({"suppkey": dev(supp["suppkey"]),
+                         "comment": P.Varbin(strings)}),
+                 orc.part_type_name)
+        got, secs = run("q16", lambda: P.pipelines.q16(*pages), args.reps)
+        exp = orc.q16(part3, ptype, ps, bbb)
+        record("q16", secs, got == exp)
+        del pages, ps, strings
+
+    if "q20" in want:
+        words = orc.gen_part_name_words(sf)
+        names = [orc.color_name(i) for i in range(92)]
+        strings = [" ".join(names[w] for w in row).encode() for row in words]
+        ps = orc.gen_partsupp(sf)
+        pages = (P.Page({"partkey": dev(np.arange(1, len(strings) + 1,
+                                                  dtype=np.int64)),
+                         "name": P.Varbin(strings)}),
+                 page({"partkey": ps["partkey"], "suppkey": ps["suppkey"],
+                       "availqty": ps["availqty"]}),
+                 page({"suppkey": supp["suppkey"],
+                       "nationkey": supp["nationkey"]}),
+                 page({"partkey": lpk, "suppkey": li["suppkey"],
+                       "quantity": li["quantity"],
+                       "shipdate": li["shipdate"]}))
+        got, secs = run("q20", lambda: P.pipelines.q20(*pages), args.reps)
+        exp = orc.q20(words, ps, li, lpk, supp)
+        record("q20", secs, got == exp)
+        del pages, ps, strings, words
 
     if "q8" in want:
         ptype = orc.gen_part_type(sf)
