@@ -1840,9 +1840,11 @@ def q10(cust_n: int, orders: Page, li: Page, limit=20):
     j2.add_input_raw(pa)
     j2.finish()
     g = j2.get_output(["custkey", "rev", "f64", "cnt"])
-    rows = sorted(((int(g["custkey"][i]), int(g["rev"][i]))
-                   for i in range(len(g["custkey"])) if g["rev"][i] > 0),
-                  key=lambda r: (-r[1], r[0]))[:limit]
+    ck, rev = g["custkey"], g["rev"]
+    nz = rev > 0
+    ck, rev = ck[nz], rev[nz]
+    top = np.lexsort((ck, -rev))[:limit]
+    rows = [(int(ck[i]), int(rev[i])) for i in top]
     j2.destroy()
     jo.destroy()
     for o in (oo, oc):
