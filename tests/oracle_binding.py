@@ -720,7 +720,9 @@ def q15(self, li, n_supp, cap=16):
     return [(int(sk[i]), int(rv[i])) for i in range(n)]
 
 
-def q20(self, words, ps, li, lpk, supp, cap=4096):
+def q20(self, words, ps, li, lpk, supp, cap=None):
+    if cap is None:
+        cap = len(supp["suppkey"]) + 1
     sk = np.empty(cap, np.int64)
     fid = self.color_id("forest")
     self.lib.oracle_q20.restype = C.c_int64
