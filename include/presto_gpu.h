@@ -275,6 +275,11 @@ typedef struct {
      *   probe columns emit_probe_cols[] + build payloads (join emit order:
      *   probe rows ascending; within a probe row, chain head first —
      *   LookupJoinPageBuilder.appendRow:75 + ArrayPositionLinks order)
+     * mode 3: fused star-join grouped SUM — probe table (chained, first
+     *   payload = an i64 grouping key, e.g. orders keyed by orderkey
+     *   with o_custkey payload), accumulate proj into table2's slots
+     *   keyed by that payload (Q10: revenue per customer in ONE pass
+     *   over lineitem).  get_output emits table2's groups page.
      * mode 1: fused grouped SUM into the build table (group = join key):
      *   for each match, table.acc += proj(probe row) in exact decimal ticks
      *   AND exact-f64 fixed-point (fixed128.h); get_output after finish

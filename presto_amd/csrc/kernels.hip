@@ -1372,6 +1372,44 @@ __global__ __launch_bounds__(256) void k_probe_agg_q3(
     }
 }
 
+/* mode 3: probe + grouped SUM keyed by the matched build row's first
+ * i64 payload, accumulated into a SECOND (agg) table — the fused
+ * star-join shape (probe li -> orders, group by o_custkey): one pass,
+ * two random lookups per row, no materialized join output. */
+__global__ __launch_bounds__(256) void k_probe_agg_pay(
+    pg_page pg, pg_plan_lookup_join plan, const int64_t* keys1,
+    const uint8_t* tags1, const int32_t* head1, int64_t mask1,
+    const int64_t* pay1, const int64_t* keys2, const uint8_t* tags2,
+    int64_t mask2, slot_acc* acc2)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < pg.n_rows; i += stride) {
+        if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
+        int64_t key = d_load_i64(pg.cols[plan.key_col], i);
+        int64_t s1 = d_tbl_find_tagged(keys1, tags1, mask1, key);
+        if (s1 < 0) continue;
+        int64_t r = head1 ? (int64_t)head1[s1] : s1;
+        if (r < 0) continue;
+        int64_t g = pay1[r];
+        int64_t s2 = d_tbl_find_tagged(keys2, tags2, mask2, g);
+        if (s2 < 0) continue;
+        pg_agg ag;
+        ag.proj = plan.proj;
+        ag.dec_scale = plan.dec_scale;
+        int64_t ticks = d_eval_proj_dec(pg, ag, i);
+        atomicAdd(&acc2[s2].dec, (unsigned long long)ticks);
+        if (!plan.dec_only) {
+            double p = d_eval_proj_f64(pg, plan.proj, i);
+            uint64_t phi, plo;
+            fx128_from_f64(p, &phi, &plo);
+            unsigned long long old = atomicAdd(&acc2[s2].flo, plo);
+            atomicAdd(&acc2[s2].fhi, phi + (old > ~plo ? 1ull : 0ull));
+        }
+        atomicAdd(&acc2[s2].cnt, 1ull);
+    }
+}
+
 /* MIN-accumulator identity init (dec = +inf sentinel, rest zero) */
 __global__ __launch_bounds__(256) void k_acc_min_init(slot_acc* acc,
                                                       int64_t cap)
@@ -2872,6 +2910,24 @@ struct JoinOp : Op {
             m2_acc.alloc(4 * 8 * 8);
             m2_acc.zero();
         }
+        if (plan.mode == 3) {
+            if (t->slot_payloads || t->ptag.empty() ||
+                t->ptag[0] != PG_T_I64)
+                throw std::runtime_error(
+                    "mode 3 needs a chained build whose first payload "
+                    "is the i64 grouping key");
+            auto it2 = g_tables.find(plan.table2);
+            if (it2 == g_tables.end() || it2->second->dense ||
+                it2->second->key_set_only)
+                throw std::runtime_error(
+                    "mode 3 needs an agg/keyed table2 for the groups");
+            t2 = it2->second.get();
+            if (!t2->acc.p) {
+                t2->acc.alloc((size_t)t2->cap * sizeof(slot_acc));
+                t2->acc.zero();
+                CHKV(hipStreamSynchronize(g_stream));
+            }
+        }
     }
     void add_input(const pg_page* in) override
     {
@@ -2887,6 +2943,21 @@ struct JoinOp : Op {
                                (const uint8_t*)t->payload[0].p,
                                (const uint8_t*)t2->payload[0].p, t2->cap,
                                a, a + 8, a + 16, a + 24);
+            hot_end();
+            CHKV(hipStreamSynchronize(g_stream));
+            return;
+        }
+        if (plan.mode == 3) {
+            hot_begin();
+            hipLaunchKernelGGL(k_probe_agg_pay, dim3(4096), dim3(256), 0,
+                               g_stream, sp.pg, plan,
+                               (const int64_t*)t->keys.p,
+                               (const uint8_t*)t->tags.p,
+                               (const int32_t*)t->head.p, t->mask,
+                               (const int64_t*)t->payload[0].p,
+                               (const int64_t*)t2->keys.p,
+                               (const uint8_t*)t2->tags.p, t2->mask,
+                               (slot_acc*)t2->acc.p);
             hot_end();
             CHKV(hipStreamSynchronize(g_stream));
             return;
@@ -3058,16 +3129,18 @@ struct JoinOp : Op {
             outq.push_back(std::move(op));
             return;
         }
-        if (plan.mode != 1) return;
-        /* extract groups: slots with count>0, slot-ascending */
-        int64_t cap = t->cap;
+        if (plan.mode != 1 && plan.mode != 3) return;
+        /* extract groups: slots with count>0, slot-ascending (mode 3
+         * groups live in table2) */
+        Table* gt = plan.mode == 3 ? t2 : t;
+        int64_t cap = gt->cap;
         int64_t chunk = (cap + FLT_NB - 1) / FLT_NB;
         chunk = (chunk + 255) / 256 * 256;
         if (chunk < 256) chunk = 256;
         DevBuf d_counts;
         d_counts.alloc(FLT_NB * 8);
         hipLaunchKernelGGL(k_groups_count, dim3(FLT_NB), dim3(256), 0,
-                           g_stream, (const slot_acc*)t->acc.p,
+                           g_stream, (const slot_acc*)gt->acc.p,
                            cap, chunk, (int64_t*)d_counts.p);
         std::vector<int64_t> h(FLT_NB);
         CHKV(hipMemcpyAsync(h.data(), d_counts.p, FLT_NB * 8,
@@ -3096,25 +3169,25 @@ struct JoinOp : Op {
         };
         int c_key = add_dev_col(PG_T_I64);
         build_payloads bp{};
-        bp.n = (int32_t)t->payload.size();
-        bp.by_slot = t->slot_payloads ? 1 : 0;
+        bp.n = (int32_t)gt->payload.size();
+        bp.by_slot = gt->slot_payloads ? 1 : 0;
         emit_outs pl_outs{};
         pl_outs.n = bp.n;
         for (int o = 0; o < bp.n; o++) {
-            bp.ptr[o] = t->payload[o].p;
-            bp.tag[o] = t->ptag[o];
-            int c = add_dev_col(t->ptag[o]);
+            bp.ptr[o] = gt->payload[o].p;
+            bp.tag[o] = gt->ptag[o];
+            int c = add_dev_col(gt->ptag[o]);
             pl_outs.ptr[o] = op.pg.cols[c].data;
-            pl_outs.tag[o] = t->ptag[o];
+            pl_outs.tag[o] = gt->ptag[o];
         }
         int c_dec = add_dev_col(PG_T_I64);
         int c_f64 = add_dev_col(PG_T_F64);
         int c_cnt = add_dev_col(PG_T_I64);
         op.pg.n_cols = nc;
         hipLaunchKernelGGL(k_groups_emit, dim3(FLT_NB), dim3(256), 0,
-                           g_stream, (const int64_t*)t->keys.p,
-                           (const int32_t*)t->head.p,
-                           (const slot_acc*)t->acc.p, bp, cap,
+                           g_stream, (const int64_t*)gt->keys.p,
+                           (const int32_t*)gt->head.p,
+                           (const slot_acc*)gt->acc.p, bp, cap,
                            chunk, (const int64_t*)d_offs.p,
                            (int64_t*)op.pg.cols[c_key].data, pl_outs,
                            (int64_t*)op.pg.cols[c_dec].data,

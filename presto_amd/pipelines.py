@@ -1806,19 +1806,6 @@ def q10(cust_n: int, orders: Page, li: Page, limit=20):
     oo.add_input(orders)
     oo.finish()
 
-    jp = PlanLookupJoin()
-    jp.table = oo.table()
-    jp.n_preds = 1
-    jp.preds[0] = Pred(li.channel("returnflag"), CMP_EQ, ord("R"), 0.0)
-    jp.key_col = li.channel("orderkey")
-    jp.mode = 0
-    jp.n_emit = 2
-    jp.emit_probe_cols[0] = li.channel("extendedprice")
-    jp.emit_probe_cols[1] = li.channel("discount")
-    jo = Operator(OP_LOOKUP_JOIN, jp)
-    jo.add_input(li)
-    pa = jo.get_output_raw()  # [ep, dc, custkey]
-
     bc = PlanHashBuild()
     bc.key_col = 0
     bc.semijoin_table = -1
@@ -1829,15 +1816,21 @@ def q10(cust_n: int, orders: Page, li: Page, limit=20):
     oc.add_input(keys)
     oc.finish()
 
+    # one fused pass: probe orders by orderkey, group the revenue by the
+    # matched o_custkey payload into the customer table (mode 3)
     ja = PlanLookupJoin()
-    ja.table = oc.table()
-    ja.key_col = 2
-    ja.mode = 1
-    ja.proj = Proj(PROJ_DISC_PRICE, 0, 1, 0)
+    ja.table = oo.table()
+    ja.table2 = oc.table()
+    ja.n_preds = 1
+    ja.preds[0] = Pred(li.channel("returnflag"), CMP_EQ, ord("R"), 0.0)
+    ja.key_col = li.channel("orderkey")
+    ja.mode = 3
+    ja.proj = Proj(PROJ_DISC_PRICE, li.channel("extendedprice"),
+                   li.channel("discount"), 0)
     ja.dec_scale = 4
     ja.dec_only = 1
     j2 = Operator(OP_LOOKUP_JOIN, ja)
-    j2.add_input_raw(pa)
+    j2.add_input(li)
     j2.finish()
     g = j2.get_output(["custkey", "rev", "f64", "cnt"])
     ck, rev = g["custkey"], g["rev"]
@@ -1846,7 +1839,6 @@ def q10(cust_n: int, orders: Page, li: Page, limit=20):
     top = np.lexsort((ck, -rev))[:limit]
     rows = [(int(ck[i]), int(rev[i])) for i in top]
     j2.destroy()
-    jo.destroy()
     for o in (oo, oc):
         lib().c.pg_table_destroy(o.table())
         o.destroy()
