@@ -1836,6 +1836,10 @@ def q10(cust_n: int, orders: Page, li: Page, limit=20):
     ck, rev = g["custkey"], g["rev"]
     nz = rev > 0
     ck, rev = ck[nz], rev[nz]
+    if len(rev) > limit:  # O(n) preselect, ties kept for the exact sort
+        kth = np.partition(rev, len(rev) - limit)[len(rev) - limit]
+        keep = rev >= kth
+        ck, rev = ck[keep], rev[keep]
     top = np.lexsort((ck, -rev))[:limit]
     rows = [(int(ck[i]), int(rev[i])) for i in top]
     j2.destroy()
