@@ -1,6 +1,6 @@
 #!/usr/bin/env python3
-"""Measurement harness for the beyond-contract query pipelines
-(q8/q9/q12/q14/q17/q18/q19/q21/q22) at a given scale factor.
+"""Measurement harness for the fifteen beyond-contract query pipelines
+(everything outside bench.py's q1/q3/q5) at a given scale factor.
 
 TEST/BENCH INFRASTRUCTURE: generates inputs with the oracle's dbgen
 restatement (like bench.py), stages the hot columns into HBM as torch
