@@ -380,7 +380,7 @@ __global__ __launch_bounds__(FT_NTHREADS) void k_agg_small(
 
     const int64_t n = pg.n_rows;
     const int64_t v = (int64_t)blockIdx.x * FT_NTHREADS + threadIdx.x;
-    unsigned long long local_bad = 0;
+    unsigned long long local_bad = 0, local_ovf = 0;
 
     for (int64_t base = 2 * v; base < n; base += 2 * FT_VL) {
         int64_t lim = base + 2 < n ? base + 2 : n;
@@ -443,8 +443,18 @@ __global__ __launch_bounds__(FT_NTHREADS) void k_agg_small(
                     else if (fn == PG_AGG_MAX)
                         acc[a][gg] =
                             m && val > acc[a][gg] ? val : acc[a][gg];
-                    else
+                    else if constexpr (DEC) {
+                        /* overflow-checked add: LongSumAggregation.java:33-37
+                         * raises via Math.addExact — wraps are surfaced as
+                         * an operator error, never silent */
+                        int64_t cur = (int64_t)acc[a][gg], nv;
+                        if (__builtin_add_overflow(
+                                cur, (int64_t)(m ? val : (T)0), &nv))
+                            local_ovf++;
+                        acc[a][gg] = (T)nv;
+                    } else {
                         acc[a][gg] += m ? val : (T)0;
+                    }
                 }
             }
         }
@@ -472,8 +482,14 @@ __global__ __launch_bounds__(FT_NTHREADS) void k_agg_small(
                     s = o < s ? o : s;
                 else if (fn == PG_AGG_MAX)
                     s = o > s ? o : s;
-                else
+                else if constexpr (DEC) {
+                    int64_t nv;
+                    if (__builtin_add_overflow((int64_t)s, (int64_t)o, &nv))
+                        local_ovf++;
+                    s = (T)nv;
+                } else {
                     s = s + o;
+                }
             }
             if (lane == 0) lds[wid][a][g] = s;
         }
@@ -497,8 +513,15 @@ __global__ __launch_bounds__(FT_NTHREADS) void k_agg_small(
                         x = o < x ? o : x;
                     else if (fn == PG_AGG_MAX)
                         x = o > x ? o : x;
-                    else
+                    else if constexpr (DEC) {
+                        int64_t nv;
+                        if (__builtin_add_overflow((int64_t)x, (int64_t)o,
+                                                   &nv))
+                            local_ovf++;
+                        x = (T)nv;
+                    } else {
                         x = x + o;
+                    }
                 }
                 if (lane == 0) {
                     size_t off =
@@ -509,8 +532,8 @@ __global__ __launch_bounds__(FT_NTHREADS) void k_agg_small(
                             *d = (int64_t)x < *d ? (int64_t)x : *d;
                         else if (fn == PG_AGG_MAX)
                             *d = (int64_t)x > *d ? (int64_t)x : *d;
-                        else
-                            *d += (int64_t)x;
+                        else if (__builtin_add_overflow(*d, (int64_t)x, d))
+                            local_ovf++;
                     } else {
                         double* d = &partials[off];
                         if (fn == PG_AGG_MIN)
@@ -525,6 +548,7 @@ __global__ __launch_bounds__(FT_NTHREADS) void k_agg_small(
         }
     }
     if (local_bad) atomicAdd(bad_keys, local_bad);
+    if (local_ovf) atomicAdd(bad_keys + 1, local_ovf);
 }
 
 
@@ -1239,13 +1263,27 @@ struct slot_acc {
     unsigned long long cnt;
 };
 
+/* overflow-checked atomic tick add: a wrapped int64 tick sum is raised as
+ * an operator error, never silent (Math.addExact semantics,
+ * LongSumAggregation.java:33-37; exact-decimal sums depend on never
+ * wrapping).  Detection is post-hoc from the returned old value. */
+__device__ inline void d_atomic_add_dec_ck(unsigned long long* slot,
+                                           int64_t ticks,
+                                           unsigned long long* ovf)
+{
+    long long old = (long long)atomicAdd(slot, (unsigned long long)ticks);
+    long long nv = old + ticks;
+    if (((old ^ nv) & (ticks ^ nv)) < 0 && ovf) atomicAdd(ovf, 1ull);
+}
+
 /* probe + fused grouped SUM into table accumulators (Q3's
  * LookupJoinOperator + HashAggregationOperator fused; revenue summed
  * exactly in decimal ticks AND in 64.64 fixed point — order-independent,
  * so atomics preserve bit-determinism) */
 __global__ __launch_bounds__(256) void k_probe_agg(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
-    const uint8_t* tags, int64_t mask, slot_acc* acc)
+    const uint8_t* tags, int64_t mask, slot_acc* acc,
+    unsigned long long* ovf)
 {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -1261,7 +1299,7 @@ __global__ __launch_bounds__(256) void k_probe_agg(
         if (plan.dec_min)
             atomicMin((long long*)&acc[s].dec, (long long)ticks);
         else
-            atomicAdd(&acc[s].dec, (unsigned long long)ticks);
+            d_atomic_add_dec_ck(&acc[s].dec, ticks, ovf);
         if (!plan.dec_only && !plan.dec_min) {
             double p = d_eval_proj_f64(pg, plan.proj, i);
             uint64_t phi, plo;
@@ -1283,7 +1321,7 @@ __global__ __launch_bounds__(256) void k_probe_agg_q3(
     const int32_t* sd /* nullable pred col */, int32_t pred_op,
     int32_t pred_val, const int64_t* okey, const double* ep,
     const double* dc, int64_t n, const int64_t* keys, const uint8_t* tags,
-    int64_t mask, slot_acc* acc)
+    int64_t mask, slot_acc* acc, unsigned long long* ovf)
 {
     typedef double vd2 __attribute__((ext_vector_type(2)));
     typedef int vi2 __attribute__((ext_vector_type(2)));
@@ -1364,7 +1402,7 @@ __global__ __launch_bounds__(256) void k_probe_agg_q3(
             double pr = e * (1.0 - d);
             uint64_t phi, plo;
             fx128_from_f64(pr, &phi, &plo);
-            atomicAdd(&acc[sl].dec, (unsigned long long)ticks);
+            d_atomic_add_dec_ck(&acc[sl].dec, ticks, ovf);
             unsigned long long old = atomicAdd(&acc[sl].flo, plo);
             atomicAdd(&acc[sl].fhi, phi + (old > ~plo ? 1ull : 0ull));
             atomicAdd(&acc[sl].cnt, 1ull);
@@ -1380,7 +1418,7 @@ __global__ __launch_bounds__(256) void k_probe_agg_pay(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys1,
     const uint8_t* tags1, const int32_t* head1, int64_t mask1,
     const int64_t* pay1, const int64_t* keys2, const uint8_t* tags2,
-    int64_t mask2, slot_acc* acc2)
+    int64_t mask2, slot_acc* acc2, unsigned long long* ovf)
 {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -1398,7 +1436,7 @@ __global__ __launch_bounds__(256) void k_probe_agg_pay(
         ag.proj = plan.proj;
         ag.dec_scale = plan.dec_scale;
         int64_t ticks = d_eval_proj_dec(pg, ag, i);
-        atomicAdd(&acc2[s2].dec, (unsigned long long)ticks);
+        d_atomic_add_dec_ck(&acc2[s2].dec, ticks, ovf);
         if (!plan.dec_only) {
             double p = d_eval_proj_f64(pg, plan.proj, i);
             uint64_t phi, plo;
@@ -1593,7 +1631,8 @@ __global__ __launch_bounds__(256) void k_probe_agg_fused2(
 #pragma unroll
         for (int g = 0; g < MAXG; g++) {
             bool m = g == gi;
-            acc[g] += m ? ticks : 0;
+            if (__builtin_add_overflow(acc[g], m ? ticks : 0, &acc[g]))
+                atomicAdd(out_cnt + MAXG, 1ull); /* ovf slot */
             cnt[g] += m ? 1 : 0;
             uint64_t nlo = flo[g] + (m ? plo : 0);
             fhi[g] += (m ? phi : 0) + (nlo < flo[g] ? 1u : 0u);
@@ -1608,7 +1647,7 @@ __global__ __launch_bounds__(256) void k_probe_agg_fused2(
         int64_t a = d_bfly_i64(acc[g]);
         int64_t c = d_bfly_i64((int64_t)cnt[g]);
         if ((threadIdx.x & 63) == 0) {
-            if (a) atomicAdd(&out_dec[g], (unsigned long long)a);
+            if (a) d_atomic_add_dec_ck(&out_dec[g], a, out_cnt + MAXG);
             if (c) atomicAdd(&out_cnt[g], (unsigned long long)c);
         }
         if (flo[g] | fhi[g]) {
@@ -2518,19 +2557,35 @@ struct AggSmallOp : Op {
         std::vector<double> hf(nm);
         std::vector<int64_t> hhi(nm);
         std::vector<uint64_t> hlo(nm);
-        unsigned long long hbad = 0;
+        unsigned long long hbad[2] = {0, 0};
         CHKV(hipMemcpyAsync(hf.data(), out_f.p, nm * 8, hipMemcpyDeviceToHost,
                             g_stream));
         CHKV(hipMemcpyAsync(hhi.data(), out_hi.p, nm * 8,
                             hipMemcpyDeviceToHost, g_stream));
         CHKV(hipMemcpyAsync(hlo.data(), out_lo.p, nm * 8,
                             hipMemcpyDeviceToHost, g_stream));
-        CHKV(hipMemcpyAsync(&hbad, bad_ptr(), 8, hipMemcpyDeviceToHost,
+        CHKV(hipMemcpyAsync(hbad, bad_ptr(), 16, hipMemcpyDeviceToHost,
                             g_stream));
         CHKV(hipStreamSynchronize(g_stream));
-        if (hbad)
+        if (hbad[0])
             throw std::runtime_error(
                 "group key value outside plan enumeration");
+        if (hbad[1])
+            throw std::runtime_error(
+                "bigint/decimal SUM overflow (Math.addExact semantics, "
+                "LongSumAggregation.java:33-37)");
+        /* a SUM(bigint)'s 128-bit total past int64 range is the same
+         * overflow Presto raises on */
+        for (int a = 0; a < user_aggs; a++)
+            if (dec && plan.aggs[a].func == PG_AGG_SUM_I64)
+                for (int g = 0; g < n_groups; g++) {
+                    int64_t hi = hhi[a * maxg + g];
+                    int64_t lo = (int64_t)hlo[a * maxg + g];
+                    if (hi != (lo < 0 ? -1 : 0))
+                        throw std::runtime_error(
+                            "BIGINT sum overflow (Math.addExact semantics, "
+                            "LongSumAggregation.java:33-37)");
+                }
         /* presence from internal count agg (index user_aggs) */
         int ic = user_aggs;
         auto cnt_of = [&](int g) -> int64_t {
@@ -2870,6 +2925,7 @@ struct JoinOp : Op {
     Table* t = nullptr;
     Table* t2 = nullptr; /* mode 2 dense table */
     DevBuf m2_acc;       /* mode 2: [dec, flo, fhi, cnt] x 8 groups */
+    DevBuf ovf;          /* tick-sum overflow counter (modes 1/3) */
     void init()
     {
         std::lock_guard<std::mutex> lk(g_mu);
@@ -2883,6 +2939,10 @@ struct JoinOp : Op {
                 "cannot probe a key-set-only table");
         /* mode 0 emit over a slot-payload table: unique keys, payloads
          * indexed by slot (no chains) */
+        if (plan.mode == 1 || plan.mode == 3) {
+            ovf.alloc(8);
+            ovf.zero();
+        }
         if (plan.mode == 1 && !t->acc.p) {
             t->acc.alloc((size_t)t->cap * sizeof(slot_acc));
             t->acc.zero();
@@ -2907,7 +2967,7 @@ struct JoinOp : Op {
             t2 = it2->second.get();
             if (plan.n_group_vals < 1 || plan.n_group_vals > 8)
                 throw std::runtime_error("n_group_vals must be 1..8");
-            m2_acc.alloc(4 * 8 * 8);
+            m2_acc.alloc(4 * 8 * 8 + 8); /* + tick-overflow counter */
             m2_acc.zero();
         }
         if (plan.mode == 3) {
@@ -2957,7 +3017,8 @@ struct JoinOp : Op {
                                (const int64_t*)t->payload[0].p,
                                (const int64_t*)t2->keys.p,
                                (const uint8_t*)t2->tags.p, t2->mask,
-                               (slot_acc*)t2->acc.p);
+                               (slot_acc*)t2->acc.p,
+                               (unsigned long long*)ovf.p);
             hot_end();
             CHKV(hipStreamSynchronize(g_stream));
             return;
@@ -2998,13 +3059,15 @@ struct JoinOp : Op {
                     (const double*)sp.pg.cols[plan.proj.b].data,
                     sp.pg.n_rows, (const int64_t*)t->keys.p,
                     (const uint8_t*)t->tags.p, t->mask,
-                    (slot_acc*)t->acc.p);
+                    (slot_acc*)t->acc.p,
+                    (unsigned long long*)ovf.p);
             } else {
                 hipLaunchKernelGGL(k_probe_agg, dim3(4096), dim3(256), 0,
                                    g_stream, sp.pg, plan,
                                    (const int64_t*)t->keys.p,
                                    (const uint8_t*)t->tags.p, t->mask,
-                                   (slot_acc*)t->acc.p);
+                                   (slot_acc*)t->acc.p,
+                                   (unsigned long long*)ovf.p);
             }
             hot_end();
             CHKV(hipStreamSynchronize(g_stream));
@@ -3096,8 +3159,12 @@ struct JoinOp : Op {
     void finish() override
     {
         if (plan.mode == 2) {
-            unsigned long long h[32];
+            unsigned long long h[33];
             CHKV(hipMemcpy(h, m2_acc.p, sizeof(h), hipMemcpyDeviceToHost));
+            if (h[32])
+                throw std::runtime_error(
+                    "decimal SUM overflow in fused probe "
+                    "(Math.addExact semantics)");
             int n_out = 0;
             for (int g = 0; g < plan.n_group_vals; g++)
                 if (h[24 + g]) n_out++;
@@ -3130,6 +3197,15 @@ struct JoinOp : Op {
             return;
         }
         if (plan.mode != 1 && plan.mode != 3) return;
+        {
+            unsigned long long h_ovf = 0;
+            CHKV(hipMemcpy(&h_ovf, ovf.p, 8, hipMemcpyDeviceToHost));
+            if (h_ovf)
+                throw std::runtime_error(
+                    "bigint/decimal SUM overflow in grouped probe "
+                    "(Math.addExact semantics, "
+                    "LongSumAggregation.java:33-37)");
+        }
         /* extract groups: slots with count>0, slot-ascending (mode 3
          * groups live in table2) */
         Table* gt = plan.mode == 3 ? t2 : t;
@@ -3352,7 +3428,11 @@ struct PartitionOp : Op {
     }
 };
 
-static std::map<int64_t, std::unique_ptr<Op>> g_ops;
+/* shared_ptr so a concurrent pg_op_destroy cannot free an Op another
+ * thread's call still holds (each C-ABI entry copies the ref; the
+ * per-handle single-threaded contract of presto_gpu.h still applies to
+ * calls on the SAME handle) */
+static std::map<int64_t, std::shared_ptr<Op>> g_ops;
 static int64_t g_next_op = 1;
 
 } /* namespace */
@@ -3429,21 +3509,21 @@ extern "C" pg_status pg_op_create(int32_t kind, const void* plan,
     }
 }
 
-static Op* find_op(pg_op h)
+static std::shared_ptr<Op> find_op(pg_op h)
 {
     std::lock_guard<std::mutex> lk(g_mu);
     auto it = g_ops.find(h);
-    return it == g_ops.end() ? nullptr : it->second.get();
+    return it == g_ops.end() ? nullptr : it->second;
 }
 
 extern "C" int32_t pg_op_needs_input(pg_op h)
 {
-    Op* op = find_op(h);
+    auto op = find_op(h);
     return op && !op->finished ? 1 : 0;
 }
 extern "C" pg_status pg_op_add_input(pg_op h, const pg_page* page)
 {
-    Op* op = find_op(h);
+    auto op = find_op(h);
     if (!op) return seterr("bad op handle");
     if (op->finished) return seterr("addInput after finish");
     try {
@@ -3455,7 +3535,7 @@ extern "C" pg_status pg_op_add_input(pg_op h, const pg_page* page)
 }
 extern "C" pg_status pg_op_get_output(pg_op h, const pg_page** out)
 {
-    Op* op = find_op(h);
+    auto op = find_op(h);
     if (!op) return seterr("bad op handle");
     try {
         *out = op->pop_output();
@@ -3466,7 +3546,7 @@ extern "C" pg_status pg_op_get_output(pg_op h, const pg_page** out)
 }
 extern "C" pg_status pg_op_finish(pg_op h)
 {
-    Op* op = find_op(h);
+    auto op = find_op(h);
     if (!op) return seterr("bad op handle");
     if (op->finished) return PG_OK; /* idempotent */
     try {
@@ -3479,7 +3559,7 @@ extern "C" pg_status pg_op_finish(pg_op h)
 }
 extern "C" int32_t pg_op_is_finished(pg_op h)
 {
-    Op* op = find_op(h);
+    auto op = find_op(h);
     return op && op->finished && op->outq.empty() ? 1 : 0;
 }
 extern "C" pg_status pg_op_destroy(pg_op h)
@@ -3490,7 +3570,7 @@ extern "C" pg_status pg_op_destroy(pg_op h)
 }
 extern "C" pg_status pg_op_table(pg_op h, int64_t* out)
 {
-    Op* op = find_op(h);
+    auto op = find_op(h);
     if (!op) return seterr("bad op handle");
     int64_t t = op->table_handle();
     if (t < 0) return seterr("op has no table (finish the build first)");
@@ -3500,7 +3580,7 @@ extern "C" pg_status pg_op_table(pg_op h, int64_t* out)
 extern "C" pg_status pg_op_partition_counts(pg_op h, int64_t* counts,
                                             int32_t n)
 {
-    Op* op = find_op(h);
+    auto op = find_op(h);
     if (!op) return seterr("bad op handle");
     try {
         op->partition_counts(counts, n);

@@ -288,10 +288,23 @@ def q5_distributed(cust_page, ord_page, supp_page, li_page, world, rank,
             dist.all_reduce(full)  # disjoint shards -> sum == union
         return full
 
-    # callers set .n_total (global table cardinality) on the dimension
-    # pages; default = local rows x world (exact when shards are even)
-    n_cust_all = getattr(cust_page, "n_total", cust_page.n_rows * world)
-    n_supp_all = getattr(supp_page, "n_total", supp_page.n_rows * world)
+    # callers may set .n_total (global table cardinality) on the dimension
+    # pages; otherwise derive it exactly as all_reduce(MAX) of the local
+    # max key (dense 1..n keys), which is correct for uneven shards too
+    def global_cardinality(page, key_name):
+        n = getattr(page, "n_total", None)
+        if n is not None:
+            return int(n)
+        local_max = int(page.cols[key_name].max().item()) \
+            if page.n_rows else 0
+        if use_dist:
+            t = torch.tensor([local_max], dtype=torch.int64, device=device)
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+            local_max = int(t.item())
+        return local_max
+
+    n_cust_all = global_cardinality(cust_page, "custkey")
+    n_supp_all = global_cardinality(supp_page, "suppkey")
     cnat = global_dense(cust_page, "custkey", "nationkey", n_cust_all)
     snat = global_dense(supp_page, "suppkey", "nationkey", n_supp_all)
 
@@ -404,12 +417,16 @@ def q5_distributed(cust_page, ord_page, supp_page, li_page, world, rank,
     for b in (bc, bs, b2):
         lib().c.pg_table_destroy(b.table())
         b.destroy()
-    # exact final combine: per-nation tick sums + counts
+    # exact final combine: per-nation tick sums.  Per-rank sums are
+    # un-wrapped (the probe kernel raises on int64 tick overflow); bound
+    # them so the cross-rank all_reduce cannot wrap either.
     ticks = torch.zeros(len(Q5Pipeline.ASIA), dtype=torch.int64,
                         device=device)
     for i in range(len(out["nationkey"])):
-        ticks[Q5Pipeline.ASIA.index(int(out["nationkey"][i]))] = \
-            int(out["rev_lo"][i])
+        t = int(out["rev_lo"][i])
+        assert abs(t) < (1 << 62) // max(world, 1), \
+            "per-rank nation revenue too large for exact int64 combine"
+        ticks[Q5Pipeline.ASIA.index(int(out["nationkey"][i]))] = t
     if use_dist:
         dist.all_reduce(ticks)
     if rank != 0:

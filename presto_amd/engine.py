@@ -338,11 +338,16 @@ class Operator:
         return list(arr)
 
     def destroy(self):
-        lib().c.pg_op_destroy(self.h)
+        h, self.h = self.h, None
+        if h is not None:
+            lib().c.pg_op_destroy(h)
 
     def __del__(self):
+        # guard on self.h so a stale finalizer can never destroy a live
+        # operator if handle reuse is ever introduced
         try:
-            if _lib is not None:
+            if _lib is not None and self.h is not None:
                 _lib.c.pg_op_destroy(self.h)
+                self.h = None
         except Exception:
             pass

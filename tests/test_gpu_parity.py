@@ -1482,3 +1482,66 @@ def test_q2_exact(P, oracle_lib):
     exp = oracle_lib.q2(part3, ptype, ps, supp, abal)
     assert got == exp
     assert len(got) > 10
+
+
+def test_sum_bigint_overflow_raises(P):
+    """SUM(bigint) overflow is a loud operator error, not a silent wrap
+    (LongSumAggregation.java:33-37 raises via Math.addExact)."""
+    n = 1024
+    vals = np.full(n, (1 << 62) + 12345, np.int64)
+    key = np.zeros(n, np.uint8)
+    page = P.Page({"k": key, "v": vals})
+    plan = P.PlanHashAggSmall()
+    plan.n_keys = 1
+    plan.key_col[0] = 0
+    plan.n_vals[0] = 1
+    plan.key_vals[0][0] = 0
+    plan.n_aggs = 1
+    plan.aggs[0] = P.Agg(P.AGG_SUM_I64, P.Proj(P.PROJ_IDENT, 1, 0, 0), 0)
+    op = P.Operator(P.OP_HASH_AGG_SMALL, plan)
+    op.add_input(page)
+    with pytest.raises(RuntimeError, match="overflow"):
+        op.finish()
+        op.get_output()
+    op.destroy()
+    # sanity: the same shape with small values still sums exactly
+    vals2 = np.arange(n, dtype=np.int64)
+    page2 = P.Page({"k": key, "v": vals2})
+    op2 = P.Operator(P.OP_HASH_AGG_SMALL, plan)
+    op2.add_input(page2)
+    op2.finish()
+    out = op2.get_output(["k", "hi", "lo"])
+    assert out["lo"][0] == vals2.sum()
+    op2.destroy()
+
+
+def test_grouped_probe_sum_overflow_raises(P):
+    """Tick-sum overflow inside a fused grouped probe raises too."""
+    bk = np.array([7], np.int64)
+    bp = P.PlanHashBuild()
+    bp.key_col = 0
+    bp.semijoin_table = -1
+    bp.capacity_hint = 16
+    bp.agg_table = 1
+    b = P.Operator(P.OP_HASH_BUILD, bp)
+    b.add_input(P.Page({"k": bk}))
+    b.finish()
+    probe_n = 8
+    pk = np.full(probe_n, 7, np.int64)
+    pv = np.full(probe_n, (1 << 62) + 99, np.int64)
+    jp = P.PlanLookupJoin()
+    jp.table = b.table()
+    jp.key_col = 0
+    jp.mode = 1
+    jp.proj = P.Proj(P.PROJ_IDENT, 1, 0, 0)
+    jp.dec_scale = 0
+    jp.dec_only = 1
+    j = P.Operator(P.OP_LOOKUP_JOIN, jp)
+    j.add_input(P.Page({"k": pk, "v": pv}))
+    with pytest.raises(RuntimeError, match="overflow"):
+        j.finish()
+        j.get_output()
+    j.destroy()
+    from presto_amd.engine import lib
+    lib().c.pg_table_destroy(b.table())
+    b.destroy()
