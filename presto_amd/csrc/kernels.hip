@@ -1112,9 +1112,18 @@ __global__ __launch_bounds__(256) void k_tbl_insert(const int64_t* in_keys,
     }
 }
 
+/* linear-probe step with partition-local wraparound: partitioned tables
+ * (radix-partitioned agg builds) wrap within their cap_p-slot
+ * region; unpartitioned tables pass lmask == mask (identical behavior) */
+__device__ inline int64_t d_probe_next(int64_t s, int64_t lmask)
+{
+    return (s & ~lmask) | ((s + 1) & lmask);
+}
+
 __device__ inline int64_t d_tbl_find_tagged(const int64_t* keys,
                                             const uint8_t* tags,
-                                            int64_t mask, int64_t key);
+                                            int64_t mask, int64_t lmask,
+                                            int64_t key);
 
 /* direct insert for agg tables: filter+semijoin+insert in ONE scan of the
  * build input; payloads stored per SLOT (keys unique).  count tracks
@@ -1128,8 +1137,8 @@ struct direct_payloads {
 __global__ __launch_bounds__(256) void k_tbl_insert_direct(
     pg_page pg, pg_plan_hash_build plan, const int64_t* set_keys,
     int64_t set_mask, const int64_t* lu_keys, const uint8_t* lu_tags,
-    int64_t lu_mask, const uint8_t* lu_payload, int64_t* keys,
-    uint8_t* tags, direct_payloads dp, int64_t mask,
+    int64_t lu_mask, int64_t lu_lmask, const uint8_t* lu_payload,
+    int64_t* keys, uint8_t* tags, direct_payloads dp, int64_t mask,
     unsigned long long* inserted, unsigned long long* overflow)
 {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -1146,7 +1155,8 @@ __global__ __launch_bounds__(256) void k_tbl_insert_direct(
                 d_load_i64(pg.cols[plan.payload_lookup_key_col], i);
             int64_t sl2;
             if (lu_keys) {
-                sl2 = d_tbl_find_tagged(lu_keys, lu_tags, lu_mask, k2);
+                sl2 = d_tbl_find_tagged(lu_keys, lu_tags, lu_mask,
+                                        lu_lmask, k2);
             } else { /* dense: payload[key-1], lu_mask = capacity */
                 sl2 = (k2 >= 1 && k2 <= lu_mask) ? k2 - 1 : -1;
             }
@@ -1224,8 +1234,261 @@ __global__ __launch_bounds__(256) void k_tbl_insert_direct(
     }
 }
 
+/* ------------------------------------------------------------------ */
+/* Radix-partitioned agg-table build (round 2).                        */
+/* The old direct insert (k_tbl_insert_direct) random-stores into the  */
+/* whole table: at SF100 the Q3 orders build fetched ~12.5 GB against  */
+/* ~3 GB of algorithmic input (profiles/r01_q3_sf100_pmc.txt) because  */
+/* every insert pulls whole 64-B lines of an HBM-resident table.       */
+/* Fix: (A) one fused filter+semijoin+lookup+hash pass scatters the    */
+/* surviving {key, payload} records into per-partition staging runs    */
+/* (partition = high bits of the masked slot index, so probes keep     */
+/* bit-identical slot math); (B) a persistent kernel then initializes  */
+/* and fills K partition regions at a time — each region sized to sit  */
+/* in the 256 MiB Infinity Cache — separated by grid barriers, so the  */
+/* random CAS/stores hit L3 instead of HBM and each table line is      */
+/* written back exactly once.                                          */
+/* ------------------------------------------------------------------ */
+#define SCAT_TILE 8192
+#define SCAT_MAXP 256
+
+struct build_row {
+    int64_t key;
+    int64_t pay[4];
+};
+
+/* shared row evaluation for the partitioned build: predicates, fused
+ * dimension lookup, semijoin, then key + payload words (f64 payloads are
+ * bit-copied).  Returns 0 when the row is dropped. */
+__device__ inline int d_build_eval(
+    const pg_page& pg, const pg_plan_hash_build& plan,
+    const int64_t* set_keys, int64_t set_mask, const int64_t* lu_keys,
+    const uint8_t* lu_tags, int64_t lu_mask, int64_t lu_lmask,
+    const uint8_t* lu_payload, const direct_payloads& dp, int64_t i,
+    bool want_pay, build_row* out)
+{
+    if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) return 0;
+    uint8_t plv = 0;
+    if (lu_payload) {
+        int64_t k2 = d_load_i64(pg.cols[plan.payload_lookup_key_col], i);
+        int64_t sl2;
+        if (lu_keys)
+            sl2 = d_tbl_find_tagged(lu_keys, lu_tags, lu_mask, lu_lmask,
+                                    k2);
+        else /* dense: payload[key-1], lu_mask = capacity */
+            sl2 = (k2 >= 1 && k2 <= lu_mask) ? k2 - 1 : -1;
+        if (sl2 < 0) return 0;
+        plv = lu_payload[sl2];
+    }
+    if (set_keys) {
+        int64_t sk = d_load_i64(pg.cols[plan.semijoin_col], i);
+        bool found;
+        if (set_mask < 0) {
+            found = sk >= 1 && sk <= -set_mask &&
+                    ((const uint8_t*)set_keys)[sk - 1];
+        } else {
+            found = false;
+            uint64_t h = pg_murmur3_finalize(pg_bigint_hash(sk));
+            int64_t s = (int64_t)(h & (uint64_t)set_mask);
+            for (;;) {
+                int64_t k = set_keys[s];
+                if (k == sk) { found = true; break; }
+                if (k == TBL_EMPTY) break;
+                s = (s + 1) & set_mask;
+            }
+        }
+        if (!found) return 0;
+    }
+    out->key = d_load_i64(pg.cols[plan.key_col], i);
+    if (want_pay) {
+        if (lu_payload && dp.n >= 1) out->pay[0] = (int64_t)plv;
+        for (int o = lu_payload ? 1 : 0; o < dp.n; o++) {
+            const pg_col& c = pg.cols[dp.src[o]];
+            if (dp.tag[o] == PG_T_F64) {
+                double v = d_load_f64(c, i);
+                int64_t b;
+                memcpy(&b, &v, 8);
+                out->pay[o] = b;
+            } else {
+                out->pay[o] = d_load_i64(c, i);
+            }
+        }
+    }
+    return 1;
+}
+
+/* (A) fused filter + partition scatter: per-tile LDS counting sort
+ * (count, one global cursor reservation per partition per tile, place),
+ * so per-partition staging runs stay dense and appends cost ~P global
+ * atomics per 8192-row tile instead of one per row. */
+__global__ __launch_bounds__(256) void k_part_scatter(
+    pg_page pg, pg_plan_hash_build plan, const int64_t* set_keys,
+    int64_t set_mask, const int64_t* lu_keys, const uint8_t* lu_tags,
+    int64_t lu_mask, int64_t lu_lmask, const uint8_t* lu_payload,
+    direct_payloads dp, int64_t mask, int32_t capp_bits, int32_t P,
+    int64_t* stage, unsigned long long* cursor, int64_t cap_stage_rows,
+    int32_t r_words, unsigned long long* stage_ovf)
+{
+    __shared__ unsigned int cnt[SCAT_MAXP];
+    __shared__ unsigned int c2[SCAT_MAXP];
+    __shared__ unsigned long long gbase[SCAT_MAXP];
+    const int64_t n = pg.n_rows;
+    for (int64_t t0 = (int64_t)blockIdx.x * SCAT_TILE; t0 < n;
+         t0 += (int64_t)gridDim.x * SCAT_TILE) {
+        const int64_t t1 = t0 + SCAT_TILE < n ? t0 + SCAT_TILE : n;
+        for (int p = threadIdx.x; p < P; p += 256) {
+            cnt[p] = 0;
+            c2[p] = 0;
+        }
+        __syncthreads();
+        for (int64_t i = t0 + threadIdx.x; i < t1; i += 256) {
+            build_row r;
+            if (!d_build_eval(pg, plan, set_keys, set_mask, lu_keys,
+                              lu_tags, lu_mask, lu_lmask, lu_payload, dp,
+                              i, false, &r))
+                continue;
+            uint64_t h = pg_murmur3_finalize(pg_bigint_hash(r.key));
+            int p = (int)((h & (uint64_t)mask) >> capp_bits);
+            atomicAdd(&cnt[p], 1u);
+        }
+        __syncthreads();
+        for (int p = threadIdx.x; p < P; p += 256) {
+            if (cnt[p]) {
+                unsigned long long b =
+                    atomicAdd(&cursor[p], (unsigned long long)cnt[p]);
+                gbase[p] = b;
+                if (b + cnt[p] > (unsigned long long)cap_stage_rows)
+                    atomicAdd(stage_ovf, 1ull);
+            }
+        }
+        __syncthreads();
+        for (int64_t i = t0 + threadIdx.x; i < t1; i += 256) {
+            build_row r;
+            if (!d_build_eval(pg, plan, set_keys, set_mask, lu_keys,
+                              lu_tags, lu_mask, lu_lmask, lu_payload, dp,
+                              i, true, &r))
+                continue;
+            uint64_t h = pg_murmur3_finalize(pg_bigint_hash(r.key));
+            int p = (int)((h & (uint64_t)mask) >> capp_bits);
+            unsigned long long row = gbase[p] + atomicAdd(&c2[p], 1u);
+            if (row >= (unsigned long long)cap_stage_rows) continue;
+            int64_t* rec = stage +
+                ((int64_t)p * cap_stage_rows + (int64_t)row) * r_words;
+            rec[0] = r.key;
+            for (int o = 0; o < dp.n; o++) rec[1 + o] = r.pay[o];
+        }
+        __syncthreads();
+    }
+}
+
+/* grid barrier for the persistent insert kernel: monotonic counter,
+ * agent-scope release before arrive / acquire after (per-XCD L2s are not
+ * coherent — MI355X_MICROARCH.md §Workgroup dispatch; the inline
+ * s_waitcnt guards the known ROCm 7.2 release-fence hazard).  Caller
+ * guarantees every block is resident (grid <= 1 block per CU). */
+__device__ inline void d_grid_barrier(unsigned long long* bar,
+                                      unsigned long long target)
+{
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __hip_atomic_fetch_add(bar, 1ull, __ATOMIC_RELAXED,
+                               __HIP_MEMORY_SCOPE_AGENT);
+        while (__hip_atomic_load(bar, __ATOMIC_RELAXED,
+                                 __HIP_MEMORY_SCOPE_AGENT) < target)
+            __builtin_amdgcn_s_sleep(32);
+        __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+    }
+    __syncthreads();
+}
+
+/* (B) persistent partitioned insert: waves of K L3-resident regions —
+ * init the region slots, barrier, CAS-insert that wave's staged rows
+ * (partition-local linear probe), barrier, next wave. */
+__global__ __launch_bounds__(256) void k_part_insert(
+    const int64_t* stage, const unsigned long long* cursor,
+    int64_t cap_stage_rows, int32_t r_words, int64_t* keys, uint8_t* tags,
+    direct_payloads dp, int64_t cap_p, int32_t P, int32_t K,
+    unsigned long long* bar, unsigned long long* inserted,
+    unsigned long long* overflow)
+{
+    const int64_t lmask = cap_p - 1;
+    const int64_t stride = (int64_t)gridDim.x * 256;
+    int64_t my_ins = 0, my_ovf = 0;
+    unsigned long long target = 0;
+    for (int32_t w0 = 0; w0 < P; w0 += K) {
+        const int32_t kmax = w0 + K <= P ? K : P - w0;
+        const int64_t tot = (int64_t)kmax * cap_p;
+        const int64_t wave_base = (int64_t)w0 * cap_p;
+        for (int64_t idx = (int64_t)blockIdx.x * 256 + threadIdx.x;
+             idx < tot; idx += stride) {
+            keys[wave_base + idx] = TBL_EMPTY;
+            if (tags) tags[wave_base + idx] = 0;
+        }
+        target += gridDim.x;
+        d_grid_barrier(bar, target);
+        for (int32_t k = 0; k < kmax; k++) {
+            const int32_t p = w0 + k;
+            int64_t n_p = (int64_t)cursor[p];
+            if (n_p > cap_stage_rows) n_p = cap_stage_rows;
+            const int64_t base_slot = (int64_t)p * cap_p;
+            const int64_t* prows = stage + (int64_t)p * cap_stage_rows *
+                                              r_words;
+            for (int64_t r = (int64_t)blockIdx.x * 256 + threadIdx.x;
+                 r < n_p; r += stride) {
+                const int64_t* rec = prows + r * r_words;
+                const int64_t key = rec[0];
+                uint64_t h = pg_murmur3_finalize(pg_bigint_hash(key));
+                int64_t off = (int64_t)(h & (uint64_t)lmask);
+                int64_t tries = 0;
+                for (;;) {
+                    int64_t s = base_slot + off;
+                    int64_t old =
+                        atomicCAS((unsigned long long*)&keys[s],
+                                  (unsigned long long)TBL_EMPTY,
+                                  (unsigned long long)key);
+                    if (old == TBL_EMPTY || old == key) {
+                        if (old == TBL_EMPTY) {
+                            my_ins++;
+                            if (tags) tags[s] = d_tbl_tag(h);
+                        }
+                        for (int o = 0; o < dp.n; o++) {
+                            int64_t w = rec[1 + o];
+                            switch (dp.tag[o]) {
+                                case PG_T_U8:
+                                    ((uint8_t*)dp.ptr[o])[s] = (uint8_t)w;
+                                    break;
+                                case PG_T_I32:
+                                    ((int32_t*)dp.ptr[o])[s] = (int32_t)w;
+                                    break;
+                                default: /* I64 and F64 bits */
+                                    ((int64_t*)dp.ptr[o])[s] = w;
+                            }
+                        }
+                        break;
+                    }
+                    if (++tries > lmask) {
+                        my_ovf++;
+                        break;
+                    }
+                    off = (off + 1) & lmask;
+                }
+            }
+        }
+        target += gridDim.x;
+        d_grid_barrier(bar, target);
+    }
+    my_ins = d_bfly_i64(my_ins);
+    my_ovf = d_bfly_i64(my_ovf);
+    if ((threadIdx.x & 63) == 0) {
+        if (my_ins) atomicAdd(inserted, (unsigned long long)my_ins);
+        if (my_ovf) atomicAdd(overflow, (unsigned long long)my_ovf);
+    }
+}
+
 __device__ inline int64_t d_tbl_find(const int64_t* keys, int64_t mask,
-                                     int64_t key)
+                                     int64_t lmask, int64_t key)
 {
     uint64_t h = pg_murmur3_finalize(pg_bigint_hash(key));
     int64_t s = (int64_t)(h & (uint64_t)mask);
@@ -1233,16 +1496,17 @@ __device__ inline int64_t d_tbl_find(const int64_t* keys, int64_t mask,
         int64_t k = keys[s];
         if (k == key) return s;
         if (k == TBL_EMPTY) return -1;
-        s = (s + 1) & mask;
+        s = d_probe_next(s, lmask);
     }
 }
 
 
 __device__ inline int64_t d_tbl_find_tagged(const int64_t* keys,
                                             const uint8_t* tags,
-                                            int64_t mask, int64_t key)
+                                            int64_t mask, int64_t lmask,
+                                            int64_t key)
 {
-    if (!tags) return d_tbl_find(keys, mask, key);
+    if (!tags) return d_tbl_find(keys, mask, lmask, key);
     uint64_t h = pg_murmur3_finalize(pg_bigint_hash(key));
     int64_t s = (int64_t)(h & (uint64_t)mask);
     uint8_t tag = d_tbl_tag(h);
@@ -1250,7 +1514,7 @@ __device__ inline int64_t d_tbl_find_tagged(const int64_t* keys,
         uint8_t t = tags[s];
         if (t == 0) return -1;
         if (t == tag && keys[s] == key) return s;
-        s = (s + 1) & mask;
+        s = d_probe_next(s, lmask);
     }
 }
 
@@ -1282,7 +1546,7 @@ __device__ inline void d_atomic_add_dec_ck(unsigned long long* slot,
  * so atomics preserve bit-determinism) */
 __global__ __launch_bounds__(256) void k_probe_agg(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
-    const uint8_t* tags, int64_t mask, slot_acc* acc,
+    const uint8_t* tags, int64_t mask, int64_t lmask, slot_acc* acc,
     unsigned long long* ovf)
 {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -1290,7 +1554,7 @@ __global__ __launch_bounds__(256) void k_probe_agg(
     for (; i < pg.n_rows; i += stride) {
         if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
         int64_t key = d_load_i64(pg.cols[plan.key_col], i);
-        int64_t s = d_tbl_find_tagged(keys, tags, mask, key);
+        int64_t s = d_tbl_find_tagged(keys, tags, mask, lmask, key);
         if (s < 0) continue;
         pg_agg ag;
         ag.proj = plan.proj;
@@ -1321,7 +1585,7 @@ __global__ __launch_bounds__(256) void k_probe_agg_q3(
     const int32_t* sd /* nullable pred col */, int32_t pred_op,
     int32_t pred_val, const int64_t* okey, const double* ep,
     const double* dc, int64_t n, const int64_t* keys, const uint8_t* tags,
-    int64_t mask, slot_acc* acc, unsigned long long* ovf)
+    int64_t mask, int64_t lmask, slot_acc* acc, unsigned long long* ovf)
 {
     typedef double vd2 __attribute__((ext_vector_type(2)));
     typedef int vi2 __attribute__((ext_vector_type(2)));
@@ -1371,7 +1635,7 @@ __global__ __launch_bounds__(256) void k_probe_agg_q3(
                     if (k == k0) { slot0 = p0; break; }
                     if (k == TBL_EMPTY) break;
                 }
-                p0 = (p0 + 1) & mask;
+                p0 = d_probe_next(p0, lmask);
             }
         }
         if (sel1) {
@@ -1388,7 +1652,7 @@ __global__ __launch_bounds__(256) void k_probe_agg_q3(
                     if (k == k1) { slot1 = p1; break; }
                     if (k == TBL_EMPTY) break;
                 }
-                p1 = (p1 + 1) & mask;
+                p1 = d_probe_next(p1, lmask);
             }
         }
 #pragma unroll
@@ -1417,20 +1681,21 @@ __global__ __launch_bounds__(256) void k_probe_agg_q3(
 __global__ __launch_bounds__(256) void k_probe_agg_pay(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys1,
     const uint8_t* tags1, const int32_t* head1, int64_t mask1,
-    const int64_t* pay1, const int64_t* keys2, const uint8_t* tags2,
-    int64_t mask2, slot_acc* acc2, unsigned long long* ovf)
+    int64_t lmask1, const int64_t* pay1, const int64_t* keys2,
+    const uint8_t* tags2, int64_t mask2, int64_t lmask2, slot_acc* acc2,
+    unsigned long long* ovf)
 {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; i < pg.n_rows; i += stride) {
         if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
         int64_t key = d_load_i64(pg.cols[plan.key_col], i);
-        int64_t s1 = d_tbl_find_tagged(keys1, tags1, mask1, key);
+        int64_t s1 = d_tbl_find_tagged(keys1, tags1, mask1, lmask1, key);
         if (s1 < 0) continue;
         int64_t r = head1 ? (int64_t)head1[s1] : s1;
         if (r < 0) continue;
         int64_t g = pay1[r];
-        int64_t s2 = d_tbl_find_tagged(keys2, tags2, mask2, g);
+        int64_t s2 = d_tbl_find_tagged(keys2, tags2, mask2, lmask2, g);
         if (s2 < 0) continue;
         pg_agg ag;
         ag.proj = plan.proj;
@@ -1468,7 +1733,7 @@ __global__ __launch_bounds__(256) void k_acc_min_init(slot_acc* acc,
 __global__ __launch_bounds__(256) void k_probe_count(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
     const uint8_t* tags, const int32_t* head, const int32_t* next,
-    int64_t mask, int64_t chunk, int64_t* block_counts)
+    int64_t mask, int64_t lmask, int64_t chunk, int64_t* block_counts)
 {
     const int64_t n = pg.n_rows;
     const int64_t lo = (int64_t)blockIdx.x * chunk;
@@ -1477,7 +1742,7 @@ __global__ __launch_bounds__(256) void k_probe_count(
     for (int64_t i = lo + threadIdx.x; i < hi; i += 256) {
         if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
         int64_t key = d_load_i64(pg.cols[plan.key_col], i);
-        int64_t sl = d_tbl_find_tagged(keys, tags, mask, key);
+        int64_t sl = d_tbl_find_tagged(keys, tags, mask, lmask, key);
         if (sl >= 0) {
             if (head)
                 for (int32_t r = head[sl]; r >= 0; r = next[r]) total++;
@@ -1503,7 +1768,7 @@ struct build_payloads {
 __global__ __launch_bounds__(256) void k_probe_emit(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
     const uint8_t* tags, const int32_t* head, const int32_t* next,
-    int64_t mask, int64_t chunk, const int64_t* block_offs,
+    int64_t mask, int64_t lmask, int64_t chunk, const int64_t* block_offs,
     emit_outs probe_outs, build_payloads bp, emit_outs build_outs)
 {
     const int64_t n = pg.n_rows;
@@ -1520,7 +1785,7 @@ __global__ __launch_bounds__(256) void k_probe_emit(
         int32_t c = 0;
         if (i < hi && d_eval_preds(pg, plan.preds, plan.n_preds, i)) {
             int64_t key = d_load_i64(pg.cols[plan.key_col], i);
-            sl = d_tbl_find_tagged(keys, tags, mask, key);
+            sl = d_tbl_find_tagged(keys, tags, mask, lmask, key);
             if (sl >= 0) {
                 if (head)
                     for (int32_t r = head[sl]; r >= 0; r = next[r]) c++;
@@ -1587,6 +1852,7 @@ __global__ __launch_bounds__(256) void k_probe_emit(
 template <int MAXG>
 __global__ __launch_bounds__(256) void k_probe_agg_fused2(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
+    int64_t lmask_in,
     const uint8_t* tags, int64_t mask, const uint8_t* slot_payload_u8,
     const uint8_t* dense_vals, int64_t dense_n,
     unsigned long long* out_dec /* [MAXG] */,
@@ -1608,7 +1874,7 @@ __global__ __launch_bounds__(256) void k_probe_agg_fused2(
     for (; i < pg.n_rows; i += stride) {
         if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
         int64_t key = d_load_i64(pg.cols[plan.key_col], i);
-        int64_t sl = d_tbl_find_tagged(keys, tags, mask, key);
+        int64_t sl = d_tbl_find_tagged(keys, tags, mask, lmask_in, key);
         if (sl < 0) continue;
         uint8_t g1 = slot_payload_u8[sl];
         int64_t k2 = d_load_i64(pg.cols[plan.table2_key_col], i);
@@ -2129,6 +2395,7 @@ struct OutPage {
 
 struct Table {
     int64_t cap = 0, mask = 0, n_rows = 0;
+    int64_t local_mask = 0; /* cap_p-1 for partitioned tables, else mask */
     bool key_set_only = false;
     DevBuf keys, head, next, tags;
     DevBuf acc; /* slot_acc[cap], interleaved */
@@ -2652,12 +2919,32 @@ struct AggSmallOp : Op {
 };
 
 /* ---------------- HASH_BUILD ---------------- */
+/* grid for persistent kernels: exactly one 256-thread block per CU so a
+ * hand-rolled grid barrier can never strand a non-resident block */
+static int persistent_grid()
+{
+    static int nb = 0;
+    if (!nb) {
+        hipDeviceProp_t p{};
+        if (hipGetDeviceProperties(&p, 0) == hipSuccess)
+            nb = p.multiProcessorCount;
+        if (nb <= 0) nb = 64;
+        if (nb > 256) nb = 256;
+    }
+    return nb;
+}
+
 struct BuildOp : Op {
     pg_plan_hash_build plan;
     int64_t tbl = -1;
     std::unique_ptr<Table> t;
     int64_t cap_rows = 0;
-    DevBuf counters; /* direct mode: [inserted, overflow] */
+    DevBuf counters; /* direct mode: [inserted, overflow, stage_ovf] */
+    /* radix-partitioned agg build (tables too big for cache-resident
+     * random stores): staging + cursors + barrier counter */
+    bool part = false;
+    int32_t P = 1, Kact = 1, capp_bits = 0;
+    DevBuf stage, cursorb, barb;
     void init()
     {
         t.reset(new Table());
@@ -2678,9 +2965,12 @@ struct BuildOp : Op {
             /* direct mode: size the table now from the hint; payloads
              * live per slot */
             t->slot_payloads = true;
-            int64_t cap = next_pow2(plan.capacity_hint * 2 + 16);
+            /* fill <= ~0.77: smaller table halves random-store working
+             * set and probe-side line pulls vs the old x2 sizing */
+            int64_t cap = next_pow2(plan.capacity_hint * 13 / 10 + 16);
             t->cap = cap;
             t->mask = cap - 1;
+            t->local_mask = cap - 1;
             t->keys.alloc((size_t)cap * 8);
             if (cap >= (64ll << 20)) {
                 t->tags.alloc((size_t)cap);
@@ -2689,10 +2979,33 @@ struct BuildOp : Op {
             /* acc is allocated lazily by the first mode-1 probe (mode-2
              * and emit probes never touch it — at SF300 the slot_acc
              * array is 8 GB of alloc+memset otherwise) */
-            counters.alloc(16);
+            counters.alloc(24);
             counters.zero();
-            hipLaunchKernelGGL(k_tbl_init, dim3(1024), dim3(256), 0,
-                               g_stream, (int64_t*)t->keys.p, nullptr, cap);
+            /* partitioned build once the random-store working set leaves
+             * the 256 MiB L3 (see k_part_scatter/insert) */
+            part = cap >= (32ll << 20);
+            if (part) {
+                int64_t slot_bytes = 8 + (t->tags.p ? 1 : 0) +
+                                     8ll * plan.n_payload;
+                const int64_t region_target = 48ll << 20;
+                while (cap / P * slot_bytes > region_target && P < 256)
+                    P <<= 1;
+                Kact = P >= 2 ? 2 : 1;
+                int64_t cap_p = cap / P;
+                capp_bits = 0;
+                while ((1ll << capp_bits) < cap_p) capp_bits++;
+                t->local_mask = cap_p - 1;
+                int32_t r_words = 1 + plan.n_payload;
+                stage.alloc((size_t)cap * r_words * 8);
+                cursorb.alloc((size_t)P * 8);
+                cursorb.zero();
+                barb.alloc(8);
+                barb.zero();
+            } else {
+                hipLaunchKernelGGL(k_tbl_init, dim3(1024), dim3(256), 0,
+                                   g_stream, (int64_t*)t->keys.p, nullptr,
+                                   cap);
+            }
             for (int i = 0; i < plan.n_payload; i++) {
                 t->payload.emplace_back();
                 t->ptag.push_back(-1);
@@ -2794,6 +3107,28 @@ struct BuildOp : Op {
                 dp.tag[o] = lu && o == 0 ? PG_T_U8 : t->ptag[o];
                 dp.src[o] = plan.payload_col[o];
             }
+            if (part) {
+                /* phase A: scatter surviving rows into partition runs;
+                 * inserts happen at finish (k_part_insert) */
+                hot_begin();
+                hipLaunchKernelGGL(
+                    k_part_scatter, dim3(2048), dim3(256), 0, g_stream,
+                    sp.pg, plan, set_ptr(semi), set_mask_of(semi),
+                    lu && !lu->dense ? (const int64_t*)lu->keys.p
+                                     : nullptr,
+                    lu && !lu->dense ? (const uint8_t*)lu->tags.p
+                                     : nullptr,
+                    lu ? (lu->dense ? lu->cap : lu->mask) : 0,
+                    lu ? (lu->dense ? lu->cap : lu->local_mask) : 0,
+                    lu ? (const uint8_t*)lu->payload[0].p : nullptr, dp,
+                    t->mask, capp_bits, P, (int64_t*)stage.p,
+                    (unsigned long long*)cursorb.p, t->cap / P,
+                    1 + plan.n_payload,
+                    (unsigned long long*)counters.p + 2);
+                hot_end();
+                CHKV(hipStreamSynchronize(g_stream));
+                return;
+            }
             hot_begin();
             hipLaunchKernelGGL(k_tbl_insert_direct, dim3(4096), dim3(256),
                                0, g_stream, sp.pg, plan, set_ptr(semi),
@@ -2805,6 +3140,8 @@ struct BuildOp : Op {
                                    ? (const uint8_t*)lu->tags.p
                                    : nullptr,
                                lu ? (lu->dense ? lu->cap : lu->mask) : 0,
+                               lu ? (lu->dense ? lu->cap : lu->local_mask)
+                                  : 0,
                                lu ? (const uint8_t*)lu->payload[0].p
                                   : nullptr,
                                (int64_t*)t->keys.p,
@@ -2864,14 +3201,44 @@ struct BuildOp : Op {
             return;
         }
         if (plan.agg_table) {
-            unsigned long long c[2];
-            CHKV(hipMemcpy(c, counters.p, 16, hipMemcpyDeviceToHost));
+            if (part) {
+                direct_payloads dp{};
+                dp.n = plan.n_payload;
+                for (int o = 0; o < dp.n; o++) {
+                    dp.ptr[o] = t->payload[o].p;
+                    dp.tag[o] = (plan.payload_lookup_table > 0 && o == 0)
+                                    ? PG_T_U8
+                                    : t->ptag[o];
+                }
+                hot_begin();
+                hipLaunchKernelGGL(k_part_insert,
+                                   dim3(persistent_grid()), dim3(256), 0,
+                                   g_stream, (const int64_t*)stage.p,
+                                   (const unsigned long long*)cursorb.p,
+                                   t->cap / P, 1 + plan.n_payload,
+                                   (int64_t*)t->keys.p,
+                                   (uint8_t*)t->tags.p, dp, t->cap / P, P,
+                                   Kact, (unsigned long long*)barb.p,
+                                   (unsigned long long*)counters.p,
+                                   (unsigned long long*)counters.p + 1);
+                hot_end();
+                CHKV(hipStreamSynchronize(g_stream));
+                stage.free();
+                cursorb.free();
+                barb.free();
+            }
+            unsigned long long c[3];
+            CHKV(hipMemcpy(c, counters.p, 24, hipMemcpyDeviceToHost));
+            if (c[2])
+                throw std::runtime_error(
+                    "partition staging overflow (unexpected hash skew): "
+                    "raise capacity_hint");
             if (c[1])
                 throw std::runtime_error(
                     "agg table overflow: capacity_hint too small");
-            if ((int64_t)c[0] * 2 > t->cap)
+            if ((int64_t)c[0] * 100 > t->cap * 85)
                 throw std::runtime_error(
-                    "agg table fill exceeded 0.5: raise capacity_hint");
+                    "agg table fill exceeded 0.85: raise capacity_hint");
             t->n_rows = (int64_t)c[0];
             std::lock_guard<std::mutex> lk(g_mu);
             tbl = g_next_table++;
@@ -2881,6 +3248,7 @@ struct BuildOp : Op {
         int64_t cap = next_pow2(t->n_rows * 2 + 16); /* fill <= 0.5 */
         t->cap = cap;
         t->mask = cap - 1;
+        t->local_mask = cap - 1;
         t->keys.alloc((size_t)cap * 8);
         if (!t->key_set_only) {
             t->head.alloc((size_t)cap * 4);
@@ -2998,7 +3366,7 @@ struct JoinOp : Op {
             hot_begin();
             hipLaunchKernelGGL(k_probe_agg_fused2<8>, dim3(4096), dim3(256),
                                0, g_stream, sp.pg, plan,
-                               (const int64_t*)t->keys.p,
+                               (const int64_t*)t->keys.p, t->local_mask,
                                (const uint8_t*)t->tags.p, t->mask,
                                (const uint8_t*)t->payload[0].p,
                                (const uint8_t*)t2->payload[0].p, t2->cap,
@@ -3014,9 +3382,11 @@ struct JoinOp : Op {
                                (const int64_t*)t->keys.p,
                                (const uint8_t*)t->tags.p,
                                (const int32_t*)t->head.p, t->mask,
+                               t->local_mask,
                                (const int64_t*)t->payload[0].p,
                                (const int64_t*)t2->keys.p,
                                (const uint8_t*)t2->tags.p, t2->mask,
+                               t2->local_mask,
                                (slot_acc*)t2->acc.p,
                                (unsigned long long*)ovf.p);
             hot_end();
@@ -3058,7 +3428,7 @@ struct JoinOp : Op {
                     (const double*)sp.pg.cols[plan.proj.a].data,
                     (const double*)sp.pg.cols[plan.proj.b].data,
                     sp.pg.n_rows, (const int64_t*)t->keys.p,
-                    (const uint8_t*)t->tags.p, t->mask,
+                    (const uint8_t*)t->tags.p, t->mask, t->local_mask,
                     (slot_acc*)t->acc.p,
                     (unsigned long long*)ovf.p);
             } else {
@@ -3066,6 +3436,7 @@ struct JoinOp : Op {
                                    g_stream, sp.pg, plan,
                                    (const int64_t*)t->keys.p,
                                    (const uint8_t*)t->tags.p, t->mask,
+                                   t->local_mask,
                                    (slot_acc*)t->acc.p,
                                    (unsigned long long*)ovf.p);
             }
@@ -3095,7 +3466,8 @@ struct JoinOp : Op {
                            g_stream, sp.pg, plan, (const int64_t*)t->keys.p,
                            (const uint8_t*)t->tags.p,
                            (const int32_t*)t->head.p,
-                           (const int32_t*)t->next.p, t->mask, chunk,
+                           (const int32_t*)t->next.p, t->mask,
+                           t->local_mask, chunk,
                            (int64_t*)d_counts.p);
         std::vector<int64_t> h(FLT_NB);
         CHKV(hipMemcpyAsync(h.data(), d_counts.p, FLT_NB * 8,
@@ -3150,7 +3522,8 @@ struct JoinOp : Op {
                            g_stream, sp.pg, plan, (const int64_t*)t->keys.p,
                            (const uint8_t*)t->tags.p,
                            (const int32_t*)t->head.p,
-                           (const int32_t*)t->next.p, t->mask, chunk,
+                           (const int32_t*)t->next.p, t->mask,
+                           t->local_mask, chunk,
                            (const int64_t*)d_offs.p, pouts, bp, bouts);
         hot_end();
         CHKV(hipStreamSynchronize(g_stream));
