@@ -1279,6 +1279,8 @@ __device__ inline int d_build_eval(
             sl2 = (k2 >= 1 && k2 <= lu_mask) ? k2 - 1 : -1;
         if (sl2 < 0) return 0;
         plv = lu_payload[sl2];
+        out->pay[0] = (int64_t)plv; /* cached even when !want_pay (the
+                                       scatter pass-1 LDS row tag) */
     }
     if (set_keys) {
         int64_t sk = d_load_i64(pg.cols[plan.semijoin_col], i);
@@ -1317,10 +1319,12 @@ __device__ inline int d_build_eval(
     return 1;
 }
 
-/* (A) fused filter + partition scatter: per-tile LDS counting sort
- * (count, one global cursor reservation per partition per tile, place),
- * so per-partition staging runs stay dense and appends cost ~P global
- * atomics per 8192-row tile instead of one per row. */
+/* (A) fused filter + partition scatter: per-tile LDS counting sort —
+ * pass 1 evaluates each row ONCE (predicates + semijoin + dimension
+ * lookup), caching {partition+1, lookup payload} per row in a 32 KB LDS
+ * tile; the cursor reservation costs ~P global atomics per 8192-row
+ * tile; pass 2 re-reads only the SELECTED rows' key/payload columns and
+ * places the records in dense per-partition runs. */
 __global__ __launch_bounds__(256) void k_part_scatter(
     pg_page pg, pg_plan_hash_build plan, const int64_t* set_keys,
     int64_t set_mask, const int64_t* lu_keys, const uint8_t* lu_tags,
@@ -1332,6 +1336,9 @@ __global__ __launch_bounds__(256) void k_part_scatter(
     __shared__ unsigned int cnt[SCAT_MAXP];
     __shared__ unsigned int c2[SCAT_MAXP];
     __shared__ unsigned long long gbase[SCAT_MAXP];
+    /* per-row eval cache: low 16 bits = partition+1 (0 = dropped),
+     * high 16 = the u8 lookup payload */
+    __shared__ unsigned int rowp[SCAT_TILE];
     const int64_t n = pg.n_rows;
     for (int64_t t0 = (int64_t)blockIdx.x * SCAT_TILE; t0 < n;
          t0 += (int64_t)gridDim.x * SCAT_TILE) {
@@ -1343,13 +1350,18 @@ __global__ __launch_bounds__(256) void k_part_scatter(
         __syncthreads();
         for (int64_t i = t0 + threadIdx.x; i < t1; i += 256) {
             build_row r;
-            if (!d_build_eval(pg, plan, set_keys, set_mask, lu_keys,
-                              lu_tags, lu_mask, lu_lmask, lu_payload, dp,
-                              i, false, &r))
-                continue;
-            uint64_t h = pg_murmur3_finalize(pg_bigint_hash(r.key));
-            int p = (int)((h & (uint64_t)mask) >> capp_bits);
-            atomicAdd(&cnt[p], 1u);
+            r.pay[0] = 0;
+            unsigned int tagw = 0;
+            if (d_build_eval(pg, plan, set_keys, set_mask, lu_keys,
+                             lu_tags, lu_mask, lu_lmask, lu_payload, dp,
+                             i, false, &r)) {
+                uint64_t h = pg_murmur3_finalize(pg_bigint_hash(r.key));
+                int p = (int)((h & (uint64_t)mask) >> capp_bits);
+                atomicAdd(&cnt[p], 1u);
+                tagw = (unsigned int)(p + 1) |
+                       ((unsigned int)(uint8_t)r.pay[0] << 16);
+            }
+            rowp[i - t0] = tagw;
         }
         __syncthreads();
         for (int p = threadIdx.x; p < P; p += 256) {
@@ -1363,19 +1375,27 @@ __global__ __launch_bounds__(256) void k_part_scatter(
         }
         __syncthreads();
         for (int64_t i = t0 + threadIdx.x; i < t1; i += 256) {
-            build_row r;
-            if (!d_build_eval(pg, plan, set_keys, set_mask, lu_keys,
-                              lu_tags, lu_mask, lu_lmask, lu_payload, dp,
-                              i, true, &r))
-                continue;
-            uint64_t h = pg_murmur3_finalize(pg_bigint_hash(r.key));
-            int p = (int)((h & (uint64_t)mask) >> capp_bits);
+            unsigned int tagw = rowp[i - t0];
+            if (!tagw) continue;
+            int p = (int)(tagw & 0xffffu) - 1;
             unsigned long long row = gbase[p] + atomicAdd(&c2[p], 1u);
             if (row >= (unsigned long long)cap_stage_rows) continue;
             int64_t* rec = stage +
                 ((int64_t)p * cap_stage_rows + (int64_t)row) * r_words;
-            rec[0] = r.key;
-            for (int o = 0; o < dp.n; o++) rec[1 + o] = r.pay[o];
+            rec[0] = d_load_i64(pg.cols[plan.key_col], i);
+            const bool lu = lu_payload != nullptr;
+            if (lu && dp.n >= 1) rec[1] = (int64_t)(tagw >> 16);
+            for (int o = lu ? 1 : 0; o < dp.n; o++) {
+                const pg_col& c = pg.cols[dp.src[o]];
+                if (dp.tag[o] == PG_T_F64) {
+                    double v = d_load_f64(c, i);
+                    int64_t b;
+                    memcpy(&b, &v, 8);
+                    rec[1 + o] = b;
+                } else {
+                    rec[1 + o] = d_load_i64(c, i);
+                }
+            }
         }
         __syncthreads();
     }
@@ -2923,13 +2943,20 @@ struct AggSmallOp : Op {
  * hand-rolled grid barrier can never strand a non-resident block */
 static int persistent_grid()
 {
+    /* 4 blocks of 256 threads per CU: enough latency hiding for the CAS
+     * chains, still guaranteed-resident for the grid barrier (8/CU is
+     * the hardware cap at <=80 sgprs; 4 leaves margin) */
     static int nb = 0;
     if (!nb) {
         hipDeviceProp_t p{};
+        int cus = 0;
         if (hipGetDeviceProperties(&p, 0) == hipSuccess)
-            nb = p.multiProcessorCount;
-        if (nb <= 0) nb = 64;
-        if (nb > 256) nb = 256;
+            cus = p.multiProcessorCount;
+        if (cus <= 0) cus = 64;
+        if (cus > 256) cus = 256;
+        nb = cus * 4;
+        const char* e = getenv("PG_PART_NB");
+        if (e && atoi(e) > 0) nb = atoi(e);
     }
     return nb;
 }
@@ -2972,25 +2999,38 @@ struct BuildOp : Op {
             t->mask = cap - 1;
             t->local_mask = cap - 1;
             t->keys.alloc((size_t)cap * 8);
-            if (cap >= (64ll << 20)) {
+            /* partitioned build once the random-store working set leaves
+             * the 256 MiB L3 (DESIGN.md; see k_part_scatter/insert) */
+            part = cap >= (32ll << 20);
+            /* byte tags reject probe misses from a cap-sized L3-resident
+             * array (8x denser than the key lines).  For partitioned
+             * builds the tag store lands in the L3-resident region wave
+             * (nearly free), so enable them there unconditionally. */
+            if (part || cap >= (64ll << 20)) {
                 t->tags.alloc((size_t)cap);
-                t->tags.zero();
+                if (!part) t->tags.zero(); /* part: k_part_insert inits */
             }
             /* acc is allocated lazily by the first mode-1 probe (mode-2
              * and emit probes never touch it — at SF300 the slot_acc
              * array is 8 GB of alloc+memset otherwise) */
             counters.alloc(24);
             counters.zero();
-            /* partitioned build once the random-store working set leaves
-             * the 256 MiB L3 (DESIGN.md; see k_part_scatter/insert) */
-            part = cap >= (32ll << 20);
             if (part) {
                 int64_t slot_bytes = 8 + (t->tags.p ? 1 : 0) +
                                      8ll * plan.n_payload;
                 const int64_t region_target = 48ll << 20;
                 while (cap / P * slot_bytes > region_target && P < 256)
                     P <<= 1;
-                Kact = P >= 2 ? 2 : 1;
+                /* K regions in flight, bounded by ~half the L3 */
+                int64_t region_bytes = cap / P * slot_bytes;
+                Kact = (int32_t)((128ll << 20) / region_bytes);
+                if (Kact < 1) Kact = 1;
+                if (Kact > P) Kact = P;
+                const char* ek = getenv("PG_PART_K"); /* tuning knob */
+                if (ek && atoi(ek) > 0) {
+                    Kact = atoi(ek);
+                    if (Kact > P) Kact = P;
+                }
                 int64_t cap_p = cap / P;
                 capp_bits = 0;
                 while ((1ll << capp_bits) < cap_p) capp_bits++;
