@@ -257,6 +257,15 @@ typedef struct {
                                    rejected.  capacity_hint must be >= the
                                    number of inserted rows (errors out
                                    otherwise). */
+    /* agg_table only: store the single payload in the low pack_bits of
+     * the key slot word (slot = key << pack_bits | payload) so one CAS
+     * carries key AND payload — one random line per insert and per probe
+     * hit instead of two.  Caller guarantees 0 <= key < 2^(63-pack_bits)
+     * and 0 <= payload < 2^pack_bits (checked; insert errors out
+     * otherwise).  The analog of the reference's SyntheticAddress
+     * packing (SyntheticAddress.java:23-36), applied to slot payloads.
+     * 0 = unpacked.  Requires n_payload == 1. */
+    int32_t pack_bits;
 } pg_plan_hash_build;
 
 typedef struct {

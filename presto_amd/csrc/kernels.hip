@@ -1123,7 +1123,7 @@ __device__ inline int64_t d_probe_next(int64_t s, int64_t lmask)
 __device__ inline int64_t d_tbl_find_tagged(const int64_t* keys,
                                             const uint8_t* tags,
                                             int64_t mask, int64_t lmask,
-                                            int64_t key);
+                                            int32_t pbits, int64_t key);
 
 /* direct insert for agg tables: filter+semijoin+insert in ONE scan of the
  * build input; payloads stored per SLOT (keys unique).  count tracks
@@ -1139,11 +1139,13 @@ __global__ __launch_bounds__(256) void k_tbl_insert_direct(
     int64_t set_mask, const int64_t* lu_keys, const uint8_t* lu_tags,
     int64_t lu_mask, int64_t lu_lmask, const uint8_t* lu_payload,
     int64_t* keys, uint8_t* tags, direct_payloads dp, int64_t mask,
-    unsigned long long* inserted, unsigned long long* overflow)
+    unsigned long long* inserted, unsigned long long* overflow,
+    unsigned long long* pack_err)
 {
+    const int32_t pbits = plan.pack_bits;
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    int64_t my_inserted = 0, my_overflow = 0;
+    int64_t my_inserted = 0, my_overflow = 0, my_packerr = 0;
     for (; i < pg.n_rows; i += stride) {
         if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
         uint8_t plv = 0;
@@ -1156,7 +1158,7 @@ __global__ __launch_bounds__(256) void k_tbl_insert_direct(
             int64_t sl2;
             if (lu_keys) {
                 sl2 = d_tbl_find_tagged(lu_keys, lu_tags, lu_mask,
-                                        lu_lmask, k2);
+                                        lu_lmask, 0, k2);
             } else { /* dense: payload[key-1], lu_mask = capacity */
                 sl2 = (k2 >= 1 && k2 <= lu_mask) ? k2 - 1 : -1;
             }
@@ -1183,18 +1185,32 @@ __global__ __launch_bounds__(256) void k_tbl_insert_direct(
             if (!found) continue;
         }
         int64_t key = d_load_i64(pg.cols[plan.key_col], i);
+        int64_t word = key;
+        if (pbits) {
+            /* packed slot: key<<pbits | payload0 (single payload) */
+            int64_t pay0 = lu_payload
+                               ? (int64_t)plv
+                               : d_load_i64(pg.cols[dp.src[0]], i);
+            if (key < 0 || key >= (1ll << (63 - pbits)) || pay0 < 0 ||
+                pay0 >= (1ll << pbits)) {
+                my_packerr++;
+                continue;
+            }
+            word = (key << pbits) | pay0;
+        }
         uint64_t h = pg_murmur3_finalize(pg_bigint_hash(key));
         int64_t s = (int64_t)(h & (uint64_t)mask);
         int64_t tries = 0;
         for (;;) {
             int64_t old = atomicCAS((unsigned long long*)&keys[s],
                                     (unsigned long long)TBL_EMPTY,
-                                    (unsigned long long)key);
-            if (old == TBL_EMPTY || old == key) {
+                                    (unsigned long long)word);
+            if (old == TBL_EMPTY || old == word) {
                 if (old == TBL_EMPTY) {
                     my_inserted++;
                     if (tags) tags[s] = d_tbl_tag(h);
                 }
+                if (pbits) break; /* payload lives in the slot word */
                 if (lu_payload && dp.n >= 1)
                     ((uint8_t*)dp.ptr[0])[s] = plv;
                 for (int o = lu_payload ? 1 : 0; o < dp.n; o++) {
@@ -1228,9 +1244,11 @@ __global__ __launch_bounds__(256) void k_tbl_insert_direct(
      * row serializes the whole grid) */
     my_inserted = d_bfly_i64(my_inserted);
     my_overflow = d_bfly_i64(my_overflow);
+    my_packerr = d_bfly_i64(my_packerr);
     if ((threadIdx.x & 63) == 0) {
         if (my_inserted) atomicAdd(inserted, (unsigned long long)my_inserted);
         if (my_overflow) atomicAdd(overflow, (unsigned long long)my_overflow);
+        if (my_packerr) atomicAdd(pack_err, (unsigned long long)my_packerr);
     }
 }
 
@@ -1274,7 +1292,7 @@ __device__ inline int d_build_eval(
         int64_t sl2;
         if (lu_keys)
             sl2 = d_tbl_find_tagged(lu_keys, lu_tags, lu_mask, lu_lmask,
-                                    k2);
+                                    0, k2);
         else /* dense: payload[key-1], lu_mask = capacity */
             sl2 = (k2 >= 1 && k2 <= lu_mask) ? k2 - 1 : -1;
         if (sl2 < 0) return 0;
@@ -1430,9 +1448,11 @@ __global__ __launch_bounds__(256) void k_part_insert(
     const int64_t* stage, const unsigned long long* cursor,
     int64_t cap_stage_rows, int32_t r_words, int64_t* keys, uint8_t* tags,
     direct_payloads dp, int64_t cap_p, int32_t P, int32_t K,
-    int32_t use_barrier, unsigned long long* bar,
-    unsigned long long* inserted, unsigned long long* overflow)
+    int32_t use_barrier, int32_t pbits, unsigned long long* bar,
+    unsigned long long* inserted, unsigned long long* overflow,
+    unsigned long long* pack_err)
 {
+    int64_t my_packerr = 0;
     /* use_barrier=1: init K L3-resident regions in-kernel between grid
      * barriers (strict cache blocking; grid must be all-resident).
      * use_barrier=0 (default): table pre-initialized by k_tbl_init; the
@@ -1467,6 +1487,16 @@ __global__ __launch_bounds__(256) void k_part_insert(
                  r < n_p; r += stride) {
                 const int64_t* rec = prows + r * r_words;
                 const int64_t key = rec[0];
+                int64_t word = key;
+                if (pbits) {
+                    int64_t pay0 = rec[1];
+                    if (key < 0 || key >= (1ll << (63 - pbits)) ||
+                        pay0 < 0 || pay0 >= (1ll << pbits)) {
+                        my_packerr++;
+                        continue;
+                    }
+                    word = (key << pbits) | pay0;
+                }
                 uint64_t h = pg_murmur3_finalize(pg_bigint_hash(key));
                 int64_t off = (int64_t)(h & (uint64_t)lmask);
                 int64_t tries = 0;
@@ -1475,12 +1505,13 @@ __global__ __launch_bounds__(256) void k_part_insert(
                     int64_t old =
                         atomicCAS((unsigned long long*)&keys[s],
                                   (unsigned long long)TBL_EMPTY,
-                                  (unsigned long long)key);
-                    if (old == TBL_EMPTY || old == key) {
+                                  (unsigned long long)word);
+                    if (old == TBL_EMPTY || old == word) {
                         if (old == TBL_EMPTY) {
                             my_ins++;
                             if (tags) tags[s] = d_tbl_tag(h);
                         }
+                        if (pbits) break;
                         for (int o = 0; o < dp.n; o++) {
                             int64_t w = rec[1 + o];
                             switch (dp.tag[o]) {
@@ -1511,21 +1542,27 @@ __global__ __launch_bounds__(256) void k_part_insert(
     }
     my_ins = d_bfly_i64(my_ins);
     my_ovf = d_bfly_i64(my_ovf);
+    my_packerr = d_bfly_i64(my_packerr);
     if ((threadIdx.x & 63) == 0) {
         if (my_ins) atomicAdd(inserted, (unsigned long long)my_ins);
         if (my_ovf) atomicAdd(overflow, (unsigned long long)my_ovf);
+        if (my_packerr)
+            atomicAdd(pack_err, (unsigned long long)my_packerr);
     }
 }
 
 __device__ inline int64_t d_tbl_find(const int64_t* keys, int64_t mask,
-                                     int64_t lmask, int64_t key)
+                                     int64_t lmask, int32_t pbits,
+                                     int64_t key)
 {
+    /* pbits > 0: slot word = key << pbits | payload (pg_plan_hash_build
+     * .pack_bits) — compare the key field only */
     uint64_t h = pg_murmur3_finalize(pg_bigint_hash(key));
     int64_t s = (int64_t)(h & (uint64_t)mask);
     for (;;) {
         int64_t k = keys[s];
-        if (k == key) return s;
         if (k == TBL_EMPTY) return -1;
+        if ((k >> pbits) == key) return s;
         s = d_probe_next(s, lmask);
     }
 }
@@ -1534,16 +1571,16 @@ __device__ inline int64_t d_tbl_find(const int64_t* keys, int64_t mask,
 __device__ inline int64_t d_tbl_find_tagged(const int64_t* keys,
                                             const uint8_t* tags,
                                             int64_t mask, int64_t lmask,
-                                            int64_t key)
+                                            int32_t pbits, int64_t key)
 {
-    if (!tags) return d_tbl_find(keys, mask, lmask, key);
+    if (!tags) return d_tbl_find(keys, mask, lmask, pbits, key);
     uint64_t h = pg_murmur3_finalize(pg_bigint_hash(key));
     int64_t s = (int64_t)(h & (uint64_t)mask);
     uint8_t tag = d_tbl_tag(h);
     for (;;) {
         uint8_t t = tags[s];
         if (t == 0) return -1;
-        if (t == tag && keys[s] == key) return s;
+        if (t == tag && (keys[s] >> pbits) == key) return s;
         s = d_probe_next(s, lmask);
     }
 }
@@ -1576,15 +1613,16 @@ __device__ inline void d_atomic_add_dec_ck(unsigned long long* slot,
  * so atomics preserve bit-determinism) */
 __global__ __launch_bounds__(256) void k_probe_agg(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
-    const uint8_t* tags, int64_t mask, int64_t lmask, slot_acc* acc,
-    unsigned long long* ovf)
+    const uint8_t* tags, int64_t mask, int64_t lmask, int32_t pbits,
+    slot_acc* acc, unsigned long long* ovf)
 {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; i < pg.n_rows; i += stride) {
         if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
         int64_t key = d_load_i64(pg.cols[plan.key_col], i);
-        int64_t s = d_tbl_find_tagged(keys, tags, mask, lmask, key);
+        int64_t s = d_tbl_find_tagged(keys, tags, mask, lmask, pbits,
+                                      key);
         if (s < 0) continue;
         pg_agg ag;
         ag.proj = plan.proj;
@@ -1615,7 +1653,8 @@ __global__ __launch_bounds__(256) void k_probe_agg_q3(
     const int32_t* sd /* nullable pred col */, int32_t pred_op,
     int32_t pred_val, const int64_t* okey, const double* ep,
     const double* dc, int64_t n, const int64_t* keys, const uint8_t* tags,
-    int64_t mask, int64_t lmask, slot_acc* acc, unsigned long long* ovf)
+    int64_t mask, int64_t lmask, int32_t pbits, slot_acc* acc,
+    unsigned long long* ovf)
 {
     typedef int vi2 __attribute__((ext_vector_type(2)));
     typedef long vl2 __attribute__((ext_vector_type(2)));
@@ -1656,11 +1695,14 @@ __global__ __launch_bounds__(256) void k_probe_agg_q3(
                 if (tags) {
                     uint8_t t = tags[p0];
                     if (t == 0) break;
-                    if (t == tg && keys[p0] == k0) { slot0 = p0; break; }
+                    if (t == tg && (keys[p0] >> pbits) == k0) {
+                        slot0 = p0;
+                        break;
+                    }
                 } else {
                     int64_t k = keys[p0];
-                    if (k == k0) { slot0 = p0; break; }
                     if (k == TBL_EMPTY) break;
+                    if ((k >> pbits) == k0) { slot0 = p0; break; }
                 }
                 p0 = d_probe_next(p0, lmask);
             }
@@ -1673,11 +1715,14 @@ __global__ __launch_bounds__(256) void k_probe_agg_q3(
                 if (tags) {
                     uint8_t t = tags[p1];
                     if (t == 0) break;
-                    if (t == tg && keys[p1] == k1) { slot1 = p1; break; }
+                    if (t == tg && (keys[p1] >> pbits) == k1) {
+                        slot1 = p1;
+                        break;
+                    }
                 } else {
                     int64_t k = keys[p1];
-                    if (k == k1) { slot1 = p1; break; }
                     if (k == TBL_EMPTY) break;
+                    if ((k >> pbits) == k1) { slot1 = p1; break; }
                 }
                 p1 = d_probe_next(p1, lmask);
             }
@@ -1724,12 +1769,14 @@ __global__ __launch_bounds__(256) void k_probe_agg_pay(
     for (; i < pg.n_rows; i += stride) {
         if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
         int64_t key = d_load_i64(pg.cols[plan.key_col], i);
-        int64_t s1 = d_tbl_find_tagged(keys1, tags1, mask1, lmask1, key);
+        int64_t s1 = d_tbl_find_tagged(keys1, tags1, mask1, lmask1, 0,
+                                       key);
         if (s1 < 0) continue;
         int64_t r = head1 ? (int64_t)head1[s1] : s1;
         if (r < 0) continue;
         int64_t g = pay1[r];
-        int64_t s2 = d_tbl_find_tagged(keys2, tags2, mask2, lmask2, g);
+        int64_t s2 = d_tbl_find_tagged(keys2, tags2, mask2, lmask2, 0,
+                                       g);
         if (s2 < 0) continue;
         pg_agg ag;
         ag.proj = plan.proj;
@@ -1776,7 +1823,7 @@ __global__ __launch_bounds__(256) void k_probe_count(
     for (int64_t i = lo + threadIdx.x; i < hi; i += 256) {
         if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
         int64_t key = d_load_i64(pg.cols[plan.key_col], i);
-        int64_t sl = d_tbl_find_tagged(keys, tags, mask, lmask, key);
+        int64_t sl = d_tbl_find_tagged(keys, tags, mask, lmask, 0, key);
         if (sl >= 0) {
             if (head)
                 for (int32_t r = head[sl]; r >= 0; r = next[r]) total++;
@@ -1798,6 +1845,8 @@ struct build_payloads {
     int32_t tag[4];
     int32_t n;
     int32_t by_slot; /* 1: payload arrays indexed by slot, not build row */
+    int32_t pack_bits; /* >0: payload[0] = keys[slot] & (2^bits-1) and the
+                          emitted key = keys[slot] >> bits */
 };
 __global__ __launch_bounds__(256) void k_probe_emit(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
@@ -1819,7 +1868,7 @@ __global__ __launch_bounds__(256) void k_probe_emit(
         int32_t c = 0;
         if (i < hi && d_eval_preds(pg, plan.preds, plan.n_preds, i)) {
             int64_t key = d_load_i64(pg.cols[plan.key_col], i);
-            sl = d_tbl_find_tagged(keys, tags, mask, lmask, key);
+            sl = d_tbl_find_tagged(keys, tags, mask, lmask, 0, key);
             if (sl >= 0) {
                 if (head)
                     for (int32_t r = head[sl]; r >= 0; r = next[r]) c++;
@@ -1886,7 +1935,7 @@ __global__ __launch_bounds__(256) void k_probe_emit(
 template <int MAXG>
 __global__ __launch_bounds__(256) void k_probe_agg_fused2(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
-    int64_t lmask_in,
+    int64_t lmask_in, int32_t pbits,
     const uint8_t* tags, int64_t mask, const uint8_t* slot_payload_u8,
     const uint8_t* dense_vals, int64_t dense_n,
     unsigned long long* out_dec /* [MAXG] */,
@@ -1908,9 +1957,11 @@ __global__ __launch_bounds__(256) void k_probe_agg_fused2(
     for (; i < pg.n_rows; i += stride) {
         if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
         int64_t key = d_load_i64(pg.cols[plan.key_col], i);
-        int64_t sl = d_tbl_find_tagged(keys, tags, mask, lmask_in, key);
+        int64_t sl = d_tbl_find_tagged(keys, tags, mask, lmask_in, pbits,
+                                       key);
         if (sl < 0) continue;
-        uint8_t g1 = slot_payload_u8[sl];
+        uint8_t g1 = pbits ? (uint8_t)(keys[sl] & ((1ll << pbits) - 1))
+                           : slot_payload_u8[sl];
         int64_t k2 = d_load_i64(pg.cols[plan.table2_key_col], i);
         if (k2 < 1 || k2 > dense_n) continue;
         uint8_t g2 = dense_vals[k2 - 1];
@@ -2006,21 +2057,31 @@ __global__ __launch_bounds__(256) void k_groups_emit(
         for (int w = 0; w < wid; w++) woff += wcnt[w];
         if (sel) {
             int64_t pos = woff + __popcll(m & ((1ull << lane) - 1));
-            out_key[pos] = keys[i];
+            int64_t kw = keys[i];
+            out_key[pos] = bp.pack_bits ? (kw >> bp.pack_bits) : kw;
             int64_t r = bp.by_slot ? i : (int64_t)head[i];
             for (int o = 0; o < bp.n; o++) {
+                int64_t pv = bp.pack_bits && o == 0
+                                 ? (kw & ((1ll << bp.pack_bits) - 1))
+                                 : 0;
                 switch (bp.tag[o]) {
                     case PG_T_U8:
                         ((uint8_t*)payload_outs.ptr[o])[pos] =
-                            ((const uint8_t*)bp.ptr[o])[r];
+                            bp.pack_bits && o == 0
+                                ? (uint8_t)pv
+                                : ((const uint8_t*)bp.ptr[o])[r];
                         break;
                     case PG_T_I32:
                         ((int32_t*)payload_outs.ptr[o])[pos] =
-                            ((const int32_t*)bp.ptr[o])[r];
+                            bp.pack_bits && o == 0
+                                ? (int32_t)pv
+                                : ((const int32_t*)bp.ptr[o])[r];
                         break;
                     case PG_T_I64:
                         ((int64_t*)payload_outs.ptr[o])[pos] =
-                            ((const int64_t*)bp.ptr[o])[r];
+                            bp.pack_bits && o == 0
+                                ? pv
+                                : ((const int64_t*)bp.ptr[o])[r];
                         break;
                     default:
                         ((double*)payload_outs.ptr[o])[pos] =
@@ -2430,6 +2491,7 @@ struct OutPage {
 struct Table {
     int64_t cap = 0, mask = 0, n_rows = 0;
     int64_t local_mask = 0; /* cap_p-1 for partitioned tables, else mask */
+    int32_t pack_bits = 0;  /* slot word = key << pack_bits | payload0 */
     bool key_set_only = false;
     DevBuf keys, head, next, tags;
     DevBuf acc; /* slot_acc[cap], interleaved */
@@ -3006,6 +3068,14 @@ struct BuildOp : Op {
             /* direct mode: size the table now from the hint; payloads
              * live per slot */
             t->slot_payloads = true;
+            if (plan.pack_bits) {
+                if (plan.n_payload != 1 || plan.pack_bits < 1 ||
+                    plan.pack_bits > 32)
+                    throw std::runtime_error(
+                        "pack_bits needs exactly one payload and "
+                        "1..32 bits");
+                t->pack_bits = plan.pack_bits;
+            }
             /* fill <= ~0.5 by default: linear-probe cluster length is
              * what the 324M-probe miss path pays for (measured: x1.3
              * sizing cost ~0.7 ms of Q3 probe); override via PG_CAP_X10 */
@@ -3019,7 +3089,11 @@ struct BuildOp : Op {
             t->keys.alloc((size_t)cap * 8);
             /* partitioned build once the random-store working set leaves
              * the 256 MiB L3 (see k_part_scatter/insert) */
-            part = cap >= (32ll << 20);
+            /* partitioned build only once the table is far beyond the
+             * 256 MiB L3 (measured: at ~0.5 GB the direct random insert
+             * is op-rate-bound, not locality-bound, and partitioning
+             * loses; at multi-GB tables the radix path wins) */
+            part = cap >= (128ll << 20);
             /* byte tags reject probe misses from a cap-sized L3-resident
              * array (8x denser than the key lines).  For partitioned
              * builds the tag store lands in the L3-resident region wave
@@ -3031,7 +3105,7 @@ struct BuildOp : Op {
             /* acc is allocated lazily by the first mode-1 probe (mode-2
              * and emit probes never touch it — at SF300 the slot_acc
              * array is 8 GB of alloc+memset otherwise) */
-            counters.alloc(24);
+            counters.alloc(32);
             counters.zero();
             if (part) {
                 /* P=256 keeps the scatter's per-row LDS counter atomics
@@ -3068,7 +3142,10 @@ struct BuildOp : Op {
             for (int i = 0; i < plan.n_payload; i++) {
                 t->payload.emplace_back();
                 t->ptag.push_back(-1);
-                t->payload.back().alloc((size_t)cap * 8);
+                /* packed payloads live in the key word — the array is a
+                 * placeholder so output plumbing keeps one column */
+                t->payload.back().alloc(
+                    plan.pack_bits ? 1 : (size_t)cap * 8);
             }
             return;
         }
@@ -3206,7 +3283,8 @@ struct BuildOp : Op {
                                (int64_t*)t->keys.p,
                                (uint8_t*)t->tags.p, dp, t->mask,
                                (unsigned long long*)counters.p,
-                               (unsigned long long*)counters.p + 1);
+                               (unsigned long long*)counters.p + 1,
+                               (unsigned long long*)counters.p + 3);
             hot_end();
             CHKV(hipStreamSynchronize(g_stream));
             return;
@@ -3291,18 +3369,23 @@ struct BuildOp : Op {
                                    t->cap / P, 1 + plan.n_payload,
                                    (int64_t*)t->keys.p,
                                    (uint8_t*)t->tags.p, dp, t->cap / P, P,
-                                   Kact, use_bar,
+                                   Kact, use_bar, plan.pack_bits,
                                    (unsigned long long*)barb.p,
                                    (unsigned long long*)counters.p,
-                                   (unsigned long long*)counters.p + 1);
+                                   (unsigned long long*)counters.p + 1,
+                                   (unsigned long long*)counters.p + 3);
                 hot_end();
                 CHKV(hipStreamSynchronize(g_stream));
                 stage.free();
                 cursorb.free();
                 barb.free();
             }
-            unsigned long long c[3];
-            CHKV(hipMemcpy(c, counters.p, 24, hipMemcpyDeviceToHost));
+            unsigned long long c[4];
+            CHKV(hipMemcpy(c, counters.p, 32, hipMemcpyDeviceToHost));
+            if (c[3])
+                throw std::runtime_error(
+                    "pack_bits violated: key or payload outside the "
+                    "declared bit ranges");
             if (c[2])
                 throw std::runtime_error(
                     "partition staging overflow (unexpected hash skew): "
@@ -3379,6 +3462,10 @@ struct JoinOp : Op {
         if (t->key_set_only)
             throw std::runtime_error(
                 "cannot probe a key-set-only table");
+        if (t->pack_bits && plan.mode != 1 && plan.mode != 2)
+            throw std::runtime_error(
+                "packed agg tables support fused-agg probes only "
+                "(modes 1/2)");
         /* mode 0 emit over a slot-payload table: unique keys, payloads
          * indexed by slot (no chains) */
         if (plan.mode == 1 || plan.mode == 3) {
@@ -3424,6 +3511,9 @@ struct JoinOp : Op {
                 throw std::runtime_error(
                     "mode 3 needs an agg/keyed table2 for the groups");
             t2 = it2->second.get();
+            if (t2->pack_bits)
+                throw std::runtime_error(
+                    "mode 3 groups table cannot be packed");
             if (!t2->acc.p) {
                 t2->acc.alloc((size_t)t2->cap * sizeof(slot_acc));
                 t2->acc.zero();
@@ -3441,6 +3531,7 @@ struct JoinOp : Op {
             hipLaunchKernelGGL(k_probe_agg_fused2<8>, dim3(4096), dim3(256),
                                0, g_stream, sp.pg, plan,
                                (const int64_t*)t->keys.p, t->local_mask,
+                               t->pack_bits,
                                (const uint8_t*)t->tags.p, t->mask,
                                (const uint8_t*)t->payload[0].p,
                                (const uint8_t*)t2->payload[0].p, t2->cap,
@@ -3503,14 +3594,14 @@ struct JoinOp : Op {
                     (const double*)sp.pg.cols[plan.proj.b].data,
                     sp.pg.n_rows, (const int64_t*)t->keys.p,
                     (const uint8_t*)t->tags.p, t->mask, t->local_mask,
-                    (slot_acc*)t->acc.p,
+                    t->pack_bits, (slot_acc*)t->acc.p,
                     (unsigned long long*)ovf.p);
             } else {
                 hipLaunchKernelGGL(k_probe_agg, dim3(4096), dim3(256), 0,
                                    g_stream, sp.pg, plan,
                                    (const int64_t*)t->keys.p,
                                    (const uint8_t*)t->tags.p, t->mask,
-                                   t->local_mask,
+                                   t->local_mask, t->pack_bits,
                                    (slot_acc*)t->acc.p,
                                    (unsigned long long*)ovf.p);
             }
@@ -3694,6 +3785,7 @@ struct JoinOp : Op {
         build_payloads bp{};
         bp.n = (int32_t)gt->payload.size();
         bp.by_slot = gt->slot_payloads ? 1 : 0;
+        bp.pack_bits = gt->pack_bits;
         emit_outs pl_outs{};
         pl_outs.n = bp.n;
         for (int o = 0; o < bp.n; o++) {

@@ -77,7 +77,7 @@ class PlanHashBuild(C.Structure):
                 ("key_set_only", C.c_int32), ("dense_array", C.c_int32),
                 ("payload_lookup_table", C.c_int64),
                 ("payload_lookup_key_col", C.c_int32),
-                ("agg_table", C.c_int32)]
+                ("agg_table", C.c_int32), ("pack_bits", C.c_int32)]
 
 
 class PlanLookupJoin(C.Structure):

@@ -118,6 +118,8 @@ class Q3Pipeline:
         b2p.capacity_hint = max(orders.n_rows // 8, 16)
         b2p.key_set_only = 0
         b2p.agg_table = 1  # direct single-scan insert (fused-agg probe only)
+        b2p.pack_bits = 16  # slot = orderkey<<16 | orderdate: one CAS
+                            # carries key AND payload (date < 2^16)
         self.b2 = Operator(OP_HASH_BUILD, b2p)
         self.b2.add_input(orders)
         self.b2.finish()
@@ -335,6 +337,7 @@ class Q5PipelineFused:
         b2.payload_lookup_key_col = orders.channel("custkey")
         b2.capacity_hint = max(orders.n_rows // 4, 64)
         b2.agg_table = 1
+        b2.pack_bits = 8  # slot = orderkey<<8 | cust_nation (u8)
         self.b2 = Operator(OP_HASH_BUILD, b2)
         self.b2.add_input(orders)
         self.b2.finish()

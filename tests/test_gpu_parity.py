@@ -1564,7 +1564,7 @@ def test_partitioned_build_parity(P):
     pk[::2] = pk[::2] + 3_000_001
     pv = rng.randint(1, 1000, probe_n).astype(np.int64)
 
-    def run(hint):
+    def run(hint, pack_bits=0):
         bp = P.PlanHashBuild()
         bp.key_col = 0
         bp.semijoin_table = -1
@@ -1572,6 +1572,7 @@ def test_partitioned_build_parity(P):
         bp.payload_col[0] = 1
         bp.capacity_hint = hint
         bp.agg_table = 1
+        bp.pack_bits = pack_bits
         b = P.Operator(P.OP_HASH_BUILD, bp)
         b.add_input(P.Page({"k": keys, "d": dates}))
         b.finish()
@@ -1593,11 +1594,15 @@ def test_partitioned_build_parity(P):
         order = np.argsort(out["key"])
         return {nm: out[nm][order] for nm in ("key", "date", "sum", "cnt")}
 
-    direct = run(n)               # cap < 32M -> direct insert
-    partd = run(26_000_000)       # cap = 32M -> partitioned build
+    direct = run(n)               # small cap -> direct insert
+    partd = run(70_000_000)       # cap = 256M slots -> partitioned build
+    packed = run(n, pack_bits=14)  # slot word = key<<14 | date
+    packed_part = run(70_000_000, pack_bits=14)
     assert len(direct["key"]) == len(partd["key"])
     for nm in ("key", "date", "sum", "cnt"):
         assert np.array_equal(direct[nm], partd[nm]), nm
+        assert np.array_equal(direct[nm], packed[nm]), nm
+        assert np.array_equal(direct[nm], packed_part[nm]), nm
     # numpy cross-check of the grouped sums
     import collections
     exp = collections.defaultdict(int)
