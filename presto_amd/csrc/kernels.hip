@@ -1660,9 +1660,12 @@ __global__ __launch_bounds__(256) void k_probe_agg_q3(
     int64_t mask, int64_t lmask, int32_t pbits, int32_t dec_only,
     slot_acc* acc, unsigned long long* ovf)
 {
-    typedef int vi4 __attribute__((ext_vector_type(4)));
+    typedef int vi2 __attribute__((ext_vector_type(2)));
     typedef long vl2 __attribute__((ext_vector_type(2)));
-    const int Q = 4;
+    const int Q = 2; /* measured: 2-row unrolled chains beat a 4-row
+                        lockstep state machine (most probes resolve in one
+                        step, so the machinery costs more than the extra
+                        memory-level parallelism buys) */
     int64_t base0 = Q * ((int64_t)blockIdx.x * blockDim.x + threadIdx.x);
     int64_t stride = Q * (int64_t)gridDim.x * blockDim.x;
     for (int64_t base = base0; base < n; base += stride) {
@@ -1671,12 +1674,10 @@ __global__ __launch_bounds__(256) void k_probe_agg_q3(
         bool sel[Q];
         if (base + Q <= n) {
             vl2 ka = __builtin_nontemporal_load((const vl2*)(okey + base));
-            vl2 kb = __builtin_nontemporal_load(
-                (const vl2*)(okey + base + 2));
-            k[0] = ka[0]; k[1] = ka[1]; k[2] = kb[0]; k[3] = kb[1];
+            k[0] = ka[0]; k[1] = ka[1];
             if (sd) {
-                vi4 ss = __builtin_nontemporal_load((const vi4*)(sd + base));
-                s[0] = ss[0]; s[1] = ss[1]; s[2] = ss[2]; s[3] = ss[3];
+                vi2 ss = __builtin_nontemporal_load((const vi2*)(sd + base));
+                s[0] = ss[0]; s[1] = ss[1];
             }
 #pragma unroll
             for (int j = 0; j < Q; j++) sel[j] = true;
@@ -1694,50 +1695,28 @@ __global__ __launch_bounds__(256) void k_probe_agg_q3(
                 sel[j] = sel[j] && (pred_op == PG_CMP_GT ? s[j] > pred_val
                                                          : s[j] < pred_val);
         }
-        /* lockstep probe: all four chains issue their next tag/key load
-         * each iteration */
-        int64_t pos[Q], slot[Q];
-        uint8_t tg[Q];
-        bool live[Q];
-        bool any = false;
+        int64_t slot[Q];
 #pragma unroll
         for (int j = 0; j < Q; j++) {
             slot[j] = -1;
-            live[j] = sel[j];
-            if (sel[j]) {
-                uint64_t h = pg_murmur3_finalize(pg_bigint_hash(k[j]));
-                pos[j] = (int64_t)(h & (uint64_t)mask);
-                tg[j] = tags ? d_tbl_tag(h) : 0;
-                any = true;
-            }
-        }
-        while (any) {
-            any = false;
-#pragma unroll
-            for (int j = 0; j < Q; j++) {
-                if (!live[j]) continue;
+            if (!sel[j]) continue;
+            uint64_t h = pg_murmur3_finalize(pg_bigint_hash(k[j]));
+            int64_t p = (int64_t)(h & (uint64_t)mask);
+            uint8_t tg = tags ? d_tbl_tag(h) : 0;
+            for (;;) {
                 if (tags) {
-                    uint8_t t = tags[pos[j]];
-                    if (t == 0) {
-                        live[j] = false;
-                    } else if (t == tg[j] &&
-                               (keys[pos[j]] >> pbits) == k[j]) {
-                        slot[j] = pos[j];
-                        live[j] = false;
+                    uint8_t t = tags[p];
+                    if (t == 0) break;
+                    if (t == tg && (keys[p] >> pbits) == k[j]) {
+                        slot[j] = p;
+                        break;
                     }
                 } else {
-                    int64_t kw = keys[pos[j]];
-                    if (kw == TBL_EMPTY) {
-                        live[j] = false;
-                    } else if ((kw >> pbits) == k[j]) {
-                        slot[j] = pos[j];
-                        live[j] = false;
-                    }
+                    int64_t kw = keys[p];
+                    if (kw == TBL_EMPTY) break;
+                    if ((kw >> pbits) == k[j]) { slot[j] = p; break; }
                 }
-                if (live[j]) {
-                    pos[j] = d_probe_next(pos[j], lmask);
-                    any = true;
-                }
+                p = d_probe_next(p, lmask);
             }
         }
 #pragma unroll
@@ -2008,6 +1987,122 @@ __global__ __launch_bounds__(256) void k_probe_agg_fused2(
     /* wave reduce ticks/count (exact), then one atomic per wave; the
      * fx128 halves go through carry-aware atomics per lane (exact,
      * order-independent) */
+#pragma unroll
+    for (int g = 0; g < MAXG; g++) {
+        int64_t a = d_bfly_i64(acc[g]);
+        int64_t c = d_bfly_i64((int64_t)cnt[g]);
+        if ((threadIdx.x & 63) == 0) {
+            if (a) d_atomic_add_dec_ck(&out_dec[g], a, out_cnt + MAXG);
+            if (c) atomicAdd(&out_cnt[g], (unsigned long long)c);
+        }
+        if (flo[g] | fhi[g]) {
+            unsigned long long old = atomicAdd(&out_flo[g], flo[g]);
+            atomicAdd(&out_fhi[g],
+                      fhi[g] + (old > ~flo[g] ? 1ull : 0ull));
+        }
+    }
+}
+
+/* specialized mode-2 probe for the Q5 local-supplier shape (the
+ * codegen-analog of k_agg_q1/k_probe_agg_q3): PACKED orders table
+ * (slot = orderkey << pbits | cust_nation), dense suppkey->nation
+ * payload, no predicates, DISC_PRICE over two f64 columns.  Two rows
+ * per thread with nontemporal vector key loads; ep/dc are loaded only
+ * for local-supplier matches (~0.6% of rows at SF100); per-thread
+ * register group accumulators, wave butterfly, one atomic per group per
+ * wave (exact ticks + exact fx128 — deterministic). */
+template <int MAXG>
+__global__ __launch_bounds__(256) void k_probe_agg_q5(
+    pg_plan_lookup_join plan, const int64_t* okey, const int64_t* skey,
+    const double* ep, const double* dc, int64_t n, const int64_t* keys,
+    const uint8_t* tags, int64_t mask, int64_t lmask, int32_t pbits,
+    const uint8_t* dense_vals, int64_t dense_n,
+    unsigned long long* out_dec, unsigned long long* out_flo,
+    unsigned long long* out_fhi, unsigned long long* out_cnt)
+{
+    typedef long vl2 __attribute__((ext_vector_type(2)));
+    const int64_t pmask = (1ll << pbits) - 1;
+    int64_t acc[MAXG];
+    uint64_t fhi[MAXG], flo[MAXG];
+    int32_t cnt[MAXG];
+#pragma unroll
+    for (int g = 0; g < MAXG; g++) {
+        acc[g] = 0;
+        fhi[g] = 0;
+        flo[g] = 0;
+        cnt[g] = 0;
+    }
+    int64_t base0 = 2 * ((int64_t)blockIdx.x * blockDim.x + threadIdx.x);
+    int64_t stride = 2 * (int64_t)gridDim.x * blockDim.x;
+    for (int64_t base = base0; base < n; base += stride) {
+        int64_t k[2], sk[2];
+        bool sel[2];
+        if (base + 2 <= n) {
+            vl2 ka = __builtin_nontemporal_load((const vl2*)(okey + base));
+            vl2 kb = __builtin_nontemporal_load((const vl2*)(skey + base));
+            k[0] = ka[0]; k[1] = ka[1];
+            sk[0] = kb[0]; sk[1] = kb[1];
+            sel[0] = sel[1] = true;
+        } else {
+#pragma unroll
+            for (int j = 0; j < 2; j++) {
+                sel[j] = base + j < n;
+                k[j] = sel[j] ? okey[base + j] : 0;
+                sk[j] = sel[j] ? skey[base + j] : 0;
+            }
+        }
+#pragma unroll
+        for (int j = 0; j < 2; j++) {
+            if (!sel[j]) continue;
+            int64_t slot = -1;
+            uint64_t h = pg_murmur3_finalize(pg_bigint_hash(k[j]));
+            int64_t p = (int64_t)(h & (uint64_t)mask);
+            uint8_t tg = tags ? d_tbl_tag(h) : 0;
+            for (;;) {
+                if (tags) {
+                    uint8_t t = tags[p];
+                    if (t == 0) break;
+                    if (t == tg && (keys[p] >> pbits) == k[j]) {
+                        slot = p;
+                        break;
+                    }
+                } else {
+                    int64_t kw = keys[p];
+                    if (kw == TBL_EMPTY) break;
+                    if ((kw >> pbits) == k[j]) { slot = p; break; }
+                }
+                p = d_probe_next(p, lmask);
+            }
+            if (slot < 0) continue;
+            uint8_t g1 = (uint8_t)(keys[slot] & pmask);
+            if (sk[j] < 1 || sk[j] > dense_n) continue;
+            uint8_t g2 = dense_vals[sk[j] - 1];
+            if (g1 != g2) continue;
+            int gi = -1;
+#pragma unroll
+            for (int g = 0; g < MAXG; g++)
+                if (g < plan.n_group_vals && plan.group_vals[g] == g2 &&
+                    gi < 0)
+                    gi = g;
+            if (gi < 0) continue;
+            double e = ep[base + j], d = dc[base + j];
+            int64_t cents = (int64_t)(e * 100.0 + 0.5);
+            int64_t di = (int64_t)(d * 100.0 + 0.5);
+            int64_t ticks = cents * (100 - di);
+            double pr = e * (1.0 - d);
+            uint64_t phi, plo;
+            fx128_from_f64(pr, &phi, &plo);
+#pragma unroll
+            for (int g = 0; g < MAXG; g++) {
+                bool m = g == gi;
+                acc[g] += m ? ticks : 0;
+                cnt[g] += m ? 1 : 0;
+                uint64_t nlo = flo[g] + (m ? plo : 0);
+                fhi[g] += (m ? phi : 0) + (nlo < flo[g] ? 1u : 0u);
+                flo[g] = nlo;
+            }
+        }
+    }
 #pragma unroll
     for (int g = 0; g < MAXG; g++) {
         int64_t a = d_bfly_i64(acc[g]);
@@ -3542,15 +3637,44 @@ struct JoinOp : Op {
         sp.stage(in);
         if (plan.mode == 2) {
             unsigned long long* a = (unsigned long long*)m2_acc.p;
+            /* specialized fast path: packed table + no predicates +
+             * DISC_PRICE over aligned f64 columns (the Q5 shape) */
+            bool spec = plan.n_preds == 0 && t->pack_bits > 0 &&
+                        plan.proj.kind == PG_PROJ_DISC_PRICE &&
+                        sp.pg.cols[plan.key_col].tag == PG_T_I64 &&
+                        sp.pg.cols[plan.table2_key_col].tag == PG_T_I64 &&
+                        sp.pg.cols[plan.proj.a].tag == PG_T_F64 &&
+                        sp.pg.cols[plan.proj.b].tag == PG_T_F64 &&
+                        plan.dec_scale == 4 &&
+                        !((uintptr_t)sp.pg.cols[plan.key_col].data & 15) &&
+                        !((uintptr_t)sp.pg.cols[plan.table2_key_col].data &
+                          15) &&
+                        !sp.pg.cols[plan.key_col].null_mask &&
+                        !sp.pg.cols[plan.table2_key_col].null_mask &&
+                        !sp.pg.cols[plan.proj.a].null_mask &&
+                        !sp.pg.cols[plan.proj.b].null_mask;
             hot_begin();
-            hipLaunchKernelGGL(k_probe_agg_fused2<8>, dim3(4096), dim3(256),
-                               0, g_stream, sp.pg, plan,
-                               (const int64_t*)t->keys.p, t->local_mask,
-                               t->pack_bits,
-                               (const uint8_t*)t->tags.p, t->mask,
-                               (const uint8_t*)t->payload[0].p,
-                               (const uint8_t*)t2->payload[0].p, t2->cap,
-                               a, a + 8, a + 16, a + 24);
+            if (spec) {
+                hipLaunchKernelGGL(
+                    k_probe_agg_q5<8>, dim3(4096), dim3(256), 0, g_stream,
+                    plan, (const int64_t*)sp.pg.cols[plan.key_col].data,
+                    (const int64_t*)sp.pg.cols[plan.table2_key_col].data,
+                    (const double*)sp.pg.cols[plan.proj.a].data,
+                    (const double*)sp.pg.cols[plan.proj.b].data,
+                    sp.pg.n_rows, (const int64_t*)t->keys.p,
+                    (const uint8_t*)t->tags.p, t->mask, t->local_mask,
+                    t->pack_bits, (const uint8_t*)t2->payload[0].p,
+                    t2->cap, a, a + 8, a + 16, a + 24);
+            } else {
+                hipLaunchKernelGGL(k_probe_agg_fused2<8>, dim3(4096),
+                                   dim3(256), 0, g_stream, sp.pg, plan,
+                                   (const int64_t*)t->keys.p,
+                                   t->local_mask, t->pack_bits,
+                                   (const uint8_t*)t->tags.p, t->mask,
+                                   (const uint8_t*)t->payload[0].p,
+                                   (const uint8_t*)t2->payload[0].p,
+                                   t2->cap, a, a + 8, a + 16, a + 24);
+            }
             hot_end();
             CHKV(hipStreamSynchronize(g_stream));
             return;
