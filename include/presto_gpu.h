@@ -266,6 +266,12 @@ typedef struct {
      * packing (SyntheticAddress.java:23-36), applied to slot payloads.
      * 0 = unpacked.  Requires n_payload == 1. */
     int32_t pack_bits;
+    /* agg_table only: capacity multiplier x10 (0 = default 20, i.e.
+     * cap = next_pow2(2 x capacity_hint), fill <= ~0.5 — sized for
+     * short linear-probe clusters on miss-heavy probes).  Probes that
+     * ALWAYS hit (e.g. lineitem -> its order) can size tighter: 13
+     * gives fill <= ~0.77 and halves the table/accumulator footprint. */
+    int32_t fill_x10;
 } pg_plan_hash_build;
 
 typedef struct {
@@ -314,6 +320,19 @@ typedef struct {
      * grouped probe — Q2's per-part minimum supplycost).  Implies
      * dec_only. */
     int32_t dec_min;
+    /* mode 1 only: MULTI-ACCUMULATOR probe — the full
+     * InMemoryHashAggregationBuilder.processPage:204 analog (one
+     * getGroupIds probe, then EVERY aggregator's addInput), optionally
+     * with per-aggregate FILTER clauses (AggregationNode's
+     * aggregation masks).  n_aggs > 0 replaces the single proj/dec legs:
+     * per-slot accumulators are n_aggs int64 tick sums plus the match
+     * count; agg_filter[a] >= 0 indexes a predicate in preds[] that
+     * gates aggregate a alone (row-level filtering still uses
+     * preds[0..n_preds)).  Overflow-checked (Math.addExact).  Output
+     * page after finish: key, payloads..., agg0..agg{n-1}, count. */
+    int32_t n_aggs; /* 0 = legacy single proj; 1..6 multi */
+    pg_agg aggs[6];
+    int32_t agg_filter[6];
 } pg_plan_lookup_join;
 
 typedef struct {

@@ -324,6 +324,11 @@ __device__ inline int64_t d_eval_proj_dec(const pg_page& pg, const pg_agg& ag,
         for (int k = 0; k < ag.dec_scale; k++) s *= 10.0;
         return (int64_t)(a * s + 0.5);
     }
+    if (p.kind == PG_PROJ_MUL && ag.dec_scale == 0 &&
+        pg.cols[p.a].tag != PG_T_F64 && pg.cols[p.b].tag != PG_T_F64)
+        /* raw integer product (e.g. Q21's exact sum of squared
+         * suppkeys) — scale-4 money products keep the cents path */
+        return d_load_i64(pg.cols[p.a], i) * d_load_i64(pg.cols[p.b], i);
     int64_t cents = (int64_t)(d_load_f64(pg.cols[p.a], i) * 100.0 + 0.5);
     int64_t d = (int64_t)(d_load_f64(pg.cols[p.b], i) * 100.0 + 0.5);
     if (p.kind == PG_PROJ_MUL) return cents * d; /* scale 4 ticks */
@@ -2119,11 +2124,113 @@ __global__ __launch_bounds__(256) void k_probe_agg_q5(
     }
 }
 
+/* MULTI-ACCUMULATOR fused-agg probe (mode 1 with n_aggs > 0): the full
+ * InMemoryHashAggregationBuilder.processPage:204 analog — ONE getGroupIds
+ * probe per row, then every aggregator accumulates, each optionally
+ * gated by its own FILTER predicate (AggregationNode masks).  Per-slot
+ * state: n_aggs overflow-checked int64 tick sums + the match count. */
+__global__ __launch_bounds__(256) void k_probe_agg_multi(
+    pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
+    const uint8_t* tags, int64_t mask, int64_t lmask, int32_t pbits,
+    unsigned long long* acc, unsigned long long* ovf)
+{
+    const int n_aggs = plan.n_aggs;
+    const int64_t stride_w = n_aggs + 1;
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < pg.n_rows; i += stride) {
+        if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
+        int64_t key = d_load_i64(pg.cols[plan.key_col], i);
+        int64_t s = d_tbl_find_tagged(keys, tags, mask, lmask, pbits, key);
+        if (s < 0) continue;
+        unsigned long long* slot = acc + (size_t)s * stride_w;
+#pragma unroll
+        for (int a = 0; a < 6; a++) {
+            if (a >= n_aggs) break;
+            int f = plan.agg_filter[a];
+            if (f >= 0 && !d_eval_preds(pg, &plan.preds[f], 1, i))
+                continue;
+            int64_t ticks = plan.aggs[a].func == PG_AGG_COUNT
+                                ? 1
+                                : d_eval_proj_dec(pg, plan.aggs[a], i);
+            if (ticks) d_atomic_add_dec_ck(slot + a, ticks, ovf);
+        }
+        atomicAdd(slot + n_aggs, 1ull);
+    }
+}
+
+/* group extraction for the multi-accumulator layout: key (+payloads),
+ * then the n_aggs tick sums and the count, slot-ascending */
+__global__ __launch_bounds__(256) void k_groups_emit_multi(
+    const int64_t* keys, const unsigned long long* acc, int64_t stride_w,
+    int32_t n_aggs, build_payloads bp, int64_t cap, int64_t chunk,
+    const int64_t* block_offs, int64_t* out_key, emit_outs payload_outs,
+    emit_outs agg_outs)
+{
+    const int64_t lo = (int64_t)blockIdx.x * chunk;
+    const int64_t hi = min(lo + chunk, cap);
+    const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
+    __shared__ int64_t wcnt[4];
+    __shared__ int64_t running;
+    if (threadIdx.x == 0) running = block_offs[blockIdx.x];
+    __syncthreads();
+    for (int64_t base = lo; base < hi; base += 256) {
+        int64_t i = base + 64 * wid + lane;
+        bool sel = i < hi && acc[i * stride_w + n_aggs] > 0;
+        uint64_t m = d_ballot(sel);
+        int wsum = __popcll(m);
+        if (lane == 0) wcnt[wid] = wsum;
+        __syncthreads();
+        int64_t woff = running;
+        for (int w = 0; w < wid; w++) woff += wcnt[w];
+        if (sel) {
+            int64_t pos = woff + __popcll(m & ((1ull << lane) - 1));
+            int64_t kw = keys[i];
+            out_key[pos] = bp.pack_bits ? (kw >> bp.pack_bits) : kw;
+            for (int o = 0; o < bp.n; o++) {
+                int64_t pv = bp.pack_bits && o == 0
+                                 ? (kw & ((1ll << bp.pack_bits) - 1))
+                                 : 0;
+                switch (bp.tag[o]) {
+                    case PG_T_U8:
+                        ((uint8_t*)payload_outs.ptr[o])[pos] =
+                            bp.pack_bits && o == 0
+                                ? (uint8_t)pv
+                                : ((const uint8_t*)bp.ptr[o])[i];
+                        break;
+                    case PG_T_I32:
+                        ((int32_t*)payload_outs.ptr[o])[pos] =
+                            bp.pack_bits && o == 0
+                                ? (int32_t)pv
+                                : ((const int32_t*)bp.ptr[o])[i];
+                        break;
+                    case PG_T_I64:
+                        ((int64_t*)payload_outs.ptr[o])[pos] =
+                            bp.pack_bits && o == 0
+                                ? pv
+                                : ((const int64_t*)bp.ptr[o])[i];
+                        break;
+                    default:
+                        ((double*)payload_outs.ptr[o])[pos] =
+                            ((const double*)bp.ptr[o])[i];
+                }
+            }
+            for (int a = 0; a <= n_aggs; a++)
+                ((int64_t*)agg_outs.ptr[a])[pos] =
+                    (int64_t)acc[i * stride_w + a];
+        }
+        __syncthreads();
+        if (threadIdx.x == 0)
+            running += wcnt[0] + wcnt[1] + wcnt[2] + wcnt[3];
+        __syncthreads();
+    }
+}
+
 /* extract group rows after fused probe-agg: slots with count>0, emitted
  * slot-ascending (stable compaction over the slot array) */
 __global__ __launch_bounds__(256) void k_groups_count(
-    const slot_acc* acc, int64_t cap, int64_t chunk,
-    int64_t* block_counts)
+    const unsigned long long* acc, int64_t stride_w, int64_t cnt_off,
+    int64_t cap, int64_t chunk, int64_t* block_counts)
 {
     const int64_t lo = (int64_t)blockIdx.x * chunk;
     const int64_t hi = min(lo + chunk, cap);
@@ -2131,7 +2238,7 @@ __global__ __launch_bounds__(256) void k_groups_count(
     int64_t cnt = 0;
     for (int64_t base = lo + 64 * wid; base < hi; base += 256) {
         int64_t i = base + lane;
-        bool sel = i < hi && acc[i].cnt > 0;
+        bool sel = i < hi && acc[i * stride_w + cnt_off] > 0;
         uint64_t m = d_ballot(sel);
         if (lane == 0) cnt += __popcll(m);
     }
@@ -2605,6 +2712,8 @@ struct Table {
     bool key_set_only = false;
     DevBuf keys, head, next, tags;
     DevBuf acc; /* slot_acc[cap], interleaved */
+    DevBuf acc_multi; /* multi-agg probes: (n_acc+1) u64 per slot */
+    int32_t n_acc = 0;
     bool slot_payloads = false; /* payloads indexed by slot (agg tables) */
     bool dense = false;          /* dense_array: payload[key-1], no hash */
     /* compacted build-row arrays: key + payloads */
@@ -3189,7 +3298,7 @@ struct BuildOp : Op {
             /* fill <= ~0.5 by default: linear-probe cluster length is
              * what the 324M-probe miss path pays for (measured: x1.3
              * sizing cost ~0.7 ms of Q3 probe); override via PG_CAP_X10 */
-            int64_t mult10 = 20;
+            int64_t mult10 = plan.fill_x10 >= 11 ? plan.fill_x10 : 20;
             if (const char* em = getenv("PG_CAP_X10"))
                 if (atoi(em) >= 11) mult10 = atoi(em);
             int64_t cap = next_pow2(plan.capacity_hint * mult10 / 10 + 16);
@@ -3582,7 +3691,26 @@ struct JoinOp : Op {
             ovf.alloc(8);
             ovf.zero();
         }
-        if (plan.mode == 1 && !t->acc.p) {
+        if (plan.mode == 1 && plan.n_aggs > 0) {
+            if (plan.n_aggs > 6)
+                throw std::runtime_error("n_aggs must be 1..6");
+            if (!t->slot_payloads)
+                throw std::runtime_error(
+                    "multi-agg probes need an agg_table build");
+            for (int a = 0; a < plan.n_aggs; a++)
+                if (plan.agg_filter[a] >= plan.n_preds &&
+                    plan.agg_filter[a] >= PG_MAX_PRED)
+                    throw std::runtime_error("agg_filter out of range");
+            if (t->acc_multi.p && t->n_acc != plan.n_aggs)
+                throw std::runtime_error(
+                    "table already carries a different multi-agg layout");
+            if (!t->acc_multi.p) {
+                t->n_acc = plan.n_aggs;
+                t->acc_multi.alloc((size_t)t->cap * (plan.n_aggs + 1) * 8);
+                t->acc_multi.zero();
+                CHKV(hipStreamSynchronize(g_stream));
+            }
+        } else if (plan.mode == 1 && !t->acc.p) {
             t->acc.alloc((size_t)t->cap * sizeof(slot_acc));
             t->acc.zero();
             CHKV(hipStreamSynchronize(g_stream));
@@ -3692,6 +3820,19 @@ struct JoinOp : Op {
                                (const uint8_t*)t2->tags.p, t2->mask,
                                t2->local_mask,
                                (slot_acc*)t2->acc.p,
+                               (unsigned long long*)ovf.p);
+            hot_end();
+            CHKV(hipStreamSynchronize(g_stream));
+            return;
+        }
+        if (plan.mode == 1 && plan.n_aggs > 0) {
+            hot_begin();
+            hipLaunchKernelGGL(k_probe_agg_multi, dim3(4096), dim3(256), 0,
+                               g_stream, sp.pg, plan,
+                               (const int64_t*)t->keys.p,
+                               (const uint8_t*)t->tags.p, t->mask,
+                               t->local_mask, t->pack_bits,
+                               (unsigned long long*)t->acc_multi.p,
                                (unsigned long long*)ovf.p);
             hot_end();
             CHKV(hipStreamSynchronize(g_stream));
@@ -3886,15 +4027,19 @@ struct JoinOp : Op {
         /* extract groups: slots with count>0, slot-ascending (mode 3
          * groups live in table2) */
         Table* gt = plan.mode == 3 ? t2 : t;
+        const bool multi = plan.mode == 1 && plan.n_aggs > 0;
         int64_t cap = gt->cap;
         int64_t chunk = (cap + FLT_NB - 1) / FLT_NB;
         chunk = (chunk + 255) / 256 * 256;
         if (chunk < 256) chunk = 256;
         DevBuf d_counts;
         d_counts.alloc(FLT_NB * 8);
-        hipLaunchKernelGGL(k_groups_count, dim3(FLT_NB), dim3(256), 0,
-                           g_stream, (const slot_acc*)gt->acc.p,
-                           cap, chunk, (int64_t*)d_counts.p);
+        hipLaunchKernelGGL(
+            k_groups_count, dim3(FLT_NB), dim3(256), 0, g_stream,
+            multi ? (const unsigned long long*)gt->acc_multi.p
+                  : (const unsigned long long*)gt->acc.p,
+            multi ? plan.n_aggs + 1 : 4, multi ? plan.n_aggs : 3, cap,
+            chunk, (int64_t*)d_counts.p);
         std::vector<int64_t> h(FLT_NB);
         CHKV(hipMemcpyAsync(h.data(), d_counts.p, FLT_NB * 8,
                             hipMemcpyDeviceToHost, g_stream));
@@ -3933,6 +4078,26 @@ struct JoinOp : Op {
             int c = add_dev_col(gt->ptag[o]);
             pl_outs.ptr[o] = op.pg.cols[c].data;
             pl_outs.tag[o] = gt->ptag[o];
+        }
+        if (multi) {
+            emit_outs agg_outs{};
+            agg_outs.n = plan.n_aggs + 1;
+            for (int a = 0; a <= plan.n_aggs; a++) {
+                int c = add_dev_col(PG_T_I64);
+                agg_outs.ptr[a] = op.pg.cols[c].data;
+                agg_outs.tag[a] = PG_T_I64;
+            }
+            op.pg.n_cols = nc;
+            hipLaunchKernelGGL(
+                k_groups_emit_multi, dim3(FLT_NB), dim3(256), 0, g_stream,
+                (const int64_t*)gt->keys.p,
+                (const unsigned long long*)gt->acc_multi.p,
+                plan.n_aggs + 1, plan.n_aggs, bp, cap, chunk,
+                (const int64_t*)d_offs.p,
+                (int64_t*)op.pg.cols[c_key].data, pl_outs, agg_outs);
+            CHKV(hipStreamSynchronize(g_stream));
+            outq.push_back(std::move(op));
+            return;
         }
         int c_dec = add_dev_col(PG_T_I64);
         int c_f64 = add_dev_col(PG_T_F64);
@@ -4285,8 +4450,11 @@ extern "C" pg_status pg_table_reset_acc(int64_t t)
                 return seterr("pg_table_reset_acc: table not found");
             tbl = it->second.get();
         }
-        if (!tbl->acc.p) return PG_OK; /* lazily allocated: nothing yet */
-        CHK(hipMemsetAsync(tbl->acc.p, 0, tbl->acc.sz, g_stream));
+        if (tbl->acc.p)
+            CHK(hipMemsetAsync(tbl->acc.p, 0, tbl->acc.sz, g_stream));
+        if (tbl->acc_multi.p)
+            CHK(hipMemsetAsync(tbl->acc_multi.p, 0, tbl->acc_multi.sz,
+                               g_stream));
         CHK(hipStreamSynchronize(g_stream));
         return PG_OK;
     } catch (const std::exception& e) {

@@ -77,7 +77,8 @@ class PlanHashBuild(C.Structure):
                 ("key_set_only", C.c_int32), ("dense_array", C.c_int32),
                 ("payload_lookup_table", C.c_int64),
                 ("payload_lookup_key_col", C.c_int32),
-                ("agg_table", C.c_int32), ("pack_bits", C.c_int32)]
+                ("agg_table", C.c_int32), ("pack_bits", C.c_int32),
+                ("fill_x10", C.c_int32)]
 
 
 class PlanLookupJoin(C.Structure):
@@ -89,7 +90,8 @@ class PlanLookupJoin(C.Structure):
                 ("table2_key_col", C.c_int32),
                 ("n_group_vals", C.c_int32),
                 ("group_vals", C.c_uint8 * 8), ("dec_only", C.c_int32),
-                ("dec_min", C.c_int32)]
+                ("dec_min", C.c_int32), ("n_aggs", C.c_int32),
+                ("aggs", Agg * 6), ("agg_filter", C.c_int32 * 6)]
 
 
 class PlanTopN(C.Structure):

@@ -338,12 +338,11 @@ class Q5PipelineFused:
         b2.payload_col[0] = 0  # sourced through the lookup instead
         b2.payload_lookup_table = self.b1.table()
         b2.payload_lookup_key_col = orders.channel("custkey")
-        # the date window passes ~15.2% of orders.  //10 deliberately
-        # sizes the packed table at fill ~0.71 (cap = n/5): the 32M-slot
-        # 256 MB key array then sits in the Infinity Cache, so the ~91M
-        # probe HITS read L3 lines instead of HBM — the byte tags already
-        # reject the misses before any key line is touched.
-        b2.capacity_hint = max(orders.n_rows // 10, 64)
+        # the date window passes ~15.2% of orders; //4 keeps fill ~0.18.
+        # Measured: smaller tables (fill 0.36 / 0.71) cost 1-17 ms of the
+        # 600M-row probe — linear-probe cluster length beats any L3
+        # residency gain, so size for SHORT clusters.
+        b2.capacity_hint = max(orders.n_rows // 4, 64)
         b2.agg_table = 1
         b2.pack_bits = 8  # slot = orderkey<<8 | cust_nation (u8)
         self.b2 = Operator(OP_HASH_BUILD, b2)
@@ -1182,127 +1181,80 @@ def q19(part: Page, li: Page):
 
 def q21(supp: Page, orders: Page, li: Page, limit=100):
     """Q21 suppliers who kept orders waiting (q21.sql).  The correlated
-    EXISTS / NOT-EXISTS pair decomposes into per-order aggregates over
-    one fused-agg probe each: (sum suppkey, count) over all lines, the
-    'F'-line count, and (sum, sum*1e4, sum-of-squares, count) over LATE
-    lines.  An order qualifies when every line is 'F', some line has a
-    different supplier, and all late lines share one supplier — checked
-    with exact integer identities (cnt*sum_sq == sum*sum_ticks for the
-    zero-variance test; s* = sum/cnt is that supplier).  numwait(s) then
-    sums cnt_late over qualifying orders with s* = s (SAUDI ARABIA
-    semijoin).  Returns [(suppkey, numwait)] sorted (numwait desc,
-    suppkey asc) LIMIT limit."""
-    import numpy as np
+    EXISTS / NOT-EXISTS pair decomposes into per-order aggregates; all of
+    them come out of ONE multi-accumulator fused-agg probe over lineitem
+    (LOOKUP_JOIN mode 1, n_aggs=5 with per-aggregate FILTER predicates —
+    the InMemoryHashAggregationBuilder + AggregationNode-mask analog):
+      a0 = sum(suppkey)            over all lines of the order
+      a1 = count                   where linestatus = 'F'
+      a2 = sum(suppkey)            where receiptdate > commitdate (late)
+      a3 = count                   where late
+      a4 = sum(suppkey*suppkey)    where late  (raw integer product)
+      cnt = count of all lines
+    An order qualifies when every line is 'F' (a1 == cnt), it has a late
+    line (a3 > 0), all late lines share one supplier (zero variance:
+    a3*a4 == a2*a2, exact in f64 — both sides < 2^53), and some line has
+    a different supplier (a2*cnt != a0*a3); s* = a2/a3 is that supplier.
+    numwait(s) = sum of a3 over qualifying orders with s* = s (counting
+    the l1 late lines, per q21.sql's count(*)), SAUDI ARABIA semijoin.
+    Returns [(suppkey, numwait)] sorted (numwait desc, suppkey asc)
+    LIMIT limit."""
     from .engine import lib
     tables = []
 
-    # ONE orders key table reused by all five fused-agg probes
-    # (lendPartitionLookupSource semantics; accumulators reset between)
     b = PlanHashBuild()
     b.key_col = orders.channel("orderkey")
     b.semijoin_table = -1
     b.capacity_hint = orders.n_rows + 64
     b.agg_table = 1
+    b.fill_x10 = 13  # every probe hits (each line has its order):
+                     # miss clusters are irrelevant, so size tight
     otbl = Operator(OP_HASH_BUILD, b)
     otbl.add_input(orders)
     otbl.finish()
     tables.append(otbl)
 
-    def agg_probe(page, key_col, proj, scale):
-        lib().c.pg_table_reset_acc(otbl.table())
-        jp = PlanLookupJoin()
-        jp.table = otbl.table()
-        jp.key_col = key_col
-        jp.mode = 1
-        jp.proj = proj
-        jp.dec_scale = scale
-        jp.dec_only = 1  # integer sums only: skip the fx128 legs
-        jo = Operator(OP_LOOKUP_JOIN, jp)
-        if isinstance(page, Page):
-            jo.add_input(page)
-        else:
-            jo.add_input_raw(page)
-        jo.finish()
-        g = jo.get_output_raw()  # [orderkey, sum_dec, sum_f64, cnt]
-        return jo, g
-
-    # prefilter the late and 'F' rows once; the per-order probes then
-    # scan the narrow subsets instead of full lineitem three more times
-    def prefilter(preds, cols):
-        fp = PlanFilterProject()
-        fp.n_preds = len(preds)
-        for i, pr in enumerate(preds):
-            fp.preds[i] = pr
-        fp.n_proj = len(cols)
-        for i, c in enumerate(cols):
-            fp.proj[i] = Proj(PROJ_IDENT, c, 0, 0)
-        f = Operator(OP_FILTER_PROJECT, fp)
-        f.add_input(li)
-        return f, f.get_output_raw()
-
-    skc = li.channel("suppkey")
+    sk = li.channel("suppkey")
+    jp = PlanLookupJoin()
+    jp.table = otbl.table()
+    jp.key_col = li.channel("orderkey")
+    jp.mode = 1
+    jp.n_preds = 0  # preds[] holds the per-aggregate FILTER predicates
+    jp.preds[0] = Pred(li.channel("linestatus"), CMP_EQ, ord("F"), 0.0)
     late = Pred(li.channel("receiptdate"), CMP_GT, 0, 0.0)
     late.rhs_col = li.channel("commitdate") + 1
-    f_late, p_late = prefilter([late], (li.channel("orderkey"), skc))
-    f_stat, p_stat = prefilter(
-        [Pred(li.channel("linestatus"), CMP_EQ, ord("F"), 0.0)],
-        (li.channel("orderkey"),))
-    j_all, g_all = agg_probe(li, li.channel("orderkey"),
-                             Proj(PROJ_IDENT, skc, 0, 0), 0)
-    j_f, g_f = agg_probe(p_stat, 0, Proj(PROJ_IDENT, 0, 0, 0), 0)
-    j_lu, g_lu = agg_probe(p_late, 0, Proj(PROJ_IDENT, 1, 0, 0), 0)
-    j_lt, g_lt = agg_probe(p_late, 0, Proj(PROJ_IDENT, 1, 0, 0), 4)
-    j_sq, g_sq = agg_probe(p_late, 0, Proj(PROJ_MUL, 1, 1, 0), 4)
-
-    def chain_join(probe_page, emit_cols, build_page, payload_cols):
-        b = PlanHashBuild()
-        b.key_col = 0
-        b.semijoin_table = -1
-        b.n_payload = len(payload_cols)
-        for i, c in enumerate(payload_cols):
-            b.payload_col[i] = c
-        b.capacity_hint = max(build_page.n_rows, 16)
-        ob = Operator(OP_HASH_BUILD, b)
-        ob.add_input_raw(build_page)
-        ob.finish()
-        tables.append(ob)
-        jp = PlanLookupJoin()
-        jp.table = ob.table()
-        jp.key_col = 0
-        jp.mode = 0
-        jp.n_emit = len(emit_cols)
-        for i, c in enumerate(emit_cols):
-            jp.emit_probe_cols[i] = c
-        jo = Operator(OP_LOOKUP_JOIN, jp)
-        jo.add_input_raw(probe_page)
-        return jo, jo.get_output_raw()
-
-    # combined per-order page
-    ja, pa = chain_join(g_all, (0, 1, 3), g_f, (3,))
-    # pa: [ok, sum_all, cnt_all, cnt_F]
-    jb, pb = chain_join(pa, (0, 1, 2, 3), g_lu, (1, 3))
-    # pb: [ok, sum_all, cnt_all, cnt_F, sum_lu, cnt_l]
-    jc, pc = chain_join(pb, (0, 1, 2, 3, 4, 5), g_lt, (1,))
-    # pc: [ok, sum_all, cnt_all, cnt_F, sum_lu, cnt_l, sum_lt]
-    jd, pd = chain_join(pc, (1, 2, 3, 4, 5, 6), g_sq, (1,))
-    # pd: [sum_all, cnt_all, cnt_F, sum_lu, cnt_l, sum_lt, sum_sq]
+    jp.preds[1] = late
+    jp.n_aggs = 5
+    jp.aggs[0] = Agg(AGG_SUM_I64, Proj(PROJ_IDENT, sk, 0, 0), 0)
+    jp.aggs[1] = Agg(AGG_COUNT, Proj(PROJ_IDENT, 0, 0, 0), 0)
+    jp.aggs[2] = Agg(AGG_SUM_I64, Proj(PROJ_IDENT, sk, 0, 0), 0)
+    jp.aggs[3] = Agg(AGG_COUNT, Proj(PROJ_IDENT, 0, 0, 0), 0)
+    jp.aggs[4] = Agg(AGG_SUM_I64, Proj(PROJ_MUL, sk, sk, 0), 0)
+    for i, f in enumerate((-1, 0, 1, 1, 1)):
+        jp.agg_filter[i] = f
+    jo = Operator(OP_LOOKUP_JOIN, jp)
+    jo.add_input(li)
+    jo.finish()
+    pg = jo.get_output_raw()
+    # pg: [ok(0), sum_all(1), cnt_F(2), sum_l(3), cnt_l(4), sq_l(5),
+    #      cnt_all(6)]
 
     f1p = PlanFilterProject()
-    f1p.n_preds = 1
-    pf = Pred(2, CMP_EQ, 0, 0.0)  # cnt_F == cnt_all (all lines 'F')
-    pf.rhs_col = 1 + 1
+    f1p.n_preds = 2
+    pf = Pred(2, CMP_EQ, 0, 0.0)   # every line 'F': cnt_F == cnt_all
+    pf.rhs_col = 6 + 1
     f1p.preds[0] = pf
+    f1p.preds[1] = Pred(4, CMP_GT, 0, 0.0)  # EXISTS late line
     f1p.n_proj = 6
-    f1p.proj[0] = Proj(PROJ_DIV, 3, 4, 0)   # s* = sum_lu / cnt_l
+    f1p.proj[0] = Proj(PROJ_DIV, 3, 4, 0)   # s* = sum_l / cnt_l
     f1p.proj[1] = Proj(PROJ_IDENT, 4, 0, 0)  # cnt_l
-    f1p.proj[2] = Proj(PROJ_MUL, 4, 6, 0)   # m1 = cnt_l * sum_sq
-    f1p.proj[3] = Proj(PROJ_MUL, 3, 5, 0)   # m2 = sum_lu * sum_lt
-    f1p.proj[4] = Proj(PROJ_MUL, 3, 1, 0)   # m3' = sum_lu * cnt_all
-    f1p.proj[5] = Proj(PROJ_MUL, 0, 4, 0)   # m4' = sum_all * cnt_l
+    f1p.proj[2] = Proj(PROJ_MUL, 4, 5, 0)   # m1 = cnt_l * sq_l  (<2^53)
+    f1p.proj[3] = Proj(PROJ_MUL, 3, 3, 0)   # m2 = sum_l^2       (<2^53)
+    f1p.proj[4] = Proj(PROJ_MUL, 3, 6, 0)   # m3 = sum_l * cnt_all
+    f1p.proj[5] = Proj(PROJ_MUL, 1, 4, 0)   # m4 = sum_all * cnt_l
     f1 = Operator(OP_FILTER_PROJECT, f1p)
-    f1.add_input_raw(pd)
-    pe = f1.get_output_raw()  # [s*, cnt_l, m1, m2, m3', m4']
-    # multi-supplier: sum_all != s* * cnt_all  <=>  m4' != m3' (x cnt_l)
+    f1.add_input_raw(pg)
+    pe = f1.get_output_raw()  # [s*, cnt_l, m1, m2, m3, m4]
 
     bsa = PlanHashBuild()
     bsa.n_preds = 1
@@ -1322,7 +1274,7 @@ def q21(supp: Page, orders: Page, li: Page, limit=100):
     p1 = Pred(2, CMP_EQ, 0, 0.0)  # zero variance: m1 == m2
     p1.rhs_col = 3 + 1
     f2p.preds[0] = p1
-    p2 = Pred(4, CMP_NE, 0, 0.0)  # some different supplier: m3' != m4'
+    p2 = Pred(4, CMP_NE, 0, 0.0)  # some different supplier: m3 != m4
     p2.rhs_col = 5 + 1
     f2p.preds[1] = p2
     f2p.n_proj = 2
@@ -1353,15 +1305,14 @@ def q21(supp: Page, orders: Page, li: Page, limit=100):
     jw.proj = Proj(PROJ_IDENT, 1, 0, 0)
     jw.dec_scale = 0
     jw.dec_only = 1
-    jo = Operator(OP_LOOKUP_JOIN, jw)
-    jo.add_input_raw(pf2)
-    jo.finish()
-    out = jo.get_output(["suppkey", "numwait", "f64", "cnt"])
-    jo.destroy()
+    jn = Operator(OP_LOOKUP_JOIN, jw)
+    jn.add_input_raw(pf2)
+    jn.finish()
+    out = jn.get_output(["suppkey", "numwait", "f64", "cnt"])
+    jn.destroy()
     f2.destroy()
     f1.destroy()
-    for j in (jd, jc, jb, ja, j_sq, j_lt, j_lu, j_f, j_all, f_late, f_stat):
-        j.destroy()
+    jo.destroy()
     for o in tables:
         lib().c.pg_table_destroy(o.table())
         o.destroy()
