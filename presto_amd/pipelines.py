@@ -448,6 +448,8 @@ def q7(cust: Page, orders: Page, supp: Page, li: Page):
     b2.payload_lookup_key_col = orders.channel("custkey")
     b2.capacity_hint = orders.n_rows + 64
     b2.agg_table = 1
+    b2.fill_x10 = 13  # every lineitem probe hits: size tight
+    b2.pack_bits = 8  # slot = orderkey<<8 | cust_nation
     o2 = Operator(OP_HASH_BUILD, b2)
     o2.add_input(orders)
     o2.finish()
@@ -1050,6 +1052,7 @@ def q18(orders: Page, li: Page, limit=100):
     bo.semijoin_table = -1
     bo.capacity_hint = orders.n_rows + 64
     bo.agg_table = 1
+    bo.fill_x10 = 13  # every lineitem probe hits: size tight
     oo = Operator(OP_HASH_BUILD, bo)
     oo.add_input(orders)
     oo.finish()
