@@ -189,6 +189,11 @@ typedef enum {
     PG_OP_TOPN = 5,           /* TopNOperator.java:32,90-111 */
     PG_OP_PARTITION = 6,      /* PartitionedOutputOperator.partitionPage:394 /
                                  LocalExchange partition split */
+    PG_OP_GROUPBY_MULTI = 7,  /* MultiChannelGroupByHash.java:300-380 +
+                                 InMemoryHashAggregationBuilder: general
+                                 grouped aggregation over 1..4 key
+                                 channels of mixed type at arbitrary
+                                 cardinality (see pg_plan_groupby) */
 } pg_op_kind;
 
 #define PG_MAX_PRED 8
@@ -334,6 +339,39 @@ typedef struct {
     pg_agg aggs[6];
     int32_t agg_filter[6];
 } pg_plan_lookup_join;
+
+typedef struct {
+    /* General multi-channel grouped aggregation — the
+     * MultiChannelGroupByHash.java:300-380 analog: arbitrary group
+     * cardinality over 1..4 key channels of mixed I64/I32/U8 or
+     * dictionary-VARBIN type (dictionary keys group by their dictionary
+     * id — DictionaryBlock identity, emitted back as I32 ids into the
+     * same dictionary; requires distinct dictionary entries, i.e. a
+     * proper dictionary).  Open-address table, linear probe, group hash
+     * = murmur3-finalized CombineHashFunction fold of the per-channel
+     * bigint hashes (CombineHashFunction.java:28-30,
+     * AbstractLongType.java:137-140); multi-word keys are claimed with
+     * a per-slot state word (CAS-claim, release-publish), so concurrent
+     * inserts are exact — never fingerprint-approximate.
+     * Aggregates: SUM_DEC/SUM_I64 (overflow-checked ticks), COUNT,
+     * SUM_F64 (exact 64.64 fixed point), MIN/MAX (integer ticks), each
+     * optionally gated by a FILTER predicate (agg_filter indexes
+     * preds[]; row-level filtering uses preds[0..n_preds)).
+     * Output after finish: key channels (input types), then per
+     * aggregate its value column (I64 ticks or F64), then the group
+     * row count; groups emitted slot-ascending (order is not part of
+     * the contract, as with the reference's parallel drivers).
+     * capacity_hint must be >= the number of distinct groups (errors
+     * out past fill 0.85, like agg tables). */
+    int32_t n_preds;
+    pg_pred preds[PG_MAX_PRED];
+    int32_t n_keys;
+    int32_t key_col[4];
+    int64_t capacity_hint;
+    int32_t n_aggs;
+    pg_agg aggs[6];
+    int32_t agg_filter[6];
+} pg_plan_groupby;
 
 typedef struct {
     /* ORDER BY value DESC, date ASC, key ASC LIMIT n  (Q3 shape) */

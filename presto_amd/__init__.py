@@ -16,13 +16,14 @@ fails loudly otherwise.
 from .engine import (  # noqa: F401
     lib, Operator, Page, Varbin, DictVarbin, PlanFilterProject,
     PlanHashAggSmall,
-    PlanHashBuild, PlanLookupJoin, PlanTopN, PlanPartition, Pred, Proj, Agg,
+    PlanHashBuild, PlanLookupJoin, PlanTopN, PlanPartition, PlanGroupBy,
+    Pred, Proj, Agg,
     CMP_LT, CMP_LE, CMP_GT, CMP_GE, CMP_EQ, CMP_NE, CMP_CONTAINS,
     CMP_PREFIX, CMP_CONTAINS2, CMP_NOT_CONTAINS2,
     PROJ_IDENT, PROJ_DISC_PRICE, PROJ_CHARGE, PROJ_MUL, PROJ_DIV,
     PROJ_KEYSHL, PROJ_SHR,
     AGG_COUNT, AGG_SUM_F64, AGG_SUM_DEC, AGG_SUM_I64, AGG_MIN, AGG_MAX,
     OP_FILTER_PROJECT, OP_HASH_AGG_SMALL, OP_HASH_BUILD, OP_LOOKUP_JOIN,
-    OP_TOPN, OP_PARTITION,
+    OP_TOPN, OP_PARTITION, OP_GROUPBY_MULTI,
 )
 from . import pipelines  # noqa: F401

@@ -2226,6 +2226,215 @@ __global__ __launch_bounds__(256) void k_groups_emit_multi(
     }
 }
 
+/* ------------------------------------------------------------------ */
+/* GROUPBY_MULTI: general multi-channel grouped aggregation            */
+/* (MultiChannelGroupByHash.java:300-380 + InMemoryHashAggregation-    */
+/*  Builder.processPage:204).  Group hash = murmur3 finalizer over the */
+/* CombineHashFunction fold (31*h + bigint_hash per channel,           */
+/* CombineHashFunction.java:28-30); open-address linear probe.        */
+/* Multi-word keys are EXACT: a per-slot state word is CAS-claimed     */
+/* (0->1), the claimer publishes the key words with system-scope       */
+/* stores + vmcnt drain + a state=2 flag (the MI355X_MICROARCH.md      */
+/* "sc1 payload -> vmcnt(0) -> sc1 flag" handoff form, so readers      */
+/* need no acquire fences), and losers compare the full key vector —  */
+/* never a fingerprint approximation.                                  */
+/* ------------------------------------------------------------------ */
+struct gb_agg_off {
+    int32_t off[6];
+};
+
+__device__ inline int64_t d_gb_key(const pg_page& pg, int32_t col,
+                                   int64_t i)
+{
+    const pg_col& c = pg.cols[col];
+    if (c.tag == PG_T_VARBIN)
+        return (int64_t)c.dict_ids[i]; /* DictionaryBlock id identity */
+    return d_load_i64(c, i);
+}
+
+__global__ __launch_bounds__(256) void k_groupby_multi(
+    pg_page pg, pg_plan_groupby plan, unsigned int* state, int64_t* kv,
+    int64_t cap, int64_t mask, unsigned long long* acc, int32_t acc_words,
+    gb_agg_off offs, unsigned long long* counters /* [ovf, full] */)
+{
+    const int n_keys = plan.n_keys;
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < pg.n_rows; i += stride) {
+        if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
+        int64_t k[4];
+        uint64_t h = 0;
+#pragma unroll
+        for (int ch = 0; ch < 4; ch++) {
+            if (ch >= n_keys) break;
+            k[ch] = d_gb_key(pg, plan.key_col[ch], i);
+            h = 31u * h + pg_bigint_hash(k[ch]);
+        }
+        int64_t pos = (int64_t)(pg_murmur3_finalize(h) & (uint64_t)mask);
+        int64_t slot = -1, tries = 0;
+        for (;;) {
+            unsigned int st = __hip_atomic_load(
+                &state[pos], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+            if (st == 2u) {
+                bool eq = true;
+#pragma unroll
+                for (int ch = 0; ch < 4; ch++) {
+                    if (ch >= n_keys) break;
+                    eq = eq && __hip_atomic_load(
+                                   &kv[(size_t)ch * cap + pos],
+                                   __ATOMIC_RELAXED,
+                                   __HIP_MEMORY_SCOPE_SYSTEM) == k[ch];
+                }
+                if (eq) {
+                    slot = pos;
+                    break;
+                }
+                pos = (pos + 1) & mask;
+                if (++tries > mask) {
+                    atomicAdd(counters + 1, 1ull);
+                    break;
+                }
+            } else if (st == 0u) {
+                unsigned int prev = atomicCAS(&state[pos], 0u, 1u);
+                if (prev == 0u) {
+#pragma unroll
+                    for (int ch = 0; ch < 4; ch++) {
+                        if (ch >= n_keys) break;
+                        __hip_atomic_store(&kv[(size_t)ch * cap + pos],
+                                           k[ch], __ATOMIC_RELAXED,
+                                           __HIP_MEMORY_SCOPE_SYSTEM);
+                    }
+                    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+                    __hip_atomic_store(&state[pos], 2u, __ATOMIC_RELAXED,
+                                       __HIP_MEMORY_SCOPE_SYSTEM);
+                    slot = pos;
+                    break;
+                }
+                /* lost the claim: retry the same position */
+            }
+            /* st == 1: publisher in flight — retry (it completes within
+             * its own branch, so intra-wave divergence cannot deadlock
+             * this loop) */
+        }
+        if (slot < 0) continue;
+        unsigned long long* s = acc + (size_t)slot * acc_words;
+#pragma unroll
+        for (int a = 0; a < 6; a++) {
+            if (a >= plan.n_aggs) break;
+            int f = plan.agg_filter[a];
+            if (f >= 0 && !d_eval_preds(pg, &plan.preds[f], 1, i))
+                continue;
+            unsigned long long* w = s + offs.off[a];
+            switch (plan.aggs[a].func) {
+                case PG_AGG_COUNT:
+                    atomicAdd(w, 1ull);
+                    break;
+                case PG_AGG_SUM_F64: {
+                    double v = d_eval_proj_f64(pg, plan.aggs[a].proj, i);
+                    uint64_t phi, plo;
+                    fx128_from_f64(v, &phi, &plo);
+                    unsigned long long old = atomicAdd(w, plo);
+                    atomicAdd(w + 1, phi + (old > ~plo ? 1ull : 0ull));
+                    break;
+                }
+                case PG_AGG_MIN:
+                    atomicMin((long long*)w,
+                              (long long)d_eval_proj_dec(pg, plan.aggs[a],
+                                                         i));
+                    break;
+                case PG_AGG_MAX:
+                    atomicMax((long long*)w,
+                              (long long)d_eval_proj_dec(pg, plan.aggs[a],
+                                                         i));
+                    break;
+                default: { /* SUM_DEC / SUM_I64: checked tick sums */
+                    int64_t ticks = d_eval_proj_dec(pg, plan.aggs[a], i);
+                    if (ticks) d_atomic_add_dec_ck(w, ticks, counters);
+                }
+            }
+        }
+        atomicAdd(s + acc_words - 1, 1ull);
+    }
+}
+
+/* MIN/MAX identity init for the groupby accumulator words */
+__global__ __launch_bounds__(256) void k_gb_acc_init(
+    unsigned long long* acc, int64_t cap, int32_t acc_words,
+    gb_agg_off offs, pg_plan_groupby plan)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < cap; i += stride) {
+        unsigned long long* s = acc + (size_t)i * acc_words;
+#pragma unroll
+        for (int a = 0; a < 6; a++) {
+            if (a >= plan.n_aggs) break;
+            if (plan.aggs[a].func == PG_AGG_MIN)
+                s[offs.off[a]] = (unsigned long long)INT64_MAX;
+            else if (plan.aggs[a].func == PG_AGG_MAX)
+                s[offs.off[a]] = (unsigned long long)INT64_MIN;
+        }
+    }
+}
+
+/* groupby extraction: occupied slots (cnt>0) slot-ascending; keys
+ * re-narrowed to their input tags, aggregate columns per layout */
+__global__ __launch_bounds__(256) void k_groupby_emit(
+    const int64_t* kv, int64_t cap, int32_t n_keys,
+    const unsigned long long* acc, int32_t acc_words, gb_agg_off offs,
+    pg_plan_groupby plan, int64_t chunk, const int64_t* block_offs,
+    emit_outs key_outs, emit_outs agg_outs)
+{
+    const int64_t lo = (int64_t)blockIdx.x * chunk;
+    const int64_t hi = min(lo + chunk, cap);
+    const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
+    __shared__ int64_t wcnt[4];
+    __shared__ int64_t running;
+    if (threadIdx.x == 0) running = block_offs[blockIdx.x];
+    __syncthreads();
+    for (int64_t base = lo; base < hi; base += 256) {
+        int64_t i = base + 64 * wid + lane;
+        bool sel = i < hi && acc[i * acc_words + acc_words - 1] > 0;
+        uint64_t m = d_ballot(sel);
+        int wsum = __popcll(m);
+        if (lane == 0) wcnt[wid] = wsum;
+        __syncthreads();
+        int64_t woff = running;
+        for (int w = 0; w < wid; w++) woff += wcnt[w];
+        if (sel) {
+            int64_t pos = woff + __popcll(m & ((1ull << lane) - 1));
+            for (int ch = 0; ch < n_keys; ch++) {
+                int64_t kvv = kv[(size_t)ch * cap + i];
+                switch (key_outs.tag[ch]) {
+                    case PG_T_U8:
+                        ((uint8_t*)key_outs.ptr[ch])[pos] = (uint8_t)kvv;
+                        break;
+                    case PG_T_I32:
+                        ((int32_t*)key_outs.ptr[ch])[pos] = (int32_t)kvv;
+                        break;
+                    default:
+                        ((int64_t*)key_outs.ptr[ch])[pos] = kvv;
+                }
+            }
+            const unsigned long long* s = acc + (size_t)i * acc_words;
+            for (int a = 0; a < plan.n_aggs; a++) {
+                if (plan.aggs[a].func == PG_AGG_SUM_F64)
+                    ((double*)agg_outs.ptr[a])[pos] =
+                        fx128_to_f64(s[offs.off[a] + 1], s[offs.off[a]]);
+                else
+                    ((int64_t*)agg_outs.ptr[a])[pos] =
+                        (int64_t)s[offs.off[a]];
+            }
+            ((int64_t*)agg_outs.ptr[plan.n_aggs])[pos] =
+                (int64_t)s[acc_words - 1];
+        }
+        __syncthreads();
+        if (threadIdx.x == 0)
+            running += wcnt[0] + wcnt[1] + wcnt[2] + wcnt[3];
+        __syncthreads();
+    }
+}
+
 /* extract group rows after fused probe-agg: slots with count>0, emitted
  * slot-ascending (stable compaction over the slot array) */
 __global__ __launch_bounds__(256) void k_groups_count(
@@ -4117,6 +4326,160 @@ struct JoinOp : Op {
     }
 };
 
+/* ---------------- GROUPBY_MULTI ---------------- */
+struct GroupByOp : Op {
+    pg_plan_groupby plan;
+    DevBuf state, kv, acc, counters;
+    int32_t acc_words = 0;
+    gb_agg_off offs{};
+    int64_t cap = 0, mask = 0;
+    int32_t key_out_tag[4] = {0, 0, 0, 0};
+    bool tags_known = false;
+    void init()
+    {
+        if (plan.n_keys < 1 || plan.n_keys > 4)
+            throw std::runtime_error("groupby needs 1..4 key channels");
+        if (plan.n_aggs < 1 || plan.n_aggs > 6)
+            throw std::runtime_error("groupby needs 1..6 aggregates");
+        for (int a = 0; a < plan.n_aggs; a++)
+            if (plan.agg_filter[a] >= PG_MAX_PRED)
+                throw std::runtime_error("agg_filter out of range");
+        if (plan.capacity_hint < 1)
+            throw std::runtime_error("capacity_hint required");
+        cap = next_pow2(plan.capacity_hint * 2 + 16);
+        mask = cap - 1;
+        int w = 0;
+        for (int a = 0; a < plan.n_aggs; a++) {
+            offs.off[a] = w;
+            w += plan.aggs[a].func == PG_AGG_SUM_F64 ? 2 : 1;
+        }
+        acc_words = w + 1; /* + group row count */
+        state.alloc((size_t)cap * 4);
+        state.zero();
+        kv.alloc((size_t)cap * 8 * plan.n_keys);
+        acc.alloc((size_t)cap * acc_words * 8);
+        acc.zero();
+        counters.alloc(16);
+        counters.zero();
+        bool mm = false;
+        for (int a = 0; a < plan.n_aggs; a++)
+            mm = mm || plan.aggs[a].func == PG_AGG_MIN ||
+                 plan.aggs[a].func == PG_AGG_MAX;
+        if (mm)
+            hipLaunchKernelGGL(k_gb_acc_init, dim3(1024), dim3(256), 0,
+                               g_stream, (unsigned long long*)acc.p, cap,
+                               acc_words, offs, plan);
+    }
+    void add_input(const pg_page* in) override
+    {
+        StagedPage sp;
+        sp.stage(in);
+        for (int ch = 0; ch < plan.n_keys; ch++) {
+            const pg_col& c = sp.pg.cols[plan.key_col[ch]];
+            if (c.tag == PG_T_F64)
+                throw std::runtime_error("f64 group keys unsupported");
+            if (c.tag == PG_T_VARBIN && !c.dict_ids)
+                throw std::runtime_error(
+                    "varbin group keys must be dictionary blocks "
+                    "(group identity = dictionary id)");
+            int tag = c.tag == PG_T_VARBIN ? PG_T_I32 : c.tag;
+            if (!tags_known)
+                key_out_tag[ch] = tag;
+            else if (key_out_tag[ch] != tag)
+                throw std::runtime_error("key column type changed");
+        }
+        tags_known = true;
+        hot_begin();
+        hipLaunchKernelGGL(k_groupby_multi, dim3(4096), dim3(256), 0,
+                           g_stream, sp.pg, plan,
+                           (unsigned int*)state.p, (int64_t*)kv.p, cap,
+                           mask, (unsigned long long*)acc.p, acc_words,
+                           offs, (unsigned long long*)counters.p);
+        hot_end();
+        CHKV(hipStreamSynchronize(g_stream));
+    }
+    void finish() override
+    {
+        unsigned long long c[2];
+        CHKV(hipMemcpy(c, counters.p, 16, hipMemcpyDeviceToHost));
+        if (c[1])
+            throw std::runtime_error(
+                "groupby table full: raise capacity_hint");
+        if (c[0])
+            throw std::runtime_error(
+                "bigint/decimal SUM overflow in groupby "
+                "(Math.addExact semantics)");
+        int64_t chunk = (cap + FLT_NB - 1) / FLT_NB;
+        chunk = (chunk + 255) / 256 * 256;
+        if (chunk < 256) chunk = 256;
+        DevBuf d_counts;
+        d_counts.alloc(FLT_NB * 8);
+        hipLaunchKernelGGL(k_groups_count, dim3(FLT_NB), dim3(256), 0,
+                           g_stream, (const unsigned long long*)acc.p,
+                           acc_words, acc_words - 1, cap, chunk,
+                           (int64_t*)d_counts.p);
+        std::vector<int64_t> h(FLT_NB);
+        CHKV(hipMemcpyAsync(h.data(), d_counts.p, FLT_NB * 8,
+                            hipMemcpyDeviceToHost, g_stream));
+        CHKV(hipStreamSynchronize(g_stream));
+        int64_t total = 0;
+        for (int b = 0; b < FLT_NB; b++) {
+            int64_t v = h[b];
+            h[b] = total;
+            total += v;
+        }
+        if (total * 100 > cap * 85)
+            throw std::runtime_error(
+                "groupby fill exceeded 0.85: raise capacity_hint");
+        DevBuf d_offs;
+        d_offs.alloc(FLT_NB * 8);
+        CHKV(hipMemcpyAsync(d_offs.p, h.data(), FLT_NB * 8,
+                            hipMemcpyHostToDevice, g_stream));
+        OutPage op;
+        op.pg.n_rows = total;
+        int nc = 0;
+        auto add_dev_col = [&](int tag) {
+            op.dev.emplace_back();
+            op.dev.back().alloc((size_t)total * type_size(tag) + 1);
+            op.pg.cols[nc].tag = tag;
+            op.pg.cols[nc].on_device = 1;
+            op.pg.cols[nc].data = op.dev.back().p;
+            return nc++;
+        };
+        emit_outs key_outs{};
+        key_outs.n = plan.n_keys;
+        for (int ch = 0; ch < plan.n_keys; ch++) {
+            int tag = tags_known ? key_out_tag[ch] : PG_T_I64;
+            int cx = add_dev_col(tag);
+            key_outs.ptr[ch] = op.pg.cols[cx].data;
+            key_outs.tag[ch] = tag;
+        }
+        emit_outs agg_outs{};
+        agg_outs.n = plan.n_aggs + 1;
+        for (int a = 0; a < plan.n_aggs; a++) {
+            int tag = plan.aggs[a].func == PG_AGG_SUM_F64 ? PG_T_F64
+                                                          : PG_T_I64;
+            int cx = add_dev_col(tag);
+            agg_outs.ptr[a] = op.pg.cols[cx].data;
+            agg_outs.tag[a] = tag;
+        }
+        {
+            int cx = add_dev_col(PG_T_I64);
+            agg_outs.ptr[plan.n_aggs] = op.pg.cols[cx].data;
+            agg_outs.tag[plan.n_aggs] = PG_T_I64;
+        }
+        op.pg.n_cols = nc;
+        hipLaunchKernelGGL(k_groupby_emit, dim3(FLT_NB), dim3(256), 0,
+                           g_stream, (const int64_t*)kv.p, cap,
+                           plan.n_keys,
+                           (const unsigned long long*)acc.p, acc_words,
+                           offs, plan, chunk, (const int64_t*)d_offs.p,
+                           key_outs, agg_outs);
+        CHKV(hipStreamSynchronize(g_stream));
+        outq.push_back(std::move(op));
+    }
+};
+
 /* ---------------- TOPN ---------------- */
 struct TopNOp : Op {
     pg_plan_topn plan;
@@ -4336,6 +4699,15 @@ extern "C" pg_status pg_op_create(int32_t kind, const void* plan,
                     return seterr("bad plan size");
                 auto* o = new PartitionOp();
                 memcpy(&o->plan, plan, sizeof(o->plan));
+                op.reset(o);
+                break;
+            }
+            case PG_OP_GROUPBY_MULTI: {
+                if (plan_bytes != sizeof(pg_plan_groupby))
+                    return seterr("bad plan size");
+                auto* o = new GroupByOp();
+                memcpy(&o->plan, plan, sizeof(o->plan));
+                o->init();
                 op.reset(o);
                 break;
             }

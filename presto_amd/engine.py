@@ -20,7 +20,7 @@ T_U8, T_I32, T_I64, T_F64, T_VARBIN = 0, 1, 2, 3, 4
 (AGG_COUNT, AGG_SUM_F64, AGG_SUM_DEC, AGG_SUM_I64,
  AGG_MIN, AGG_MAX) = range(6)
 (OP_FILTER_PROJECT, OP_HASH_AGG_SMALL, OP_HASH_BUILD, OP_LOOKUP_JOIN,
- OP_TOPN, OP_PARTITION) = range(1, 7)
+ OP_TOPN, OP_PARTITION, OP_GROUPBY_MULTI) = range(1, 8)
 
 _NP_TAG = {np.dtype(np.uint8): T_U8, np.dtype(np.int32): T_I32,
            np.dtype(np.int64): T_I64, np.dtype(np.float64): T_F64}
@@ -91,6 +91,13 @@ class PlanLookupJoin(C.Structure):
                 ("n_group_vals", C.c_int32),
                 ("group_vals", C.c_uint8 * 8), ("dec_only", C.c_int32),
                 ("dec_min", C.c_int32), ("n_aggs", C.c_int32),
+                ("aggs", Agg * 6), ("agg_filter", C.c_int32 * 6)]
+
+
+class PlanGroupBy(C.Structure):
+    _fields_ = [("n_preds", C.c_int32), ("preds", Pred * 8),
+                ("n_keys", C.c_int32), ("key_col", C.c_int32 * 4),
+                ("capacity_hint", C.c_int64), ("n_aggs", C.c_int32),
                 ("aggs", Agg * 6), ("agg_filter", C.c_int32 * 6)]
 
 
