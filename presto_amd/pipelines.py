@@ -14,8 +14,8 @@ import ctypes as C
 
 from .engine import (
     Operator, Page, PlanFilterProject, PlanHashAggSmall, PlanHashBuild,
-    PlanLookupJoin, PlanTopN, PlanPartition, Pred, Proj, Agg,
-    OP_FILTER_PROJECT,
+    PlanLookupJoin, PlanTopN, PlanPartition, PlanGroupBy, Pred, Proj, Agg,
+    OP_FILTER_PROJECT, OP_GROUPBY_MULTI,
     CMP_LE, CMP_LT, CMP_GT, CMP_GE, CMP_EQ, CMP_NE, CMP_CONTAINS,
     CMP_PREFIX, CMP_CONTAINS2, CMP_NOT_CONTAINS2,
     PROJ_IDENT, PROJ_DISC_PRICE, PROJ_CHARGE, PROJ_MUL, PROJ_DIV,
@@ -1659,91 +1659,55 @@ def q16(part: Page, ps: Page, supp: Page, type_name):
     pa = ja.get_output_raw()  # [sk, brand, type, size]
 
     f1p = PlanFilterProject()
-    f1p.n_proj = 3
-    f1p.proj[0] = Proj(PROJ_KEYSHL, 1, 2, 8)   # bt = brand<<8|type
-    f1p.proj[1] = Proj(PROJ_IDENT, 3, 0, 0)    # size
-    f1p.proj[2] = Proj(PROJ_IDENT, 0, 0, 0)    # sk
+    f1p.n_proj = 4
+    f1p.proj[0] = Proj(PROJ_IDENT, 1, 0, 0)    # brand
+    f1p.proj[1] = Proj(PROJ_IDENT, 2, 0, 0)    # type
+    f1p.proj[2] = Proj(PROJ_IDENT, 3, 0, 0)    # size
+    f1p.proj[3] = Proj(PROJ_IDENT, 0, 0, 0)    # sk
     f1p.semijoin_table = ocompl.table()
     f1p.semijoin_col = 0
     f1p.semijoin_anti = 1
     f1 = Operator(OP_FILTER_PROJECT, f1p)
     f1.add_input_raw(pa)
-    pb = f1.get_output_raw()  # [bt, size, sk]
+    pb = f1.get_output_raw()  # [brand, type, size, sk]
 
-    f2p = PlanFilterProject()
-    f2p.n_proj = 2
-    f2p.proj[0] = Proj(PROJ_KEYSHL, 0, 1, 8)   # g = bt<<8|size
-    f2p.proj[1] = Proj(PROJ_IDENT, 2, 0, 0)
-    f2 = Operator(OP_FILTER_PROJECT, f2p)
-    f2.add_input_raw(pb)
-    pc2 = f2.get_output_raw()  # [g, sk]
+    # count(DISTINCT suppkey) per (brand, type, size) as two general
+    # multi-channel group-bys (MultiChannelGroupByHash analog):
+    # distinct (brand,type,size,sk) rows, then supplier count per group
+    g1p = PlanGroupBy()
+    g1p.n_keys = 4
+    for i in range(4):
+        g1p.key_col[i] = i
+    g1p.capacity_hint = max(pb.n_rows, 1024)
+    g1p.n_aggs = 1
+    g1p.aggs[0] = Agg(AGG_COUNT, Proj(PROJ_IDENT, 0, 0, 0), 0)
+    g1p.agg_filter[0] = -1
+    g1 = Operator(OP_GROUPBY_MULTI, g1p)
+    g1.add_input_raw(pb)
+    g1.finish()
+    pdist = g1.get_output_raw()  # [brand, type, size, sk, cnt, cnt]
 
-    f3p = PlanFilterProject()
-    f3p.n_proj = 1
-    f3p.proj[0] = Proj(PROJ_KEYSHL, 0, 1, 32)  # gk = g<<32|sk
-    f3 = Operator(OP_FILTER_PROJECT, f3p)
-    f3.add_input_raw(pc2)
-    pd = f3.get_output_raw()  # [gk]
-
-    # distinct (group, supplier) pairs: a chained table keyed by the
-    # composite (duplicates share a slot) + a fused-agg probe — the
-    # groups ARE the distinct pairs
-    bgk = PlanHashBuild()
-    bgk.key_col = 0
-    bgk.semijoin_table = -1
-    bgk.capacity_hint = max(pd.n_rows, 1024)
-    ogk = Operator(OP_HASH_BUILD, bgk)
-    ogk.add_input_raw(pd)
-    ogk.finish()
-
-    jgk = PlanLookupJoin()
-    jgk.table = ogk.table()
-    jgk.key_col = 0
-    jgk.mode = 1
-    jgk.proj = Proj(PROJ_IDENT, 0, 0, 0)
-    jgk.dec_scale = 0
-    jgk.dec_only = 1
-    jo = Operator(OP_LOOKUP_JOIN, jgk)
-    jo.add_input_raw(pd)
-    jo.finish()
-    gk_groups = jo.get_output_raw()  # [gk, sum, f64, cnt] distinct gk rows
-
-    f4p = PlanFilterProject()
-    f4p.n_proj = 1
-    f4p.proj[0] = Proj(PROJ_SHR, 0, 0, 32)  # g
-    f4 = Operator(OP_FILTER_PROJECT, f4p)
-    f4.add_input_raw(gk_groups)
-    pe = f4.get_output_raw()  # [g] one row per distinct (g, sk)
-
-    bg2 = PlanHashBuild()
-    bg2.key_col = 0
-    bg2.semijoin_table = -1
-    bg2.capacity_hint = max(pe.n_rows, 1024)
-    og2 = Operator(OP_HASH_BUILD, bg2)
-    og2.add_input_raw(pe)
-    og2.finish()
-
-    jg2 = PlanLookupJoin()
-    jg2.table = og2.table()
-    jg2.key_col = 0
-    jg2.mode = 1
-    jg2.proj = Proj(PROJ_IDENT, 0, 0, 0)
-    jg2.dec_scale = 0
-    jg2.dec_only = 1
-    jo2 = Operator(OP_LOOKUP_JOIN, jg2)
-    jo2.add_input_raw(pe)
-    jo2.finish()
-    out = jo2.get_output(["g", "sum", "f64", "cnt"])
+    g2p = PlanGroupBy()
+    g2p.n_keys = 3
+    for i in range(3):
+        g2p.key_col[i] = i
+    g2p.capacity_hint = max(pdist.n_rows, 1024)
+    g2p.n_aggs = 1
+    g2p.aggs[0] = Agg(AGG_COUNT, Proj(PROJ_IDENT, 0, 0, 0), 0)
+    g2p.agg_filter[0] = -1
+    g2 = Operator(OP_GROUPBY_MULTI, g2p)
+    g2.add_input_raw(pdist)
+    g2.finish()
+    out = g2.get_output(["brand", "type", "size", "scnt", "cnt"])
 
     rows = []
-    for i in range(len(out["g"])):
-        g = int(out["g"][i])
-        rows.append((g >> 16, (g >> 8) & 0xFF, g & 0xFF,
-                     int(out["cnt"][i])))
+    for i in range(len(out["brand"])):
+        rows.append((int(out["brand"][i]), int(out["type"][i]),
+                     int(out["size"][i]), int(out["scnt"][i])))
     rows.sort(key=lambda r: (-r[3], r[0], type_name(r[1]), r[2]))
-    for op_ in (jo2, f4, jo, f3, f2, f1, ja):
+    for op_ in (g2, g1, f1, ja):
         op_.destroy()
-    for o in (oqual, ocompl, oattr, ogk, og2):
+    for o in (oqual, ocompl, oattr):
         lib().c.pg_table_destroy(o.table())
         o.destroy()
     return rows
