@@ -449,7 +449,8 @@ def q7(cust: Page, orders: Page, supp: Page, li: Page):
     b2.capacity_hint = orders.n_rows + 64
     b2.agg_table = 1
     b2.fill_x10 = 13  # every lineitem probe hits: size tight
-    b2.pack_bits = 8  # slot = orderkey<<8 | cust_nation
+    # (no pack_bits: this table feeds an EMIT-mode join, which reads
+    # payloads through the slot arrays)
     o2 = Operator(OP_HASH_BUILD, b2)
     o2.add_input(orders)
     o2.finish()
@@ -1444,38 +1445,32 @@ def q9(part: Page, supp: Page, orders: Page, ps: Page, li: Page):
     pd = jc.get_output_raw()  # [qty, ep, dc, cost, nat, odate]
 
     profit = [[0] * 7 for _ in range(25)]
-    for nat in range(25):
-        # one nation-split scan, then the 7 year aggregations run over
-        # the ~1/25-sized slice (175 full-page scans would dominate the
-        # whole pipeline — see profiles/r01_q9_sf10_kernel_stats.txt)
-        fn = PlanFilterProject()
-        fn.n_preds = 1
-        fn.preds[0] = Pred(4, CMP_EQ, nat, 0.0)
-        fn.n_proj = 5
-        for i, c in enumerate((0, 1, 2, 3, 5)):
-            fn.proj[i] = Proj(PROJ_IDENT, c, 0, 0)
-        fnat = Operator(OP_FILTER_PROJECT, fn)
-        fnat.add_input_raw(pd)
-        pnat = fnat.get_output_raw()  # [qty, ep, dc, cost, odate]
-        for y in range(7):
-            p = PlanHashAggSmall()
-            p.n_preds = 2
-            p.preds[0] = Pred(4, CMP_GE, Q9_YEAR_BOUNDS[y], 0.0)
-            p.preds[1] = Pred(4, CMP_LT, Q9_YEAR_BOUNDS[y + 1], 0.0)
-            p.n_keys = 0
-            p.n_aggs = 2
-            p.aggs[0] = Agg(AGG_SUM_DEC, Proj(PROJ_DISC_PRICE, 1, 2, 0), 4)
-            p.aggs[1] = Agg(AGG_SUM_DEC, Proj(PROJ_MUL, 3, 0, 0), 4)
-            a = Operator(OP_HASH_AGG_SMALL, p)
-            a.add_input_raw(pnat)
-            a.finish()
-            r = a.get_output(["rhi", "rlo", "chi", "clo"])
-            if len(r["rlo"]):
-                rev = (int(r["rhi"][0]) << 64) | int(np.uint64(r["rlo"][0]))
-                cst = (int(r["chi"][0]) << 64) | int(np.uint64(r["clo"][0]))
-                profit[nat][y] = rev - cst
-            a.destroy()
-        fnat.destroy()
+    # ONE general multi-channel group-by over (nation, orderdate) — the
+    # MultiChannelGroupByHash analog replaces the former 25 nation-split
+    # scans x 7 year aggregations (175 launches); the date groups fold
+    # into years on the host (exact integer ticks end to end)
+    gq = PlanGroupBy()
+    gq.n_keys = 2
+    gq.key_col[0] = 4  # nat
+    gq.key_col[1] = 5  # odate
+    gq.capacity_hint = 25 * 2500 + 1024
+    gq.n_aggs = 2
+    gq.aggs[0] = Agg(AGG_SUM_DEC, Proj(PROJ_DISC_PRICE, 1, 2, 0), 4)
+    gq.aggs[1] = Agg(AGG_SUM_DEC, Proj(PROJ_MUL, 3, 0, 0), 4)
+    gq.agg_filter[0] = -1
+    gq.agg_filter[1] = -1
+    gop = Operator(OP_GROUPBY_MULTI, gq)
+    gop.add_input_raw(pd)
+    gop.finish()
+    gout = gop.get_output(["nat", "odate", "rev", "cost", "cnt"])
+    gop.destroy()
+    years = np.searchsorted(np.asarray(Q9_YEAR_BOUNDS[1:]),
+                            np.asarray(gout["odate"]), side="right")
+    for i in range(len(gout["nat"])):
+        y = int(years[i])
+        if y < 7:
+            profit[int(gout["nat"][i])][y] += \
+                int(gout["rev"][i]) - int(gout["cost"][i])
 
     jc.destroy()
     jb.destroy()
@@ -1808,10 +1803,16 @@ def q15(supp: Page, li: Page):
     jo.add_input(li)
     jo.finish()
     g = jo.get_output(["suppkey", "rev", "f64", "cnt"])
-    mx = max((int(v) for v in g["rev"]), default=0)
-    rows = sorted((int(g["suppkey"][i]), int(g["rev"][i]))
-                  for i in range(len(g["suppkey"]))
-                  if int(g["rev"][i]) == mx and mx > 0)
+    import numpy as np
+    rev = np.asarray(g["rev"])
+    rows = []
+    if len(rev):
+        mx = int(rev.max())
+        if mx > 0:
+            sel = np.nonzero(rev == mx)[0]
+            sk = np.asarray(g["suppkey"])[sel]
+            order = np.argsort(sk)
+            rows = [(int(sk[i]), mx) for i in order]
     jo.destroy()
     lib().c.pg_table_destroy(os_.table())
     os_.destroy()
