@@ -129,9 +129,11 @@ typedef struct {
                       CONTAINS2 packs both patterns) */
     int32_t slen;
     int32_t rhs_col; /* 0: compare against the constant; k>0: compare
-                        against integer channel k-1 (filter expression
-                        col OP col, e.g. Q5's c_nationkey = s_nationkey).
-                        Zero-initialized plans keep constant semantics. */
+                        against integer channel k-1 PLUS the constant
+                        (col OP col + ival / dval — e.g. Q5's
+                        c_nationkey = s_nationkey, TPC-DS Q72's
+                        d3.d_date > d1.d_date + 5).  Zero-initialized
+                        plans keep constant semantics. */
 } pg_pred;
 
 /* projection expressions (PageProjection analogs) */
@@ -150,6 +152,11 @@ typedef enum {
                                  key packing for multi-channel group-bys,
                                  e.g. Q16's (brand,type,size,suppkey)) */
     PG_PROJ_SHR = 6,          /* a >> c — composite key extraction */
+    PG_PROJ_SUBDIV = 7,       /* (a - b) / c with b, c CONSTANTS (i64
+                                 emit; floor division of non-negative
+                                 differences) — planner expressions like
+                                 date -> week_seq ((date - day0) / 7) or
+                                 multiplicity - 1 */
 } pg_proj_kind;
 
 typedef struct {

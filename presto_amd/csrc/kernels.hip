@@ -258,7 +258,7 @@ __device__ inline bool d_eval_preds(const pg_page& pg, const pg_pred* preds,
             if (pr.rhs_col > 0) {
                 const pg_col& rc = pg.cols[pr.rhs_col - 1];
                 if (rc.null_mask && rc.null_mask[i]) return false;
-                x = d_load_f64(rc, i);
+                x = d_load_f64(rc, i) + pr.dval;
             } else {
                 x = pr.dval;
             }
@@ -276,7 +276,7 @@ __device__ inline bool d_eval_preds(const pg_page& pg, const pg_pred* preds,
             if (pr.rhs_col > 0) {
                 const pg_col& rc = pg.cols[pr.rhs_col - 1];
                 if (rc.null_mask && rc.null_mask[i]) return false;
-                x = d_load_i64(rc, i);
+                x = d_load_i64(rc, i) + pr.ival; /* col OP col + const */
             } else {
                 x = pr.ival;
             }
@@ -324,6 +324,9 @@ __device__ inline int64_t d_eval_proj_dec(const pg_page& pg, const pg_agg& ag,
         for (int k = 0; k < ag.dec_scale; k++) s *= 10.0;
         return (int64_t)(a * s + 0.5);
     }
+    if (p.kind == PG_PROJ_SUBDIV)
+        return (d_load_i64(pg.cols[p.a], i) - (int64_t)p.b) /
+               (p.c ? (int64_t)p.c : 1);
     if (p.kind == PG_PROJ_MUL && ag.dec_scale == 0 &&
         pg.cols[p.a].tag != PG_T_F64 && pg.cols[p.b].tag != PG_T_F64)
         /* raw integer product (e.g. Q21's exact sum of squared
@@ -923,6 +926,12 @@ __device__ inline void d_emit_val(const pg_page& pg, const pg_proj& p,
     }
     if (p.kind == PG_PROJ_SHR) {
         ((int64_t*)out)[pos] = d_load_i64(pg.cols[p.a], i) >> p.c;
+        return;
+    }
+    if (p.kind == PG_PROJ_SUBDIV) {
+        ((int64_t*)out)[pos] =
+            (d_load_i64(pg.cols[p.a], i) - (int64_t)p.b) /
+            (p.c ? (int64_t)p.c : 1);
         return;
     }
     if (p.kind == PG_PROJ_IDENT) {

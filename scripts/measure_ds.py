@@ -1,0 +1,126 @@
+#!/usr/bin/env python3
+"""TPC-DS config-5 measurement (BASELINE.json configs[4]): Q17 + Q72 at
+--sf on one MI355X, verified integer-exact against the CPU oracle.
+
+TEST/BENCH INFRASTRUCTURE: data comes from oracle/tpcds.c (the single
+parity-pinned data source — see oracle/tpcds.h); inputs are staged into
+HBM before the timed region; the step is the full query pipeline.
+Writes one JSON line per query to stdout and gpurun_out/ds_q.json.
+"""
+import argparse
+import json
+import pathlib
+import sys
+import time
+
+sys.path.insert(0, str(pathlib.Path(__file__).resolve().parent.parent))
+
+import numpy as np  # noqa: E402
+import torch  # noqa: E402
+
+import presto_amd as P  # noqa: E402
+from presto_amd.tpcds import DsGen, ds_q17, ds_q72  # noqa: E402
+
+REPO = pathlib.Path(__file__).resolve().parent.parent
+
+
+def dev(a):
+    return torch.from_numpy(np.ascontiguousarray(a)).cuda()
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--sf", type=float, default=100.0)
+    ap.add_argument("--reps", type=int, default=2)
+    ap.add_argument("--queries", default="q17,q72")
+    ap.add_argument("--skip-verify", action="store_true")
+    args = ap.parse_args()
+    want = set(args.queries.split(","))
+    sf = args.sf
+    gen = DsGen(REPO / "oracle" / "liboracle.so")
+    results = []
+    outp = REPO / "gpurun_out"
+    outp.mkdir(exist_ok=True)
+
+    sks = np.arange(1, 73050, dtype=np.int64)
+    d_year, d_qname, _ = gen.date_dim()
+
+    def run(name, fn):
+        best, out = None, None
+        for _ in range(args.reps):
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            out = fn()
+            torch.cuda.synchronize()
+            dt = time.perf_counter() - t0
+            best = dt if best is None else min(best, dt)
+        return out, best
+
+    if "q17" in want:
+        print(f"gen q17 inputs sf={sf} ...", file=sys.stderr, flush=True)
+        ss = gen.store_sales(sf)
+        sr = gen.store_returns(sf)
+        cs = gen.catalog_sales(sf)
+        n_fact = len(ss["date"]) + len(sr["date"]) + len(cs["sold"])
+        ss_p = P.Page({k: dev(ss[k]) for k in
+                       ("date", "item", "cust", "store", "ticket", "qty")})
+        sr_p = P.Page({k: dev(sr[k]) for k in
+                       ("date", "item", "cust", "ticket", "qty")})
+        cs_p = P.Page({k: dev(cs[k]) for k in
+                       ("sold", "item", "cust", "qty")})
+        date_q = P.Page({"sk": sks, "qname": d_qname})
+        q0 = 2001 * 4
+        got, secs = run("q17", lambda: ds_q17(gen, sf, ss_p, sr_p, cs_p,
+                                              date_q, q0))
+        ok = True
+        if not args.skip_verify:
+            exp = gen.q17(sf, q0)
+            ok = got == exp
+        rec = {"query": "tpcds_q17", "sf": sf, "ms": round(secs * 1e3, 3),
+               "fact_rows": n_fact,
+               "grows_per_s": round(n_fact / secs / 1e9, 3), "exact": ok,
+               "groups": len(got)}
+        print(json.dumps(rec), flush=True)
+        results.append(rec)
+        del ss, sr, cs, ss_p, sr_p, cs_p
+
+    if "q72" in want:
+        print(f"gen q72 inputs sf={sf} ...", file=sys.stderr, flush=True)
+        cs = gen.catalog_sales(sf, want_all=True)
+        cr = gen.catalog_returns(sf)
+        inv = gen.inventory(sf)
+        n_fact = len(cs["sold"]) + len(inv["date"])
+        cs_p = P.Page({k: dev(cs[k]) for k in
+                       ("sold", "ship", "item", "order", "qty", "cdemo",
+                        "hdemo", "promo")})
+        cr_p = P.Page({k: dev(cr[k]) for k in ("item", "order")})
+        inv_p = P.Page({k: dev(inv[k]) for k in
+                        ("date", "item", "wh", "qoh")})
+        date_y = P.Page({"sk": sks, "year": d_year})
+        cdemo = P.Page({"sk": np.arange(1, 1920801, dtype=np.int64),
+                        "m": gen.cdemo_marital()})
+        hdemo = P.Page({"sk": np.arange(1, 7201, dtype=np.int64),
+                        "b": gen.hdemo_buypot()})
+        got, secs = run("q72", lambda: ds_q72(gen, sf, cs_p, [inv_p],
+                                              cr_p, date_y, cdemo, hdemo,
+                                              1999, 2, 3))
+        ok = True
+        if not args.skip_verify:
+            exp = gen.q72(sf, 1999, 2, 3)
+            ok = got == exp
+        rec = {"query": "tpcds_q72", "sf": sf, "ms": round(secs * 1e3, 3),
+               "fact_rows": n_fact,
+               "grows_per_s": round(n_fact / secs / 1e9, 3), "exact": ok,
+               "groups": len(got)}
+        print(json.dumps(rec), flush=True)
+        results.append(rec)
+
+    (outp / "ds_q.json").write_text(
+        "\n".join(json.dumps(r) for r in results) + "\n")
+    bad = [r["query"] for r in results if not r["exact"]]
+    print(f"done: {len(results)} queries, inexact: {bad or 'none'}",
+          file=sys.stderr)
+
+
+if __name__ == "__main__":
+    main()
