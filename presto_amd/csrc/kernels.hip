@@ -3137,7 +3137,8 @@ struct FilterOp : Op {
             int tag = plan.proj[o].kind == PG_PROJ_IDENT
                           ? sp.pg.cols[plan.proj[o].a].tag
                       : (plan.proj[o].kind == PG_PROJ_KEYSHL ||
-                         plan.proj[o].kind == PG_PROJ_SHR)
+                         plan.proj[o].kind == PG_PROJ_SHR ||
+                         plan.proj[o].kind == PG_PROJ_SUBDIV)
                           ? PG_T_I64
                           : PG_T_F64;
             if (tag == PG_T_VARBIN) {
