@@ -42,7 +42,10 @@ def main():
     outp = REPO / "gpurun_out"
     outp.mkdir(exist_ok=True)
 
-    sks = np.arange(1, 73050, dtype=np.int64)
+    # fact date columns store raw day indexes, so the date
+    # dimension is keyed by day index too (day 0 never occurs
+    # in facts)
+    sks = np.arange(0, 73049, dtype=np.int64)
     d_year, d_qname, _ = gen.date_dim()
 
     def run(name, fn):

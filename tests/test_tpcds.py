@@ -32,7 +32,10 @@ def _pages(P, dsgen, sf, torch=None):
     cs72 = dsgen.catalog_sales(sf, want_all=True)
     cr = dsgen.catalog_returns(sf)
     inv = dsgen.inventory(sf)
-    sks = np.arange(1, 73050, dtype=np.int64)
+    # fact date columns store raw day indexes, so the date
+    # dimension is keyed by day index too (day 0 never occurs
+    # in facts)
+    sks = np.arange(0, 73049, dtype=np.int64)
     _, qname, _ = dsgen.date_dim()
     year, _, _ = dsgen.date_dim()
     date_q = P.Page({"sk": sks, "qname": qname})
