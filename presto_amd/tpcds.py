@@ -517,21 +517,30 @@ def ds_q72(gen, sf, cs_page, inv_pages, cr_page, date_page, cdemo_page,
         lib().c.pg_table_destroy(t.table())
         t.destroy()
 
-    # host fold: merge base + extras, item_sk -> item_id
-    agg = {}
-    for i in range(len(base["item"])):
-        key = ((int(base["item"][i]) + 1) // 2, int(base["wh"][i]),
-               int(base["wk"][i]))
-        cur = agg.setdefault(key, [0, 0, 0])
-        cur[0] += int(base["no_promo"][i])
-        cur[1] += int(base["promo"][i])
-        cur[2] += int(base["cnt"][i])
+    # host fold (display side), vectorized: item_sk -> item_id, merge
+    # base + extra group rows, sort by (item_id, wh, week)
+    def cols(d, names):
+        return [np.asarray(d[n]) for n in names]
+
+    b_item, b_wh, b_wk, b_np_, b_p, b_c = cols(
+        base, ("item", "wh", "wk", "no_promo", "promo", "cnt"))
+    parts = [((b_item + 1) // 2, b_wh, b_wk, b_np_, b_p, b_c)]
     if extra is not None:
-        for i in range(len(extra["item"])):
-            key = ((int(extra["item"][i]) + 1) // 2, int(extra["wh"][i]),
-                   int(extra["wk"][i]))
-            cur = agg.setdefault(key, [0, 0, 0])
-            cur[0] += int(extra["e_np"][i])
-            cur[1] += int(extra["e_p"][i])
-            cur[2] += int(extra["e_t"][i])
-    return [k + tuple(v) for k, v in sorted(agg.items())]
+        e_item, e_wh, e_wk, e_np_, e_p, e_t = cols(
+            extra, ("item", "wh", "wk", "e_np", "e_p", "e_t"))
+        parts.append(((e_item + 1) // 2, e_wh, e_wk, e_np_, e_p, e_t))
+    item = np.concatenate([p[0] for p in parts])
+    wh = np.concatenate([p[1] for p in parts])
+    wk = np.concatenate([p[2] for p in parts]).astype(np.int64)
+    vnp = np.concatenate([p[3] for p in parts])
+    vp = np.concatenate([p[4] for p in parts])
+    vt = np.concatenate([p[5] for p in parts])
+    key = (item << 32) | (wh << 16) | wk
+    ukey, inv_ = np.unique(key, return_inverse=True)
+    s_np = np.bincount(inv_, vnp, minlength=len(ukey)).astype(np.int64)
+    s_p = np.bincount(inv_, vp, minlength=len(ukey)).astype(np.int64)
+    s_t = np.bincount(inv_, vt, minlength=len(ukey)).astype(np.int64)
+    return list(zip((ukey >> 32).tolist(),
+                    ((ukey >> 16) & 0xffff).tolist(),
+                    (ukey & 0xffff).tolist(),
+                    s_np.tolist(), s_p.tolist(), s_t.tolist()))
