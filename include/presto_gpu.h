@@ -71,6 +71,14 @@ typedef enum {
     PG_T_I32 = 1, /* IntArrayBlock / DateType days */
     PG_T_I64 = 2, /* LongArrayBlock / BigintType */
     PG_T_F64 = 3, /* LongArrayBlock bits / DoubleType */
+    PG_T_I128 = 5, /* Int128ArrayBlock (Int128ArrayBlock.java): 16-byte
+                       little-endian (low, high) pairs — decimal(p>18,s)
+                       unscaled values.  Carried through staging, IDENT
+                       projection/emit and the INT128_ARRAY wire encoding;
+                       decimal arithmetic accumulates through the exact
+                       int128 totals of the aggregation paths
+                       (fixed128.h, UnscaledDecimal128Arithmetic.java:770
+                       add semantics). */
     PG_T_VARBIN = 4, /* VariableWidthBlock (VariableWidthBlock.java:48-61):
                         data = bytes, offsets = int32[n_rows+1] (element i
                         spans bytes [offsets[i], offsets[i+1])).  Supports
@@ -445,6 +453,14 @@ pg_status pg_table_reset_acc(int64_t table);
  * Host-side (the node-boundary seam stays on the host). */
 pg_status pg_page_serialize(const pg_page* page /* host cols */,
                             void* out, int64_t cap, int64_t* out_len);
+/* round-2 wire scope: compress=1 runs the body through the LZ4 block
+ * codec and sets PageCodecMarker.COMPRESSED when the ratio clears
+ * PagesSerde.java:41's MINIMUM_COMPRESSION_RATIO (0.9); encodings now
+ * cover LONG/INT/BYTE/INT128_ARRAY, VARIABLE_WIDTH, DICTIONARY
+ * (varbin dictionaries travel as dictionaries, fixed-width ones expand
+ * on read) and RLE (expanded on read). */
+pg_status pg_page_serialize2(const pg_page* page, int32_t compress,
+                             void* out, int64_t cap, int64_t* out_len);
 /* parses and verifies; fills *out with malloc-backed host columns
  * (free with pg_page_free). F64 consumers reinterpret LONG_ARRAY bits. */
 pg_status pg_page_deserialize(const void* buf, int64_t len, pg_page* out);

@@ -946,6 +946,13 @@ __device__ inline void d_emit_val(const pg_page& pg, const pg_proj& p,
             case PG_T_I64:
                 ((int64_t*)out)[pos] = ((const int64_t*)c.data)[i];
                 return;
+            case PG_T_I128: {
+                const int64_t* s = (const int64_t*)c.data + 2 * i;
+                int64_t* d = (int64_t*)out + 2 * pos;
+                d[0] = s[0];
+                d[1] = s[1];
+                return;
+            }
             default:
                 ((double*)out)[pos] = ((const double*)c.data)[i];
                 return;
@@ -2916,6 +2923,7 @@ static size_t type_size(int tag)
     switch (tag) {
         case PG_T_U8: return 1;
         case PG_T_I32: return 4;
+        case PG_T_I128: return 16;
         default: return 8;
     }
 }
@@ -4996,6 +5004,120 @@ static const char* enc_name(int tag)
     }
 }
 
+/* ---- LZ4 block format codec (host side) ----
+ * The reference compresses SerializedPage bodies with airlift
+ * aircompressor's Lz4Compressor (PagesSerde.java:67-95); this is an
+ * independent implementation of the same LZ4 block format (the wire
+ * LAYOUT — marker bit, sizes, checksum — is byte-pinned against
+ * PagesSerdeUtil; the compressed byte stream itself is codec output and
+ * interoperable with any LZ4 block decoder, not byte-identical to
+ * aircompressor's encoder choices). */
+static int64_t lz4_compress(const uint8_t* src, int64_t n, uint8_t* dst,
+                            int64_t cap)
+{
+    if (n == 0) return 0;
+    const int64_t HASH_SIZE = 1 << 14;
+    static thread_local std::vector<int32_t> table;
+    table.assign(HASH_SIZE, -1);
+    int64_t ip = 0, op = 0, anchor = 0;
+    const int64_t mflimit = n - 12; /* last match must end 12 before n */
+    auto hash4 = [&](int64_t p) {
+        uint32_t v;
+        memcpy(&v, src + p, 4);
+        return (int64_t)((v * 2654435761u) >> 18);
+    };
+    auto emit = [&](int64_t lit_len, int64_t lit_at, int64_t mlen,
+                    int64_t dist) -> bool {
+        int64_t need = 1 + lit_len + lit_len / 255 + 1 + 2 + mlen / 255 + 1;
+        if (op + need > cap) return false;
+        uint8_t* tok = dst + op++;
+        int64_t l = lit_len;
+        *tok = (uint8_t)((l >= 15 ? 15 : l) << 4);
+        if (l >= 15) {
+            l -= 15;
+            while (l >= 255) { dst[op++] = 255; l -= 255; }
+            dst[op++] = (uint8_t)l;
+        }
+        memcpy(dst + op, src + lit_at, lit_len);
+        op += lit_len;
+        if (mlen > 0) {
+            dst[op++] = (uint8_t)dist;
+            dst[op++] = (uint8_t)(dist >> 8);
+            int64_t m = mlen - 4;
+            *tok |= (uint8_t)(m >= 15 ? 15 : m);
+            if (m >= 15) {
+                m -= 15;
+                while (m >= 255) { dst[op++] = 255; m -= 255; }
+                dst[op++] = (uint8_t)m;
+            }
+        }
+        return true;
+    };
+    while (ip < mflimit) {
+        int64_t h = hash4(ip);
+        int64_t ref = table[h];
+        table[h] = (int32_t)ip;
+        uint32_t a, b;
+        if (ref >= 0 && ip - ref < 65536) {
+            memcpy(&a, src + ref, 4);
+            memcpy(&b, src + ip, 4);
+            if (a == b) {
+                int64_t mlen = 4;
+                while (ip + mlen < n - 5 &&
+                       src[ref + mlen] == src[ip + mlen])
+                    mlen++;
+                if (!emit(ip - anchor, anchor, mlen, ip - ref)) return -1;
+                ip += mlen;
+                anchor = ip;
+                continue;
+            }
+        }
+        ip++;
+    }
+    /* final literals */
+    if (!emit(n - anchor, anchor, 0, 0)) return -1;
+    return op;
+}
+
+static bool lz4_decompress(const uint8_t* src, int64_t n, uint8_t* dst,
+                           int64_t dn)
+{
+    int64_t ip = 0, op = 0;
+    while (ip < n) {
+        uint8_t tok = src[ip++];
+        int64_t lit = tok >> 4;
+        if (lit == 15) {
+            uint8_t b;
+            do {
+                if (ip >= n) return false;
+                b = src[ip++];
+                lit += b;
+            } while (b == 255);
+        }
+        if (ip + lit > n || op + lit > dn) return false;
+        memcpy(dst + op, src + ip, lit);
+        ip += lit;
+        op += lit;
+        if (ip >= n) break; /* last sequence has no match */
+        if (ip + 2 > n) return false;
+        int64_t dist = src[ip] | ((int64_t)src[ip + 1] << 8);
+        ip += 2;
+        if (dist == 0 || dist > op) return false;
+        int64_t mlen = (tok & 15) + 4;
+        if ((tok & 15) == 15) {
+            uint8_t b;
+            do {
+                if (ip >= n) return false;
+                b = src[ip++];
+                mlen += b;
+            } while (b == 255);
+        }
+        if (op + mlen > dn) return false;
+        for (int64_t j = 0; j < mlen; j++, op++) dst[op] = dst[op - dist];
+    }
+    return op == dn;
+}
+
 /* encodeNullsAsBits — EncoderUtil.java:31-63 (MSB-first per byte) */
 static void write_null_bits(ByteWriter& w, const uint8_t* mask, int64_t n)
 {
@@ -5017,73 +5139,320 @@ static void write_null_bits(ByteWriter& w, const uint8_t* mask, int64_t n)
 
 } /* namespace */
 
+namespace {
+
+/* write one block (writeRawPage writes blockCount then per-block a
+ * length-prefixed encoding name + body — BlockEncodingManager.java:96-99) */
+static void write_block(ByteWriter& w, const pg_col& col, int64_t n)
+{
+    if (col.tag == PG_T_VARBIN && col.dict_ids) {
+        /* DICTIONARY (DictionaryBlockEncoding.java:32-55): positionCount,
+         * nested dictionary block, raw int ids, then the 3-long
+         * DictionaryId (most/least significant bits + sequence id) */
+        static const char* dn = "DICTIONARY";
+        w.i32((int32_t)strlen(dn));
+        w.bytes(dn, strlen(dn));
+        w.i32((int32_t)n);
+        pg_col dict = col;
+        dict.dict_ids = nullptr;
+        write_block(w, dict, col.dict_n);
+        w.bytes(col.dict_ids, n * 4);
+        w.i64(0x50472d414d442d31ll); /* fixed instance id: readers treat
+                                        it as an opaque dedup token */
+        w.i64(0x4449435400000000ll);
+        w.i64(1);
+        return;
+    }
+    if (col.tag == PG_T_I128) {
+        /* INT128_ARRAY (Int128ArrayBlockEncoding.java:36-50): two longs
+         * per non-null position after the null bits */
+        static const char* in = "INT128_ARRAY";
+        w.i32((int32_t)strlen(in));
+        w.bytes(in, strlen(in));
+        w.i32((int32_t)n);
+        write_null_bits(w, col.null_mask, n);
+        const uint8_t* d = (const uint8_t*)col.data;
+        if (!col.null_mask) {
+            w.bytes(d, n * 16);
+        } else {
+            for (int64_t i = 0; i < n; i++)
+                if (!col.null_mask[i]) w.bytes(d + i * 16, 16);
+        }
+        return;
+    }
+    const char* name = enc_name(col.tag);
+    int32_t nl = (int32_t)strlen(name);
+    w.i32(nl);
+    w.bytes(name, nl);
+    w.i32((int32_t)n);
+    if (col.tag == PG_T_VARBIN) {
+        /* VariableWidthBlockEncoding.writeBlock:37-58 */
+        for (int64_t i = 0; i < n; i++) w.i32(col.offsets[i + 1]);
+        write_null_bits(w, col.null_mask, n);
+        int32_t total = col.offsets[n];
+        w.i32(total);
+        w.bytes(col.data, total);
+        return;
+    }
+    write_null_bits(w, col.null_mask, n);
+    size_t esz = type_size(col.tag);
+    if (!col.null_mask) {
+        w.bytes(col.data, n * esz);
+    } else {
+        for (int64_t i = 0; i < n; i++)
+            if (!col.null_mask[i])
+                w.bytes((const uint8_t*)col.data + i * esz, esz);
+    }
+}
+
+static bool read_null_bits(ByteReader& r, int64_t n, uint8_t** out_mask)
+{
+    uint8_t may = r.u8();
+    *out_mask = nullptr;
+    if (!may) return r.ok;
+    uint8_t* mask = (uint8_t*)calloc(n ? n : 1, 1);
+    for (int64_t base = 0; base < n; base += 8) {
+        uint8_t v = r.u8();
+        for (int64_t j = base; j < base + 8 && j < n; j++)
+            mask[j] = (v >> (7 - (j - base))) & 1;
+    }
+    *out_mask = mask;
+    return r.ok;
+}
+
+static pg_status read_block(ByteReader& r, pg_col* out, int64_t* out_n);
+
+/* expand an RLE value block to n positions (materialized on read, the
+ * way LazyBlock consumers see a flat block) */
+static pg_status expand_rle(const pg_col& v, int64_t vn, int64_t n,
+                            pg_col* out)
+{
+    (void)vn;
+    memset(out, 0, sizeof(*out));
+    out->tag = v.tag;
+    bool isnull = v.null_mask && v.null_mask[0];
+    if (v.tag == PG_T_VARBIN) {
+        int32_t len = isnull ? 0 : v.offsets[1];
+        int32_t* offs = (int32_t*)calloc((size_t)n + 1, 4);
+        uint8_t* data = (uint8_t*)calloc((size_t)len * (n ? n : 1) + 1, 1);
+        for (int64_t i = 0; i < n; i++) {
+            offs[i + 1] = offs[i] + len;
+            if (len) memcpy(data + offs[i], v.data, len);
+        }
+        out->offsets = offs;
+        out->data = data;
+    } else {
+        size_t esz = type_size(v.tag);
+        uint8_t* data = (uint8_t*)calloc(n ? n : 1, esz);
+        if (!isnull)
+            for (int64_t i = 0; i < n; i++)
+                memcpy(data + i * esz, v.data, esz);
+        out->data = data;
+    }
+    if (isnull) {
+        uint8_t* mask = (uint8_t*)malloc(n ? n : 1);
+        memset(mask, 1, n ? n : 1);
+        out->null_mask = mask;
+    }
+    free((void*)v.data);
+    free((void*)v.null_mask);
+    free((void*)v.offsets);
+    return PG_OK;
+}
+
+static pg_status read_block(ByteReader& r, pg_col* out, int64_t* out_n)
+{
+    memset(out, 0, sizeof(*out));
+    int32_t nl = r.i32();
+    if (!r.ok || nl < 0 || nl > 64)
+        return seterr("deserialize: bad encoding name");
+    char name[65] = {0};
+    const void* np_ = r.bytes(nl);
+    if (np_) memcpy(name, np_, nl);
+    if (!strcmp(name, "RLE")) {
+        /* RunLengthBlockEncoding.java:31-51: run length then the
+         * single-position value block */
+        int32_t n = r.i32();
+        pg_col v;
+        int64_t vn = 0;
+        pg_status st = read_block(r, &v, &vn);
+        if (st != PG_OK) return st;
+        if (vn != 1) return seterr("deserialize: RLE value not 1 row");
+        *out_n = n;
+        return expand_rle(v, vn, n, out);
+    }
+    if (!strcmp(name, "DICTIONARY")) {
+        int32_t n = r.i32();
+        pg_col dict;
+        int64_t dn = 0;
+        pg_status st = read_block(r, &dict, &dn);
+        if (st != PG_OK) return st;
+        int32_t* ids = (int32_t*)calloc(n ? n : 1, 4);
+        const void* idsrc = r.bytes((int64_t)n * 4);
+        if (idsrc) memcpy(ids, idsrc, (size_t)n * 4);
+        r.i64();
+        r.i64();
+        r.i64(); /* DictionaryId: opaque */
+        if (!r.ok) {
+            free(ids);
+            return seterr("deserialize: truncated DICTIONARY block");
+        }
+        *out_n = n;
+        if (dict.tag == PG_T_VARBIN) {
+            /* preserved in dictionary form (DictionaryBlock semantics) */
+            *out = dict;
+            out->dict_ids = ids;
+            out->dict_n = (int32_t)dn;
+            return PG_OK;
+        }
+        /* fixed-width dictionaries are expanded on read */
+        size_t esz = type_size(dict.tag);
+        uint8_t* data = (uint8_t*)calloc(n ? n : 1, esz);
+        uint8_t* mask = nullptr;
+        for (int64_t i = 0; i < n; i++) {
+            int32_t id = ids[i];
+            if (id < 0 || id >= dn) {
+                free(ids);
+                free(data);
+                free(mask);
+                free((void*)dict.data);
+                free((void*)dict.null_mask);
+                return seterr("deserialize: dictionary id out of range");
+            }
+            memcpy(data + i * esz, (const uint8_t*)dict.data + id * esz,
+                   esz);
+            if (dict.null_mask && dict.null_mask[id]) {
+                if (!mask) mask = (uint8_t*)calloc(n, 1);
+                mask[i] = 1;
+            }
+        }
+        free(ids);
+        free((void*)dict.data);
+        free((void*)dict.null_mask);
+        free((void*)dict.offsets);
+        out->tag = dict.tag;
+        out->data = data;
+        out->null_mask = mask;
+        return PG_OK;
+    }
+    int tag;
+    if (!strcmp(name, "LONG_ARRAY")) tag = PG_T_I64;
+    else if (!strcmp(name, "INT_ARRAY")) tag = PG_T_I32;
+    else if (!strcmp(name, "BYTE_ARRAY")) tag = PG_T_U8;
+    else if (!strcmp(name, "INT128_ARRAY")) tag = PG_T_I128;
+    else if (!strcmp(name, "VARIABLE_WIDTH")) tag = PG_T_VARBIN;
+    else return seterr("deserialize: unsupported block encoding");
+    int32_t n = r.i32();
+    *out_n = n;
+    if (tag == PG_T_VARBIN) {
+        /* VariableWidthBlockEncoding.readBlock:62-76 */
+        int32_t* offs = (int32_t*)calloc((size_t)n + 1, 4);
+        for (int64_t i = 0; i < n; i++) offs[i + 1] = r.i32();
+        uint8_t* vmask = nullptr;
+        read_null_bits(r, n, &vmask);
+        int32_t nb = r.i32();
+        uint8_t* vb = (uint8_t*)calloc(nb ? nb : 1, 1);
+        const void* vsrc = r.bytes(nb);
+        if (vsrc) memcpy(vb, vsrc, (size_t)nb);
+        if (!r.ok || nb != offs[n]) {
+            free(offs);
+            free(vmask);
+            free(vb);
+            return seterr("deserialize: bad VARIABLE_WIDTH block");
+        }
+        out->tag = PG_T_VARBIN;
+        out->data = vb;
+        out->offsets = offs;
+        out->null_mask = vmask;
+        return PG_OK;
+    }
+    uint8_t* mask = nullptr;
+    read_null_bits(r, n, &mask);
+    size_t esz = type_size(tag);
+    uint8_t* vals = (uint8_t*)calloc(n ? n : 1, esz);
+    if (!mask) {
+        const void* vsrc = r.bytes((int64_t)n * esz);
+        if (vsrc) memcpy(vals, vsrc, (size_t)n * esz);
+    } else {
+        for (int64_t i = 0; i < n; i++)
+            if (!mask[i]) {
+                const void* vsrc = r.bytes(esz);
+                if (vsrc) memcpy(vals + i * esz, vsrc, esz);
+            }
+    }
+    if (!r.ok) {
+        free(mask);
+        free(vals);
+        return seterr("deserialize: truncated block");
+    }
+    out->tag = tag;
+    out->data = vals;
+    out->null_mask = mask;
+    return PG_OK;
+}
+
+} /* namespace */
+
+extern "C" pg_status pg_page_serialize2(const pg_page* page,
+                                        int32_t compress, void* out,
+                                        int64_t cap, int64_t* out_len)
+{
+    for (int c = 0; c < page->n_cols; c++)
+        if (page->cols[c].on_device)
+            return seterr("pg_page_serialize: host columns required");
+    /* body built aside so the whole page can go through the compressor
+     * (PagesSerde.java:67-95) */
+    std::vector<uint8_t> body(64 + (size_t)page->n_rows * 64);
+    ByteWriter w{body.data(), (int64_t)body.size()};
+    for (;;) {
+        w = ByteWriter{body.data(), (int64_t)body.size()};
+        w.i32(page->n_cols);
+        for (int c = 0; c < page->n_cols; c++)
+            write_block(w, page->cols[c], page->n_rows);
+        if (w.ok) break;
+        body.resize(body.size() * 2);
+    }
+    int64_t usize = w.off;
+    uint8_t marker = 0;
+    const uint8_t* stored = body.data();
+    int64_t ssize = usize;
+    std::vector<uint8_t> cbuf;
+    if (compress) {
+        cbuf.resize((size_t)usize + usize / 16 + 256);
+        int64_t cs = lz4_compress(body.data(), usize, cbuf.data(),
+                                  (int64_t)cbuf.size());
+        /* MINIMUM_COMPRESSION_RATIO = 0.9 (PagesSerde.java:41) */
+        if (cs > 0 && (double)cs / (double)usize <= 0.9) {
+            marker = 1; /* PageCodecMarker.COMPRESSED */
+            stored = cbuf.data();
+            ssize = cs;
+        }
+    }
+    ByteWriter m{(uint8_t*)out, cap};
+    m.i32((int32_t)page->n_rows);
+    m.u8(marker);
+    m.i32((int32_t)usize);
+    m.i32((int32_t)ssize);
+    /* computeSerializedPageChecksum:109-120 */
+    uint32_t crc = pg_crc32(0, stored, ssize);
+    uint8_t tail[9];
+    tail[0] = marker;
+    int32_t pc = (int32_t)page->n_rows;
+    int32_t us32 = (int32_t)usize;
+    memcpy(tail + 1, &pc, 4);
+    memcpy(tail + 5, &us32, 4);
+    crc = pg_crc32(crc, tail, 9);
+    m.i64((int64_t)(uint32_t)crc);
+    m.bytes(stored, ssize);
+    if (!m.ok) return seterr("pg_page_serialize: buffer too small");
+    *out_len = m.off;
+    return PG_OK;
+}
+
 extern "C" pg_status pg_page_serialize(const pg_page* page, void* out,
                                        int64_t cap, int64_t* out_len)
 {
-    for (int c = 0; c < page->n_cols; c++) {
-        if (page->cols[c].on_device)
-            return seterr("pg_page_serialize: host columns required");
-        if (page->cols[c].dict_ids)
-            return seterr(
-                "pg_page_serialize: dictionary blocks must be expanded "
-                "before serialization (v1 wire scope)");
-    }
-    ByteWriter w{(uint8_t*)out, cap};
-    /* metadata placeholder, filled after the body is written */
-    int64_t meta_at = w.off;
-    w.off += 4 + 1 + 4 + 4 + 8;
-    int64_t body_at = w.off;
-    w.i32(page->n_cols); /* writeRawPage: block count */
-    for (int c = 0; c < page->n_cols; c++) {
-        const pg_col& col = page->cols[c];
-        const char* name = enc_name(col.tag);
-        int32_t nl = (int32_t)strlen(name);
-        w.i32(nl);
-        w.bytes(name, nl);
-        w.i32((int32_t)page->n_rows);
-        if (col.tag == PG_T_VARBIN) {
-            /* VariableWidthBlockEncoding.writeBlock:37-58: cumulative
-             * lengths for every position, null bits, total length, then
-             * the raw byte region */
-            for (int64_t i = 0; i < page->n_rows; i++)
-                w.i32(col.offsets[i + 1]);
-            write_null_bits(w, col.null_mask, page->n_rows);
-            int32_t total = col.offsets[page->n_rows];
-            w.i32(total);
-            w.bytes(col.data, total);
-            continue;
-        }
-        write_null_bits(w, col.null_mask, page->n_rows);
-        size_t esz = type_size(col.tag);
-        if (!col.null_mask) {
-            w.bytes(col.data, page->n_rows * esz);
-        } else {
-            for (int64_t i = 0; i < page->n_rows; i++)
-                if (!col.null_mask[i])
-                    w.bytes((const uint8_t*)col.data + i * esz, esz);
-        }
-    }
-    if (!w.ok) return seterr("pg_page_serialize: buffer too small");
-    int32_t body = (int32_t)(w.off - body_at);
-    /* metadata: PagesSerdeUtil.writeSerializedPageMetadata:70-77 */
-    ByteWriter m{(uint8_t*)out + meta_at, 4 + 1 + 4 + 4 + 8};
-    m.i32((int32_t)page->n_rows);
-    m.u8(0); /* codec markers: uncompressed, unencrypted, no checksum bit —
-                the checksum field is still always present on the wire */
-    m.i32(body); /* uncompressed size */
-    m.i32(body); /* size */
-    /* computeSerializedPageChecksum:109-120: crc32(data) then marker byte
-     * then positionCount and uncompressedSize as 4 LE bytes each */
-    uint32_t crc = pg_crc32(0, (uint8_t*)out + body_at, body);
-    uint8_t tail[9];
-    tail[0] = 0;
-    int32_t pc = (int32_t)page->n_rows;
-    memcpy(tail + 1, &pc, 4);
-    memcpy(tail + 5, &body, 4);
-    crc = pg_crc32(crc, tail, 9);
-    m.i64((int64_t)(uint32_t)crc);
-    *out_len = w.off;
-    return PG_OK;
+    return pg_page_serialize2(page, 0, out, cap, out_len);
 }
 
 extern "C" pg_status pg_page_deserialize(const void* buf, int64_t len,
@@ -5096,12 +5465,10 @@ extern "C" pg_status pg_page_deserialize(const void* buf, int64_t len,
     int32_t size = r.i32();
     int64_t checksum = r.i64();
     if (!r.ok) return seterr("deserialize: truncated metadata");
-    if (marker != 0)
-        return seterr("deserialize: compressed/encrypted pages unsupported "
-                      "(codec marker != 0)");
-    if (size != usize || r.off + size > len)
-        return seterr("deserialize: bad sizes");
-    /* verify checksum */
+    if (marker & ~1)
+        return seterr("deserialize: encrypted/checksummed markers "
+                      "unsupported");
+    if (r.off + size > len) return seterr("deserialize: bad sizes");
     uint32_t crc = pg_crc32(0, r.p + r.off, size);
     uint8_t tail[9];
     tail[0] = marker;
@@ -5110,89 +5477,33 @@ extern "C" pg_status pg_page_deserialize(const void* buf, int64_t len,
     crc = pg_crc32(crc, tail, 9);
     if ((int64_t)(uint32_t)crc != checksum)
         return seterr("deserialize: checksum mismatch");
-    int32_t n_blocks = r.i32();
+    std::vector<uint8_t> ubuf;
+    ByteReader body{r.p + r.off, size};
+    if (marker & 1) {
+        ubuf.resize(usize ? usize : 1);
+        if (!lz4_decompress(r.p + r.off, size, ubuf.data(), usize))
+            return seterr("deserialize: LZ4 stream corrupt");
+        body = ByteReader{ubuf.data(), usize};
+    } else if (size != usize) {
+        return seterr("deserialize: bad sizes");
+    }
+    int32_t n_blocks = body.i32();
     if (n_blocks < 0 || n_blocks > 16)
         return seterr("deserialize: unsupported block count");
     memset(out, 0, sizeof(*out));
     out->n_rows = pos_count;
     out->n_cols = n_blocks;
     for (int c = 0; c < n_blocks; c++) {
-        int32_t nl = r.i32();
-        if (!r.ok || nl < 0 || nl > 64)
-            return seterr("deserialize: bad encoding name");
-        char name[65] = {0};
-        memcpy(name, r.bytes(nl), r.ok ? nl : 0);
-        int tag;
-        if (!strcmp(name, "LONG_ARRAY")) tag = PG_T_I64;
-        else if (!strcmp(name, "INT_ARRAY")) tag = PG_T_I32;
-        else if (!strcmp(name, "BYTE_ARRAY")) tag = PG_T_U8;
-        else if (!strcmp(name, "VARIABLE_WIDTH")) tag = PG_T_VARBIN;
-        else return seterr("deserialize: unsupported block encoding");
-        int32_t n = r.i32();
-        if (n != pos_count)
+        int64_t bn = 0;
+        pg_status st = read_block(body, &out->cols[c], &bn);
+        if (st != PG_OK) {
+            pg_page_free(out);
+            return st;
+        }
+        if (bn != pos_count) {
+            pg_page_free(out);
             return seterr("deserialize: block position count mismatch");
-        if (tag == PG_T_VARBIN) {
-            /* VariableWidthBlockEncoding.readBlock:62-76 */
-            int32_t* offs = (int32_t*)calloc((size_t)n + 1, 4);
-            for (int64_t i = 0; i < n; i++) offs[i + 1] = r.i32();
-            uint8_t vb_null = r.u8();
-            uint8_t* vmask = nullptr;
-            if (vb_null) {
-                vmask = (uint8_t*)calloc(n, 1);
-                for (int64_t base = 0; base < n; base += 8) {
-                    uint8_t v = r.u8();
-                    for (int64_t j = base; j < base + 8 && j < n; j++)
-                        vmask[j] = (v >> (7 - (j - base))) & 1;
-                }
-            }
-            int32_t nb = r.i32();
-            uint8_t* vb = (uint8_t*)calloc(nb ? nb : 1, 1);
-            const void* src = r.bytes(nb);
-            if (src) memcpy(vb, src, (size_t)nb);
-            if (!r.ok || nb != offs[n]) {
-                free(offs);
-                free(vmask);
-                free(vb);
-                return seterr("deserialize: bad VARIABLE_WIDTH block");
-            }
-            out->cols[c].tag = PG_T_VARBIN;
-            out->cols[c].on_device = 0;
-            out->cols[c].data = vb;
-            out->cols[c].offsets = offs;
-            out->cols[c].null_mask = vmask;
-            continue;
         }
-        uint8_t may_null = r.u8();
-        uint8_t* mask = nullptr;
-        if (may_null) {
-            mask = (uint8_t*)calloc(n, 1);
-            for (int64_t base = 0; base < n; base += 8) {
-                uint8_t v = r.u8();
-                for (int64_t j = base; j < base + 8 && j < n; j++)
-                    mask[j] = (v >> (7 - (j - base))) & 1;
-            }
-        }
-        size_t esz = type_size(tag);
-        uint8_t* vals = (uint8_t*)calloc(n ? n : 1, esz);
-        if (!mask) {
-            const void* src = r.bytes((int64_t)n * esz);
-            if (src) memcpy(vals, src, (size_t)n * esz);
-        } else {
-            for (int64_t i = 0; i < n; i++)
-                if (!mask[i]) {
-                    const void* src = r.bytes(esz);
-                    if (src) memcpy(vals + i * esz, src, esz);
-                }
-        }
-        if (!r.ok) {
-            free(mask);
-            free(vals);
-            return seterr("deserialize: truncated block");
-        }
-        out->cols[c].tag = tag;
-        out->cols[c].on_device = 0;
-        out->cols[c].data = vals;
-        out->cols[c].null_mask = mask;
     }
     return PG_OK;
 }
@@ -5203,9 +5514,11 @@ extern "C" pg_status pg_page_free(pg_page* page)
         free(page->cols[c].data);
         free((void*)page->cols[c].null_mask);
         free((void*)page->cols[c].offsets);
+        free((void*)page->cols[c].dict_ids);
         page->cols[c].data = nullptr;
         page->cols[c].null_mask = nullptr;
         page->cols[c].offsets = nullptr;
+        page->cols[c].dict_ids = nullptr;
     }
     return PG_OK;
 }

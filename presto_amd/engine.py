@@ -12,7 +12,7 @@ _HERE = pathlib.Path(__file__).resolve().parent
 _SO = _HERE / "libpresto_gpu.so"
 
 # ---- enums (presto_gpu.h) ----
-T_U8, T_I32, T_I64, T_F64, T_VARBIN = 0, 1, 2, 3, 4
+T_U8, T_I32, T_I64, T_F64, T_VARBIN, T_I128 = 0, 1, 2, 3, 4, 5
 (CMP_LT, CMP_LE, CMP_GT, CMP_GE, CMP_EQ, CMP_NE, CMP_CONTAINS,
  CMP_PREFIX, CMP_CONTAINS2, CMP_NOT_CONTAINS2) = range(10)
 (PROJ_IDENT, PROJ_DISC_PRICE, PROJ_CHARGE, PROJ_MUL, PROJ_DIV,
