@@ -107,7 +107,7 @@ typedef struct {
 typedef struct {
     int64_t n_rows;
     int32_t n_cols;
-    pg_col cols[16];
+    pg_col cols[32];
 } pg_page;
 
 /* ---- plan blobs ----
@@ -133,7 +133,7 @@ typedef struct {
     int32_t op;    /* pg_cmp */
     int64_t ival;  /* compare value for integer columns */
     double dval;   /* compare value for f64 columns */
-    char sval[24]; /* VARBIN: compare bytes (EQ/NE/CONTAINS/PREFIX;
+    char sval[40]; /* VARBIN: compare bytes (EQ/NE/CONTAINS/PREFIX;
                       CONTAINS2 packs both patterns) */
     int32_t slen;
     int32_t rhs_col; /* 0: compare against the constant; k>0: compare

@@ -2726,6 +2726,69 @@ __global__ __launch_bounds__(256) void k_topn(const void* vals, int is_f64,
     }
 }
 
+/* large-limit TopN (limit > TOPN_MAXL): histogram preselect.
+ * Pass 1 histograms the 16 high bits of the order-preserving key
+ * transform; the host scans from the top bin until >= limit rows are
+ * covered; pass 2 collects every row at or above that bin's floor
+ * (wave-aggregated append), and the host does the exact final sort of
+ * the <= limit + bin-slop candidates — the same candidate-merge step
+ * the block-tournament path ends with. */
+__device__ inline uint64_t d_topn_sortable(int is_f64, int64_t bits)
+{
+    if (!is_f64) return (uint64_t)bits ^ 0x8000000000000000ull;
+    uint64_t u = (uint64_t)bits;
+    return (u & 0x8000000000000000ull) ? ~u : (u | 0x8000000000000000ull);
+}
+
+__global__ __launch_bounds__(256) void k_topn_hist(
+    const void* vals, int is_f64, int64_t n,
+    unsigned long long* hist /* [65536] */)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) {
+        int64_t b = is_f64
+                        ? __double_as_longlong(((const double*)vals)[i])
+                        : ((const int64_t*)vals)[i];
+        uint64_t s = d_topn_sortable(is_f64, b);
+        atomicAdd(&hist[(unsigned)(s >> 48)], 1ull);
+    }
+}
+
+__global__ __launch_bounds__(256) void k_topn_collect(
+    const void* vals, int is_f64, const int32_t* dates,
+    const int64_t* keys, int64_t n, uint64_t floor_sortable,
+    topn_cand* out, int64_t cap, unsigned long long* cursor)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < n; i += stride) {
+        int64_t b = is_f64
+                        ? __double_as_longlong(((const double*)vals)[i])
+                        : ((const int64_t*)vals)[i];
+        uint64_t s = d_topn_sortable(is_f64, b);
+        bool sel = s >= floor_sortable;
+        uint64_t m = d_ballot(sel);
+        int lane = threadIdx.x & 63;
+        unsigned long long base = 0;
+        if (lane == 0 && m)
+            base = atomicAdd(cursor, (unsigned long long)__popcll(m));
+        base = __shfl((long long)base, 0, 64);
+        if (sel) {
+            int64_t pos =
+                (int64_t)base + __popcll(m & ((1ull << lane) - 1));
+            if (pos < cap) {
+                topn_cand c;
+                c.val_bits = b;
+                c.date = dates[i];
+                c.key = keys[i];
+                c.valid = 1;
+                out[pos] = c;
+            }
+        }
+    }
+}
+
 /* ------------------------------------------------------------------ */
 /* partition split — PartitionedOutputOperator.partitionPage:394 /    */
 /* OptimizedPartitionedOutputOperator.java:86 columnar split;         */
@@ -2735,7 +2798,7 @@ __global__ __launch_bounds__(256) void k_topn(const void* vals, int is_f64,
 /* per-partition position lists (PartitioningExchanger.java:73).      */
 /* ------------------------------------------------------------------ */
 #define PART_NB 1024
-#define PART_MAXP 64
+#define PART_MAXP 256
 
 __device__ inline int d_part_id(const pg_page& pg, int key_col, int np,
                                 int64_t i)
@@ -4605,7 +4668,54 @@ struct TopNOp : Op {
         sp.stage(in);
         is_f64 = sp.pg.cols[plan.val_col].tag == PG_T_F64;
         int L = plan.limit;
-        if (L > TOPN_MAXL) throw std::runtime_error("limit > 16 unsupported");
+        if (L > (1 << 20))
+            throw std::runtime_error("limit > 1M unsupported");
+        if (L > TOPN_MAXL) {
+            /* histogram preselect (see k_topn_hist) */
+            DevBuf hist;
+            hist.alloc(65536 * 8);
+            hist.zero();
+            hipLaunchKernelGGL(k_topn_hist, dim3(2048), dim3(256), 0,
+                               g_stream, sp.pg.cols[plan.val_col].data,
+                               is_f64 ? 1 : 0, sp.pg.n_rows,
+                               (unsigned long long*)hist.p);
+            std::vector<unsigned long long> h(65536);
+            CHKV(hipMemcpyAsync(h.data(), hist.p, 65536 * 8,
+                                hipMemcpyDeviceToHost, g_stream));
+            CHKV(hipStreamSynchronize(g_stream));
+            int64_t cum = 0;
+            int bin = 65535;
+            for (; bin >= 0; bin--) {
+                cum += (int64_t)h[bin];
+                if (cum >= L) break;
+            }
+            if (bin < 0) bin = 0;
+            uint64_t floor_sortable = (uint64_t)bin << 48;
+            DevBuf cand, cur;
+            cand.alloc((size_t)(cum ? cum : 1) * sizeof(topn_cand));
+            cur.alloc(8);
+            cur.zero();
+            hipLaunchKernelGGL(k_topn_collect, dim3(2048), dim3(256), 0,
+                               g_stream, sp.pg.cols[plan.val_col].data,
+                               is_f64 ? 1 : 0,
+                               (const int32_t*)
+                                   sp.pg.cols[plan.date_col].data,
+                               (const int64_t*)
+                                   sp.pg.cols[plan.key_col].data,
+                               sp.pg.n_rows, floor_sortable,
+                               (topn_cand*)cand.p, cum,
+                               (unsigned long long*)cur.p);
+            std::vector<topn_cand> hc(cum);
+            if (cum)
+                CHKV(hipMemcpyAsync(hc.data(), cand.p,
+                                    (size_t)cum * sizeof(topn_cand),
+                                    hipMemcpyDeviceToHost, g_stream));
+            CHKV(hipStreamSynchronize(g_stream));
+            for (auto& c : hc)
+                if (c.valid)
+                    cands.push_back({c.val_bits, c.key, c.date, is_f64});
+            return;
+        }
         DevBuf bout;
         bout.alloc((size_t)TOPN_NB * L * sizeof(topn_cand));
         hipLaunchKernelGGL(k_topn, dim3(TOPN_NB), dim3(256), 0, g_stream,
@@ -5488,7 +5598,7 @@ extern "C" pg_status pg_page_deserialize(const void* buf, int64_t len,
         return seterr("deserialize: bad sizes");
     }
     int32_t n_blocks = body.i32();
-    if (n_blocks < 0 || n_blocks > 16)
+    if (n_blocks < 0 || n_blocks > 32)
         return seterr("deserialize: unsupported block count");
     memset(out, 0, sizeof(*out));
     out->n_rows = pos_count;

@@ -36,12 +36,12 @@ class PgCol(C.Structure):
 
 class PgPage(C.Structure):
     _fields_ = [("n_rows", C.c_int64), ("n_cols", C.c_int32),
-                ("cols", PgCol * 16)]
+                ("cols", PgCol * 32)]
 
 
 class Pred(C.Structure):
     _fields_ = [("col", C.c_int32), ("op", C.c_int32), ("ival", C.c_int64),
-                ("dval", C.c_double), ("sval", C.c_char * 24),
+                ("dval", C.c_double), ("sval", C.c_char * 40),
                 ("slen", C.c_int32), ("rhs_col", C.c_int32)]
 
 
