@@ -1483,7 +1483,7 @@ def q9(part: Page, supp: Page, orders: Page, ps: Page, li: Page):
     return profit
 
 
-def q13(n_cust: int, orders: Page, max_count=64):
+def q13(n_cust: int, orders, max_count=64):
     """Q13 customer distribution (q13.sql): the NOT LIKE
     '%special%requests%' comment filter runs as the ordered
     two-substring VARBIN predicate (CONTAINS2 negated); per-customer
@@ -1491,21 +1491,16 @@ def q13(n_cust: int, orders: Page, max_count=64):
     table; the count histogram is a second fused-agg probe keyed by the
     count value.  The LEFT OUTER zero bucket is n_cust minus the
     customers with qualifying orders.  Returns [(c_count, custdist)]
-    sorted (custdist desc, c_count desc)."""
+    sorted (custdist desc, c_count desc).
+
+    orders: one Page or an iterable of Pages — the o_comment
+    VariableWidthBlock has int32 offsets (the reference's Slice cap), so
+    beyond ~SF20 the orders table arrives as MULTIPLE pages, exactly as
+    Presto's Driver would feed them; every operator here accumulates
+    across addInput calls."""
     import numpy as np
     from .engine import lib
-
-    fp = PlanFilterProject()
-    fp.n_preds = 1
-    pr = Pred(orders.channel("comment"), CMP_NOT_CONTAINS2, 8, 0.0)
-    pr.sval = b"special" + b"requests"
-    pr.slen = 7
-    fp.preds[0] = pr
-    fp.n_proj = 1
-    fp.proj[0] = Proj(PROJ_IDENT, orders.channel("custkey"), 0, 0)
-    f = Operator(OP_FILTER_PROJECT, fp)
-    f.add_input(orders)
-    okp = f.get_output_raw()  # [custkey] of qualifying orders
+    order_pages = [orders] if isinstance(orders, Page) else list(orders)
 
     bc = PlanHashBuild()
     bc.key_col = 0
@@ -1525,7 +1520,19 @@ def q13(n_cust: int, orders: Page, max_count=64):
     jp.dec_scale = 0
     jp.dec_only = 1
     jo = Operator(OP_LOOKUP_JOIN, jp)
-    jo.add_input_raw(okp)
+    fp = PlanFilterProject()
+    fp.n_preds = 1
+    pr = Pred(order_pages[0].channel("comment"), CMP_NOT_CONTAINS2, 8, 0.0)
+    pr.sval = b"special" + b"requests"
+    pr.slen = 7
+    fp.preds[0] = pr
+    fp.n_proj = 1
+    fp.proj[0] = Proj(PROJ_IDENT, order_pages[0].channel("custkey"), 0, 0)
+    f = Operator(OP_FILTER_PROJECT, fp)
+    for opg in order_pages:
+        f.add_input(opg)
+        okp = f.get_output_raw()  # [custkey] of qualifying orders
+        jo.add_input_raw(okp)
     jo.finish()
     groups = jo.get_output_raw()  # [custkey, sum, f64, cnt]
     n_with_orders = groups.n_rows

@@ -116,22 +116,34 @@ def main():
         del pages
 
     if "q13" in want:
-        # the o_comment VariableWidthBlock has int32 offsets: one block
-        # caps near SF24 (2^31 bytes); measure q13 at <= SF20
+        # the o_comment VariableWidthBlock has int32 offsets (the
+        # reference Slice cap), so at full SF the orders table arrives
+        # as MULTIPLE pages of <= 30M orders — Driver-style paging, no
+        # scale cap
         from presto_amd.engine import Varbin
-        sf13 = min(sf, 20.0)
-        orders13 = orc.gen_orders(sf13) if sf13 != sf else orders
-        data, offs = orc.gen_orders_comment_varbin(sf13)
-        cm = Varbin.__new__(Varbin)
-        cm.data, cm.offsets, cm.n = data, offs, len(orders13["custkey"])
-        n_cust = int(150000 * sf13)
-        pages = (P.Page({"custkey": dev(orders13["custkey"]),
-                         "comment": cm}),)
-        got, secs = run("q13", lambda: P.pipelines.q13(n_cust, *pages),
+        n_ord = len(orders["custkey"])
+        chunk = 30_000_000
+        off_all, ln_all = orc.gen_orders_comment(sf)
+        pool = np.frombuffer(orc.text_pool(), np.uint8)
+        opages = []
+        for a in range(0, n_ord, chunk):
+            b = min(a + chunk, n_ord)
+            ln = ln_all[a:b]
+            offs = np.zeros(b - a + 1, np.int32)
+            np.cumsum(ln, out=offs[1:])
+            starts = np.repeat(off_all[a:b] -
+                               offs[:-1].astype(np.int64), ln)
+            data = pool[starts + np.arange(int(offs[-1]))]
+            cm = Varbin.__new__(Varbin)
+            cm.data, cm.offsets, cm.n = data, offs, b - a
+            opages.append(P.Page({"custkey": dev(orders["custkey"][a:b]),
+                                  "comment": cm}))
+        n_cust = int(150000 * sf)
+        got, secs = run("q13", lambda: P.pipelines.q13(n_cust, opages),
                         args.reps)
-        exp = orc.q13(sf13, orders13)
-        record("q13", secs, got == exp, {"sf": sf13})
-        del pages, data, offs
+        exp = orc.q13(sf, orders)
+        record("q13", secs, got == exp, {"sf": sf})
+        del opages
 
     if "q15" in want:
         pages = (page({"suppkey": supp["suppkey"],
