@@ -289,12 +289,21 @@ class Bench:
         for _ in range(a.warmup):
             step()
         self.lib.c.pg_hot_reset()
+        if self.world > 1:
+            from presto_amd import dist as pdist
+            pdist.exchange_bytes_reset()
         self.barrier_sync()
         t0 = time.time()
         for _ in range(a.steps):
             step()
         self.barrier_sync()
         elapsed = time.time() - t0
+        if self.world > 1:
+            from presto_amd import dist as pdist
+            log(f"rank {self.rank}: RCCL all_to_all xGMI-crossing volume "
+                f"{pdist.EXCHANGE_BYTES / a.steps / 1e9:.3f} GB/step "
+                f"(SURVEY.md §8d config-4 scale estimate: ~10.8 GB total "
+                f"at SF300 Q3 across 8 ranks)")
         if self.world > 1:
             import torch.distributed as dist
             e = self.torch.tensor([elapsed], device=self.device)

@@ -8,6 +8,16 @@ hop.  Backend-agnostic (gloo on CPU for tests)."""
 import torch
 import torch.distributed as dist
 
+# cumulative bytes this process moved through all_to_all exchanges — the
+# xGMI-crossing volume (read by bench.py; compare against SURVEY.md §8d
+# config 4's estimate, e.g. ~10.8 GB total for SF300 Q3)
+EXCHANGE_BYTES = 0
+
+
+def exchange_bytes_reset():
+    global EXCHANGE_BYTES
+    EXCHANGE_BYTES = 0
+
 
 def exchange_split_counts(send_counts):
     """all_to_all of row counts: returns recv_counts (rows arriving from
@@ -29,6 +39,8 @@ def exchange_columns(cols, send_counts, recv_counts=None, device=None):
         recv_counts = exchange_split_counts(send_counts)
     out = {}
     n_recv = sum(recv_counts)
+    global EXCHANGE_BYTES
+    me = dist.get_rank()
     for name, t in cols.items():
         r = torch.empty(n_recv, dtype=t.dtype,
                         device=device if device is not None else t.device)
@@ -36,6 +48,8 @@ def exchange_columns(cols, send_counts, recv_counts=None, device=None):
                                output_split_sizes=recv_counts,
                                input_split_sizes=list(send_counts))
         out[name] = r
+        # bytes leaving this rank for OTHER ranks (xGMI-crossing)
+        EXCHANGE_BYTES += (t.numel() - send_counts[me]) * t.element_size()
     return out
 
 
