@@ -1608,6 +1608,11 @@ __device__ inline int64_t d_tbl_find_tagged(const int64_t* keys,
 
 /* per-slot accumulators, interleaved so one probe hit touches ONE cache
  * line (32 B of one 64-B line) instead of four separate arrays */
+/* per-slot accumulators: 4 words [dec ticks, fx128 lo, fx128 hi, cnt],
+ * or the SLIM 2-word layout [dec ticks, cnt] when every probe of the
+ * table is dec_only/dec_min (integer consumers — halves the zeroing and
+ * extraction traffic).  Kernels take the word stride (aw); cnt is the
+ * last word. */
 struct slot_acc {
     unsigned long long dec;  /* exact decimal ticks */
     unsigned long long flo;  /* 64.64 fixed-point f64 sum, low word */
@@ -1635,7 +1640,7 @@ __device__ inline void d_atomic_add_dec_ck(unsigned long long* slot,
 __global__ __launch_bounds__(256) void k_probe_agg(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
     const uint8_t* tags, int64_t mask, int64_t lmask, int32_t pbits,
-    slot_acc* acc, unsigned long long* ovf)
+    unsigned long long* acc, int32_t aw, unsigned long long* ovf)
 {
     /* 4 CONSECUTIVE rows per thread with run dedup: neighboring rows
      * often share the join key (lineitem is clustered by orderkey), so
@@ -1656,18 +1661,18 @@ __global__ __launch_bounds__(256) void k_probe_agg(
         int run_cnt = 0;
         auto flush = [&]() {
             if (cur_slot < 0 || run_cnt == 0) return;
+            unsigned long long* s = acc + (size_t)cur_slot * aw;
             if (plan.dec_min) {
-                atomicMin((long long*)&acc[cur_slot].dec, run_min);
+                atomicMin((long long*)&s[0], run_min);
             } else if (run_dec) {
-                d_atomic_add_dec_ck(&acc[cur_slot].dec, run_dec, ovf);
+                d_atomic_add_dec_ck(&s[0], run_dec, ovf);
             }
             if (!plan.dec_only && !plan.dec_min && (run_flo | run_fhi)) {
-                unsigned long long old =
-                    atomicAdd(&acc[cur_slot].flo, run_flo);
-                atomicAdd(&acc[cur_slot].fhi,
+                unsigned long long old = atomicAdd(&s[1], run_flo);
+                atomicAdd(&s[2],
                           run_fhi + (old > ~run_flo ? 1ull : 0ull));
             }
-            atomicAdd(&acc[cur_slot].cnt, (unsigned long long)run_cnt);
+            atomicAdd(&s[aw - 1], (unsigned long long)run_cnt);
         };
         for (int64_t i = base; i < lim; i++) {
             if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
@@ -1721,7 +1726,7 @@ __global__ __launch_bounds__(256) void k_probe_agg_q3(
     int32_t pred_val, const int64_t* okey, const double* ep,
     const double* dc, int64_t n, const int64_t* keys, const uint8_t* tags,
     int64_t mask, int64_t lmask, int32_t pbits, int32_t dec_only,
-    slot_acc* acc, unsigned long long* ovf)
+    unsigned long long* acc, int32_t aw, unsigned long long* ovf)
 {
     typedef int vi2 __attribute__((ext_vector_type(2)));
     typedef long vl2 __attribute__((ext_vector_type(2)));
@@ -1797,15 +1802,16 @@ __global__ __launch_bounds__(256) void k_probe_agg_q3(
              * far beyond a page.  The generic k_probe_agg keeps the
              * checked add. */
             (void)ovf;
-            atomicAdd(&acc[sl].dec, (unsigned long long)ticks);
+            unsigned long long* s = acc + (size_t)sl * aw;
+            atomicAdd(&s[0], (unsigned long long)ticks);
             if (!dec_only) {
                 double pr = e * (1.0 - d);
                 uint64_t phi, plo;
                 fx128_from_f64(pr, &phi, &plo);
-                unsigned long long old = atomicAdd(&acc[sl].flo, plo);
-                atomicAdd(&acc[sl].fhi, phi + (old > ~plo ? 1ull : 0ull));
+                unsigned long long old = atomicAdd(&s[1], plo);
+                atomicAdd(&s[2], phi + (old > ~plo ? 1ull : 0ull));
             }
-            atomicAdd(&acc[sl].cnt, 1ull);
+            atomicAdd(&s[aw - 1], 1ull);
         }
     }
 }
@@ -1818,8 +1824,8 @@ __global__ __launch_bounds__(256) void k_probe_agg_pay(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys1,
     const uint8_t* tags1, const int32_t* head1, int64_t mask1,
     int64_t lmask1, const int64_t* pay1, const int64_t* keys2,
-    const uint8_t* tags2, int64_t mask2, int64_t lmask2, slot_acc* acc2,
-    unsigned long long* ovf)
+    const uint8_t* tags2, int64_t mask2, int64_t lmask2,
+    unsigned long long* acc2, int32_t aw, unsigned long long* ovf)
 {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -1839,29 +1845,29 @@ __global__ __launch_bounds__(256) void k_probe_agg_pay(
         ag.proj = plan.proj;
         ag.dec_scale = plan.dec_scale;
         int64_t ticks = d_eval_proj_dec(pg, ag, i);
-        d_atomic_add_dec_ck(&acc2[s2].dec, ticks, ovf);
+        unsigned long long* s = acc2 + (size_t)s2 * aw;
+        d_atomic_add_dec_ck(&s[0], ticks, ovf);
         if (!plan.dec_only) {
             double p = d_eval_proj_f64(pg, plan.proj, i);
             uint64_t phi, plo;
             fx128_from_f64(p, &phi, &plo);
-            unsigned long long old = atomicAdd(&acc2[s2].flo, plo);
-            atomicAdd(&acc2[s2].fhi, phi + (old > ~plo ? 1ull : 0ull));
+            unsigned long long old = atomicAdd(&s[1], plo);
+            atomicAdd(&s[2], phi + (old > ~plo ? 1ull : 0ull));
         }
-        atomicAdd(&acc2[s2].cnt, 1ull);
+        atomicAdd(&s[aw - 1], 1ull);
     }
 }
 
 /* MIN-accumulator identity init (dec = +inf sentinel, rest zero) */
-__global__ __launch_bounds__(256) void k_acc_min_init(slot_acc* acc,
-                                                      int64_t cap)
+__global__ __launch_bounds__(256) void k_acc_min_init(
+    unsigned long long* acc, int32_t aw, int64_t cap)
 {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; i < cap; i += stride) {
-        acc[i].dec = 0x7f7f7f7f7f7f7f7fULL;
-        acc[i].fhi = 0;
-        acc[i].flo = 0;
-        acc[i].cnt = 0;
+        unsigned long long* s = acc + (size_t)i * aw;
+        s[0] = 0x7f7f7f7f7f7f7f7fULL;
+        for (int w = 1; w < aw; w++) s[w] = 0;
     }
 }
 
@@ -2547,11 +2553,11 @@ __global__ __launch_bounds__(256) void k_groups_count(
 }
 
 __global__ __launch_bounds__(256) void k_groups_emit(
-    const int64_t* keys, const int32_t* head, const slot_acc* acc,
-    build_payloads bp, int64_t cap,
-    int64_t chunk, const int64_t* block_offs, int64_t* out_key,
-    emit_outs payload_outs, int64_t* out_dec, double* out_f64,
-    int64_t* out_cnt)
+    const int64_t* keys, const int32_t* head,
+    const unsigned long long* acc, int32_t aw, build_payloads bp,
+    int64_t cap, int64_t chunk, const int64_t* block_offs,
+    int64_t* out_key, emit_outs payload_outs, int64_t* out_dec,
+    double* out_f64, int64_t* out_cnt)
 {
     const int64_t lo = (int64_t)blockIdx.x * chunk;
     const int64_t hi = min(lo + chunk, cap);
@@ -2562,7 +2568,7 @@ __global__ __launch_bounds__(256) void k_groups_emit(
     __syncthreads();
     for (int64_t base = lo; base < hi; base += 256) {
         int64_t i = base + 64 * wid + lane;
-        bool sel = i < hi && acc[i].cnt > 0;
+        bool sel = i < hi && acc[i * aw + aw - 1] > 0;
         uint64_t m = d_ballot(sel);
         int wsum = __popcll(m);
         if (lane == 0) wcnt[wid] = wsum;
@@ -2602,9 +2608,10 @@ __global__ __launch_bounds__(256) void k_groups_emit(
                             ((const double*)bp.ptr[o])[r];
                 }
             }
-            out_dec[pos] = (int64_t)acc[i].dec;
-            out_f64[pos] = fx128_to_f64(acc[i].fhi, acc[i].flo);
-            out_cnt[pos] = (int64_t)acc[i].cnt;
+            const unsigned long long* s = acc + (size_t)i * aw;
+            out_dec[pos] = (int64_t)s[0];
+            out_f64[pos] = aw == 4 ? fx128_to_f64(s[2], s[1]) : 0.0;
+            out_cnt[pos] = (int64_t)s[aw - 1];
         }
         __syncthreads();
         if (threadIdx.x == 0)
@@ -3072,7 +3079,8 @@ struct Table {
     int32_t pack_bits = 0;  /* slot word = key << pack_bits | payload0 */
     bool key_set_only = false;
     DevBuf keys, head, next, tags;
-    DevBuf acc; /* slot_acc[cap], interleaved */
+    DevBuf acc; /* per-slot accumulators, acc_words u64 each */
+    int32_t acc_words = 0; /* 4 = [dec, flo, fhi, cnt]; 2 = [dec, cnt] */
     DevBuf acc_multi; /* multi-agg probes: (n_acc+1) u64 per slot */
     int32_t n_acc = 0;
     bool slot_payloads = false; /* payloads indexed by slot (agg tables) */
@@ -4075,16 +4083,29 @@ struct JoinOp : Op {
                 t->acc_multi.zero();
                 CHKV(hipStreamSynchronize(g_stream));
             }
-        } else if (plan.mode == 1 && !t->acc.p) {
-            t->acc.alloc((size_t)t->cap * sizeof(slot_acc));
-            t->acc.zero();
-            CHKV(hipStreamSynchronize(g_stream));
+        } else if (plan.mode == 1) {
+            /* slim 2-word accumulators when this consumer reads only the
+             * tick sum + count (dec_only / dec_min) — halves the zeroing
+             * and extraction traffic; a later fx128 consumer of the same
+             * table errors out loudly */
+            int32_t want_aw = (plan.dec_only || plan.dec_min) ? 2 : 4;
+            if (!t->acc.p) {
+                t->acc_words = want_aw;
+                t->acc.alloc((size_t)t->cap * t->acc_words * 8);
+                t->acc.zero();
+                CHKV(hipStreamSynchronize(g_stream));
+            } else if (want_aw == 4 && t->acc_words == 2) {
+                throw std::runtime_error(
+                    "table accumulators were allocated dec-only; fx128 "
+                    "probes need a fresh table");
+            }
         }
         if (plan.mode == 1 && plan.dec_min) {
             /* MIN identity: dec = large positive sentinel; cnt
              * distinguishes matched groups */
             hipLaunchKernelGGL(k_acc_min_init, dim3(2048), dim3(256), 0,
-                               g_stream, (slot_acc*)t->acc.p, t->cap);
+                               g_stream, (unsigned long long*)t->acc.p,
+                               t->acc_words, t->cap);
             CHKV(hipStreamSynchronize(g_stream));
         }
         if (plan.mode == 2) {
@@ -4118,7 +4139,8 @@ struct JoinOp : Op {
                 throw std::runtime_error(
                     "mode 3 groups table cannot be packed");
             if (!t2->acc.p) {
-                t2->acc.alloc((size_t)t2->cap * sizeof(slot_acc));
+                t2->acc_words = 4;
+                t2->acc.alloc((size_t)t2->cap * 4 * 8);
                 t2->acc.zero();
                 CHKV(hipStreamSynchronize(g_stream));
             }
@@ -4184,7 +4206,8 @@ struct JoinOp : Op {
                                (const int64_t*)t2->keys.p,
                                (const uint8_t*)t2->tags.p, t2->mask,
                                t2->local_mask,
-                               (slot_acc*)t2->acc.p,
+                               (unsigned long long*)t2->acc.p,
+                               t2->acc_words,
                                (unsigned long long*)ovf.p);
             hot_end();
             CHKV(hipStreamSynchronize(g_stream));
@@ -4239,7 +4262,8 @@ struct JoinOp : Op {
                     (const double*)sp.pg.cols[plan.proj.b].data,
                     sp.pg.n_rows, (const int64_t*)t->keys.p,
                     (const uint8_t*)t->tags.p, t->mask, t->local_mask,
-                    t->pack_bits, plan.dec_only, (slot_acc*)t->acc.p,
+                    t->pack_bits, plan.dec_only,
+                    (unsigned long long*)t->acc.p, t->acc_words,
                     (unsigned long long*)ovf.p);
             } else {
                 hipLaunchKernelGGL(k_probe_agg, dim3(4096), dim3(256), 0,
@@ -4247,7 +4271,8 @@ struct JoinOp : Op {
                                    (const int64_t*)t->keys.p,
                                    (const uint8_t*)t->tags.p, t->mask,
                                    t->local_mask, t->pack_bits,
-                                   (slot_acc*)t->acc.p,
+                                   (unsigned long long*)t->acc.p,
+                                   t->acc_words,
                                    (unsigned long long*)ovf.p);
             }
             hot_end();
@@ -4403,7 +4428,8 @@ struct JoinOp : Op {
             k_groups_count, dim3(FLT_NB), dim3(256), 0, g_stream,
             multi ? (const unsigned long long*)gt->acc_multi.p
                   : (const unsigned long long*)gt->acc.p,
-            multi ? plan.n_aggs + 1 : 4, multi ? plan.n_aggs : 3, cap,
+            multi ? plan.n_aggs + 1 : gt->acc_words,
+            multi ? plan.n_aggs : gt->acc_words - 1, cap,
             chunk, (int64_t*)d_counts.p);
         std::vector<int64_t> h(FLT_NB);
         CHKV(hipMemcpyAsync(h.data(), d_counts.p, FLT_NB * 8,
@@ -4471,7 +4497,8 @@ struct JoinOp : Op {
         hipLaunchKernelGGL(k_groups_emit, dim3(FLT_NB), dim3(256), 0,
                            g_stream, (const int64_t*)gt->keys.p,
                            (const int32_t*)gt->head.p,
-                           (const slot_acc*)gt->acc.p, bp, cap,
+                           (const unsigned long long*)gt->acc.p,
+                           gt->acc_words, bp, cap,
                            chunk, (const int64_t*)d_offs.p,
                            (int64_t*)op.pg.cols[c_key].data, pl_outs,
                            (int64_t*)op.pg.cols[c_dec].data,
