@@ -1803,3 +1803,100 @@ def test_many_partitions(P):
         assert (_partition_math(np.asarray(ks), npart) == p).all()
         # stable: source order preserved within a partition
         assert np.all(np.diff(vs) > 0)
+
+
+def test_multi_acc_probe_fuzz(P):
+    """Multi-accumulator probe (mode 1, n_aggs with FILTER masks) vs a
+    numpy restatement — the InMemoryHashAggregationBuilder +
+    AggregationNode-mask analog used by q21."""
+    rng = np.random.RandomState(21)
+    nb = 5_000
+    bk = rng.permutation(100_000)[:nb].astype(np.int64) + 1
+    bp = P.PlanHashBuild()
+    bp.key_col = 0
+    bp.semijoin_table = -1
+    bp.capacity_hint = nb
+    bp.agg_table = 1
+    b = P.Operator(P.OP_HASH_BUILD, bp)
+    b.add_input(P.Page({"k": bk}))
+    b.finish()
+    n = 300_000
+    pk = bk[rng.randint(0, nb, n)]
+    pk[::3] += 100_001  # a third miss
+    v = rng.randint(-50, 50, n).astype(np.int64)
+    flag = rng.randint(0, 3, n).astype(np.int64)
+    jp = P.PlanLookupJoin()
+    jp.table = b.table()
+    jp.key_col = 0
+    jp.mode = 1
+    jp.n_preds = 0
+    jp.preds[0] = P.Pred(2, P.CMP_EQ, 1, 0.0)   # flag == 1
+    jp.preds[1] = P.Pred(1, P.CMP_GT, 0, 0.0)   # v > 0
+    jp.n_aggs = 3
+    jp.aggs[0] = P.Agg(P.AGG_SUM_I64, P.Proj(P.PROJ_IDENT, 1, 0, 0), 0)
+    jp.aggs[1] = P.Agg(P.AGG_COUNT, P.Proj(P.PROJ_IDENT, 0, 0, 0), 0)
+    jp.aggs[2] = P.Agg(P.AGG_SUM_I64, P.Proj(P.PROJ_MUL, 1, 1, 0), 0)
+    for i, f in enumerate((-1, 0, 1)):
+        jp.agg_filter[i] = f
+    j = P.Operator(P.OP_LOOKUP_JOIN, jp)
+    j.add_input(P.Page({"k": pk, "v": v, "f": flag}))
+    j.finish()
+    out = j.get_output(["k", "s", "c1", "sq", "cnt"])
+    j.destroy()
+    from presto_amd.engine import lib
+    lib().c.pg_table_destroy(b.table())
+    b.destroy()
+    import collections
+    ref = collections.defaultdict(lambda: [0, 0, 0, 0])
+    bset = set(bk.tolist())
+    for k, vv, ff in zip(pk.tolist(), v.tolist(), flag.tolist()):
+        if k not in bset:
+            continue
+        r = ref[k]
+        r[0] += vv
+        if ff == 1:
+            r[1] += 1
+        if vv > 0:
+            r[2] += vv * vv
+        r[3] += 1
+    assert len(out["k"]) == len(ref)
+    for i in range(len(out["k"])):
+        r = ref[int(out["k"][i])]
+        assert [int(out["s"][i]), int(out["c1"][i]), int(out["sq"][i]),
+                int(out["cnt"][i])] == r
+
+
+def test_groupby_capacity_error(P):
+    """GroupBy past its declared capacity fails loudly, never silently."""
+    n = 50_000
+    keys = np.arange(n, dtype=np.int64)
+    plan = P.PlanGroupBy()
+    plan.n_keys = 1
+    plan.key_col[0] = 0
+    plan.capacity_hint = 100  # far too small
+    plan.n_aggs = 1
+    plan.aggs[0] = P.Agg(P.AGG_COUNT, P.Proj(P.PROJ_IDENT, 0, 0, 0), 0)
+    plan.agg_filter[0] = -1
+    op = P.Operator(P.OP_GROUPBY_MULTI, plan)
+    with pytest.raises(RuntimeError,
+                       match="full|capacity|fill"):
+        op.add_input(P.Page({"k": keys}))
+        op.finish()
+        op.get_output()
+    op.destroy()
+
+
+def test_groupby_empty_input(P):
+    plan = P.PlanGroupBy()
+    plan.n_keys = 1
+    plan.key_col[0] = 0
+    plan.capacity_hint = 64
+    plan.n_aggs = 1
+    plan.aggs[0] = P.Agg(P.AGG_COUNT, P.Proj(P.PROJ_IDENT, 0, 0, 0), 0)
+    plan.agg_filter[0] = -1
+    op = P.Operator(P.OP_GROUPBY_MULTI, plan)
+    op.add_input(P.Page({"k": np.empty(0, np.int64)}))
+    op.finish()
+    out = op.get_output(["k", "c", "cnt"])
+    op.destroy()
+    assert len(out["k"]) == 0
