@@ -2258,7 +2258,7 @@ __global__ __launch_bounds__(256) void k_probe_agg_multi(
 __global__ __launch_bounds__(256) void k_groups_emit_multi(
     const int64_t* keys, const unsigned long long* acc, int64_t stride_w,
     int32_t n_aggs, build_payloads bp, int64_t cap, int64_t chunk,
-    const int64_t* block_offs, int64_t* out_key, emit_outs payload_outs,
+    unsigned long long* cursor, int64_t* out_key, emit_outs payload_outs,
     emit_outs agg_outs)
 {
     const int64_t lo = (int64_t)blockIdx.x * chunk;
@@ -2266,7 +2266,20 @@ __global__ __launch_bounds__(256) void k_groups_emit_multi(
     const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
     __shared__ int64_t wcnt[4];
     __shared__ int64_t running;
-    if (threadIdx.x == 0) running = block_offs[blockIdx.x];
+    {
+        int64_t mycnt = 0;
+        for (int64_t i = lo + threadIdx.x; i < hi; i += 256)
+            mycnt += acc[i * stride_w + n_aggs] > 0;
+        mycnt = d_bfly_i64(mycnt);
+        if (lane == 0) wcnt[wid] = mycnt;
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            int64_t t = wcnt[0] + wcnt[1] + wcnt[2] + wcnt[3];
+            running = t ? (int64_t)atomicAdd(cursor,
+                                             (unsigned long long)t)
+                        : 0;
+        }
+    }
     __syncthreads();
     for (int64_t base = lo; base < hi; base += 256) {
         int64_t i = base + 64 * wid + lane;
@@ -2401,6 +2414,7 @@ __global__ __launch_bounds__(256) void k_groupby_multi(
                     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
                     __hip_atomic_store(&state[pos], 2u, __ATOMIC_RELAXED,
                                        __HIP_MEMORY_SCOPE_SYSTEM);
+                    atomicAdd(counters + 2, 1ull); /* distinct groups */
                     slot = pos;
                     break;
                 }
@@ -2476,7 +2490,7 @@ __global__ __launch_bounds__(256) void k_gb_acc_init(
 __global__ __launch_bounds__(256) void k_groupby_emit(
     const int64_t* kv, int64_t cap, int32_t n_keys,
     const unsigned long long* acc, int32_t acc_words, gb_agg_off offs,
-    pg_plan_groupby plan, int64_t chunk, const int64_t* block_offs,
+    pg_plan_groupby plan, int64_t chunk, unsigned long long* cursor,
     emit_outs key_outs, emit_outs agg_outs)
 {
     const int64_t lo = (int64_t)blockIdx.x * chunk;
@@ -2484,7 +2498,20 @@ __global__ __launch_bounds__(256) void k_groupby_emit(
     const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
     __shared__ int64_t wcnt[4];
     __shared__ int64_t running;
-    if (threadIdx.x == 0) running = block_offs[blockIdx.x];
+    {
+        int64_t mycnt = 0;
+        for (int64_t i = lo + threadIdx.x; i < hi; i += 256)
+            mycnt += acc[i * acc_words + acc_words - 1] > 0;
+        mycnt = d_bfly_i64(mycnt);
+        if (lane == 0) wcnt[wid] = mycnt;
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            int64_t t = wcnt[0] + wcnt[1] + wcnt[2] + wcnt[3];
+            running = t ? (int64_t)atomicAdd(cursor,
+                                             (unsigned long long)t)
+                        : 0;
+        }
+    }
     __syncthreads();
     for (int64_t base = lo; base < hi; base += 256) {
         int64_t i = base + 64 * wid + lane;
@@ -2552,10 +2579,15 @@ __global__ __launch_bounds__(256) void k_groups_count(
         block_counts[blockIdx.x] = lds[0] + lds[1] + lds[2] + lds[3];
 }
 
+/* ONE pass per block: count the block's chunk (first sweep, lines land
+ * in L2), reserve a contiguous span with one atomicAdd, then emit into
+ * it (second sweep hits L2).  Group output order is block-interleaved —
+ * result-set semantics, like the reference's parallel drivers; every
+ * consumer (TopN, joins, host folds) is order-free. */
 __global__ __launch_bounds__(256) void k_groups_emit(
     const int64_t* keys, const int32_t* head,
     const unsigned long long* acc, int32_t aw, build_payloads bp,
-    int64_t cap, int64_t chunk, const int64_t* block_offs,
+    int64_t cap, int64_t chunk, unsigned long long* cursor,
     int64_t* out_key, emit_outs payload_outs, int64_t* out_dec,
     double* out_f64, int64_t* out_cnt)
 {
@@ -2564,7 +2596,22 @@ __global__ __launch_bounds__(256) void k_groups_emit(
     const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
     __shared__ int64_t wcnt[4];
     __shared__ int64_t running;
-    if (threadIdx.x == 0) running = block_offs[blockIdx.x];
+    {
+        int64_t mycnt = 0;
+        for (int64_t i = lo + threadIdx.x; i < hi; i += 256)
+            mycnt += acc[i * aw + aw - 1] > 0;
+        mycnt = d_bfly_i64(mycnt);
+        if (lane == 0) wcnt[wid] = mycnt;
+        __syncthreads();
+        if (threadIdx.x == 0)
+            running = wcnt[0] + wcnt[1] + wcnt[2] + wcnt[3]
+                          ? (int64_t)atomicAdd(
+                                cursor, (unsigned long long)(wcnt[0] +
+                                                             wcnt[1] +
+                                                             wcnt[2] +
+                                                             wcnt[3]))
+                          : 0;
+    }
     __syncthreads();
     for (int64_t base = lo; base < hi; base += 256) {
         int64_t i = base + 64 * wid + lane;
@@ -4422,29 +4469,13 @@ struct JoinOp : Op {
         int64_t chunk = (cap + FLT_NB - 1) / FLT_NB;
         chunk = (chunk + 255) / 256 * 256;
         if (chunk < 256) chunk = 256;
-        DevBuf d_counts;
-        d_counts.alloc(FLT_NB * 8);
-        hipLaunchKernelGGL(
-            k_groups_count, dim3(FLT_NB), dim3(256), 0, g_stream,
-            multi ? (const unsigned long long*)gt->acc_multi.p
-                  : (const unsigned long long*)gt->acc.p,
-            multi ? plan.n_aggs + 1 : gt->acc_words,
-            multi ? plan.n_aggs : gt->acc_words - 1, cap,
-            chunk, (int64_t*)d_counts.p);
-        std::vector<int64_t> h(FLT_NB);
-        CHKV(hipMemcpyAsync(h.data(), d_counts.p, FLT_NB * 8,
-                            hipMemcpyDeviceToHost, g_stream));
-        CHKV(hipStreamSynchronize(g_stream));
-        int64_t total = 0;
-        for (int b = 0; b < FLT_NB; b++) {
-            int64_t v = h[b];
-            h[b] = total;
-            total += v;
-        }
-        DevBuf d_offs;
-        d_offs.alloc(FLT_NB * 8);
-        CHKV(hipMemcpyAsync(d_offs.p, h.data(), FLT_NB * 8,
-                            hipMemcpyHostToDevice, g_stream));
+        /* single-pass extraction: groups <= inserted build rows, so the
+         * output buffers are sized at that bound and the emit kernel
+         * reserves per-block spans off one cursor (n_rows read back) */
+        int64_t total = gt->n_rows;
+        DevBuf d_cursor;
+        d_cursor.alloc(8);
+        d_cursor.zero();
         OutPage op;
         op.pg.n_rows = total;
         int nc = 0;
@@ -4484,9 +4515,13 @@ struct JoinOp : Op {
                 (const int64_t*)gt->keys.p,
                 (const unsigned long long*)gt->acc_multi.p,
                 plan.n_aggs + 1, plan.n_aggs, bp, cap, chunk,
-                (const int64_t*)d_offs.p,
+                (unsigned long long*)d_cursor.p,
                 (int64_t*)op.pg.cols[c_key].data, pl_outs, agg_outs);
+            unsigned long long n_out = 0;
+            CHKV(hipMemcpyAsync(&n_out, d_cursor.p, 8,
+                                hipMemcpyDeviceToHost, g_stream));
             CHKV(hipStreamSynchronize(g_stream));
+            op.pg.n_rows = (int64_t)n_out;
             outq.push_back(std::move(op));
             return;
         }
@@ -4499,12 +4534,16 @@ struct JoinOp : Op {
                            (const int32_t*)gt->head.p,
                            (const unsigned long long*)gt->acc.p,
                            gt->acc_words, bp, cap,
-                           chunk, (const int64_t*)d_offs.p,
+                           chunk, (unsigned long long*)d_cursor.p,
                            (int64_t*)op.pg.cols[c_key].data, pl_outs,
                            (int64_t*)op.pg.cols[c_dec].data,
                            (double*)op.pg.cols[c_f64].data,
                            (int64_t*)op.pg.cols[c_cnt].data);
+        unsigned long long n_out = 0;
+        CHKV(hipMemcpyAsync(&n_out, d_cursor.p, 8, hipMemcpyDeviceToHost,
+                            g_stream));
         CHKV(hipStreamSynchronize(g_stream));
+        op.pg.n_rows = (int64_t)n_out;
         outq.push_back(std::move(op));
     }
 };
@@ -4542,7 +4581,7 @@ struct GroupByOp : Op {
         kv.alloc((size_t)cap * 8 * plan.n_keys);
         acc.alloc((size_t)cap * acc_words * 8);
         acc.zero();
-        counters.alloc(16);
+        counters.alloc(24); /* [tick ovf, table full, distinct groups] */
         counters.zero();
         bool mm = false;
         for (int a = 0; a < plan.n_aggs; a++)
@@ -4583,8 +4622,8 @@ struct GroupByOp : Op {
     }
     void finish() override
     {
-        unsigned long long c[2];
-        CHKV(hipMemcpy(c, counters.p, 16, hipMemcpyDeviceToHost));
+        unsigned long long c[3];
+        CHKV(hipMemcpy(c, counters.p, 24, hipMemcpyDeviceToHost));
         if (c[1])
             throw std::runtime_error(
                 "groupby table full: raise capacity_hint");
@@ -4592,32 +4631,17 @@ struct GroupByOp : Op {
             throw std::runtime_error(
                 "bigint/decimal SUM overflow in groupby "
                 "(Math.addExact semantics)");
-        int64_t chunk = (cap + FLT_NB - 1) / FLT_NB;
-        chunk = (chunk + 255) / 256 * 256;
-        if (chunk < 256) chunk = 256;
-        DevBuf d_counts;
-        d_counts.alloc(FLT_NB * 8);
-        hipLaunchKernelGGL(k_groups_count, dim3(FLT_NB), dim3(256), 0,
-                           g_stream, (const unsigned long long*)acc.p,
-                           acc_words, acc_words - 1, cap, chunk,
-                           (int64_t*)d_counts.p);
-        std::vector<int64_t> h(FLT_NB);
-        CHKV(hipMemcpyAsync(h.data(), d_counts.p, FLT_NB * 8,
-                            hipMemcpyDeviceToHost, g_stream));
-        CHKV(hipStreamSynchronize(g_stream));
-        int64_t total = 0;
-        for (int b = 0; b < FLT_NB; b++) {
-            int64_t v = h[b];
-            h[b] = total;
-            total += v;
-        }
+        int64_t total = (int64_t)c[2]; /* distinct groups, counted at
+                                          claim time */
         if (total * 100 > cap * 85)
             throw std::runtime_error(
                 "groupby fill exceeded 0.85: raise capacity_hint");
-        DevBuf d_offs;
-        d_offs.alloc(FLT_NB * 8);
-        CHKV(hipMemcpyAsync(d_offs.p, h.data(), FLT_NB * 8,
-                            hipMemcpyHostToDevice, g_stream));
+        int64_t chunk = (cap + FLT_NB - 1) / FLT_NB;
+        chunk = (chunk + 255) / 256 * 256;
+        if (chunk < 256) chunk = 256;
+        DevBuf d_cursor;
+        d_cursor.alloc(8);
+        d_cursor.zero();
         OutPage op;
         op.pg.n_rows = total;
         int nc = 0;
@@ -4656,9 +4680,14 @@ struct GroupByOp : Op {
                            g_stream, (const int64_t*)kv.p, cap,
                            plan.n_keys,
                            (const unsigned long long*)acc.p, acc_words,
-                           offs, plan, chunk, (const int64_t*)d_offs.p,
+                           offs, plan, chunk,
+                           (unsigned long long*)d_cursor.p,
                            key_outs, agg_outs);
+        unsigned long long n_out = 0;
+        CHKV(hipMemcpyAsync(&n_out, d_cursor.p, 8, hipMemcpyDeviceToHost,
+                            g_stream));
         CHKV(hipStreamSynchronize(g_stream));
+        op.pg.n_rows = (int64_t)n_out;
         outq.push_back(std::move(op));
     }
 };
