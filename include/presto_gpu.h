@@ -165,6 +165,11 @@ typedef enum {
                                  differences) — planner expressions like
                                  date -> week_seq ((date - day0) / 7) or
                                  multiplicity - 1 */
+    PG_PROJ_KEYSHL_DIV = 8,   /* (a << shift) | (b / div) with
+                                 c = (shift << 16) | div — composite keys
+                                 over a derived dimension (e.g. TPC-DS
+                                 Q72's (item, week) key in ONE pass:
+                                 item << 14 | date/7) */
 } pg_proj_kind;
 
 typedef struct {

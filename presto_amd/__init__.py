@@ -21,7 +21,7 @@ from .engine import (  # noqa: F401
     CMP_LT, CMP_LE, CMP_GT, CMP_GE, CMP_EQ, CMP_NE, CMP_CONTAINS,
     CMP_PREFIX, CMP_CONTAINS2, CMP_NOT_CONTAINS2,
     PROJ_IDENT, PROJ_DISC_PRICE, PROJ_CHARGE, PROJ_MUL, PROJ_DIV,
-    PROJ_KEYSHL, PROJ_SHR, PROJ_SUBDIV,
+    PROJ_KEYSHL, PROJ_SHR, PROJ_SUBDIV, PROJ_KEYSHL_DIV,
     AGG_COUNT, AGG_SUM_F64, AGG_SUM_DEC, AGG_SUM_I64, AGG_MIN, AGG_MAX,
     OP_FILTER_PROJECT, OP_HASH_AGG_SMALL, OP_HASH_BUILD, OP_LOOKUP_JOIN,
     OP_TOPN, OP_PARTITION, OP_GROUPBY_MULTI,

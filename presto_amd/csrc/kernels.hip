@@ -934,6 +934,13 @@ __device__ inline void d_emit_val(const pg_page& pg, const pg_proj& p,
             (p.c ? (int64_t)p.c : 1);
         return;
     }
+    if (p.kind == PG_PROJ_KEYSHL_DIV) {
+        int32_t shift = p.c >> 16, dv = p.c & 0xffff;
+        ((int64_t*)out)[pos] =
+            (d_load_i64(pg.cols[p.a], i) << shift) |
+            (d_load_i64(pg.cols[p.b], i) / (dv ? dv : 1));
+        return;
+    }
     if (p.kind == PG_PROJ_IDENT) {
         const pg_col& c = pg.cols[p.a];
         switch (c.tag) {
@@ -1963,14 +1970,22 @@ __global__ __launch_bounds__(256) void k_probe_emit(
                            pos);
             }
             for (int o = 0; o < bp.n; o++) {
+                int64_t pv = bp.pack_bits && o == 0
+                                 ? (keys[r] &
+                                    ((1ll << bp.pack_bits) - 1))
+                                 : 0;
                 switch (bp.tag[o]) {
                     case PG_T_I32:
                         ((int32_t*)build_outs.ptr[o])[pos] =
-                            ((const int32_t*)bp.ptr[o])[r];
+                            bp.pack_bits && o == 0
+                                ? (int32_t)pv
+                                : ((const int32_t*)bp.ptr[o])[r];
                         break;
                     case PG_T_I64:
                         ((int64_t*)build_outs.ptr[o])[pos] =
-                            ((const int64_t*)bp.ptr[o])[r];
+                            bp.pack_bits && o == 0
+                                ? pv
+                                : ((const int64_t*)bp.ptr[o])[r];
                         break;
                     case PG_T_F64:
                         ((double*)build_outs.ptr[o])[pos] =
@@ -1978,7 +1993,9 @@ __global__ __launch_bounds__(256) void k_probe_emit(
                         break;
                     default:
                         ((uint8_t*)build_outs.ptr[o])[pos] =
-                            ((const uint8_t*)bp.ptr[o])[r];
+                            bp.pack_bits && o == 0
+                                ? (uint8_t)pv
+                                : ((const uint8_t*)bp.ptr[o])[r];
                 }
             }
             pos++;
@@ -3264,7 +3281,8 @@ struct FilterOp : Op {
                           ? sp.pg.cols[plan.proj[o].a].tag
                       : (plan.proj[o].kind == PG_PROJ_KEYSHL ||
                          plan.proj[o].kind == PG_PROJ_SHR ||
-                         plan.proj[o].kind == PG_PROJ_SUBDIV)
+                         plan.proj[o].kind == PG_PROJ_SUBDIV ||
+                         plan.proj[o].kind == PG_PROJ_KEYSHL_DIV)
                           ? PG_T_I64
                           : PG_T_F64;
             if (tag == PG_T_VARBIN) {
@@ -4101,10 +4119,12 @@ struct JoinOp : Op {
         if (t->key_set_only)
             throw std::runtime_error(
                 "cannot probe a key-set-only table");
-        if (t->pack_bits && plan.mode != 1 && plan.mode != 2)
+        if (t->pack_bits &&
+            !(plan.mode == 1 || plan.mode == 2 ||
+              (plan.mode == 0 && t->slot_payloads)))
             throw std::runtime_error(
-                "packed agg tables support fused-agg probes only "
-                "(modes 1/2)");
+                "packed tables support fused-agg probes and slot-payload "
+                "emit joins only");
         /* mode 0 emit over a slot-payload table: unique keys, payloads
          * indexed by slot (no chains) */
         if (plan.mode == 1 || plan.mode == 3) {
@@ -4385,6 +4405,7 @@ struct JoinOp : Op {
         build_payloads bp{};
         bp.n = (int32_t)t->payload.size();
         bp.by_slot = t->slot_payloads ? 1 : 0;
+        bp.pack_bits = t->pack_bits;
         emit_outs bouts{};
         bouts.n = bp.n;
         for (int o = 0; o < bp.n; o++) {

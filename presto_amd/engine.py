@@ -16,7 +16,7 @@ T_U8, T_I32, T_I64, T_F64, T_VARBIN, T_I128 = 0, 1, 2, 3, 4, 5
 (CMP_LT, CMP_LE, CMP_GT, CMP_GE, CMP_EQ, CMP_NE, CMP_CONTAINS,
  CMP_PREFIX, CMP_CONTAINS2, CMP_NOT_CONTAINS2) = range(10)
 (PROJ_IDENT, PROJ_DISC_PRICE, PROJ_CHARGE, PROJ_MUL, PROJ_DIV,
- PROJ_KEYSHL, PROJ_SHR, PROJ_SUBDIV) = range(8)
+ PROJ_KEYSHL, PROJ_SHR, PROJ_SUBDIV, PROJ_KEYSHL_DIV) = range(9)
 (AGG_COUNT, AGG_SUM_F64, AGG_SUM_DEC, AGG_SUM_I64,
  AGG_MIN, AGG_MAX) = range(6)
 (OP_FILTER_PROJECT, OP_HASH_AGG_SMALL, OP_HASH_BUILD, OP_LOOKUP_JOIN,

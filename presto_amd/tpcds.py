@@ -24,7 +24,7 @@ from .engine import (
     PlanGroupBy, Pred, Proj, Agg,
     OP_FILTER_PROJECT, OP_HASH_BUILD, OP_LOOKUP_JOIN, OP_GROUPBY_MULTI,
     CMP_LT, CMP_GT, CMP_GE, CMP_LE, CMP_EQ, CMP_NE,
-    PROJ_IDENT, PROJ_MUL, PROJ_KEYSHL, PROJ_SUBDIV,
+    PROJ_IDENT, PROJ_MUL, PROJ_KEYSHL, PROJ_SUBDIV, PROJ_KEYSHL_DIV,
     AGG_COUNT, AGG_SUM_I64, lib,
 )
 
@@ -403,20 +403,17 @@ def ds_q72(gen, sf, cs_page, inv_pages, cr_page, date_page, cdemo_page,
     b3 = _chain_build(f4[1], 0, (1, 2, 3), f4[1].n_rows)
     tables.append(b3)
 
-    # inventory probes the (item, week) table; on-hand < ordered
+    # inventory probes the (item, week) table; on-hand < ordered.
+    # ONE pass: the (item<<14 | date/7) key and the week channel are both
+    # derived in the same projection set (KEYSHL_DIV / SUBDIV)
     joined_parts = []
     for inv in inv_pages:
-        fi1 = _filter(inv, projs=(Proj(PROJ_SUBDIV, 0, 0, 7),
-                                  Proj(PROJ_IDENT, 1, 0, 0),
-                                  Proj(PROJ_IDENT, 2, 0, 0),
-                                  Proj(PROJ_IDENT, 3, 0, 0)))
-        fi2 = _filter(fi1[1], projs=(Proj(PROJ_KEYSHL, 1, 0, 14),
-                                     Proj(PROJ_IDENT, 1, 0, 0),
-                                     Proj(PROJ_IDENT, 0, 0, 0),
-                                     Proj(PROJ_IDENT, 2, 0, 0),
-                                     Proj(PROJ_IDENT, 3, 0, 0)),
-                      raw=True)
-        fi1[0].destroy()
+        fi2 = _filter(inv,
+                      projs=(Proj(PROJ_KEYSHL_DIV, 1, 0, (14 << 16) | 7),
+                             Proj(PROJ_IDENT, 1, 0, 0),
+                             Proj(PROJ_SUBDIV, 0, 0, 7),
+                             Proj(PROJ_IDENT, 2, 0, 0),
+                             Proj(PROJ_IDENT, 3, 0, 0)))
         jj, pj = _emit_join(fi2[1], b3.table(), 0, (1, 2, 3, 4))
         fi2[0].destroy()
         # pj: [item, wk, wh, qoh, qty, promo, crkey]
