@@ -1884,7 +1884,8 @@ __global__ __launch_bounds__(256) void k_acc_min_init(
 __global__ __launch_bounds__(256) void k_probe_count(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
     const uint8_t* tags, const int32_t* head, const int32_t* next,
-    int64_t mask, int64_t lmask, int64_t chunk, int64_t* block_counts)
+    int64_t mask, int64_t lmask, int32_t pbits, int64_t chunk,
+    int64_t* block_counts)
 {
     const int64_t n = pg.n_rows;
     const int64_t lo = (int64_t)blockIdx.x * chunk;
@@ -1893,7 +1894,8 @@ __global__ __launch_bounds__(256) void k_probe_count(
     for (int64_t i = lo + threadIdx.x; i < hi; i += 256) {
         if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
         int64_t key = d_load_i64(pg.cols[plan.key_col], i);
-        int64_t sl = d_tbl_find_tagged(keys, tags, mask, lmask, 0, key);
+        int64_t sl = d_tbl_find_tagged(keys, tags, mask, lmask, pbits,
+                                       key);
         if (sl >= 0) {
             if (head)
                 for (int32_t r = head[sl]; r >= 0; r = next[r]) total++;
@@ -1921,8 +1923,9 @@ struct build_payloads {
 __global__ __launch_bounds__(256) void k_probe_emit(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
     const uint8_t* tags, const int32_t* head, const int32_t* next,
-    int64_t mask, int64_t lmask, int64_t chunk, const int64_t* block_offs,
-    emit_outs probe_outs, build_payloads bp, emit_outs build_outs)
+    int64_t mask, int64_t lmask, int32_t pbits, int64_t chunk,
+    const int64_t* block_offs, emit_outs probe_outs, build_payloads bp,
+    emit_outs build_outs)
 {
     const int64_t n = pg.n_rows;
     const int64_t lo = (int64_t)blockIdx.x * chunk;
@@ -1938,7 +1941,7 @@ __global__ __launch_bounds__(256) void k_probe_emit(
         int32_t c = 0;
         if (i < hi && d_eval_preds(pg, plan.preds, plan.n_preds, i)) {
             int64_t key = d_load_i64(pg.cols[plan.key_col], i);
-            sl = d_tbl_find_tagged(keys, tags, mask, lmask, 0, key);
+            sl = d_tbl_find_tagged(keys, tags, mask, lmask, pbits, key);
             if (sl >= 0) {
                 if (head)
                     for (int32_t r = head[sl]; r >= 0; r = next[r]) c++;
@@ -4369,7 +4372,7 @@ struct JoinOp : Op {
                            (const uint8_t*)t->tags.p,
                            (const int32_t*)t->head.p,
                            (const int32_t*)t->next.p, t->mask,
-                           t->local_mask, chunk,
+                           t->local_mask, t->pack_bits, chunk,
                            (int64_t*)d_counts.p);
         std::vector<int64_t> h(FLT_NB);
         CHKV(hipMemcpyAsync(h.data(), d_counts.p, FLT_NB * 8,
@@ -4426,7 +4429,7 @@ struct JoinOp : Op {
                            (const uint8_t*)t->tags.p,
                            (const int32_t*)t->head.p,
                            (const int32_t*)t->next.p, t->mask,
-                           t->local_mask, chunk,
+                           t->local_mask, t->pack_bits, chunk,
                            (const int64_t*)d_offs.p, pouts, bp, bouts);
         hot_end();
         CHKV(hipStreamSynchronize(g_stream));
