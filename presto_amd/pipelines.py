@@ -1357,107 +1357,92 @@ def q9(part: Page, supp: Page, orders: Page, ps: Page, li: Page):
     og.add_input(part)
     og.finish()
 
-    # lineitem: one green-semijoin pass emitting the packed (pk,sk)
-    # composite key alongside the money/order channels
     fl = PlanFilterProject()
     fl.n_proj = 6
-    fl.proj[0] = Proj(PROJ_KEYSHL, li.channel("partkey"),
-                      li.channel("suppkey"), 20)  # pks = pk<<20|sk
-    for i, name in enumerate(("suppkey", "orderkey", "quantity",
+    for i, name in enumerate(("partkey", "suppkey", "orderkey", "quantity",
                               "extendedprice", "discount")):
-        fl.proj[1 + i] = Proj(PROJ_IDENT, li.channel(name), 0, 0)
+        fl.proj[i] = Proj(PROJ_IDENT, li.channel(name), 0, 0)
     fl.semijoin_table = og.table()
     fl.semijoin_col = li.channel("partkey")
     f = Operator(OP_FILTER_PROJECT, fl)
     f.add_input(li)
-    gli = f.get_output_raw()  # [pks, sk, ok, qty, ep, dc]
-
-    # partsupp as a DIRECT composite-key agg table (unique (pk,sk) pairs;
-    # every surviving lineitem hits, so fill is tight and tags are off) —
-    # replaces the chained partkey build + 4-way chain walk + suppkey
-    # equality filter of round 1
-    fps = PlanFilterProject()
-    fps.n_proj = 2
-    fps.proj[0] = Proj(PROJ_KEYSHL, ps.channel("partkey"),
-                       ps.channel("suppkey"), 20)
-    fps.proj[1] = Proj(PROJ_IDENT, ps.channel("supplycost"), 0, 0)
-    fps.semijoin_table = og.table()
-    fps.semijoin_col = ps.channel("partkey")
-    fo = Operator(OP_FILTER_PROJECT, fps)
-    fo.add_input(ps)
-    psp = fo.get_output_raw()  # green rows only: [pks, cost]
+    gli = f.get_output_raw()  # [pk, sk, ok, qty, ep, dc]
 
     bp = PlanHashBuild()
-    bp.key_col = 0
+    bp.key_col = ps.channel("partkey")
     bp.semijoin_table = -1
-    bp.n_payload = 1
-    bp.payload_col[0] = 1
-    bp.capacity_hint = max(psp.n_rows, 16)
-    bp.agg_table = 1
-    bp.fill_x10 = 13
+    bp.n_payload = 2
+    bp.payload_col[0] = ps.channel("suppkey")
+    bp.payload_col[1] = ps.channel("supplycost")
+    bp.capacity_hint = ps.n_rows
     ops_ = Operator(OP_HASH_BUILD, bp)
-    ops_.add_input_raw(psp)
+    ops_.add_input(ps)
     ops_.finish()
-    fo.destroy()
 
     j1 = PlanLookupJoin()
     j1.table = ops_.table()
     j1.key_col = 0
     j1.mode = 0
-    j1.n_emit = 5
-    for i, c in enumerate((1, 2, 3, 4, 5)):
-        j1.emit_probe_cols[i] = c
+    j1.n_emit = 6
+    for i in range(6):
+        j1.emit_probe_cols[i] = i
     ja = Operator(OP_LOOKUP_JOIN, j1)
     ja.add_input_raw(gli)
-    pb = ja.get_output_raw()  # [sk, ok, qty, ep, dc, cost]
+    pa = ja.get_output_raw()  # [pk, sk, ok, qty, ep, dc, ps_sk, cost]
 
-    # orders as a PACKED agg table (slot = orderkey<<16 | orderdate, one
-    # CAS per insert; all-hit probes) — replaces the chained 150M build
+    fe = PlanFilterProject()
+    fe.n_preds = 1
+    pe = Pred(1, CMP_EQ, 0, 0.0)
+    pe.rhs_col = 6 + 1
+    fe.preds[0] = pe
+    fe.n_proj = 6
+    for i, c in enumerate((1, 2, 3, 4, 5, 7)):
+        fe.proj[i] = Proj(PROJ_IDENT, c, 0, 0)
+    f2 = Operator(OP_FILTER_PROJECT, fe)
+    f2.add_input_raw(pa)
+    pb = f2.get_output_raw()  # [sk, ok, qty, ep, dc, cost]
+
+    bs = PlanHashBuild()
+    bs.key_col = supp.channel("suppkey")
+    bs.semijoin_table = -1
+    bs.n_payload = 1
+    bs.payload_col[0] = supp.channel("nationkey")
+    bs.capacity_hint = supp.n_rows
+    os_ = Operator(OP_HASH_BUILD, bs)
+    os_.add_input(supp)
+    os_.finish()
+
+    j2 = PlanLookupJoin()
+    j2.table = os_.table()
+    j2.key_col = 0
+    j2.mode = 0
+    j2.n_emit = 5
+    for i, c in enumerate((1, 2, 3, 4, 5)):
+        j2.emit_probe_cols[i] = c
+    jb = Operator(OP_LOOKUP_JOIN, j2)
+    jb.add_input_raw(pb)
+    pc = jb.get_output_raw()  # [ok, qty, ep, dc, cost, nat]
+
     bo = PlanHashBuild()
     bo.key_col = orders.channel("orderkey")
     bo.semijoin_table = -1
     bo.n_payload = 1
     bo.payload_col[0] = orders.channel("orderdate")
-    bo.capacity_hint = orders.n_rows + 64
-    bo.agg_table = 1
-    bo.fill_x10 = 13
-    bo.pack_bits = 16
+    bo.capacity_hint = orders.n_rows
     oo = Operator(OP_HASH_BUILD, bo)
     oo.add_input(orders)
     oo.finish()
 
     j3 = PlanLookupJoin()
     j3.table = oo.table()
-    j3.key_col = 1
+    j3.key_col = 0
     j3.mode = 0
     j3.n_emit = 5
-    for i, c in enumerate((0, 2, 3, 4, 5)):
+    for i, c in enumerate((1, 2, 3, 4, 5)):
         j3.emit_probe_cols[i] = c
     jc = Operator(OP_LOOKUP_JOIN, j3)
-    jc.add_input_raw(pb)
-    pc2 = jc.get_output_raw()  # [sk, qty, ep, dc, cost, odate]
-
-    # attach supplier nation via a tiny chained supplier join (1M-entry
-    # build; the probe side is only the green-part survivors)
-    bs2 = PlanHashBuild()
-    bs2.key_col = supp.channel("suppkey")
-    bs2.semijoin_table = -1
-    bs2.n_payload = 1
-    bs2.payload_col[0] = supp.channel("nationkey")
-    bs2.capacity_hint = supp.n_rows
-    os2 = Operator(OP_HASH_BUILD, bs2)
-    os2.add_input(supp)
-    os2.finish()
-    j4 = PlanLookupJoin()
-    j4.table = os2.table()
-    j4.key_col = 0
-    j4.mode = 0
-    j4.n_emit = 5
-    for i, c in enumerate((1, 2, 3, 4, 5)):
-        j4.emit_probe_cols[i] = c
-    jd = Operator(OP_LOOKUP_JOIN, j4)
-    jd.add_input_raw(pc2)
-    pd = jd.get_output_raw()  # [qty, ep, dc, cost, odate, nat]
+    jc.add_input_raw(pc)
+    pd = jc.get_output_raw()  # [qty, ep, dc, cost, nat, odate]
 
     profit = [[0] * 7 for _ in range(25)]
     # ONE general multi-channel group-by over (nation, orderdate) — the
@@ -1466,8 +1451,8 @@ def q9(part: Page, supp: Page, orders: Page, ps: Page, li: Page):
     # into years on the host (exact integer ticks end to end)
     gq = PlanGroupBy()
     gq.n_keys = 2
-    gq.key_col[0] = 5  # nat
-    gq.key_col[1] = 4  # odate
+    gq.key_col[0] = 4  # nat
+    gq.key_col[1] = 5  # odate
     gq.capacity_hint = 25 * 2500 + 1024
     gq.n_aggs = 2
     gq.aggs[0] = Agg(AGG_SUM_DEC, Proj(PROJ_DISC_PRICE, 1, 2, 0), 4)
@@ -1487,11 +1472,12 @@ def q9(part: Page, supp: Page, orders: Page, ps: Page, li: Page):
             profit[int(gout["nat"][i])][y] += \
                 int(gout["rev"][i]) - int(gout["cost"][i])
 
-    jd.destroy()
     jc.destroy()
+    jb.destroy()
+    f2.destroy()
     ja.destroy()
     f.destroy()
-    for o in (og, ops_, os2, oo):
+    for o in (og, ops_, os_, oo):
         lib().c.pg_table_destroy(o.table())
         o.destroy()
     return profit
