@@ -2124,19 +2124,20 @@ __global__ __launch_bounds__(256) void k_probe_agg_q5(
     int64_t base0 = 2 * ((int64_t)blockIdx.x * blockDim.x + threadIdx.x);
     int64_t stride = 2 * (int64_t)gridDim.x * blockDim.x;
     for (int64_t base = base0; base < n; base += stride) {
-        int64_t k[2];
+        int64_t k[2], sk[2];
         bool sel[2];
-        /* suppkey and the money columns are loaded only on ORDER hits
-         * (~15% of rows) / local-supplier matches (~0.6%) */
         if (base + 2 <= n) {
             vl2 ka = __builtin_nontemporal_load((const vl2*)(okey + base));
+            vl2 kb = __builtin_nontemporal_load((const vl2*)(skey + base));
             k[0] = ka[0]; k[1] = ka[1];
+            sk[0] = kb[0]; sk[1] = kb[1];
             sel[0] = sel[1] = true;
         } else {
 #pragma unroll
             for (int j = 0; j < 2; j++) {
                 sel[j] = base + j < n;
                 k[j] = sel[j] ? okey[base + j] : 0;
+                sk[j] = sel[j] ? skey[base + j] : 0;
             }
         }
 #pragma unroll
@@ -2163,9 +2164,8 @@ __global__ __launch_bounds__(256) void k_probe_agg_q5(
             }
             if (slot < 0) continue;
             uint8_t g1 = (uint8_t)(keys[slot] & pmask);
-            int64_t skj = skey[base + j];
-            if (skj < 1 || skj > dense_n) continue;
-            uint8_t g2 = dense_vals[skj - 1];
+            if (sk[j] < 1 || sk[j] > dense_n) continue;
+            uint8_t g2 = dense_vals[sk[j] - 1];
             if (g1 != g2) continue;
             int gi = -1;
 #pragma unroll
