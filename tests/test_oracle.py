@@ -615,3 +615,29 @@ def test_q2_sf1_golden(oracle_lib):
         # g[5] = s_address: unpinned
         assert phones[sk - 1] == g[6]
         assert pool[soff[sk-1]:soff[sk-1]+sln[sk-1]].decode() == g[7]
+
+
+def test_tpcds_date_dim_math():
+    """The tpcds.c proleptic calendar (year/quarter/week per day index)
+    cross-checked against Python's datetime."""
+    import ctypes as C
+    import datetime
+    import pathlib
+    import numpy as np
+    so = pathlib.Path(__file__).resolve().parent.parent / "oracle" / \
+        "liboracle.so"
+    L = C.CDLL(str(so))
+    n = 73049
+    year = np.zeros(n, np.int32)
+    qname = np.zeros(n, np.int32)
+    week = np.zeros(n, np.int32)
+    L.dsgen_date_dim(C.c_void_p(year.ctypes.data),
+                     C.c_void_p(qname.ctypes.data),
+                     C.c_void_p(week.ctypes.data))
+    d0 = datetime.date(1900, 1, 1)
+    for i in list(range(0, n, 997)) + [0, 58, 59, 60, 36158, n - 1]:
+        d = d0 + datetime.timedelta(days=i)
+        assert year[i] == d.year, i
+        q = (d.month - 1) // 3
+        assert qname[i] == d.year * 4 + q, i
+        assert week[i] == i // 7
