@@ -58,8 +58,8 @@ def main():
     ap.add_argument("--sf", type=float, default=30.0)
     ap.add_argument("--reps", type=int, default=3)
     ap.add_argument("--queries",
-                    default="q2,q8,q9,q10,q12,q13,q14,q15,q16,q17,q18,"
-                            "q19,q20,q21,q22")
+                    default="q2,q4,q6,q7,q8,q9,q10,q11,q12,q13,q14,q15,"
+                            "q16,q17,q18,q19,q20,q21,q22")
     args = ap.parse_args()
     sf = args.sf
     orc = OracleLib(str(REPO / "oracle" / "liboracle.so"))
@@ -82,6 +82,67 @@ def main():
             r.update(extra)
         results.append(r)
         print(json.dumps(r), flush=True)
+
+    if "q6" in want:
+        pages = (page({k: li[k] for k in
+                       ("quantity", "extendedprice", "discount",
+                        "shipdate")}),)
+        got, secs = run("q6", lambda: P.pipelines.q6(*pages), args.reps)
+        rev, cnt = orc.q6(li)
+        ok = (len(got["rev_lo"]) == 1 and int(got["rev_lo"][0]) == rev
+              and int(got["count"][0]) == cnt)
+        record("q6", secs, ok)
+
+    if "q4" in want:
+        pri = orc.gen_orders_priority(sf)
+        lid = orc.gen_lineitem_dates(sf)
+        pages = (page({"orderkey": orders["orderkey"],
+                       "orderdate": orders["orderdate"],
+                       "priority": pri}),
+                 page({k: lid[k] for k in ("orderkey", "commitdate",
+                                           "receiptdate")}))
+        got, secs = run("q4", lambda: P.pipelines.q4(*pages), args.reps)
+        exp = orc.q4(orders, pri, lid)
+        record("q4", secs, got == exp)
+        del pages, lid
+
+    if "q7" in want:
+        pages = (page({"custkey": cust["custkey"],
+                       "nationkey": cust["nationkey"]}),
+                 page({k: orders[k] for k in ("orderkey", "custkey")}),
+                 page({"suppkey": supp["suppkey"],
+                       "nationkey": supp["nationkey"]}),
+                 page({k: li[k] for k in
+                       ("orderkey", "suppkey", "extendedprice",
+                        "discount", "shipdate")}))
+        got, secs = run("q7", lambda: P.pipelines.q7(*pages), args.reps)
+        exp = orc.q7(cust, orders, li, supp)
+        exp_t = [(r.supp_nation, r.cust_nation, r.year, r.revenue_1e4)
+                 for r in exp]
+        record("q7", secs, sorted(got) == sorted(exp_t))
+        del pages
+
+    if "q11" in want:
+        ps11 = orc.gen_partsupp(sf)
+        n_part = int(200000 * sf)
+        pages = (page({"suppkey": supp["suppkey"],
+                       "nationkey": supp["nationkey"]}),
+                 page({"partkey": ps11["partkey"],
+                       "suppkey": ps11["suppkey"],
+                       "supplycost": ps11["supplycost_cents"] / 100.0,
+                       "availqty":
+                           ps11["availqty"].astype(np.float64)}),)
+        got_pk, got_val = None, None
+
+        def _q11():
+            return P.pipelines.q11(pages[0], pages[1], n_part)
+        got, secs = run("q11", _q11, args.reps)
+        got_pk, got_val = got
+        exp_pk, exp_val = orc.q11(ps11, supp, n_part)
+        ok = (list(got_pk) == list(exp_pk) and
+              got_val.tolist() == list(exp_val))
+        record("q11", secs, ok)
+        del pages, ps11
 
     if "q2" in want:
         part3 = orc.gen_part3(sf)
