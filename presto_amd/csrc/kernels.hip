@@ -883,8 +883,15 @@ __global__ __launch_bounds__(256) void k_sel_count(pg_page pg,
                                                    int64_t* block_counts,
                                                    const int64_t* set_keys,
                                                    int64_t set_mask,
-                                                   int32_t set_col)
+                                                   int32_t set_col,
+                                                   unsigned long long*
+                                                       selbits)
 {
+    /* the per-window ballot masks are saved so the emit pass never
+     * re-evaluates the predicates — for expensive VARBIN scans (q13's
+     * NOT LIKE) that halves the whole filter cost; the mask array is
+     * n/8 bytes.  Chunks are 256-aligned, so each wave owns one
+     * 64-aligned mask word. */
     const int64_t n = pg.n_rows;
     const int64_t lo = (int64_t)blockIdx.x * chunk;
     const int64_t hi = min(lo + chunk, n);
@@ -897,7 +904,10 @@ __global__ __launch_bounds__(256) void k_sel_count(pg_page pg,
             sel = d_semi_ok(pg, set_col, i, set_keys, set_mask,
                             plan.semijoin_anti);
         uint64_t m = d_ballot(sel);
-        if (lane == 0) cnt += __popcll(m);
+        if (lane == 0) {
+            cnt += __popcll(m);
+            if (selbits) selbits[base >> 6] = m;
+        }
     }
     __shared__ int64_t lds[4];
     if (lane == 0) lds[wid] = cnt;
@@ -983,7 +993,9 @@ __global__ __launch_bounds__(256) void k_sel_emit(pg_page pg,
                                                   emit_outs outs,
                                                   const int64_t* set_keys,
                                                   int64_t set_mask,
-                                                  int32_t set_col)
+                                                  int32_t set_col,
+                                                  const unsigned long long*
+                                                      selbits)
 {
     const int64_t n = pg.n_rows;
     const int64_t lo = (int64_t)blockIdx.x * chunk;
@@ -995,11 +1007,19 @@ __global__ __launch_bounds__(256) void k_sel_emit(pg_page pg,
     __syncthreads();
     for (int64_t base = lo; base < hi; base += 256) {
         int64_t i = base + 64 * wid + lane;
-        bool sel = i < hi && d_eval_preds(pg, plan.preds, plan.n_preds, i);
-        if (sel && set_keys)
-            sel = d_semi_ok(pg, set_col, i, set_keys, set_mask,
-                            plan.semijoin_anti);
-        uint64_t m = d_ballot(sel);
+        uint64_t m;
+        bool sel;
+        if (selbits) { /* verdicts cached by the count pass */
+            m = selbits[(base + 64 * wid) >> 6];
+            sel = i < hi && ((m >> lane) & 1);
+        } else {
+            sel = i < hi &&
+                  d_eval_preds(pg, plan.preds, plan.n_preds, i);
+            if (sel && set_keys)
+                sel = d_semi_ok(pg, set_col, i, set_keys, set_mask,
+                                plan.semijoin_anti);
+            m = d_ballot(sel);
+        }
         int wsum = __popcll(m);
         if (lane == 0) wcnt[wid] = wsum;
         __syncthreads();
@@ -3196,6 +3216,7 @@ static int64_t next_pow2(int64_t x)
 struct SelResult {
     int64_t n = 0;
     std::vector<int64_t> block_offs; /* host */
+    std::shared_ptr<DevBuf> selbits; /* per-window ballot masks */
 };
 
 /* semijoin set arguments: dense sets pass the flag array with a negative
@@ -3219,14 +3240,18 @@ static SelResult sel_count(const pg_page& pg,
 {
     DevBuf d_counts;
     d_counts.alloc(FLT_NB * sizeof(int64_t));
+    auto selbits = std::make_shared<DevBuf>();
+    selbits->alloc(((size_t)pg.n_rows / 64 + 2) * 8);
     hipLaunchKernelGGL(k_sel_count, dim3(FLT_NB), dim3(256), 0, g_stream, pg,
                        plan, chunk, (int64_t*)d_counts.p, set_ptr(semi),
-                       set_mask_of(semi), semi_col);
+                       set_mask_of(semi), semi_col,
+                       (unsigned long long*)selbits->p);
     std::vector<int64_t> h(FLT_NB);
     CHKV(hipMemcpyAsync(h.data(), d_counts.p, FLT_NB * 8,
                         hipMemcpyDeviceToHost, g_stream));
     CHKV(hipStreamSynchronize(g_stream));
     SelResult r;
+    r.selbits = selbits;
     r.block_offs.resize(FLT_NB);
     for (int b = 0; b < FLT_NB; b++) {
         r.block_offs[b] = r.n;
@@ -3251,7 +3276,9 @@ static void sel_emit(const pg_page& pg, const pg_plan_filter_project& plan,
                         hipMemcpyHostToDevice, g_stream));
     hipLaunchKernelGGL(k_sel_emit, dim3(FLT_NB), dim3(256), 0, g_stream, pg,
                        plan, chunk, (const int64_t*)d_offs.p, outs,
-                       set_ptr(semi), set_mask_of(semi), semi_col);
+                       set_ptr(semi), set_mask_of(semi), semi_col,
+                       r.selbits ? (const unsigned long long*)r.selbits->p
+                                 : nullptr);
     CHKV(hipStreamSynchronize(g_stream));
 }
 
