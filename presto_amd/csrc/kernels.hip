@@ -195,6 +195,47 @@ __device__ inline int64_t d_load_i64(const pg_col& c, int64_t i)
     }
 }
 
+/* substring search with a SWAR first-byte skip: one unaligned u32 load
+ * covers four candidate offsets in the common no-match case (the
+ * LikeFunctions.likeVarchar positional scan, vectorized).  Returns the
+ * leftmost match offset >= from, or -1. */
+__device__ inline int32_t d_find(const uint8_t* d, int32_t n,
+                                 const char* pat, int32_t len,
+                                 int32_t from)
+{
+    if (len <= 0) return from <= n ? from : -1;
+    const uint8_t c0 = (uint8_t)pat[0];
+    const uint32_t c4 = 0x01010101u * c0;
+    int32_t last = n - len;
+    int32_t s = from;
+    for (; s + 3 <= last; ) {
+        uint32_t w;
+        memcpy(&w, d + s, 4);
+        uint32_t x = w ^ c4;
+        uint32_t t = (x - 0x01010101u) & ~x & 0x80808080u;
+        if (!t) {
+            s += 4;
+            continue;
+        }
+        /* some byte equals c0: test the 4 offsets the window covers */
+        for (int k = 0; k < 4; k++, s++) {
+            if (((w >> (8 * k)) & 0xff) != c0) continue;
+            bool e = true;
+            for (int j = 1; j < len; j++)
+                e = e && d[s + j] == (uint8_t)pat[j];
+            if (e) return s;
+        }
+    }
+    for (; s <= last; s++) {
+        if (d[s] != c0) continue;
+        bool e = true;
+        for (int j = 1; j < len; j++)
+            e = e && d[s + j] == (uint8_t)pat[j];
+        if (e) return s;
+    }
+    return -1;
+}
+
 /* predicate conjunction — PageFilter semantics (PageProcessor.java:299-343:
  * null comparison result excludes the row; our round-1 columns are
  * non-null, null_mask honored as exclusion) */
@@ -216,35 +257,21 @@ __device__ inline bool d_eval_preds(const pg_page& pg, const pg_pred* preds,
             int32_t b0 = c.offsets[e], b1 = c.offsets[e + 1];
             const uint8_t* d = (const uint8_t*)c.data + b0;
             int32_t n = b1 - b0;
-            if (pr.op == PG_CMP_CONTAINS || pr.op == PG_CMP_PREFIX) {
-                bool m = false;
-                int32_t last = pr.op == PG_CMP_PREFIX ? 0 : n - pr.slen;
-                for (int32_t s = 0; !m && s <= last; s++) {
-                    bool e = s + pr.slen <= n;
-                    for (int j = 0; e && j < pr.slen; j++)
-                        e = d[s + j] == (uint8_t)pr.sval[j];
-                    m = e;
-                }
-                ok = m;
+            if (pr.op == PG_CMP_CONTAINS) {
+                ok = d_find(d, n, pr.sval, pr.slen, 0) >= 0;
+            } else if (pr.op == PG_CMP_PREFIX) {
+                bool e = pr.slen <= n;
+                for (int j = 0; e && j < pr.slen; j++)
+                    e = d[j] == (uint8_t)pr.sval[j];
+                ok = e;
             } else if (pr.op == PG_CMP_CONTAINS2 ||
                        pr.op == PG_CMP_NOT_CONTAINS2) {
                 /* ordered '%a%b%': find a, then b after it (leftmost-a
                  * suffices: any later a leaves less room for b) */
                 int32_t la = pr.slen, lb = (int32_t)pr.ival;
-                bool m = false;
-                for (int32_t s = 0; !m && s + la <= n; s++) {
-                    bool e = true;
-                    for (int j = 0; e && j < la; j++)
-                        e = d[s + j] == (uint8_t)pr.sval[j];
-                    if (!e) continue;
-                    for (int32_t t = s + la; !m && t + lb <= n; t++) {
-                        bool f = true;
-                        for (int j = 0; f && j < lb; j++)
-                            f = d[t + j] == (uint8_t)pr.sval[la + j];
-                        m = f;
-                    }
-                    break; /* leftmost a checked; b not found after it */
-                }
+                int32_t a = d_find(d, n, pr.sval, la, 0);
+                bool m = a >= 0 &&
+                         d_find(d, n, pr.sval + la, lb, a + la) >= 0;
                 ok = (pr.op == PG_CMP_CONTAINS2) ? m : !m;
             } else {
                 bool eq = n == pr.slen;
