@@ -28,7 +28,9 @@ def _worker(rank, world, port):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     torch.distributed.init_process_group("gloo", rank=rank, world_size=world)
+    from presto_amd import dist as pdist
     from presto_amd.dist import exchange_columns, exchange_split_counts
+    pdist.exchange_bytes_reset()
 
     rng = np.random.default_rng(100 + rank)
     n = 5000
@@ -52,6 +54,10 @@ def _worker(rank, world, port):
     tot = torch.tensor([got["k"].numel()])
     torch.distributed.all_reduce(tot)
     assert tot.item() == n * world
+    # the xGMI-volume counter saw exactly the bytes leaving this rank
+    # (two i64 columns, rows not destined for self)
+    sent_away = n - send_counts[rank]
+    assert pdist.EXCHANGE_BYTES == 2 * 8 * sent_away
     # stability: rows from each source rank arrive in that source's
     # partition-major order (ascending original order within partition)
     off = 0
