@@ -3804,7 +3804,10 @@ struct BuildOp : Op {
              * 256 MiB L3 (measured: at ~0.5 GB the direct random insert
              * is op-rate-bound, not locality-bound, and partitioning
              * loses; at multi-GB tables the radix path wins) */
-            part = cap >= (128ll << 20);
+            int64_t part_min = 128ll << 20;
+            if (const char* ep = getenv("PG_PART_MIN_SLOTS"))
+                if (atoll(ep) > 0) part_min = atoll(ep);
+            part = cap >= part_min;
             /* byte tags reject probe misses from a cap-sized L3-resident
              * array (8x denser than the key lines).  For partitioned
              * builds the tag store lands in the L3-resident region wave
