@@ -1595,14 +1595,19 @@ def test_partitioned_build_parity(P):
         return {nm: out[nm][order] for nm in ("key", "date", "sum", "cnt")}
 
     direct = run(n)               # small cap -> direct insert
-    partd = run(70_000_000)       # cap = 256M slots -> partitioned build
+    import os
+    os.environ["PG_PART_MIN_SLOTS"] = "33554432"
+    try:
+        partd = run(70_000_000)   # cap 256M >= forced threshold
+        packed_part2 = run(70_000_000, pack_bits=14)
+    finally:
+        del os.environ["PG_PART_MIN_SLOTS"]
     packed = run(n, pack_bits=14)  # slot word = key<<14 | date
-    packed_part = run(70_000_000, pack_bits=14)
     assert len(direct["key"]) == len(partd["key"])
     for nm in ("key", "date", "sum", "cnt"):
         assert np.array_equal(direct[nm], partd[nm]), nm
         assert np.array_equal(direct[nm], packed[nm]), nm
-        assert np.array_equal(direct[nm], packed_part[nm]), nm
+        assert np.array_equal(direct[nm], packed_part2[nm]), nm
     # numpy cross-check of the grouped sums
     import collections
     exp = collections.defaultdict(int)
