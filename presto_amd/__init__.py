@@ -14,7 +14,8 @@ without a GPU) but creating any operator requires an AMD gfx950 GPU and
 fails loudly otherwise.
 """
 from .engine import (  # noqa: F401
-    lib, Operator, Page, Varbin, DictVarbin, PlanFilterProject,
+    lib, Operator, Page, Varbin, DictVarbin, DeviceVarbin,
+    PlanFilterProject,
     PlanHashAggSmall,
     PlanHashBuild, PlanLookupJoin, PlanTopN, PlanPartition, PlanGroupBy,
     Pred, Proj, Agg,

@@ -190,10 +190,37 @@ class DictVarbin:
         return len(self.ids)
 
 
+class DeviceVarbin:
+    """A VariableWidthBlock resident in HBM: torch-cuda uint8 bytes +
+    int32 offsets (+ optional int32 dictionary ids for DictionaryBlock
+    columns).  The boundary form of Varbin/DictVarbin once a page has
+    been staged."""
+
+    def __init__(self, data, offsets, dict_ids=None, dict_n=0):
+        self.data = data          # torch.uint8 cuda tensor
+        self.offsets = offsets    # torch.int32 cuda tensor (n+1)
+        self.dict_ids = dict_ids  # torch.int32 cuda tensor or None
+        self.dict_n = dict_n
+
+    @classmethod
+    def from_host(cls, v):
+        import torch
+        if isinstance(v, DictVarbin):
+            return cls(torch.from_numpy(v.dictionary.data).cuda(),
+                       torch.from_numpy(v.dictionary.offsets).cuda(),
+                       torch.from_numpy(v.ids).cuda(), v.dictionary.n)
+        return cls(torch.from_numpy(v.data).cuda(),
+                   torch.from_numpy(v.offsets).cuda())
+
+    def __len__(self):
+        n = len(self.dict_ids) if self.dict_ids is not None             else len(self.offsets) - 1
+        return n
+
+
 class Page:
     """A Presto Page: named columns backed by numpy (host) or torch-cuda
-    (device) arrays (Varbin for variable-width columns).  Column order is
-    the channel order."""
+    (device) arrays (Varbin / DeviceVarbin for variable-width columns).
+    Column order is the channel order."""
 
     def __init__(self, cols, n_rows=None):
         self.names = list(cols.keys())
@@ -211,7 +238,15 @@ class Page:
         for i, name in enumerate(self.names):
             a = self.cols[name]
             col = PgCol()
-            if isinstance(a, Varbin):
+            if isinstance(a, DeviceVarbin):
+                col.tag = T_VARBIN
+                col.on_device = 1
+                col.data = a.data.data_ptr()
+                col.offsets = a.offsets.data_ptr()
+                if a.dict_ids is not None:
+                    col.dict_ids = a.dict_ids.data_ptr()
+                    col.dict_n = a.dict_n
+            elif isinstance(a, Varbin):
                 col.tag = T_VARBIN
                 col.on_device = 0
                 col.data = a.data.ctypes.data

@@ -36,8 +36,14 @@ def dev(a):
 
 
 def page(cols):
-    return P.Page({k: (v if isinstance(v, P.Varbin) else dev(v))
-                   for k, v in cols.items()})
+    """Stage every column into HBM — including variable-width ones (the
+    boundary of SURVEY.md §8d: data starts device-resident; re-uploading
+    VARBIN bytes per step would put PCIe inside the timed region)."""
+    def up(v):
+        if isinstance(v, (P.Varbin, P.DictVarbin)):
+            return P.DeviceVarbin.from_host(v)
+        return dev(v)
+    return P.Page({k: up(v) for k, v in cols.items()})
 
 
 def run(name, fn, reps):
@@ -198,7 +204,8 @@ def main():
             cm = Varbin.__new__(Varbin)
             cm.data, cm.offsets, cm.n = data, offs, b - a
             opages.append(P.Page({"custkey": dev(orders["custkey"][a:b]),
-                                  "comment": cm}))
+                                  "comment":
+                                      P.DeviceVarbin.from_host(cm)}))
         n_cust = int(150000 * sf)
         got, secs = run("q13", lambda: P.pipelines.q13(n_cust, opages),
                         args.reps)
