@@ -1905,3 +1905,43 @@ def test_groupby_empty_input(P):
     out = op.get_output(["k", "c", "cnt"])
     op.destroy()
     assert len(out["k"]) == 0
+
+
+def test_device_varbin_page(P):
+    """DeviceVarbin (HBM-resident VariableWidthBlock) must behave exactly
+    like its host twin through predicates and emit."""
+    rng = np.random.RandomState(31)
+    words = [b"alpha", b"special sauce", b"beta", b"requests here",
+             b"plain", b"special then requests", b""]
+    n = 20_000
+    ids = rng.randint(0, len(words), n)
+    strings = [words[i] for i in ids]
+    host_v = P.Varbin(strings)
+    dev_v = P.DeviceVarbin.from_host(host_v)
+    vals = np.arange(n, dtype=np.int64)
+
+    def run(v):
+        fp = P.PlanFilterProject()
+        fp.n_preds = 1
+        pr = P.Pred(1, P.CMP_CONTAINS2, 8, 0.0)
+        pr.sval = b"special" + b"requests"
+        pr.slen = 7
+        fp.preds[0] = pr
+        fp.n_proj = 1
+        fp.proj[0] = P.Proj(P.PROJ_IDENT, 0, 0, 0)
+        f = P.Operator(P.OP_FILTER_PROJECT, fp)
+        f.add_input(P.Page({"v": vals, "s": v}))
+        out = f.get_output(["v"])
+        f.destroy()
+        return out["v"]
+
+    got_h = run(host_v)
+    got_d = run(dev_v)
+    exp = vals[np.array([b"special then requests" == s for s in strings])]
+    assert np.array_equal(got_h, exp)
+    assert np.array_equal(got_d, exp)
+    # dictionary form too
+    dict_host = P.DictVarbin(words, ids)
+    dict_dev = P.DeviceVarbin.from_host(dict_host)
+    assert np.array_equal(run(dict_host), exp)
+    assert np.array_equal(run(dict_dev), exp)

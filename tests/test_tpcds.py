@@ -83,9 +83,11 @@ def test_ds_q72_exact(dsgen):
     from presto_amd.tpcds import ds_q72
     sf = 1.0
     pages = _pages(P, dsgen, sf, torch=torch)
-    got = ds_q72(dsgen, sf, pages["cs72"], [pages["inv"]], pages["cr"],
-                 pages["date_y"], pages["cdemo"], pages["hdemo"],
-                 1999, 2, 3)
-    exp = dsgen.q72(sf, 1999, 2, 3)
-    assert len(got) == len(exp), (len(got), len(exp))
-    assert got == exp
+    for year, marital, buypot in ((1999, 2, 3), (2000, 0, 5),
+                                  (2001, 4, 0)):
+        got = ds_q72(dsgen, sf, pages["cs72"], [pages["inv"]],
+                     pages["cr"], pages["date_y"], pages["cdemo"],
+                     pages["hdemo"], year, marital, buypot)
+        exp = dsgen.q72(sf, year, marital, buypot)
+        assert len(got) == len(exp), (year, len(got), len(exp))
+        assert got == exp, (year, marital, buypot)
