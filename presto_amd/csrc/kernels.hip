@@ -1159,17 +1159,41 @@ __global__ __launch_bounds__(256) void k_set_insert(const int64_t* in_keys,
 }
 
 /* insert build rows (already filtered/compacted) with chains */
+/* key-presence bitmap over [1, bmax] (pg_plan_hash_build.bitmap_max_key):
+ * the bigint dynamic-filter analog.  Tested BEFORE the bucket hash on
+ * probes; a clear bit is a definitive miss. */
+__device__ inline void d_kbit_set(unsigned long long* kb, int64_t key)
+{
+    atomicOr(&kb[(uint64_t)key >> 6], 1ull << (key & 63));
+}
+__device__ inline bool d_kbit_test(const unsigned long long* kb,
+                                   int64_t bmax, int64_t key)
+{
+    if ((uint64_t)(key - 1) >= (uint64_t)bmax) return false;
+    return (kb[(uint64_t)key >> 6] >> (key & 63)) & 1;
+}
+
 __global__ __launch_bounds__(256) void k_tbl_insert(const int64_t* in_keys,
                                                     int64_t n, int64_t* keys,
                                                     uint8_t* tags,
                                                     int32_t* head,
                                                     int32_t* next,
-                                                    int64_t mask)
+                                                    int64_t mask,
+                                                    unsigned long long* kbits,
+                                                    int64_t bmax,
+                                                    unsigned long long* bm_err)
 {
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int64_t stride = (int64_t)gridDim.x * blockDim.x;
     for (; i < n; i += stride) {
         int64_t key = in_keys[i];
+        if (kbits) {
+            if ((uint64_t)(key - 1) >= (uint64_t)bmax) {
+                atomicAdd(bm_err, 1ull);
+                continue;
+            }
+            d_kbit_set(kbits, key);
+        }
         uint64_t h = pg_murmur3_finalize(pg_bigint_hash(key));
         int64_t s = (int64_t)(h & (uint64_t)mask);
         for (;;) {
@@ -1215,7 +1239,8 @@ __global__ __launch_bounds__(256) void k_tbl_insert_direct(
     int64_t lu_mask, int64_t lu_lmask, const uint8_t* lu_payload,
     int64_t* keys, uint8_t* tags, direct_payloads dp, int64_t mask,
     unsigned long long* inserted, unsigned long long* overflow,
-    unsigned long long* pack_err)
+    unsigned long long* pack_err, unsigned long long* kbits, int64_t bmax,
+    unsigned long long* bm_err)
 {
     const int32_t pbits = plan.pack_bits;
     int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -1272,6 +1297,13 @@ __global__ __launch_bounds__(256) void k_tbl_insert_direct(
                 continue;
             }
             word = (key << pbits) | pay0;
+        }
+        if (kbits) {
+            if ((uint64_t)(key - 1) >= (uint64_t)bmax) {
+                atomicAdd(bm_err, 1ull);
+                continue;
+            }
+            d_kbit_set(kbits, key);
         }
         uint64_t h = pg_murmur3_finalize(pg_bigint_hash(key));
         int64_t s = (int64_t)(h & (uint64_t)mask);
@@ -1525,7 +1557,8 @@ __global__ __launch_bounds__(256) void k_part_insert(
     direct_payloads dp, int64_t cap_p, int32_t P, int32_t K,
     int32_t use_barrier, int32_t pbits, unsigned long long* bar,
     unsigned long long* inserted, unsigned long long* overflow,
-    unsigned long long* pack_err)
+    unsigned long long* pack_err, unsigned long long* kbits, int64_t bmax,
+    unsigned long long* bm_err)
 {
     int64_t my_packerr = 0;
     /* use_barrier=1: init K L3-resident regions in-kernel between grid
@@ -1571,6 +1604,13 @@ __global__ __launch_bounds__(256) void k_part_insert(
                         continue;
                     }
                     word = (key << pbits) | pay0;
+                }
+                if (kbits) {
+                    if ((uint64_t)(key - 1) >= (uint64_t)bmax) {
+                        atomicAdd(bm_err, 1ull);
+                        continue;
+                    }
+                    d_kbit_set(kbits, key);
                 }
                 uint64_t h = pg_murmur3_finalize(pg_bigint_hash(key));
                 int64_t off = (int64_t)(h & (uint64_t)lmask);
@@ -1694,6 +1734,7 @@ __device__ inline void d_atomic_add_dec_ck(unsigned long long* slot,
 __global__ __launch_bounds__(256) void k_probe_agg(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
     const uint8_t* tags, int64_t mask, int64_t lmask, int32_t pbits,
+    const unsigned long long* kbits, int64_t bmax,
     unsigned long long* acc, int32_t aw, unsigned long long* ovf)
 {
     /* 4 CONSECUTIVE rows per thread with run dedup: neighboring rows
@@ -1734,8 +1775,10 @@ __global__ __launch_bounds__(256) void k_probe_agg(
             if (cur_slot == -2 || key != cur_key) {
                 flush();
                 cur_key = key;
-                cur_slot = d_tbl_find_tagged(keys, tags, mask, lmask,
-                                             pbits, key);
+                cur_slot = kbits && !d_kbit_test(kbits, bmax, key)
+                               ? -1
+                               : d_tbl_find_tagged(keys, tags, mask,
+                                                   lmask, pbits, key);
                 run_dec = 0;
                 run_flo = run_fhi = 0;
                 run_cnt = 0;
@@ -1779,7 +1822,8 @@ __global__ __launch_bounds__(256) void k_probe_agg_q3(
     const int32_t* sd /* nullable pred col */, int32_t pred_op,
     int32_t pred_val, const int64_t* okey, const double* ep,
     const double* dc, int64_t n, const int64_t* keys, const uint8_t* tags,
-    int64_t mask, int64_t lmask, int32_t pbits, int32_t dec_only,
+    int64_t mask, int64_t lmask, int32_t pbits,
+    const unsigned long long* kbits, int64_t bmax, int32_t dec_only,
     unsigned long long* acc, int32_t aw, unsigned long long* ovf)
 {
     typedef int vi2 __attribute__((ext_vector_type(2)));
@@ -1822,6 +1866,9 @@ __global__ __launch_bounds__(256) void k_probe_agg_q3(
         for (int j = 0; j < Q; j++) {
             slot[j] = -1;
             if (!sel[j]) continue;
+            /* dynamic-filter bitmap: one (usually L3-hot) load rejects
+             * ~91% of rows before the hash + tag/key chain */
+            if (kbits && !d_kbit_test(kbits, bmax, k[j])) continue;
             uint64_t h = pg_murmur3_finalize(pg_bigint_hash(k[j]));
             int64_t p = (int64_t)(h & (uint64_t)mask);
             uint8_t tg = tags ? d_tbl_tag(h) : 0;
@@ -1877,7 +1924,8 @@ __global__ __launch_bounds__(256) void k_probe_agg_q3(
 __global__ __launch_bounds__(256) void k_probe_agg_pay(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys1,
     const uint8_t* tags1, const int32_t* head1, int64_t mask1,
-    int64_t lmask1, const int64_t* pay1, const int64_t* keys2,
+    int64_t lmask1, const unsigned long long* kbits1, int64_t bmax1,
+    const int64_t* pay1, const int64_t* keys2,
     const uint8_t* tags2, int64_t mask2, int64_t lmask2,
     unsigned long long* acc2, int32_t aw, unsigned long long* ovf)
 {
@@ -1886,6 +1934,7 @@ __global__ __launch_bounds__(256) void k_probe_agg_pay(
     for (; i < pg.n_rows; i += stride) {
         if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
         int64_t key = d_load_i64(pg.cols[plan.key_col], i);
+        if (kbits1 && !d_kbit_test(kbits1, bmax1, key)) continue;
         int64_t s1 = d_tbl_find_tagged(keys1, tags1, mask1, lmask1, 0,
                                        key);
         if (s1 < 0) continue;
@@ -1931,7 +1980,8 @@ __global__ __launch_bounds__(256) void k_acc_min_init(
 __global__ __launch_bounds__(256) void k_probe_count(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
     const uint8_t* tags, const int32_t* head, const int32_t* next,
-    int64_t mask, int64_t lmask, int32_t pbits, int64_t chunk,
+    int64_t mask, int64_t lmask, int32_t pbits,
+    const unsigned long long* kbits, int64_t bmax, int64_t chunk,
     int64_t* block_counts)
 {
     const int64_t n = pg.n_rows;
@@ -1941,6 +1991,7 @@ __global__ __launch_bounds__(256) void k_probe_count(
     for (int64_t i = lo + threadIdx.x; i < hi; i += 256) {
         if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
         int64_t key = d_load_i64(pg.cols[plan.key_col], i);
+        if (kbits && !d_kbit_test(kbits, bmax, key)) continue;
         int64_t sl = d_tbl_find_tagged(keys, tags, mask, lmask, pbits,
                                        key);
         if (sl >= 0) {
@@ -1970,7 +2021,8 @@ struct build_payloads {
 __global__ __launch_bounds__(256) void k_probe_emit(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
     const uint8_t* tags, const int32_t* head, const int32_t* next,
-    int64_t mask, int64_t lmask, int32_t pbits, int64_t chunk,
+    int64_t mask, int64_t lmask, int32_t pbits,
+    const unsigned long long* kbits, int64_t bmax, int64_t chunk,
     const int64_t* block_offs, emit_outs probe_outs, build_payloads bp,
     emit_outs build_outs)
 {
@@ -1988,7 +2040,10 @@ __global__ __launch_bounds__(256) void k_probe_emit(
         int32_t c = 0;
         if (i < hi && d_eval_preds(pg, plan.preds, plan.n_preds, i)) {
             int64_t key = d_load_i64(pg.cols[plan.key_col], i);
-            sl = d_tbl_find_tagged(keys, tags, mask, lmask, pbits, key);
+            sl = kbits && !d_kbit_test(kbits, bmax, key)
+                     ? -1
+                     : d_tbl_find_tagged(keys, tags, mask, lmask, pbits,
+                                         key);
             if (sl >= 0) {
                 if (head)
                     for (int32_t r = head[sl]; r >= 0; r = next[r]) c++;
@@ -2152,6 +2207,7 @@ __global__ __launch_bounds__(256) void k_probe_agg_q5(
     pg_plan_lookup_join plan, const int64_t* okey, const int64_t* skey,
     const double* ep, const double* dc, int64_t n, const int64_t* keys,
     const uint8_t* tags, int64_t mask, int64_t lmask, int32_t pbits,
+    const unsigned long long* kbits, int64_t bmax,
     const uint8_t* dense_vals, int64_t dense_n,
     unsigned long long* out_dec, unsigned long long* out_flo,
     unsigned long long* out_fhi, unsigned long long* out_cnt)
@@ -2190,6 +2246,7 @@ __global__ __launch_bounds__(256) void k_probe_agg_q5(
 #pragma unroll
         for (int j = 0; j < 2; j++) {
             if (!sel[j]) continue;
+            if (kbits && !d_kbit_test(kbits, bmax, k[j])) continue;
             int64_t slot = -1;
             uint64_t h = pg_murmur3_finalize(pg_bigint_hash(k[j]));
             int64_t p = (int64_t)(h & (uint64_t)mask);
@@ -2263,6 +2320,7 @@ __global__ __launch_bounds__(256) void k_probe_agg_q5(
 __global__ __launch_bounds__(256) void k_probe_agg_multi(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
     const uint8_t* tags, int64_t mask, int64_t lmask, int32_t pbits,
+    const unsigned long long* kbits, int64_t bmax,
     unsigned long long* acc, unsigned long long* ovf)
 {
     const int n_aggs = plan.n_aggs;
@@ -2294,8 +2352,10 @@ __global__ __launch_bounds__(256) void k_probe_agg_multi(
             if (cur_slot == -2 || key != cur_key) {
                 flush();
                 cur_key = key;
-                cur_slot = d_tbl_find_tagged(keys, tags, mask, lmask,
-                                             pbits, key);
+                cur_slot = kbits && !d_kbit_test(kbits, bmax, key)
+                               ? -1
+                               : d_tbl_find_tagged(keys, tags, mask,
+                                                   lmask, pbits, key);
 #pragma unroll
                 for (int a = 0; a < 6; a++) run[a] = 0;
                 run_cnt = 0;
@@ -3197,6 +3257,8 @@ struct Table {
     int32_t acc_words = 0; /* 4 = [dec, flo, fhi, cnt]; 2 = [dec, cnt] */
     DevBuf acc_multi; /* multi-agg probes: (n_acc+1) u64 per slot */
     int32_t n_acc = 0;
+    DevBuf kbits;     /* key-presence bitmap (dynamic-filter analog) */
+    int64_t bmax = 0; /* bitmap covers keys [1, bmax]; 0 = none */
     bool slot_payloads = false; /* payloads indexed by slot (agg tables) */
     bool dense = false;          /* dense_array: payload[key-1], no hash */
     /* compacted build-row arrays: key + payloads */
@@ -3823,8 +3885,13 @@ struct BuildOp : Op {
             /* acc is allocated lazily by the first mode-1 probe (mode-2
              * and emit probes never touch it — at SF300 the slot_acc
              * array is 8 GB of alloc+memset otherwise) */
-            counters.alloc(32);
+            counters.alloc(40); /* [+4] = bitmap-range errors */
             counters.zero();
+            if (plan.bitmap_max_key > 0) {
+                t->bmax = plan.bitmap_max_key;
+                t->kbits.alloc((size_t)((t->bmax >> 6) + 1) * 8);
+                t->kbits.zero();
+            }
             if (part) {
                 /* P=256 keeps the scatter's per-row LDS counter atomics
                  * nearly conflict-free (≈1 lane per counter per wave) */
@@ -4002,7 +4069,9 @@ struct BuildOp : Op {
                                (uint8_t*)t->tags.p, dp, t->mask,
                                (unsigned long long*)counters.p,
                                (unsigned long long*)counters.p + 1,
-                               (unsigned long long*)counters.p + 3);
+                               (unsigned long long*)counters.p + 3,
+                               (unsigned long long*)t->kbits.p, t->bmax,
+                               (unsigned long long*)counters.p + 4);
             hot_end();
             CHKV(hipStreamSynchronize(g_stream));
             return;
@@ -4091,15 +4160,22 @@ struct BuildOp : Op {
                                    (unsigned long long*)barb.p,
                                    (unsigned long long*)counters.p,
                                    (unsigned long long*)counters.p + 1,
-                                   (unsigned long long*)counters.p + 3);
+                                   (unsigned long long*)counters.p + 3,
+                                   (unsigned long long*)t->kbits.p,
+                                   t->bmax,
+                                   (unsigned long long*)counters.p + 4);
                 hot_end();
                 CHKV(hipStreamSynchronize(g_stream));
                 stage.free();
                 cursorb.free();
                 barb.free();
             }
-            unsigned long long c[4];
-            CHKV(hipMemcpy(c, counters.p, 32, hipMemcpyDeviceToHost));
+            unsigned long long c[5];
+            CHKV(hipMemcpy(c, counters.p, 40, hipMemcpyDeviceToHost));
+            if (c[4])
+                throw std::runtime_error(
+                    "bitmap_max_key violated: build key outside "
+                    "[1, bitmap_max_key]");
             if (c[3])
                 throw std::runtime_error(
                     "pack_bits violated: key or payload outside the "
@@ -4141,6 +4217,14 @@ struct BuildOp : Op {
                            (int64_t*)t->keys.p,
                            t->key_set_only ? nullptr : (int32_t*)t->head.p,
                            cap);
+        DevBuf bm_err;
+        if (plan.bitmap_max_key > 0 && !t->key_set_only) {
+            t->bmax = plan.bitmap_max_key;
+            t->kbits.alloc((size_t)((t->bmax >> 6) + 1) * 8);
+            t->kbits.zero();
+            bm_err.alloc(8);
+            bm_err.zero();
+        }
         if (t->n_rows) {
             if (t->key_set_only)
                 hipLaunchKernelGGL(k_set_insert, dim3(2048), dim3(256), 0,
@@ -4152,9 +4236,20 @@ struct BuildOp : Op {
                                    t->n_rows, (int64_t*)t->keys.p,
                                    (uint8_t*)t->tags.p,
                                    (int32_t*)t->head.p, (int32_t*)t->next.p,
-                                   t->mask);
+                                   t->mask,
+                                   (unsigned long long*)t->kbits.p,
+                                   t->bmax,
+                                   (unsigned long long*)bm_err.p);
         }
         CHKV(hipStreamSynchronize(g_stream));
+        if (bm_err.p) {
+            unsigned long long e = 0;
+            CHKV(hipMemcpy(&e, bm_err.p, 8, hipMemcpyDeviceToHost));
+            if (e)
+                throw std::runtime_error(
+                    "bitmap_max_key violated: build key outside "
+                    "[1, bitmap_max_key]");
+        }
         std::lock_guard<std::mutex> lk(g_mu);
         tbl = g_next_table++;
         g_tables[tbl] = std::move(t);
@@ -4306,7 +4401,8 @@ struct JoinOp : Op {
                     (const double*)sp.pg.cols[plan.proj.b].data,
                     sp.pg.n_rows, (const int64_t*)t->keys.p,
                     (const uint8_t*)t->tags.p, t->mask, t->local_mask,
-                    t->pack_bits, (const uint8_t*)t2->payload[0].p,
+                    t->pack_bits, (const unsigned long long*)t->kbits.p, t->bmax,
+                    (const uint8_t*)t2->payload[0].p,
                     t2->cap, a, a + 8, a + 16, a + 24);
             } else {
                 hipLaunchKernelGGL(k_probe_agg_fused2<8>, dim3(4096),
@@ -4330,6 +4426,7 @@ struct JoinOp : Op {
                                (const uint8_t*)t->tags.p,
                                (const int32_t*)t->head.p, t->mask,
                                t->local_mask,
+                               (const unsigned long long*)t->kbits.p, t->bmax,
                                (const int64_t*)t->payload[0].p,
                                (const int64_t*)t2->keys.p,
                                (const uint8_t*)t2->tags.p, t2->mask,
@@ -4348,6 +4445,7 @@ struct JoinOp : Op {
                                (const int64_t*)t->keys.p,
                                (const uint8_t*)t->tags.p, t->mask,
                                t->local_mask, t->pack_bits,
+                               (const unsigned long long*)t->kbits.p, t->bmax,
                                (unsigned long long*)t->acc_multi.p,
                                (unsigned long long*)ovf.p);
             hot_end();
@@ -4390,7 +4488,7 @@ struct JoinOp : Op {
                     (const double*)sp.pg.cols[plan.proj.b].data,
                     sp.pg.n_rows, (const int64_t*)t->keys.p,
                     (const uint8_t*)t->tags.p, t->mask, t->local_mask,
-                    t->pack_bits, plan.dec_only,
+                    t->pack_bits, (const unsigned long long*)t->kbits.p, t->bmax, plan.dec_only,
                     (unsigned long long*)t->acc.p, t->acc_words,
                     (unsigned long long*)ovf.p);
             } else {
@@ -4399,6 +4497,7 @@ struct JoinOp : Op {
                                    (const int64_t*)t->keys.p,
                                    (const uint8_t*)t->tags.p, t->mask,
                                    t->local_mask, t->pack_bits,
+                                   (const unsigned long long*)t->kbits.p, t->bmax,
                                    (unsigned long long*)t->acc.p,
                                    t->acc_words,
                                    (unsigned long long*)ovf.p);
@@ -4430,7 +4529,8 @@ struct JoinOp : Op {
                            (const uint8_t*)t->tags.p,
                            (const int32_t*)t->head.p,
                            (const int32_t*)t->next.p, t->mask,
-                           t->local_mask, t->pack_bits, chunk,
+                           t->local_mask, t->pack_bits,
+                           (const unsigned long long*)t->kbits.p, t->bmax, chunk,
                            (int64_t*)d_counts.p);
         std::vector<int64_t> h(FLT_NB);
         CHKV(hipMemcpyAsync(h.data(), d_counts.p, FLT_NB * 8,
@@ -4487,7 +4587,8 @@ struct JoinOp : Op {
                            (const uint8_t*)t->tags.p,
                            (const int32_t*)t->head.p,
                            (const int32_t*)t->next.p, t->mask,
-                           t->local_mask, t->pack_bits, chunk,
+                           t->local_mask, t->pack_bits,
+                           (const unsigned long long*)t->kbits.p, t->bmax, chunk,
                            (const int64_t*)d_offs.p, pouts, bp, bouts);
         hot_end();
         CHKV(hipStreamSynchronize(g_stream));

@@ -1945,3 +1945,89 @@ def test_device_varbin_page(P):
     dict_dev = P.DeviceVarbin.from_host(dict_host)
     assert np.array_equal(run(dict_host), exp)
     assert np.array_equal(run(dict_dev), exp)
+
+
+def test_bitmap_prefilter_parity(P):
+    """bitmap_max_key (the dynamic-filter bitmap in front of probes) must
+    not change any result: fused-agg probes, multi-agg probes and
+    emit-mode joins all run with and without it on the same data; and a
+    build key outside [1, bitmap_max_key] must raise loudly (a silent
+    skip would turn probes of that key into wrong misses)."""
+    rng = np.random.RandomState(23)
+    n = 150_000
+    keys = (rng.permutation(2_000_000)[:n].astype(np.int64) + 1)
+    dates = rng.randint(1, 1 << 14, n).astype(np.int32)
+    probe_n = 300_000
+    pk = keys[rng.randint(0, n, probe_n)].astype(np.int64)
+    pk[::2] = rng.randint(1, 3_000_000, probe_n // 2 + probe_n % 2)
+    pv = rng.randint(1, 1000, probe_n).astype(np.int64)
+
+    def build(bmax, pack_bits=0, chained=False):
+        bp = P.PlanHashBuild()
+        bp.key_col = 0
+        bp.semijoin_table = -1
+        bp.n_payload = 1
+        bp.payload_col[0] = 1
+        bp.capacity_hint = n
+        bp.agg_table = 0 if chained else 1
+        bp.pack_bits = pack_bits
+        bp.bitmap_max_key = bmax
+        b = P.Operator(P.OP_HASH_BUILD, bp)
+        b.add_input(P.Page({"k": keys, "d": dates}))
+        b.finish()
+        return b
+
+    def agg_probe(b):
+        jp = P.PlanLookupJoin()
+        jp.table = b.table()
+        jp.key_col = 0
+        jp.mode = 1
+        jp.proj = P.Proj(P.PROJ_IDENT, 1, 0, 0)
+        jp.dec_only = 1
+        j = P.Operator(P.OP_LOOKUP_JOIN, jp)
+        j.add_input(P.Page({"k": pk, "v": pv}))
+        j.finish()
+        out = j.get_output(["key", "date", "sum", "f64", "cnt"])
+        j.destroy()
+        order = np.argsort(out["key"])
+        return {nm: out[nm][order] for nm in ("key", "date", "sum", "cnt")}
+
+    def emit_probe(b):
+        jp = P.PlanLookupJoin()
+        jp.table = b.table()
+        jp.key_col = 0
+        jp.mode = 0
+        jp.n_emit = 2
+        jp.emit_probe_cols[0] = 0
+        jp.emit_probe_cols[1] = 1
+        j = P.Operator(P.OP_LOOKUP_JOIN, jp)
+        j.add_input(P.Page({"k": pk, "v": pv}))
+        j.finish()
+        out = j.get_output(["k", "v", "bd"])
+        j.destroy()
+        order = np.lexsort((out["v"], out["k"]))
+        return {nm: out[nm][order] for nm in ("k", "v", "bd")}
+
+    def close(b):
+        from presto_amd.engine import lib
+        lib().c.pg_table_destroy(b.table())
+        b.destroy()
+
+    bmax = 2_000_001
+    for pack in (0, 14):
+        b0, b1 = build(0, pack), build(bmax, pack)
+        r0, r1 = agg_probe(b0), agg_probe(b1)
+        for nm in ("key", "date", "sum", "cnt"):
+            assert np.array_equal(r0[nm], r1[nm]), (pack, nm)
+        close(b0), close(b1)
+    # chained (emit-mode) build + probe
+    b0, b1 = build(0, chained=True), build(bmax, chained=True)
+    e0, e1 = emit_probe(b0), emit_probe(b1)
+    for nm in ("k", "v", "bd"):
+        assert np.array_equal(e0[nm], e1[nm]), nm
+    close(b0), close(b1)
+    # out-of-range build key: loud error, both build modes
+    for chained in (False, True):
+        with pytest.raises(RuntimeError, match="bitmap_max_key"):
+            b = build(1_000_000, chained=chained)  # half the keys outside
+            close(b)
