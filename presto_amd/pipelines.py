@@ -598,6 +598,7 @@ def q8(cust: Page, orders: Page, supp: Page, part: Page, li: Page):
     bo.n_payload = 1
     bo.payload_col[0] = orders.channel("orderdate")
     bo.capacity_hint = max(orders.n_rows // 8, 16)
+    bo.bitmap_max_key = okey_max(orders.n_rows)
     oo = Operator(OP_HASH_BUILD, bo)
     oo.add_input(orders)
     oo.finish()
@@ -863,6 +864,7 @@ def q17(part: Page, li: Page):
     bt.semijoin_table = -1
     bt.capacity_hint = max(part.n_rows // 16, 4096)
     bt.agg_table = 1
+    bt.bitmap_max_key = part.n_rows  # ~0.1% of 600M probes hit
     ot = Operator(OP_HASH_BUILD, bt)
     ot.add_input(part)
     ot.finish()
@@ -1652,6 +1654,7 @@ def q16(part: Page, ps: Page, supp: Page, type_name):
     ba.key_col = part.channel("partkey")
     ba.semijoin_table = oqual.table()
     ba.semijoin_col = part.channel("partkey")
+    ba.bitmap_max_key = part.n_rows
     ba.n_payload = 3
     ba.payload_col[0] = brandc
     ba.payload_col[1] = typec
@@ -1745,6 +1748,7 @@ def q10(cust_n: int, orders: Page, li: Page, limit=20):
     bo.n_payload = 1
     bo.payload_col[0] = orders.channel("custkey")
     bo.capacity_hint = max(orders.n_rows // 8, 16)
+    bo.bitmap_max_key = okey_max(orders.n_rows)
     oo = Operator(OP_HASH_BUILD, bo)
     oo.add_input(orders)
     oo.finish()
@@ -2093,6 +2097,7 @@ def q2(part: Page, ps: Page, supp: Page, s_abal, s_nat, limit=100):
     bm.semijoin_col = part.channel("partkey")
     bm.capacity_hint = max(part.n_rows // 64, 4096)
     bm.agg_table = 1
+    bm.bitmap_max_key = part.n_rows
     om = Operator(OP_HASH_BUILD, bm)
     om.add_input(part)
     om.finish()
