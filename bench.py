@@ -367,12 +367,15 @@ def setup_q3(B):
     ocols, ccols = B.orders_customer()
     ord_page = P.Page(ocols)
     cust_page = P.Page(ccols)
+    okb = pipelines.okey_max(
+        B.orc.lib.tpch_orders_count(C.c_double(B.sf_total)))
 
     def step():
         if B.world > 1:
             from presto_amd.dist import q3_distributed
             return q3_distributed(cust_page, ord_page, li_page, B.world,
-                                  B.rank, B.device, mode="dec")
+                                  B.rank, B.device, mode="dec",
+                                  okey_bound=okb)
         return pipelines.q3(cust_page, ord_page, li_page, mode="dec")
 
     return dict(step=step, n_rows=n_rows,
@@ -426,7 +429,10 @@ def setup_q5(B):
         if B.world > 1:
             from presto_amd.dist import q5_distributed
             return q5_distributed(cust_page, ord_page, supp_page, li5,
-                                  B.world, B.rank, B.device)
+                                  B.world, B.rank, B.device,
+                                  okey_bound=pipelines.okey_max(
+                                      B.orc.lib.tpch_orders_count(
+                                          C.c_double(B.sf_total))))
         return pipelines.q5(cust_page, ord_page, supp_page, li5)
 
     return dict(step=step, n_rows=n_rows,

@@ -83,7 +83,7 @@ def _gather_padded(t, cap, device, pad_value=0):
 
 
 def q3_distributed(cust_page, ord_page, li_page, world, rank, device,
-                   limit=10, mode="dec"):
+                   limit=10, mode="dec", okey_bound=0):
     """Distributed Q3 (BASELINE config 4): the repartition seam over RCCL.
 
     customer: local filter (mktsegment='BUILDING') -> broadcast key set
@@ -198,6 +198,8 @@ def q3_distributed(cust_page, ord_page, li_page, world, rank, device,
     b2p.payload_col[0] = 1
     b2p.capacity_hint = max(int(ocols["c0"].numel()) // 4, 16)
     b2p.agg_table = 1
+    b2p.bitmap_max_key = okey_bound  # global orderkey range (keys are
+    # hash-partitioned, so every rank sees global-range keys); 0 = off
     b2 = P.Operator(P.OP_HASH_BUILD, b2p)
     b2.add_input(P.Page({k: v for k, v in ocols.items()}))
     b2.finish()
@@ -275,7 +277,7 @@ def q3_distributed(cust_page, ord_page, li_page, world, rank, device,
 
 
 def q5_distributed(cust_page, ord_page, supp_page, li_page, world, rank,
-                   device, sf_hint=None):
+                   device, sf_hint=None, okey_bound=0):
     """Distributed Q5 (BASELINE config 4): the nation dimensions are
     replicated (all_reduce of the dense custkey/suppkey -> nationkey
     arrays — the broadcast-join analog), orders and lineitem are
@@ -399,6 +401,7 @@ def q5_distributed(cust_page, ord_page, supp_page, li_page, world, rank,
     b2p.payload_lookup_key_col = 1
     b2p.capacity_hint = max(int(ocols["c0"].numel()) + 64, 64)
     b2p.agg_table = 1
+    b2p.bitmap_max_key = okey_bound
     b2 = P.Operator(P.OP_HASH_BUILD, b2p)
     b2.add_input(P.Page(ocols))
     b2.finish()
