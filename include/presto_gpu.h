@@ -309,6 +309,18 @@ typedef struct {
      * [1, bitmap_max_key] are an error (the bitmap cannot represent
      * them, so probes would wrongly reject).  0 = no bitmap. */
     int64_t bitmap_max_key;
+    /* 1: a RANGE-GROUP table — the group-by domain is the dense key
+     * range [1, capacity_hint] itself (the perfect-hash GroupByHash
+     * analog for primary-key-shaped bigint group channels, cf.
+     * BigintGroupByHash.java:66-117 whose probe is exactly a
+     * value->bucket map).  No build input is scanned and no key slots
+     * are stored: a mode-1 fused-agg probe indexes its accumulators by
+     * key-1 directly (keys outside the range are misses, matching
+     * inner-join-with-FK semantics), and group extraction reconstructs
+     * the key from the slot index.  For probe inputs CLUSTERED by the
+     * key (lineitem by orderkey) the accumulator atomics become
+     * near-sequential instead of hash-scattered. */
+    int32_t range_group;
 } pg_plan_hash_build;
 
 typedef struct {

@@ -1775,10 +1775,15 @@ __global__ __launch_bounds__(256) void k_probe_agg(
             if (cur_slot == -2 || key != cur_key) {
                 flush();
                 cur_key = key;
-                cur_slot = kbits && !d_kbit_test(kbits, bmax, key)
-                               ? -1
-                               : d_tbl_find_tagged(keys, tags, mask,
-                                                   lmask, pbits, key);
+                if (!keys) /* range-group: acc index = key-1 */
+                    cur_slot = (uint64_t)(key - 1) <= (uint64_t)mask
+                                   ? key - 1
+                                   : -1;
+                else
+                    cur_slot = kbits && !d_kbit_test(kbits, bmax, key)
+                                   ? -1
+                                   : d_tbl_find_tagged(keys, tags, mask,
+                                                       lmask, pbits, key);
                 run_dec = 0;
                 run_flo = run_fhi = 0;
                 run_cnt = 0;
@@ -2352,10 +2357,15 @@ __global__ __launch_bounds__(256) void k_probe_agg_multi(
             if (cur_slot == -2 || key != cur_key) {
                 flush();
                 cur_key = key;
-                cur_slot = kbits && !d_kbit_test(kbits, bmax, key)
-                               ? -1
-                               : d_tbl_find_tagged(keys, tags, mask,
-                                                   lmask, pbits, key);
+                if (!keys) /* range-group: acc index = key-1 */
+                    cur_slot = (uint64_t)(key - 1) <= (uint64_t)mask
+                                   ? key - 1
+                                   : -1;
+                else
+                    cur_slot = kbits && !d_kbit_test(kbits, bmax, key)
+                                   ? -1
+                                   : d_tbl_find_tagged(keys, tags, mask,
+                                                       lmask, pbits, key);
 #pragma unroll
                 for (int a = 0; a < 6; a++) run[a] = 0;
                 run_cnt = 0;
@@ -2419,7 +2429,8 @@ __global__ __launch_bounds__(256) void k_groups_emit_multi(
         for (int w = 0; w < wid; w++) woff += wcnt[w];
         if (sel) {
             int64_t pos = woff + __popcll(m & ((1ull << lane) - 1));
-            int64_t kw = keys[i];
+            /* range-group tables store no keys: the slot index IS key-1 */
+            int64_t kw = keys ? keys[i] : i + 1;
             out_key[pos] = bp.pack_bits ? (kw >> bp.pack_bits) : kw;
             for (int o = 0; o < bp.n; o++) {
                 int64_t pv = bp.pack_bits && o == 0
@@ -2751,7 +2762,8 @@ __global__ __launch_bounds__(256) void k_groups_emit(
         for (int w = 0; w < wid; w++) woff += wcnt[w];
         if (sel) {
             int64_t pos = woff + __popcll(m & ((1ull << lane) - 1));
-            int64_t kw = keys[i];
+            /* range-group tables store no keys: the slot index IS key-1 */
+            int64_t kw = keys ? keys[i] : i + 1;
             out_key[pos] = bp.pack_bits ? (kw >> bp.pack_bits) : kw;
             int64_t r = bp.by_slot ? i : (int64_t)head[i];
             for (int o = 0; o < bp.n; o++) {
@@ -3259,6 +3271,8 @@ struct Table {
     int32_t n_acc = 0;
     DevBuf kbits;     /* key-presence bitmap (dynamic-filter analog) */
     int64_t bmax = 0; /* bitmap covers keys [1, bmax]; 0 = none */
+    bool range_group = false; /* dense-range group domain [1, cap]:
+                                 keys.p stays null, acc index = key-1 */
     bool slot_payloads = false; /* payloads indexed by slot (agg tables) */
     bool dense = false;          /* dense_array: payload[key-1], no hash */
     /* compacted build-row arrays: key + payloads */
@@ -3825,6 +3839,17 @@ struct BuildOp : Op {
     {
         t.reset(new Table());
         t->key_set_only = plan.key_set_only != 0;
+        if (plan.range_group) {
+            if (plan.capacity_hint <= 0)
+                throw std::runtime_error(
+                    "range_group needs capacity_hint = max key");
+            t->range_group = true;
+            t->slot_payloads = true; /* unique "rows" = the range */
+            t->cap = plan.capacity_hint;
+            t->mask = t->cap - 1;       /* bound, not a hash mask */
+            t->local_mask = t->mask;
+            return;
+        }
         if (plan.dense_array) {
             if (plan.n_payload != 1 && !plan.key_set_only)
                 throw std::runtime_error(
@@ -3976,6 +4001,10 @@ struct BuildOp : Op {
     }
     void add_input(const pg_page* in) override
     {
+        if (plan.range_group)
+            throw std::runtime_error(
+                "range_group tables take no build input (the domain is "
+                "the key range itself)");
         StagedPage sp;
         sp.stage(in);
         /* resolve payload tags on first page (payload 0 is u8 when it
@@ -4118,7 +4147,7 @@ struct BuildOp : Op {
     }
     void finish() override
     {
-        if (plan.dense_array) {
+        if (plan.dense_array || plan.range_group) {
             std::lock_guard<std::mutex> lk(g_mu);
             tbl = g_next_table++;
             g_tables[tbl] = std::move(t);

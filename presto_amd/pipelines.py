@@ -1220,15 +1220,16 @@ def q21(supp: Page, orders: Page, li: Page, limit=100):
     from .engine import lib
     tables = []
 
+    # The probe takes NOTHING from orders (no payload — the table is only
+    # the group-by domain, and every lineitem orderkey exists in orders by
+    # FK): a RANGE-GROUP table over the orderkey domain replaces the whole
+    # 150M-row build, and the orderkey-clustered lineitem makes the
+    # accumulator atomics near-sequential instead of hash-scattered.
     b = PlanHashBuild()
-    b.key_col = orders.channel("orderkey")
     b.semijoin_table = -1
-    b.capacity_hint = orders.n_rows + 64
-    b.agg_table = 1
-    b.fill_x10 = 13  # every probe hits (each line has its order):
-                     # miss clusters are irrelevant, so size tight
+    b.capacity_hint = okey_max(orders.n_rows)
+    b.range_group = 1
     otbl = Operator(OP_HASH_BUILD, b)
-    otbl.add_input(orders)
     otbl.finish()
     tables.append(otbl)
 

@@ -2031,3 +2031,70 @@ def test_bitmap_prefilter_parity(P):
         with pytest.raises(RuntimeError, match="bitmap_max_key"):
             b = build(1_000_000, chained=chained)  # half the keys outside
             close(b)
+
+
+def test_range_group_parity(P):
+    """A range_group table (dense-range group domain, no build input)
+    must produce exactly the groups a real agg-table build + multi-agg
+    probe produces when the build rows are the keys [1, K]; probe keys
+    outside the range are misses."""
+    rng = np.random.RandomState(41)
+    K = 100_000
+    probe_n = 400_000
+    pk = rng.randint(1, K + 1, probe_n).astype(np.int64)
+    pk[::7] = rng.randint(K + 1, K * 3, (probe_n + 6) // 7)  # out of range
+    pk.sort()  # clustered, like lineitem by orderkey
+    v1 = rng.randint(0, 100, probe_n).astype(np.int64)
+    flag = rng.randint(0, 2, probe_n).astype(np.int32)
+
+    def probe(tbl_op):
+        jp = P.PlanLookupJoin()
+        jp.table = tbl_op.table()
+        jp.key_col = 0
+        jp.mode = 1
+        jp.n_preds = 0
+        jp.preds[0] = P.Pred(2, P.CMP_EQ, 1, 0.0)
+        jp.n_aggs = 2
+        jp.aggs[0] = P.Agg(P.AGG_SUM_I64, P.Proj(P.PROJ_IDENT, 1, 0, 0), 0)
+        jp.aggs[1] = P.Agg(P.AGG_COUNT, P.Proj(P.PROJ_IDENT, 0, 0, 0), 0)
+        jp.agg_filter[0] = -1
+        jp.agg_filter[1] = 0
+        j = P.Operator(P.OP_LOOKUP_JOIN, jp)
+        j.add_input(P.Page({"k": pk, "v": v1, "f": flag}))
+        j.finish()
+        out = j.get_output(["key", "s", "c", "cnt"])
+        j.destroy()
+        order = np.argsort(out["key"])
+        return {nm: out[nm][order] for nm in ("key", "s", "c", "cnt")}
+
+    bp = P.PlanHashBuild()
+    bp.key_col = 0
+    bp.semijoin_table = -1
+    bp.capacity_hint = K
+    bp.agg_table = 1
+    b = P.Operator(P.OP_HASH_BUILD, bp)
+    b.add_input(P.Page({"k": np.arange(1, K + 1, dtype=np.int64)}))
+    b.finish()
+    br = P.PlanHashBuild()
+    br.semijoin_table = -1
+    br.capacity_hint = K
+    br.range_group = 1
+    r = P.Operator(P.OP_HASH_BUILD, br)
+    r.finish()
+    got_t, got_r = probe(b), probe(r)
+    from presto_amd.engine import lib
+    lib().c.pg_table_destroy(b.table())
+    lib().c.pg_table_destroy(r.table())
+    b.destroy(), r.destroy()
+    for nm in ("key", "s", "c", "cnt"):
+        assert np.array_equal(got_t[nm], got_r[nm]), nm
+    # numpy cross-check
+    sel = pk <= K
+    exp_keys = np.unique(pk[sel])
+    assert np.array_equal(got_r["key"], exp_keys)
+    import collections
+    es = collections.defaultdict(int)
+    for k, v in zip(pk[sel].tolist(), v1[sel].tolist()):
+        es[k] += v
+    assert np.array_equal(got_r["s"],
+                          np.array([es[k] for k in exp_keys.tolist()]))
