@@ -4683,8 +4683,29 @@ struct JoinOp : Op {
         if (chunk < 256) chunk = 256;
         /* single-pass extraction: groups <= inserted build rows, so the
          * output buffers are sized at that bound and the emit kernel
-         * reserves per-block spans off one cursor (n_rows read back) */
+         * reserves per-block spans off one cursor (n_rows read back).
+         * Range-group tables have no build rows — their bound is the
+         * whole range, far beyond the real group count, so size them
+         * with an exact count pass (one scan of the accumulators). */
         int64_t total = gt->n_rows;
+        if (gt->range_group) {
+            DevBuf d_counts;
+            d_counts.alloc(FLT_NB * 8);
+            const unsigned long long* cacc =
+                multi ? (const unsigned long long*)gt->acc_multi.p
+                      : (const unsigned long long*)gt->acc.p;
+            int64_t stride_w =
+                multi ? plan.n_aggs + 1 : gt->acc_words;
+            hipLaunchKernelGGL(k_groups_count, dim3(FLT_NB), dim3(256),
+                               0, g_stream, cacc, stride_w, stride_w - 1,
+                               cap, chunk, (int64_t*)d_counts.p);
+            std::vector<int64_t> hc(FLT_NB);
+            CHKV(hipMemcpyAsync(hc.data(), d_counts.p, FLT_NB * 8,
+                                hipMemcpyDeviceToHost, g_stream));
+            CHKV(hipStreamSynchronize(g_stream));
+            total = 0;
+            for (int b = 0; b < FLT_NB; b++) total += hc[b];
+        }
         DevBuf d_cursor;
         d_cursor.alloc(8);
         d_cursor.zero();
