@@ -382,6 +382,28 @@ typedef struct {
     int32_t n_aggs; /* 0 = legacy single proj; 1..6 multi */
     pg_agg aggs[6];
     int32_t agg_filter[6];
+    /* Multi-agg accumulator PACKING (the slot-word analog of pack_bits,
+     * applied to the accumulator state): aggregates whose per-GROUP
+     * totals the caller can bound (integrity facts like "<= 7 lineitems
+     * per order" and "suppkey < supplier count") share ONE u64 word as
+     * bit fields, so a probe flush issues one atomicAdd on one line
+     * instead of one per aggregate — the multi-agg probe is atomic-op-
+     * rate bound, not bandwidth bound.  acc_pack = 1 enables it;
+     * acc_pack_shift[a] is the field's bit offset in word 0 or -1 for a
+     * full u64 word of its own (appended after word 0 in agg order);
+     * acc_pack_width[a] the field width.  The count field lives at
+     * acc_pack_cnt_shift/width in word 0.  CONTRACT: the caller
+     * guarantees every per-group TOTAL fits its declared field; each
+     * probe flush additionally checks its own contribution against the
+     * width and raises the overflow error on violation (gross width
+     * mistakes fail loudly; the bound itself is the caller's integrity
+     * fact, as with pack_bits).  Output schema is unchanged — group
+     * extraction unpacks the fields. */
+    int32_t acc_pack;
+    int32_t acc_pack_shift[6];
+    int32_t acc_pack_width[6];
+    int32_t acc_pack_cnt_shift;
+    int32_t acc_pack_cnt_width;
 } pg_plan_lookup_join;
 
 typedef struct {

@@ -2329,7 +2329,14 @@ __global__ __launch_bounds__(256) void k_probe_agg_multi(
     unsigned long long* acc, unsigned long long* ovf)
 {
     const int n_aggs = plan.n_aggs;
-    const int64_t stride_w = n_aggs + 1;
+    /* packed layout: word 0 carries the bit-field aggs + count, own-word
+     * aggs follow in agg order (plan.acc_pack; see presto_gpu.h) */
+    int64_t stride_w = n_aggs + 1;
+    if (plan.acc_pack) {
+        stride_w = 1;
+        for (int a = 0; a < n_aggs; a++)
+            if (plan.acc_pack_shift[a] < 0) stride_w++;
+    }
     /* consecutive-row run dedup, as in k_probe_agg: one probe and one
      * set of atomics per run of equal keys */
     const int C = 4;
@@ -2344,12 +2351,40 @@ __global__ __launch_bounds__(256) void k_probe_agg_multi(
         auto flush = [&]() {
             if (cur_slot < 0 || run_cnt == 0) return;
             unsigned long long* slot = acc + (size_t)cur_slot * stride_w;
+            if (plan.acc_pack) {
+                unsigned long long w0 =
+                    (unsigned long long)run_cnt
+                    << plan.acc_pack_cnt_shift;
+                unsigned long long bad =
+                    (unsigned long long)run_cnt >>
+                    plan.acc_pack_cnt_width;
+                int w = 1;
 #pragma unroll
-            for (int a = 0; a < 6; a++) {
-                if (a >= n_aggs) break;
-                if (run[a]) d_atomic_add_dec_ck(slot + a, run[a], ovf);
+                for (int a = 0; a < 6; a++) {
+                    if (a >= n_aggs) break;
+                    if (plan.acc_pack_shift[a] >= 0) {
+                        /* negative contributions wrap the uint64 and
+                         * trip the width check -> loud overflow */
+                        bad |= (unsigned long long)run[a] >>
+                               plan.acc_pack_width[a];
+                        w0 += (unsigned long long)run[a]
+                              << plan.acc_pack_shift[a];
+                    } else {
+                        if (run[a])
+                            d_atomic_add_dec_ck(slot + w, run[a], ovf);
+                        w++;
+                    }
+                }
+                atomicAdd(&slot[0], w0);
+                if (bad && ovf) atomicAdd(ovf, 1ull);
+            } else {
+#pragma unroll
+                for (int a = 0; a < 6; a++) {
+                    if (a >= n_aggs) break;
+                    if (run[a]) d_atomic_add_dec_ck(slot + a, run[a], ovf);
+                }
+                atomicAdd(slot + n_aggs, (unsigned long long)run_cnt);
             }
-            atomicAdd(slot + n_aggs, (unsigned long long)run_cnt);
         };
         for (int64_t i = base; i < lim; i++) {
             if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
@@ -2392,12 +2427,35 @@ __global__ __launch_bounds__(256) void k_probe_agg_multi(
 
 /* group extraction for the multi-accumulator layout: key (+payloads),
  * then the n_aggs tick sums and the count, slot-ascending */
+struct acc_pack_desc {
+    int32_t on;        /* 0 = unpacked (cnt at word n_aggs) */
+    int32_t shift[6];  /* bit offset in word0, or -1 = own word */
+    int32_t width[6];
+    int32_t cnt_shift, cnt_width;
+};
+__device__ inline int64_t d_acc_field(const unsigned long long* slot,
+                                      const acc_pack_desc& pk, int a)
+{
+    if (pk.shift[a] >= 0)
+        return (int64_t)((slot[0] >> pk.shift[a]) &
+                         ((1ull << pk.width[a]) - 1));
+    int w = 1;
+    for (int b = 0; b < a; b++)
+        if (pk.shift[b] < 0) w++;
+    return (int64_t)slot[w];
+}
 __global__ __launch_bounds__(256) void k_groups_emit_multi(
     const int64_t* keys, const unsigned long long* acc, int64_t stride_w,
     int32_t n_aggs, build_payloads bp, int64_t cap, int64_t chunk,
     unsigned long long* cursor, int64_t* out_key, emit_outs payload_outs,
-    emit_outs agg_outs)
+    emit_outs agg_outs, acc_pack_desc pk)
 {
+    auto slot_cnt = [&](int64_t i) -> unsigned long long {
+        const unsigned long long* sl = acc + i * stride_w;
+        return pk.on ? (sl[0] >> pk.cnt_shift) &
+                           ((1ull << pk.cnt_width) - 1)
+                     : sl[n_aggs];
+    };
     const int64_t lo = (int64_t)blockIdx.x * chunk;
     const int64_t hi = min(lo + chunk, cap);
     const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
@@ -2406,7 +2464,7 @@ __global__ __launch_bounds__(256) void k_groups_emit_multi(
     {
         int64_t mycnt = 0;
         for (int64_t i = lo + threadIdx.x; i < hi; i += 256)
-            mycnt += acc[i * stride_w + n_aggs] > 0;
+            mycnt += slot_cnt(i) > 0;
         mycnt = d_bfly_i64(mycnt);
         if (lane == 0) wcnt[wid] = mycnt;
         __syncthreads();
@@ -2420,7 +2478,7 @@ __global__ __launch_bounds__(256) void k_groups_emit_multi(
     __syncthreads();
     for (int64_t base = lo; base < hi; base += 256) {
         int64_t i = base + 64 * wid + lane;
-        bool sel = i < hi && acc[i * stride_w + n_aggs] > 0;
+        bool sel = i < hi && slot_cnt(i) > 0;
         uint64_t m = d_ballot(sel);
         int wsum = __popcll(m);
         if (lane == 0) wcnt[wid] = wsum;
@@ -2460,9 +2518,18 @@ __global__ __launch_bounds__(256) void k_groups_emit_multi(
                             ((const double*)bp.ptr[o])[i];
                 }
             }
-            for (int a = 0; a <= n_aggs; a++)
-                ((int64_t*)agg_outs.ptr[a])[pos] =
-                    (int64_t)acc[i * stride_w + a];
+            if (pk.on) {
+                const unsigned long long* sl = acc + i * stride_w;
+                for (int a = 0; a < n_aggs; a++)
+                    ((int64_t*)agg_outs.ptr[a])[pos] =
+                        d_acc_field(sl, pk, a);
+                ((int64_t*)agg_outs.ptr[n_aggs])[pos] =
+                    (int64_t)slot_cnt(i);
+            } else {
+                for (int a = 0; a <= n_aggs; a++)
+                    ((int64_t*)agg_outs.ptr[a])[pos] =
+                        (int64_t)acc[i * stride_w + a];
+            }
         }
         __syncthreads();
         if (threadIdx.x == 0)
@@ -2698,6 +2765,7 @@ __global__ __launch_bounds__(256) void k_groupby_emit(
  * slot-ascending (stable compaction over the slot array) */
 __global__ __launch_bounds__(256) void k_groups_count(
     const unsigned long long* acc, int64_t stride_w, int64_t cnt_off,
+    int32_t cnt_shift, unsigned long long cnt_mask,
     int64_t cap, int64_t chunk, int64_t* block_counts)
 {
     const int64_t lo = (int64_t)blockIdx.x * chunk;
@@ -2706,7 +2774,9 @@ __global__ __launch_bounds__(256) void k_groups_count(
     int64_t cnt = 0;
     for (int64_t base = lo + 64 * wid; base < hi; base += 256) {
         int64_t i = base + lane;
-        bool sel = i < hi && acc[i * stride_w + cnt_off] > 0;
+        bool sel = i < hi &&
+                   ((acc[i * stride_w + cnt_off] >> cnt_shift) &
+                    cnt_mask) > 0;
         uint64_t m = d_ballot(sel);
         if (lane == 0) cnt += __popcll(m);
     }
@@ -3267,8 +3337,10 @@ struct Table {
     DevBuf keys, head, next, tags;
     DevBuf acc; /* per-slot accumulators, acc_words u64 each */
     int32_t acc_words = 0; /* 4 = [dec, flo, fhi, cnt]; 2 = [dec, cnt] */
-    DevBuf acc_multi; /* multi-agg probes: (n_acc+1) u64 per slot */
+    DevBuf acc_multi; /* multi-agg probes: acc_stride u64 per slot
+                         (n_acc+1 unpacked; fewer when acc_pack) */
     int32_t n_acc = 0;
+    int32_t acc_stride = 0;
     DevBuf kbits;     /* key-presence bitmap (dynamic-filter analog) */
     int64_t bmax = 0; /* bitmap covers keys [1, bmax]; 0 = none */
     bool range_group = false; /* dense-range group domain [1, cap]:
@@ -4326,12 +4398,43 @@ struct JoinOp : Op {
                 if (plan.agg_filter[a] >= plan.n_preds &&
                     plan.agg_filter[a] >= PG_MAX_PRED)
                     throw std::runtime_error("agg_filter out of range");
-            if (t->acc_multi.p && t->n_acc != plan.n_aggs)
+            int32_t want_stride = plan.n_aggs + 1;
+            if (plan.acc_pack) {
+                want_stride = 1;
+                unsigned long long used =
+                    ((1ull << plan.acc_pack_cnt_width) - 1)
+                    << plan.acc_pack_cnt_shift;
+                if (plan.acc_pack_cnt_width < 1 ||
+                    plan.acc_pack_cnt_shift +
+                        plan.acc_pack_cnt_width > 64)
+                    throw std::runtime_error("acc_pack: bad cnt field");
+                for (int a = 0; a < plan.n_aggs; a++) {
+                    if (plan.acc_pack_shift[a] < 0) {
+                        want_stride++;
+                        continue;
+                    }
+                    if (plan.acc_pack_width[a] < 1 ||
+                        plan.acc_pack_shift[a] +
+                            plan.acc_pack_width[a] > 64)
+                        throw std::runtime_error(
+                            "acc_pack: bad field bounds");
+                    unsigned long long fm =
+                        ((1ull << plan.acc_pack_width[a]) - 1)
+                        << plan.acc_pack_shift[a];
+                    if (used & fm)
+                        throw std::runtime_error(
+                            "acc_pack: overlapping fields");
+                    used |= fm;
+                }
+            }
+            if (t->acc_multi.p && (t->n_acc != plan.n_aggs ||
+                                   t->acc_stride != want_stride))
                 throw std::runtime_error(
                     "table already carries a different multi-agg layout");
             if (!t->acc_multi.p) {
                 t->n_acc = plan.n_aggs;
-                t->acc_multi.alloc((size_t)t->cap * (plan.n_aggs + 1) * 8);
+                t->acc_stride = want_stride;
+                t->acc_multi.alloc((size_t)t->cap * want_stride * 8);
                 t->acc_multi.zero();
                 CHKV(hipStreamSynchronize(g_stream));
             }
@@ -4695,9 +4798,18 @@ struct JoinOp : Op {
                 multi ? (const unsigned long long*)gt->acc_multi.p
                       : (const unsigned long long*)gt->acc.p;
             int64_t stride_w =
-                multi ? plan.n_aggs + 1 : gt->acc_words;
+                multi ? gt->acc_stride : gt->acc_words;
+            int64_t cnt_off = stride_w - 1;
+            int32_t cnt_shift = 0;
+            unsigned long long cnt_mask = ~0ull;
+            if (multi && plan.acc_pack) {
+                cnt_off = 0;
+                cnt_shift = plan.acc_pack_cnt_shift;
+                cnt_mask = (1ull << plan.acc_pack_cnt_width) - 1;
+            }
             hipLaunchKernelGGL(k_groups_count, dim3(FLT_NB), dim3(256),
-                               0, g_stream, cacc, stride_w, stride_w - 1,
+                               0, g_stream, cacc, stride_w, cnt_off,
+                               cnt_shift, cnt_mask,
                                cap, chunk, (int64_t*)d_counts.p);
             std::vector<int64_t> hc(FLT_NB);
             CHKV(hipMemcpyAsync(hc.data(), d_counts.p, FLT_NB * 8,
@@ -4743,13 +4855,21 @@ struct JoinOp : Op {
                 agg_outs.tag[a] = PG_T_I64;
             }
             op.pg.n_cols = nc;
+            acc_pack_desc pk{};
+            pk.on = plan.acc_pack;
+            for (int a = 0; a < 6; a++) {
+                pk.shift[a] = plan.acc_pack_shift[a];
+                pk.width[a] = plan.acc_pack_width[a];
+            }
+            pk.cnt_shift = plan.acc_pack_cnt_shift;
+            pk.cnt_width = plan.acc_pack_cnt_width;
             hipLaunchKernelGGL(
                 k_groups_emit_multi, dim3(FLT_NB), dim3(256), 0, g_stream,
                 (const int64_t*)gt->keys.p,
                 (const unsigned long long*)gt->acc_multi.p,
-                plan.n_aggs + 1, plan.n_aggs, bp, cap, chunk,
+                gt->acc_stride, plan.n_aggs, bp, cap, chunk,
                 (unsigned long long*)d_cursor.p,
-                (int64_t*)op.pg.cols[c_key].data, pl_outs, agg_outs);
+                (int64_t*)op.pg.cols[c_key].data, pl_outs, agg_outs, pk);
             unsigned long long n_out = 0;
             CHKV(hipMemcpyAsync(&n_out, d_cursor.p, 8,
                                 hipMemcpyDeviceToHost, g_stream));

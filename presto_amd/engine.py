@@ -92,7 +92,11 @@ class PlanLookupJoin(C.Structure):
                 ("n_group_vals", C.c_int32),
                 ("group_vals", C.c_uint8 * 8), ("dec_only", C.c_int32),
                 ("dec_min", C.c_int32), ("n_aggs", C.c_int32),
-                ("aggs", Agg * 6), ("agg_filter", C.c_int32 * 6)]
+                ("aggs", Agg * 6), ("agg_filter", C.c_int32 * 6),
+                ("acc_pack", C.c_int32), ("acc_pack_shift", C.c_int32 * 6),
+                ("acc_pack_width", C.c_int32 * 6),
+                ("acc_pack_cnt_shift", C.c_int32),
+                ("acc_pack_cnt_width", C.c_int32)]
 
 
 class PlanGroupBy(C.Structure):

@@ -1251,6 +1251,26 @@ def q21(supp: Page, orders: Page, li: Page, limit=100):
     jp.aggs[4] = Agg(AGG_SUM_I64, Proj(PROJ_MUL, sk, sk, 0), 0)
     for i, f in enumerate((-1, 0, 1, 1, 1)):
         jp.agg_filter[i] = f
+    # accumulator packing: the probe is atomic-op-rate bound, and every
+    # per-order total is bounded by integrity facts — at most 7 lineitems
+    # per order (TPC-H PK) and suppkey < supplier count — so a0/a2 (sums
+    # of suppkeys) and the three counts share ONE u64 word; only a4
+    # (sum of squared suppkeys) needs its own.  One flush = 1-2 atomics
+    # on 1-2 adjacent words instead of up to 6.
+    wsum = (7 * supp.n_rows).bit_length()
+    if 2 * wsum + 12 <= 64:
+        jp.acc_pack = 1
+        jp.acc_pack_shift[0] = 0           # a0: sum(suppkey) all
+        jp.acc_pack_width[0] = wsum
+        jp.acc_pack_shift[2] = wsum        # a2: sum(suppkey) late
+        jp.acc_pack_width[2] = wsum
+        jp.acc_pack_shift[1] = 2 * wsum    # a1: cnt F
+        jp.acc_pack_width[1] = 4
+        jp.acc_pack_shift[3] = 2 * wsum + 4  # a3: cnt late
+        jp.acc_pack_width[3] = 4
+        jp.acc_pack_shift[4] = -1          # a4: sum(sk^2) -> own word
+        jp.acc_pack_cnt_shift = 2 * wsum + 8
+        jp.acc_pack_cnt_width = 4
     jo = Operator(OP_LOOKUP_JOIN, jp)
     jo.add_input(li)
     jo.finish()

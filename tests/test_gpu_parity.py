@@ -2047,7 +2047,7 @@ def test_range_group_parity(P):
     v1 = rng.randint(0, 100, probe_n).astype(np.int64)
     flag = rng.randint(0, 2, probe_n).astype(np.int32)
 
-    def probe(tbl_op):
+    def probe(tbl_op, pack=False):
         jp = P.PlanLookupJoin()
         jp.table = tbl_op.table()
         jp.key_col = 0
@@ -2059,6 +2059,15 @@ def test_range_group_parity(P):
         jp.aggs[1] = P.Agg(P.AGG_COUNT, P.Proj(P.PROJ_IDENT, 0, 0, 0), 0)
         jp.agg_filter[0] = -1
         jp.agg_filter[1] = 0
+        if pack:
+            # per-group bounds: <= probe_n rows/group of v < 100 each
+            jp.acc_pack = 1
+            jp.acc_pack_shift[0] = 0
+            jp.acc_pack_width[0] = 30
+            jp.acc_pack_shift[1] = 30
+            jp.acc_pack_width[1] = 16
+            jp.acc_pack_cnt_shift = 46
+            jp.acc_pack_cnt_width = 16
         j = P.Operator(P.OP_LOOKUP_JOIN, jp)
         j.add_input(P.Page({"k": pk, "v": v1, "f": flag}))
         j.finish()
@@ -2081,13 +2090,22 @@ def test_range_group_parity(P):
     br.range_group = 1
     r = P.Operator(P.OP_HASH_BUILD, br)
     r.finish()
+    rp = P.PlanHashBuild()
+    rp.semijoin_table = -1
+    rp.capacity_hint = K
+    rp.range_group = 1
+    r2 = P.Operator(P.OP_HASH_BUILD, rp)
+    r2.finish()
     got_t, got_r = probe(b), probe(r)
+    got_p = probe(r2, pack=True)  # packed accumulators, same results
     from presto_amd.engine import lib
     lib().c.pg_table_destroy(b.table())
     lib().c.pg_table_destroy(r.table())
-    b.destroy(), r.destroy()
+    lib().c.pg_table_destroy(r2.table())
+    b.destroy(), r.destroy(), r2.destroy()
     for nm in ("key", "s", "c", "cnt"):
         assert np.array_equal(got_t[nm], got_r[nm]), nm
+        assert np.array_equal(got_t[nm], got_p[nm]), ("packed", nm)
     # numpy cross-check
     sel = pk <= K
     exp_keys = np.unique(pk[sel])
