@@ -1062,14 +1062,15 @@ def q18(orders: Page, li: Page, limit=100):
     import numpy as np
     from .engine import lib
 
+    # per-order qty sums take NOTHING from orders: range-group domain
+    # over the orderkey range replaces the 150M-row build, and the
+    # orderkey-clustered lineitem makes the accumulator atomics
+    # near-sequential (see q21)
     bo = PlanHashBuild()
-    bo.key_col = orders.channel("orderkey")
     bo.semijoin_table = -1
-    bo.capacity_hint = orders.n_rows + 64
-    bo.agg_table = 1
-    bo.fill_x10 = 13  # every lineitem probe hits: size tight
+    bo.capacity_hint = okey_max(orders.n_rows)
+    bo.range_group = 1
     oo = Operator(OP_HASH_BUILD, bo)
-    oo.add_input(orders)
     oo.finish()
 
     jp = PlanLookupJoin()
@@ -1100,6 +1101,7 @@ def q18(orders: Page, li: Page, limit=100):
     bb.n_payload = 1
     bb.payload_col[0] = 1
     bb.capacity_hint = max(big.n_rows, 16)
+    bb.bitmap_max_key = okey_max(orders.n_rows)  # ~6K hits of 150M probes
     ob = Operator(OP_HASH_BUILD, bb)
     ob.add_input_raw(big)
     ob.finish()
