@@ -1539,14 +1539,13 @@ def q13(n_cust: int, orders, max_count=64):
     from .engine import lib
     order_pages = [orders] if isinstance(orders, Page) else list(orders)
 
+    # the "table" is the keys 1..n_cust themselves: a range-group
+    # domain (no build, no hash; acc index = custkey-1)
     bc = PlanHashBuild()
-    bc.key_col = 0
     bc.semijoin_table = -1
-    bc.capacity_hint = n_cust + 64
-    bc.agg_table = 1
-    keys = Page({"custkey": np.arange(1, n_cust + 1, dtype=np.int64)})
+    bc.capacity_hint = n_cust
+    bc.range_group = 1
     oc = Operator(OP_HASH_BUILD, bc)
-    oc.add_input(keys)
     oc.finish()
 
     jp = PlanLookupJoin()
@@ -1575,13 +1574,10 @@ def q13(n_cust: int, orders, max_count=64):
     n_with_orders = groups.n_rows
 
     bh = PlanHashBuild()
-    bh.key_col = 0
     bh.semijoin_table = -1
-    bh.capacity_hint = max_count + 64
-    bh.agg_table = 1
-    counts = Page({"c": np.arange(1, max_count + 1, dtype=np.int64)})
+    bh.capacity_hint = max_count  # counts beyond max_count: range miss
+    bh.range_group = 1
     oh = Operator(OP_HASH_BUILD, bh)
-    oh.add_input(counts)
     oh.finish()
 
     jh = PlanLookupJoin()
@@ -2312,8 +2308,13 @@ def q4(orders: Page, li_dates: Page):
     bs.preds[0] = p
     bs.key_col = li_dates.channel("orderkey")
     bs.semijoin_table = -1
-    bs.capacity_hint = max(li_dates.n_rows // 2, 16)
+    # dense flag set over the orderkey range (one byte per possible key,
+    # ~600 MB at SF100): one predicated scan + flag stores replaces the
+    # 380M-key chained set insert, and the semijoin probe is one byte
+    # load (all lineitem orderkeys are in range by FK)
+    bs.capacity_hint = okey_max(orders.n_rows)
     bs.key_set_only = 1
+    bs.dense_array = 1
     b = Operator(OP_HASH_BUILD, bs)
     b.add_input(li_dates)
     b.finish()
