@@ -1127,6 +1127,19 @@ __global__ __launch_bounds__(256) void k_dense_fill(
     }
 }
 
+__global__ __launch_bounds__(256) void k_dense_fill32(
+    pg_page pg, pg_plan_hash_build plan, const int32_t* vals, int32_t* out,
+    int64_t cap)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < pg.n_rows; i += stride) {
+        if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
+        int64_t k = d_load_i64(pg.cols[plan.key_col], i);
+        if (k >= 1 && k <= cap) out[k - 1] = vals[i];
+    }
+}
+
 __global__ __launch_bounds__(256) void k_tbl_init(int64_t* keys, int32_t* head,
                                                   int64_t cap)
 {
@@ -1982,12 +1995,25 @@ __global__ __launch_bounds__(256) void k_acc_min_init(
 /* probe match counting (emit mode, pass 1): per-BLOCK totals over the
  * chunked stable geometry — no per-row counts array; the emit pass
  * recomputes per-row counts and places rows with in-window prefix sums */
+/* dense-array probe: present iff the payload value is nonzero (u8 flag
+ * sets store 1; i32 payloads carry values the caller guarantees nonzero,
+ * e.g. epoch-day dates) */
+__device__ inline int64_t d_dense_find(const uint8_t* dv8,
+                                       const int32_t* dv32, int64_t dcap,
+                                       int64_t key)
+{
+    if ((uint64_t)(key - 1) >= (uint64_t)dcap) return -1;
+    if (dv32) return dv32[key - 1] ? key - 1 : -1;
+    return dv8[key - 1] ? key - 1 : -1;
+}
+
 __global__ __launch_bounds__(256) void k_probe_count(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
     const uint8_t* tags, const int32_t* head, const int32_t* next,
     int64_t mask, int64_t lmask, int32_t pbits,
-    const unsigned long long* kbits, int64_t bmax, int64_t chunk,
-    int64_t* block_counts)
+    const unsigned long long* kbits, int64_t bmax,
+    const uint8_t* dv8, const int32_t* dv32, int64_t dcap,
+    int64_t chunk, int64_t* block_counts)
 {
     const int64_t n = pg.n_rows;
     const int64_t lo = (int64_t)blockIdx.x * chunk;
@@ -1996,9 +2022,13 @@ __global__ __launch_bounds__(256) void k_probe_count(
     for (int64_t i = lo + threadIdx.x; i < hi; i += 256) {
         if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
         int64_t key = d_load_i64(pg.cols[plan.key_col], i);
-        if (kbits && !d_kbit_test(kbits, bmax, key)) continue;
-        int64_t sl = d_tbl_find_tagged(keys, tags, mask, lmask, pbits,
-                                       key);
+        int64_t sl;
+        if (dv8 || dv32) {
+            sl = d_dense_find(dv8, dv32, dcap, key);
+        } else {
+            if (kbits && !d_kbit_test(kbits, bmax, key)) continue;
+            sl = d_tbl_find_tagged(keys, tags, mask, lmask, pbits, key);
+        }
         if (sl >= 0) {
             if (head)
                 for (int32_t r = head[sl]; r >= 0; r = next[r]) total++;
@@ -2027,7 +2057,9 @@ __global__ __launch_bounds__(256) void k_probe_emit(
     pg_page pg, pg_plan_lookup_join plan, const int64_t* keys,
     const uint8_t* tags, const int32_t* head, const int32_t* next,
     int64_t mask, int64_t lmask, int32_t pbits,
-    const unsigned long long* kbits, int64_t bmax, int64_t chunk,
+    const unsigned long long* kbits, int64_t bmax,
+    const uint8_t* dv8, const int32_t* dv32, int64_t dcap,
+    int64_t chunk,
     const int64_t* block_offs, emit_outs probe_outs, build_payloads bp,
     emit_outs build_outs)
 {
@@ -2045,10 +2077,13 @@ __global__ __launch_bounds__(256) void k_probe_emit(
         int32_t c = 0;
         if (i < hi && d_eval_preds(pg, plan.preds, plan.n_preds, i)) {
             int64_t key = d_load_i64(pg.cols[plan.key_col], i);
-            sl = kbits && !d_kbit_test(kbits, bmax, key)
-                     ? -1
-                     : d_tbl_find_tagged(keys, tags, mask, lmask, pbits,
-                                         key);
+            if (dv8 || dv32)
+                sl = d_dense_find(dv8, dv32, dcap, key);
+            else
+                sl = kbits && !d_kbit_test(kbits, bmax, key)
+                         ? -1
+                         : d_tbl_find_tagged(keys, tags, mask, lmask,
+                                             pbits, key);
             if (sl >= 0) {
                 if (head)
                     for (int32_t r = head[sl]; r >= 0; r = next[r]) c++;
@@ -3345,6 +3380,8 @@ struct Table {
     int64_t bmax = 0; /* bitmap covers keys [1, bmax]; 0 = none */
     bool range_group = false; /* dense-range group domain [1, cap]:
                                  keys.p stays null, acc index = key-1 */
+    bool dense_alloc_pending = false; /* dense payload tag set on first
+                                         input page */
     bool slot_payloads = false; /* payloads indexed by slot (agg tables) */
     bool dense = false;          /* dense_array: payload[key-1], no hash */
     /* compacted build-row arrays: key + payloads */
@@ -3929,9 +3966,16 @@ struct BuildOp : Op {
             t->dense = true;
             t->cap = plan.capacity_hint;
             t->payload.emplace_back();
+            /* payload tag resolved on first add_input (U8 or I32 —
+             * an I32 dense payload doubles as the presence flag, so
+             * its values must be nonzero; U8 flag sets stay as-is) */
             t->ptag.push_back(PG_T_U8);
-            t->payload.back().alloc((size_t)t->cap);
-            if (plan.key_set_only) t->payload.back().zero();
+            t->dense_alloc_pending = true;
+            if (plan.key_set_only) {
+                t->payload.back().alloc((size_t)t->cap);
+                t->payload.back().zero();
+                t->dense_alloc_pending = false;
+            }
             return;
         }
         if (plan.agg_table) {
@@ -4094,16 +4138,38 @@ struct BuildOp : Op {
                     "pre-filter with a FILTER_PROJECT semijoin instead");
             if (sp.pg.cols[plan.key_col].tag != PG_T_I64 ||
                 (!plan.key_set_only &&
-                 sp.pg.cols[plan.payload_col[0]].tag != PG_T_U8))
+                 sp.pg.cols[plan.payload_col[0]].tag != PG_T_U8 &&
+                 sp.pg.cols[plan.payload_col[0]].tag != PG_T_I32))
                 throw std::runtime_error(
-                    "dense_array expects I64 keys and a U8 payload");
-            hipLaunchKernelGGL(
-                k_dense_fill, dim3(2048), dim3(256), 0, g_stream, sp.pg,
-                plan,
-                plan.key_set_only
-                    ? nullptr
-                    : (const uint8_t*)sp.pg.cols[plan.payload_col[0]].data,
-                (uint8_t*)t->payload[0].p, t->cap);
+                    "dense_array expects I64 keys and a U8/I32 payload");
+            int32_t ptag = plan.key_set_only
+                               ? PG_T_U8
+                               : sp.pg.cols[plan.payload_col[0]].tag;
+            if (t->dense_alloc_pending) {
+                t->ptag[0] = ptag;
+                t->payload[0].alloc((size_t)t->cap *
+                                    (ptag == PG_T_I32 ? 4 : 1));
+                t->payload[0].zero();
+                t->dense_alloc_pending = false;
+            } else if (t->ptag[0] != ptag) {
+                throw std::runtime_error(
+                    "dense_array payload tag changed across pages");
+            }
+            if (ptag == PG_T_I32)
+                hipLaunchKernelGGL(
+                    k_dense_fill32, dim3(2048), dim3(256), 0, g_stream,
+                    sp.pg, plan,
+                    (const int32_t*)sp.pg.cols[plan.payload_col[0]].data,
+                    (int32_t*)t->payload[0].p, t->cap);
+            else
+                hipLaunchKernelGGL(
+                    k_dense_fill, dim3(2048), dim3(256), 0, g_stream,
+                    sp.pg, plan,
+                    plan.key_set_only
+                        ? nullptr
+                        : (const uint8_t*)
+                              sp.pg.cols[plan.payload_col[0]].data,
+                    (uint8_t*)t->payload[0].p, t->cap);
             CHKV(hipStreamSynchronize(g_stream));
             return;
         }
@@ -4376,6 +4442,9 @@ struct JoinOp : Op {
         if (t->key_set_only)
             throw std::runtime_error(
                 "cannot probe a key-set-only table");
+        if (t->dense && plan.mode != 0)
+            throw std::runtime_error(
+                "dense-array tables support emit-mode joins only");
         if (t->pack_bits &&
             !(plan.mode == 1 || plan.mode == 2 ||
               (plan.mode == 0 && t->slot_payloads)))
@@ -4662,7 +4731,15 @@ struct JoinOp : Op {
                            (const int32_t*)t->head.p,
                            (const int32_t*)t->next.p, t->mask,
                            t->local_mask, t->pack_bits,
-                           (const unsigned long long*)t->kbits.p, t->bmax, chunk,
+                           (const unsigned long long*)t->kbits.p, t->bmax,
+                           (const uint8_t*)(t->dense && t->ptag[0] == PG_T_U8
+                                              ? t->payload[0].p
+                                              : nullptr),
+                           (const int32_t*)(t->dense &&
+                                                    t->ptag[0] == PG_T_I32
+                                                ? t->payload[0].p
+                                                : nullptr),
+                           t->dense ? t->cap : 0, chunk,
                            (int64_t*)d_counts.p);
         std::vector<int64_t> h(FLT_NB);
         CHKV(hipMemcpyAsync(h.data(), d_counts.p, FLT_NB * 8,
@@ -4697,7 +4774,7 @@ struct JoinOp : Op {
         }
         build_payloads bp{};
         bp.n = (int32_t)t->payload.size();
-        bp.by_slot = t->slot_payloads ? 1 : 0;
+        bp.by_slot = (t->slot_payloads || t->dense) ? 1 : 0;
         bp.pack_bits = t->pack_bits;
         emit_outs bouts{};
         bouts.n = bp.n;
@@ -4720,7 +4797,15 @@ struct JoinOp : Op {
                            (const int32_t*)t->head.p,
                            (const int32_t*)t->next.p, t->mask,
                            t->local_mask, t->pack_bits,
-                           (const unsigned long long*)t->kbits.p, t->bmax, chunk,
+                           (const unsigned long long*)t->kbits.p, t->bmax,
+                           (const uint8_t*)(t->dense && t->ptag[0] == PG_T_U8
+                                              ? t->payload[0].p
+                                              : nullptr),
+                           (const int32_t*)(t->dense &&
+                                                    t->ptag[0] == PG_T_I32
+                                                ? t->payload[0].p
+                                                : nullptr),
+                           t->dense ? t->cap : 0, chunk,
                            (const int64_t*)d_offs.p, pouts, bp, bouts);
         hot_end();
         CHKV(hipStreamSynchronize(g_stream));

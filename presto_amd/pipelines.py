@@ -1460,12 +1460,17 @@ def q9(part: Page, supp: Page, orders: Page, ps: Page, li: Page):
     jb.add_input_raw(pb)
     pc = jb.get_output_raw()  # [ok, qty, ep, dc, cost, nat]
 
+    # orderkey -> orderdate as a dense i32 array over the orderkey
+    # range (2.4 GB at SF100): one scan + direct stores replace the
+    # 150M-row chained insert, and each probe is one i32 load (presence
+    # = nonzero payload; epoch-day dates are always nonzero)
     bo = PlanHashBuild()
     bo.key_col = orders.channel("orderkey")
     bo.semijoin_table = -1
     bo.n_payload = 1
     bo.payload_col[0] = orders.channel("orderdate")
-    bo.capacity_hint = orders.n_rows
+    bo.capacity_hint = okey_max(orders.n_rows)
+    bo.dense_array = 1
     oo = Operator(OP_HASH_BUILD, bo)
     oo.add_input(orders)
     oo.finish()

@@ -2116,3 +2116,51 @@ def test_range_group_parity(P):
         es[k] += v
     assert np.array_equal(got_r["s"],
                           np.array([es[k] for k in exp_keys.tolist()]))
+
+
+def test_dense_emit_join_parity(P):
+    """Emit-mode joins over dense-array tables (i32 payload; presence =
+    nonzero value) must match the chained-table join on the same data."""
+    rng = np.random.RandomState(53)
+    K = 50_000
+    bkeys = np.unique(rng.randint(1, K + 1, K // 2)).astype(np.int64)
+    bvals = rng.randint(1, 10_000, len(bkeys)).astype(np.int32)
+    pk = rng.randint(1, 2 * K, 200_000).astype(np.int64)
+    pidx = np.arange(200_000, dtype=np.int64)
+
+    def run(dense):
+        bp = P.PlanHashBuild()
+        bp.key_col = 0
+        bp.semijoin_table = -1
+        bp.n_payload = 1
+        bp.payload_col[0] = 1
+        bp.capacity_hint = K if dense else len(bkeys)
+        bp.dense_array = 1 if dense else 0
+        b = P.Operator(P.OP_HASH_BUILD, bp)
+        b.add_input(P.Page({"k": bkeys, "v": bvals}))
+        b.finish()
+        jp = P.PlanLookupJoin()
+        jp.table = b.table()
+        jp.key_col = 0
+        jp.mode = 0
+        jp.n_emit = 1
+        jp.emit_probe_cols[0] = 1
+        j = P.Operator(P.OP_LOOKUP_JOIN, jp)
+        j.add_input(P.Page({"k": pk, "i": pidx}))
+        out = j.get_output(["i", "bv"])
+        j.destroy()
+        from presto_amd.engine import lib
+        lib().c.pg_table_destroy(b.table())
+        b.destroy()
+        order = np.argsort(out["i"])
+        return out["i"][order], out["bv"][order]
+
+    ih, vh = run(False)
+    id_, vd = run(True)
+    assert np.array_equal(ih, id_)
+    assert np.array_equal(vh, vd)
+    # numpy cross-check
+    m = {int(k): int(v) for k, v in zip(bkeys, bvals)}
+    sel = np.array([int(k) in m for k in pk])
+    assert np.array_equal(ih, pidx[sel])
+    assert np.array_equal(vh, np.array([m[int(k)] for k in pk[sel]]))
