@@ -1959,8 +1959,9 @@ __global__ __launch_bounds__(256) void k_probe_agg_pay(
         int64_t r = head1 ? (int64_t)head1[s1] : s1;
         if (r < 0) continue;
         int64_t g = pay1[r];
-        int64_t s2 = d_tbl_find_tagged(keys2, tags2, mask2, lmask2, 0,
-                                       g);
+        int64_t s2 =
+            keys2 ? d_tbl_find_tagged(keys2, tags2, mask2, lmask2, 0, g)
+                  : ((uint64_t)(g - 1) <= (uint64_t)mask2 ? g - 1 : -1);
         if (s2 < 0) continue;
         pg_agg ag;
         ag.proj = plan.proj;

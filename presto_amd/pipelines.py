@@ -1777,14 +1777,12 @@ def q10(cust_n: int, orders: Page, li: Page, limit=20):
     oo.add_input(orders)
     oo.finish()
 
+    # the customer group domain is the keys 1..cust_n themselves
     bc = PlanHashBuild()
-    bc.key_col = 0
     bc.semijoin_table = -1
-    bc.capacity_hint = cust_n + 64
-    bc.agg_table = 1
-    keys = Page({"custkey": np.arange(1, cust_n + 1, dtype=np.int64)})
+    bc.capacity_hint = cust_n
+    bc.range_group = 1
     oc = Operator(OP_HASH_BUILD, bc)
-    oc.add_input(keys)
     oc.finish()
 
     # one fused pass: probe orders by orderkey, group the revenue by the
