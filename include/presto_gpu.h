@@ -321,6 +321,12 @@ typedef struct {
      * key (lineitem by orderkey) the accumulator atomics become
      * near-sequential instead of hash-scattered. */
     int32_t range_group;
+    /* dense_array only: constant added to the payload value at fill
+     * time (probe-side consumers see value+bias).  Lets 0-valued
+     * payloads (nationkey 0, priority 0) coexist with "presence =
+     * nonzero" dense probing; the pipeline accounts for the bias in
+     * its downstream predicates. */
+    int32_t dense_payload_bias;
 } pg_plan_hash_build;
 
 typedef struct {

@@ -1123,7 +1123,27 @@ __global__ __launch_bounds__(256) void k_dense_fill(
     for (; i < pg.n_rows; i += stride) {
         if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
         int64_t k = d_load_i64(pg.cols[plan.key_col], i);
-        if (k >= 1 && k <= cap) out[k - 1] = vals ? vals[i] : 1;
+        if (k >= 1 && k <= cap)
+            out[k - 1] =
+                vals ? (uint8_t)(vals[i] + plan.dense_payload_bias) : 1;
+    }
+}
+
+/* dense fill with the payload fetched THROUGH a dense u8 dimension
+ * (q7's orderkey -> customer nation: out[ok-1] = cust_nat[ck-1]+bias) */
+__global__ __launch_bounds__(256) void k_dense_fill_lu(
+    pg_page pg, pg_plan_hash_build plan, const uint8_t* lu, int64_t lu_cap,
+    uint8_t* out, int64_t cap)
+{
+    int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; i < pg.n_rows; i += stride) {
+        if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
+        int64_t k = d_load_i64(pg.cols[plan.key_col], i);
+        int64_t k2 = d_load_i64(pg.cols[plan.payload_lookup_key_col], i);
+        if (k >= 1 && k <= cap && k2 >= 1 && k2 <= lu_cap)
+            out[k - 1] =
+                (uint8_t)(lu[k2 - 1] + plan.dense_payload_bias);
     }
 }
 
@@ -1136,7 +1156,8 @@ __global__ __launch_bounds__(256) void k_dense_fill32(
     for (; i < pg.n_rows; i += stride) {
         if (!d_eval_preds(pg, plan.preds, plan.n_preds, i)) continue;
         int64_t k = d_load_i64(pg.cols[plan.key_col], i);
-        if (k >= 1 && k <= cap) out[k - 1] = vals[i];
+        if (k >= 1 && k <= cap)
+            out[k - 1] = vals[i] + plan.dense_payload_bias;
     }
 }
 
@@ -4156,7 +4177,22 @@ struct BuildOp : Op {
                 throw std::runtime_error(
                     "dense_array payload tag changed across pages");
             }
-            if (ptag == PG_T_I32)
+            if (plan.payload_lookup_table > 0) {
+                const Table* lu;
+                {
+                    std::lock_guard<std::mutex> lk(g_mu);
+                    auto it = g_tables.find(plan.payload_lookup_table);
+                    if (it == g_tables.end() || !it->second->dense ||
+                        it->second->ptag[0] != PG_T_U8)
+                        throw std::runtime_error(
+                            "dense fill lookup needs a dense u8 table");
+                    lu = it->second.get();
+                }
+                hipLaunchKernelGGL(
+                    k_dense_fill_lu, dim3(2048), dim3(256), 0, g_stream,
+                    sp.pg, plan, (const uint8_t*)lu->payload[0].p,
+                    lu->cap, (uint8_t*)t->payload[0].p, t->cap);
+            } else if (ptag == PG_T_I32)
                 hipLaunchKernelGGL(
                     k_dense_fill32, dim3(2048), dim3(256), 0, g_stream,
                     sp.pg, plan,

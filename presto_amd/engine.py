@@ -79,7 +79,8 @@ class PlanHashBuild(C.Structure):
                 ("payload_lookup_key_col", C.c_int32),
                 ("agg_table", C.c_int32), ("pack_bits", C.c_int32),
                 ("fill_x10", C.c_int32), ("bitmap_max_key", C.c_int64),
-                ("range_group", C.c_int32)]
+                ("range_group", C.c_int32),
+                ("dense_payload_bias", C.c_int32)]
 
 
 class PlanLookupJoin(C.Structure):

@@ -451,6 +451,11 @@ def q7(cust: Page, orders: Page, supp: Page, li: Page):
     o1.add_input(cust)
     o1.finish()
 
+    # orderkey -> customer nation as a dense u8 array over the orderkey
+    # range (one scan + one byte per possible key; nation is fetched
+    # through the dense customer dimension during the fill).  Presence =
+    # nonzero, so the stored value is nation+1 — downstream predicates
+    # are biased accordingly.
     b2 = PlanHashBuild()
     b2.key_col = orders.channel("orderkey")
     b2.semijoin_table = -1
@@ -458,11 +463,9 @@ def q7(cust: Page, orders: Page, supp: Page, li: Page):
     b2.payload_col[0] = 0
     b2.payload_lookup_table = o1.table()
     b2.payload_lookup_key_col = orders.channel("custkey")
-    b2.capacity_hint = orders.n_rows + 64
-    b2.agg_table = 1
-    b2.fill_x10 = 13  # every lineitem probe hits: size tight
-    # (no pack_bits: this table feeds an EMIT-mode join, which reads
-    # payloads through the slot arrays)
+    b2.capacity_hint = okey_max(orders.n_rows)
+    b2.dense_array = 1
+    b2.dense_payload_bias = 1
     o2 = Operator(OP_HASH_BUILD, b2)
     o2.add_input(orders)
     o2.finish()
@@ -513,7 +516,7 @@ def q7(cust: Page, orders: Page, supp: Page, li: Page):
             p = PlanHashAggSmall()
             p.n_preds = 4
             p.preds[0] = Pred(4, CMP_EQ, sn, 0.0)
-            p.preds[1] = Pred(3, CMP_EQ, cn, 0.0)
+            p.preds[1] = Pred(3, CMP_EQ, cn + 1, 0.0)  # dense bias
             p.preds[2] = Pred(2, CMP_GE, ylo, 0.0)
             p.preds[3] = Pred(2, CMP_LE, yhi, 0.0)
             p.n_keys = 0
@@ -773,12 +776,17 @@ def q12(orders: Page, li: Page):
     """Q12 shipmode priority (q12.sql): late-commit lineitems received in
     1994 joined to orders; counts per (shipmode in {MAIL=4, SHIP=6},
     priority class).  Returns {mode_id: (high, low)}."""
+    # orderkey -> priority as a dense u8 array over the orderkey range
+    # (presence = nonzero, so priority+1 is stored; the class-split
+    # predicates below are biased accordingly)
     bo = PlanHashBuild()
     bo.key_col = orders.channel("orderkey")
     bo.semijoin_table = -1
     bo.n_payload = 1
     bo.payload_col[0] = orders.channel("priority")
-    bo.capacity_hint = orders.n_rows
+    bo.capacity_hint = okey_max(orders.n_rows)
+    bo.dense_array = 1
+    bo.dense_payload_bias = 1
     oo = Operator(OP_HASH_BUILD, bo)
     oo.add_input(orders)
     oo.finish()
@@ -813,7 +821,8 @@ def q12(orders: Page, li: Page):
     jpage = jo.get_output_raw()  # [shipmode, priority]
 
     res = {4: [0, 0], 6: [0, 0]}
-    for cls, (op_, val) in enumerate(((CMP_LE, 1), (CMP_GE, 2))):
+    for cls, (op_, val) in enumerate(((CMP_LE, 2), (CMP_GE, 3))):
+        # priority stored +1 (dense bias): URGENT/HIGH {0,1} -> {1,2}
         ag = PlanHashAggSmall()
         ag.n_preds = 1
         ag.preds[0] = Pred(1, op_, val, 0.0)  # priority class split
