@@ -4159,14 +4159,15 @@ struct BuildOp : Op {
                     "dense_array builds do not evaluate semijoins — "
                     "pre-filter with a FILTER_PROJECT semijoin instead");
             if (sp.pg.cols[plan.key_col].tag != PG_T_I64 ||
-                (!plan.key_set_only &&
+                (!plan.key_set_only && !plan.payload_lookup_table &&
                  sp.pg.cols[plan.payload_col[0]].tag != PG_T_U8 &&
                  sp.pg.cols[plan.payload_col[0]].tag != PG_T_I32))
                 throw std::runtime_error(
                     "dense_array expects I64 keys and a U8/I32 payload");
-            int32_t ptag = plan.key_set_only
-                               ? PG_T_U8
-                               : sp.pg.cols[plan.payload_col[0]].tag;
+            int32_t ptag =
+                plan.key_set_only || plan.payload_lookup_table
+                    ? PG_T_U8
+                    : sp.pg.cols[plan.payload_col[0]].tag;
             if (t->dense_alloc_pending) {
                 t->ptag[0] = ptag;
                 t->payload[0].alloc((size_t)t->cap *
