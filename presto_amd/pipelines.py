@@ -1562,13 +1562,21 @@ def q13(n_cust: int, orders, max_count=64):
     oc = Operator(OP_HASH_BUILD, bc)
     oc.finish()
 
+    # count-only probe with a packed 1-word accumulator: per-customer
+    # order counts are bounded far below 2^16, so count+cnt share one
+    # u64 (one atomic per row into a ~120 MB L3-resident array)
     jp = PlanLookupJoin()
     jp.table = oc.table()
     jp.key_col = 0
     jp.mode = 1
-    jp.proj = Proj(PROJ_IDENT, 0, 0, 0)
-    jp.dec_scale = 0
-    jp.dec_only = 1
+    jp.n_aggs = 1
+    jp.aggs[0] = Agg(AGG_COUNT, Proj(PROJ_IDENT, 0, 0, 0), 0)
+    jp.agg_filter[0] = -1
+    jp.acc_pack = 1
+    jp.acc_pack_shift[0] = 0
+    jp.acc_pack_width[0] = 16
+    jp.acc_pack_cnt_shift = 16
+    jp.acc_pack_cnt_width = 16
     jo = Operator(OP_LOOKUP_JOIN, jp)
     fp = PlanFilterProject()
     fp.n_preds = 1
@@ -1584,7 +1592,7 @@ def q13(n_cust: int, orders, max_count=64):
         okp = f.get_output_raw()  # [custkey] of qualifying orders
         jo.add_input_raw(okp)
     jo.finish()
-    groups = jo.get_output_raw()  # [custkey, sum, f64, cnt]
+    groups = jo.get_output_raw()  # [custkey, n_orders, cnt]
     n_with_orders = groups.n_rows
 
     bh = PlanHashBuild()
@@ -1596,15 +1604,20 @@ def q13(n_cust: int, orders, max_count=64):
 
     jh = PlanLookupJoin()
     jh.table = oh.table()
-    jh.key_col = 3  # the per-customer count
+    jh.key_col = 1  # the per-customer count
     jh.mode = 1
-    jh.proj = Proj(PROJ_IDENT, 3, 0, 0)
-    jh.dec_scale = 0
-    jh.dec_only = 1
+    jh.n_aggs = 1
+    jh.aggs[0] = Agg(AGG_COUNT, Proj(PROJ_IDENT, 0, 0, 0), 0)
+    jh.agg_filter[0] = -1
+    jh.acc_pack = 1
+    jh.acc_pack_shift[0] = 0
+    jh.acc_pack_width[0] = 32
+    jh.acc_pack_cnt_shift = 32
+    jh.acc_pack_cnt_width = 32
     jo2 = Operator(OP_LOOKUP_JOIN, jh)
     jo2.add_input_raw(groups)
     jo2.finish()
-    hist = jo2.get_output(["c_count", "sum", "f64", "custdist"])
+    hist = jo2.get_output(["c_count", "n_cust", "custdist"])
     rows = [(int(hist["c_count"][i]), int(hist["custdist"][i]))
             for i in range(len(hist["c_count"]))]
     rows.append((0, n_cust - n_with_orders))
