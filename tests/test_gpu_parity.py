@@ -2164,3 +2164,42 @@ def test_dense_emit_join_parity(P):
     sel = np.array([int(k) in m for k in pk])
     assert np.array_equal(ih, pidx[sel])
     assert np.array_equal(vh, np.array([m[int(k)] for k in pk[sel]]))
+
+
+def test_packed_emit_join(P):
+    """Emit-mode join over a PACKED slot-payload table (payload in the
+    key word): q9's composite partsupp shape in miniature."""
+    rng = np.random.RandomState(61)
+    K = 40_000
+    bkeys = np.arange(1, K + 1, dtype=np.int64)
+    bvals = rng.randint(1, 1 << 14, K).astype(np.int64)
+    pk = rng.randint(1, K + 1, 150_000).astype(np.int64)
+    pidx = np.arange(150_000, dtype=np.int64)
+    bp = P.PlanHashBuild()
+    bp.key_col = 0
+    bp.semijoin_table = -1
+    bp.n_payload = 1
+    bp.payload_col[0] = 1
+    bp.capacity_hint = K
+    bp.agg_table = 1
+    bp.pack_bits = 14
+    bp.fill_x10 = 13
+    b = P.Operator(P.OP_HASH_BUILD, bp)
+    b.add_input(P.Page({"k": bkeys, "v": bvals}))
+    b.finish()
+    jp = P.PlanLookupJoin()
+    jp.table = b.table()
+    jp.key_col = 0
+    jp.mode = 0
+    jp.n_emit = 1
+    jp.emit_probe_cols[0] = 1
+    j = P.Operator(P.OP_LOOKUP_JOIN, jp)
+    j.add_input(P.Page({"k": pk, "i": pidx}))
+    out = j.get_output(["i", "bv"])
+    j.destroy()
+    from presto_amd.engine import lib
+    lib().c.pg_table_destroy(b.table())
+    b.destroy()
+    order = np.argsort(out["i"])
+    assert np.array_equal(out["i"][order], pidx)
+    assert np.array_equal(out["bv"][order], bvals[pk - 1])
