@@ -1403,63 +1403,50 @@ def q9(part: Page, supp: Page, orders: Page, ps: Page, li: Page):
     og.add_input(part)
     og.finish()
 
-    # (partkey, suppkey) as ONE packed composite key (partkey <<
-    # suppkey-bits | suppkey, the SyntheticAddress shape): partsupp rows
-    # are unique on the pair, so the lookup is a single-word direct
-    # table — no 4-candidate chain walk, no 4x emitted rows, no
-    # post-join suppkey equality pass.  supplycost ticks (< 2^17) ride
-    # in the slot word when the three fields fit 63 bits; beyond that
-    # (very large SF) the cost stays a slot payload array.
-    skbits = max(supp.n_rows.bit_length(), 1)
-    ckbits = part.n_rows.bit_length() + skbits
-    cost_pack = 17 if ckbits + 17 <= 63 else 0
-
     fl = PlanFilterProject()
     fl.n_proj = 6
-    fl.proj[0] = Proj(PROJ_KEYSHL, li.channel("partkey"),
-                      li.channel("suppkey"), skbits)
-    for i, name in enumerate(("suppkey", "orderkey", "quantity",
-                              "extendedprice", "discount"), start=1):
+    for i, name in enumerate(("partkey", "suppkey", "orderkey", "quantity",
+                              "extendedprice", "discount")):
         fl.proj[i] = Proj(PROJ_IDENT, li.channel(name), 0, 0)
     fl.semijoin_table = og.table()
     fl.semijoin_col = li.channel("partkey")
     f = Operator(OP_FILTER_PROJECT, fl)
     f.add_input(li)
-    gli = f.get_output_raw()  # [ck, sk, ok, qty, ep, dc]
-
-    fps = PlanFilterProject()
-    fps.n_proj = 2
-    fps.proj[0] = Proj(PROJ_KEYSHL, ps.channel("partkey"),
-                       ps.channel("suppkey"), skbits)
-    fps.proj[1] = Proj(PROJ_IDENT, ps.channel("supplycost"), 0, 0)
-    fpso = Operator(OP_FILTER_PROJECT, fps)
-    fpso.add_input(ps)
-    psraw = fpso.get_output_raw()  # [ck, cost]
+    gli = f.get_output_raw()  # [pk, sk, ok, qty, ep, dc]
 
     bp = PlanHashBuild()
-    bp.key_col = 0
+    bp.key_col = ps.channel("partkey")
     bp.semijoin_table = -1
-    bp.n_payload = 1
-    bp.payload_col[0] = 1
+    bp.n_payload = 2
+    bp.payload_col[0] = ps.channel("suppkey")
+    bp.payload_col[1] = ps.channel("supplycost")
     bp.capacity_hint = ps.n_rows
-    bp.agg_table = 1
-    bp.pack_bits = cost_pack
-    bp.fill_x10 = 13  # every probe hits (FK): size tight
     ops_ = Operator(OP_HASH_BUILD, bp)
-    ops_.add_input_raw(psraw)
+    ops_.add_input(ps)
     ops_.finish()
-    fpso.destroy()
 
     j1 = PlanLookupJoin()
     j1.table = ops_.table()
     j1.key_col = 0
     j1.mode = 0
-    j1.n_emit = 5
-    for i in range(5):
-        j1.emit_probe_cols[i] = i + 1
+    j1.n_emit = 6
+    for i in range(6):
+        j1.emit_probe_cols[i] = i
     ja = Operator(OP_LOOKUP_JOIN, j1)
     ja.add_input_raw(gli)
-    pb = ja.get_output_raw()  # [sk, ok, qty, ep, dc, cost]
+    pa = ja.get_output_raw()  # [pk, sk, ok, qty, ep, dc, ps_sk, cost]
+
+    fe = PlanFilterProject()
+    fe.n_preds = 1
+    pe = Pred(1, CMP_EQ, 0, 0.0)
+    pe.rhs_col = 6 + 1
+    fe.preds[0] = pe
+    fe.n_proj = 6
+    for i, c in enumerate((1, 2, 3, 4, 5, 7)):
+        fe.proj[i] = Proj(PROJ_IDENT, c, 0, 0)
+    f2 = Operator(OP_FILTER_PROJECT, fe)
+    f2.add_input_raw(pa)
+    pb = f2.get_output_raw()  # [sk, ok, qty, ep, dc, cost]
 
     bs = PlanHashBuild()
     bs.key_col = supp.channel("suppkey")
@@ -1538,6 +1525,7 @@ def q9(part: Page, supp: Page, orders: Page, ps: Page, li: Page):
 
     jc.destroy()
     jb.destroy()
+    f2.destroy()
     ja.destroy()
     f.destroy()
     for o in (og, ops_, os_, oo):
