@@ -516,6 +516,14 @@ pg_status pg_page_serialize(const pg_page* page /* host cols */,
  * cover LONG/INT/BYTE/INT128_ARRAY, VARIABLE_WIDTH, DICTIONARY
  * (varbin dictionaries travel as dictionaries, fixed-width ones expand
  * on read) and RLE (expanded on read). */
+/* ABI drift guard: fills out[0..n) with sizeof each public struct in
+ * the order {pg_col, pg_page, pg_pred, pg_proj, pg_agg,
+ * pg_plan_filter_project, pg_plan_hash_agg_small, pg_plan_hash_build,
+ * pg_plan_lookup_join, pg_plan_groupby, pg_plan_topn,
+ * pg_plan_partition}; returns how many it would fill.  Bindings compare
+ * against their own struct sizes at load time. */
+int32_t pg_abi_struct_sizes(int32_t* out, int32_t n);
+
 pg_status pg_page_serialize2(const pg_page* page, int32_t compress,
                              void* out, int64_t cap, int64_t* out_len);
 /* parses and verifies; fills *out with malloc-backed host columns

@@ -6035,6 +6035,27 @@ static pg_status read_block(ByteReader& r, pg_col* out, int64_t* out_n)
 
 } /* namespace */
 
+extern "C" int32_t pg_abi_struct_sizes(int32_t* out, int32_t n)
+{
+    const int32_t sz[] = {
+        (int32_t)sizeof(pg_col),
+        (int32_t)sizeof(pg_page),
+        (int32_t)sizeof(pg_pred),
+        (int32_t)sizeof(pg_proj),
+        (int32_t)sizeof(pg_agg),
+        (int32_t)sizeof(pg_plan_filter_project),
+        (int32_t)sizeof(pg_plan_hash_agg_small),
+        (int32_t)sizeof(pg_plan_hash_build),
+        (int32_t)sizeof(pg_plan_lookup_join),
+        (int32_t)sizeof(pg_plan_groupby),
+        (int32_t)sizeof(pg_plan_topn),
+        (int32_t)sizeof(pg_plan_partition),
+    };
+    const int32_t count = (int32_t)(sizeof(sz) / sizeof(sz[0]));
+    for (int32_t i = 0; i < n && i < count; i++) out[i] = sz[i];
+    return count;
+}
+
 extern "C" pg_status pg_page_serialize2(const pg_page* page,
                                         int32_t compress, void* out,
                                         int64_t cap, int64_t* out_len)

@@ -62,3 +62,21 @@ def test_no_gpu_fails_loudly(solib):
     n = C.c_int32()
     solib.pg_device_count(C.byref(n))
     assert n.value == 0
+
+
+def test_struct_sizes_match_ctypes(solib):
+    """The ctypes mirrors in presto_amd/engine.py must match the C
+    struct layouts exactly — a drift here corrupts every plan field
+    after the divergence point."""
+    from presto_amd import engine as E
+    mirrors = [E.PgCol, E.PgPage, E.Pred, E.Proj, E.Agg,
+               E.PlanFilterProject, E.PlanHashAggSmall, E.PlanHashBuild,
+               E.PlanLookupJoin, E.PlanGroupBy, E.PlanTopN,
+               E.PlanPartition]
+    solib.pg_abi_struct_sizes.restype = C.c_int32
+    n = solib.pg_abi_struct_sizes(None, 0)
+    assert n == len(mirrors)
+    sizes = (C.c_int32 * n)()
+    solib.pg_abi_struct_sizes(sizes, n)
+    for m, sz in zip(mirrors, sizes):
+        assert C.sizeof(m) == sz, (m.__name__, C.sizeof(m), sz)
