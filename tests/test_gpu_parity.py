@@ -2203,3 +2203,89 @@ def test_packed_emit_join(P):
     order = np.argsort(out["i"])
     assert np.array_equal(out["i"][order], pidx)
     assert np.array_equal(out["bv"][order], bvals[pk - 1])
+
+
+@pytest.mark.skip(reason="known issue (DESIGN.md): this exact flow "
+                  "faults on GPU while test_packed_emit_join passes; "
+                  "kept as the bisection starting point")
+def test_q9_composite_flow_repro(P):
+    """q9's reverted composite-partsupp wiring in miniature: semijoin +
+    KEYSHL filter -> raw page -> packed build from raw -> emit join."""
+    rng = np.random.RandomState(67)
+    n_part, n_supp = 2000, 100
+    skbits = n_supp.bit_length()
+    # dense green-part flag set
+    bg = P.PlanHashBuild()
+    bg.key_col = 0
+    bg.semijoin_table = -1
+    bg.capacity_hint = n_part
+    bg.key_set_only = 1
+    bg.dense_array = 1
+    og = P.Operator(P.OP_HASH_BUILD, bg)
+    green = np.unique(rng.randint(1, n_part + 1, n_part // 18))
+    og.add_input(P.Page({"pk": green.astype(np.int64)}))
+    og.finish()
+    # ps rows: 4 suppliers per part
+    pk_ps = np.repeat(np.arange(1, n_part + 1, dtype=np.int64), 4)
+    sk_ps = ((pk_ps + np.tile(np.arange(4), n_part)) % n_supp) + 1
+    cost = rng.randint(1, 100000, len(pk_ps)).astype(np.int64)
+    fps = P.PlanFilterProject()
+    fps.n_proj = 2
+    fps.proj[0] = P.Proj(P.PROJ_KEYSHL, 0, 1, skbits)
+    fps.proj[1] = P.Proj(P.PROJ_IDENT, 2, 0, 0)
+    fpso = P.Operator(P.OP_FILTER_PROJECT, fps)
+    fpso.add_input(P.Page({"pk": pk_ps, "sk": sk_ps, "cost": cost}))
+    psraw = fpso.get_output_raw()
+    bp = P.PlanHashBuild()
+    bp.key_col = 0
+    bp.semijoin_table = -1
+    bp.n_payload = 1
+    bp.payload_col[0] = 1
+    bp.capacity_hint = len(pk_ps)
+    bp.agg_table = 1
+    bp.pack_bits = 17
+    bp.fill_x10 = 13
+    ops_ = P.Operator(P.OP_HASH_BUILD, bp)
+    ops_.add_input_raw(psraw)
+    ops_.finish()
+    fpso.destroy()
+    # lineitem-like probe input through the semijoin + KEYSHL filter
+    n_li = 40_000
+    pk_li = rng.randint(1, n_part + 1, n_li).astype(np.int64)
+    sk_li = ((pk_li + rng.randint(0, 4, n_li)) % n_supp) + 1
+    v = rng.randint(1, 1000, n_li).astype(np.int64)
+    fl = P.PlanFilterProject()
+    fl.n_proj = 3
+    fl.proj[0] = P.Proj(P.PROJ_KEYSHL, 0, 1, skbits)
+    fl.proj[1] = P.Proj(P.PROJ_IDENT, 1, 0, 0)
+    fl.proj[2] = P.Proj(P.PROJ_IDENT, 2, 0, 0)
+    fl.semijoin_table = og.table()
+    fl.semijoin_col = 0
+    f = P.Operator(P.OP_FILTER_PROJECT, fl)
+    f.add_input(P.Page({"pk": pk_li, "sk": sk_li, "v": v}))
+    gli = f.get_output_raw()
+    j1 = P.PlanLookupJoin()
+    j1.table = ops_.table()
+    j1.key_col = 0
+    j1.mode = 0
+    j1.n_emit = 2
+    j1.emit_probe_cols[0] = 1
+    j1.emit_probe_cols[1] = 2
+    ja = P.Operator(P.OP_LOOKUP_JOIN, j1)
+    ja.add_input_raw(gli)
+    out = ja.get_output(["sk", "v", "cost"])
+    # numpy expectation
+    gsel = np.isin(pk_li, green)
+    ck = (pk_li << skbits) | sk_li
+    cm = {int(k): int(c) for k, c in zip((pk_ps << skbits) | sk_ps, cost)}
+    exp_v = v[gsel]
+    got_pairs = sorted(zip(out["v"].tolist(), out["cost"].tolist()))
+    exp_pairs = sorted(zip(exp_v.tolist(),
+                           [cm[int(k)] for k in ck[gsel]]))
+    assert got_pairs == exp_pairs
+    for o in (ja, f):
+        o.destroy()
+    from presto_amd.engine import lib
+    for o in (og, ops_):
+        lib().c.pg_table_destroy(o.table())
+        o.destroy()
