@@ -369,7 +369,7 @@ def test_serde_fuzz_roundtrip(L):
         pg.n_cols = ncols
         spec = []
         for c in range(ncols):
-            kind = rng.randint(0, 4)
+            kind = rng.randint(0, 5)
             nulls = None
             if rng.randint(0, 2) and n:
                 nulls = (rng.randint(0, 4, n) == 0).astype(np.uint8)
@@ -382,14 +382,27 @@ def test_serde_fuzz_roundtrip(L):
             elif kind == 2:
                 a = rng.randint(0, 256, n).astype(np.uint8)
                 tag = 0
-            else:
+            elif kind == 3:
                 a = rng.randint(-2**60, 2**60, 2 * n).astype(np.int64)
                 tag = 5  # i128 (lo, hi) pairs
+            else:
+                # VARBIN: random lengths incl. empty strings
+                lens = rng.randint(0, 12, n)
+                offs = np.zeros(n + 1, np.int32)
+                np.cumsum(lens, out=offs[1:])
+                data = rng.randint(0, 256,
+                                   int(offs[-1])).astype(np.uint8)
+                a = (data, offs)
+                tag = 4
             keep.append((a, nulls))
             spec.append((tag, a, nulls))
             pg.cols[c].tag = tag
             pg.cols[c].on_device = 0
-            pg.cols[c].data = a.ctypes.data
+            if tag == 4:
+                pg.cols[c].data = a[0].ctypes.data
+                pg.cols[c].offsets = a[1].ctypes.data
+            else:
+                pg.cols[c].data = a.ctypes.data
             pg.cols[c].null_mask = (nulls.ctypes.data
                                     if nulls is not None else None)
         wire = _serialize2(L, pg, int(rng.randint(0, 2)))
@@ -398,6 +411,27 @@ def test_serde_fuzz_roundtrip(L):
         assert st == 0, L.pg_last_error()
         assert out.n_rows == n and out.n_cols == ncols
         for c, (tag, a, nulls) in enumerate(spec):
+            if tag == 4:
+                data, offs = a
+                roffs = np.ctypeslib.as_array(
+                    C.cast(C.c_void_p(out.cols[c].offsets),
+                           C.POINTER(C.c_int32)),
+                    shape=(n + 1,)).copy() if n else offs[:1]
+                if n:
+                    m = (nulls.astype(bool) if nulls is not None
+                         else np.zeros(n, bool))
+                    for i in range(n):
+                        if m[i]:
+                            continue
+                        exp = data[offs[i]:offs[i + 1]].tobytes()
+                        got_b = bytes(np.ctypeslib.as_array(
+                            C.cast(out.cols[c].data,
+                                   C.POINTER(C.c_uint8)),
+                            shape=(max(int(roffs[-1]), 1),))
+                            [roffs[i]:roffs[i + 1]]) \
+                            if roffs[-1] else b""
+                        assert got_b == exp, (trial, c, i)
+                continue
             esz = {0: 1, 1: 4, 2: 8, 5: 16}[tag]
             cnt = n * (2 if tag == 5 else 1)
             dt = {0: np.uint8, 1: np.int32, 2: np.int64,
@@ -416,5 +450,4 @@ def test_serde_fuzz_roundtrip(L):
                         assert np.array_equal(got[~m2], a[~m2])
                     else:
                         assert np.array_equal(got[~m], a[~m])
-            L and None
         L.pg_page_free(C.byref(out))
