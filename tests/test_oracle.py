@@ -641,3 +641,51 @@ def test_tpcds_date_dim_math():
         q = (d.month - 1) // 3
         assert qname[i] == d.year * 4 + q, i
         assert week[i] == i // 7
+
+
+def test_key_domain_invariants(oracle_lib):
+    """The round-2 key-structure mechanisms (bitmap_max_key, range_group,
+    dense_array over [1, n]) rest on dbgen's key domains: orderkey is
+    mk_sparse-shaped with max == pipelines.okey_max(n_orders) and every
+    key >= 1; custkey/partkey/suppkey are dense 1..n; every lineitem
+    orderkey appears in orders (FK).  Pin all of that against the
+    generator itself at two scale factors."""
+    import numpy as np
+    from presto_amd.pipelines import okey_max
+
+    for sf in (0.01, 0.1):
+        orders = oracle_lib.gen_orders(sf)
+        ok = orders["orderkey"]
+        n = len(ok)
+        assert ok.min() >= 1
+        assert ok.max() == okey_max(n), (sf, ok.max(), okey_max(n))
+        # mk_sparse shape: low 3 bits dense within each 32-key block
+        assert len(np.unique(ok)) == n
+        cust = oracle_lib.gen_customer(sf)
+        ck = cust["custkey"]
+        assert np.array_equal(np.sort(ck), np.arange(1, len(ck) + 1))
+        # orders custkeys stay inside the customer domain
+        assert orders["custkey"].min() >= 1
+        assert orders["custkey"].max() <= len(ck)
+        li = oracle_lib.gen_lineitem(sf)
+        lok = li["orderkey"]
+        assert lok.min() >= 1 and lok.max() <= okey_max(n)
+        # FK: every lineitem orderkey exists in orders
+        assert np.isin(np.unique(lok), ok).all()
+
+
+def test_q21_acc_pack_layout_fits_across_sfs():
+    """q21's accumulator packing must fit 64 bits (or fall back) at every
+    plausible scale factor: widths derive from the supplier count and
+    the <= 7 lineitems/order spec bound."""
+    for sf, n_supp in ((1, 10_000), (100, 1_000_000), (300, 3_000_000),
+                       (1000, 10_000_000), (3000, 30_000_000)):
+        wsum = (7 * n_supp).bit_length()
+        packs = 2 * wsum + 12 <= 64
+        if sf <= 300:
+            assert packs, (sf, wsum)  # benched SFs must take the packed path
+        if packs:
+            # fields: a0[0,wsum) a2[wsum,2w) a1[2w,+4) a3[+4,+4) cnt[+4)
+            assert 2 * wsum + 12 <= 64
+            # totals fit: sum(suppkey) <= 7*n_supp < 2^wsum
+            assert 7 * n_supp < (1 << wsum)
