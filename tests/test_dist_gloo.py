@@ -73,3 +73,23 @@ def test_exchange_gloo_world2():
     ctx = torch.multiprocessing.start_processes(
         _worker, args=(2, port), nprocs=2, join=True,
         start_method="spawn")
+
+
+def test_bench_dist_call_signatures_bind():
+    """bench.py's N>1 steps call the dist entry points with these exact
+    argument lists; a drift here would only surface on the driver's
+    8-GPU window, so bind them against the signatures on CPU."""
+    import inspect
+    from presto_amd import dist as pdist
+
+    s3 = inspect.signature(pdist.q3_distributed)
+    s3.bind("cust", "ord", "li", 8, 0, "cuda:0", mode="dec",
+            okey_bound=600_000_007)
+    s5 = inspect.signature(pdist.q5_distributed)
+    s5.bind("cust", "ord", "supp", "li", 8, 0, "cuda:0",
+            okey_bound=600_000_007)
+    # and the single-GPU pipeline entry points bench/measure call
+    from presto_amd import pipelines as pl
+    inspect.signature(pl.q3).bind("c", "o", "l", mode="dec")
+    inspect.signature(pl.q5).bind("c", "o", "s", "l")
+    inspect.signature(pl.q4).bind("orders", "li_dates")
